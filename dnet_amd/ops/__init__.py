@@ -1,0 +1,111 @@
+"""dnet_amd.ops — gfx950 HIP kernels with CPU reference fallback.
+
+GPU tensors dispatch to the in-tree ``_C.so`` extension (hand-written
+CDNA4/HIP kernels); if the extension is missing on a GPU machine the ops
+FAIL LOUDLY — there is no silent eager fallback on the GPU path. CPU
+tensors use the fp32 reference implementations (tests / no-GPU machines).
+"""
+from __future__ import annotations
+
+import importlib.util
+from pathlib import Path
+
+import torch
+
+from . import reference as ref
+
+_SO = Path(__file__).resolve().parent / "_C.so"
+_C = None
+_load_error: Exception | None = None
+
+
+def _try_load():
+    global _C, _load_error
+    if _C is not None or _load_error is not None:
+        return _C
+    try:
+        spec = importlib.util.spec_from_file_location("dnet_amd.ops._C", _SO)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _C = mod
+    except Exception as e:  # pragma: no cover
+        _load_error = e
+    return _C
+
+
+def has_native() -> bool:
+    return _try_load() is not None
+
+
+def _native():
+    c = _try_load()
+    if c is None:
+        raise RuntimeError(
+            f"dnet_amd native extension not available ({_SO}): {_load_error}. "
+            "Run `python -m dnet_amd.ops.build` — GPU tensors never fall back "
+            "to eager.")
+    return c
+
+
+def rmsnorm(x: torch.Tensor, residual: torch.Tensor | None, w: torch.Tensor,
+            eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        y = torch.empty_like(x)
+        _native().rmsnorm(x, residual, w, y, eps)
+        return y
+    return ref.rmsnorm(x, residual, w, eps)
+
+
+def gemv_bf16(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
+        _native().gemv_bf16(x, w, out)
+        return out
+    return ref.gemv_bf16(x, w)
+
+
+def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+              group: int) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
+        _native().gemv_int8(x, w, scales, out, group)
+        return out
+    return ref.gemv_int8(x, w, scales, group)
+
+
+def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Tensor:
+    if w.is_cuda:
+        out = torch.empty(w.shape, dtype=torch.bfloat16, device=w.device)
+        _native().dequant_int8(w, scales, out, group)
+        return out
+    return ref.dequant_int8(w, scales, group)
+
+
+def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+                pos: torch.Tensor, scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().attn_decode(q, kcache, vcache, pos, out, scale)
+        return out
+    return ref.attn_decode(q, kcache, vcache, pos, scale)
+
+
+def rope_append(q, k, v, kcache, vcache, pos, cos, sin) -> None:
+    if q.is_cuda:
+        _native().rope_append(q, k, v, kcache, vcache, pos, cos, sin)
+        return
+    ref.rope_append(q, k, v, kcache, vcache, pos, cos, sin)
+
+
+def swiglu(gu: torch.Tensor) -> torch.Tensor:
+    if gu.is_cuda:
+        i = gu.shape[-1] // 2
+        y = torch.empty(*gu.shape[:-1], i, dtype=gu.dtype, device=gu.device)
+        _native().swiglu(gu, y)
+        return y
+    return ref.swiglu(gu)
+
+
+quantize_int8 = ref.quantize_int8
+rope_tables = ref.rope_tables
+rope_apply = ref.rope_apply

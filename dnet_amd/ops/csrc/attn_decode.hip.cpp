@@ -1,0 +1,178 @@
+// Single-token (decode) attention against the KV cache, GQA-aware.
+//
+//   out[b,h,:] = softmax(q[b,h,:] . K[b,kv(h),:len,:]^T * scale) @ V[b,kv(h),:len,:]
+//
+// One workgroup per (batch, kv_head): the K/V read — the memory bound — is
+// staged through LDS once and shared by all query heads of the group and all
+// 4 waves. Online softmax (flash-decode style) over 64-position chunks;
+// per-lane layout: score phase = one seq position per lane, accumulate
+// phase = D/64 output dims per lane with the softmax weight broadcast by
+// shuffle. Sequence length comes from a device tensor (`pos`) so the kernel
+// is hipGraph-replayable across decode steps without recapture.
+//
+// Replaces the reference's MLX scaled_dot_product_attention decode path
+// (reference: src/dnet/core/models/llama.py apply_single_layer).
+#include "common.h"
+
+namespace dnet {
+
+// Row pitch in bf16 elements for a D-wide LDS tile such that the per-lane
+// column reads are bank-conflict-free (pitch in dwords must be ≡ 2 mod 4
+// and 8B-aligned; 66 dwords for D=128 → 2l mod 64 distinct per 32-lane group).
+__device__ __host__ __forceinline__ constexpr int lds_pitch(int D) {
+  return D + 4;  // bf16 elems; D=128 -> 264 B = 66 dwords; D=64 -> 136 B = 34 dwords
+}
+
+__device__ __forceinline__ float wave_allreduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+__device__ __forceinline__ float wave_allreduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+constexpr int kChunk = 64;
+
+template <int D>
+__global__ void attn_decode_kernel(const short* __restrict__ q,
+                                   const short* __restrict__ kc,
+                                   const short* __restrict__ vc,
+                                   const int* __restrict__ pos,
+                                   short* __restrict__ out, const int Hq,
+                                   const int Hkv, const int Smax,
+                                   const float scale) {
+  constexpr int P = lds_pitch(D);
+  constexpr int DPL = D / kWave;  // output dims per lane (1 or 2)
+  const int b = blockIdx.x / Hkv;
+  const int hkv = blockIdx.x % Hkv;
+  const int G = Hq / Hkv;  // query heads per kv head
+  const int wid = threadIdx.x / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = reinterpret_cast<short*>(smem);               // [kChunk][P]
+  short* v_lds = k_lds + kChunk * P;                           // [kChunk][P]
+  short* q_lds = v_lds + kChunk * P;                           // [G][D]
+
+  const int len = pos[b];
+  const int64_t kvbase = ((int64_t)b * Hkv + hkv) * Smax * D;
+
+  // Stage this group's q rows.
+  for (int i = threadIdx.x; i < G * D / 8; i += blockDim.x) {
+    const int g = i / (D / 8);
+    reinterpret_cast<short8*>(q_lds)[i] =
+        reinterpret_cast<const short8*>(q + ((int64_t)b * Hq + hkv * G + g) * D)[i % (D / 8)];
+  }
+
+  // Per-wave head list: g = wid, wid+4, ... (max 4 heads per wave).
+  float m[4], l[4], acc[4][DPL];
+  const int nh = (G - wid + 3) / 4;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    m[i] = -1e30f;
+    l[i] = 0.f;
+#pragma unroll
+    for (int j = 0; j < DPL; ++j) acc[i][j] = 0.f;
+  }
+  __syncthreads();
+
+  const int nchunks = (len + kChunk - 1) / kChunk;
+  for (int c = 0; c < nchunks; ++c) {
+    const int s0 = c * kChunk;
+    const int valid = min(kChunk, len - s0);
+    // Stage K and V chunk: 16 B per thread per iteration, coalesced rows.
+    for (int i = threadIdx.x; i < kChunk * D / 8; i += blockDim.x) {
+      const int row = i / (D / 8);
+      const int col = (i % (D / 8)) * 8;
+      short8 kv8, vv8;
+      if (row < valid) {
+        kv8 = *reinterpret_cast<const short8*>(kc + kvbase + (int64_t)(s0 + row) * D + col);
+        vv8 = *reinterpret_cast<const short8*>(vc + kvbase + (int64_t)(s0 + row) * D + col);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) { kv8.x[j] = 0; vv8.x[j] = 0; }
+      }
+      *reinterpret_cast<short8*>(&k_lds[row * P + col]) = kv8;
+      *reinterpret_cast<short8*>(&v_lds[row * P + col]) = vv8;
+    }
+    __syncthreads();
+
+    for (int hi = 0; hi < nh; ++hi) {
+      const int g = wid + hi * 4;
+      // Score for position s = lane.
+      float s_val = 0.f;
+      const short* krow = &k_lds[lane * P];
+      const short* qrow = &q_lds[g * D];
+      for (int d = 0; d < D; d += 4) {
+        const short4v kv = *reinterpret_cast<const short4v*>(&krow[d]);
+        const short4v qv = *reinterpret_cast<const short4v*>(&qrow[d]);  // broadcast
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          s_val = fmaf(bits2f(kv.x[j]), bits2f(qv.x[j]), s_val);
+      }
+      s_val *= scale;
+      if (lane >= valid) s_val = -1e30f;
+      // Online softmax update.
+      const float cmax = wave_allreduce_max(s_val);
+      const float mn = fmaxf(m[hi], cmax);
+      const float alpha = __expf(m[hi] - mn);
+      const float p = (lane < valid) ? __expf(s_val - mn) : 0.f;
+      l[hi] = l[hi] * alpha + wave_allreduce_sum(p);
+      m[hi] = mn;
+#pragma unroll
+      for (int j = 0; j < DPL; ++j) acc[hi][j] *= alpha;
+      // Accumulate P @ V: lane owns dims d = DPL*lane + j.
+      for (int s = 0; s < valid; ++s) {
+        const float ps = __shfl(p, s, 64);
+        const short* vrow = &v_lds[s * P + DPL * lane];
+#pragma unroll
+        for (int j = 0; j < DPL; ++j)
+          acc[hi][j] = fmaf(ps, bits2f(vrow[j]), acc[hi][j]);
+      }
+    }
+    __syncthreads();
+  }
+
+  for (int hi = 0; hi < nh; ++hi) {
+    const int g = wid + hi * 4;
+    const float inv = (l[hi] > 0.f) ? 1.f / l[hi] : 0.f;
+    short* orow = out + ((int64_t)b * Hq + hkv * G + g) * D + DPL * lane;
+#pragma unroll
+    for (int j = 0; j < DPL; ++j) orow[j] = f2bits(acc[hi][j] * inv);
+  }
+}
+
+void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
+                 torch::Tensor pos, torch::Tensor out, double scale) {
+  const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int64_t Hkv = kcache.size(1), Smax = kcache.size(2);
+  DNET_CHECK(kcache.size(0) == B && kcache.size(3) == D, "kcache shape");
+  DNET_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 16, "GQA group <= 16");
+  DNET_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  DNET_CHECK(pos.dtype() == torch::kInt32, "pos int32");
+  DNET_CHECK(q.is_contiguous() && kcache.is_contiguous() && vcache.is_contiguous() &&
+                 out.is_contiguous(), "contig");
+  auto stream = current_stream();
+  const int G = (int)(Hq / Hkv);
+  const int P = (int)D + 4;
+  const size_t lds = (2 * kChunk * P + G * D) * sizeof(short);
+  const dim3 grid((unsigned)(B * Hkv));
+  if (D == 128) {
+    hipLaunchKernelGGL((attn_decode_kernel<128>), grid, dim3(256), lds, stream,
+                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                       (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
+                       (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
+                       (float)scale);
+  } else {
+    hipLaunchKernelGGL((attn_decode_kernel<64>), grid, dim3(256), lds, stream,
+                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
+                       (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
+                       (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
+                       (float)scale);
+  }
+}
+
+}  // namespace dnet

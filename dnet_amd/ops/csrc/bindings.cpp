@@ -1,0 +1,29 @@
+// Python bindings for the dnet_amd gfx950 kernel extension.
+#include <torch/extension.h>
+
+namespace dnet {
+void rmsnorm(torch::Tensor x, c10::optional<torch::Tensor> residual,
+             torch::Tensor w, torch::Tensor y, double eps);
+void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out);
+void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
+               torch::Tensor out, int64_t group);
+void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
+                 torch::Tensor pos, torch::Tensor out, double scale);
+void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                 torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
+                 torch::Tensor cos_table, torch::Tensor sin_table);
+void swiglu(torch::Tensor gu, torch::Tensor y);
+void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
+                  int64_t group);
+}  // namespace dnet
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dnet_amd gfx950 (MI355X/CDNA4) kernels";
+  m.def("rmsnorm", &dnet::rmsnorm, "fused RMSNorm (+residual)");
+  m.def("gemv_bf16", &dnet::gemv_bf16, "bf16 decode GEMV");
+  m.def("gemv_int8", &dnet::gemv_int8, "grouped-int8 W8A16 decode GEMV");
+  m.def("attn_decode", &dnet::attn_decode, "GQA decode attention vs KV cache");
+  m.def("rope_append", &dnet::rope_append, "fused RoPE + KV append (decode)");
+  m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
+  m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");
+}

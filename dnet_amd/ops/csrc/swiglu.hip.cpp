@@ -1,0 +1,85 @@
+// Fused SwiGLU: y = silu(gate) * up, reading the fused [T, 2I] gate/up
+// projection output (gate = [:, :I], up = [:, I:]).  Memory-bound
+// elementwise; vectorized 16 B loads, grid-stride (CDNA4 guide G13).
+//
+// Reference equivalent: mlx_lm MLP silu(gate)*up (reference:
+// src/dnet/core/models/llama.py TransformerBlock MLP).
+#include "common.h"
+
+namespace dnet {
+
+__global__ void swiglu_kernel(const short* __restrict__ gu,
+                              short* __restrict__ y, const int64_t I,
+                              const int64_t T) {
+  const int64_t nvec = T * I / 8;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < nvec;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = idx / (I / 8);
+    const int64_t col = idx % (I / 8);
+    const short8 g = reinterpret_cast<const short8*>(gu + row * 2 * I)[col];
+    const short8 u = reinterpret_cast<const short8*>(gu + row * 2 * I + I)[col];
+    short8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = bits2f(g.x[j]);
+      const float uf = bits2f(u.x[j]);
+      const float silu = gf / (1.f + __expf(-gf));
+      o.x[j] = f2bits(silu * uf);
+    }
+    reinterpret_cast<short8*>(y + row * I)[col] = o;
+  }
+}
+
+void swiglu(torch::Tensor gu, torch::Tensor y) {
+  const int64_t I = y.size(-1);
+  const int64_t T = y.numel() / I;
+  DNET_CHECK(gu.size(-1) == 2 * I, "gu last dim must be 2*I");
+  DNET_CHECK(I % 8 == 0, "I % 8");
+  DNET_CHECK(gu.is_contiguous() && y.is_contiguous(), "contig");
+  auto stream = current_stream();
+  const int64_t nvec = T * I / 8;
+  const int grid = (int)std::min<int64_t>((nvec + 255) / 256, 2048);
+  hipLaunchKernelGGL(swiglu_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const short*)gu.data_ptr(), (short*)y.data_ptr(), I, T);
+}
+
+__global__ void dequant_int8_kernel(const int8_t* __restrict__ w,
+                                    const short* __restrict__ scales,
+                                    short* __restrict__ out, const int64_t K,
+                                    const int G, const int64_t N) {
+  const int64_t nvec = N * K / 16;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < nvec;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t n = idx / (K / 16);
+    const int64_t k0 = (idx % (K / 16)) * 16;
+    const int4 wv = reinterpret_cast<const int4*>(w + n * K)[k0 / 16];
+    const int8_t* q = reinterpret_cast<const int8_t*>(&wv);
+    const float s = bits2f(scales[n * (K / G) + k0 / G]);
+    short8 o0, o1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      o0.x[j] = f2bits((float)q[j] * s);
+      o1.x[j] = f2bits((float)q[8 + j] * s);
+    }
+    reinterpret_cast<short8*>(out + n * K)[k0 / 8] = o0;
+    reinterpret_cast<short8*>(out + n * K)[k0 / 8 + 1] = o1;
+  }
+}
+
+// Dequantize grouped-int8 weights to bf16 (prefill path: the dequantized
+// tile feeds a hipBLASLt GEMM via torch.matmul).
+void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
+                  int64_t group) {
+  const int64_t N = w.size(0), K = w.size(1);
+  DNET_CHECK(K % 16 == 0 && group % 16 == 0, "K align");
+  DNET_CHECK(out.size(0) == N && out.size(1) == K, "out shape");
+  DNET_CHECK(w.is_contiguous() && scales.is_contiguous() && out.is_contiguous(), "contig");
+  auto stream = current_stream();
+  const int64_t nvec = N * K / 16;
+  const int grid = (int)std::min<int64_t>((nvec + 255) / 256, 2048);
+  hipLaunchKernelGGL(dequant_int8_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const int8_t*)w.data_ptr(), (const short*)scales.data_ptr(),
+                     (short*)out.data_ptr(), K, (int)group, N);
+}
+
+}  // namespace dnet

@@ -1,0 +1,134 @@
+"""Plain-PyTorch fp32 reference implementations of every dnet_amd HIP kernel.
+
+These serve two purposes:
+  * golden references for the GPU numerics tests (kernel vs fp32 eager), and
+  * the CPU execution path, so the whole engine (ring executor, policies,
+    API server) runs and is testable on machines with no GPU.
+
+They intentionally mirror the kernel contracts exactly (shapes, in-place
+semantics, bf16 rounding points).
+"""
+from __future__ import annotations
+
+import torch
+
+
+def rmsnorm(x: torch.Tensor, residual: torch.Tensor | None, w: torch.Tensor,
+            eps: float) -> torch.Tensor:
+    """y = rms_norm(x or x+residual) * w; residual (if given) updated in place."""
+    if residual is not None:
+        h = (x.float() + residual.float()).to(x.dtype)
+        residual.copy_(h)
+        src = h.float()
+    else:
+        src = x.float()
+    scale = torch.rsqrt(src.pow(2).mean(-1, keepdim=True) + eps)
+    return (src * scale * w.float()).to(x.dtype)
+
+
+def gemv_bf16(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    return (x.float() @ w.float().t()).to(x.dtype)
+
+
+def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Tensor:
+    n, k = w.shape
+    wf = w.float().view(n, k // group, group)
+    return (wf * scales.float().unsqueeze(-1)).view(n, k).to(torch.bfloat16)
+
+
+def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+              group: int) -> torch.Tensor:
+    wd = dequant_int8(w, scales, group)
+    return (x.float() @ wd.float().t()).to(x.dtype)
+
+
+def quantize_int8(w: torch.Tensor, group: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Symmetric per-group int8 quantization along the K (last) dim."""
+    n, k = w.shape
+    assert k % group == 0
+    wf = w.float().view(n, k // group, group)
+    amax = wf.abs().amax(dim=-1).clamp_min(1e-8)
+    scales = (amax / 127.0).to(torch.bfloat16)
+    q = torch.round(wf / scales.float().unsqueeze(-1)).clamp(-127, 127).to(torch.int8)
+    return q.view(n, k), scales
+
+
+def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+                pos: torch.Tensor, scale: float) -> torch.Tensor:
+    """out[b,h] = softmax(q . K^T * scale) @ V over len=pos[b] positions."""
+    B, Hq, D = q.shape
+    Hkv = kcache.shape[1]
+    G = Hq // Hkv
+    out = torch.zeros_like(q)
+    for b in range(B):
+        ln = int(pos[b])
+        if ln == 0:
+            continue
+        k = kcache[b, :, :ln].float()          # [Hkv, ln, D]
+        v = vcache[b, :, :ln].float()
+        qq = q[b].float().view(Hkv, G, D)      # [Hkv, G, D]
+        s = torch.einsum("hgd,hld->hgl", qq, k) * scale
+        p = torch.softmax(s, dim=-1)
+        o = torch.einsum("hgl,hld->hgd", p, v)
+        out[b] = o.reshape(Hq, D).to(q.dtype)
+    return out
+
+
+def rope_tables(smax: int, d: int, theta: float,
+                device=None, scaling: dict | None = None) -> tuple[torch.Tensor, torch.Tensor]:
+    """Precompute [Smax, D/2] f32 cos/sin tables (neox convention).
+
+    ``scaling`` supports the llama3 rope-scaling dict
+    (factor / low_freq_factor / high_freq_factor / original_max_position_embeddings).
+    """
+    half = d // 2
+    inv = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float32) / half))
+    if scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
+        factor = scaling["factor"]
+        lo = scaling.get("low_freq_factor", 1.0)
+        hi = scaling.get("high_freq_factor", 4.0)
+        orig = scaling.get("original_max_position_embeddings", 8192)
+        wavelen = 2 * torch.pi / inv
+        low_wl = orig / lo
+        high_wl = orig / hi
+        smooth = (orig / wavelen - lo) / (hi - lo)
+        scaled = torch.where(wavelen > low_wl, inv / factor, inv)
+        mid = (1 - smooth) * inv / factor + smooth * inv
+        use_mid = (wavelen <= low_wl) & (wavelen >= high_wl)
+        inv = torch.where(use_mid, mid, scaled)
+    t = torch.arange(smax, dtype=torch.float32)
+    freqs = torch.outer(t, inv)
+    return freqs.cos().to(device), freqs.sin().to(device)
+
+
+def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+               positions: torch.Tensor) -> torch.Tensor:
+    """Neox rotate-half rope. x: [..., T, H, D] or [B, H, D] with positions [T]/[B]."""
+    half = x.shape[-1] // 2
+    c = cos[positions].float()  # [T, half]
+    s = sin[positions].float()
+    while c.dim() < x.dim():
+        c = c.unsqueeze(-2)
+        s = s.unsqueeze(-2)
+    x1 = x[..., :half].float()
+    x2 = x[..., half:].float()
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
+
+
+def rope_append(q, k, v, kcache, vcache, pos, cos, sin):
+    """Decode-step fused rope+append reference (in-place on q, k and caches)."""
+    B = q.shape[0]
+    positions = pos.long()
+    q.copy_(rope_apply(q, cos, sin, positions))
+    k.copy_(rope_apply(k, cos, sin, positions))
+    for b in range(B):
+        p = int(pos[b])
+        kcache[b, :, p] = k[b]
+        vcache[b, :, p] = v[b]
+
+
+def swiglu(gu: torch.Tensor) -> torch.Tensor:
+    i = gu.shape[-1] // 2
+    g = gu[..., :i].float()
+    u = gu[..., i:].float()
+    return (torch.nn.functional.silu(g) * u).to(gu.dtype)

@@ -1,0 +1,52 @@
+"""CPU tests of the reference op implementations (no GPU needed)."""
+import torch
+
+import dnet_amd.ops as ops
+from dnet_amd.ops import reference as ref
+
+
+def test_quantize_roundtrip():
+    torch.manual_seed(0)
+    w = torch.randn(64, 256, dtype=torch.bfloat16)
+    q, s = ops.quantize_int8(w, 64)
+    assert q.dtype == torch.int8 and s.shape == (64, 4)
+    wd = ref.dequant_int8(q, s, 64)
+    assert (wd.float() - w.float()).abs().max().item() < 0.05
+
+
+def test_rmsnorm_residual_inplace():
+    x = torch.randn(4, 64, dtype=torch.bfloat16)
+    r = torch.randn(4, 64, dtype=torch.bfloat16)
+    r0 = r.clone()
+    y = ops.rmsnorm(x, r, torch.ones(64, dtype=torch.bfloat16), 1e-6)
+    assert torch.allclose(r.float(), (x.float() + r0.float()), atol=2e-2)
+    assert y.shape == x.shape
+
+
+def test_attn_decode_matches_sdpa():
+    torch.manual_seed(1)
+    B, Hq, Hkv, D, S = 2, 4, 2, 64, 32
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, 128, D, dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, 128, D, dtype=torch.bfloat16)
+    pos = torch.tensor([S, 7], dtype=torch.int32)
+    out = ref.attn_decode(q, kc, vc, pos, D ** -0.5)
+    # cross-check against torch sdpa for batch 0
+    G = Hq // Hkv
+    qe = q[0].float().view(Hkv, G, D)
+    o = torch.nn.functional.scaled_dot_product_attention(
+        qe, kc[0, :, :S].float(), vc[0, :, :S].float())
+    assert torch.allclose(out[0].float(), o.reshape(Hq, D), atol=2e-2)
+
+
+def test_rope_tables_llama3_scaling():
+    cos, sin = ops.rope_tables(64, 128, 500000.0, scaling={
+        "rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+        "high_freq_factor": 4.0, "original_max_position_embeddings": 8192})
+    assert cos.shape == (64, 64) and torch.isfinite(cos).all()
+
+
+def test_swiglu_shapes():
+    gu = torch.randn(3, 5, 2 * 32, dtype=torch.bfloat16)
+    y = ops.swiglu(gu)
+    assert y.shape == (3, 5, 32)

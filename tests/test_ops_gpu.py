@@ -1,0 +1,117 @@
+"""GPU numerics tests: every HIP kernel vs its plain-PyTorch fp32 reference.
+
+Mirrors the reference repo's kernel test strategy (golden tests on tiny
+tensors; see SURVEY.md §4 carry-over note).
+"""
+import pytest
+import torch
+
+import dnet_amd.ops as ops
+from dnet_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+def test_native_loaded():
+    # On a GPU box the in-tree extension must be what runs — no eager fallback.
+    assert ops.has_native(), "native _C.so must load on the GPU box"
+
+
+def test_rmsnorm():
+    torch.manual_seed(0)
+    x = torch.randn(33, 5120, dtype=torch.bfloat16, device=_dev())
+    w = torch.randn(5120, dtype=torch.bfloat16, device=_dev())
+    y = ops.rmsnorm(x.clone(), None, w, 1e-6)
+    y_ref = ref.rmsnorm(x.cpu().clone(), None, w.cpu(), 1e-6)
+    assert torch.allclose(y.float().cpu(), y_ref.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_rmsnorm_residual():
+    torch.manual_seed(1)
+    x = torch.randn(9, 1024, dtype=torch.bfloat16, device=_dev())
+    r = torch.randn(9, 1024, dtype=torch.bfloat16, device=_dev())
+    w = torch.randn(1024, dtype=torch.bfloat16, device=_dev())
+    r_ref = r.cpu().clone()
+    y_ref = ref.rmsnorm(x.cpu().clone(), r_ref, w.cpu(), 1e-6)
+    y = ops.rmsnorm(x, r, w, 1e-6)
+    assert torch.allclose(y.float().cpu(), y_ref.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(r.float().cpu(), r_ref.float(), atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("m", [1, 2, 7, 16, 33])
+def test_gemv_bf16(m):
+    torch.manual_seed(2)
+    K, N = 2048, 1536
+    x = torch.randn(m, K, dtype=torch.bfloat16, device=_dev())
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=_dev()) / 30
+    out = ops.gemv_bf16(x, w)
+    out_ref = ref.gemv_bf16(x.cpu(), w.cpu())
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=5e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("m,group", [(1, 64), (4, 128), (13, 64)])
+def test_gemv_int8(m, group):
+    torch.manual_seed(3)
+    K, N = 1024, 768
+    x = torch.randn(m, K, dtype=torch.bfloat16, device=_dev())
+    wf = torch.randn(N, K, dtype=torch.bfloat16, device=_dev()) / 30
+    q, scales = ops.quantize_int8(wf, group)
+    out = ops.gemv_int8(x, q, scales, group)
+    out_ref = ref.gemv_int8(x.cpu(), q.cpu(), scales.cpu(), group)
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=5e-2, rtol=3e-2)
+
+
+def test_dequant_int8():
+    torch.manual_seed(4)
+    wf = torch.randn(256, 512, dtype=torch.bfloat16, device=_dev())
+    q, scales = ops.quantize_int8(wf, 64)
+    wd = ops.dequant_int8(q, scales, 64)
+    wd_ref = ref.dequant_int8(q.cpu(), scales.cpu(), 64)
+    assert torch.allclose(wd.float().cpu(), wd_ref.float(), atol=2e-2, rtol=2e-2)
+    # quantization round trip stays close to the original
+    assert (wd.float() - wf.float()).abs().max().item() < 0.05
+
+
+@pytest.mark.parametrize("d,hq,hkv", [(128, 10, 2), (64, 8, 8), (128, 40, 8)])
+def test_attn_decode(d, hq, hkv):
+    torch.manual_seed(5)
+    B, Smax = 3, 256
+    q = torch.randn(B, hq, d, dtype=torch.bfloat16, device=_dev())
+    kc = torch.randn(B, hkv, Smax, d, dtype=torch.bfloat16, device=_dev())
+    vc = torch.randn(B, hkv, Smax, d, dtype=torch.bfloat16, device=_dev())
+    pos = torch.tensor([5, 200, 64], dtype=torch.int32, device=_dev())[:B]
+    scale = d ** -0.5
+    out = ops.attn_decode(q, kc, vc, pos, scale)
+    out_ref = ref.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), pos.cpu(), scale)
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_rope_append():
+    torch.manual_seed(6)
+    B, Hq, Hkv, D, Smax = 4, 8, 2, 128, 128
+    cos, sin = ops.rope_tables(Smax, D, 10000.0)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=_dev())
+    k = torch.randn(B, Hkv, D, dtype=torch.bfloat16, device=_dev())
+    v = torch.randn(B, Hkv, D, dtype=torch.bfloat16, device=_dev())
+    kc = torch.zeros(B, Hkv, Smax, D, dtype=torch.bfloat16, device=_dev())
+    vc = torch.zeros_like(kc)
+    pos = torch.tensor([0, 3, 77, 127], dtype=torch.int32, device=_dev())
+    qr, kr = q.cpu().clone(), k.cpu().clone()
+    kcr, vcr = kc.cpu().clone(), vc.cpu().clone()
+    ref.rope_append(qr, kr, v.cpu().clone(), kcr, vcr, pos.cpu(), cos, sin)
+    ops.rope_append(q, k, v, kc, vc, pos, cos.to(_dev()), sin.to(_dev()))
+    assert torch.allclose(q.float().cpu(), qr.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(kc.float().cpu(), kcr.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(vc.float().cpu(), vcr.float(), atol=1e-3)
+
+
+def test_swiglu():
+    torch.manual_seed(7)
+    gu = torch.randn(17, 2 * 1024, dtype=torch.bfloat16, device=_dev())
+    y = ops.swiglu(gu)
+    y_ref = ref.swiglu(gu.cpu())
+    assert torch.allclose(y.float().cpu(), y_ref.float(), atol=2e-2, rtol=2e-2)
