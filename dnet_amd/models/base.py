@@ -1,0 +1,377 @@
+"""Ring-model base: layer weights, fused linears, decode/prefill windows.
+
+The MI355X counterpart of the reference's BaseRingModel
+(reference: src/dnet/core/models/base.py) — but instead of nn.Modules it
+manages raw weight tensors per layer so the windowed-residency weight cache
+can bind/evict them, runs decode through the hand-written CDNA4 kernels
+(fused QKV GEMV, rope+append, GQA decode attention, fused SwiGLU) and
+prefill through hipBLASLt GEMMs (torch.matmul) with chunked causal attention.
+
+Weight-name mapping accepts HF-style absolute names
+(``model.layers.N.self_attn.q_proj.weight`` or ``layers.N...``) and keeps
+only locally-assigned layers (reference: base.py load_weights abs->local
+remap).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional, Sequence
+
+import torch
+
+from .. import ops
+from .config import ModelConfig, QuantConfig
+
+# Decode GEMV handles M<=16 tiles; beyond that the dequant+GEMM path wins.
+GEMV_MAX_M = 16
+
+
+class Linear:
+    """A weight-only-quantizable linear: y = x @ W^T + b.
+
+    Small-M (decode) goes through the fused HIP GEMV; large-M (prefill)
+    through hipBLASLt via torch.matmul (dequantizing int8 tiles to a scratch
+    buffer first).
+    """
+
+    def __init__(self, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
+                 scales: Optional[torch.Tensor] = None, group: int = 0):
+        self.w = w
+        self.bias = bias
+        self.scales = scales
+        self.group = group
+
+    @property
+    def is_quant(self) -> bool:
+        return self.scales is not None
+
+    @property
+    def out_features(self) -> int:
+        return self.w.shape[0]
+
+    @property
+    def in_features(self) -> int:
+        return self.w.shape[1]
+
+    @classmethod
+    def make(cls, w: torch.Tensor, bias: Optional[torch.Tensor],
+             quant: Optional[QuantConfig]) -> "Linear":
+        if quant is not None and quant.bits == 8:
+            q, s = ops.quantize_int8(w, quant.group)
+            return cls(q, bias, s, quant.group)
+        return cls(w.to(torch.bfloat16), bias)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        m = x.shape[0]
+        if m <= GEMV_MAX_M:
+            if self.is_quant:
+                return ops.gemv_int8(x, self.w, self.scales, self.group, self.bias)
+            return ops.gemv_bf16(x, self.w, self.bias)
+        wd = ops.dequant_int8(self.w, self.scales, self.group) if self.is_quant else self.w
+        y = x @ wd.t()
+        if self.bias is not None:
+            y = y + self.bias
+        return y
+
+    def nbytes(self) -> int:
+        n = self.w.numel() * self.w.element_size()
+        if self.scales is not None:
+            n += self.scales.numel() * self.scales.element_size()
+        if self.bias is not None:
+            n += self.bias.numel() * self.bias.element_size()
+        return n
+
+
+@dataclass
+class LayerWeights:
+    attn_norm: torch.Tensor = None
+    qkv: Linear = None
+    o: Linear = None
+    mlp_norm: torch.Tensor = None
+    gateup: Linear = None          # dense MLP; None for MoE layers
+    down: Linear = None
+    q_norm: Optional[torch.Tensor] = None   # qwen3
+    k_norm: Optional[torch.Tensor] = None
+    # MoE
+    router: Optional[Linear] = None
+    experts_gateup: Optional[list] = None   # list[Linear] per expert
+    experts_down: Optional[list] = None
+
+    def nbytes(self) -> int:
+        n = 0
+        for t in (self.attn_norm, self.mlp_norm, self.q_norm, self.k_norm):
+            if t is not None:
+                n += t.numel() * t.element_size()
+        for l in (self.qkv, self.o, self.gateup, self.down, self.router):
+            if l is not None:
+                n += l.nbytes()
+        for lst in (self.experts_gateup, self.experts_down):
+            if lst:
+                n += sum(l.nbytes() for l in lst)
+        return n
+
+
+class KVCache:
+    """Per-request-group KV cache: [n_local_layers, B, Hkv, Smax, D] x 2.
+
+    ``pos`` is a device int32 tensor [B] (lengths) so decode kernels are
+    hipGraph-replayable. Reference counterpart: per-nonce mlx KVCache dict
+    (reference: src/dnet/shard/runtime.py get_or_make_kv).
+    """
+
+    def __init__(self, cfg: ModelConfig, layer_ids: Sequence[int], batch: int,
+                 smax: int, device):
+        self.layer_ids = list(layer_ids)
+        self.local = {g: i for i, g in enumerate(self.layer_ids)}
+        L = len(self.layer_ids)
+        self.k = torch.zeros(L, batch, cfg.num_kv_heads, smax, cfg.head_dim,
+                             dtype=torch.bfloat16, device=device)
+        self.v = torch.zeros_like(self.k)
+        self.pos = torch.zeros(batch, dtype=torch.int32, device=device)
+        self.smax = smax
+        self.batch = batch
+
+    def reset(self):
+        self.pos.zero_()
+
+    def nbytes(self) -> int:
+        return 2 * self.k.numel() * self.k.element_size()
+
+
+def _chunked_causal_attention(q, k, v, scale, q_offsets):
+    """Prefill attention with explicit GEMMs (hipBLASLt), causal.
+
+    q: [B, Hq, T, D]; k/v: [B, Hkv, S, D] where S >= T and queries occupy
+    positions q_offsets..q_offsets+T-1 (same offset for all batches).
+    Chunked over queries to bound the score matrix.
+    """
+    B, Hq, T, D = q.shape
+    Hkv = k.shape[1]
+    S = k.shape[2]
+    G = Hq // Hkv
+    out = torch.empty_like(q)
+    qc = 512
+    kk = k.unsqueeze(2)  # [B, Hkv, 1, S, D]
+    vv = v.unsqueeze(2)
+    qg = q.view(B, Hkv, G, T, D)
+    og = out.view(B, Hkv, G, T, D)
+    pos_k = torch.arange(S, device=q.device)
+    for t0 in range(0, T, qc):
+        t1 = min(T, t0 + qc)
+        scores = torch.einsum("bhgtd,bhgsd->bhgts",
+                              qg[:, :, :, t0:t1].float(), kk.float()) * scale
+        pos_q = q_offsets + torch.arange(t0, t1, device=q.device)
+        mask = pos_k.view(1, -1) > pos_q.view(-1, 1)
+        scores.masked_fill_(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        og[:, :, :, t0:t1] = torch.einsum("bhgts,bhgsd->bhgtd", p,
+                                          vv.float()).to(q.dtype)
+    return out
+
+
+class RingModel:
+    """Llama-family ring model (llama / qwen2 / qwen3 / mistral; MoE in
+    subclass). Owns only its assigned layers plus the API-layer weights
+    (embedding if it owns layer 0; final norm + lm_head if it owns the last
+    layer) — reference: src/dnet/core/models/llama.py + base.py.
+    """
+
+    model_type = "llama"
+
+    def __init__(self, cfg: ModelConfig, layer_ids: Sequence[int], device,
+                 is_first: bool, is_last: bool, smax: int = 4096):
+        self.cfg = cfg
+        self.layer_ids = list(layer_ids)
+        self.device = torch.device(device)
+        self.is_first = is_first
+        self.is_last = is_last
+        self.smax = smax
+        self.layers: dict[int, LayerWeights] = {}
+        self.embed: Optional[torch.Tensor] = None
+        self.final_norm: Optional[torch.Tensor] = None
+        self.lm_head: Optional[Linear] = None
+        cos, sin = ops.rope_tables(smax, cfg.head_dim, cfg.rope_theta,
+                                   scaling=cfg.rope_scaling)
+        self.cos = cos.to(self.device)
+        self.sin = sin.to(self.device)
+
+    # ---------- weight init / loading ----------
+
+    def init_random(self, seed: int = 0):
+        """Random-init all owned weights (synthetic benchmarking; no network)."""
+        g = torch.Generator().manual_seed(seed)
+        c = self.cfg
+
+        def rand(*shape):
+            std = 0.02
+            return torch.randn(*shape, generator=g, dtype=torch.float32).mul_(std).to(torch.bfloat16)
+
+        for lid in self.layer_ids:
+            self.layers[lid] = self._init_layer(rand, lid)
+        if self.is_first:
+            self.embed = rand(c.vocab_size, c.hidden_size).to(self.device)
+        if self.is_last:
+            self.final_norm = torch.ones(c.hidden_size, dtype=torch.bfloat16,
+                                         device=self.device)
+            if c.tie_word_embeddings and self.embed is not None:
+                self.lm_head = Linear(self.embed)
+            else:
+                self.lm_head = Linear(rand(c.vocab_size, c.hidden_size).to(self.device))
+
+    def _init_layer(self, rand, lid: int) -> LayerWeights:
+        c = self.cfg
+        dev = self.device
+        ones = lambda n: torch.ones(n, dtype=torch.bfloat16, device=dev)
+        lw = LayerWeights(
+            attn_norm=ones(c.hidden_size),
+            qkv=Linear.make(rand(c.qkv_out, c.hidden_size).to(dev),
+                            rand(c.qkv_out).to(dev) if c.attention_bias else None,
+                            c.quant),
+            o=Linear.make(rand(c.hidden_size, c.num_q_heads * c.head_dim).to(dev),
+                          None, c.quant),
+            mlp_norm=ones(c.hidden_size),
+            gateup=Linear.make(rand(2 * c.intermediate_size, c.hidden_size).to(dev),
+                               None, c.quant),
+            down=Linear.make(rand(c.hidden_size, c.intermediate_size).to(dev),
+                             None, c.quant),
+        )
+        if c.qk_norm:
+            lw.q_norm = ones(c.head_dim)
+            lw.k_norm = ones(c.head_dim)
+        return lw
+
+    def load_state_dict(self, sd: dict):
+        """Load HF-style weights, keeping only local layers (abs->local map)."""
+        c = self.cfg
+
+        def get(name):
+            for pref in ("model.", ""):
+                if pref + name in sd:
+                    return sd[pref + name].to(torch.bfloat16)
+            return None
+
+        for lid in self.layer_ids:
+            p = f"layers.{lid}."
+            qw, kw, vw = (get(p + f"self_attn.{x}_proj.weight") for x in "qkv")
+            qb, kb, vb = (get(p + f"self_attn.{x}_proj.bias") for x in "qkv")
+            bias = None
+            if qb is not None:
+                bias = torch.cat([qb, kb, vb]).to(self.device)
+            lw = LayerWeights(
+                attn_norm=get(p + "input_layernorm.weight").to(self.device),
+                qkv=Linear.make(torch.cat([qw, kw, vw]).to(self.device), bias, c.quant),
+                o=Linear.make(get(p + "self_attn.o_proj.weight").to(self.device), None, c.quant),
+                mlp_norm=get(p + "post_attention_layernorm.weight").to(self.device),
+                gateup=Linear.make(
+                    torch.cat([get(p + "mlp.gate_proj.weight"),
+                               get(p + "mlp.up_proj.weight")]).to(self.device),
+                    None, c.quant),
+                down=Linear.make(get(p + "mlp.down_proj.weight").to(self.device), None, c.quant),
+            )
+            qn = get(p + "self_attn.q_norm.weight")
+            if qn is not None:
+                lw.q_norm = qn.to(self.device)
+                lw.k_norm = get(p + "self_attn.k_norm.weight").to(self.device)
+            self.layers[lid] = lw
+        if self.is_first:
+            self.embed = get("embed_tokens.weight").to(self.device)
+        if self.is_last:
+            self.final_norm = get("norm.weight").to(self.device)
+            head = sd.get("lm_head.weight")
+            if head is None or c.tie_word_embeddings:
+                emb = get("embed_tokens.weight").to(self.device)
+                self.lm_head = Linear(emb)
+            else:
+                self.lm_head = Linear(head.to(torch.bfloat16).to(self.device))
+
+    # ---------- forward ----------
+
+    def embed_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
+        assert self.embed is not None, "this shard does not own the embedding"
+        return torch.nn.functional.embedding(tokens, self.embed)
+
+    def _qk_norm(self, q, k, lw):
+        c = self.cfg
+        qh = q.reshape(-1, c.head_dim)
+        q.copy_(ops.rmsnorm(qh.contiguous(), None, lw.q_norm, c.rms_eps).view_as(q))
+        kh = k.reshape(-1, c.head_dim)
+        k.copy_(ops.rmsnorm(kh.contiguous(), None, lw.k_norm, c.rms_eps).view_as(k))
+
+    def decode_window(self, h: torch.Tensor, layer_ids: Sequence[int],
+                      kv: KVCache) -> torch.Tensor:
+        """One decode step over a window of local layers. h: [B, H] bf16
+        (residual stream, modified in place). kv.pos must already hold the
+        write position for this token."""
+        c = self.cfg
+        B = h.shape[0]
+        nq, nkv, d = c.num_q_heads, c.num_kv_heads, c.head_dim
+        len_t = kv.pos + 1  # attend over lengths including the token being written
+        delta = None
+        for lid in layer_ids:
+            lw = self.layers[lid]
+            y = ops.rmsnorm(delta if delta is not None else h,
+                            h if delta is not None else None,
+                            lw.attn_norm, c.rms_eps)
+            qkv = lw.qkv(y)
+            q = qkv[:, :nq * d].view(B, nq, d)
+            k = qkv[:, nq * d:(nq + nkv) * d].view(B, nkv, d)
+            v = qkv[:, (nq + nkv) * d:].view(B, nkv, d)
+            if c.qk_norm:
+                self._qk_norm(q, k, lw)
+            li = kv.local[lid]
+            ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos, self.sin)
+            attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5)
+            o = lw.o(attn.view(B, nq * d))
+            y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
+            delta = self._mlp(y2, lw)
+        h.add_(delta)
+        return h
+
+    def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
+        gu = lw.gateup(y)
+        a = ops.swiglu(gu)
+        return lw.down(a)
+
+    def prefill_window(self, h: torch.Tensor, layer_ids: Sequence[int],
+                       kv: KVCache, p0: int) -> torch.Tensor:
+        """Prefill T tokens. h: [B, T, H]; tokens occupy positions p0..p0+T-1.
+        kv.pos is advanced by the caller after the full shard window."""
+        c = self.cfg
+        B, T, H = h.shape
+        nq, nkv, d = c.num_q_heads, c.num_kv_heads, c.head_dim
+        positions = torch.arange(p0, p0 + T, device=h.device)
+        for lid in layer_ids:
+            lw = self.layers[lid]
+            flat = h.view(B * T, H)
+            y = ops.rmsnorm(flat, None, lw.attn_norm, c.rms_eps)
+            qkv = lw.qkv(y)
+            q = qkv[:, :nq * d].view(B, T, nq, d)
+            k = qkv[:, nq * d:(nq + nkv) * d].view(B, T, nkv, d)
+            v = qkv[:, (nq + nkv) * d:].view(B, T, nkv, d)
+            if c.qk_norm:
+                q = ops.rmsnorm(q.reshape(-1, d).contiguous(), None, lw.q_norm,
+                                c.rms_eps).view(B, T, nq, d)
+                k = ops.rmsnorm(k.reshape(-1, d).contiguous(), None, lw.k_norm,
+                                c.rms_eps).view(B, T, nkv, d)
+            q = ops.rope_apply(q, self.cos, self.sin, positions)
+            k = ops.rope_apply(k, self.cos, self.sin, positions)
+            li = kv.local[lid]
+            kv.k[li][:, :, p0:p0 + T] = k.transpose(1, 2)
+            kv.v[li][:, :, p0:p0 + T] = v.transpose(1, 2)
+            attn = _chunked_causal_attention(
+                q.transpose(1, 2), kv.k[li][:, :, :p0 + T],
+                kv.v[li][:, :, :p0 + T], d ** -0.5, p0)
+            o = lw.o(attn.transpose(1, 2).reshape(B * T, nq * d).contiguous())
+            y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)
+            delta = self._mlp(y2, lw)
+            flat.add_(delta)
+        return h
+
+    def normalize_project(self, h: torch.Tensor) -> torch.Tensor:
+        """Final RMSNorm + lm head -> logits [B, V] (last shard only)."""
+        assert self.final_norm is not None and self.lm_head is not None
+        y = ops.rmsnorm(h, None, self.final_norm, self.cfg.rms_eps)
+        return self.lm_head(y)
+
+    def weight_bytes(self) -> int:
+        return sum(lw.nbytes() for lw in self.layers.values())

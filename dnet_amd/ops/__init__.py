@@ -56,21 +56,22 @@ def rmsnorm(x: torch.Tensor, residual: torch.Tensor | None, w: torch.Tensor,
     return ref.rmsnorm(x, residual, w, eps)
 
 
-def gemv_bf16(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
+              bias: torch.Tensor | None = None) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        _native().gemv_bf16(x, w, out)
+        _native().gemv_bf16(x, w, out, bias)
         return out
-    return ref.gemv_bf16(x, w)
+    return ref.gemv_bf16(x, w, bias)
 
 
 def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
-              group: int) -> torch.Tensor:
+              group: int, bias: torch.Tensor | None = None) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        _native().gemv_int8(x, w, scales, out, group)
+        _native().gemv_int8(x, w, scales, out, group, bias)
         return out
-    return ref.gemv_int8(x, w, scales, group)
+    return ref.gemv_int8(x, w, scales, group, bias)
 
 
 def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Tensor:
@@ -84,7 +85,8 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Ten
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 pos: torch.Tensor, scale: float) -> torch.Tensor:
     if q.is_cuda:
-        out = torch.empty_like(q)
+        # q may be a strided slice of the fused QKV buffer; out is contiguous.
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _native().attn_decode(q, kcache, vcache, pos, out, scale)
         return out
     return ref.attn_decode(q, kcache, vcache, pos, scale)

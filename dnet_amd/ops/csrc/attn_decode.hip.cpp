@@ -43,7 +43,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
                                    const int* __restrict__ pos,
                                    short* __restrict__ out, const int Hq,
                                    const int Hkv, const int Smax,
-                                   const float scale) {
+                                   const float scale, const int ldq) {
   constexpr int P = lds_pitch(D);
   constexpr int DPL = D / kWave;  // output dims per lane (1 or 2)
   const int b = blockIdx.x / Hkv;
@@ -64,7 +64,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   for (int i = threadIdx.x; i < G * D / 8; i += blockDim.x) {
     const int g = i / (D / 8);
     reinterpret_cast<short8*>(q_lds)[i] =
-        reinterpret_cast<const short8*>(q + ((int64_t)b * Hq + hkv * G + g) * D)[i % (D / 8)];
+        reinterpret_cast<const short8*>(q + (int64_t)b * ldq + (hkv * G + g) * D)[i % (D / 8)];
   }
 
   // Per-wave head list: g = wid, wid+4, ... (max 4 heads per wave).
@@ -145,6 +145,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   }
 }
 
+// q may be a column slice of a fused-QKV buffer: strides (ldq, D, 1).
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
@@ -153,25 +154,27 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   DNET_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 16, "GQA group <= 16");
   DNET_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   DNET_CHECK(pos.dtype() == torch::kInt32, "pos int32");
-  DNET_CHECK(q.is_contiguous() && kcache.is_contiguous() && vcache.is_contiguous() &&
+  DNET_CHECK(q.stride(2) == 1 && q.stride(1) == D, "q inner dims contiguous");
+  DNET_CHECK(kcache.is_contiguous() && vcache.is_contiguous() &&
                  out.is_contiguous(), "contig");
   auto stream = current_stream();
   const int G = (int)(Hq / Hkv);
   const int P = (int)D + 4;
   const size_t lds = (2 * kChunk * P + G * D) * sizeof(short);
   const dim3 grid((unsigned)(B * Hkv));
+  const int ldq = (int)q.stride(0);
   if (D == 128) {
     hipLaunchKernelGGL((attn_decode_kernel<128>), grid, dim3(256), lds, stream,
                        (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                        (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
                        (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
-                       (float)scale);
+                       (float)scale, ldq);
   } else {
     hipLaunchKernelGGL((attn_decode_kernel<64>), grid, dim3(256), lds, stream,
                        (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                        (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
                        (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
-                       (float)scale);
+                       (float)scale, ldq);
   }
 }
 

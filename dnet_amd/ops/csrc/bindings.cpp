@@ -4,9 +4,11 @@
 namespace dnet {
 void rmsnorm(torch::Tensor x, c10::optional<torch::Tensor> residual,
              torch::Tensor w, torch::Tensor y, double eps);
-void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out);
+void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out,
+               c10::optional<torch::Tensor> bias);
 void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
-               torch::Tensor out, int64_t group);
+               torch::Tensor out, int64_t group,
+               c10::optional<torch::Tensor> bias);
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,

@@ -23,6 +23,7 @@ constexpr int kRowsPerBlock = 4;  // 4 waves, one output row each
 template <int M>
 __global__ void gemv_bf16_kernel(const short* __restrict__ x,
                                  const short* __restrict__ w,
+                                 const short* __restrict__ bias,
                                  short* __restrict__ out, const int K,
                                  const int N, const int ldx) {
   const int wid = threadIdx.x / kWave;
@@ -46,8 +47,11 @@ __global__ void gemv_bf16_kernel(const short* __restrict__ x,
   }
 #pragma unroll
   for (int m = 0; m < M; ++m) {
-    const float r = wave_reduce_sum(acc[m]);
-    if (lane == 0) out[(int64_t)m * N + n] = f2bits(r);
+    float r = wave_reduce_sum(acc[m]);
+    if (lane == 0) {
+      if (bias != nullptr) r += bits2f(bias[n]);
+      out[(int64_t)m * N + n] = f2bits(r);
+    }
   }
 }
 
@@ -55,6 +59,7 @@ template <int M>
 __global__ void gemv_int8_kernel(const short* __restrict__ x,
                                  const int8_t* __restrict__ w,
                                  const short* __restrict__ scales,
+                                 const short* __restrict__ bias,
                                  short* __restrict__ out, const int K,
                                  const int N, const int G, const int ldx) {
   const int wid = threadIdx.x / kWave;
@@ -87,8 +92,11 @@ __global__ void gemv_int8_kernel(const short* __restrict__ x,
   }
 #pragma unroll
   for (int m = 0; m < M; ++m) {
-    const float r = wave_reduce_sum(acc[m]);
-    if (lane == 0) out[(int64_t)m * N + n] = f2bits(r);
+    float r = wave_reduce_sum(acc[m]);
+    if (lane == 0) {
+      if (bias != nullptr) r += bits2f(bias[n]);
+      out[(int64_t)m * N + n] = f2bits(r);
+    }
   }
 }
 
@@ -109,12 +117,14 @@ static void dispatch_m(int M, LaunchFn&& fn) {
 }
 
 // Host entry: loops over M in tiles of <=16 rows.
-void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out) {
+void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out,
+               c10::optional<torch::Tensor> bias) {
   const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
   DNET_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N, "shape");
   DNET_CHECK(K % 8 == 0, "K % 8");
   DNET_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous(), "contig");
   auto stream = current_stream();
+  const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   const int grid = cdiv(N, kRowsPerBlock);
   int64_t m0 = 0;
   while (m0 < M) {
@@ -126,7 +136,7 @@ void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out) {
       hipLaunchKernelGGL((gemv_bf16_kernel<decltype(mc)::value>), dim3(grid),
                          dim3(kRowsPerBlock * kWave), 0, stream,
                          (const short*)x.data_ptr() + m0 * K,
-                         (const short*)w.data_ptr(),
+                         (const short*)w.data_ptr(), bptr,
                          (short*)out.data_ptr() + m0 * N, (int)K, (int)N, (int)K);
     });
     m0 += mt;
@@ -134,7 +144,8 @@ void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out) {
 }
 
 void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
-               torch::Tensor out, int64_t group) {
+               torch::Tensor out, int64_t group,
+               c10::optional<torch::Tensor> bias) {
   const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
   DNET_CHECK(w.size(1) == K && out.size(1) == N, "shape");
   DNET_CHECK(K % 16 == 0 && group % 16 == 0 && K % group == 0, "K/group align");
@@ -142,6 +153,7 @@ void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
   DNET_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous() &&
                  scales.is_contiguous(), "contig");
   auto stream = current_stream();
+  const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   const int grid = cdiv(N, kRowsPerBlock);
   int64_t m0 = 0;
   while (m0 < M) {
@@ -154,7 +166,7 @@ void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
                          dim3(kRowsPerBlock * kWave), 0, stream,
                          (const short*)x.data_ptr() + m0 * K,
                          (const int8_t*)w.data_ptr(),
-                         (const short*)scales.data_ptr(),
+                         (const short*)scales.data_ptr(), bptr,
                          (short*)out.data_ptr() + m0 * N, (int)K, (int)N,
                          (int)group, (int)K);
     });

@@ -26,8 +26,12 @@ def rmsnorm(x: torch.Tensor, residual: torch.Tensor | None, w: torch.Tensor,
     return (src * scale * w.float()).to(x.dtype)
 
 
-def gemv_bf16(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    return (x.float() @ w.float().t()).to(x.dtype)
+def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
+              bias: torch.Tensor | None = None) -> torch.Tensor:
+    y = x.float() @ w.float().t()
+    if bias is not None:
+        y = y + bias.float()
+    return y.to(x.dtype)
 
 
 def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Tensor:
@@ -37,9 +41,12 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Ten
 
 
 def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
-              group: int) -> torch.Tensor:
+              group: int, bias: torch.Tensor | None = None) -> torch.Tensor:
     wd = dequant_int8(w, scales, group)
-    return (x.float() @ wd.float().t()).to(x.dtype)
+    y = x.float() @ wd.float().t()
+    if bias is not None:
+        y = y + bias.float()
+    return y.to(x.dtype)
 
 
 def quantize_int8(w: torch.Tensor, group: int) -> tuple[torch.Tensor, torch.Tensor]:
@@ -59,7 +66,7 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     B, Hq, D = q.shape
     Hkv = kcache.shape[1]
     G = Hq // Hkv
-    out = torch.zeros_like(q)
+    out = torch.zeros(q.shape, dtype=q.dtype, device=q.device)
     for b in range(B):
         ln = int(pos[b])
         if ln == 0:
@@ -105,11 +112,12 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
                positions: torch.Tensor) -> torch.Tensor:
     """Neox rotate-half rope. x: [..., T, H, D] or [B, H, D] with positions [T]/[B]."""
     half = x.shape[-1] // 2
-    c = cos[positions].float()  # [T, half]
+    c = cos[positions].float()  # [P, half]
     s = sin[positions].float()
-    while c.dim() < x.dim():
-        c = c.unsqueeze(-2)
-        s = s.unsqueeze(-2)
+    # positions index the dim at x.dim()-3 (T for [B,T,H,D], B for [B,H,D])
+    shape = [1] * (x.dim() - 3) + [c.shape[0], 1, half]
+    c = c.view(shape)
+    s = s.view(shape)
     x1 = x[..., :half].float()
     x2 = x[..., half:].float()
     return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
