@@ -1,0 +1,132 @@
+#!/usr/bin/env python3
+"""dnet_amd flagship benchmark: Qwen-2.5 32B int8 pipelined-ring decode.
+
+Measures the BASELINE.json headline metric — output tokens/sec (whole node)
+plus p50 TTFT — on synthetic data with random-init weights (no network).
+
+Single GPU:   python bench.py --steps 32 --warmup 8
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One rank per GPU; ring hops are RCCL send/recv over xGMI. A "step" = every
+sequence in every microbatch advances one token (one full ring round).
+Rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from dnet_amd.models import ModelConfig, PRESETS, QuantConfig
+from dnet_amd.parallel.comm import init_from_env
+from dnet_amd.parallel.ring import RingExecutor
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=32)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--model", default="qwen-2.5-32b")
+    ap.add_argument("--quant", default="int8", choices=["int8", "bf16"])
+    ap.add_argument("--mb-size", type=int, default=8,
+                    help="sequences per microbatch")
+    ap.add_argument("--mb-per-rank", type=int, default=2,
+                    help="microbatches = mb_per_rank * world (pipeline fill)")
+    ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--smax", type=int, default=1024)
+    ap.add_argument("--layers", type=int, default=0,
+                    help="override layer count (debug only; invalid for scoring)")
+    ap.add_argument("--no-graphs", action="store_true")
+    args = ap.parse_args()
+
+    rank, world, device = init_from_env()
+    on_gpu = device.type == "cuda"
+
+    quant = QuantConfig(8, 128) if args.quant == "int8" else None
+    hf = dict(PRESETS[args.model])
+    if args.layers:
+        hf["num_hidden_layers"] = args.layers
+    cfg = ModelConfig.from_hf(hf, quant=quant)
+
+    mb_count = max(args.mb_per_rank * world, 1)
+    ex = RingExecutor(cfg, rank, world, device, mb_count=mb_count,
+                      mb_size=args.mb_size, smax=args.smax, seed=1234,
+                      use_graphs=on_gpu and not args.no_graphs)
+
+    g = torch.Generator().manual_seed(7)
+    tokens = torch.randint(0, cfg.vocab_size,
+                           (mb_count, args.mb_size, args.prompt_len),
+                           generator=g).to(device)
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    # TTFT: prefill + first token
+    barrier_sync()
+    t0 = time.perf_counter()
+    ex.prefill(tokens)
+    barrier_sync()
+    ttft_ms = (time.perf_counter() - t0) * 1e3
+
+    # warmup decode
+    ex.decode_rounds(args.warmup, collect=False)
+    barrier_sync()
+
+    t0 = time.perf_counter()
+    ex.decode_rounds(args.steps, collect=False)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    total_seqs = mb_count * args.mb_size
+    tokens_generated = total_seqs * args.steps
+    toks_per_s = tokens_generated / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        result = {
+            "metric": f"output tokens/sec (whole node), {args.model} "
+                      f"{args.quant} pipelined ring",
+            "value": round(toks_per_s, 2),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if quant is None else "int8-w8a16(bf16 compute)",
+            "data": "synthetic (random tokens, random-init weights)",
+            "ttft_ms": round(ttft_ms, 1),
+            "config": {
+                "model": args.model,
+                "global_batch": total_seqs,
+                "seq_len": args.prompt_len,
+                "gen_len": args.steps,
+                "parallelism": f"ring-pp{world}",
+                "microbatches": mb_count,
+                "mb_size": args.mb_size,
+                "layers": cfg.num_layers,
+            },
+        }
+        print(json.dumps(result))
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,64 @@
+"""Token sampling: temperature / top-k / top-p / min-p + logprobs.
+
+Reference counterpart: src/dnet/core/decoding/sampler.py (mlx_lm
+make_sampler); here implemented on torch logits [B, V].
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+
+@dataclass
+class DecodingConfig:
+    temperature: float = 0.0
+    top_k: int = 0
+    top_p: float = 1.0
+    min_p: float = 0.0
+    logprobs: bool = False
+    top_logprobs: int = 0
+
+
+class Sampler:
+    def __init__(self, cfg: DecodingConfig | None = None,
+                 generator: Optional[torch.Generator] = None):
+        self.cfg = cfg or DecodingConfig()
+        self.generator = generator
+
+    def sample(self, logits: torch.Tensor):
+        """logits [B, V] -> (tokens [B] int64, logprob [B] or None,
+        top_logprobs list or None)."""
+        c = self.cfg
+        lf = logits.float()
+        if c.temperature <= 0.0:
+            tok = lf.argmax(dim=-1)
+        else:
+            x = lf / c.temperature
+            if c.top_k and c.top_k > 0:
+                kth = torch.topk(x, min(c.top_k, x.shape[-1]), dim=-1).values[..., -1:]
+                x = x.masked_fill(x < kth, float("-inf"))
+            if c.top_p < 1.0:
+                sx, si = torch.sort(x, descending=True, dim=-1)
+                probs = torch.softmax(sx, dim=-1)
+                cum = probs.cumsum(dim=-1)
+                keep = cum - probs < c.top_p  # keep at least the top token
+                sx = sx.masked_fill(~keep, float("-inf"))
+                x = torch.full_like(x, float("-inf")).scatter_(-1, si, sx)
+            if c.min_p > 0.0:
+                p = torch.softmax(x, dim=-1)
+                x = x.masked_fill(p < c.min_p * p.amax(-1, keepdim=True),
+                                  float("-inf"))
+            probs = torch.softmax(x, dim=-1)
+            tok = torch.multinomial(probs, 1, generator=self.generator).squeeze(-1)
+        if not c.logprobs:
+            return tok, None, None
+        logp = lf - torch.logsumexp(lf, dim=-1, keepdim=True)
+        chosen = logp.gather(-1, tok.unsqueeze(-1)).squeeze(-1)
+        tops = None
+        if c.top_logprobs:
+            tv, ti = torch.topk(logp, c.top_logprobs, dim=-1)
+            tops = [{int(i): float(v) for v, i in zip(tv[b], ti[b])}
+                    for b in range(logits.shape[0])]
+        return tok, chosen, tops

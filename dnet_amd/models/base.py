@@ -198,25 +198,35 @@ class RingModel:
     # ---------- weight init / loading ----------
 
     def init_random(self, seed: int = 0):
-        """Random-init all owned weights (synthetic benchmarking; no network)."""
-        g = torch.Generator().manual_seed(seed)
+        """Random-init all owned weights (synthetic benchmarking; no network).
+
+        Seeding is per layer id, so any sharding of the same (model, seed)
+        produces identical global weights — ring-vs-single-shard tests rely
+        on this.
+        """
         c = self.cfg
 
-        def rand(*shape):
-            std = 0.02
-            return torch.randn(*shape, generator=g, dtype=torch.float32).mul_(std).to(torch.bfloat16)
+        def rand_for(sub_seed):
+            g = torch.Generator().manual_seed(seed * 100003 + sub_seed)
+
+            def rand(*shape):
+                return torch.randn(*shape, generator=g,
+                                   dtype=torch.float32).mul_(0.02).to(torch.bfloat16)
+            return rand
 
         for lid in self.layer_ids:
-            self.layers[lid] = self._init_layer(rand, lid)
+            self.layers[lid] = self._init_layer(rand_for(lid), lid)
         if self.is_first:
-            self.embed = rand(c.vocab_size, c.hidden_size).to(self.device)
+            self.embed = rand_for(99991)(c.vocab_size, c.hidden_size).to(self.device)
         if self.is_last:
             self.final_norm = torch.ones(c.hidden_size, dtype=torch.bfloat16,
                                          device=self.device)
-            if c.tie_word_embeddings and self.embed is not None:
-                self.lm_head = Linear(self.embed)
+            if c.tie_word_embeddings:
+                emb = self.embed if self.embed is not None else \
+                    rand_for(99991)(c.vocab_size, c.hidden_size).to(self.device)
+                self.lm_head = Linear(emb)
             else:
-                self.lm_head = Linear(rand(c.vocab_size, c.hidden_size).to(self.device))
+                self.lm_head = Linear(rand_for(99992)(c.vocab_size, c.hidden_size).to(self.device))
 
     def _init_layer(self, rand, lid: int) -> LayerWeights:
         c = self.cfg
