@@ -206,12 +206,19 @@ class RingModel:
         """
         c = self.cfg
 
+        on_gpu = self.device.type == "cuda"
+
         def rand_for(sub_seed):
-            g = torch.Generator().manual_seed(seed * 100003 + sub_seed)
+            if on_gpu:
+                g = torch.Generator(device=self.device)
+            else:
+                g = torch.Generator()
+            g.manual_seed(seed * 100003 + sub_seed)
 
             def rand(*shape):
-                return torch.randn(*shape, generator=g,
-                                   dtype=torch.float32).mul_(0.02).to(torch.bfloat16)
+                t = torch.randn(*shape, generator=g, dtype=torch.float32,
+                                device=self.device if on_gpu else "cpu")
+                return t.mul_(0.02).to(torch.bfloat16)
             return rand
 
         for lid in self.layer_ids:

@@ -131,9 +131,14 @@ class RingExecutor:
         log.info("captured %d decode graphs", self.mb_count)
 
     def _run_decode(self, mb: int):
-        if self.use_graphs:
-            if not self._graphs:
+        if self.use_graphs and not self._graphs:
+            try:
                 self._capture_graphs()
+            except Exception as e:  # pragma: no cover
+                log.warning("hipGraph capture failed (%s); falling back to eager", e)
+                self.use_graphs = False
+                self._graphs = []
+        if self.use_graphs:
             self._graphs[mb].replay()
         else:
             self._decode_body(mb)
