@@ -56,11 +56,25 @@ def rmsnorm(x: torch.Tensor, residual: torch.Tensor | None, w: torch.Tensor,
     return ref.rmsnorm(x, residual, w, eps)
 
 
+# split-K f32 scratch, cached per device (max 16 rows x N<8192 cols)
+_scratch: dict = {}
+
+
+def _get_scratch(device) -> torch.Tensor:
+    key = str(device)
+    if key not in _scratch:
+        _scratch[key] = torch.empty(16 * 8192, dtype=torch.float32, device=device)
+    return _scratch[key]
+
+
 def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
               bias: torch.Tensor | None = None) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        _native().gemv_bf16(x, w, out, bias)
+        if x.shape[1] % 32 == 0:
+            _native().gemm_m16(x, w, None, bias, out, _get_scratch(x.device), 0)
+        else:
+            _native().gemv_bf16(x, w, out, bias)
         return out
     return ref.gemv_bf16(x, w, bias)
 
@@ -69,7 +83,11 @@ def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
               group: int, bias: torch.Tensor | None = None) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        _native().gemv_int8(x, w, scales, out, group, bias)
+        if x.shape[1] % 32 == 0 and group % 8 == 0:
+            _native().gemm_m16(x, w, scales, bias, out, _get_scratch(x.device),
+                               group)
+        else:
+            _native().gemv_int8(x, w, scales, out, group, bias)
         return out
     return ref.gemv_int8(x, w, scales, group, bias)
 

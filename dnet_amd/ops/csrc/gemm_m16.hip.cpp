@@ -1,0 +1,183 @@
+// MFMA decode GEMM for M<=16: out[M,N] = x[M,K] @ W[N,K]^T (+bias).
+//
+// The scalar GEMV (gemv.hip.cpp) is VALU-bound at M>=4: every weight element
+// costs ~2 VALU ops per output row. Here one v_mfma_f32_16x16x32_bf16
+// covers a 16(M)x16(N)x32(K) tile, so the per-weight-element cost collapses
+// to the in-register int8->bf16 dequant (QUANT path) or nothing (bf16 path),
+// and the kernel runs at the weight-read HBM bound. Fragment layout for
+// 16x16x32 (cdna_hip_programming.md §3): lane l holds A[row=l&15][k=(l>>4)*8+j],
+// B[col=l&15][k=(l>>4)*8+j], C[col=l&15][row=(l>>4)*4+r].
+//
+// Wave tile: 16 N-columns; block = 4 waves = 64 columns; gridDim.y = SPLITK
+// K-splits (chosen so blocks ~ fill 256 CUs; partials combined by f32
+// atomics into a scratch accumulator, then a tiny convert+bias kernel).
+#include "common.h"
+
+namespace dnet {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+template <bool QUANT>
+__global__ void gemm_m16_kernel(const short* __restrict__ x,
+                                const void* __restrict__ w,
+                                const short* __restrict__ scales,
+                                const short* __restrict__ bias,
+                                short* __restrict__ out,        // SPLITK==1
+                                float* __restrict__ out_f32,    // SPLITK>1
+                                const int M, const int K, const int N,
+                                const int G, const int splitk) {
+  const int wave = threadIdx.x / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int n0 = (blockIdx.x * 4 + wave) * 16;
+  if (n0 >= N) return;
+  const int row = lane & 15;          // A row (x row = output m), B col
+  const int ks = (lane >> 4) * 8;     // k-offset of this lane's 8-elem slice
+  const bool row_ok = row < M;
+  const int n_w = min(n0 + row, N - 1);     // this lane's W row
+  const bool n_ok = n0 + row < N;
+
+  const int chunks = K / 32;
+  const int per_split = chunks / splitk;
+  const int c_begin = blockIdx.y * per_split;
+  const int c_end = (blockIdx.y == splitk - 1) ? chunks : c_begin + per_split;
+
+  const short* xrow = x + (int64_t)(row_ok ? row : 0) * K;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+
+  if (QUANT) {
+    const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
+    const short* srow = scales + (int64_t)n_w * (K / G);
+    for (int c = c_begin; c < c_end; ++c) {
+      const int k = c * 32 + ks;
+      bf16x8 a;
+      if (row_ok) {
+        a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) a[j] = (__bf16)0.f;
+      }
+      // 8 int8 -> bf16 with the per-group scale folded in
+      const int2 wq8 = *reinterpret_cast<const int2*>(&wrow[k]);
+      const int8_t* q = reinterpret_cast<const int8_t*>(&wq8);
+      const float s = bits2f(srow[k / G]);
+      bf16x8 b;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) b[j] = (__bf16)((float)q[j] * s);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+  } else {
+    const short* wrow = (const short*)w + (int64_t)n_w * K;
+    for (int c = c_begin; c < c_end; ++c) {
+      const int k = c * 32 + ks;
+      bf16x8 a;
+      if (row_ok) {
+        a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) a[j] = (__bf16)0.f;
+      }
+      const bf16x8 b = *reinterpret_cast<const bf16x8*>(&wrow[k]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+  }
+
+  // C write: lane covers col = n0 + (lane&15), rows (lane>>4)*4 + 0..3.
+  const int n = n0 + (lane & 15);
+  if (n >= N) return;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = (lane >> 4) * 4 + r;
+    if (m >= M) continue;
+    if (splitk > 1) {
+      atomicAdd(out_f32 + (int64_t)m * N + n, acc[r]);
+    } else {
+      float v = acc[r];
+      if (bias != nullptr) v += bits2f(bias[n]);
+      out[(int64_t)m * N + n] = f2bits(v);
+    }
+  }
+}
+
+__global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
+                                        const short* __restrict__ bias,
+                                        short* __restrict__ out,
+                                        const int64_t total, const int N) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = in[i];
+    if (bias != nullptr) v += bits2f(bias[i % N]);
+    out[i] = f2bits(v);
+  }
+}
+
+static int pick_splitk(int64_t N, int64_t K) {
+  // Aim for 1/2..1x of the 256 CUs (cdna_hip_programming.md "sampling GEMM"
+  // decomposition rule); the last split absorbs any chunk remainder.
+  const int blocks = (int)((N + 63) / 64);
+  int sk = 1;
+  while (sk < 8 && blocks * sk * 2 < 256 && (K / 32) / (sk * 2) > 0) sk *= 2;
+  return sk;
+}
+
+// out must be [M<=16, N]; scratch_f32 (nullable) must be [M, N] f32 if
+// splitk>1 would be chosen (the wrapper supplies it).
+static void launch_m16(torch::Tensor x, torch::Tensor w,
+                       c10::optional<torch::Tensor> scales,
+                       c10::optional<torch::Tensor> bias, torch::Tensor out,
+                       c10::optional<torch::Tensor> scratch, int group,
+                       int64_t m0, int M) {
+  const int64_t K = x.size(1), N = w.size(0);
+  auto stream = current_stream();
+  const bool quant = scales.has_value();
+  const int sk = pick_splitk(N, K);
+  const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
+  const dim3 grid((unsigned)((N + 63) / 64), sk);
+  const short* xp = (const short*)x.data_ptr() + m0 * K;
+  short* op = (short*)out.data_ptr() + m0 * N;
+  float* fp = nullptr;
+  if (sk > 1) {
+    TORCH_CHECK(scratch.has_value() && scratch->numel() >= (int64_t)M * N,
+                "split-k scratch required");
+    fp = (float*)scratch->data_ptr();
+    DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * M * N, stream));
+  }
+  if (quant) {
+    hipLaunchKernelGGL((gemm_m16_kernel<true>), grid, dim3(256), 0, stream,
+                       xp, w.data_ptr(), (const short*)scales->data_ptr(),
+                       sk > 1 ? nullptr : bptr, op, fp, M, (int)K, (int)N,
+                       group, sk);
+  } else {
+    hipLaunchKernelGGL((gemm_m16_kernel<false>), grid, dim3(256), 0, stream,
+                       xp, w.data_ptr(), nullptr, sk > 1 ? nullptr : bptr, op,
+                       fp, M, (int)K, (int)N, group, sk);
+  }
+  if (sk > 1) {
+    const int64_t total = (int64_t)M * N;
+    const int cgrid = (int)std::min<int64_t>((total + 255) / 256, 2048);
+    hipLaunchKernelGGL(f32_to_bf16_bias_kernel, dim3(cgrid), dim3(256), 0,
+                       stream, fp, bptr, op, total, (int)N);
+  }
+}
+
+void gemm_m16(torch::Tensor x, torch::Tensor w,
+              c10::optional<torch::Tensor> scales,
+              c10::optional<torch::Tensor> bias, torch::Tensor out,
+              c10::optional<torch::Tensor> scratch, int64_t group) {
+  const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+  DNET_CHECK(K % 32 == 0, "K % 32 == 0 required for the MFMA path");
+  DNET_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N, "shape");
+  DNET_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous(), "contig");
+  if (scales.has_value()) {
+    DNET_CHECK(group % 8 == 0 && K % group == 0, "group align");
+    DNET_CHECK(scales->is_contiguous(), "scales contig");
+  }
+  int64_t m0 = 0;
+  while (m0 < M) {
+    const int mt = (int)std::min<int64_t>(M - m0, 16);
+    launch_m16(x, w, scales, bias, out, scratch, (int)group, m0, mt);
+    m0 += mt;
+  }
+}
+
+}  // namespace dnet

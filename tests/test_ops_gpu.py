@@ -115,3 +115,27 @@ def test_swiglu():
     y = ops.swiglu(gu)
     y_ref = ref.swiglu(gu.cpu())
     assert torch.allclose(y.float().cpu(), y_ref.float(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("m,n,k,group,bias", [
+    (1, 1000, 2048, 0, True),
+    (8, 1000, 2048, 64, True),
+    (16, 256, 4096, 128, False),
+    (33, 4096, 5120, 128, True),
+])
+def test_gemm_m16_shapes(m, n, k, group, bias):
+    """MFMA decode GEMM path: odd N (tail waves), split-K shapes, bias."""
+    torch.manual_seed(42)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=_dev())
+    b = (torch.randn(n, dtype=torch.bfloat16, device=_dev()) if bias else None)
+    if group:
+        wf = torch.randn(n, k, dtype=torch.bfloat16, device=_dev()) / 30
+        q, scales = ops.quantize_int8(wf, group)
+        out = ops.gemv_int8(x, q, scales, group, b)
+        out_ref = ref.gemv_int8(x.cpu(), q.cpu(), scales.cpu(), group,
+                                b.cpu() if bias else None)
+    else:
+        w = torch.randn(n, k, dtype=torch.bfloat16, device=_dev()) / 30
+        out = ops.gemv_bf16(x, w, b)
+        out_ref = ref.gemv_bf16(x.cpu(), w.cpu(), b.cpu() if bias else None)
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=6e-2, rtol=3e-2)
