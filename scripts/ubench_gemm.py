@@ -3,7 +3,11 @@
 Prints achieved weight-read GB/s for the MFMA path vs the scalar GEMV for
 each Qwen-32B layer shape, plus graph-capture status for one decode layer.
 """
+import pathlib
+import sys
 import time
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
 
 import torch
 
