@@ -33,54 +33,87 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   if (n0 >= N) return;
   const int row = lane & 15;          // A row (x row = output m), B col
   const int ks = (lane >> 4) * 8;     // k-offset of this lane's 8-elem slice
-  const bool row_ok = row < M;
   const int n_w = min(n0 + row, N - 1);     // this lane's W row
-  const bool n_ok = n0 + row < N;
 
   const int chunks = K / 32;
   const int per_split = chunks / splitk;
   const int c_begin = blockIdx.y * per_split;
   const int c_end = (blockIdx.y == splitk - 1) ? chunks : c_begin + per_split;
 
-  const short* xrow = x + (int64_t)(row_ok ? row : 0) * K;
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  // Rows >= M read a clamped (valid) row and produce garbage C rows that are
+  // never stored — cheaper than per-load zero-masking.
+  const short* xrow = x + (int64_t)min(row, M - 1) * K;
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
 
+  // 4-chunk unrolled main loop: all 8-12 loads issue before the first
+  // dequant+MFMA, so ~4 HBM loads stay in flight per wave (the single-chunk
+  // loop was load-latency-bound at ~1 chunk / 970 cycles).
   if (QUANT) {
     const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
     const short* srow = scales + (int64_t)n_w * (K / G);
-    for (int c = c_begin; c < c_end; ++c) {
-      const int k = c * 32 + ks;
-      bf16x8 a;
-      if (row_ok) {
-        a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
-      } else {
+    int c = c_begin;
+    for (; c + 4 <= c_end; c += 4) {
+      bf16x8 a[4];
+      int2 wq[4];
+      float s[4];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) a[j] = (__bf16)0.f;
+      for (int u = 0; u < 4; ++u) {
+        const int k = (c + u) * 32 + ks;
+        a[u] = *reinterpret_cast<const bf16x8*>(&xrow[k]);
+        wq[u] = *reinterpret_cast<const int2*>(&wrow[k]);
+        s[u] = bits2f(srow[k / G]);
       }
-      // 8 int8 -> bf16 with the per-group scale folded in
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int8_t* q = reinterpret_cast<const int8_t*>(&wq[u]);
+        bf16x8 b;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) b[j] = (__bf16)((float)q[j] * s[u]);
+        if (u & 1)
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b, acc1, 0, 0, 0);
+        else
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b, acc0, 0, 0, 0);
+      }
+    }
+    for (; c < c_end; ++c) {
+      const int k = c * 32 + ks;
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
       const int2 wq8 = *reinterpret_cast<const int2*>(&wrow[k]);
       const int8_t* q = reinterpret_cast<const int8_t*>(&wq8);
       const float s = bits2f(srow[k / G]);
       bf16x8 b;
 #pragma unroll
       for (int j = 0; j < 8; ++j) b[j] = (__bf16)((float)q[j] * s);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
     }
   } else {
     const short* wrow = (const short*)w + (int64_t)n_w * K;
-    for (int c = c_begin; c < c_end; ++c) {
-      const int k = c * 32 + ks;
-      bf16x8 a;
-      if (row_ok) {
-        a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
-      } else {
+    int c = c_begin;
+    for (; c + 4 <= c_end; c += 4) {
+      bf16x8 a[4], b[4];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) a[j] = (__bf16)0.f;
+      for (int u = 0; u < 4; ++u) {
+        const int k = (c + u) * 32 + ks;
+        a[u] = *reinterpret_cast<const bf16x8*>(&xrow[k]);
+        b[u] = *reinterpret_cast<const bf16x8*>(&wrow[k]);
       }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        if (u & 1)
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc1, 0, 0, 0);
+        else
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc0, 0, 0, 0);
+      }
+    }
+    for (; c < c_end; ++c) {
+      const int k = c * 32 + ks;
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
       const bf16x8 b = *reinterpret_cast<const bf16x8*>(&wrow[k]);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
     }
   }
+  const f32x4 acc = acc0 + acc1;
 
   // C write: lane covers col = n0 + (lane&15), rows (lane>>4)*4 + 0..3.
   const int n = n0 + (lane & 15);
@@ -111,12 +144,14 @@ __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
   }
 }
 
-static int pick_splitk(int64_t N, int64_t K) {
-  // Aim for 1/2..1x of the 256 CUs (cdna_hip_programming.md "sampling GEMM"
-  // decomposition rule); the last split absorbs any chunk remainder.
+static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems) {
+  // Target ~320-640 blocks (~5-10 waves per CU) so load latency is covered;
+  // the last split absorbs any chunk remainder. Split-K needs the f32
+  // scratch to hold the [M, N] accumulator.
   const int blocks = (int)((N + 63) / 64);
   int sk = 1;
-  while (sk < 8 && blocks * sk * 2 < 256 && (K / 32) / (sk * 2) > 0) sk *= 2;
+  while (sk < 16 && blocks * sk < 320 && (K / 32) / (sk * 2) > 0) sk *= 2;
+  if (sk > 1 && M * N > scratch_elems) sk = 1;
   return sk;
 }
 
@@ -130,7 +165,8 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   const int64_t K = x.size(1), N = w.size(0);
   auto stream = current_stream();
   const bool quant = scales.has_value();
-  const int sk = pick_splitk(N, K);
+  const int64_t scratch_elems = scratch.has_value() ? scratch->numel() : 0;
+  const int sk = pick_splitk(N, K, M, scratch_elems);
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   const dim3 grid((unsigned)((N + 63) / 64), sk);
   const short* xp = (const short*)x.data_ptr() + m0 * K;
