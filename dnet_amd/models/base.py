@@ -35,11 +35,13 @@ class Linear:
     """
 
     def __init__(self, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
-                 scales: Optional[torch.Tensor] = None, group: int = 0):
+                 scales: Optional[torch.Tensor] = None, group: int = 0,
+                 packed: bool = False):
         self.w = w
         self.bias = bias
         self.scales = scales
         self.group = group
+        self.packed = packed  # MFMA chunk-pair weight layout (GPU int8 path)
 
     @property
     def is_quant(self) -> bool:
@@ -58,16 +60,22 @@ class Linear:
              quant: Optional[QuantConfig]) -> "Linear":
         if quant is not None and quant.bits == 8:
             q, s = ops.quantize_int8(w, quant.group)
-            return cls(q, bias, s, quant.group)
+            packed = (w.is_cuda and w.shape[1] % 64 == 0
+                      and quant.group % 64 == 0)
+            if packed:
+                q = ops.pack_int8_mfma(q)
+            return cls(q, bias, s, quant.group, packed)
         return cls(w.to(torch.bfloat16), bias)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         m = x.shape[0]
         if m <= GEMV_MAX_M:
             if self.is_quant:
-                return ops.gemv_int8(x, self.w, self.scales, self.group, self.bias)
+                return ops.gemv_int8(x, self.w, self.scales, self.group,
+                                     self.bias, self.packed)
             return ops.gemv_bf16(x, self.w, self.bias)
-        wd = ops.dequant_int8(self.w, self.scales, self.group) if self.is_quant else self.w
+        wd = (ops.dequant_int8(self.w, self.scales, self.group, self.packed)
+              if self.is_quant else self.w)
         y = x @ wd.t()
         if self.bias is not None:
             y = y + self.bias

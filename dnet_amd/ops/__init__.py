@@ -72,7 +72,8 @@ def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
         if x.shape[1] % 32 == 0:
-            _native().gemm_m16(x, w, None, bias, out, _get_scratch(x.device), 0)
+            _native().gemm_m16(x, w, None, bias, out, _get_scratch(x.device),
+                               0, False)
         else:
             _native().gemv_bf16(x, w, out, bias)
         return out
@@ -80,23 +81,30 @@ def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
 
 
 def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
-              group: int, bias: torch.Tensor | None = None) -> torch.Tensor:
+              group: int, bias: torch.Tensor | None = None,
+              packed: bool = False) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
         if x.shape[1] % 32 == 0 and group % 8 == 0:
             _native().gemm_m16(x, w, scales, bias, out, _get_scratch(x.device),
-                               group)
+                               group, packed)
         else:
+            assert not packed, "packed layout needs the MFMA path"
             _native().gemv_int8(x, w, scales, out, group, bias)
         return out
+    if packed:
+        w = ref.unpack_int8_mfma(w)
     return ref.gemv_int8(x, w, scales, group, bias)
 
 
-def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int) -> torch.Tensor:
+def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
+                 packed: bool = False) -> torch.Tensor:
     if w.is_cuda:
         out = torch.empty(w.shape, dtype=torch.bfloat16, device=w.device)
-        _native().dequant_int8(w, scales, out, group)
+        _native().dequant_int8(w, scales, out, group, packed)
         return out
+    if packed:
+        w = ref.unpack_int8_mfma(w)
     return ref.dequant_int8(w, scales, group)
 
 
@@ -127,5 +135,7 @@ def swiglu(gu: torch.Tensor) -> torch.Tensor:
 
 
 quantize_int8 = ref.quantize_int8
+pack_int8_mfma = ref.pack_int8_mfma
+unpack_int8_mfma = ref.unpack_int8_mfma
 rope_tables = ref.rope_tables
 rope_apply = ref.rope_apply

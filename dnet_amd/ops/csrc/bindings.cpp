@@ -16,11 +16,11 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor cos_table, torch::Tensor sin_table);
 void swiglu(torch::Tensor gu, torch::Tensor y);
 void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
-                  int64_t group);
+                  int64_t group, bool packed);
 void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
-              c10::optional<torch::Tensor> scratch, int64_t group);
+              c10::optional<torch::Tensor> scratch, int64_t group, bool packed);
 }  // namespace dnet
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

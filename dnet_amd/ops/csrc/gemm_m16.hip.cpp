@@ -18,7 +18,14 @@ namespace dnet {
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-template <bool QUANT>
+__device__ __forceinline__ bf16x8 deq8(const int8_t* q, const float s) {
+  bf16x8 b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) b[j] = (__bf16)((float)q[j] * s);
+  return b;
+}
+
+template <bool QUANT, bool PACKED>
 __global__ void gemm_m16_kernel(const short* __restrict__ x,
                                 const void* __restrict__ w,
                                 const short* __restrict__ scales,
@@ -49,7 +56,47 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   // 4-chunk unrolled main loop: all 8-12 loads issue before the first
   // dequant+MFMA, so ~4 HBM loads stay in flight per wave (the single-chunk
   // loop was load-latency-bound at ~1 chunk / 970 cycles).
-  if (QUANT) {
+  if (QUANT && PACKED) {
+    // W stored in MFMA chunk-pair order (pack_int8_mfma): one b128 load per
+    // lane covers its B slices of two adjacent chunks -> full 64 B bursts.
+    const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
+    const short* srow = scales + (int64_t)n_w * (K / G);
+    const int pairs = K / 64;
+    const int pp = pairs / splitk;
+    int p = blockIdx.y * pp;
+    const int p_end = (blockIdx.y == splitk - 1) ? pairs : p + pp;
+    const int woff = (lane >> 4) * 16;
+    for (; p + 2 <= p_end; p += 2) {
+      bf16x8 a[4];
+      int4 wq[2];
+      float s2[2];
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        wq[u] = *reinterpret_cast<const int4*>(&wrow[(p + u) * 64 + woff]);
+        s2[u] = bits2f(srow[((p + u) * 64) / G]);
+        a[2 * u] = *reinterpret_cast<const bf16x8*>(&xrow[(p + u) * 64 + ks]);
+        a[2 * u + 1] = *reinterpret_cast<const bf16x8*>(&xrow[(p + u) * 64 + 32 + ks]);
+      }
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int8_t* q = reinterpret_cast<const int8_t*>(&wq[u]);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u], deq8(q, s2[u]),
+                                                       acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u + 1],
+                                                       deq8(q + 8, s2[u]),
+                                                       acc1, 0, 0, 0);
+      }
+    }
+    for (; p < p_end; ++p) {
+      const int4 wq = *reinterpret_cast<const int4*>(&wrow[p * 64 + woff]);
+      const int8_t* q = reinterpret_cast<const int8_t*>(&wq);
+      const float sv = bits2f(srow[(p * 64) / G]);
+      const bf16x8 al = *reinterpret_cast<const bf16x8*>(&xrow[p * 64 + ks]);
+      const bf16x8 ah = *reinterpret_cast<const bf16x8*>(&xrow[p * 64 + 32 + ks]);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(al, deq8(q, sv), acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ah, deq8(q + 8, sv), acc1, 0, 0, 0);
+    }
+  } else if (QUANT) {
     const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
     const short* srow = scales + (int64_t)n_w * (K / G);
     int c = c_begin;
@@ -161,7 +208,7 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
                        c10::optional<torch::Tensor> scales,
                        c10::optional<torch::Tensor> bias, torch::Tensor out,
                        c10::optional<torch::Tensor> scratch, int group,
-                       int64_t m0, int M) {
+                       int64_t m0, int M, bool packed) {
   const int64_t K = x.size(1), N = w.size(0);
   auto stream = current_stream();
   const bool quant = scales.has_value();
@@ -178,15 +225,21 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
     fp = (float*)scratch->data_ptr();
     DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * M * N, stream));
   }
-  if (quant) {
-    hipLaunchKernelGGL((gemm_m16_kernel<true>), grid, dim3(256), 0, stream,
-                       xp, w.data_ptr(), (const short*)scales->data_ptr(),
+  if (quant && packed) {
+    hipLaunchKernelGGL((gemm_m16_kernel<true, true>), grid, dim3(256), 0,
+                       stream, xp, w.data_ptr(), (const short*)scales->data_ptr(),
+                       sk > 1 ? nullptr : bptr, op, fp, M, (int)K, (int)N,
+                       group, sk);
+  } else if (quant) {
+    hipLaunchKernelGGL((gemm_m16_kernel<true, false>), grid, dim3(256), 0,
+                       stream, xp, w.data_ptr(), (const short*)scales->data_ptr(),
                        sk > 1 ? nullptr : bptr, op, fp, M, (int)K, (int)N,
                        group, sk);
   } else {
-    hipLaunchKernelGGL((gemm_m16_kernel<false>), grid, dim3(256), 0, stream,
-                       xp, w.data_ptr(), nullptr, sk > 1 ? nullptr : bptr, op,
-                       fp, M, (int)K, (int)N, group, sk);
+    hipLaunchKernelGGL((gemm_m16_kernel<false, false>), grid, dim3(256), 0,
+                       stream, xp, w.data_ptr(), nullptr,
+                       sk > 1 ? nullptr : bptr, op, fp, M, (int)K, (int)N,
+                       group, sk);
   }
   if (sk > 1) {
     const int64_t total = (int64_t)M * N;
@@ -199,7 +252,8 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
 void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
-              c10::optional<torch::Tensor> scratch, int64_t group) {
+              c10::optional<torch::Tensor> scratch, int64_t group,
+              bool packed) {
   const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
   DNET_CHECK(K % 32 == 0, "K % 32 == 0 required for the MFMA path");
   DNET_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N, "shape");
@@ -207,11 +261,12 @@ void gemm_m16(torch::Tensor x, torch::Tensor w,
   if (scales.has_value()) {
     DNET_CHECK(group % 8 == 0 && K % group == 0, "group align");
     DNET_CHECK(scales->is_contiguous(), "scales contig");
+    if (packed) DNET_CHECK(K % 64 == 0 && group % 64 == 0, "packed layout align");
   }
   int64_t m0 = 0;
   while (m0 < M) {
     const int mt = (int)std::min<int64_t>(M - m0, 16);
-    launch_m16(x, w, scales, bias, out, scratch, (int)group, m0, mt);
+    launch_m16(x, w, scales, bias, out, scratch, (int)group, m0, mt, packed);
     m0 += mt;
   }
 }

@@ -43,6 +43,7 @@ void swiglu(torch::Tensor gu, torch::Tensor y) {
                      (const short*)gu.data_ptr(), (short*)y.data_ptr(), I, T);
 }
 
+template <bool PACKED>
 __global__ void dequant_int8_kernel(const int8_t* __restrict__ w,
                                     const short* __restrict__ scales,
                                     short* __restrict__ out, const int64_t K,
@@ -61,25 +62,39 @@ __global__ void dequant_int8_kernel(const int8_t* __restrict__ w,
       o0.x[j] = f2bits((float)q[j] * s);
       o1.x[j] = f2bits((float)q[8 + j] * s);
     }
-    reinterpret_cast<short8*>(out + n * K)[k0 / 8] = o0;
-    reinterpret_cast<short8*>(out + n * K)[k0 / 8 + 1] = o1;
+    if (PACKED) {
+      // packed vec covers orig k = p*64 + s*8 (low 8) and +32 (high 8)
+      const int64_t pr = k0 / 64, sl = (k0 / 16) % 4;
+      const int64_t ka = pr * 64 + sl * 8;
+      reinterpret_cast<short8*>(out + n * K)[ka / 8] = o0;
+      reinterpret_cast<short8*>(out + n * K)[(ka + 32) / 8] = o1;
+    } else {
+      reinterpret_cast<short8*>(out + n * K)[k0 / 8] = o0;
+      reinterpret_cast<short8*>(out + n * K)[k0 / 8 + 1] = o1;
+    }
   }
 }
 
 // Dequantize grouped-int8 weights to bf16 (prefill path: the dequantized
 // tile feeds a hipBLASLt GEMM via torch.matmul).
 void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
-                  int64_t group) {
+                  int64_t group, bool packed) {
   const int64_t N = w.size(0), K = w.size(1);
   DNET_CHECK(K % 16 == 0 && group % 16 == 0, "K align");
+  if (packed) DNET_CHECK(K % 64 == 0 && group % 64 == 0, "packed align");
   DNET_CHECK(out.size(0) == N && out.size(1) == K, "out shape");
   DNET_CHECK(w.is_contiguous() && scales.is_contiguous() && out.is_contiguous(), "contig");
   auto stream = current_stream();
   const int64_t nvec = N * K / 16;
   const int grid = (int)std::min<int64_t>((nvec + 255) / 256, 2048);
-  hipLaunchKernelGGL(dequant_int8_kernel, dim3(grid), dim3(256), 0, stream,
-                     (const int8_t*)w.data_ptr(), (const short*)scales.data_ptr(),
-                     (short*)out.data_ptr(), K, (int)group, N);
+  if (packed)
+    hipLaunchKernelGGL(dequant_int8_kernel<true>, dim3(grid), dim3(256), 0, stream,
+                       (const int8_t*)w.data_ptr(), (const short*)scales.data_ptr(),
+                       (short*)out.data_ptr(), K, (int)group, N);
+  else
+    hipLaunchKernelGGL(dequant_int8_kernel<false>, dim3(grid), dim3(256), 0, stream,
+                       (const int8_t*)w.data_ptr(), (const short*)scales.data_ptr(),
+                       (short*)out.data_ptr(), K, (int)group, N);
 }
 
 }  // namespace dnet

@@ -49,6 +49,26 @@ def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
     return y.to(x.dtype)
 
 
+def pack_int8_mfma(q: torch.Tensor) -> torch.Tensor:
+    """Permute each row's K dim into MFMA chunk-pair order so a lane's
+    16 B load covers its 8-elem B-fragment slices of two adjacent K=32
+    chunks (full 64 B HBM bursts per 16-row group instead of 32 B halves).
+
+    orig [N, cp, half, slice, j] -> packed [N, cp, slice, half, j]
+    (cp = 64-k pair, half = which chunk, slice = lane>>4, j = 0..7).
+    """
+    n, k = q.shape
+    assert k % 64 == 0
+    v = q.view(n, k // 64, 2, 4, 8)
+    return v.permute(0, 1, 3, 2, 4).reshape(n, k).contiguous()
+
+
+def unpack_int8_mfma(qp: torch.Tensor) -> torch.Tensor:
+    n, k = qp.shape
+    v = qp.view(n, k // 64, 4, 2, 8)
+    return v.permute(0, 1, 3, 2, 4).reshape(n, k).contiguous()
+
+
 def quantize_int8(w: torch.Tensor, group: int) -> tuple[torch.Tensor, torch.Tensor]:
     """Symmetric per-group int8 quantization along the K (last) dim."""
     n, k = w.shape

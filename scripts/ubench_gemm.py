@@ -40,6 +40,7 @@ def main():
     for name, N, K in SHAPES:
         wf = torch.randn(N, K, dtype=torch.bfloat16, device=dev) / 30
         q, scales = ops.quantize_int8(wf, 128)
+        qp = ops.pack_int8_mfma(q)
         wbytes_i8 = q.numel() + scales.numel() * 2
         wbytes_bf = wf.numel() * 2
         for M in (1, 8, 16):
@@ -48,14 +49,17 @@ def main():
             scratch = ops._get_scratch(dev)
 
             t_mfma = bench(lambda: nat.gemm_m16(x, q, scales, None, out,
-                                                scratch, 128))
+                                                scratch, 128, False))
+            t_pk = bench(lambda: nat.gemm_m16(x, qp, scales, None, out,
+                                              scratch, 128, True))
             t_old = bench(lambda: nat.gemv_int8(x, q, scales, out, 128, None))
             t_mfma_bf = bench(lambda: nat.gemm_m16(x, wf, None, None, out,
-                                                   scratch, 0))
+                                                   scratch, 0, False))
             print(f"{name:8s} N={N:6d} K={K:6d} M={M:2d}  "
-                  f"int8-mfma {t_mfma*1e6:7.1f}us {wbytes_i8/t_mfma/1e9:7.1f}GB/s | "
-                  f"int8-scalar {t_old*1e6:7.1f}us {wbytes_i8/t_old/1e9:7.1f}GB/s | "
-                  f"bf16-mfma {t_mfma_bf*1e6:7.1f}us {wbytes_bf/t_mfma_bf/1e9:7.1f}GB/s")
+                  f"i8-mfma {wbytes_i8/t_mfma/1e9:6.0f} | "
+                  f"i8-packed {t_pk*1e6:7.1f}us {wbytes_i8/t_pk/1e9:6.0f}GB/s | "
+                  f"i8-scalar {wbytes_i8/t_old/1e9:6.0f} | "
+                  f"bf16-mfma {t_mfma_bf*1e6:7.1f}us {wbytes_bf/t_mfma_bf/1e9:6.0f}GB/s")
         del wf, q, scales
         torch.cuda.empty_cache()
 

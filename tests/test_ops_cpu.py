@@ -50,3 +50,10 @@ def test_swiglu_shapes():
     gu = torch.randn(3, 5, 2 * 32, dtype=torch.bfloat16)
     y = ops.swiglu(gu)
     assert y.shape == (3, 5, 32)
+
+
+def test_pack_unpack_int8_mfma():
+    q = torch.randint(-127, 128, (4, 256), dtype=torch.int8)
+    p = ref.pack_int8_mfma(q)
+    assert not torch.equal(p, q)
+    assert torch.equal(ref.unpack_int8_mfma(p), q)

@@ -139,3 +139,21 @@ def test_gemm_m16_shapes(m, n, k, group, bias):
         out = ops.gemv_bf16(x, w, b)
         out_ref = ref.gemv_bf16(x.cpu(), w.cpu(), b.cpu() if bias else None)
     assert torch.allclose(out.float().cpu(), out_ref.float(), atol=6e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("m,n,k,group", [(1, 512, 1024, 64), (8, 1000, 5120, 128),
+                                         (16, 4096, 27648, 128)])
+def test_gemm_m16_packed(m, n, k, group):
+    """Packed MFMA weight layout must match the plain-layout reference."""
+    torch.manual_seed(9)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=_dev())
+    wf = torch.randn(n, k, dtype=torch.bfloat16, device=_dev()) / 30
+    q, scales = ops.quantize_int8(wf, group)
+    qp = ops.pack_int8_mfma(q)
+    out = ops.gemv_int8(x, qp, scales, group, None, packed=True)
+    out_ref = ref.gemv_int8(x.cpu(), q.cpu(), scales.cpu(), group)
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=6e-2, rtol=3e-2)
+    # packed dequant inverts the layout
+    wd = ops.dequant_int8(qp, scales, group, packed=True)
+    wd_ref = ref.dequant_int8(q.cpu(), scales.cpu(), group)
+    assert torch.allclose(wd.float().cpu(), wd_ref.float(), atol=2e-2, rtol=2e-2)
