@@ -66,19 +66,19 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
     int p = blockIdx.y * pp;
     const int p_end = (blockIdx.y == splitk - 1) ? pairs : p + pp;
     const int woff = (lane >> 4) * 16;
-    for (; p + 2 <= p_end; p += 2) {
-      bf16x8 a[4];
-      int4 wq[2];
-      float s2[2];
+    for (; p + 4 <= p_end; p += 4) {
+      bf16x8 a[8];
+      int4 wq[4];
+      float s2[4];
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
+      for (int u = 0; u < 4; ++u) {
         wq[u] = *reinterpret_cast<const int4*>(&wrow[(p + u) * 64 + woff]);
         s2[u] = bits2f(srow[((p + u) * 64) / G]);
         a[2 * u] = *reinterpret_cast<const bf16x8*>(&xrow[(p + u) * 64 + ks]);
         a[2 * u + 1] = *reinterpret_cast<const bf16x8*>(&xrow[(p + u) * 64 + 32 + ks]);
       }
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
+      for (int u = 0; u < 4; ++u) {
         const int8_t* q = reinterpret_cast<const int8_t*>(&wq[u]);
         acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u], deq8(q, s2[u]),
                                                        acc0, 0, 0, 0);
@@ -192,12 +192,13 @@ __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
 }
 
 static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems) {
-  // Target ~320-640 blocks (~5-10 waves per CU) so load latency is covered;
-  // the last split absorbs any chunk remainder. Split-K needs the f32
-  // scratch to hold the [M, N] accumulator.
+  // Target ~1024 blocks (~4 blocks / 16 waves per CU) so HBM latency is
+  // covered by wave overlap; the last split absorbs any chunk remainder.
+  // Split-K needs the f32 scratch to hold the [M, N] accumulator, and each
+  // split should keep >= 8 K-chunks of work.
   const int blocks = (int)((N + 63) / 64);
   int sk = 1;
-  while (sk < 16 && blocks * sk < 320 && (K / 32) / (sk * 2) > 0) sk *= 2;
+  while (sk < 32 && blocks * sk < 1024 && (K / 32) / (sk * 2) >= 8) sk *= 2;
   if (sk > 1 && M * N > scratch_elems) sk = 1;
   return sk;
 }
