@@ -37,7 +37,9 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   const int wave = threadIdx.x / kWave;
   const int lane = threadIdx.x & (kWave - 1);
   const int n0 = (blockIdx.x * 4 + wave) * 16;
-  if (n0 >= N) return;
+  // PACKED path has block-wide barriers: OOB waves must stay alive (their
+  // stores are skipped); the barrier-free paths may exit early.
+  if (!PACKED && n0 >= N) return;
   const int row = lane & 15;          // A row (x row = output m), B col
   const int ks = (lane >> 4) * 8;     // k-offset of this lane's 8-elem slice
   const int n_w = min(n0 + row, N - 1);     // this lane's W row
@@ -59,42 +61,65 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   if (QUANT && PACKED) {
     // W stored in MFMA chunk-pair order (pack_int8_mfma): one b128 load per
     // lane covers its B slices of two adjacent chunks -> full 64 B bursts.
+    // x is staged tile-wise into LDS with coalesced full-line loads (the
+    // fragment-shaped direct read touches 16 scattered rows per instruction
+    // and costs up to +45% — cdna_hip_programming.md §5 M=256 GEMM row).
+    constexpr int XT = 1024;              // k values per x tile
+    constexpr int SE = XT + 8;            // row stride in bf16 elems (16B pad)
+    __shared__ short x_lds[16 * SE];
     const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
     const short* srow = scales + (int64_t)n_w * (K / G);
     const int pairs = K / 64;
     const int pp = pairs / splitk;
-    int p = blockIdx.y * pp;
-    const int p_end = (blockIdx.y == splitk - 1) ? pairs : p + pp;
+    const int p_begin = blockIdx.y * pp;
+    const int p_end = (blockIdx.y == splitk - 1) ? pairs : p_begin + pp;
     const int woff = (lane >> 4) * 16;
-    for (; p + 4 <= p_end; p += 4) {
-      bf16x8 a[8];
-      int4 wq[4];
-      float s2[4];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        wq[u] = *reinterpret_cast<const int4*>(&wrow[(p + u) * 64 + woff]);
-        s2[u] = bits2f(srow[((p + u) * 64) / G]);
-        a[2 * u] = *reinterpret_cast<const bf16x8*>(&xrow[(p + u) * 64 + ks]);
-        a[2 * u + 1] = *reinterpret_cast<const bf16x8*>(&xrow[(p + u) * 64 + 32 + ks]);
+    for (int k0 = p_begin * 64; k0 < p_end * 64; k0 += XT) {
+      const int tk = min(XT, p_end * 64 - k0);
+      __syncthreads();
+      for (int idx = threadIdx.x; idx < 16 * (tk / 8); idx += 256) {
+        const int r = idx / (tk / 8);
+        const int vec = idx % (tk / 8);
+        *reinterpret_cast<short8*>(&x_lds[r * SE + vec * 8]) =
+            *reinterpret_cast<const short8*>(
+                &x[(int64_t)min(r, M - 1) * K + k0 + vec * 8]);
       }
+      __syncthreads();
+      const short* arow = &x_lds[row * SE];
+      int pl = 0;
+      const int pl_end = tk / 64;
+      for (; pl + 4 <= pl_end; pl += 4) {
+        bf16x8 a[8];
+        int4 wq[4];
+        float s2[4];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int8_t* q = reinterpret_cast<const int8_t*>(&wq[u]);
-        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u], deq8(q, s2[u]),
-                                                       acc0, 0, 0, 0);
-        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u + 1],
-                                                       deq8(q + 8, s2[u]),
-                                                       acc1, 0, 0, 0);
+        for (int u = 0; u < 4; ++u) {
+          const int p = (k0 / 64) + pl + u;
+          wq[u] = *reinterpret_cast<const int4*>(&wrow[p * 64 + woff]);
+          s2[u] = bits2f(srow[(p * 64) / G]);
+          a[2 * u] = *reinterpret_cast<const bf16x8*>(&arow[(pl + u) * 64 + ks]);
+          a[2 * u + 1] = *reinterpret_cast<const bf16x8*>(&arow[(pl + u) * 64 + 32 + ks]);
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int8_t* q = reinterpret_cast<const int8_t*>(&wq[u]);
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u], deq8(q, s2[u]),
+                                                         acc0, 0, 0, 0);
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[2 * u + 1],
+                                                         deq8(q + 8, s2[u]),
+                                                         acc1, 0, 0, 0);
+        }
       }
-    }
-    for (; p < p_end; ++p) {
-      const int4 wq = *reinterpret_cast<const int4*>(&wrow[p * 64 + woff]);
-      const int8_t* q = reinterpret_cast<const int8_t*>(&wq);
-      const float sv = bits2f(srow[(p * 64) / G]);
-      const bf16x8 al = *reinterpret_cast<const bf16x8*>(&xrow[p * 64 + ks]);
-      const bf16x8 ah = *reinterpret_cast<const bf16x8*>(&xrow[p * 64 + 32 + ks]);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(al, deq8(q, sv), acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ah, deq8(q + 8, sv), acc1, 0, 0, 0);
+      for (; pl < pl_end; ++pl) {
+        const int p = (k0 / 64) + pl;
+        const int4 wq = *reinterpret_cast<const int4*>(&wrow[p * 64 + woff]);
+        const int8_t* q = reinterpret_cast<const int8_t*>(&wq);
+        const float sv = bits2f(srow[(p * 64) / G]);
+        const bf16x8 al = *reinterpret_cast<const bf16x8*>(&arow[pl * 64 + ks]);
+        const bf16x8 ah = *reinterpret_cast<const bf16x8*>(&arow[pl * 64 + 32 + ks]);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(al, deq8(q, sv), acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ah, deq8(q + 8, sv), acc1, 0, 0, 0);
+      }
     }
   } else if (QUANT) {
     const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
