@@ -1,0 +1,143 @@
+"""ClusterManager: discovery scan, health/profile/latency fan-out, topology
+solving (reference: src/dnet/api/cluster.py).
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Optional
+
+import httpx
+
+from ..core.types import LayerAssignment, TopologyInfo
+from ..models import ModelConfig, PRESETS, QuantConfig
+from ..parallel.profiler import DeviceProfile
+from ..parallel.solver import (compute_layer_assignments, halda_solve,
+                               postprocess_single_round)
+from ..utils.hostfile import DeviceProperties
+from ..utils.logger import get_logger
+
+log = get_logger("api")
+
+
+def estimate_layer_bytes(cfg: ModelConfig) -> int:
+    """Weight bytes per transformer layer for the configured quantization."""
+    c = cfg
+    per_elem = 1.0 if c.quant else 2.0
+    n_mlp = 2 * c.intermediate_size * c.hidden_size * 1.5  # gate+up+down
+    if c.num_experts:
+        inter = c.moe_intermediate_size or c.intermediate_size
+        n_mlp = c.num_experts * 3 * inter * c.hidden_size
+    n_attn = (c.qkv_out + c.num_q_heads * c.head_dim) * c.hidden_size
+    n = n_attn + n_mlp
+    scale_overhead = 1.02 if c.quant else 1.0
+    return int(n * per_elem * scale_overhead)
+
+
+class ClusterManager:
+    def __init__(self, discovery, solver_settings=None):
+        self.discovery = discovery
+        self.devices: dict[str, DeviceProperties] = {}
+        self.profiles: dict[str, DeviceProfile] = {}
+        self.topology: Optional[TopologyInfo] = None
+
+    async def scan_devices(self) -> dict[str, DeviceProperties]:
+        self.devices = await self.discovery.async_get_properties()
+        return self.devices
+
+    def shard_devices(self) -> list[DeviceProperties]:
+        return [d for d in self.devices.values() if not d.is_manager]
+
+    async def profile_cluster(self, parallel: bool = True) -> dict:
+        """Health-check then /profile each shard; merge latency medians."""
+        await self.scan_devices()
+        shards = self.shard_devices()
+        async with httpx.AsyncClient(timeout=120.0) as client:
+            healthy = []
+            for d in shards:
+                try:
+                    r = await client.get(
+                        f"http://{d.local_ip}:{d.server_port}/health")
+                    if r.status_code == 200:
+                        healthy.append(d)
+                except httpx.HTTPError:
+                    log.warning("shard %s unreachable", d.instance)
+            results = await asyncio.gather(*[
+                client.post(f"http://{d.local_ip}:{d.server_port}/profile")
+                for d in healthy], return_exceptions=True)
+            for d, r in zip(healthy, results):
+                if isinstance(r, Exception):
+                    continue
+                self.profiles[d.instance] = DeviceProfile.from_dict(r.json())
+            # latency sweep: ask the first shard to probe everyone
+            if len(healthy) > 1:
+                peers = [{"instance": d.instance, "host": d.local_ip,
+                          "port": d.shard_port} for d in healthy]
+                try:
+                    r = await client.post(
+                        f"http://{healthy[0].local_ip}:{healthy[0].server_port}"
+                        "/measure_latency",
+                        json={"peers": peers, "payload_sizes": [65536],
+                              "reps": 5})
+                    lat = r.json().get("latencies", {})
+                    for inst, sizes in lat.items():
+                        med = next(iter(sizes.values()), {}).get("median_ms")
+                        if inst in self.profiles and med is not None:
+                            self.profiles[inst].t_comm_ms = med
+                except httpx.HTTPError:
+                    pass
+        return {k: v.to_dict() for k, v in self.profiles.items()}
+
+    def get_head_node(self) -> Optional[DeviceProperties]:
+        """Owner of layer 0 (reference: cluster.py get_head_node)."""
+        if not self.topology or not self.topology.assignments:
+            return None
+        for a in self.topology.assignments:
+            if a.layers and a.layers[0] and a.layers[0][0] == 0:
+                return self.devices.get(a.instance)
+        return self.devices.get(self.topology.assignments[0].instance)
+
+    def solve_topology(self, model_id: str, cfg: ModelConfig,
+                       master_port: int = 29500, kv_bits: int = 16,
+                       batch: int = 1, seq_len: int = 4096) -> TopologyInfo:
+        shards = self.shard_devices()
+        assert shards, "no shard devices discovered"
+        profs = []
+        for d in shards:
+            p = self.profiles.get(d.instance)
+            if p is None:
+                p = DeviceProfile(instance=d.instance, hbm_gbps=5000.0,
+                                  h2d_gbps=50.0, hbm_free_gb=280.0)
+            profs.append(p)
+        lb = estimate_layer_bytes(cfg)
+        kv_per_layer = (2 * batch * cfg.num_kv_heads * seq_len * cfg.head_dim
+                        * (kv_bits / 8))
+        res = halda_solve(profs, cfg.num_layers, lb, kv_per_layer,
+                          kv_bits=kv_bits)
+        w = res.w
+        if res.k == 1:
+            w = postprocess_single_round(w, profs)
+        # drop devices with zero layers, keep order
+        active = [i for i, x in enumerate(w) if x > 0]
+        w_active = [w[i] for i in active]
+        assigns_lists = compute_layer_assignments(w_active, res.k,
+                                                  cfg.num_layers)
+        assignments = []
+        for j, i in enumerate(active):
+            d = shards[i]
+            nxt = shards[active[(j + 1) % len(active)]].instance
+            assignments.append(LayerAssignment(
+                instance=d.instance, layers=assigns_lists[j],
+                next_instance=nxt,
+                window_size=min(4, max(w_active[j], 1)),
+                residency_size=res.n[i], gpu_index=max(d.gpu_index, 0)))
+        head = assignments[0].instance if assignments else ""
+        head_dev = next((d for d in shards if d.instance == head), shards[0])
+        topo = TopologyInfo(
+            model=model_id, kv_bits=kv_bits, num_layers=cfg.num_layers,
+            devices=[a.instance for a in assignments],
+            assignments=assignments,
+            solution={"w": res.w, "n": res.n, "k": res.k,
+                      "obj_value_ms": res.obj_value, "sets": res.sets},
+            master_addr=head_dev.local_ip, master_port=master_port)
+        self.topology = topo
+        return topo

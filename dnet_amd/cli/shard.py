@@ -1,0 +1,45 @@
+"""dnet-shard entry point (reference: src/cli/shard.py).
+
+Usage:
+  python -m dnet_amd.cli.shard --name shard0 [--host 0.0.0.0]
+      [--http-port 8081] [--wire-port 50052]
+
+The driver loop (model + process group owner) runs on the main thread;
+HTTP control + wire data servers run on a background thread.
+"""
+from __future__ import annotations
+
+import argparse
+import faulthandler
+import signal
+
+from ..config import get_settings
+from ..utils.logger import get_logger
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser("dnet-shard")
+    s = get_settings()
+    ap.add_argument("--name", default="shard0")
+    ap.add_argument("--host", default=s.shard.host)
+    ap.add_argument("--http-port", type=int, default=s.shard.http_port)
+    ap.add_argument("--wire-port", type=int, default=s.shard.grpc_port)
+    args = ap.parse_args(argv)
+
+    from ..shard.runtime import ShardRuntime
+    from ..shard.server import start_servers
+
+    log = get_logger("shard")
+    faulthandler.register(signal.SIGUSR1)
+    rt = ShardRuntime(instance=args.name)
+    start_servers(rt, args.host, args.http_port, args.wire_port)
+    log.info("dnet-shard %s on http://%s:%d (wire %d)", args.name, args.host,
+             args.http_port, args.wire_port)
+    try:
+        rt.run()
+    except KeyboardInterrupt:
+        pass
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,125 @@
+"""Device + fabric profiler: feeds the layer-assignment solver.
+
+MI355X replacement of the reference's distilp device profiler + Thunderbolt
+latency sweep (reference: lib/distilp profile_device, shard /measure_latency
++ /profile): measures GEMM throughput, HBM read bandwidth, host->device
+(pinned) bandwidth and free HBM capacity on the local GPU, and — inside an
+active torch.distributed group — the xGMI p2p link latency/bandwidth per
+payload size with RCCL send/recv.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+
+import torch
+
+
+@dataclass
+class DeviceProfile:
+    instance: str = ""
+    device_name: str = ""
+    is_head: bool = False
+    gemm_tflops: float = 0.0
+    hbm_gbps: float = 0.0
+    h2d_gbps: float = 0.0
+    hbm_total_gb: float = 0.0
+    hbm_free_gb: float = 0.0
+    t_comm_ms: float = 0.0           # median link latency to peers
+    link_gbps: float = 0.0           # p2p bandwidth to next rank
+
+    def to_dict(self) -> dict:
+        return self.__dict__.copy()
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "DeviceProfile":
+        p = cls()
+        for k, v in d.items():
+            if hasattr(p, k):
+                setattr(p, k, v)
+        return p
+
+
+def _time_op(fn, reps=10, warmup=3, sync=True) -> float:
+    for _ in range(warmup):
+        fn()
+    if sync and torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        fn()
+    if sync and torch.cuda.is_available():
+        torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / reps
+
+
+def profile_device(instance: str = "", device: str = "cuda:0",
+                   quick: bool = False) -> DeviceProfile:
+    p = DeviceProfile(instance=instance)
+    if not torch.cuda.is_available():
+        p.device_name = "cpu"
+        p.gemm_tflops = 0.1
+        p.hbm_gbps = 10.0
+        p.h2d_gbps = 10.0
+        p.hbm_total_gb = 16.0
+        p.hbm_free_gb = 8.0
+        return p
+    dev = torch.device(device)
+    p.device_name = torch.cuda.get_device_name(dev)
+    free, total = torch.cuda.mem_get_info(dev)
+    p.hbm_total_gb = total / 1e9
+    p.hbm_free_gb = free / 1e9
+    n = 4096 if quick else 8192
+    a = torch.randn(n, n, dtype=torch.bfloat16, device=dev)
+    b = torch.randn(n, n, dtype=torch.bfloat16, device=dev)
+    t = _time_op(lambda: a @ b)
+    p.gemm_tflops = 2 * n ** 3 / t / 1e12
+    big = torch.empty(512 * 1024 * 1024, dtype=torch.uint8, device=dev)
+    dst = torch.empty_like(big)
+    t = _time_op(lambda: dst.copy_(big))
+    p.hbm_gbps = 2 * big.numel() / t / 1e9
+    host = torch.empty(256 * 1024 * 1024, dtype=torch.uint8, pin_memory=True)
+    hdst = torch.empty(host.numel(), dtype=torch.uint8, device=dev)
+    t = _time_op(lambda: hdst.copy_(host, non_blocking=True))
+    p.h2d_gbps = host.numel() / t / 1e9
+    del a, b, big, dst, hdst
+    torch.cuda.empty_cache()
+    return p
+
+
+def measure_ring_links(rank: int, world: int, device,
+                       sizes=(4 * 1024, 1024 * 1024, 16 * 1024 * 1024),
+                       reps: int = 20) -> dict:
+    """xGMI p2p sweep inside an active process group: rank r <-> r+1
+    ping-pong per payload size. Returns {size: {latency_ms, gbps}} for this
+    rank's link to the next rank. Collective — all ranks must call."""
+    import torch.distributed as dist
+    results = {}
+    nxt, prv = (rank + 1) % world, (rank - 1) % world
+    for size in sizes:
+        buf = torch.zeros(size, dtype=torch.uint8, device=device)
+        # warmup + timed ping-pong around the full ring
+        for _ in range(3):
+            _ring_pass(buf, rank, nxt, prv)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            _ring_pass(buf, rank, nxt, prv)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / reps
+        per_hop = dt / world
+        results[size] = {"latency_ms": per_hop * 1e3,
+                         "gbps": size / per_hop / 1e9}
+    return results
+
+
+def _ring_pass(buf, rank, nxt, prv):
+    import torch.distributed as dist
+    if rank == 0:
+        dist.send(buf, nxt)
+        dist.recv(buf, prv)
+    else:
+        dist.recv(buf, prv)
+        dist.send(buf, nxt)

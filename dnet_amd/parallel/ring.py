@@ -62,7 +62,7 @@ class RingExecutor:
                  device: torch.device, plan: Optional[RingPlan] = None,
                  mb_count: int = 1, mb_size: int = 1, smax: int = 4096,
                  seed: int = 0, decoding: Optional[DecodingConfig] = None,
-                 use_graphs: Optional[bool] = None):
+                 use_graphs: Optional[bool] = None, init_weights: bool = True):
         self.cfg = cfg
         self.rank = rank
         self.world = world
@@ -79,7 +79,8 @@ class RingExecutor:
         cls = get_ring_model(cfg.model_type)
         self.model = cls(cfg, self.my_layers, self.device, self.is_first,
                          self.is_last, smax=smax)
-        self.model.init_random(seed)
+        if init_weights:
+            self.model.init_random(seed)
         self.kvs = [KVCache(cfg, self.my_layers, mb_size, smax, self.device)
                     for _ in range(mb_count)]
         self.sampler = Sampler(decoding or DecodingConfig())
@@ -98,6 +99,8 @@ class RingExecutor:
             use_graphs = self.device.type == "cuda"
         self.use_graphs = use_graphs
         self._graphs: list = []
+        self.last_logprob = None   # last sample's logprobs (last rank only)
+        self.last_tops = None
 
     # ------------- one-rank step bodies (graph-capturable) -------------
 
@@ -230,3 +233,52 @@ class RingExecutor:
     def reset(self):
         for kv in self.kvs:
             kv.reset()
+
+    def set_decoding(self, cfg: DecodingConfig):
+        self.sampler = Sampler(cfg)
+
+    def decode_stream(self, max_tokens: int, stop_ids=(), on_token=None,
+                      mb: int = 0):
+        """Serving decode: one microbatch, token broadcast from the last rank
+        to ALL ranks each step (so every rank stops identically on EOS), and
+        ``on_token(step, tokens_tensor)`` called per step on every rank.
+
+        rank 0's tokbuf must hold the current token (set by prefill). The
+        first generated token is the prefill's sample; this generates up to
+        ``max_tokens - 1`` more.
+        """
+        import torch.distributed as dist
+        stop = torch.tensor(sorted(stop_ids), dtype=torch.int64,
+                            device=self.device) if stop_ids else None
+
+        def is_stop(tok: torch.Tensor) -> bool:
+            if stop is None or stop.numel() == 0:
+                return False
+            return bool(torch.isin(tok, stop).all())
+
+        produced = 0
+        for s in range(max_tokens - 1):
+            if self.is_first:
+                self._run_decode(mb)
+                if self.world > 1:
+                    self.ring.send(self.hbuf[mb])
+            elif self.world > 1:
+                self.ring.recv(self.hbuf[mb])
+                self._run_decode(mb)
+                if not self.is_last:
+                    self.ring.send(self.hbuf[mb])
+            if self.is_last:
+                tok, logprob, tops = self.sampler.sample(
+                    self.logits_buf[mb].float())
+                self.last_logprob, self.last_tops = logprob, tops
+                self.tokbuf[mb].copy_(tok)
+            if self.world > 1:
+                dist.broadcast(self.tokbuf[mb], src=self.world - 1)
+            self.kvs[mb].pos.add_(1)
+            produced += 1
+            done = is_stop(self.tokbuf[mb])
+            if on_token is not None:
+                on_token(s, self.tokbuf[mb], done or s == max_tokens - 2)
+            if done:
+                break
+        return produced

@@ -1,0 +1,135 @@
+"""Heterogeneity-aware layer-assignment solver (HALDA-style w/n/k).
+
+MI355X re-derivation of the reference's MILP layer distribution
+(reference: lib/distilp halda_solve -> HALDAResult{w, n, k, obj_value, sets};
+consumed in src/dnet/api/strategies/ring.py:59-65): per-device layer counts
+``w``, GPU-resident counts ``n`` and ring-round count ``k``, fed by the
+MI355X profiler's numbers (HBM bandwidth/capacity, host-DRAM H2D bandwidth,
+xGMI hop latency) instead of UMA/disk/Thunderbolt.
+
+Decode is weight-read bound, so per-layer cost on device i is
+    c_i = layer_bytes/hbm_bw_i              (resident layer)
+    c_i' = layer_bytes/h2d_bw_i             (host-swapped layer; copy-stream
+                                             overlap hides compute)
+The makespan objective min max_i t_i with identical layers is solved exactly
+by greedy list assignment on uniform machines; residency n_i then follows
+from HBM capacity, swapped layers re-priced at c_i', and a second greedy
+pass rebalances if swapping changed the ordering. k>1 (multiple ring rounds)
+is chosen so each round's window fits residency.
+"""
+from __future__ import annotations
+
+import heapq
+from dataclasses import dataclass, field
+
+from .profiler import DeviceProfile
+
+
+@dataclass
+class SolveResult:
+    w: list            # layers per device
+    n: list            # GPU-resident layers per device
+    k: int             # ring rounds
+    obj_value: float   # estimated per-token latency (ms)
+    sets: dict = field(default_factory=dict)   # device classes M1/M2/M3
+
+
+def halda_solve(devices: list[DeviceProfile], num_layers: int,
+                layer_bytes: float, kv_bytes_per_layer: float = 0.0,
+                overhead_gb: float = 4.0, hop_ms: float = 0.02,
+                kv_bits: int = 16) -> SolveResult:
+    """Assign ``num_layers`` identical layers across ``devices``.
+
+    layer_bytes: weight bytes per layer (after quantization).
+    kv_bytes_per_layer: KV cache bytes per layer for the planned batch/seq.
+    """
+    nd = len(devices)
+    assert nd > 0
+    c_res = []      # ms per resident layer
+    c_swap = []     # ms per host-swapped layer
+    cap_layers = []
+    for d in devices:
+        bw = max(d.hbm_gbps, 1.0) * 1e9
+        c_res.append(layer_bytes / bw * 1e3)
+        c_swap.append(layer_bytes / (max(d.h2d_gbps, 0.5) * 1e9) * 1e3)
+        usable = max(d.hbm_free_gb - overhead_gb, 0.5) * 1e9
+        cap_layers.append(max(int(usable // max(layer_bytes + kv_bytes_per_layer, 1)), 1))
+
+    def per_layer_cost(i: int, count: int) -> float:
+        """Average per-layer cost if device i holds `count` layers."""
+        resident = min(count, cap_layers[i])
+        swapped = count - resident
+        return (resident * c_res[i] + swapped * c_swap[i]) / max(count, 1)
+
+    # Greedy list scheduling on uniform machines: repeatedly give the next
+    # layer to the device whose finish time after taking it is smallest.
+    w = [0] * nd
+    heap = [(c_res[i], i) for i in range(nd)]
+    heapq.heapify(heap)
+    for _ in range(num_layers):
+        _, i = heapq.heappop(heap)
+        w[i] += 1
+        nxt = w[i] + 1
+        resident = min(nxt, cap_layers[i])
+        t_next = (min(nxt, cap_layers[i]) * c_res[i]
+                  + max(nxt - cap_layers[i], 0) * c_swap[i])
+        heapq.heappush(heap, (t_next, i))
+
+    n = [min(w[i], cap_layers[i]) for i in range(nd)]
+    # k rounds: if any device swaps, use enough rounds that one round's
+    # window fits residency (reference: k-round interleaving per prima.cpp).
+    k = 1
+    over = [w[i] / max(n[i], 1) for i in range(nd)]
+    if max(over) > 1.0:
+        k = min(int(max(over).__ceil__()), 8)
+
+    t_devices = [n[i] * c_res[i] + (w[i] - n[i]) * c_swap[i] for i in range(nd)]
+    obj = sum(t_devices) + hop_ms * nd * k
+    sets = {
+        "M1": [devices[i].instance for i in range(nd) if w[i] == n[i]],
+        "M2": [devices[i].instance for i in range(nd)
+               if n[i] < w[i] <= 2 * n[i]],
+        "M3": [devices[i].instance for i in range(nd) if w[i] > 2 * n[i]],
+    }
+    return SolveResult(w=w, n=n, k=k, obj_value=obj, sets=sets)
+
+
+def postprocess_single_round(w: list, devices: list[DeviceProfile]) -> list:
+    """Fold single-layer devices into their lighter neighbor when k == 1
+    (reference: src/dnet/api/utils.py postprocess_single_round)."""
+    if len(w) <= 1 or sum(1 for x in w if x > 0) <= 1:
+        return w
+    w = list(w)
+    while True:
+        try:
+            i = next(i for i, x in enumerate(w) if 0 < x <= 1 and len([y for y in w if y > 0]) > 1)
+        except StopIteration:
+            return w
+        others = [(w[j], j) for j in range(len(w)) if j != i and w[j] > 0]
+        if not others:
+            return w
+        _, j = min(others)
+        w[j] += w[i]
+        w[i] = 0
+
+
+def compute_layer_assignments(w: list, k: int, num_layers: int) -> list:
+    """Round-robin k rounds of w[i] layers per device -> per-device list of
+    per-round layer lists (reference: src/dnet/api/utils.py
+    compute_layer_assignments)."""
+    nd = len(w)
+    per_round = [[max(w[i] // k, 0) for i in range(nd)] for _ in range(k)]
+    # distribute remainders into the earliest rounds
+    for i in range(nd):
+        rem = w[i] - sum(pr[i] for pr in per_round)
+        for r in range(rem):
+            per_round[r % k][i] += 1
+    out = [[] for _ in range(nd)]
+    nxt = 0
+    for r in range(k):
+        for i in range(nd):
+            cnt = per_round[r][i]
+            out[i].append(list(range(nxt, min(nxt + cnt, num_layers))))
+            nxt += cnt
+    assert nxt == num_layers, f"assigned {nxt} != {num_layers}"
+    return out

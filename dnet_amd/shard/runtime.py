@@ -1,0 +1,363 @@
+"""Shard runtime: one process per GPU; driver thread owns all torch state.
+
+MI355X redesign of the reference's ShardRuntime + RingAdapter + policies
+(reference: src/dnet/shard/runtime.py, adapters/ring.py): instead of gRPC
+activation frames between asyncio workers, the shard joins a
+torch.distributed group (RCCL over xGMI on GPU, gloo on CPU rigs) at
+load_model time; rank 0 pulls inference requests from the wire-protocol
+data server and drives the ring with collective broadcasts, so every rank
+runs the same schedule. The reference's three-lock concurrency model is
+replaced by ONE driver thread that owns the model, the process group and
+the GPU (SURVEY.md §7 hard-part (3)).
+"""
+from __future__ import annotations
+
+import json
+import queue
+import socket
+import struct
+import threading
+import time
+from pathlib import Path
+from typing import Optional
+
+import msgpack
+import torch
+
+from ..config import get_settings
+from ..core.sampler import DecodingConfig
+from ..core.types import ShardLoadModelRequest
+from ..models import ModelConfig, PRESETS, QuantConfig
+from ..parallel.profiler import DeviceProfile, profile_device
+from ..parallel.ring import RingExecutor, RingPlan
+from ..utils.logger import get_logger
+from ..utils.model_meta import get_model_metadata, load_tensors
+
+log = get_logger("shard")
+
+CMD_NOOP, CMD_INFER, CMD_UNLOAD, CMD_SHUTDOWN = 0, 1, 2, 3
+
+
+class SyncWireClient:
+    """Blocking wire-protocol client for the driver thread (token path)."""
+
+    def __init__(self, host: str, port: int):
+        self.host, self.port = host, port
+        self.sock: Optional[socket.socket] = None
+
+    def send(self, frame: dict):
+        if self.sock is None:
+            self.sock = socket.create_connection((self.host, self.port),
+                                                 timeout=30)
+        body = msgpack.packb(frame, use_bin_type=True)
+        self.sock.sendall(struct.pack(">I", len(body)) + body)
+
+    def close(self):
+        if self.sock is not None:
+            try:
+                self.sock.close()
+            finally:
+                self.sock = None
+
+
+class ShardRuntime:
+    def __init__(self, instance: str = "shard0"):
+        self.instance = instance
+        self.settings = get_settings()
+        self.executor: Optional[RingExecutor] = None
+        self.load_req: Optional[ShardLoadModelRequest] = None
+        self.model_name: str = ""
+        self.ctrl_q: "queue.Queue[tuple]" = queue.Queue()
+        self.infer_q: "queue.Queue[dict]" = queue.Queue()
+        self._callback: Optional[SyncWireClient] = None
+        self._stop = threading.Event()
+        self.status = "idle"
+        self.last_error = ""
+
+    # ---------- control-plane entry points (called from HTTP threads) ----------
+
+    def submit_load(self, req: ShardLoadModelRequest) -> None:
+        done = threading.Event()
+        box: dict = {}
+        self.ctrl_q.put(("load", req, done, box))
+        done.wait(timeout=1800)
+        if box.get("error"):
+            raise RuntimeError(box["error"])
+
+    def submit_unload(self) -> None:
+        done = threading.Event()
+        box: dict = {}
+        self.ctrl_q.put(("unload", None, done, box))
+        done.wait(timeout=120)
+
+    def submit_infer(self, frame: dict) -> None:
+        if self.executor is None:
+            raise RuntimeError("no model loaded")
+        log.debug("infer queued nonce=%s prompt_len=%s", frame.get("nonce"),
+                 frame.get("prompt_len"))
+        self.infer_q.put(frame)
+
+    def shutdown(self) -> None:
+        self._stop.set()
+
+    def profile(self, quick: bool = True) -> DeviceProfile:
+        dev = "cuda:0" if torch.cuda.is_available() else "cpu"
+        return profile_device(self.instance, dev, quick=quick)
+
+    # ---------- model load / unload (driver thread) ----------
+
+    def _resolve_config(self, req: ShardLoadModelRequest) -> ModelConfig:
+        quant = None
+        if req.quant.startswith("int8"):
+            group = int(req.quant.split("-g")[1]) if "-g" in req.quant else 128
+            quant = QuantConfig(8, group)
+        p = Path(req.model_path).expanduser()
+        if (p / "config.json").exists():
+            return ModelConfig.from_hf(json.loads((p / "config.json").read_text()),
+                                       quant=quant)
+        name = req.model_name or req.model_path
+        if name in PRESETS:
+            return ModelConfig.from_hf(dict(PRESETS[name]), quant=quant)
+        raise FileNotFoundError(
+            f"model not found: {req.model_path!r} (no config.json, not a preset)")
+
+    def _load(self, req: ShardLoadModelRequest) -> None:
+        import torch.distributed as dist
+        cfg = self._resolve_config(req)
+        if torch.cuda.is_available():
+            device = torch.device(f"cuda:{req.gpu_index}")
+            torch.cuda.set_device(device)
+        else:
+            device = torch.device("cpu")
+        if req.world_size > 1 and not dist.is_initialized():
+            backend = "nccl" if device.type == "cuda" else "gloo"
+            dist.init_process_group(
+                backend=backend,
+                init_method=f"tcp://{req.master_addr}:{req.master_port}",
+                rank=req.rank, world_size=req.world_size)
+        plan = RingPlan([[sorted(req.layers)]] if req.world_size == 1 else None)
+        if req.world_size > 1:
+            # every rank gets the full plan via its own request's layer list;
+            # build a world-sized plan with only our slot filled (the executor
+            # only reads its own slot)
+            plan = RingPlan([[[]] for _ in range(req.world_size)])
+            plan.assignments[req.rank] = [sorted(req.layers)]
+        synthetic = not (Path(req.model_path).expanduser() / "config.json").exists()
+        ex = RingExecutor(cfg, req.rank, req.world_size, device, plan=plan,
+                          mb_count=1, mb_size=req.max_batch,
+                          smax=req.max_seq,
+                          use_graphs=(device.type == "cuda"
+                                      and self.settings.compute.use_graphs),
+                          init_weights=synthetic)
+        if not synthetic:
+            self._load_weights(ex, req)
+        self.executor = ex
+        self.load_req = req
+        self.model_name = req.model_name or req.model_path
+        if ex.is_last and req.api_callback_address:
+            host, _, port = req.api_callback_address.rpartition(":")
+            self._callback = SyncWireClient(host or "127.0.0.1", int(port))
+        self.status = "loaded"
+        log.info("model loaded: %s rank %d/%d layers %s", self.model_name,
+                 req.rank, req.world_size, req.layers[:4])
+
+    def _load_weights(self, ex: RingExecutor, req: ShardLoadModelRequest) -> None:
+        meta = get_model_metadata(req.model_path)
+        names: list[str] = []
+        for lid in ex.my_layers:
+            names += meta.layers.get(lid, [])
+        if ex.is_first:
+            names += meta.embed
+        if ex.is_last:
+            names += meta.final_norm + meta.lm_head + meta.embed
+        sd = load_tensors(meta, sorted(set(names)))
+        ex.model.load_state_dict(sd)
+
+    def _unload(self) -> None:
+        import torch.distributed as dist
+        self.executor = None
+        self.load_req = None
+        self.model_name = ""
+        if self._callback:
+            self._callback.close()
+            self._callback = None
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+        self.status = "idle"
+
+    # ---------- the driver loop ----------
+
+    def run(self) -> None:
+        """Main driver loop — owns the model and the process group."""
+        while not self._stop.is_set():
+            try:
+                kind, arg, done, box = self.ctrl_q.get(timeout=0.02)
+                try:
+                    if kind == "load":
+                        self._load(arg)
+                    elif kind == "unload":
+                        self._broadcast_cmd(CMD_UNLOAD)
+                        self._unload()
+                except Exception as e:  # surface to the HTTP caller
+                    log.exception("control command failed")
+                    box["error"] = str(e)
+                    self.last_error = str(e)
+                finally:
+                    done.set()
+                continue
+            except queue.Empty:
+                pass
+            ex = self.executor
+            if ex is None:
+                continue
+            if ex.rank == 0:
+                try:
+                    frame = self.infer_q.get(timeout=0.02)
+                except queue.Empty:
+                    if ex.world > 1:
+                        self._broadcast_cmd(CMD_NOOP)
+                    continue
+                try:
+                    log.debug("running infer nonce=%s", frame.get("nonce"))
+                    self._run_infer_rank0(frame)
+                except Exception:
+                    log.exception("inference failed")
+                    self._send_error(frame.get("nonce", ""))
+            else:
+                cmd = self._recv_cmd()
+                if cmd[0] == CMD_INFER:
+                    self._run_infer_follower(cmd)
+                elif cmd[0] == CMD_UNLOAD:
+                    self._unload()
+
+    # ---------- collective command plumbing ----------
+
+    _CMD_LEN = 16
+
+    def _cmd_tensor(self, vals=()):
+        t = torch.zeros(self._CMD_LEN, dtype=torch.float64,
+                        device=self._comm_device())
+        for i, v in enumerate(vals):
+            t[i] = float(v)
+        return t
+
+    def _comm_device(self):
+        ex = self.executor
+        return ex.device if (ex and ex.device.type == "cuda") else torch.device("cpu")
+
+    def _broadcast_cmd(self, *vals):
+        import torch.distributed as dist
+        if self.executor is None or self.executor.world == 1:
+            return
+        t = self._cmd_tensor(vals)
+        dist.broadcast(t, src=0)
+
+    def _recv_cmd(self):
+        import torch.distributed as dist
+        t = self._cmd_tensor()
+        dist.broadcast(t, src=0)
+        return [t[i].item() for i in range(self._CMD_LEN)]
+
+    # ---------- inference ----------
+
+    def _run_infer_rank0(self, frame: dict) -> None:
+        import numpy as np
+        ex = self.executor
+        tokens = torch.from_numpy(
+            np.frombuffer(frame["tokens"], dtype=np.int32).copy()).long()
+        T = int(frame.get("prompt_len", tokens.numel()))
+        tokens = tokens.view(1, 1, T)
+        p = frame.get("params", {})
+        max_tokens = int(frame.get("max_tokens", 128))
+        stop_ids = list(frame.get("stop_ids", []))
+        nonce = frame.get("nonce", "")
+        nonce_ids = list(nonce.encode("utf-8"))[:64]
+        if ex.world > 1:
+            self._broadcast_cmd(CMD_INFER, T, max_tokens, len(stop_ids),
+                                p.get("temperature", 0.0), p.get("top_p", 1.0),
+                                p.get("top_k", 0), p.get("min_p", 0.0),
+                                int(p.get("logprobs", False)),
+                                int(p.get("top_logprobs", 0)), len(nonce_ids))
+            import torch.distributed as dist
+            payload = torch.cat([
+                tokens.flatten().to(self._comm_device()),
+                torch.tensor(stop_ids + nonce_ids, dtype=torch.int64,
+                             device=self._comm_device())])
+            dist.broadcast(payload, src=0)
+        self._execute_infer(nonce, tokens, max_tokens, stop_ids, p)
+
+    def _run_infer_follower(self, cmd) -> None:
+        import torch.distributed as dist
+        ex = self.executor
+        T, max_tokens, n_stop = int(cmd[1]), int(cmd[2]), int(cmd[3])
+        n_nonce = int(cmd[10])
+        p = {"temperature": cmd[4], "top_p": cmd[5], "top_k": int(cmd[6]),
+             "min_p": cmd[7], "logprobs": bool(cmd[8]),
+             "top_logprobs": int(cmd[9])}
+        payload = torch.zeros(T + n_stop + n_nonce, dtype=torch.int64,
+                              device=self._comm_device())
+        dist.broadcast(payload, src=0)
+        tokens = payload[:T].view(1, 1, T).cpu()
+        stop_ids = payload[T:T + n_stop].tolist()
+        nonce = bytes(payload[T + n_stop:].tolist()).decode("utf-8", "replace")
+        self._execute_infer(nonce, tokens, max_tokens, stop_ids, p)
+
+    def _execute_infer(self, nonce: str, tokens: torch.Tensor,
+                       max_tokens: int, stop_ids: list, p: dict) -> None:
+        ex = self.executor
+        ex.reset()
+        ex.set_decoding(DecodingConfig(
+            temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
+            top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0),
+            logprobs=bool(p.get("logprobs", False)),
+            top_logprobs=int(p.get("top_logprobs", 0))))
+        # pad the single sequence to the executor's batch width
+        B = ex.mb_size
+        toks = tokens.expand(1, B, tokens.shape[-1]).contiguous().to(ex.device)
+        first = ex.prefill(toks)
+        # every rank must agree on EOS-after-first-token before entering the
+        # collective decode loop; rank 0 holds the first token in tokbuf.
+        tok0_t = ex.tokbuf[0][:1].clone()
+        if ex.world > 1:
+            import torch.distributed as dist
+            dist.broadcast(tok0_t, src=0)
+        tok0 = int(tok0_t[0])
+        done0 = tok0 in stop_ids or max_tokens <= 1
+        if ex.is_last:
+            self._emit_token(nonce, tok0, finished=done0)
+        if done0:
+            return
+
+        def on_token(step, tokt, last):
+            if ex.is_last:
+                self._emit_token(nonce, int(tokt[0]),
+                                 finished=last or int(tokt[0]) in stop_ids)
+
+        ex.decode_stream(max_tokens, stop_ids=stop_ids, on_token=on_token)
+
+    def _emit_token(self, nonce: str, token_id: int, finished: bool = False):
+        ex = self.executor
+        if self._callback is None:
+            return
+        frame = {"t": "token", "nonce": nonce, "token_id": token_id,
+                 "ts_ms": int(time.time() * 1e3), "finished": finished}
+        if ex.last_logprob is not None:
+            frame["logprob"] = float(ex.last_logprob[0])
+        if ex.last_tops is not None:
+            frame["top_logprobs"] = ex.last_tops[0]
+        try:
+            self._callback.send(frame)
+        except OSError:
+            log.warning("token callback failed (api down?)")
+            self._callback.close()
+
+    def _send_error(self, nonce: str):
+        if self._callback is not None:
+            try:
+                self._callback.send({"t": "error", "nonce": nonce,
+                                     "failed_node": self.instance,
+                                     "code": 500, "error": self.last_error})
+            except OSError:
+                pass

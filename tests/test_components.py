@@ -1,0 +1,167 @@
+"""Unit tests: config tree, hostfile discovery, solver, wire protocol,
+tokenizer/detokenizer, repack, metadata parsing."""
+import asyncio
+import json
+import os
+
+import pytest
+import torch
+
+from dnet_amd.config import get_settings, reset_settings
+from dnet_amd.parallel.profiler import DeviceProfile
+from dnet_amd.parallel.solver import (compute_layer_assignments, halda_solve,
+                                      postprocess_single_round)
+from dnet_amd.utils.hostfile import StaticDiscovery, load_hostfile
+
+
+def test_settings_env_override(monkeypatch):
+    monkeypatch.setenv("DNET_API_PORT", "9999")
+    monkeypatch.setenv("DNET_TRANSPORT_COMPRESS", "true")
+    reset_settings()
+    s = get_settings()
+    assert s.api.port == 9999
+    assert s.transport.compress is True
+    monkeypatch.delenv("DNET_API_PORT")
+    monkeypatch.delenv("DNET_TRANSPORT_COMPRESS")
+    reset_settings()
+
+
+def test_hostfile_ssh_style(tmp_path):
+    hf = tmp_path / "hosts"
+    hf.write_text("# comment\nshard0 127.0.0.1 8081 50052 0\n"
+                  "shard1 127.0.0.1 8181 50152 1\n")
+    devs = load_hostfile(str(hf))
+    assert [d.instance for d in devs] == ["shard0", "shard1"]
+    assert devs[1].gpu_index == 1
+
+
+def test_hostfile_json(tmp_path):
+    hf = tmp_path / "hosts.json"
+    hf.write_text(json.dumps([
+        {"name": "a", "ip": "10.0.0.1", "http_port": 1, "grpc_port": 2}]))
+    devs = load_hostfile(str(hf))
+    assert devs[0].local_ip == "10.0.0.1"
+
+
+def test_static_discovery(tmp_path):
+    hf = tmp_path / "hosts"
+    hf.write_text("shard0 127.0.0.1 8081 50052\n")
+    d = StaticDiscovery(str(hf), own_instance="api", own_http_port=8080,
+                        own_grpc_port=50051)
+    props = asyncio.run(d.async_get_properties())
+    assert set(props) == {"shard0", "api"}
+
+
+def _profiles(n, bw=None):
+    bw = bw or [5000.0] * n
+    return [DeviceProfile(instance=f"s{i}", hbm_gbps=bw[i], h2d_gbps=50.0,
+                          hbm_free_gb=280.0) for i in range(n)]
+
+
+def test_solver_homogeneous():
+    res = halda_solve(_profiles(4), 64, 500e6)
+    assert sorted(res.w) == [16, 16, 16, 16]
+    assert res.k == 1 and res.n == res.w
+
+
+def test_solver_heterogeneous():
+    res = halda_solve(_profiles(2, bw=[6000.0, 2000.0]), 64, 500e6)
+    assert res.w[0] > res.w[1]
+    assert sum(res.w) == 64
+
+
+def test_solver_capacity_offload():
+    profs = _profiles(1)
+    profs[0].hbm_free_gb = 10.0   # fits ~12 layers of 0.5 GB after overhead
+    res = halda_solve(profs, 64, 500e6)
+    assert res.n[0] < res.w[0] == 64
+    assert res.k > 1
+    assert profs[0].instance in res.sets["M3"]
+
+
+def test_postprocess_single_round():
+    profs = _profiles(3)
+    assert postprocess_single_round([10, 1, 9], profs) == [10, 0, 10]
+
+
+def test_compute_layer_assignments():
+    out = compute_layer_assignments([3, 2], 1, 5)
+    assert out == [[[0, 1, 2]], [[3, 4]]]
+    out2 = compute_layer_assignments([2, 2], 2, 4)
+    flat = [l for dev in out2 for r in dev for l in r]
+    assert sorted(flat) == [0, 1, 2, 3]
+
+
+def test_wire_roundtrip():
+    from dnet_amd.protos.wire import WireClient, WireServer
+
+    async def run():
+        seen = []
+
+        async def handler(frame, writer):
+            seen.append(frame)
+            return {"t": "pong", "echo": frame.get("x")}
+
+        srv = WireServer("127.0.0.1", 29877, handler)
+        await srv.start()
+        cli = WireClient("127.0.0.1", 29877)
+        resp = await cli.request({"t": "ping", "x": 42,
+                                  "blob": b"\x00\x01" * 100})
+        assert resp["echo"] == 42
+        assert seen[0]["blob"] == b"\x00\x01" * 100
+        await cli.close()
+        await srv.stop()
+
+    asyncio.run(run())
+
+
+def test_byte_tokenizer_roundtrip():
+    from dnet_amd.api.tokenizer import ByteTokenizer, Detokenizer
+    t = ByteTokenizer(512)
+    ids = t.encode("hello world")
+    assert ids[0] == t.BOS
+    assert t.decode(ids) == "hello world"
+    d = Detokenizer(t)
+    out = "".join(d.add_token(i) for i in ids)
+    assert out == "hello world"
+
+
+def test_metadata_and_repack(tmp_path, monkeypatch):
+    from safetensors.torch import save_file
+
+    from dnet_amd.utils.model_meta import get_model_metadata
+    from dnet_amd.utils.repack import (delete_repacked_layers,
+                                       ensure_repacked_for_layers)
+    mdir = tmp_path / "model"
+    mdir.mkdir()
+    (mdir / "config.json").write_text(json.dumps({"model_type": "llama"}))
+    sd = {
+        "model.embed_tokens.weight": torch.randn(8, 4),
+        "model.layers.0.self_attn.q_proj.weight": torch.randn(4, 4),
+        "model.layers.1.self_attn.q_proj.weight": torch.randn(4, 4),
+        "model.norm.weight": torch.randn(4),
+    }
+    save_file(sd, str(mdir / "model.safetensors"))
+    meta = get_model_metadata(str(mdir))
+    assert meta.num_layers == 2
+    assert meta.layer_bytes(0) == 4 * 4 * 4
+    assert meta.embed and meta.final_norm
+    monkeypatch.setenv("DNET_STORAGE_REPACK_DIR", str(tmp_path / "repack"))
+    reset_settings()
+    out = ensure_repacked_for_layers(str(mdir), "m", [0])
+    assert (out / "layer_0000.safetensors").exists()
+    assert (out / "repack-manifest.json").exists()
+    # idempotent
+    out2 = ensure_repacked_for_layers(str(mdir), "m", [0])
+    assert out2 == out
+    assert delete_repacked_layers("m") == 1
+    reset_settings()
+
+
+def test_estimate_layer_bytes():
+    from dnet_amd.api.cluster import estimate_layer_bytes
+    from dnet_amd.models import ModelConfig, PRESETS, QuantConfig
+    cfg = ModelConfig.from_hf(dict(PRESETS["qwen-2.5-32b"]),
+                              quant=QuantConfig(8, 128))
+    b = estimate_layer_bytes(cfg)
+    assert 400e6 < b < 600e6  # ~487 MB per layer int8
