@@ -45,6 +45,9 @@ def main():
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (debug only; invalid for scoring)")
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--residency", type=int, default=0,
+                    help="GPU-resident layers per rank (0=all; <local layers "
+                         "enables host-DRAM weight streaming)")
     args = ap.parse_args()
 
     rank, world, device = init_from_env()
@@ -59,7 +62,8 @@ def main():
     mb_count = max(args.mb_per_rank * world, 1)
     ex = RingExecutor(cfg, rank, world, device, mb_count=mb_count,
                       mb_size=args.mb_size, smax=args.smax, seed=1234,
-                      use_graphs=on_gpu and not args.no_graphs)
+                      use_graphs=on_gpu and not args.no_graphs,
+                      residency=args.residency)
 
     g = torch.Generator().manual_seed(7)
     tokens = torch.randint(0, cfg.vocab_size,
@@ -120,7 +124,9 @@ def main():
                 "global_batch": total_seqs,
                 "seq_len": args.prompt_len,
                 "gen_len": args.steps,
-                "parallelism": f"ring-pp{world}",
+                "parallelism": f"ring-pp{world}"
+                               + (f"+offload(res={args.residency})"
+                                  if args.residency else ""),
                 "microbatches": mb_count,
                 "mb_size": args.mb_size,
                 "layers": cfg.num_layers,

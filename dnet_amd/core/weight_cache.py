@@ -1,0 +1,131 @@
+"""Windowed weight residency: pinned-host layer store + HBM slot cache.
+
+MI355X redesign of the reference's WeightCache/LayerManager (reference:
+src/dnet/core/memory/weight_cache.py + utils/layer_manager.py): instead of
+disk->UMA mmap/madvise streaming, quantized layer weights live in PINNED
+host DRAM and stream into a ring of pre-allocated HBM slots on a dedicated
+copy stream (async hipMemcpy overlapped with compute); the compute stream
+waits on a per-slot event before binding. The reference's in-flight-Future
+and three-lock design collapses to single-owner semantics: the driver
+thread is the only caller (SURVEY.md §7 hard-part (3)).
+
+Emits the reference's [PROFILE][MATERIALIZE]/[PREFETCH] log protocol.
+"""
+from __future__ import annotations
+
+import time
+from typing import Callable, Optional
+
+import torch
+
+from ..utils.logger import get_logger
+
+log = get_logger("weights")
+
+
+class PinnedLayerStore:
+    """Per-layer weight tensors in pinned host memory (CPU fallback: plain
+    CPU tensors). Tensors are stored in their compute format (int8+scales
+    packed, bf16 norms/bias) so the H2D copy is the only work."""
+
+    def __init__(self, pin: bool):
+        self.pin = pin and torch.cuda.is_available()
+        self.layers: dict[int, dict[str, torch.Tensor]] = {}
+
+    def put_layer(self, lid: int, tensors: dict[str, torch.Tensor]):
+        out = {}
+        for k, t in tensors.items():
+            t = t.detach().cpu().contiguous()
+            if self.pin:
+                t = t.pin_memory()
+            out[k] = t
+        self.layers[lid] = out
+
+    def layer_bytes(self, lid: int) -> int:
+        return sum(t.numel() * t.element_size()
+                   for t in self.layers[lid].values())
+
+
+class WeightCache:
+    """N HBM slots over a pinned-host layer store, ring-prefetched.
+
+    All layers must share one tensor-shape template (true for the uniform
+    transformer stack). ``bind(lid)`` returns the device tensors for the
+    layer, blocking the CURRENT stream (not the host) on the slot's copy
+    event; ``prefetch(lid)`` schedules the H2D copy on the copy stream.
+    """
+
+    def __init__(self, store: PinnedLayerStore, residency: int,
+                 device: torch.device):
+        self.store = store
+        self.device = device
+        self.on_gpu = device.type == "cuda"
+        self.residency = max(residency, 2)
+        template = next(iter(store.layers.values()))
+        self.slots: list[dict[str, torch.Tensor]] = []
+        for _ in range(self.residency):
+            self.slots.append({k: torch.empty_like(t, device=device)
+                               for k, t in template.items()})
+        self.slot_layer: list[Optional[int]] = [None] * self.residency
+        self.slot_event: list = [None] * self.residency
+        self.layer_slot: dict[int, int] = {}
+        self._clock = 0
+        self._use_tick: list[int] = [0] * self.residency
+        self.copy_stream = torch.cuda.Stream(device) if self.on_gpu else None
+        self.hits = 0
+        self.misses = 0
+
+    def _pick_slot(self) -> int:
+        # LRU over slots (sequential layer access -> effectively a ring)
+        i = min(range(self.residency), key=lambda s: self._use_tick[s])
+        old = self.slot_layer[i]
+        if old is not None:
+            self.layer_slot.pop(old, None)
+        return i
+
+    def _copy_into(self, slot: int, lid: int):
+        src = self.store.layers[lid]
+        dst = self.slots[slot]
+        t0 = time.perf_counter()
+        if self.on_gpu:
+            with torch.cuda.stream(self.copy_stream):
+                for k, t in src.items():
+                    dst[k].copy_(t, non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record(self.copy_stream)
+            self.slot_event[slot] = ev
+        else:
+            for k, t in src.items():
+                dst[k].copy_(t)
+            self.slot_event[slot] = None
+        self.slot_layer[slot] = lid
+        self.layer_slot[lid] = slot
+        mb = self.store.layer_bytes(lid) / 1e6
+        log.info("[PROFILE][PREFETCH] layer=%d ms=%.2f bytes=%.1fMB (issued)",
+                 lid, (time.perf_counter() - t0) * 1e3, mb)
+
+    def prefetch(self, lid: int):
+        if lid in self.layer_slot:
+            return
+        self._copy_into(self._pick_slot(), lid)
+
+    def bind(self, lid: int) -> dict[str, torch.Tensor]:
+        """Device tensors for layer ``lid``; current stream waits on the
+        in-flight copy if needed."""
+        t0 = time.perf_counter()
+        slot = self.layer_slot.get(lid)
+        if slot is None:
+            self.misses += 1
+            slot = self._pick_slot()
+            self._copy_into(slot, lid)
+        else:
+            self.hits += 1
+        self._clock += 1
+        self._use_tick[slot] = self._clock
+        ev = self.slot_event[slot]
+        if ev is not None:
+            torch.cuda.current_stream(self.device).wait_event(ev)
+        wait_ms = (time.perf_counter() - t0) * 1e3
+        if wait_ms > 1.0:
+            log.info("[PROFILE][WAIT-WEIGHT] layer=%d ms=%.2f", lid, wait_ms)
+        return self.slots[slot]

@@ -105,6 +105,40 @@ class LayerWeights:
     experts_gateup: Optional[list] = None   # list[Linear] per expert
     experts_down: Optional[list] = None
 
+    _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm")
+    _LINEAR_FIELDS = ("qkv", "o", "gateup", "down")
+
+    def to_tensor_dict(self) -> dict:
+        """Flat name->tensor dict (weight-cache slot format)."""
+        out = {}
+        for f in self._TENSOR_FIELDS:
+            t = getattr(self, f)
+            if t is not None:
+                out[f] = t
+        for f in self._LINEAR_FIELDS:
+            l = getattr(self, f)
+            if l is None:
+                continue
+            out[f + ".w"] = l.w
+            if l.scales is not None:
+                out[f + ".scales"] = l.scales
+            if l.bias is not None:
+                out[f + ".bias"] = l.bias
+        return out
+
+    @classmethod
+    def from_tensor_dict(cls, d: dict, group: int, packed: bool) -> "LayerWeights":
+        lw = cls()
+        for f in cls._TENSOR_FIELDS:
+            if f in d:
+                setattr(lw, f, d[f])
+        for f in cls._LINEAR_FIELDS:
+            if f + ".w" in d:
+                setattr(lw, f, Linear(d[f + ".w"], d.get(f + ".bias"),
+                                      d.get(f + ".scales"), group,
+                                      packed and (f + ".scales") in d))
+        return lw
+
     def nbytes(self) -> int:
         n = 0
         for t in (self.attn_norm, self.mlp_norm, self.q_norm, self.k_norm):
@@ -202,6 +236,14 @@ class RingModel:
                                    scaling=cfg.rope_scaling)
         self.cos = cos.to(self.device)
         self.sin = sin.to(self.device)
+        # offload policy hook: when set, layer weights come from the
+        # windowed weight cache instead of self.layers
+        self.weight_provider = None  # Callable[[int], LayerWeights] | None
+
+    def _layer(self, lid: int) -> "LayerWeights":
+        if self.weight_provider is not None:
+            return self.weight_provider(lid)
+        return self.layers[lid]
 
     # ---------- weight init / loading ----------
 
@@ -333,7 +375,7 @@ class RingModel:
         len_t = kv.pos + 1  # attend over lengths including the token being written
         delta = None
         for lid in layer_ids:
-            lw = self.layers[lid]
+            lw = self._layer(lid)
             y = ops.rmsnorm(delta if delta is not None else h,
                             h if delta is not None else None,
                             lw.attn_norm, c.rms_eps)
@@ -366,7 +408,7 @@ class RingModel:
         nq, nkv, d = c.num_q_heads, c.num_kv_heads, c.head_dim
         positions = torch.arange(p0, p0 + T, device=h.device)
         for lid in layer_ids:
-            lw = self.layers[lid]
+            lw = self._layer(lid)
             flat = h.view(B * T, H)
             y = ops.rmsnorm(flat, None, lw.attn_norm, c.rms_eps)
             qkv = lw.qkv(y)

@@ -62,7 +62,8 @@ class RingExecutor:
                  device: torch.device, plan: Optional[RingPlan] = None,
                  mb_count: int = 1, mb_size: int = 1, smax: int = 4096,
                  seed: int = 0, decoding: Optional[DecodingConfig] = None,
-                 use_graphs: Optional[bool] = None, init_weights: bool = True):
+                 use_graphs: Optional[bool] = None, init_weights: bool = True,
+                 residency: int = 0):
         self.cfg = cfg
         self.rank = rank
         self.world = world
@@ -81,6 +82,11 @@ class RingExecutor:
                          self.is_last, smax=smax)
         if init_weights:
             self.model.init_random(seed)
+        self.weight_cache = None
+        if init_weights and 0 < residency < len(self.my_layers):
+            from ..shard.policies import enable_offload  # lazy: avoids cycle
+            self.weight_cache = enable_offload(self.model, residency)
+            use_graphs = False  # slot addresses change per step
         self.kvs = [KVCache(cfg, self.my_layers, mb_size, smax, self.device)
                     for _ in range(mb_count)]
         self.sampler = Sampler(decoding or DecodingConfig())

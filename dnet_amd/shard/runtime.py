@@ -143,14 +143,21 @@ class ShardRuntime:
             plan = RingPlan([[[]] for _ in range(req.world_size)])
             plan.assignments[req.rank] = [sorted(req.layers)]
         synthetic = not (Path(req.model_path).expanduser() / "config.json").exists()
+        residency = req.residency_size if \
+            0 < req.residency_size < len(req.layers) else 0
         ex = RingExecutor(cfg, req.rank, req.world_size, device, plan=plan,
                           mb_count=1, mb_size=req.max_batch,
                           smax=req.max_seq,
                           use_graphs=(device.type == "cuda"
-                                      and self.settings.compute.use_graphs),
-                          init_weights=synthetic)
+                                      and self.settings.compute.use_graphs
+                                      and residency == 0),
+                          init_weights=synthetic, residency=residency)
         if not synthetic:
             self._load_weights(ex, req)
+            if residency:
+                from .policies import enable_offload
+                ex.weight_cache = enable_offload(ex.model, residency)
+                ex.use_graphs = False
         self.executor = ex
         self.load_req = req
         self.model_name = req.model_name or req.model_path

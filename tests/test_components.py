@@ -165,3 +165,30 @@ def test_estimate_layer_bytes():
                               quant=QuantConfig(8, 128))
     b = estimate_layer_bytes(cfg)
     assert 400e6 < b < 600e6  # ~487 MB per layer int8
+
+
+def test_offload_policy_matches_fit():
+    """Offload (windowed weight cache) must produce the same tokens as the
+    fully-resident path for identical weights."""
+    from dnet_amd.models import ModelConfig, PRESETS
+    from dnet_amd.parallel.ring import RingExecutor
+    from dnet_amd.shard.policies import plan_policy
+
+    assert plan_policy(4, 2, 4) == "fit"
+    assert plan_policy(8, 2, 4) == "offload"
+    assert plan_policy(8, 4, 2) == "sliding_fit"
+
+    cfg = ModelConfig.from_hf(dict(PRESETS["tiny"]))
+    toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
+                         generator=torch.Generator().manual_seed(3))
+
+    def run(residency):
+        ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=2, smax=32,
+                          seed=11, use_graphs=False, residency=residency)
+        first = ex.prefill(toks.clone())
+        gen = ex.decode_rounds(4)
+        return torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+    fit = run(0)
+    off = run(2)
+    assert torch.equal(fit, off)

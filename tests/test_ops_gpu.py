@@ -157,3 +157,21 @@ def test_gemm_m16_packed(m, n, k, group):
     wd = ops.dequant_int8(qp, scales, group, packed=True)
     wd_ref = ref.dequant_int8(q.cpu(), scales.cpu(), group)
     assert torch.allclose(wd.float().cpu(), wd_ref.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_offload_gpu_matches_fit():
+    """GPU: windowed host->HBM weight streaming == fully resident."""
+    from dnet_amd.models import ModelConfig, PRESETS, QuantConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    cfg = ModelConfig.from_hf(dict(PRESETS["tiny"]), quant=QuantConfig(8, 64))
+    toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
+                         generator=torch.Generator().manual_seed(3)).cuda()
+
+    def run(residency):
+        ex = RingExecutor(cfg, 0, 1, "cuda:0", mb_count=1, mb_size=2, smax=32,
+                          seed=11, use_graphs=False, residency=residency)
+        first = ex.prefill(toks.clone())
+        gen = ex.decode_rounds(4)
+        return torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+    assert torch.equal(run(0), run(2))
