@@ -37,9 +37,9 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   const int wave = threadIdx.x / kWave;
   const int lane = threadIdx.x & (kWave - 1);
   const int n0 = (blockIdx.x * 4 + wave) * 16;
-  // PACKED path has block-wide barriers: OOB waves must stay alive (their
-  // stores are skipped); the barrier-free paths may exit early.
-  if (!PACKED && n0 >= N) return;
+  // LDS-staged paths have block-wide barriers: OOB waves must stay alive
+  // (their stores are skipped); only the barrier-free path may exit early.
+  if (QUANT && !PACKED && n0 >= N) return;
   const int row = lane & 15;          // A row (x row = output m), B col
   const int ks = (lane >> 4) * 8;     // k-offset of this lane's 8-elem slice
   const int n_w = min(n0 + row, N - 1);     // this lane's W row
@@ -58,7 +58,7 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   // 4-chunk unrolled main loop: all 8-12 loads issue before the first
   // dequant+MFMA, so ~4 HBM loads stay in flight per wave (the single-chunk
   // loop was load-latency-bound at ~1 chunk / 970 cycles).
-  if (QUANT && PACKED) {
+  if constexpr (QUANT && PACKED) {
     // W stored in MFMA chunk-pair order (pack_int8_mfma): one b128 load per
     // lane covers its B slices of two adjacent chunks -> full 64 B bursts.
     // x is staged tile-wise into LDS with coalesced full-line loads (the
@@ -121,7 +121,7 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
         acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ah, deq8(q + 8, sv), acc1, 0, 0, 0);
       }
     }
-  } else if (QUANT) {
+  } else if constexpr (QUANT) {
     const int8_t* wrow = (const int8_t*)w + (int64_t)n_w * K;
     const short* srow = scales + (int64_t)n_w * (K / G);
     int c = c_begin;
@@ -160,29 +160,46 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
     }
   } else {
+    // bf16 path: same LDS x-staging as the packed int8 path (direct
+    // fragment-shaped x reads touch 16 scattered rows per instruction).
+    constexpr int XT = 1024;
+    constexpr int SE = XT + 8;
+    __shared__ short x_lds_b[16 * SE];
     const short* wrow = (const short*)w + (int64_t)n_w * K;
-    int c = c_begin;
-    for (; c + 4 <= c_end; c += 4) {
-      bf16x8 a[4], b[4];
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int k = (c + u) * 32 + ks;
-        a[u] = *reinterpret_cast<const bf16x8*>(&xrow[k]);
-        b[u] = *reinterpret_cast<const bf16x8*>(&wrow[k]);
+    for (int k0 = c_begin * 32; k0 < c_end * 32; k0 += XT) {
+      const int tk = min(XT, c_end * 32 - k0);
+      __syncthreads();
+      for (int idx = threadIdx.x; idx < 16 * (tk / 8); idx += 256) {
+        const int r = idx / (tk / 8);
+        const int vec = idx % (tk / 8);
+        *reinterpret_cast<short8*>(&x_lds_b[r * SE + vec * 8]) =
+            *reinterpret_cast<const short8*>(
+                &x[(int64_t)min(r, M - 1) * K + k0 + vec * 8]);
       }
+      __syncthreads();
+      const short* arow = &x_lds_b[row * SE];
+      int cl = 0;
+      const int cl_end = tk / 32;
+      for (; cl + 4 <= cl_end; cl += 4) {
+        bf16x8 a[4], b[4];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        if (u & 1)
-          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc1, 0, 0, 0);
-        else
-          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc0, 0, 0, 0);
+        for (int u = 0; u < 4; ++u) {
+          a[u] = *reinterpret_cast<const bf16x8*>(&arow[(cl + u) * 32 + ks]);
+          b[u] = *reinterpret_cast<const bf16x8*>(&wrow[k0 + (cl + u) * 32 + ks]);
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (u & 1)
+            acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc1, 0, 0, 0);
+          else
+            acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b[u], acc0, 0, 0, 0);
+        }
       }
-    }
-    for (; c < c_end; ++c) {
-      const int k = c * 32 + ks;
-      const bf16x8 a = *reinterpret_cast<const bf16x8*>(&xrow[k]);
-      const bf16x8 b = *reinterpret_cast<const bf16x8*>(&wrow[k]);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
+      for (; cl < cl_end; ++cl) {
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(&arow[cl * 32 + ks]);
+        const bf16x8 b = *reinterpret_cast<const bf16x8*>(&wrow[k0 + cl * 32 + ks]);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
+      }
     }
   }
   const f32x4 acc = acc0 + acc1;
