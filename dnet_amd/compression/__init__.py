@@ -1,0 +1,101 @@
+"""Activation wire compression: column sparsification (+ wire format).
+
+Reference counterpart: src/dnet/compression/ (column_sparsify_tensor,
+compress_tensor_to_protobuf_data, decompress_tensor_from_protobuf_data,
+sparse_v1 wire format with dtype-string metadata). Drop the
+smallest-L2-norm columns of an activation before a slow (cross-node) hop;
+the intra-node xGMI ring keeps a FIXED keep-count so compressed hops stay
+RCCL-shaped (two fixed-size tensors: column indices + packed values).
+
+Config-disabled by default (DNET_TRANSPORT_COMPRESS=false), like the
+reference.
+"""
+from __future__ import annotations
+
+import struct
+
+import msgpack
+import torch
+
+from .. import ops
+
+FMT_SPARSE_V1 = "sparse_v1"
+
+
+def keep_count(d: int, keep_ratio: float) -> int:
+    return max(1, min(d, int(round(d * keep_ratio))))
+
+
+def column_sparsify(x: torch.Tensor, keep_ratio: float
+                    ) -> tuple[torch.Tensor, torch.Tensor]:
+    """x [R, D] bf16 -> (idx int32 [K] sorted asc, packed bf16 [R, K]).
+
+    Keeps the K = round(D*keep_ratio) columns with the largest L2 norm.
+    """
+    d = x.shape[-1]
+    k = keep_count(d, keep_ratio)
+    if x.is_cuda:
+        norms = torch.empty(d, dtype=torch.float32, device=x.device)
+        ops._native().col_norm2(x, norms)
+    else:
+        norms = x.float().pow(2).sum(dim=tuple(range(x.dim() - 1)))
+    idx = torch.topk(norms, k).indices.sort().values.to(torch.int32)
+    if x.is_cuda:
+        packed = torch.empty(*x.shape[:-1], k, dtype=x.dtype, device=x.device)
+        ops._native().gather_cols(x, idx, packed)
+    else:
+        packed = x.index_select(-1, idx.long())
+    return idx, packed
+
+
+def column_unsparsify(packed: torch.Tensor, idx: torch.Tensor,
+                      d: int) -> torch.Tensor:
+    out = torch.zeros(*packed.shape[:-1], d, dtype=packed.dtype,
+                      device=packed.device)
+    if packed.is_cuda:
+        ops._native().scatter_cols(packed, idx, out)
+    else:
+        out.index_copy_(-1, idx.long(), packed)
+    return out
+
+
+def dtype_string(orig_dtype: str, keep_ratio: float) -> str:
+    """Reference-style dtype metadata, e.g. 'bfloat16|90.0|fmt=sparse_v1'."""
+    return f"{orig_dtype}|{keep_ratio * 100:.1f}|fmt={FMT_SPARSE_V1}"
+
+
+def is_compressed_dtype(dtype: str) -> bool:
+    return "|" in dtype
+
+
+def compress_tensor_to_bytes(x: torch.Tensor, keep_ratio: float) -> bytes:
+    """Self-describing wire blob: msgpack header + idx + packed values."""
+    idx, packed = column_sparsify(x, keep_ratio)
+    header = {"fmt": FMT_SPARSE_V1, "shape": list(x.shape),
+              "dtype": dtype_string(str(x.dtype).replace("torch.", ""),
+                                    keep_ratio),
+              "k": int(idx.numel())}
+    h = msgpack.packb(header, use_bin_type=True)
+    return (struct.pack(">I", len(h)) + h
+            + idx.cpu().numpy().tobytes()
+            + packed.contiguous().cpu().view(torch.int16).numpy().tobytes())
+
+
+def decompress_tensor_from_bytes(blob: bytes, device="cpu") -> torch.Tensor:
+    import numpy as np
+    (hlen,) = struct.unpack(">I", blob[:4])
+    header = msgpack.unpackb(blob[4:4 + hlen], raw=False)
+    assert header["fmt"] == FMT_SPARSE_V1
+    shape = header["shape"]
+    k = header["k"]
+    off = 4 + hlen
+    idx = torch.from_numpy(np.frombuffer(blob[off:off + 4 * k],
+                                         dtype=np.int32).copy())
+    off += 4 * k
+    rows = 1
+    for s in shape[:-1]:
+        rows *= s
+    vals = torch.from_numpy(np.frombuffer(blob[off:off + rows * k * 2],
+                                          dtype=np.int16).copy())
+    packed = vals.view(*shape[:-1], k).view(torch.bfloat16).to(device)
+    return column_unsparsify(packed, idx.to(device), shape[-1])

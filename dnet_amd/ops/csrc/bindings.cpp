@@ -21,6 +21,9 @@ void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
               c10::optional<torch::Tensor> scratch, int64_t group, bool packed);
+void col_norm2(torch::Tensor x, torch::Tensor norms);
+void gather_cols(torch::Tensor x, torch::Tensor idx, torch::Tensor out);
+void scatter_cols(torch::Tensor in, torch::Tensor idx, torch::Tensor out);
 }  // namespace dnet
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -33,4 +36,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");
   m.def("gemm_m16", &dnet::gemm_m16, "MFMA decode GEMM (M<=16, bf16 or int8)");
+  m.def("col_norm2", &dnet::col_norm2, "per-column L2 norms");
+  m.def("gather_cols", &dnet::gather_cols, "pack kept columns");
+  m.def("scatter_cols", &dnet::scatter_cols, "zero + scatter kept columns");
 }

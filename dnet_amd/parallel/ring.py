@@ -63,7 +63,7 @@ class RingExecutor:
                  mb_count: int = 1, mb_size: int = 1, smax: int = 4096,
                  seed: int = 0, decoding: Optional[DecodingConfig] = None,
                  use_graphs: Optional[bool] = None, init_weights: bool = True,
-                 residency: int = 0):
+                 residency: int = 0, compress_ratio: float = 0.0):
         self.cfg = cfg
         self.rank = rank
         self.world = world
@@ -107,6 +107,37 @@ class RingExecutor:
         self._graphs: list = []
         self.last_logprob = None   # last sample's logprobs (last rank only)
         self.last_tops = None
+        # optional column-sparsified activation hops (fixed keep-count so
+        # the wire stays RCCL-shaped); reference: DNET_TRANSPORT_COMPRESS
+        self.compress_ratio = compress_ratio if 0.0 < compress_ratio < 1.0 else 0.0
+        if self.compress_ratio:
+            from ..compression import keep_count
+            k = keep_count(cfg.hidden_size, self.compress_ratio)
+            self._cidx = [torch.zeros(k, dtype=torch.int32, device=self.device)
+                          for _ in range(mb_count)]
+            self._cpacked = [torch.zeros(mb_size, k, dtype=torch.bfloat16,
+                                         device=self.device)
+                             for _ in range(mb_count)]
+
+    def _send_hidden(self, mb: int):
+        if not self.compress_ratio:
+            self.ring.send(self.hbuf[mb])
+            return
+        from ..compression import column_sparsify
+        idx, packed = column_sparsify(self.hbuf[mb], self.compress_ratio)
+        self.ring.send(idx)
+        self.ring.send(packed)
+
+    def _recv_hidden(self, mb: int):
+        if not self.compress_ratio:
+            self.ring.recv(self.hbuf[mb])
+            return
+        from ..compression import column_unsparsify
+        self.ring.recv(self._cidx[mb])
+        self.ring.recv(self._cpacked[mb])
+        self.hbuf[mb].copy_(
+            column_unsparsify(self._cpacked[mb], self._cidx[mb],
+                              self.cfg.hidden_size))
 
     # ------------- one-rank step bodies (graph-capturable) -------------
 
@@ -210,14 +241,14 @@ class RingExecutor:
                         out[mb, :, s - 1] = self.tokbuf[mb]
                     self._run_decode(mb)
                     if self.world > 1:
-                        self.ring.send(self.hbuf[mb])
+                        self._send_hidden(mb)
                         tok_req[mb] = self.ring.irecv(self.tokbuf[mb],
                                                       src=self.world - 1)
                 if self.world > 1 and not self.is_first:
-                    self.ring.recv(self.hbuf[mb])
+                    self._recv_hidden(mb)
                     self._run_decode(mb)
                     if not self.is_last:
-                        self.ring.send(self.hbuf[mb])
+                        self._send_hidden(mb)
                 if self.is_last:
                     tok, _, _ = self.sampler.sample(self.logits_buf[mb].float())
                     if self.world > 1:
@@ -267,12 +298,12 @@ class RingExecutor:
             if self.is_first:
                 self._run_decode(mb)
                 if self.world > 1:
-                    self.ring.send(self.hbuf[mb])
+                    self._send_hidden(mb)
             elif self.world > 1:
-                self.ring.recv(self.hbuf[mb])
+                self._recv_hidden(mb)
                 self._run_decode(mb)
                 if not self.is_last:
-                    self.ring.send(self.hbuf[mb])
+                    self._send_hidden(mb)
             if self.is_last:
                 tok, logprob, tops = self.sampler.sample(
                     self.logits_buf[mb].float())

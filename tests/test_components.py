@@ -192,3 +192,21 @@ def test_offload_policy_matches_fit():
     fit = run(0)
     off = run(2)
     assert torch.equal(fit, off)
+
+
+def test_compression_roundtrip():
+    from dnet_amd import compression as cz
+    torch.manual_seed(0)
+    x = torch.randn(4, 64, dtype=torch.bfloat16)
+    x[:, :8] *= 10  # dominant columns must survive
+    idx, packed = cz.column_sparsify(x, 0.5)
+    assert idx.numel() == 32 and packed.shape == (4, 32)
+    assert set(range(8)) <= set(idx.tolist())
+    y = cz.column_unsparsify(packed, idx, 64)
+    assert torch.equal(y.index_select(-1, idx.long()), packed)
+    # wire blob roundtrip
+    blob = cz.compress_tensor_to_bytes(x, 0.5)
+    y2 = cz.decompress_tensor_from_bytes(blob)
+    assert torch.equal(y, y2)
+    assert len(blob) < x.numel() * 2  # actually smaller
+    assert cz.is_compressed_dtype(cz.dtype_string("bfloat16", 0.5))

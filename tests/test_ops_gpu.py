@@ -175,3 +175,15 @@ def test_offload_gpu_matches_fit():
         return torch.cat([first.unsqueeze(-1), gen], dim=-1)
 
     assert torch.equal(run(0), run(2))
+
+
+def test_compression_kernels_gpu():
+    from dnet_amd import compression as cz
+    torch.manual_seed(1)
+    x = torch.randn(8, 512, dtype=torch.bfloat16, device=_dev())
+    idx, packed = cz.column_sparsify(x, 0.25)
+    idx_c, packed_c = cz.column_sparsify(x.cpu(), 0.25)
+    assert torch.equal(idx.cpu(), idx_c)
+    assert torch.equal(packed.cpu(), packed_c)
+    y = cz.column_unsparsify(packed, idx, 512)
+    assert torch.equal(y.cpu(), cz.column_unsparsify(packed_c, idx_c, 512))

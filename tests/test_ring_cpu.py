@@ -74,3 +74,40 @@ def test_two_rank_ring_matches_single():
         p.join(timeout=60)
         assert p.exitcode == 0
     assert torch.equal(out, single), f"ring-2 != single:\n{out}\n{single}"
+
+
+@pytest.mark.timeout(180)
+def test_two_rank_ring_compressed_hops():
+    """Ring with column-sparsified activation hops still generates sane
+    tokens (lossy, so no exact-match; shapes/flow must hold)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_compressed, args=(r, 2, 29681, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert out.shape == (MB_COUNT, MB_SIZE, NGEN + 1)
+
+
+def _rank_main_compressed(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, rank, world, "cpu", mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False,
+                      compress_ratio=0.9)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
