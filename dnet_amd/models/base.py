@@ -198,16 +198,17 @@ def _chunked_causal_attention(q, k, v, scale, q_offsets):
     qg = q.view(B, Hkv, G, T, D)
     og = out.view(B, Hkv, G, T, D)
     pos_k = torch.arange(S, device=q.device)
+    compute_dtype = q.dtype  # bf16 GEMMs (hipBLASLt MFMA); fp32 softmax
     for t0 in range(0, T, qc):
         t1 = min(T, t0 + qc)
         scores = torch.einsum("bhgtd,bhgsd->bhgts",
-                              qg[:, :, :, t0:t1].float(), kk.float()) * scale
+                              qg[:, :, :, t0:t1].to(compute_dtype),
+                              kk.to(compute_dtype)).float() * scale
         pos_q = q_offsets + torch.arange(t0, t1, device=q.device)
         mask = pos_k.view(1, -1) > pos_q.view(-1, 1)
         scores.masked_fill_(mask, float("-inf"))
-        p = torch.softmax(scores, dim=-1)
-        og[:, :, :, t0:t1] = torch.einsum("bhgts,bhgsd->bhgtd", p,
-                                          vv.float()).to(q.dtype)
+        p = torch.softmax(scores, dim=-1).to(compute_dtype)
+        og[:, :, :, t0:t1] = torch.einsum("bhgts,bhgsd->bhgtd", p, vv)
     return out
 
 
