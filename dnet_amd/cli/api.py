@@ -22,6 +22,7 @@ def main(argv=None):
     ap.add_argument("--port", type=int, default=s.api.port)
     ap.add_argument("--wire-port", type=int, default=s.api.grpc_port)
     ap.add_argument("--callback-addr", default=s.api.callback_addr)
+    ap.add_argument("--tui", action="store_true")
     args = ap.parse_args(argv)
     if args.callback_addr:
         s.api.callback_addr = args.callback_addr
@@ -48,6 +49,17 @@ async def serve(args):
     await wire.start()
     log.info("dnet-api on http://%s:%d (wire %d)", args.host, args.port,
              args.wire_port)
+    if getattr(args, "tui", False):
+        import threading
+
+        from ..tui import DnetTUI, HAS_RICH
+        if HAS_RICH:
+            tui = DnetTUI("api", lambda: {
+                "model": state.models.loaded_model or "-",
+                "devices": len(cluster.devices),
+                "pending": len(state.inference.pending)})
+            threading.Thread(target=tui.run_forever, daemon=True,
+                             name="tui").start()
     config = uvicorn.Config(app, host=args.host, port=args.port,
                             log_level="warning")
     await uvicorn.Server(config).serve()

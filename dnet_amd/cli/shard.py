@@ -24,6 +24,7 @@ def main(argv=None):
     ap.add_argument("--host", default=s.shard.host)
     ap.add_argument("--http-port", type=int, default=s.shard.http_port)
     ap.add_argument("--wire-port", type=int, default=s.shard.grpc_port)
+    ap.add_argument("--tui", action="store_true")
     args = ap.parse_args(argv)
 
     from ..shard.runtime import ShardRuntime
@@ -35,6 +36,17 @@ def main(argv=None):
     start_servers(rt, args.host, args.http_port, args.wire_port)
     log.info("dnet-shard %s on http://%s:%d (wire %d)", args.name, args.host,
              args.http_port, args.wire_port)
+    if args.tui:
+        import threading
+
+        from ..tui import DnetTUI, HAS_RICH
+        if HAS_RICH:
+            tui = DnetTUI("shard", lambda: {
+                "instance": rt.instance, "status": rt.status,
+                "model": rt.model_name or "-",
+                "queue": rt.infer_q.qsize(), "error": rt.last_error or "-"})
+            threading.Thread(target=tui.run_forever, daemon=True,
+                             name="tui").start()
     try:
         rt.run()
     except KeyboardInterrupt:
