@@ -31,5 +31,9 @@ from .moe import MoERingModel  # noqa: E402
 
 register(MoERingModel)  # mixtral / qwen2_moe / qwen3_moe
 
+from .gpt_oss import GptOssRingModel  # noqa: E402
+
+register(GptOssRingModel)
+
 __all__ = ["ModelConfig", "QuantConfig", "RingModel", "MoERingModel", "KVCache",
            "Linear", "PRESETS", "get_ring_model", "register"]

@@ -104,6 +104,7 @@ class LayerWeights:
     router: Optional[Linear] = None
     experts_gateup: Optional[list] = None   # list[Linear] per expert
     experts_down: Optional[list] = None
+    sinks: Optional[torch.Tensor] = None    # gpt-oss attention sinks [Hq]
 
     _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm")
     _LINEAR_FIELDS = ("qkv", "o", "gateup", "down")
@@ -180,7 +181,8 @@ class KVCache:
         return 2 * self.k.numel() * self.k.element_size()
 
 
-def _chunked_causal_attention(q, k, v, scale, q_offsets):
+def _chunked_causal_attention(q, k, v, scale, q_offsets, window=0,
+                              sinks=None):
     """Prefill attention with explicit GEMMs (hipBLASLt), causal.
 
     q: [B, Hq, T, D]; k/v: [B, Hkv, S, D] where S >= T and queries occupy
@@ -206,8 +208,16 @@ def _chunked_causal_attention(q, k, v, scale, q_offsets):
                               kk.to(compute_dtype)).float() * scale
         pos_q = q_offsets + torch.arange(t0, t1, device=q.device)
         mask = pos_k.view(1, -1) > pos_q.view(-1, 1)
+        if window and window > 0:
+            mask |= pos_k.view(1, -1) <= pos_q.view(-1, 1) - window
         scores.masked_fill_(mask, float("-inf"))
-        p = torch.softmax(scores, dim=-1).to(compute_dtype)
+        if sinks is not None:
+            sk = sinks.float().view(1, Hkv, G, 1, 1).expand(
+                B, Hkv, G, t1 - t0, 1)
+            p = torch.softmax(torch.cat([scores, sk], dim=-1),
+                              dim=-1)[..., :-1].to(compute_dtype)
+        else:
+            p = torch.softmax(scores, dim=-1).to(compute_dtype)
         og[:, :, :, t0:t1] = torch.einsum("bhgts,bhgsd->bhgtd", p, vv)
     return out
 
@@ -358,6 +368,11 @@ class RingModel:
         assert self.embed is not None, "this shard does not own the embedding"
         return torch.nn.functional.embedding(tokens, self.embed)
 
+    def _attn_params(self, lid: int):
+        """(window, sinks) for layer lid — overridden by sliding-window /
+        sink models (gpt-oss)."""
+        return self.cfg.sliding_window or 0, None
+
     def _qk_norm(self, q, k, lw):
         c = self.cfg
         qh = q.reshape(-1, c.head_dim)
@@ -388,7 +403,9 @@ class RingModel:
                 self._qk_norm(q, k, lw)
             li = kv.local[lid]
             ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos, self.sin)
-            attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5)
+            window, sinks = self._attn_params(lid)
+            attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5,
+                                   window, sinks)
             o = lw.o(attn.view(B, nq * d))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
             delta = self._mlp(y2, lw)
@@ -426,9 +443,10 @@ class RingModel:
             li = kv.local[lid]
             kv.k[li][:, :, p0:p0 + T] = k.transpose(1, 2)
             kv.v[li][:, :, p0:p0 + T] = v.transpose(1, 2)
+            window, sinks = self._attn_params(lid)
             attn = _chunked_causal_attention(
                 q.transpose(1, 2), kv.k[li][:, :, :p0 + T],
-                kv.v[li][:, :, :p0 + T], d ** -0.5, p0)
+                kv.v[li][:, :, :p0 + T], d ** -0.5, p0, window, sinks)
             o = lw.o(attn.transpose(1, 2).reshape(B * T, nq * d).contiguous())
             y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)
             delta = self._mlp(y2, lw)

@@ -66,15 +66,19 @@ class ModelConfig:
             intermediate_size=cfg.get("intermediate_size", 4 * hidden),
             vocab_size=cfg.get("vocab_size", 32000),
             rms_eps=cfg.get("rms_norm_eps", 1e-5),
-            rope_theta=cfg.get("rope_theta", 10000.0),
-            rope_scaling=cfg.get("rope_scaling"),
+            rope_theta=(cfg.get("rope_theta")
+                        or (cfg.get("rope_parameters") or {}).get("rope_theta")
+                        or 10000.0),
+            rope_scaling=cfg.get("rope_scaling") or cfg.get("rope_parameters"),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", mt == "qwen2"),
             qk_norm=mt in ("qwen3", "qwen3_moe"),
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
-            sliding_window=cfg.get("sliding_window") or 0,
+            sliding_window=(cfg.get("sliding_window") or 0
+                            if cfg.get("use_sliding_window", True) else 0),
             quant=quant,
         )
+        mc.sliding_window_pattern = cfg.get("layer_types")
         if mt in ("mixtral", "qwen2_moe", "qwen3_moe", "gpt_oss"):
             mc.num_experts = cfg.get("num_local_experts",
                                      cfg.get("num_experts", 8))

@@ -1,0 +1,135 @@
+"""gpt-oss ring model: MoE + alternating sliding/full attention + sinks.
+
+Reference counterpart: src/dnet/core/models/gpt_oss.py (mlx_lm gpt_oss ring
+wrapper with per-type masks and RotatingKVCache). MI355X version: the
+sliding window and the learned attention-sink logit are handled inside the
+decode-attention HIP kernel (window/sinks params); experts use the gpt-oss
+clamped-GLU activation; router is softmax-over-top-k with bias. The
+checkpoint's interleaved gate/up expert layout is de-interleaved at load to
+the [gate; up] concat the kernels use. MXFP4 checkpoints are dequantized to
+bf16/int8 at load (no MXFP4 runtime yet).
+"""
+from __future__ import annotations
+
+import torch
+
+from .base import LayerWeights, Linear, RingModel
+from .config import ModelConfig
+from .moe import MoERingModel
+
+
+def gpt_oss_glu(gu: torch.Tensor, alpha: float = 1.702,
+                limit: float = 7.0) -> torch.Tensor:
+    i = gu.shape[-1] // 2
+    gate = gu[..., :i].float().clamp(max=limit)
+    up = gu[..., i:].float().clamp(min=-limit, max=limit)
+    glu = gate * torch.sigmoid(gate * alpha)
+    return ((up + 1.0) * glu).to(gu.dtype)
+
+
+class GptOssRingModel(MoERingModel):
+    model_type = "gpt_oss"
+    model_types = ["gpt_oss"]
+
+    def _attn_params(self, lid: int):
+        types = self.cfg.sliding_window_pattern
+        if types is not None and lid < len(types):
+            sliding = types[lid] == "sliding_attention"
+        else:
+            sliding = lid % 2 == 0  # gpt-oss default: even layers slide
+        lw = self.layers.get(lid) if self.weight_provider is None else None
+        sinks = getattr(lw, "sinks", None) if lw is not None else None
+        return (self.cfg.sliding_window if sliding else 0), sinks
+
+    def _init_layer(self, rand, lid: int) -> LayerWeights:
+        lw = super()._init_layer(rand, lid)
+        c = self.cfg
+        dev = self.device
+        # gpt-oss: biases everywhere + learned sink logit per q head
+        lw.qkv.bias = rand(c.qkv_out).to(dev)
+        lw.o.bias = rand(c.hidden_size).to(dev)
+        lw.router.bias = rand(c.num_experts).to(dev)
+        lw.sinks = rand(c.num_q_heads).to(dev)
+        lw.expert_biases_gu = [rand(2 * (c.moe_intermediate_size
+                                         or c.intermediate_size)).to(dev)
+                               for _ in range(c.num_experts)]
+        lw.expert_biases_down = [rand(c.hidden_size).to(dev)
+                                 for _ in range(c.num_experts)]
+        for e in range(c.num_experts):
+            lw.experts_gateup[e].bias = lw.expert_biases_gu[e]
+            lw.experts_down[e].bias = lw.expert_biases_down[e]
+        return lw
+
+    def load_state_dict(self, sd: dict):
+        """HF gpt-oss layout: batched expert tensors
+        mlp.experts.gate_up_proj [E, H, 2I] (interleaved gate/up, input-major)
+        + per-expert biases; self_attn.sinks [Hq]."""
+        c = self.cfg
+
+        def get(name):
+            for pref in ("model.", ""):
+                if pref + name in sd:
+                    return sd[pref + name].to(torch.bfloat16)
+            return None
+
+        inter = c.moe_intermediate_size or c.intermediate_size
+        for lid in self.layer_ids:
+            p = f"layers.{lid}."
+            qw, kw, vw = (get(p + f"self_attn.{x}_proj.weight") for x in "qkv")
+            qb, kb, vb = (get(p + f"self_attn.{x}_proj.bias") for x in "qkv")
+            bias = torch.cat([qb, kb, vb]).to(self.device) if qb is not None else None
+            lw = LayerWeights(
+                attn_norm=get(p + "input_layernorm.weight").to(self.device),
+                qkv=Linear.make(torch.cat([qw, kw, vw]).to(self.device), bias,
+                                c.quant),
+                o=Linear.make(get(p + "self_attn.o_proj.weight").to(self.device),
+                              get(p + "self_attn.o_proj.bias"), c.quant),
+                mlp_norm=get(p + "post_attention_layernorm.weight").to(self.device),
+            )
+            lw.router = Linear(get(p + "mlp.router.weight").to(self.device),
+                               get(p + "mlp.router.bias"))
+            gu = get(p + "mlp.experts.gate_up_proj")        # [E, H, 2I]
+            gub = get(p + "mlp.experts.gate_up_proj_bias")  # [E, 2I]
+            dn = get(p + "mlp.experts.down_proj")           # [E, I, H]
+            dnb = get(p + "mlp.experts.down_proj_bias")     # [E, H]
+            lw.experts_gateup, lw.experts_down = [], []
+            for e in range(c.num_experts):
+                w_e = gu[e].t().contiguous()                # [2I, H] interleaved
+                # de-interleave rows: gate = even rows, up = odd rows
+                w_e = torch.cat([w_e[0::2], w_e[1::2]])     # [2I, H] concat
+                b_e = torch.cat([gub[e][0::2], gub[e][1::2]]) if gub is not None else None
+                lw.experts_gateup.append(Linear.make(
+                    w_e.to(self.device),
+                    b_e.to(self.device) if b_e is not None else None, c.quant))
+                lw.experts_down.append(Linear.make(
+                    dn[e].t().contiguous().to(self.device),
+                    dnb[e].to(self.device) if dnb is not None else None,
+                    c.quant))
+            lw.sinks = get(p + "self_attn.sinks").to(self.device)
+            self.layers[lid] = lw
+        if self.is_first:
+            self.embed = get("embed_tokens.weight").to(self.device)
+        if self.is_last:
+            self.final_norm = get("norm.weight").to(self.device)
+            head = sd.get("lm_head.weight")
+            emb = get("embed_tokens.weight")
+            self.lm_head = Linear((head.to(torch.bfloat16)
+                                   if head is not None else emb).to(self.device))
+
+    def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
+        c = self.cfg
+        logits = lw.router(y).float()
+        vals, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
+        weights = torch.softmax(vals, dim=-1)
+        out = torch.zeros_like(y, dtype=torch.float32)
+        for e in range(c.num_experts):
+            mask = (idx == e).any(dim=-1)
+            if not bool(mask.any()):
+                continue
+            rows = mask.nonzero(as_tuple=True)[0]
+            xe = y[rows].contiguous()
+            a = gpt_oss_glu(lw.experts_gateup[e](xe))
+            d = lw.experts_down[e](a)
+            we = (weights * (idx == e)).sum(dim=-1)[rows]
+            out[rows] += d.float() * we.unsqueeze(-1)
+        return out.to(y.dtype)

@@ -109,13 +109,14 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
 
 
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
-                pos: torch.Tensor, scale: float) -> torch.Tensor:
+                pos: torch.Tensor, scale: float, window: int = 0,
+                sinks: torch.Tensor | None = None) -> torch.Tensor:
     if q.is_cuda:
         # q may be a strided slice of the fused QKV buffer; out is contiguous.
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        _native().attn_decode(q, kcache, vcache, pos, out, scale)
+        _native().attn_decode(q, kcache, vcache, pos, out, scale, window, sinks)
         return out
-    return ref.attn_decode(q, kcache, vcache, pos, scale)
+    return ref.attn_decode(q, kcache, vcache, pos, scale, window, sinks)
 
 
 def rope_append(q, k, v, kcache, vcache, pos, cos, sin) -> None:

@@ -43,7 +43,9 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
                                    const int* __restrict__ pos,
                                    short* __restrict__ out, const int Hq,
                                    const int Hkv, const int Smax,
-                                   const float scale, const int ldq) {
+                                   const float scale, const int ldq,
+                                   const int window,
+                                   const short* __restrict__ sinks) {
   constexpr int P = lds_pitch(D);
   constexpr int DPL = D / kWave;  // output dims per lane (1 or 2)
   const int b = blockIdx.x / Hkv;
@@ -58,6 +60,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   short* q_lds = v_lds + kChunk * P;                           // [G][D]
 
   const int len = pos[b];
+  const int start = (window > 0) ? max(0, len - window) : 0;  // sliding
   const int64_t kvbase = ((int64_t)b * Hkv + hkv) * Smax * D;
 
   // Stage this group's q rows.
@@ -80,7 +83,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   __syncthreads();
 
   const int nchunks = (len + kChunk - 1) / kChunk;
-  for (int c = 0; c < nchunks; ++c) {
+  for (int c = start / kChunk; c < nchunks; ++c) {
     const int s0 = c * kChunk;
     const int valid = min(kChunk, len - s0);
     // Stage K and V chunk: 16 B per thread per iteration, coalesced rows.
@@ -114,12 +117,13 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
           s_val = fmaf(bits2f(kv.x[j]), bits2f(qv.x[j]), s_val);
       }
       s_val *= scale;
-      if (lane >= valid) s_val = -1e30f;
+      const bool in_win = (lane < valid) && (s0 + lane >= start);
+      if (!in_win) s_val = -1e30f;
       // Online softmax update.
       const float cmax = wave_allreduce_max(s_val);
       const float mn = fmaxf(m[hi], cmax);
       const float alpha = __expf(m[hi] - mn);
-      const float p = (lane < valid) ? __expf(s_val - mn) : 0.f;
+      const float p = in_win ? __expf(s_val - mn) : 0.f;
       l[hi] = l[hi] * alpha + wave_allreduce_sum(p);
       m[hi] = mn;
 #pragma unroll
@@ -138,7 +142,18 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
 
   for (int hi = 0; hi < nh; ++hi) {
     const int g = wid + hi * 4;
-    const float inv = (l[hi] > 0.f) ? 1.f / l[hi] : 0.f;
+    float inv;
+    if (sinks != nullptr) {
+      // gpt-oss attention sink: one extra learned softmax logit per q head
+      // whose probability mass is dropped from the output.
+      const float sk = bits2f(sinks[hkv * G + g]);
+      const float mx = fmaxf(m[hi], sk);
+      const float num = __expf(m[hi] - mx);
+      const float denom = l[hi] * num + __expf(sk - mx);
+      inv = (denom > 0.f) ? num / denom : 0.f;
+    } else {
+      inv = (l[hi] > 0.f) ? 1.f / l[hi] : 0.f;
+    }
     short* orow = out + ((int64_t)b * Hq + hkv * G + g) * D + DPL * lane;
 #pragma unroll
     for (int j = 0; j < DPL; ++j) orow[j] = f2bits(acc[hi][j] * inv);
@@ -147,7 +162,8 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
 
 // q may be a column slice of a fused-QKV buffer: strides (ldq, D, 1).
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
-                 torch::Tensor pos, torch::Tensor out, double scale) {
+                 torch::Tensor pos, torch::Tensor out, double scale,
+                 int64_t window, c10::optional<torch::Tensor> sinks) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = kcache.size(1), Smax = kcache.size(2);
   DNET_CHECK(kcache.size(0) == B && kcache.size(3) == D, "kcache shape");
@@ -163,18 +179,20 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const size_t lds = (2 * kChunk * P + G * D) * sizeof(short);
   const dim3 grid((unsigned)(B * Hkv));
   const int ldq = (int)q.stride(0);
+  const short* skp = sinks.has_value() ? (const short*)sinks->data_ptr()
+                                       : nullptr;
   if (D == 128) {
     hipLaunchKernelGGL((attn_decode_kernel<128>), grid, dim3(256), lds, stream,
                        (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                        (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
                        (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
-                       (float)scale, ldq);
+                       (float)scale, ldq, (int)window, skp);
   } else {
     hipLaunchKernelGGL((attn_decode_kernel<64>), grid, dim3(256), lds, stream,
                        (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
                        (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
                        (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
-                       (float)scale, ldq);
+                       (float)scale, ldq, (int)window, skp);
   }
 }
 

@@ -10,7 +10,8 @@ void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
                torch::Tensor out, int64_t group,
                c10::optional<torch::Tensor> bias);
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
-                 torch::Tensor pos, torch::Tensor out, double scale);
+                 torch::Tensor pos, torch::Tensor out, double scale,
+                 int64_t window, c10::optional<torch::Tensor> sinks);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
                  torch::Tensor cos_table, torch::Tensor sin_table);

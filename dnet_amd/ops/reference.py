@@ -81,8 +81,10 @@ def quantize_int8(w: torch.Tensor, group: int) -> tuple[torch.Tensor, torch.Tens
 
 
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
-                pos: torch.Tensor, scale: float) -> torch.Tensor:
-    """out[b,h] = softmax(q . K^T * scale) @ V over len=pos[b] positions."""
+                pos: torch.Tensor, scale: float, window: int = 0,
+                sinks: torch.Tensor | None = None) -> torch.Tensor:
+    """out[b,h] = softmax(q . K^T * scale) @ V over positions
+    [max(0, len-window), len) (window=0 -> full)."""
     B, Hq, D = q.shape
     Hkv = kcache.shape[1]
     G = Hq // Hkv
@@ -91,11 +93,16 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         ln = int(pos[b])
         if ln == 0:
             continue
-        k = kcache[b, :, :ln].float()          # [Hkv, ln, D]
-        v = vcache[b, :, :ln].float()
+        s0 = max(0, ln - window) if window > 0 else 0
+        k = kcache[b, :, s0:ln].float()          # [Hkv, ln-s0, D]
+        v = vcache[b, :, s0:ln].float()
         qq = q[b].float().view(Hkv, G, D)      # [Hkv, G, D]
         s = torch.einsum("hgd,hld->hgl", qq, k) * scale
-        p = torch.softmax(s, dim=-1)
+        if sinks is not None:
+            sk = sinks.float().view(Hkv, G, 1)
+            p = torch.softmax(torch.cat([s, sk], dim=-1), dim=-1)[..., :-1]
+        else:
+            p = torch.softmax(s, dim=-1)
         o = torch.einsum("hgl,hld->hgd", p, v)
         out[b] = o.reshape(Hq, D).to(q.dtype)
     return out
@@ -110,6 +117,35 @@ def rope_tables(smax: int, d: int, theta: float,
     """
     half = d // 2
     inv = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float32) / half))
+    attn_factor = 1.0
+    if scaling and scaling.get("rope_type", scaling.get("type")) == "yarn":
+        # YaRN (gpt-oss / deepseek style): NTK-by-parts interpolation +
+        # attention temperature on cos/sin.
+        import math
+        factor = scaling.get("factor", 1.0)
+        orig = scaling.get("original_max_position_embeddings", 4096)
+        beta_fast = scaling.get("beta_fast", 32.0)
+        beta_slow = scaling.get("beta_slow", 1.0)
+
+        def corr_dim(n_rot):
+            return (d * math.log(orig / (n_rot * 2 * math.pi))
+                    / (2 * math.log(theta)))
+
+        low = max(math.floor(corr_dim(beta_fast)), 0)
+        high = min(math.ceil(corr_dim(beta_slow)), half - 1)
+        rng = torch.arange(half, dtype=torch.float32)
+        ramp = ((rng - low) / max(high - low, 1e-3)).clamp(0, 1)
+        mask = 1 - ramp  # 1 = no interpolation (high freq), 0 = full
+        inv = inv / factor * (1 - mask) + inv * mask
+        attn_factor = scaling.get("attention_factor") or \
+            (0.1 * math.log(factor) + 1.0 if factor > 1 else 1.0)
+        mscale = scaling.get("mscale")
+        if mscale is not None:
+            # deepseek variant: factor from mscale/mscale_all_dim
+            def ds_mscale(scale, m):
+                return 1.0 if scale <= 1 else 0.1 * m * math.log(scale) + 1.0
+            attn_factor = (ds_mscale(factor, mscale)
+                           / ds_mscale(factor, scaling.get("mscale_all_dim", 0)))
     if scaling and scaling.get("rope_type", scaling.get("type")) == "llama3":
         factor = scaling["factor"]
         lo = scaling.get("low_freq_factor", 1.0)
@@ -125,7 +161,8 @@ def rope_tables(smax: int, d: int, theta: float,
         inv = torch.where(use_mid, mid, scaled)
     t = torch.arange(smax, dtype=torch.float32)
     freqs = torch.outer(t, inv)
-    return freqs.cos().to(device), freqs.sin().to(device)
+    return (freqs.cos().mul_(attn_factor).to(device),
+            freqs.sin().mul_(attn_factor).to(device))
 
 
 def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,

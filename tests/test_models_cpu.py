@@ -87,3 +87,42 @@ def test_moe_runs():
     m.decode_window(hd, m.layer_ids, kv)
     logits = m.normalize_project(hd)
     assert torch.isfinite(logits.float()).all()
+
+
+def test_vs_transformers_gpt_oss():
+    transformers = pytest.importorskip("transformers")
+    if not hasattr(transformers, "GptOssForCausalLM"):
+        pytest.skip("no gpt_oss in transformers")
+    import warnings
+    warnings.filterwarnings("ignore")
+    torch.manual_seed(5)
+    tc = transformers.GptOssConfig(
+        hidden_size=64, num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=64, intermediate_size=64,
+        num_local_experts=4, num_experts_per_tok=2, vocab_size=128,
+        sliding_window=8, max_position_embeddings=64, tie_word_embeddings=False)
+    hf = transformers.GptOssForCausalLM(tc).eval().float()
+    cfg = ModelConfig.from_hf(tc.to_dict())
+    assert cfg.sliding_window == 8 and cfg.num_experts == 4
+    m = get_ring_model("gpt_oss")(cfg, range(cfg.num_layers), "cpu",
+                                  True, True, smax=64)
+    m.load_state_dict(dict(hf.state_dict()))
+    kv = KVCache(cfg, range(cfg.num_layers), 1, 64, "cpu")
+    tokens = torch.randint(0, 128, (1, 20))
+    with torch.no_grad():
+        ref_logits = hf(tokens).logits[:, -1].float()
+    h = m.embed_tokens(tokens).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    ours = m.normalize_project(h[:, -1].contiguous()).float()
+    cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
+    assert (cos > 0.98).all(), f"gpt_oss vs transformers: cos={cos}"
+    # decode path consistency: prefill 19 + decode 1 == prefill 20
+    kv2 = KVCache(cfg, range(cfg.num_layers), 1, 64, "cpu")
+    h2 = m.embed_tokens(tokens[:, :-1]).clone()
+    m.prefill_window(h2, m.layer_ids, kv2, 0)
+    kv2.pos.fill_(19)
+    hd = m.embed_tokens(tokens[:, -1]).clone()
+    m.decode_window(hd, m.layer_ids, kv2)
+    dec = m.normalize_project(hd).float()
+    cos2 = torch.nn.functional.cosine_similarity(dec, ours, dim=-1)
+    assert (cos2 > 0.995).all(), f"gpt_oss decode vs prefill: cos={cos2}"
