@@ -35,5 +35,9 @@ from .gpt_oss import GptOssRingModel  # noqa: E402
 
 register(GptOssRingModel)
 
+from .deepseek_v2 import DeepseekV2RingModel  # noqa: E402
+
+register(DeepseekV2RingModel)
+
 __all__ = ["ModelConfig", "QuantConfig", "RingModel", "MoERingModel", "KVCache",
            "Linear", "PRESETS", "get_ring_model", "register"]

@@ -193,12 +193,13 @@ def _chunked_causal_attention(q, k, v, scale, q_offsets, window=0,
     Hkv = k.shape[1]
     S = k.shape[2]
     G = Hq // Hkv
-    out = torch.empty_like(q)
+    Dv = v.shape[-1]  # may differ from D (deepseek MLA)
+    out = torch.empty(B, Hq, T, Dv, dtype=q.dtype, device=q.device)
     qc = 512
     kk = k.unsqueeze(2)  # [B, Hkv, 1, S, D]
     vv = v.unsqueeze(2)
     qg = q.view(B, Hkv, G, T, D)
-    og = out.view(B, Hkv, G, T, D)
+    og = out.view(B, Hkv, G, T, Dv)
     pos_k = torch.arange(S, device=q.device)
     compute_dtype = q.dtype  # bf16 GEMMs (hipBLASLt MFMA); fp32 softmax
     for t0 in range(0, T, qc):
@@ -367,6 +368,9 @@ class RingModel:
     def embed_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
         assert self.embed is not None, "this shard does not own the embedding"
         return torch.nn.functional.embedding(tokens, self.embed)
+
+    def make_kv_cache(self, batch: int, smax: int) -> "KVCache":
+        return KVCache(self.cfg, self.layer_ids, batch, smax, self.device)
 
     def _attn_params(self, lid: int):
         """(window, sinks) for layer lid — overridden by sliding-window /

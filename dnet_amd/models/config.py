@@ -44,6 +44,16 @@ class ModelConfig:
     # sliding-window attention (gpt-oss / mistral): 0 = disabled
     sliding_window: int = 0
     sliding_window_pattern: Optional[list] = None  # per-layer types
+    # deepseek-v2 MLA
+    q_lora_rank: int = 0
+    kv_lora_rank: int = 0
+    qk_nope_head_dim: int = 0
+    qk_rope_head_dim: int = 0
+    v_head_dim: int = 0
+    first_k_dense_replace: int = 0
+    n_shared_experts: int = 0
+    routed_scaling_factor: float = 1.0
+    norm_topk_prob: bool = False
     quant: Optional[QuantConfig] = None
 
     @property
@@ -79,6 +89,19 @@ class ModelConfig:
             quant=quant,
         )
         mc.sliding_window_pattern = cfg.get("layer_types")
+        if mt in ("deepseek_v2", "deepseek_v3"):
+            mc.q_lora_rank = cfg.get("q_lora_rank") or 0
+            mc.kv_lora_rank = cfg.get("kv_lora_rank", 512)
+            mc.qk_nope_head_dim = cfg.get("qk_nope_head_dim", 128)
+            mc.qk_rope_head_dim = cfg.get("qk_rope_head_dim", 64)
+            mc.v_head_dim = cfg.get("v_head_dim", 128)
+            mc.first_k_dense_replace = cfg.get("first_k_dense_replace", 0)
+            mc.n_shared_experts = cfg.get("n_shared_experts") or 0
+            mc.routed_scaling_factor = cfg.get("routed_scaling_factor", 1.0)
+            mc.norm_topk_prob = cfg.get("norm_topk_prob", False)
+            mc.num_experts = cfg.get("n_routed_experts") or 0
+            mc.num_experts_per_tok = cfg.get("num_experts_per_tok", 0)
+            mc.moe_intermediate_size = cfg.get("moe_intermediate_size", 0)
         if mt in ("mixtral", "qwen2_moe", "qwen3_moe", "gpt_oss"):
             mc.num_experts = cfg.get("num_local_experts",
                                      cfg.get("num_experts", 8))

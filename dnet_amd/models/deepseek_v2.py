@@ -1,0 +1,287 @@
+"""DeepSeek-V2 ring model: MLA (multi-head latent attention) + MoE.
+
+Reference counterpart: src/dnet/core/models/deepseek_v2.py (mlx_lm wrapper;
+qk_nope + qk_rope split head dims). The MLA projections (q, kv_a with
+shared rope key, kv_a layernorm, kv_b) run through the fused GEMV/GEMM
+linears; attention itself uses the torch path because the MLA head dim
+(qk_nope+qk_rope, e.g. 192) is outside the 64/128 decode-kernel shapes —
+a dedicated MLA decode kernel is roadmap. KV is cached per head
+(uncompressed k/v; latent-cache compression is roadmap). MoE layers use
+routed top-k (softmax scoring, greedy) * routed_scaling_factor + shared
+experts; the first ``first_k_dense_replace`` layers are dense.
+"""
+from __future__ import annotations
+
+import torch
+
+from .. import ops
+from .base import KVCache, LayerWeights, Linear, RingModel, _chunked_causal_attention
+from .config import ModelConfig
+
+
+class MLAKVCache(KVCache):
+    """k and v have different head dims in MLA."""
+
+    def __init__(self, cfg, layer_ids, batch, smax, device):
+        self.layer_ids = list(layer_ids)
+        self.local = {g: i for i, g in enumerate(self.layer_ids)}
+        L = len(self.layer_ids)
+        kd = cfg.qk_nope_head_dim + cfg.qk_rope_head_dim
+        self.k = torch.zeros(L, batch, cfg.num_q_heads, smax, kd,
+                             dtype=torch.bfloat16, device=device)
+        self.v = torch.zeros(L, batch, cfg.num_q_heads, smax,
+                             cfg.v_head_dim, dtype=torch.bfloat16,
+                             device=device)
+        self.pos = torch.zeros(batch, dtype=torch.int32, device=device)
+        self.smax = smax
+        self.batch = batch
+
+
+class DeepseekV2RingModel(RingModel):
+    model_type = "deepseek_v2"
+    model_types = ["deepseek_v2", "deepseek_v3"]
+
+    def __init__(self, cfg: ModelConfig, layer_ids, device, is_first, is_last,
+                 smax: int = 4096):
+        super().__init__(cfg, layer_ids, device, is_first, is_last, smax)
+        # rope tables over the rope sub-dim only
+        cos, sin = ops.rope_tables(smax, cfg.qk_rope_head_dim, cfg.rope_theta,
+                                   scaling=cfg.rope_scaling)
+        self.cos = cos.to(self.device)
+        self.sin = sin.to(self.device)
+        kd = cfg.qk_nope_head_dim + cfg.qk_rope_head_dim
+        self.scale = kd ** -0.5
+
+    def make_kv_cache(self, batch: int, smax: int) -> MLAKVCache:
+        return MLAKVCache(self.cfg, self.layer_ids, batch, smax, self.device)
+
+    # ---------- weights ----------
+
+    def _init_layer(self, rand, lid: int) -> LayerWeights:
+        c = self.cfg
+        dev = self.device
+        ones = lambda n: torch.ones(n, dtype=torch.bfloat16, device=dev)
+        kd = c.qk_nope_head_dim + c.qk_rope_head_dim
+        lw = LayerWeights(
+            attn_norm=ones(c.hidden_size),
+            mlp_norm=ones(c.hidden_size),
+        )
+        if c.q_lora_rank:
+            lw.q_a = Linear.make(rand(c.q_lora_rank, c.hidden_size).to(dev),
+                                 None, c.quant)
+            lw.q_a_norm = ones(c.q_lora_rank)
+            lw.q_b = Linear.make(rand(c.num_q_heads * kd, c.q_lora_rank).to(dev),
+                                 None, c.quant)
+        else:
+            lw.q = Linear.make(rand(c.num_q_heads * kd, c.hidden_size).to(dev),
+                               None, c.quant)
+        lw.kv_a = Linear.make(
+            rand(c.kv_lora_rank + c.qk_rope_head_dim, c.hidden_size).to(dev),
+            None, None)
+        lw.kv_a_norm = ones(c.kv_lora_rank)
+        lw.kv_b = Linear.make(
+            rand(c.num_q_heads * (c.qk_nope_head_dim + c.v_head_dim),
+                 c.kv_lora_rank).to(dev), None, c.quant)
+        lw.o = Linear.make(
+            rand(c.hidden_size, c.num_q_heads * c.v_head_dim).to(dev),
+            None, c.quant)
+        inter = c.moe_intermediate_size or c.intermediate_size
+        if c.num_experts and lid >= c.first_k_dense_replace:
+            lw.router = Linear(rand(c.num_experts, c.hidden_size).to(dev))
+            lw.experts_gateup = [
+                Linear.make(rand(2 * inter, c.hidden_size).to(dev), None, c.quant)
+                for _ in range(c.num_experts)]
+            lw.experts_down = [
+                Linear.make(rand(c.hidden_size, inter).to(dev), None, c.quant)
+                for _ in range(c.num_experts)]
+            if c.n_shared_experts:
+                si = inter * c.n_shared_experts
+                lw.shared_gateup = Linear.make(
+                    rand(2 * si, c.hidden_size).to(dev), None, c.quant)
+                lw.shared_down = Linear.make(
+                    rand(c.hidden_size, si).to(dev), None, c.quant)
+        else:
+            lw.gateup = Linear.make(rand(2 * c.intermediate_size,
+                                         c.hidden_size).to(dev), None, c.quant)
+            lw.down = Linear.make(rand(c.hidden_size,
+                                       c.intermediate_size).to(dev), None, c.quant)
+        return lw
+
+    def load_state_dict(self, sd: dict):
+        c = self.cfg
+
+        def get(name):
+            for pref in ("model.", ""):
+                if pref + name in sd:
+                    return sd[pref + name].to(torch.bfloat16)
+            return None
+
+        dev = self.device
+        inter = c.moe_intermediate_size or c.intermediate_size
+        for lid in self.layer_ids:
+            p = f"layers.{lid}."
+            lw = LayerWeights(
+                attn_norm=get(p + "input_layernorm.weight").to(dev),
+                mlp_norm=get(p + "post_attention_layernorm.weight").to(dev))
+            if get(p + "self_attn.q_proj.weight") is not None:
+                lw.q = Linear.make(get(p + "self_attn.q_proj.weight").to(dev),
+                                   None, c.quant)
+            else:
+                lw.q_a = Linear.make(
+                    get(p + "self_attn.q_a_proj.weight").to(dev), None, c.quant)
+                lw.q_a_norm = get(p + "self_attn.q_a_layernorm.weight").to(dev)
+                lw.q_b = Linear.make(
+                    get(p + "self_attn.q_b_proj.weight").to(dev), None, c.quant)
+            lw.kv_a = Linear.make(
+                get(p + "self_attn.kv_a_proj_with_mqa.weight").to(dev), None, None)
+            lw.kv_a_norm = get(p + "self_attn.kv_a_layernorm.weight").to(dev)
+            lw.kv_b = Linear.make(get(p + "self_attn.kv_b_proj.weight").to(dev),
+                                  None, c.quant)
+            lw.o = Linear.make(get(p + "self_attn.o_proj.weight").to(dev),
+                               None, c.quant)
+            if get(p + "mlp.gate_proj.weight") is not None:     # dense layer
+                lw.gateup = Linear.make(
+                    torch.cat([get(p + "mlp.gate_proj.weight"),
+                               get(p + "mlp.up_proj.weight")]).to(dev),
+                    None, c.quant)
+                lw.down = Linear.make(get(p + "mlp.down_proj.weight").to(dev),
+                                      None, c.quant)
+            else:                                               # MoE layer
+                lw.router = Linear(get(p + "mlp.gate.weight").to(dev))
+                gu = get(p + "mlp.experts.gate_up_proj")    # [E, 2I, H]
+                dn = get(p + "mlp.experts.down_proj")       # [E, H, I]
+                lw.experts_gateup, lw.experts_down = [], []
+                for e in range(c.num_experts):
+                    lw.experts_gateup.append(Linear.make(
+                        gu[e].contiguous().to(dev), None, c.quant))
+                    lw.experts_down.append(Linear.make(
+                        dn[e].contiguous().to(dev), None, c.quant))
+                if get(p + "mlp.shared_experts.gate_proj.weight") is not None:
+                    lw.shared_gateup = Linear.make(
+                        torch.cat([get(p + "mlp.shared_experts.gate_proj.weight"),
+                                   get(p + "mlp.shared_experts.up_proj.weight")]
+                                  ).to(dev), None, c.quant)
+                    lw.shared_down = Linear.make(
+                        get(p + "mlp.shared_experts.down_proj.weight").to(dev),
+                        None, c.quant)
+            self.layers[lid] = lw
+        if self.is_first:
+            self.embed = get("embed_tokens.weight").to(dev)
+        if self.is_last:
+            self.final_norm = get("norm.weight").to(dev)
+            head = sd.get("lm_head.weight")
+            emb = get("embed_tokens.weight")
+            self.lm_head = Linear((head.to(torch.bfloat16)
+                                   if head is not None else emb).to(dev))
+
+    # ---------- forward ----------
+
+    def _mla_qkv(self, y: torch.Tensor, lw, positions):
+        """y [T, H] -> q [T, Hq, kd], k [T, Hq, kd], v [T, Hq, vd] (roped)."""
+        c = self.cfg
+        T = y.shape[0]
+        nope, rope, vd = c.qk_nope_head_dim, c.qk_rope_head_dim, c.v_head_dim
+        kd = nope + rope
+        if getattr(lw, "q_a", None) is not None:
+            qa = ops.rmsnorm(lw.q_a(y), None, lw.q_a_norm, c.rms_eps)
+            q = lw.q_b(qa).view(T, c.num_q_heads, kd)
+        else:
+            q = lw.q(y).view(T, c.num_q_heads, kd)
+        comp = lw.kv_a(y)                              # [T, lora + rope]
+        c_kv = ops.rmsnorm(comp[:, :c.kv_lora_rank].contiguous(), None,
+                           lw.kv_a_norm, c.rms_eps)
+        k_pe = comp[:, c.kv_lora_rank:].view(T, 1, rope)
+        kv = lw.kv_b(c_kv).view(T, c.num_q_heads, nope + vd)
+        k_nope, v = kv[..., :nope], kv[..., nope:]
+        q_pe = ops.rope_apply(q[..., nope:].contiguous(), self.cos, self.sin,
+                              positions)
+        k_pe = ops.rope_apply(k_pe.contiguous(), self.cos, self.sin, positions)
+        q = torch.cat([q[..., :nope], q_pe], dim=-1)
+        k = torch.cat([k_nope, k_pe.expand(T, c.num_q_heads, rope)], dim=-1)
+        return q, k, v.contiguous()
+
+    def decode_window(self, h, layer_ids, kv):
+        c = self.cfg
+        B = h.shape[0]
+        delta = None
+        for lid in layer_ids:
+            lw = self._layer(lid)
+            y = ops.rmsnorm(delta if delta is not None else h,
+                            h if delta is not None else None,
+                            lw.attn_norm, c.rms_eps)
+            q, k, v = self._mla_qkv(y, lw, kv.pos.long())
+            li = kv.local[lid]
+            # vectorized append at per-batch positions
+            idx = kv.pos.long().view(B, 1, 1, 1)
+            kv.k[li].scatter_(2, idx.expand(B, c.num_q_heads, 1, k.shape[-1]),
+                              k.unsqueeze(2))
+            kv.v[li].scatter_(2, idx.expand(B, c.num_q_heads, 1, v.shape[-1]),
+                              v.unsqueeze(2))
+            len_t = kv.pos + 1
+            attn = self._attn_decode_torch(q, kv.k[li], kv.v[li], len_t)
+            o = lw.o(attn.reshape(B, -1))
+            y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
+            delta = self._mlp_for(lid, y2, lw)
+        h.add_(delta)
+        return h
+
+    def _attn_decode_torch(self, q, kcache, vcache, len_t):
+        """q [B, Hq, kd]; per-head full cache; masked softmax attention."""
+        B, Hq, kd = q.shape
+        S = kcache.shape[2]
+        scores = torch.einsum("bhd,bhsd->bhs", q.float(), kcache.float())
+        scores *= self.scale
+        mask = (torch.arange(S, device=q.device).view(1, 1, S)
+                >= len_t.view(B, 1, 1))
+        scores.masked_fill_(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        out = torch.einsum("bhs,bhsd->bhd", p, vcache.float())
+        return out.to(q.dtype)
+
+    def prefill_window(self, h, layer_ids, kv, p0: int):
+        c = self.cfg
+        B, T, H = h.shape
+        positions = torch.arange(p0, p0 + T, device=h.device)
+        for lid in layer_ids:
+            lw = self._layer(lid)
+            flat = h.view(B * T, H)
+            y = ops.rmsnorm(flat, None, lw.attn_norm, c.rms_eps)
+            q, k, v = self._mla_qkv(y, lw, positions.repeat(B))
+            kd, vd = k.shape[-1], v.shape[-1]
+            q = q.view(B, T, c.num_q_heads, kd).transpose(1, 2)
+            k = k.view(B, T, c.num_q_heads, kd).transpose(1, 2)
+            v = v.view(B, T, c.num_q_heads, vd).transpose(1, 2)
+            li = kv.local[lid]
+            kv.k[li][:, :, p0:p0 + T] = k
+            kv.v[li][:, :, p0:p0 + T] = v
+            attn = _chunked_causal_attention(
+                q, kv.k[li][:, :, :p0 + T], kv.v[li][:, :, :p0 + T],
+                self.scale, p0)
+            o = lw.o(attn.transpose(1, 2).reshape(B * T, -1).contiguous())
+            y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)
+            flat.add_(self._mlp_for(lid, y2, lw))
+        return h
+
+    def _mlp_for(self, lid: int, y: torch.Tensor, lw) -> torch.Tensor:
+        c = self.cfg
+        if lw.gateup is not None:
+            return self._mlp(y, lw)
+        # MoE: softmax scoring -> top-k (greedy) * routed_scaling + shared
+        scores = torch.softmax(lw.router(y).float(), dim=-1)
+        weights, idx = torch.topk(scores, c.num_experts_per_tok, dim=-1)
+        if c.norm_topk_prob:
+            weights = weights / weights.sum(-1, keepdim=True)
+        out = torch.zeros_like(y, dtype=torch.float32)
+        for e in range(c.num_experts):
+            mask = (idx == e).any(dim=-1)
+            if not bool(mask.any()):
+                continue
+            rows = mask.nonzero(as_tuple=True)[0]
+            xe = y[rows].contiguous()
+            d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](xe)))
+            we = (weights * (idx == e)).sum(dim=-1)[rows]
+            out[rows] += d.float() * we.unsqueeze(-1)
+        out *= c.routed_scaling_factor
+        if getattr(lw, "shared_gateup", None) is not None:
+            out += lw.shared_down(ops.swiglu(lw.shared_gateup(y))).float()
+        return out.to(y.dtype)

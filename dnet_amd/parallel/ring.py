@@ -87,7 +87,7 @@ class RingExecutor:
             from ..shard.policies import enable_offload  # lazy: avoids cycle
             self.weight_cache = enable_offload(self.model, residency)
             use_graphs = False  # slot addresses change per step
-        self.kvs = [KVCache(cfg, self.my_layers, mb_size, smax, self.device)
+        self.kvs = [self.model.make_kv_cache(mb_size, smax)
                     for _ in range(mb_count)]
         self.sampler = Sampler(decoding or DecodingConfig())
         H = cfg.hidden_size

@@ -126,3 +126,45 @@ def test_vs_transformers_gpt_oss():
     dec = m.normalize_project(hd).float()
     cos2 = torch.nn.functional.cosine_similarity(dec, ours, dim=-1)
     assert (cos2 > 0.995).all(), f"gpt_oss decode vs prefill: cos={cos2}"
+
+
+def test_vs_transformers_deepseek_v2():
+    transformers = pytest.importorskip("transformers")
+    if not hasattr(transformers, "DeepseekV2ForCausalLM"):
+        pytest.skip("no deepseek_v2 in transformers")
+    import warnings
+    warnings.filterwarnings("ignore")
+    torch.manual_seed(6)
+    tc = transformers.DeepseekV2Config(
+        hidden_size=64, num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=4, vocab_size=128, intermediate_size=96,
+        q_lora_rank=None, kv_lora_rank=32, qk_nope_head_dim=32,
+        qk_rope_head_dim=16, v_head_dim=32, n_routed_experts=4,
+        num_experts_per_tok=2, n_shared_experts=1, moe_intermediate_size=32,
+        first_k_dense_replace=1, topk_method="greedy", n_group=1,
+        topk_group=1, max_position_embeddings=64, tie_word_embeddings=False)
+    hf = transformers.DeepseekV2ForCausalLM(tc).eval().float()
+    cfg = ModelConfig.from_hf(tc.to_dict())
+    assert cfg.kv_lora_rank == 32 and cfg.num_experts == 4
+    m = get_ring_model("deepseek_v2")(cfg, range(cfg.num_layers), "cpu",
+                                      True, True, smax=64)
+    m.load_state_dict(dict(hf.state_dict()))
+    kv = m.make_kv_cache(1, 64)
+    tokens = torch.randint(0, 128, (1, 14))
+    with torch.no_grad():
+        ref_logits = hf(tokens).logits[:, -1].float()
+    h = m.embed_tokens(tokens).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    ours = m.normalize_project(h[:, -1].contiguous()).float()
+    cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
+    assert (cos > 0.98).all(), f"deepseek_v2 vs transformers: cos={cos}"
+    # decode-vs-prefill consistency
+    kv2 = m.make_kv_cache(1, 64)
+    h2 = m.embed_tokens(tokens[:, :-1]).clone()
+    m.prefill_window(h2, m.layer_ids, kv2, 0)
+    kv2.pos.fill_(13)
+    hd = m.embed_tokens(tokens[:, -1]).clone()
+    m.decode_window(hd, m.layer_ids, kv2)
+    dec = m.normalize_project(hd).float()
+    cos2 = torch.nn.functional.cosine_similarity(dec, ours, dim=-1)
+    assert (cos2 > 0.995).all(), f"deepseek decode vs prefill: cos={cos2}"
