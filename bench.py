@@ -45,6 +45,8 @@ def main():
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (debug only; invalid for scoring)")
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree inside each pipeline stage")
     ap.add_argument("--residency", type=int, default=0,
                     help="GPU-resident layers per rank (0=all; <local layers "
                          "enables host-DRAM weight streaming)")
@@ -62,8 +64,8 @@ def main():
     mb_count = max(args.mb_per_rank * world, 1)
     ex = RingExecutor(cfg, rank, world, device, mb_count=mb_count,
                       mb_size=args.mb_size, smax=args.smax, seed=1234,
-                      use_graphs=on_gpu and not args.no_graphs,
-                      residency=args.residency)
+                      use_graphs=on_gpu and not args.no_graphs and args.tp == 1,
+                      residency=args.residency, tp=args.tp)
 
     g = torch.Generator().manual_seed(7)
     tokens = torch.randint(0, cfg.vocab_size,
@@ -124,7 +126,8 @@ def main():
                 "global_batch": total_seqs,
                 "seq_len": args.prompt_len,
                 "gen_len": args.steps,
-                "parallelism": f"ring-pp{world}"
+                "parallelism": (f"ring-pp{world // args.tp}"
+                                + (f"xtp{args.tp}" if args.tp > 1 else ""))
                                + (f"+offload(res={args.residency})"
                                   if args.residency else ""),
                 "microbatches": mb_count,

@@ -233,13 +233,25 @@ class RingModel:
     model_type = "llama"
 
     def __init__(self, cfg: ModelConfig, layer_ids: Sequence[int], device,
-                 is_first: bool, is_last: bool, smax: int = 4096):
+                 is_first: bool, is_last: bool, smax: int = 4096,
+                 tp_rank: int = 0, tp_size: int = 1, tp_group=None):
         self.cfg = cfg
         self.layer_ids = list(layer_ids)
         self.device = torch.device(device)
         self.is_first = is_first
         self.is_last = is_last
         self.smax = smax
+        # tensor parallelism inside a pipeline stage: attention heads and
+        # MLP intermediate dims are column/row-sharded; o-proj and mlp
+        # outputs are partial sums all-reduced over tp_group (RCCL within
+        # the stage's xGMI neighborhood). MoE layers shard experts (EP).
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.tp_group = tp_group
+        if tp_size > 1:
+            assert cfg.num_q_heads % tp_size == 0
+            assert cfg.num_kv_heads % tp_size == 0
+            assert cfg.intermediate_size % tp_size == 0
         self.layers: dict[int, LayerWeights] = {}
         self.embed: Optional[torch.Tensor] = None
         self.final_norm: Optional[torch.Tensor] = None
@@ -301,18 +313,25 @@ class RingModel:
         c = self.cfg
         dev = self.device
         ones = lambda n: torch.ones(n, dtype=torch.bfloat16, device=dev)
+        qkv_b = rand(c.qkv_out) if c.attention_bias else None
         lw = LayerWeights(
             attn_norm=ones(c.hidden_size),
-            qkv=Linear.make(rand(c.qkv_out, c.hidden_size).to(dev),
-                            rand(c.qkv_out).to(dev) if c.attention_bias else None,
+            qkv=Linear.make(self._slice_qkv(rand(c.qkv_out, c.hidden_size)).to(dev),
+                            self._slice_qkv(qkv_b.view(-1, 1)).view(-1).to(dev)
+                            if qkv_b is not None else None,
                             c.quant),
-            o=Linear.make(rand(c.hidden_size, c.num_q_heads * c.head_dim).to(dev),
-                          None, c.quant),
+            o=Linear.make(
+                self._slice_cols(rand(c.hidden_size,
+                                      c.num_q_heads * c.head_dim)).to(dev),
+                None, c.quant),
             mlp_norm=ones(c.hidden_size),
-            gateup=Linear.make(rand(2 * c.intermediate_size, c.hidden_size).to(dev),
-                               None, c.quant),
-            down=Linear.make(rand(c.hidden_size, c.intermediate_size).to(dev),
-                             None, c.quant),
+            gateup=Linear.make(
+                self._slice_gateup(rand(2 * c.intermediate_size,
+                                        c.hidden_size)).to(dev), None, c.quant),
+            down=Linear.make(
+                self._slice_cols(rand(c.hidden_size,
+                                      c.intermediate_size)).to(dev),
+                None, c.quant),
         )
         if c.qk_norm:
             lw.q_norm = ones(c.head_dim)
@@ -369,8 +388,59 @@ class RingModel:
         assert self.embed is not None, "this shard does not own the embedding"
         return torch.nn.functional.embedding(tokens, self.embed)
 
+    @property
+    def nq_local(self) -> int:
+        return self.cfg.num_q_heads // self.tp_size
+
+    @property
+    def nkv_local(self) -> int:
+        return self.cfg.num_kv_heads // self.tp_size
+
+    def _tp_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.tp_size > 1:
+            import torch.distributed as dist
+            dist.all_reduce(t, group=self.tp_group)
+        return t
+
+    def _slice_qkv(self, w: torch.Tensor) -> torch.Tensor:
+        """Slice the fused [Q;K;V] rows to this tp rank's heads."""
+        if self.tp_size == 1:
+            return w
+        c = self.cfg
+        d = c.head_dim
+        q, k, v = torch.split(w, [c.num_q_heads * d, c.num_kv_heads * d,
+                                  c.num_kv_heads * d])
+        r, t = self.tp_rank, self.tp_size
+        return torch.cat([q[r * q.shape[0] // t:(r + 1) * q.shape[0] // t],
+                          k[r * k.shape[0] // t:(r + 1) * k.shape[0] // t],
+                          v[r * v.shape[0] // t:(r + 1) * v.shape[0] // t]])
+
+    def _slice_rows(self, w: torch.Tensor) -> torch.Tensor:
+        if self.tp_size == 1:
+            return w
+        n = w.shape[0] // self.tp_size
+        return w[self.tp_rank * n:(self.tp_rank + 1) * n]
+
+    def _slice_gateup(self, w: torch.Tensor) -> torch.Tensor:
+        if self.tp_size == 1:
+            return w
+        g, u = torch.chunk(w, 2)
+        return torch.cat([self._slice_rows(g), self._slice_rows(u)])
+
+    def _slice_cols(self, w: torch.Tensor) -> torch.Tensor:
+        if self.tp_size == 1:
+            return w
+        n = w.shape[1] // self.tp_size
+        return w[:, self.tp_rank * n:(self.tp_rank + 1) * n].contiguous()
+
     def make_kv_cache(self, batch: int, smax: int) -> "KVCache":
-        return KVCache(self.cfg, self.layer_ids, batch, smax, self.device)
+        cfg = self.cfg
+        if self.tp_size == 1:
+            return KVCache(cfg, self.layer_ids, batch, smax, self.device)
+        import copy
+        c2 = copy.copy(cfg)
+        c2.num_kv_heads = self.nkv_local
+        return KVCache(c2, self.layer_ids, batch, smax, self.device)
 
     def _attn_params(self, lid: int):
         """(window, sinks) for layer lid — overridden by sliding-window /
@@ -391,7 +461,7 @@ class RingModel:
         write position for this token."""
         c = self.cfg
         B = h.shape[0]
-        nq, nkv, d = c.num_q_heads, c.num_kv_heads, c.head_dim
+        nq, nkv, d = self.nq_local, self.nkv_local, c.head_dim
         len_t = kv.pos + 1  # attend over lengths including the token being written
         delta = None
         for lid in layer_ids:
@@ -410,9 +480,9 @@ class RingModel:
             window, sinks = self._attn_params(lid)
             attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5,
                                    window, sinks)
-            o = lw.o(attn.view(B, nq * d))
+            o = self._tp_reduce(lw.o(attn.view(B, nq * d)))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
-            delta = self._mlp(y2, lw)
+            delta = self._tp_reduce(self._mlp(y2, lw))
         h.add_(delta)
         return h
 
@@ -427,7 +497,7 @@ class RingModel:
         kv.pos is advanced by the caller after the full shard window."""
         c = self.cfg
         B, T, H = h.shape
-        nq, nkv, d = c.num_q_heads, c.num_kv_heads, c.head_dim
+        nq, nkv, d = self.nq_local, self.nkv_local, c.head_dim
         positions = torch.arange(p0, p0 + T, device=h.device)
         for lid in layer_ids:
             lw = self._layer(lid)
@@ -451,9 +521,10 @@ class RingModel:
             attn = _chunked_causal_attention(
                 q.transpose(1, 2), kv.k[li][:, :, :p0 + T],
                 kv.v[li][:, :, :p0 + T], d ** -0.5, p0, window, sinks)
-            o = lw.o(attn.transpose(1, 2).reshape(B * T, nq * d).contiguous())
+            o = self._tp_reduce(
+                lw.o(attn.transpose(1, 2).reshape(B * T, nq * d).contiguous()))
             y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)
-            delta = self._mlp(y2, lw)
+            delta = self._tp_reduce(self._mlp(y2, lw))
             flat.add_(delta)
         return h
 

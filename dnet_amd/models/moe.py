@@ -47,6 +47,8 @@ class MoERingModel(RingModel):
         weights = torch.softmax(weights, dim=-1)
         out = torch.zeros_like(y, dtype=torch.float32)
         for e in range(c.num_experts):
+            if self.tp_size > 1 and e % self.tp_size != self.tp_rank:
+                continue  # expert parallelism: partial sum reduced by caller
             mask = (idx == e).any(dim=-1)
             if not bool(mask.any()):
                 continue

@@ -63,23 +63,46 @@ class RingExecutor:
                  mb_count: int = 1, mb_size: int = 1, smax: int = 4096,
                  seed: int = 0, decoding: Optional[DecodingConfig] = None,
                  use_graphs: Optional[bool] = None, init_weights: bool = True,
-                 residency: int = 0, compress_ratio: float = 0.0):
+                 residency: int = 0, compress_ratio: float = 0.0,
+                 tp: int = 1):
         self.cfg = cfg
         self.rank = rank
         self.world = world
         self.device = torch.device(device)
-        self.plan = plan or RingPlan.contiguous(cfg.num_layers, world)
+        # tensor parallelism inside each pipeline stage: ranks are laid out
+        # stage-major (rank = stage*tp + tp_rank); ring hops are pairwise
+        # rank -> rank+tp; all-reduce runs in the per-stage group.
+        assert world % tp == 0, "world must be a multiple of tp"
+        self.tp = tp
+        self.stages = world // tp
+        self.stage = rank // tp
+        self.tp_rank = rank % tp
+        self.plan = plan or RingPlan.contiguous(cfg.num_layers, self.stages)
         assert self.plan.rounds == 1, "k>1 rounds land with the solver milestone"
-        self.my_layers = [l for w in self.plan.assignments[rank] for l in w]
+        self.my_layers = [l for w in self.plan.assignments[self.stage] for l in w]
         self.mb_count = mb_count
         self.mb_size = mb_size
         self.smax = smax
-        self.is_first = rank == 0
-        self.is_last = rank == world - 1
-        self.ring = Ring(rank, world, self.device) if world > 1 else None
+        self.is_first = self.stage == 0
+        self.is_last = self.stage == self.stages - 1
+        self.ring = None
+        tp_group = None
+        if world > 1:
+            self.ring = Ring(rank, world, self.device)
+            self.ring.next = (rank + tp) % world
+            self.ring.prev = (rank - tp) % world
+        if tp > 1:
+            for st in range(self.stages):  # collective: create every group
+                g = dist.new_group(list(range(st * tp, (st + 1) * tp)))
+                if st == self.stage:
+                    tp_group = g
+        # token return path: last-stage rank pairs with its first-stage peer
+        self.token_src = (self.stages - 1) * tp + self.tp_rank
+        self.token_dst = self.tp_rank
         cls = get_ring_model(cfg.model_type)
         self.model = cls(cfg, self.my_layers, self.device, self.is_first,
-                         self.is_last, smax=smax)
+                         self.is_last, smax=smax, tp_rank=self.tp_rank,
+                         tp_size=tp, tp_group=tp_group)
         if init_weights:
             self.model.init_random(seed)
         self.weight_cache = None
@@ -204,16 +227,17 @@ class RingExecutor:
             self.model.prefill_window(h, self.my_layers, self.kvs[mb], 0)
             self.kvs[mb].pos.fill_(T)
             if not self.is_last:
-                self.ring.send(h)
+                self.ring.send(h)  # stages > 1 by construction here
                 if self.is_first:
                     tok_reqs.append(self.ring.irecv(first_tokens[mb],
-                                                    src=self.world - 1))
+                                                    src=self.token_src))
             else:
                 logits = self.model.normalize_project(h[:, -1].contiguous())
                 tok, _, _ = self.sampler.sample(logits)
                 first_tokens[mb] = tok
                 if not self.is_first:
-                    self.ring.send(first_tokens[mb], dst=0)
+                    self.ring.send(first_tokens[mb], dst=self.token_dst)
+                    # (single-stage TP: is_first too, so no self-send)
         for r in tok_reqs:
             r.wait()
         if self.is_first:
@@ -240,26 +264,26 @@ class RingExecutor:
                     if out is not None and s > 0:
                         out[mb, :, s - 1] = self.tokbuf[mb]
                     self._run_decode(mb)
-                    if self.world > 1:
+                    if self.stages > 1:
                         self._send_hidden(mb)
                         tok_req[mb] = self.ring.irecv(self.tokbuf[mb],
-                                                      src=self.world - 1)
-                if self.world > 1 and not self.is_first:
+                                                      src=self.token_src)
+                if self.stages > 1 and not self.is_first:
                     self._recv_hidden(mb)
                     self._run_decode(mb)
                     if not self.is_last:
                         self._send_hidden(mb)
                 if self.is_last:
                     tok, _, _ = self.sampler.sample(self.logits_buf[mb].float())
-                    if self.world > 1:
-                        self.ring.send(tok, dst=0)
+                    if self.stages > 1:
+                        self.ring.send(tok, dst=self.token_dst)
                     else:
                         self.tokbuf[mb].copy_(tok)
-                        if out is not None:
+                        if out is not None and self.rank == 0:
                             out[mb, :, s] = tok
                 self.kvs[mb].pos.add_(1)
         # drain last round's tokens on rank 0
-        if self.is_first and self.world > 1:
+        if self.is_first and self.stages > 1:
             for mb in range(M):
                 if mb in tok_req:
                     tok_req.pop(mb).wait()
@@ -297,9 +321,9 @@ class RingExecutor:
         for s in range(max_tokens - 1):
             if self.is_first:
                 self._run_decode(mb)
-                if self.world > 1:
+                if self.stages > 1:
                     self._send_hidden(mb)
-            elif self.world > 1:
+            elif self.stages > 1:
                 self._recv_hidden(mb)
                 self._run_decode(mb)
                 if not self.is_last:
@@ -310,7 +334,7 @@ class RingExecutor:
                 self.last_logprob, self.last_tops = logprob, tops
                 self.tokbuf[mb].copy_(tok)
             if self.world > 1:
-                dist.broadcast(self.tokbuf[mb], src=self.world - 1)
+                dist.broadcast(self.tokbuf[mb], src=(self.stages - 1) * self.tp)
             self.kvs[mb].pos.add_(1)
             produced += 1
             done = is_stop(self.tokbuf[mb])

@@ -111,3 +111,78 @@ def _rank_main_compressed(rank, world, port, q):
     if rank == 0:
         q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
     dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_tp2_matches_single():
+    """One stage with TP=2 (sharded heads/MLP + gloo all-reduce) must
+    generate exactly the single-rank tokens (same deterministic weights)."""
+    single = _run_single()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_tp, args=(r, 2, 29691, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert torch.equal(out, single), f"tp2 != single:\n{out}\n{single}"
+
+
+def _rank_main_tp(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, rank, world, "cpu", mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False,
+                      tp=2)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_pp2_tp2_matches_single():
+    """4 ranks = 2 pipeline stages x TP 2 must equal the single-rank run."""
+    single = _run_single()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_pp_tp, args=(r, 4, 29701, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=200)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert torch.equal(out, single)
+
+
+def _rank_main_pp_tp(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, rank, world, "cpu", mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False,
+                      tp=2)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
