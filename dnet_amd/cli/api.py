@@ -39,7 +39,8 @@ async def serve(args):
 
     log = get_logger("api")
     s = get_settings()
-    discovery = StaticDiscovery(args.hostfile, own_instance="api")
+    discovery = StaticDiscovery(args.hostfile, own_instance="api",
+                                own_is_manager=True)
     await discovery.async_start()
     cluster = ClusterManager(discovery)
     state = ApiState(cluster, s)

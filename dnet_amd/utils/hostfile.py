@@ -63,14 +63,19 @@ class StaticDiscovery:
 
     def __init__(self, hostfile: str, own_instance: str = "",
                  own_http_port: int = 0, own_grpc_port: int = 0,
-                 own_ip: str = "127.0.0.1"):
+                 own_ip: str = "127.0.0.1", own_is_manager: bool = False):
         self.devices = load_hostfile(hostfile)
         self.own_instance = own_instance
         if own_instance and not any(d.instance == own_instance
                                     for d in self.devices):
             self.devices.append(DeviceProperties(
                 instance=own_instance, local_ip=own_ip,
-                server_port=own_http_port, shard_port=own_grpc_port))
+                server_port=own_http_port, shard_port=own_grpc_port,
+                is_manager=own_is_manager))
+        else:
+            for d in self.devices:
+                if d.instance == own_instance:
+                    d.is_manager = own_is_manager
         self._running = False
 
     async def async_start(self):

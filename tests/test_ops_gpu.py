@@ -213,7 +213,7 @@ def test_gpt_oss_decode_gpu():
     torch.manual_seed(12)
     hf = dict(model_type="gpt_oss", hidden_size=128, num_hidden_layers=2,
               num_attention_heads=4, num_key_value_heads=2, head_dim=64,
-              intermediate_size=128, num_local_experts=4,
+              intermediate_size=128, num_local_experts=2,
               num_experts_per_tok=2, vocab_size=256, sliding_window=8,
               layer_types=["sliding_attention", "full_attention"])
     cfg = ModelConfig.from_hf(hf)
