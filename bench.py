@@ -45,6 +45,7 @@ def main():
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (debug only; invalid for scoring)")
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--kv-bits", type=int, default=16, choices=[8, 16])
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree inside each pipeline stage")
     ap.add_argument("--residency", type=int, default=0,
@@ -65,7 +66,8 @@ def main():
     ex = RingExecutor(cfg, rank, world, device, mb_count=mb_count,
                       mb_size=args.mb_size, smax=args.smax, seed=1234,
                       use_graphs=on_gpu and not args.no_graphs and args.tp == 1,
-                      residency=args.residency, tp=args.tp)
+                      residency=args.residency, tp=args.tp,
+                      kv_bits=args.kv_bits)
 
     g = torch.Generator().manual_seed(7)
     tokens = torch.randint(0, cfg.vocab_size,

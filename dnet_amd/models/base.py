@@ -163,22 +163,51 @@ class KVCache:
     """
 
     def __init__(self, cfg: ModelConfig, layer_ids: Sequence[int], batch: int,
-                 smax: int, device):
+                 smax: int, device, kv_bits: int = 16):
         self.layer_ids = list(layer_ids)
         self.local = {g: i for i, g in enumerate(self.layer_ids)}
+        self.kv_bits = kv_bits
         L = len(self.layer_ids)
-        self.k = torch.zeros(L, batch, cfg.num_kv_heads, smax, cfg.head_dim,
-                             dtype=torch.bfloat16, device=device)
-        self.v = torch.zeros_like(self.k)
+        shape = (L, batch, cfg.num_kv_heads, smax, cfg.head_dim)
+        if kv_bits == 8:
+            ng = cfg.head_dim // 64
+            self.k = torch.zeros(shape, dtype=torch.int8, device=device)
+            self.v = torch.zeros_like(self.k)
+            self.kscale = torch.zeros(L, batch, cfg.num_kv_heads, smax, ng,
+                                      dtype=torch.bfloat16, device=device)
+            self.vscale = torch.zeros_like(self.kscale)
+        else:
+            self.k = torch.zeros(shape, dtype=torch.bfloat16, device=device)
+            self.v = torch.zeros_like(self.k)
+            self.kscale = self.vscale = None
         self.pos = torch.zeros(batch, dtype=torch.int32, device=device)
         self.smax = smax
         self.batch = batch
+
+    @property
+    def quantized(self) -> bool:
+        return self.k.dtype == torch.int8
+
+    def k_deq(self, li: int) -> torch.Tensor:
+        if not self.quantized:
+            return self.k[li]
+        from ..ops import reference as _r
+        return _r.dequant_kv(self.k[li], self.kscale[li])
+
+    def v_deq(self, li: int) -> torch.Tensor:
+        if not self.quantized:
+            return self.v[li]
+        from ..ops import reference as _r
+        return _r.dequant_kv(self.v[li], self.vscale[li])
 
     def reset(self):
         self.pos.zero_()
 
     def nbytes(self) -> int:
-        return 2 * self.k.numel() * self.k.element_size()
+        n = 2 * self.k.numel() * self.k.element_size()
+        if self.kscale is not None:
+            n += 2 * self.kscale.numel() * self.kscale.element_size()
+        return n
 
 
 def _chunked_causal_attention(q, k, v, scale, q_offsets, window=0,
@@ -433,14 +462,16 @@ class RingModel:
         n = w.shape[1] // self.tp_size
         return w[:, self.tp_rank * n:(self.tp_rank + 1) * n].contiguous()
 
+    kv_bits: int = 16  # set by the executor (8 = int8 group-64 KV cache)
+
     def make_kv_cache(self, batch: int, smax: int) -> "KVCache":
         cfg = self.cfg
-        if self.tp_size == 1:
-            return KVCache(cfg, self.layer_ids, batch, smax, self.device)
-        import copy
-        c2 = copy.copy(cfg)
-        c2.num_kv_heads = self.nkv_local
-        return KVCache(c2, self.layer_ids, batch, smax, self.device)
+        if self.tp_size > 1:
+            import copy
+            cfg = copy.copy(cfg)
+            cfg.num_kv_heads = self.nkv_local
+        return KVCache(cfg, self.layer_ids, batch, smax, self.device,
+                       kv_bits=self.kv_bits)
 
     def _attn_params(self, lid: int):
         """(window, sinks) for layer lid — overridden by sliding-window /
@@ -476,10 +507,15 @@ class RingModel:
             if c.qk_norm:
                 self._qk_norm(q, k, lw)
             li = kv.local[lid]
-            ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos, self.sin)
+            ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos,
+                            self.sin,
+                            kv.kscale[li] if kv.quantized else None,
+                            kv.vscale[li] if kv.quantized else None)
             window, sinks = self._attn_params(lid)
             attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5,
-                                   window, sinks)
+                                   window, sinks,
+                                   kv.kscale[li] if kv.quantized else None,
+                                   kv.vscale[li] if kv.quantized else None)
             o = self._tp_reduce(lw.o(attn.view(B, nq * d)))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
             delta = self._tp_reduce(self._mlp(y2, lw))
@@ -515,12 +551,21 @@ class RingModel:
             q = ops.rope_apply(q, self.cos, self.sin, positions)
             k = ops.rope_apply(k, self.cos, self.sin, positions)
             li = kv.local[lid]
-            kv.k[li][:, :, p0:p0 + T] = k.transpose(1, 2)
-            kv.v[li][:, :, p0:p0 + T] = v.transpose(1, 2)
+            if kv.quantized:
+                from ..ops import reference as _r
+                kc, ks = _r.quantize_kv_rows(k.transpose(1, 2))
+                vc, vs = _r.quantize_kv_rows(v.transpose(1, 2))
+                kv.k[li][:, :, p0:p0 + T] = kc
+                kv.kscale[li][:, :, p0:p0 + T] = ks
+                kv.v[li][:, :, p0:p0 + T] = vc
+                kv.vscale[li][:, :, p0:p0 + T] = vs
+            else:
+                kv.k[li][:, :, p0:p0 + T] = k.transpose(1, 2)
+                kv.v[li][:, :, p0:p0 + T] = v.transpose(1, 2)
             window, sinks = self._attn_params(lid)
             attn = _chunked_causal_attention(
-                q.transpose(1, 2), kv.k[li][:, :, :p0 + T],
-                kv.v[li][:, :, :p0 + T], d ** -0.5, p0, window, sinks)
+                q.transpose(1, 2), kv.k_deq(li)[:, :, :p0 + T],
+                kv.v_deq(li)[:, :, :p0 + T], d ** -0.5, p0, window, sinks)
             o = self._tp_reduce(
                 lw.o(attn.transpose(1, 2).reshape(B * T, nq * d).contiguous()))
             y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)

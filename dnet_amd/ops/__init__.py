@@ -110,20 +110,26 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
 
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 pos: torch.Tensor, scale: float, window: int = 0,
-                sinks: torch.Tensor | None = None) -> torch.Tensor:
+                sinks: torch.Tensor | None = None,
+                kscale: torch.Tensor | None = None,
+                vscale: torch.Tensor | None = None) -> torch.Tensor:
     if q.is_cuda:
         # q may be a strided slice of the fused QKV buffer; out is contiguous.
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        _native().attn_decode(q, kcache, vcache, pos, out, scale, window, sinks)
+        _native().attn_decode(q, kcache, vcache, pos, out, scale, window,
+                              sinks, kscale, vscale)
         return out
-    return ref.attn_decode(q, kcache, vcache, pos, scale, window, sinks)
+    return ref.attn_decode(q, kcache, vcache, pos, scale, window, sinks,
+                           kscale, vscale)
 
 
-def rope_append(q, k, v, kcache, vcache, pos, cos, sin) -> None:
+def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
+                kscale=None, vscale=None) -> None:
     if q.is_cuda:
-        _native().rope_append(q, k, v, kcache, vcache, pos, cos, sin)
+        _native().rope_append(q, k, v, kcache, vcache, pos, cos, sin,
+                              kscale, vscale)
         return
-    ref.rope_append(q, k, v, kcache, vcache, pos, cos, sin)
+    ref.rope_append(q, k, v, kcache, vcache, pos, cos, sin, kscale, vscale)
 
 
 def swiglu(gu: torch.Tensor) -> torch.Tensor:

@@ -36,10 +36,12 @@ __device__ __forceinline__ float wave_allreduce_max(float v) {
 
 constexpr int kChunk = 64;
 
-template <int D>
+template <int D, bool Q8>
 __global__ void attn_decode_kernel(const short* __restrict__ q,
-                                   const short* __restrict__ kc,
-                                   const short* __restrict__ vc,
+                                   const void* __restrict__ kc,
+                                   const void* __restrict__ vc,
+                                   const short* __restrict__ kscale,
+                                   const short* __restrict__ vscale,
                                    const int* __restrict__ pos,
                                    short* __restrict__ out, const int Hq,
                                    const int Hkv, const int Smax,
@@ -86,14 +88,36 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   for (int c = start / kChunk; c < nchunks; ++c) {
     const int s0 = c * kChunk;
     const int valid = min(kChunk, len - s0);
-    // Stage K and V chunk: 16 B per thread per iteration, coalesced rows.
+    // Stage K and V chunk into bf16 LDS; Q8 caches dequantize while staging
+    // (HBM reads stay int8 — half the KV bandwidth).
     for (int i = threadIdx.x; i < kChunk * D / 8; i += blockDim.x) {
       const int row = i / (D / 8);
       const int col = (i % (D / 8)) * 8;
       short8 kv8, vv8;
       if (row < valid) {
-        kv8 = *reinterpret_cast<const short8*>(kc + kvbase + (int64_t)(s0 + row) * D + col);
-        vv8 = *reinterpret_cast<const short8*>(vc + kvbase + (int64_t)(s0 + row) * D + col);
+        if (Q8) {
+          constexpr int NG = D / 64;
+          const int64_t rb = kvbase / D * (int64_t)NG
+                             + (int64_t)(s0 + row) * NG + col / 64;
+          const float ks = bits2f(kscale[rb]);
+          const float vs = bits2f(vscale[rb]);
+          const int2 kq = *reinterpret_cast<const int2*>(
+              (const int8_t*)kc + kvbase + (int64_t)(s0 + row) * D + col);
+          const int2 vq = *reinterpret_cast<const int2*>(
+              (const int8_t*)vc + kvbase + (int64_t)(s0 + row) * D + col);
+          const int8_t* kb = reinterpret_cast<const int8_t*>(&kq);
+          const int8_t* vb = reinterpret_cast<const int8_t*>(&vq);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            kv8.x[j] = f2bits((float)kb[j] * ks);
+            vv8.x[j] = f2bits((float)vb[j] * vs);
+          }
+        } else {
+          kv8 = *reinterpret_cast<const short8*>(
+              (const short*)kc + kvbase + (int64_t)(s0 + row) * D + col);
+          vv8 = *reinterpret_cast<const short8*>(
+              (const short*)vc + kvbase + (int64_t)(s0 + row) * D + col);
+        }
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) { kv8.x[j] = 0; vv8.x[j] = 0; }
@@ -163,7 +187,9 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
 // q may be a column slice of a fused-QKV buffer: strides (ldq, D, 1).
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale,
-                 int64_t window, c10::optional<torch::Tensor> sinks) {
+                 int64_t window, c10::optional<torch::Tensor> sinks,
+                 c10::optional<torch::Tensor> kscale,
+                 c10::optional<torch::Tensor> vscale) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = kcache.size(1), Smax = kcache.size(2);
   DNET_CHECK(kcache.size(0) == B && kcache.size(3) == D, "kcache shape");
@@ -181,19 +207,20 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const int ldq = (int)q.stride(0);
   const short* skp = sinks.has_value() ? (const short*)sinks->data_ptr()
                                        : nullptr;
-  if (D == 128) {
-    hipLaunchKernelGGL((attn_decode_kernel<128>), grid, dim3(256), lds, stream,
-                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                       (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
-                       (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
-                       (float)scale, ldq, (int)window, skp);
-  } else {
-    hipLaunchKernelGGL((attn_decode_kernel<64>), grid, dim3(256), lds, stream,
-                       (const short*)q.data_ptr(), (const short*)kcache.data_ptr(),
-                       (const short*)vcache.data_ptr(), (const int*)pos.data_ptr(),
-                       (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,
-                       (float)scale, ldq, (int)window, skp);
-  }
+  const bool q8 = kcache.dtype() == torch::kInt8;
+  const short* ksp = q8 ? (const short*)kscale->data_ptr() : nullptr;
+  const short* vsp = q8 ? (const short*)vscale->data_ptr() : nullptr;
+#define LAUNCH_ATTN(DD, QQ)                                                   \
+  hipLaunchKernelGGL((attn_decode_kernel<DD, QQ>), grid, dim3(256), lds,      \
+                     stream, (const short*)q.data_ptr(), kcache.data_ptr(),   \
+                     vcache.data_ptr(), ksp, vsp, (const int*)pos.data_ptr(), \
+                     (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,    \
+                     (float)scale, ldq, (int)window, skp)
+  if (D == 128 && q8) LAUNCH_ATTN(128, true);
+  else if (D == 128) LAUNCH_ATTN(128, false);
+  else if (q8) LAUNCH_ATTN(64, true);
+  else LAUNCH_ATTN(64, false);
+#undef LAUNCH_ATTN
 }
 
 }  // namespace dnet

@@ -11,10 +11,14 @@ void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
                c10::optional<torch::Tensor> bias);
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale,
-                 int64_t window, c10::optional<torch::Tensor> sinks);
+                 int64_t window, c10::optional<torch::Tensor> sinks,
+                 c10::optional<torch::Tensor> kscale,
+                 c10::optional<torch::Tensor> vscale);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
-                 torch::Tensor cos_table, torch::Tensor sin_table);
+                 torch::Tensor cos_table, torch::Tensor sin_table,
+                 c10::optional<torch::Tensor> kscale,
+                 c10::optional<torch::Tensor> vscale);
 void swiglu(torch::Tensor gu, torch::Tensor y);
 void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group, bool packed);

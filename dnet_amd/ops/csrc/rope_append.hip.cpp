@@ -13,11 +13,16 @@
 namespace dnet {
 
 // grid: (B, Hq + Hkv); block: D/2 lanes, lane i rotates pair (i, i + D/2).
+// Q8: KV cache stored int8 with one bf16 scale per 64-dim group
+// (reference: utils/model.py make_cache kv quantization, 8-bit group 64).
+template <bool Q8>
 __global__ void rope_append_kernel(short* __restrict__ q,
                                    short* __restrict__ k,
                                    const short* __restrict__ v,
-                                   short* __restrict__ kcache,
-                                   short* __restrict__ vcache,
+                                   void* __restrict__ kcache,
+                                   void* __restrict__ vcache,
+                                   short* __restrict__ kscale,
+                                   short* __restrict__ vscale,
                                    const int* __restrict__ pos,
                                    const float* __restrict__ cost,
                                    const float* __restrict__ sint,
@@ -26,37 +31,90 @@ __global__ void rope_append_kernel(short* __restrict__ q,
                                    const int ldv) {
   const int b = blockIdx.x;
   const int h = blockIdx.y;
-  const int i = threadIdx.x;  // 0 .. D/2-1
+  const int i = threadIdx.x;  // 0 .. max(D/2,64)-1; lanes >= D/2 idle
   const int p = pos[b];
   const int half = D / 2;
-  const float c = cost[(int64_t)p * half + i];
-  const float s = sint[(int64_t)p * half + i];
+  const bool act = i < half;
+  const float c = act ? cost[(int64_t)p * half + i] : 0.f;
+  const float s = act ? sint[(int64_t)p * half + i] : 0.f;
   if (h < Hq) {
+    if (!act) return;
     short* row = q + (int64_t)b * ldq + h * D;
     const float x1 = bits2f(row[i]), x2 = bits2f(row[i + half]);
     row[i] = f2bits(x1 * c - x2 * s);
     row[i + half] = f2bits(x2 * c + x1 * s);
-  } else {
-    const int hk = h - Hq;
-    short* krow = k + (int64_t)b * ldk + hk * D;
+    return;
+  }
+  const int hk = h - Hq;
+  short* krow = k + (int64_t)b * ldk + hk * D;
+  float k1 = 0.f, k2 = 0.f, v1 = 0.f, v2 = 0.f;
+  if (act) {
     const float x1 = bits2f(krow[i]), x2 = bits2f(krow[i + half]);
-    const short r1 = f2bits(x1 * c - x2 * s);
-    const short r2 = f2bits(x2 * c + x1 * s);
-    krow[i] = r1;
-    krow[i + half] = r2;
-    short* kdst = kcache + (((int64_t)b * Hkv + hk) * Smax + p) * D;
-    kdst[i] = r1;
-    kdst[i + half] = r2;
+    k1 = x1 * c - x2 * s;
+    k2 = x2 * c + x1 * s;
+    krow[i] = f2bits(k1);
+    krow[i + half] = f2bits(k2);
     const short* vrow = v + (int64_t)b * ldv + hk * D;
-    short* vdst = vcache + (((int64_t)b * Hkv + hk) * Smax + p) * D;
-    vdst[i] = vrow[i];
-    vdst[i + half] = vrow[i + half];
+    v1 = bits2f(vrow[i]);
+    v2 = bits2f(vrow[i + half]);
+  }
+  const int64_t rowbase = ((int64_t)b * Hkv + hk) * Smax + p;
+  if (!Q8) {
+    if (!act) return;
+    short* kdst = (short*)kcache + rowbase * D;
+    kdst[i] = f2bits(k1);
+    kdst[i + half] = f2bits(k2);
+    short* vdst = (short*)vcache + rowbase * D;
+    vdst[i] = f2bits(v1);
+    vdst[i + half] = f2bits(v2);
+    return;
+  }
+  // int8 group-64 quantization. For D=128 (half=64): elem i is in group 0,
+  // elem i+half in group 1. For D=64 (half=32): both elems in group 0.
+  // One full wave participates in the reductions (idle lanes contribute 0).
+  const int ng = D / 64;
+  auto wave_max = [](float m) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      m = fmaxf(m, __shfl_xor(m, off, 64));
+    return m;
+  };
+  auto q8scale = [](float amax) {
+    return bits2f(f2bits(fmaxf(amax, 1e-8f) / 127.f));  // bf16-rounded
+  };
+  {
+    const float a0 = wave_max(ng == 2 ? fabsf(k1) : fmaxf(fabsf(k1), fabsf(k2)));
+    const float a1 = ng == 2 ? wave_max(fabsf(k2)) : a0;
+    if (act) {
+      const float s1 = q8scale(a0), s2 = q8scale(a1);
+      int8_t* kdst = (int8_t*)kcache + rowbase * D;
+      kdst[i] = (int8_t)min(max(__float2int_rn(k1 / s1), -127), 127);
+      kdst[i + half] = (int8_t)min(max(__float2int_rn(k2 / s2), -127), 127);
+      short* ksrow = kscale + rowbase * ng;
+      if (i == 0) ksrow[0] = f2bits(q8scale(a0));
+      if (i == 0 && ng == 2) ksrow[1] = f2bits(q8scale(a1));
+    }
+  }
+  {
+    const float a0 = wave_max(ng == 2 ? fabsf(v1) : fmaxf(fabsf(v1), fabsf(v2)));
+    const float a1 = ng == 2 ? wave_max(fabsf(v2)) : a0;
+    if (act) {
+      const float s1 = q8scale(a0), s2 = q8scale(a1);
+      int8_t* vdst = (int8_t*)vcache + rowbase * D;
+      vdst[i] = (int8_t)min(max(__float2int_rn(v1 / s1), -127), 127);
+      vdst[i + half] = (int8_t)min(max(__float2int_rn(v2 / s2), -127), 127);
+      short* vsrow = vscale + rowbase * ng;
+      if (i == 0) vsrow[0] = f2bits(q8scale(a0));
+      if (i == 0 && ng == 2) vsrow[1] = f2bits(q8scale(a1));
+    }
   }
 }
 
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
-                 torch::Tensor cos_table, torch::Tensor sin_table) {
+                 torch::Tensor cos_table, torch::Tensor sin_table,
+                 c10::optional<torch::Tensor> kscale,
+                 c10::optional<torch::Tensor> vscale) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = k.size(1), Smax = kcache.size(2);
   DNET_CHECK(D % 2 == 0 && D / 2 <= 1024, "head_dim");
@@ -66,14 +124,33 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   DNET_CHECK(k.stride(2) == 1 && k.stride(1) == D, "k inner contiguous");
   DNET_CHECK(v.stride(2) == 1 && v.stride(1) == D, "v inner contiguous");
   auto stream = current_stream();
-  hipLaunchKernelGGL(rope_append_kernel, dim3((unsigned)B, (unsigned)(Hq + Hkv)),
-                     dim3((unsigned)(D / 2)), 0, stream, (short*)q.data_ptr(),
-                     (short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     (short*)kcache.data_ptr(), (short*)vcache.data_ptr(),
-                     (const int*)pos.data_ptr(), (const float*)cos_table.data_ptr(),
-                     (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
-                     (int)Smax, (int)D, (int)q.stride(0), (int)k.stride(0),
-                     (int)v.stride(0));
+  const bool q8 = kcache.dtype() == torch::kInt8;
+  if (q8) {
+    DNET_CHECK(kscale.has_value() && vscale.has_value(), "q8 needs scales");
+    DNET_CHECK(D == 64 || D == 128, "q8 kv needs D 64/128");
+    hipLaunchKernelGGL(rope_append_kernel<true>,
+                       dim3((unsigned)B, (unsigned)(Hq + Hkv)),
+                       dim3(64), 0, stream, (short*)q.data_ptr(),
+                       (short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       kcache.data_ptr(), vcache.data_ptr(),
+                       (short*)kscale->data_ptr(), (short*)vscale->data_ptr(),
+                       (const int*)pos.data_ptr(),
+                       (const float*)cos_table.data_ptr(),
+                       (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
+                       (int)Smax, (int)D, (int)q.stride(0), (int)k.stride(0),
+                       (int)v.stride(0));
+  } else {
+    hipLaunchKernelGGL(rope_append_kernel<false>,
+                       dim3((unsigned)B, (unsigned)(Hq + Hkv)),
+                       dim3((unsigned)(D / 2)), 0, stream, (short*)q.data_ptr(),
+                       (short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       kcache.data_ptr(), vcache.data_ptr(), nullptr, nullptr,
+                       (const int*)pos.data_ptr(),
+                       (const float*)cos_table.data_ptr(),
+                       (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
+                       (int)Smax, (int)D, (int)q.stride(0), (int)k.stride(0),
+                       (int)v.stride(0));
+  }
 }
 
 }  // namespace dnet

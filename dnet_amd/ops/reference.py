@@ -80,9 +80,30 @@ def quantize_int8(w: torch.Tensor, group: int) -> tuple[torch.Tensor, torch.Tens
     return q.view(n, k), scales
 
 
+def quantize_kv_rows(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """int8 group-64 KV quantization over the last dim (matches the
+    rope_append kernel: bf16-rounded scale = amax/127)."""
+    *lead, d = x.shape
+    ng = d // 64
+    xf = x.float().view(*lead, ng, 64)
+    amax = xf.abs().amax(dim=-1).clamp_min(1e-8)
+    scales = (amax / 127.0).to(torch.bfloat16)
+    codes = torch.round(xf / scales.float().unsqueeze(-1)).clamp(-127, 127)
+    return codes.to(torch.int8).view(*lead, d), scales.view(*lead, ng)
+
+
+def dequant_kv(codes: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    *lead, d = codes.shape
+    ng = d // 64
+    xf = codes.float().view(*lead, ng, 64) * scales.float().unsqueeze(-1)
+    return xf.view(*lead, d).to(torch.bfloat16)
+
+
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 pos: torch.Tensor, scale: float, window: int = 0,
-                sinks: torch.Tensor | None = None) -> torch.Tensor:
+                sinks: torch.Tensor | None = None,
+                kscale: torch.Tensor | None = None,
+                vscale: torch.Tensor | None = None) -> torch.Tensor:
     """out[b,h] = softmax(q . K^T * scale) @ V over positions
     [max(0, len-window), len) (window=0 -> full)."""
     B, Hq, D = q.shape
@@ -94,8 +115,12 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         if ln == 0:
             continue
         s0 = max(0, ln - window) if window > 0 else 0
-        k = kcache[b, :, s0:ln].float()          # [Hkv, ln-s0, D]
-        v = vcache[b, :, s0:ln].float()
+        if kcache.dtype == torch.int8:
+            k = dequant_kv(kcache[b, :, s0:ln], kscale[b, :, s0:ln]).float()
+            v = dequant_kv(vcache[b, :, s0:ln], vscale[b, :, s0:ln]).float()
+        else:
+            k = kcache[b, :, s0:ln].float()          # [Hkv, ln-s0, D]
+            v = vcache[b, :, s0:ln].float()
         qq = q[b].float().view(Hkv, G, D)      # [Hkv, G, D]
         s = torch.einsum("hgd,hld->hgl", qq, k) * scale
         if sinks is not None:
@@ -180,16 +205,26 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
     return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
 
 
-def rope_append(q, k, v, kcache, vcache, pos, cos, sin):
+def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
+                kscale=None, vscale=None):
     """Decode-step fused rope+append reference (in-place on q, k and caches)."""
     B = q.shape[0]
     positions = pos.long()
     q.copy_(rope_apply(q, cos, sin, positions))
     k.copy_(rope_apply(k, cos, sin, positions))
+    q8 = kcache.dtype == torch.int8
     for b in range(B):
         p = int(pos[b])
-        kcache[b, :, p] = k[b]
-        vcache[b, :, p] = v[b]
+        if q8:
+            kc, ks = quantize_kv_rows(k[b])
+            vc, vs = quantize_kv_rows(v[b])
+            kcache[b, :, p] = kc
+            kscale[b, :, p] = ks
+            vcache[b, :, p] = vc
+            vscale[b, :, p] = vs
+        else:
+            kcache[b, :, p] = k[b]
+            vcache[b, :, p] = v[b]
 
 
 def swiglu(gu: torch.Tensor) -> torch.Tensor:

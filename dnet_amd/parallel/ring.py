@@ -64,7 +64,7 @@ class RingExecutor:
                  seed: int = 0, decoding: Optional[DecodingConfig] = None,
                  use_graphs: Optional[bool] = None, init_weights: bool = True,
                  residency: int = 0, compress_ratio: float = 0.0,
-                 tp: int = 1):
+                 tp: int = 1, kv_bits: int = 16):
         self.cfg = cfg
         self.rank = rank
         self.world = world
@@ -103,6 +103,7 @@ class RingExecutor:
         self.model = cls(cfg, self.my_layers, self.device, self.is_first,
                          self.is_last, smax=smax, tp_rank=self.tp_rank,
                          tp_size=tp, tp_group=tp_group)
+        self.model.kv_bits = kv_bits
         if init_weights:
             self.model.init_random(seed)
         self.weight_cache = None

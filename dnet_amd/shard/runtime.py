@@ -151,7 +151,8 @@ class ShardRuntime:
                           use_graphs=(device.type == "cuda"
                                       and self.settings.compute.use_graphs
                                       and residency == 0),
-                          init_weights=synthetic, residency=residency)
+                          init_weights=synthetic, residency=residency,
+                          kv_bits=req.kv_bits)
         if not synthetic:
             self._load_weights(ex, req)
             if residency:

@@ -168,3 +168,25 @@ def test_vs_transformers_deepseek_v2():
     dec = m.normalize_project(hd).float()
     cos2 = torch.nn.functional.cosine_similarity(dec, ours, dim=-1)
     assert (cos2 > 0.995).all(), f"deepseek decode vs prefill: cos={cos2}"
+
+
+def test_kv_cache_quantized_close_to_fp():
+    """int8 group-64 KV cache tracks the bf16 cache closely (CPU ref)."""
+    torch.manual_seed(8)
+    cfg = _tiny()
+    m, _ = _build(cfg)
+    toks = torch.randint(0, cfg.vocab_size, (2, 10))
+
+    def run(bits):
+        m.kv_bits = bits
+        kv = m.make_kv_cache(2, 64)
+        h = m.embed_tokens(toks).clone()
+        m.prefill_window(h, m.layer_ids, kv, 0)
+        kv.pos.fill_(10)
+        hd = m.embed_tokens(toks[:, -1]).clone()
+        m.decode_window(hd, m.layer_ids, kv)
+        return m.normalize_project(hd).float()
+
+    a, b = run(16), run(8)
+    cos = torch.nn.functional.cosine_similarity(a, b, dim=-1)
+    assert (cos > 0.99).all(), f"kv8 vs kv16: {cos}"

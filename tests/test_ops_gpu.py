@@ -240,3 +240,44 @@ def test_gpt_oss_decode_gpu():
     lc = mc.normalize_project(hdc).float()
     cos = torch.nn.functional.cosine_similarity(lg, lc, dim=-1)
     assert (cos > 0.99).all(), f"gpu vs cpu gpt_oss decode: {cos}"
+
+
+def test_kv8_kernels_vs_ref():
+    """Quantized-KV rope_append + attn_decode kernels vs the CPU reference."""
+    torch.manual_seed(13)
+    B, Hq, Hkv, D, Smax = 2, 8, 2, 128, 128
+    cos, sin = ops.rope_tables(Smax, D, 10000.0)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=_dev())
+    k = torch.randn(B, Hkv, D, dtype=torch.bfloat16, device=_dev())
+    v = torch.randn(B, Hkv, D, dtype=torch.bfloat16, device=_dev())
+    kc = torch.zeros(B, Hkv, Smax, D, dtype=torch.int8, device=_dev())
+    vc = torch.zeros_like(kc)
+    ks = torch.zeros(B, Hkv, Smax, 2, dtype=torch.bfloat16, device=_dev())
+    vs = torch.zeros_like(ks)
+    pos = torch.tensor([0, 77], dtype=torch.int32, device=_dev())
+    qr, kr = q.cpu().clone(), k.cpu().clone()
+    kcr, vcr = kc.cpu().clone(), vc.cpu().clone()
+    ksr, vsr = ks.cpu().clone(), vs.cpu().clone()
+    ref.rope_append(qr, kr, v.cpu().clone(), kcr, vcr, pos.cpu(), cos, sin,
+                    ksr, vsr)
+    ops.rope_append(q, k, v, kc, vc, pos, cos.to(_dev()), sin.to(_dev()),
+                    ks, vs)
+    assert torch.equal(kc.cpu(), kcr) or \
+        (kc.cpu().int() - kcr.int()).abs().max() <= 1  # rounding edge
+    assert torch.allclose(ks.float().cpu(), ksr.float(), atol=1e-3, rtol=1e-2)
+    # decode attention over a random quantized cache
+    torch.manual_seed(14)
+    kc2 = torch.randint(-127, 128, (B, Hkv, Smax, D), dtype=torch.int8,
+                        device=_dev())
+    vc2 = torch.randint(-127, 128, (B, Hkv, Smax, D), dtype=torch.int8,
+                        device=_dev())
+    ks2 = (torch.rand(B, Hkv, Smax, 2, device=_dev()) * 0.02 + 0.005
+           ).to(torch.bfloat16)
+    vs2 = (torch.rand(B, Hkv, Smax, 2, device=_dev()) * 0.02 + 0.005
+           ).to(torch.bfloat16)
+    pos2 = torch.tensor([100, 5], dtype=torch.int32, device=_dev())
+    out = ops.attn_decode(q, kc2, vc2, pos2, D ** -0.5, 0, None, ks2, vs2)
+    out_ref = ref.attn_decode(q.cpu(), kc2.cpu(), vc2.cpu(), pos2.cpu(),
+                              D ** -0.5, 0, None, ks2.cpu(), vs2.cpu())
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=3e-2,
+                          rtol=3e-2)
