@@ -207,6 +207,35 @@ def test_attn_decode_window_sinks(window, use_sinks):
                           rtol=3e-2)
 
 
+def _clone_model(src, dst):
+    """Copy every layer tensor/Linear of src into dst (cross-device) —
+    init_random uses device RNG, so same-seed CPU/GPU weights differ."""
+    from dnet_amd.models import Linear
+
+    def conv(x, dev):
+        if isinstance(x, torch.Tensor):
+            return x.detach().to(dev)
+        if isinstance(x, Linear):
+            return Linear(conv(x.w, dev), conv(x.bias, dev) if x.bias is not None else None,
+                          conv(x.scales, dev) if x.scales is not None else None,
+                          x.group, x.packed)
+        if isinstance(x, list):
+            return [conv(e, dev) for e in x]
+        return x
+
+    dev = dst.device
+    for lid, lw in src.layers.items():
+        tgt = dst.layers[lid]
+        for name, val in vars(lw).items():
+            setattr(tgt, name, conv(val, dev))
+    if src.embed is not None:
+        dst.embed = conv(src.embed, dev)
+    if src.final_norm is not None:
+        dst.final_norm = conv(src.final_norm, dev)
+    if src.lm_head is not None:
+        dst.lm_head = conv(src.lm_head, dev)
+
+
 def test_gpt_oss_decode_gpu():
     """gpt-oss decode on the HIP kernel path matches the CPU reference."""
     from dnet_amd.models import ModelConfig, get_ring_model
@@ -224,6 +253,7 @@ def test_gpt_oss_decode_gpu():
         return m
 
     mg, mc = build("cuda:0"), build("cpu")
+    _clone_model(mg, mc)  # identical weights on both devices
     toks = torch.randint(0, 256, (2, 10))
     kvg, kvc = mg.make_kv_cache(2, 64), mc.make_kv_cache(2, 64)
     hg = mg.embed_tokens(toks.cuda()).clone()
