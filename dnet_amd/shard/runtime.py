@@ -166,6 +166,18 @@ class ShardRuntime:
             host, _, port = req.api_callback_address.rpartition(":")
             self._callback = SyncWireClient(host or "127.0.0.1", int(port))
         self.status = "loaded"
+        self.link_profile = {}
+        if req.world_size > 1:
+            try:
+                from ..parallel.profiler import measure_ring_links
+                self.link_profile = measure_ring_links(
+                    req.rank, req.world_size, ex.device,
+                    sizes=(65536,), reps=5)
+                for size, r in self.link_profile.items():
+                    log.info("[PROFILE][XGMI] payload=%d latency=%.3fms "
+                             "bw=%.1fGB/s", size, r["latency_ms"], r["gbps"])
+            except Exception:
+                log.exception("xGMI link probe failed (non-fatal)")
         log.info("model loaded: %s rank %d/%d layers %s", self.model_name,
                  req.rank, req.world_size, req.layers[:4])
 

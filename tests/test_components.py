@@ -210,3 +210,27 @@ def test_compression_roundtrip():
     assert torch.equal(y, y2)
     assert len(blob) < x.numel() * 2  # actually smaller
     assert cz.is_compressed_dtype(cz.dtype_string("bfloat16", 0.5))
+
+
+def test_memory_pool():
+    from dnet_amd.core.memory_pool import LayerAwareMemoryPool
+    pool = LayerAwareMemoryPool(max_bytes=1024)
+    a = pool.acquire((8, 8), torch.bfloat16)
+    pool.release(a)
+    b = pool.acquire_for_layer(3, (8, 8), torch.bfloat16)
+    assert b.data_ptr() == a.data_ptr()      # reused
+    assert pool.hits == 1 and pool.misses == 1
+    assert pool.layer_stats[3]["count"] == 1
+    # budget eviction: release more than max_bytes
+    for _ in range(12):
+        pool.release(torch.empty(8, 8, dtype=torch.bfloat16))
+    assert pool.free_bytes <= 1024
+
+
+def test_serialization_roundtrip():
+    from dnet_amd.utils.serialization import bytes_to_tensor, tensor_to_bytes
+    for dtype in (torch.bfloat16, torch.float32, torch.int32):
+        t = (torch.randn(3, 5) * 10).to(dtype)
+        data, name, shape = tensor_to_bytes(t)
+        t2 = bytes_to_tensor(data, name, shape)
+        assert torch.equal(t, t2), dtype
