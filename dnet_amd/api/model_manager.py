@@ -68,6 +68,7 @@ class ModelManager:
                     model_name=entry.preset or entry.id,
                     total_layers=topology.num_layers,
                     layers=[l for round_ in a.layers for l in round_],
+                    layer_rounds=a.layers,
                     next_node=a.next_instance,
                     window_size=a.window_size,
                     residency_size=a.residency_size,

@@ -65,6 +65,7 @@ class ShardLoadModelRequest(BaseModel):
     model_name: str = ""
     total_layers: int = 0
     layers: list[int] = Field(default_factory=list)   # flattened local layers
+    layer_rounds: Optional[list] = None               # per-round windows (k>1)
     next_node: str = ""
     window_size: int = 0
     residency_size: int = 0

@@ -490,6 +490,8 @@ class RingModel:
         """One decode step over a window of local layers. h: [B, H] bf16
         (residual stream, modified in place). kv.pos must already hold the
         write position for this token."""
+        if not layer_ids:
+            return h
         c = self.cfg
         B = h.shape[0]
         nq, nkv, d = self.nq_local, self.nkv_local, c.head_dim
@@ -531,6 +533,8 @@ class RingModel:
                        kv: KVCache, p0: int) -> torch.Tensor:
         """Prefill T tokens. h: [B, T, H]; tokens occupy positions p0..p0+T-1.
         kv.pos is advanced by the caller after the full shard window."""
+        if not layer_ids:
+            return h
         c = self.cfg
         B, T, H = h.shape
         nq, nkv, d = self.nq_local, self.nkv_local, c.head_dim

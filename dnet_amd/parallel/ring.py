@@ -78,8 +78,15 @@ class RingExecutor:
         self.stage = rank // tp
         self.tp_rank = rank % tp
         self.plan = plan or RingPlan.contiguous(cfg.num_layers, self.stages)
-        assert self.plan.rounds == 1, "k>1 rounds land with the solver milestone"
-        self.my_layers = [l for w in self.plan.assignments[self.stage] for l in w]
+        self.rounds = self.plan.rounds
+        # per-round layer windows for this stage (padded to k with empty
+        # pass-through windows); one token step = k laps of the ring
+        # (reference: api/utils.py compute_layer_assignments k-round
+        # interleaving per prima.cpp)
+        self.windows = list(self.plan.assignments[self.stage])
+        while len(self.windows) < self.rounds:
+            self.windows.append([])
+        self.my_layers = [l for w in self.windows for l in w]
         self.mb_count = mb_count
         self.mb_size = mb_size
         self.smax = smax
@@ -165,12 +172,14 @@ class RingExecutor:
 
     # ------------- one-rank step bodies (graph-capturable) -------------
 
-    def _decode_body(self, mb: int):
-        if self.is_first:
+    def _decode_body(self, mb: int, r: int = 0):
+        if self.is_first and r == 0:
             self.hbuf[mb].copy_(
                 torch.nn.functional.embedding(self.tokbuf[mb], self.model.embed))
-        self.model.decode_window(self.hbuf[mb], self.my_layers, self.kvs[mb])
-        if self.is_last:
+        if self.windows[r]:
+            self.model.decode_window(self.hbuf[mb], self.windows[r],
+                                     self.kvs[mb])
+        if self.is_last and r == self.rounds - 1:
             self.logits_buf[mb].copy_(
                 self.model.normalize_project(self.hbuf[mb]))
 
@@ -180,32 +189,35 @@ class RingExecutor:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for mb in range(self.mb_count):
-                for _ in range(2):  # warmup
-                    self._decode_body(mb)
+                for r in range(self.rounds):
+                    for _ in range(2):  # warmup
+                        self._decode_body(mb, r)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         # warmup advanced nothing persistent except KV garbage at pos; KV pos
         # unchanged (we don't advance pos in the body), cache rows at pos get
         # rewritten by real steps.
+        self._graphs = {}
         for mb in range(self.mb_count):
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g, pool=pool):
-                self._decode_body(mb)
-            self._graphs.append(g)
-        log.info("captured %d decode graphs", self.mb_count)
+            for r in range(self.rounds):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g, pool=pool):
+                    self._decode_body(mb, r)
+                self._graphs[(mb, r)] = g
+        log.info("captured %d decode graphs", len(self._graphs))
 
-    def _run_decode(self, mb: int):
+    def _run_decode(self, mb: int, r: int = 0):
         if self.use_graphs and not self._graphs:
             try:
                 self._capture_graphs()
             except Exception as e:  # pragma: no cover
                 log.warning("hipGraph capture failed (%s); falling back to eager", e)
                 self.use_graphs = False
-                self._graphs = []
+                self._graphs = {}
         if self.use_graphs:
-            self._graphs[mb].replay()
+            self._graphs[(mb, r)].replay()
         else:
-            self._decode_body(mb)
+            self._decode_body(mb, r)
 
     # ------------- collective schedules -------------
 
@@ -219,16 +231,25 @@ class RingExecutor:
         H = self.cfg.hidden_size
         tok_reqs = []
         first_tokens = torch.zeros(M, B, dtype=torch.int64, device=self.device)
+        last_r = self.rounds - 1
         for mb in range(M):
-            if self.is_first:
-                h = self.model.embed_tokens(tokens[mb].to(self.device)).clone()
-            else:
-                h = torch.empty(B, T, H, dtype=torch.bfloat16, device=self.device)
-                self.ring.recv(h)
-            self.model.prefill_window(h, self.my_layers, self.kvs[mb], 0)
+            h = None
+            for r in range(self.rounds):
+                if self.is_first and r == 0:
+                    h = self.model.embed_tokens(tokens[mb].to(self.device)).clone()
+                else:
+                    if h is None:
+                        h = torch.empty(B, T, H, dtype=torch.bfloat16,
+                                        device=self.device)
+                    self.ring.recv(h)
+                if self.windows[r]:
+                    self.model.prefill_window(h, self.windows[r],
+                                              self.kvs[mb], 0)
+                if not (self.is_last and r == last_r):
+                    if self.stages > 1:
+                        self.ring.send(h)
             self.kvs[mb].pos.fill_(T)
             if not self.is_last:
-                self.ring.send(h)  # stages > 1 by construction here
                 if self.is_first:
                     tok_reqs.append(self.ring.irecv(first_tokens[mb],
                                                     src=self.token_src))
@@ -257,6 +278,7 @@ class RingExecutor:
                            device=self.device) if (collect and self.is_first)
                else None)
         tok_req: dict[int, object] = {}
+        last_r = self.rounds - 1
         for s in range(n):
             for mb in range(M):
                 if self.is_first:
@@ -264,16 +286,21 @@ class RingExecutor:
                         tok_req.pop(mb).wait()
                     if out is not None and s > 0:
                         out[mb, :, s - 1] = self.tokbuf[mb]
-                    self._run_decode(mb)
+                    for r in range(self.rounds):
+                        if r > 0:
+                            self._recv_hidden(mb)   # lap wrap from last stage
+                        self._run_decode(mb, r)
+                        if self.stages > 1 and not (self.is_last and r == last_r):
+                            self._send_hidden(mb)
                     if self.stages > 1:
-                        self._send_hidden(mb)
                         tok_req[mb] = self.ring.irecv(self.tokbuf[mb],
                                                       src=self.token_src)
                 if self.stages > 1 and not self.is_first:
-                    self._recv_hidden(mb)
-                    self._run_decode(mb)
-                    if not self.is_last:
-                        self._send_hidden(mb)
+                    for r in range(self.rounds):
+                        self._recv_hidden(mb)
+                        self._run_decode(mb, r)
+                        if not (self.is_last and r == last_r):
+                            self._send_hidden(mb)
                 if self.is_last:
                     tok, _, _ = self.sampler.sample(self.logits_buf[mb].float())
                     if self.stages > 1:
@@ -320,15 +347,20 @@ class RingExecutor:
 
         produced = 0
         for s in range(max_tokens - 1):
+            last_r = self.rounds - 1
             if self.is_first:
-                self._run_decode(mb)
-                if self.stages > 1:
-                    self._send_hidden(mb)
+                for r in range(self.rounds):
+                    if r > 0:
+                        self._recv_hidden(mb)
+                    self._run_decode(mb, r)
+                    if self.stages > 1 and not (self.is_last and r == last_r):
+                        self._send_hidden(mb)
             elif self.stages > 1:
-                self._recv_hidden(mb)
-                self._run_decode(mb)
-                if not self.is_last:
-                    self._send_hidden(mb)
+                for r in range(self.rounds):
+                    self._recv_hidden(mb)
+                    self._run_decode(mb, r)
+                    if not (self.is_last and r == last_r):
+                        self._send_hidden(mb)
             if self.is_last:
                 tok, logprob, tops = self.sampler.sample(
                     self.logits_buf[mb].float())

@@ -135,13 +135,17 @@ class ShardRuntime:
                 backend=backend,
                 init_method=f"tcp://{req.master_addr}:{req.master_port}",
                 rank=req.rank, world_size=req.world_size)
-        plan = RingPlan([[sorted(req.layers)]] if req.world_size == 1 else None)
-        if req.world_size > 1:
-            # every rank gets the full plan via its own request's layer list;
-            # build a world-sized plan with only our slot filled (the executor
-            # only reads its own slot)
-            plan = RingPlan([[[]] for _ in range(req.world_size)])
-            plan.assignments[req.rank] = [sorted(req.layers)]
+        rounds = req.layer_rounds or [sorted(req.layers)]
+        if req.world_size == 1:
+            plan = RingPlan([rounds])
+        else:
+            # every rank gets the full plan via its own request's windows;
+            # the executor only reads its own slot (pad others to the same
+            # round count so plan.rounds is consistent)
+            k = len(rounds)
+            plan = RingPlan([[[] for _ in range(k)]
+                             for _ in range(req.world_size)])
+            plan.assignments[req.rank] = rounds
         synthetic = not (Path(req.model_path).expanduser() / "config.json").exists()
         residency = req.residency_size if \
             0 < req.residency_size < len(req.layers) else 0

@@ -186,3 +186,42 @@ def _rank_main_pp_tp(rank, world, port, q):
     if rank == 0:
         q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
     dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_two_rank_two_rounds_matches_single():
+    """k=2 ring rounds (interleaved layer windows, two laps per token) must
+    equal the single-rank run (prima.cpp-style k-round pipelining)."""
+    single = _run_single()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_rounds, args=(r, 2, 29711, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert torch.equal(out, single), f"k2 != single:\n{out}\n{single}"
+
+
+def _rank_main_rounds(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor, RingPlan
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(CFG)
+    # tiny has 4 layers: rank0 rounds [0],[2]; rank1 rounds [1],[3]
+    plan = RingPlan([[[0], [2]], [[1], [3]]])
+    ex = RingExecutor(cfg, rank, world, "cpu", plan=plan, mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
