@@ -36,10 +36,11 @@ def main():
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="qwen-2.5-32b")
     ap.add_argument("--quant", default="int8", choices=["int8", "bf16"])
-    ap.add_argument("--mb-size", type=int, default=16,
+    ap.add_argument("--mb-size", type=int, default=32,
                     help="sequences per microbatch")
-    ap.add_argument("--mb-per-rank", type=int, default=2,
-                    help="microbatches = mb_per_rank * world (pipeline fill)")
+    ap.add_argument("--mb-per-rank", type=int, default=0,
+                    help="microbatches = mb_per_rank * world (pipeline fill); "
+                         "0 = auto (1 on a single GPU, 2 per rank otherwise)")
     ap.add_argument("--prompt-len", type=int, default=128)
     ap.add_argument("--smax", type=int, default=1024)
     ap.add_argument("--layers", type=int, default=0,
@@ -62,7 +63,8 @@ def main():
         hf["num_hidden_layers"] = args.layers
     cfg = ModelConfig.from_hf(hf, quant=quant)
 
-    mb_count = max(args.mb_per_rank * world, 1)
+    mb_per_rank = args.mb_per_rank or (1 if world == 1 else 2)
+    mb_count = max(mb_per_rank * world, 1)
     ex = RingExecutor(cfg, rank, world, device, mb_count=mb_count,
                       mb_size=args.mb_size, smax=args.smax, seed=1234,
                       use_graphs=on_gpu and not args.no_graphs and args.tp == 1,

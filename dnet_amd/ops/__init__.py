@@ -56,14 +56,14 @@ def rmsnorm(x: torch.Tensor, residual: torch.Tensor | None, w: torch.Tensor,
     return ref.rmsnorm(x, residual, w, eps)
 
 
-# split-K f32 scratch, cached per device (max 16 rows x N<8192 cols)
+# split-K f32 scratch, cached per device (max 64 rows x N<8192 cols)
 _scratch: dict = {}
 
 
 def _get_scratch(device) -> torch.Tensor:
     key = str(device)
     if key not in _scratch:
-        _scratch[key] = torch.empty(16 * 8192, dtype=torch.float32, device=device)
+        _scratch[key] = torch.empty(64 * 8192, dtype=torch.float32, device=device)
     return _scratch[key]
 
 
@@ -71,7 +71,7 @@ def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
               bias: torch.Tensor | None = None) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        if x.shape[1] % 32 == 0:
+        if x.shape[1] % 64 == 0:
             _native().gemm_m16(x, w, None, bias, out, _get_scratch(x.device),
                                0, False)
         else:
@@ -85,11 +85,10 @@ def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
               packed: bool = False) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        if x.shape[1] % 32 == 0 and group % 8 == 0:
+        if packed:
             _native().gemm_m16(x, w, scales, bias, out, _get_scratch(x.device),
-                               group, packed)
+                               group, True)
         else:
-            assert not packed, "packed layout needs the MFMA path"
             _native().gemv_int8(x, w, scales, out, group, bias)
         return out
     if packed:

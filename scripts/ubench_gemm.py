@@ -43,7 +43,7 @@ def main():
         qp = ops.pack_int8_mfma(q)
         wbytes_i8 = q.numel() + scales.numel() * 2
         wbytes_bf = wf.numel() * 2
-        for M in (1, 8, 16):
+        for M in (1, 16, 32, 64):
             x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
             out = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
             scratch = ops._get_scratch(dev)
