@@ -22,8 +22,9 @@ import torch
 from .. import ops
 from .config import ModelConfig, QuantConfig
 
-# Decode GEMV handles M<=16 tiles; beyond that the dequant+GEMM path wins.
-GEMV_MAX_M = 16
+# Decode GEMM handles M<=64 (stacked MFMA M-tiles); beyond that the
+# dequant+hipBLASLt path wins (prefill).
+GEMV_MAX_M = 64
 
 
 class Linear:

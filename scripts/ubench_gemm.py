@@ -48,15 +48,12 @@ def main():
             out = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
             scratch = ops._get_scratch(dev)
 
-            t_mfma = bench(lambda: nat.gemm_m16(x, q, scales, None, out,
-                                                scratch, 128, False))
             t_pk = bench(lambda: nat.gemm_m16(x, qp, scales, None, out,
                                               scratch, 128, True))
             t_old = bench(lambda: nat.gemv_int8(x, q, scales, out, 128, None))
             t_mfma_bf = bench(lambda: nat.gemm_m16(x, wf, None, None, out,
                                                    scratch, 0, False))
             print(f"{name:8s} N={N:6d} K={K:6d} M={M:2d}  "
-                  f"i8-mfma {wbytes_i8/t_mfma/1e9:6.0f} | "
                   f"i8-packed {t_pk*1e6:7.1f}us {wbytes_i8/t_pk/1e9:6.0f}GB/s | "
                   f"i8-scalar {wbytes_i8/t_old/1e9:6.0f} | "
                   f"bf16-mfma {t_mfma_bf*1e6:7.1f}us {wbytes_bf/t_mfma_bf/1e9:6.0f}GB/s")
