@@ -36,7 +36,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="qwen-2.5-32b")
     ap.add_argument("--quant", default="int8", choices=["int8", "bf16"])
-    ap.add_argument("--mb-size", type=int, default=32,
+    ap.add_argument("--mb-size", type=int, default=64,
                     help="sequences per microbatch")
     ap.add_argument("--mb-per-rank", type=int, default=0,
                     help="microbatches = mb_per_rank * world (pipeline fill); "

@@ -42,7 +42,7 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
                                 float* __restrict__ out_f32,    // SPLITK>1
                                 const int M, const int K, const int N,
                                 const int G, const int splitk) {
-  constexpr int XT = 1024 / MT;        // k values per x tile (LDS ~33 KB)
+  constexpr int XT = 1024 / MT;        // k per x tile (LDS ~33 KB, 4 blocks/CU)
   constexpr int SE = XT + 8;           // row stride in bf16 elems (16B pad)
   __shared__ short x_lds[16 * MT * SE];
   const int wave = threadIdx.x / kWave;
