@@ -107,6 +107,14 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
     return ref.dequant_int8(w, scales, group)
 
 
+def _attn_splits(b: int, hkv: int, smax: int) -> int:
+    # fill ~1024 blocks (4/CU); each split needs >= 1 chunk of 64 positions
+    splits = 1
+    while (b * hkv * splits * 2 <= 1024 and splits * 2 <= (smax + 63) // 64):
+        splits *= 2
+    return splits
+
+
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 pos: torch.Tensor, scale: float, window: int = 0,
                 sinks: torch.Tensor | None = None,
@@ -115,8 +123,14 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     if q.is_cuda:
         # q may be a strided slice of the fused QKV buffer; out is contiguous.
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        B, Hq, D = q.shape
+        splits = _attn_splits(B, kcache.shape[1], kcache.shape[2])
+        partials = None
+        if splits > 1:
+            partials = torch.empty(B * Hq * splits * (D + 2),
+                                   dtype=torch.float32, device=q.device)
         _native().attn_decode(q, kcache, vcache, pos, out, scale, window,
-                              sinks, kscale, vscale)
+                              sinks, kscale, vscale, partials, splits)
         return out
     return ref.attn_decode(q, kcache, vcache, pos, scale, window, sinks,
                            kscale, vscale)

@@ -36,6 +36,11 @@ __device__ __forceinline__ float wave_allreduce_max(float v) {
 
 constexpr int kChunk = 64;
 
+// splits > 1: flash-decode S-partitioning — gridDim.y splits each take
+// chunks c = c0 + split, c0 + split + splits, ...; unnormalized partials
+// (acc, m, l) go to scratch [B, Hq, splits, D+2] f32 and a combine kernel
+// merges them (sink logit folded there). Raises the block count from
+// B*Hkv to B*Hkv*splits so small batches still fill 256 CUs.
 template <int D, bool Q8>
 __global__ void attn_decode_kernel(const short* __restrict__ q,
                                    const void* __restrict__ kc,
@@ -47,7 +52,9 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
                                    const int Hkv, const int Smax,
                                    const float scale, const int ldq,
                                    const int window,
-                                   const short* __restrict__ sinks) {
+                                   const short* __restrict__ sinks,
+                                   float* __restrict__ partials,
+                                   const int splits) {
   constexpr int P = lds_pitch(D);
   constexpr int DPL = D / kWave;  // output dims per lane (1 or 2)
   const int b = blockIdx.x / Hkv;
@@ -85,7 +92,8 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   __syncthreads();
 
   const int nchunks = (len + kChunk - 1) / kChunk;
-  for (int c = start / kChunk; c < nchunks; ++c) {
+  const int split = blockIdx.y;
+  for (int c = start / kChunk + split; c < nchunks; c += splits) {
     const int s0 = c * kChunk;
     const int valid = min(kChunk, len - s0);
     // Stage K and V chunk into bf16 LDS; Q8 caches dequantize while staging
@@ -164,6 +172,21 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
     __syncthreads();
   }
 
+  if (splits > 1) {
+    // write unnormalized partials; combine kernel normalizes + sinks
+    for (int hi = 0; hi < nh; ++hi) {
+      const int g = wid + hi * 4;
+      float* prow = partials
+          + (((int64_t)b * Hq + hkv * G + g) * splits + split) * (D + 2);
+#pragma unroll
+      for (int j = 0; j < DPL; ++j) prow[DPL * lane + j] = acc[hi][j];
+      if (lane == 0) {
+        prow[D] = m[hi];
+        prow[D + 1] = l[hi];
+      }
+    }
+    return;
+  }
   for (int hi = 0; hi < nh; ++hi) {
     const int g = wid + hi * 4;
     float inv;
@@ -184,12 +207,44 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   }
 }
 
+// merge the per-split unnormalized partials: one 64-lane block per (b, h).
+template <int D>
+__global__ void attn_combine_kernel(const float* __restrict__ partials,
+                                    const short* __restrict__ sinks,
+                                    short* __restrict__ out, const int Hq,
+                                    const int splits) {
+  constexpr int DPL = D / kWave;
+  const int bh = blockIdx.x;          // b * Hq + h
+  const int h = bh % Hq;
+  const int lane = threadIdx.x;
+  const float* base = partials + (int64_t)bh * splits * (D + 2);
+  float M = -1e30f;
+  for (int s = 0; s < splits; ++s) M = fmaxf(M, base[s * (D + 2) + D]);
+  if (sinks != nullptr) M = fmaxf(M, bits2f(sinks[h]));
+  float denom = (sinks != nullptr) ? __expf(bits2f(sinks[h]) - M) : 0.f;
+  float acc[DPL];
+#pragma unroll
+  for (int j = 0; j < DPL; ++j) acc[j] = 0.f;
+  for (int s = 0; s < splits; ++s) {
+    const float* prow = base + s * (D + 2);
+    const float w = __expf(prow[D] - M);
+    denom += prow[D + 1] * w;
+#pragma unroll
+    for (int j = 0; j < DPL; ++j) acc[j] += prow[DPL * lane + j] * w;
+  }
+  const float inv = denom > 0.f ? 1.f / denom : 0.f;
+  short* orow = out + (int64_t)bh * D + DPL * lane;
+#pragma unroll
+  for (int j = 0; j < DPL; ++j) orow[j] = f2bits(acc[j] * inv);
+}
+
 // q may be a column slice of a fused-QKV buffer: strides (ldq, D, 1).
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale,
                  int64_t window, c10::optional<torch::Tensor> sinks,
                  c10::optional<torch::Tensor> kscale,
-                 c10::optional<torch::Tensor> vscale) {
+                 c10::optional<torch::Tensor> vscale,
+                 c10::optional<torch::Tensor> partials, int64_t splits) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = kcache.size(1), Smax = kcache.size(2);
   DNET_CHECK(kcache.size(0) == B && kcache.size(3) == D, "kcache shape");
@@ -203,24 +258,41 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const int G = (int)(Hq / Hkv);
   const int P = (int)D + 4;
   const size_t lds = (2 * kChunk * P + G * D) * sizeof(short);
-  const dim3 grid((unsigned)(B * Hkv));
+  if (splits > 1) {
+    DNET_CHECK(partials.has_value()
+                   && partials->numel() >= B * Hq * splits * (D + 2),
+               "split-S partials scratch required");
+  }
+  const dim3 grid((unsigned)(B * Hkv), (unsigned)splits);
   const int ldq = (int)q.stride(0);
   const short* skp = sinks.has_value() ? (const short*)sinks->data_ptr()
                                        : nullptr;
   const bool q8 = kcache.dtype() == torch::kInt8;
   const short* ksp = q8 ? (const short*)kscale->data_ptr() : nullptr;
   const short* vsp = q8 ? (const short*)vscale->data_ptr() : nullptr;
+  float* pp = splits > 1 ? (float*)partials->data_ptr() : nullptr;
 #define LAUNCH_ATTN(DD, QQ)                                                   \
   hipLaunchKernelGGL((attn_decode_kernel<DD, QQ>), grid, dim3(256), lds,      \
                      stream, (const short*)q.data_ptr(), kcache.data_ptr(),   \
                      vcache.data_ptr(), ksp, vsp, (const int*)pos.data_ptr(), \
                      (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,    \
-                     (float)scale, ldq, (int)window, skp)
+                     (float)scale, ldq, (int)window,                          \
+                     (splits > 1 ? nullptr : skp), pp, (int)splits)
   if (D == 128 && q8) LAUNCH_ATTN(128, true);
   else if (D == 128) LAUNCH_ATTN(128, false);
   else if (q8) LAUNCH_ATTN(64, true);
   else LAUNCH_ATTN(64, false);
 #undef LAUNCH_ATTN
+  if (splits > 1) {
+    if (D == 128)
+      hipLaunchKernelGGL((attn_combine_kernel<128>), dim3((unsigned)(B * Hq)),
+                         dim3(kWave), 0, stream, pp, skp,
+                         (short*)out.data_ptr(), (int)Hq, (int)splits);
+    else
+      hipLaunchKernelGGL((attn_combine_kernel<64>), dim3((unsigned)(B * Hq)),
+                         dim3(kWave), 0, stream, pp, skp,
+                         (short*)out.data_ptr(), (int)Hq, (int)splits);
+  }
 }
 
 }  // namespace dnet

@@ -13,7 +13,8 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale,
                  int64_t window, c10::optional<torch::Tensor> sinks,
                  c10::optional<torch::Tensor> kscale,
-                 c10::optional<torch::Tensor> vscale);
+                 c10::optional<torch::Tensor> vscale,
+                 c10::optional<torch::Tensor> partials, int64_t splits);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
                  torch::Tensor cos_table, torch::Tensor sin_table,

@@ -311,3 +311,17 @@ def test_kv8_kernels_vs_ref():
                               D ** -0.5, 0, None, ks2.cpu(), vs2.cpu())
     assert torch.allclose(out.float().cpu(), out_ref.float(), atol=3e-2,
                           rtol=3e-2)
+
+
+def test_attn_decode_split_s_long():
+    """Long-context decode exercises the split-S partials + combine path."""
+    torch.manual_seed(21)
+    B, Hq, Hkv, D, Smax = 2, 40, 8, 128, 2048
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=_dev())
+    kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=_dev())
+    vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=_dev())
+    pos = torch.tensor([2000, 37], dtype=torch.int32, device=_dev())
+    out = ops.attn_decode(q, kc, vc, pos, D ** -0.5)
+    out_ref = ref.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), pos.cpu(), D ** -0.5)
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=3e-2,
+                          rtol=3e-2)
