@@ -293,6 +293,12 @@ class RingModel:
         # offload policy hook: when set, layer weights come from the
         # windowed weight cache instead of self.layers
         self.weight_provider = None  # Callable[[int], LayerWeights] | None
+        # observability: forced per-layer sync + [PROFILE][LAYER] timing
+        # (reference: core/observability.py sync_per_layer / sync_every_n);
+        # effective only on the eager (non-graph) path.
+        from ..config import get_settings
+        obs = get_settings().observability
+        self._obs_sync = obs.enabled and obs.sync_per_layer
 
     def _layer(self, lid: int) -> "LayerWeights":
         if self.weight_provider is not None:
@@ -522,8 +528,22 @@ class RingModel:
             o = self._tp_reduce(lw.o(attn.view(B, nq * d)))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
             delta = self._tp_reduce(self._mlp(y2, lw))
+            if self._obs_sync:
+                self._profile_layer_sync(lid)
         h.add_(delta)
         return h
+
+    def _profile_layer_sync(self, lid: int):
+        import time
+
+        from ..utils.logger import logger
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        now = time.perf_counter()
+        last = getattr(self, "_obs_t0", now)
+        logger.info("[PROFILE][LAYER] layer=%d ms=%.3f", lid,
+                    (now - last) * 1e3)
+        self._obs_t0 = time.perf_counter()
 
     def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
         gu = lw.gateup(y)

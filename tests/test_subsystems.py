@@ -1,0 +1,157 @@
+"""Subsystem tests with fakes (reference tier: tests/subsystems/ — servers
+and managers exercised without network or GPU)."""
+import asyncio
+import json
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from dnet_amd.api.cluster import ClusterManager
+from dnet_amd.api.models import ChatRequestModel
+from dnet_amd.api.server import ApiState, build_api_app
+from dnet_amd.config import get_settings
+from dnet_amd.core.sampler import DecodingConfig, Sampler
+from dnet_amd.utils.hostfile import DeviceProperties
+
+
+class FakeDiscovery:
+    """Static device map (reference: tests/fakes/discovery.py)."""
+
+    def __init__(self, devices):
+        self._devices = {d.instance: d for d in devices}
+
+    async def async_get_properties(self):
+        return dict(self._devices)
+
+    async def async_start(self):
+        pass
+
+
+def _devices(n=2):
+    return [DeviceProperties(instance=f"shard{i}", local_ip="127.0.0.1",
+                             server_port=8081 + i, shard_port=50052 + i,
+                             gpu_index=i) for i in range(n)]
+
+
+@pytest.fixture
+def api_client():
+    cluster = ClusterManager(FakeDiscovery(_devices()))
+    state = ApiState(cluster, get_settings())
+    return TestClient(build_api_app(state)), state
+
+
+def test_models_and_devices_routes(api_client):
+    client, _ = api_client
+    ids = [m["id"] for m in client.get("/v1/models").json()["data"]]
+    assert "tiny-random" in ids and "qwen-2.5-32b-int8-synthetic" in ids
+    devs = client.get("/v1/devices").json()
+    assert set(devs) == {"shard0", "shard1"}
+
+
+def test_manual_topology_validation(api_client):
+    client, state = api_client
+    # missing layers rejected
+    r = client.post("/v1/prepare_topology_manual", json={
+        "model": "tiny-random",
+        "assignments": [{"instance": "shard0", "layers": [0, 1]}]})
+    assert r.status_code == 400
+    # unknown device rejected
+    r = client.post("/v1/prepare_topology_manual", json={
+        "model": "tiny-random",
+        "assignments": [{"instance": "nope", "layers": [0, 1, 2, 3]}]})
+    assert r.status_code == 400
+    # valid topology, ring auto-closed, re-fetchable
+    r = client.post("/v1/prepare_topology_manual", json={
+        "model": "tiny-random",
+        "assignments": [{"instance": "shard0", "layers": [0, 1]},
+                        {"instance": "shard1", "layers": [2, 3]}]})
+    assert r.status_code == 200
+    topo = r.json()
+    assert topo["assignments"][1]["next_instance"] == "shard0"
+    assert client.get("/v1/topology").json() == topo
+
+
+def test_chat_requires_model(api_client):
+    client, _ = api_client
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-random", "messages": [{"role": "user", "content": "x"}]})
+    assert r.status_code == 400
+
+
+def test_solver_topology_from_profiles():
+    from dnet_amd.api.model_manager import resolve_model_config
+    from dnet_amd.api.catalog import get_entry
+    from dnet_amd.parallel.profiler import DeviceProfile
+    cluster = ClusterManager(FakeDiscovery(_devices(3)))
+    asyncio.run(cluster.scan_devices())
+    for i, bw in enumerate([6000.0, 6000.0, 2000.0]):
+        cluster.profiles[f"shard{i}"] = DeviceProfile(
+            instance=f"shard{i}", hbm_gbps=bw, h2d_gbps=50, hbm_free_gb=280)
+    cfg = resolve_model_config(get_entry("qwen-2.5-32b-int8-synthetic"))
+    topo = cluster.solve_topology("qwen-2.5-32b-int8-synthetic", cfg)
+    w = topo.solution["w"]
+    assert sum(w) == 64 and w[2] < w[0]
+    assert topo.assignments[0].next_instance == topo.assignments[1].instance
+    head = cluster.get_head_node()
+    assert head is not None and head.instance == topo.assignments[0].instance
+
+
+def test_inference_manager_token_flow():
+    """Token frames resolved through the pending map -> SSE chunk stream
+    (reference: test_inference_manager with FakeStrategyAdapter)."""
+    from dnet_amd.api.inference import InferenceManager
+    from dnet_amd.api.tokenizer import ByteTokenizer
+
+    class MM:
+        tokenizer = ByteTokenizer(512)
+        stop_ids = [ByteTokenizer(512).EOS]
+
+    class FakeHead:
+        def __init__(self, im):
+            self.im = im
+
+        async def request(self, frame):
+            async def feed():
+                nonce = frame["nonce"]
+                for ch in b"ok":
+                    self.im.resolve_token({"t": "token", "nonce": nonce,
+                                           "token_id": int(ch),
+                                           "finished": False})
+                self.im.resolve_token({"t": "token", "nonce": nonce,
+                                       "token_id": MM.stop_ids[0],
+                                       "finished": True})
+            asyncio.get_event_loop().create_task(feed())
+            return {"t": "ack"}
+
+    im = InferenceManager(MM(), token_timeout_s=10)
+    im.head_client = FakeHead(im)
+    im.callback_addr = "127.0.0.1:1"
+
+    async def run():
+        req = ChatRequestModel(model="tiny-random", profile=True,
+                               messages=[{"role": "user", "content": "hi"}])
+        return await im.chat_completions(req)
+
+    resp = asyncio.run(run())
+    assert resp.choices[0].message.content == "ok"
+    assert resp.usage.completion_tokens == 3
+    assert resp.metrics["tokens_generated"] == 3
+
+
+def test_sampler_modes():
+    torch.manual_seed(0)
+    logits = torch.tensor([[0.1, 5.0, 0.2, 0.3]])
+    tok, lp, tops = Sampler(DecodingConfig()).sample(logits)
+    assert int(tok[0]) == 1 and lp is None
+    tok, lp, tops = Sampler(DecodingConfig(temperature=0.0, logprobs=True,
+                                           top_logprobs=2)).sample(logits)
+    assert lp is not None and len(tops[0]) == 2
+    s = Sampler(DecodingConfig(temperature=1.0, top_k=1))
+    for _ in range(5):
+        tok, _, _ = s.sample(logits)
+        assert int(tok[0]) == 1  # top-1 restricted
+    s = Sampler(DecodingConfig(temperature=1.0, top_p=0.01))
+    for _ in range(5):
+        tok, _, _ = s.sample(logits)
+        assert int(tok[0]) == 1
