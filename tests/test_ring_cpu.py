@@ -4,9 +4,18 @@ each other for the same seed — the layer-seeded random init guarantees
 identical global weights under any sharding."""
 import multiprocessing as mp
 import os
+import socket
 
 import pytest
 import torch
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
 
 from dnet_amd.models import ModelConfig, PRESETS
 from dnet_amd.parallel.ring import RingExecutor, RingPlan, split_layers
@@ -64,7 +73,7 @@ def test_two_rank_ring_matches_single():
     single = _run_single()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29671
+    port = _free_port()
     procs = [ctx.Process(target=_rank_main, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
@@ -82,7 +91,8 @@ def test_two_rank_ring_compressed_hops():
     tokens (lossy, so no exact-match; shapes/flow must hold)."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main_compressed, args=(r, 2, 29681, q))
+    port = _free_port()
+    procs = [ctx.Process(target=_rank_main_compressed, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -120,7 +130,8 @@ def test_tp2_matches_single():
     single = _run_single()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main_tp, args=(r, 2, 29691, q))
+    port = _free_port()
+    procs = [ctx.Process(target=_rank_main_tp, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -157,7 +168,8 @@ def test_pp2_tp2_matches_single():
     single = _run_single()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main_pp_tp, args=(r, 4, 29701, q))
+    port = _free_port()
+    procs = [ctx.Process(target=_rank_main_pp_tp, args=(r, 4, port, q))
              for r in range(4)]
     for p in procs:
         p.start()
@@ -195,7 +207,8 @@ def test_two_rank_two_rounds_matches_single():
     single = _run_single()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main_rounds, args=(r, 2, 29711, q))
+    port = _free_port()
+    procs = [ctx.Process(target=_rank_main_rounds, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()
