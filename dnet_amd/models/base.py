@@ -22,9 +22,9 @@ import torch
 from .. import ops
 from .config import ModelConfig, QuantConfig
 
-# Decode GEMM handles M<=64 (stacked MFMA M-tiles); beyond that the
-# dequant+hipBLASLt path wins (prefill).
-GEMV_MAX_M = 64
+# The MFMA decode GEMM loops stacked 64-row passes up to M=256; above that
+# the dequant+hipBLASLt path reads less weight traffic (large prefill).
+GEMV_MAX_M = 256
 
 
 class Linear:
