@@ -22,9 +22,10 @@ import torch
 from .. import ops
 from .config import ModelConfig, QuantConfig
 
-# The MFMA decode GEMM loops stacked 64-row passes up to M=256; above that
-# the dequant+hipBLASLt path reads less weight traffic (large prefill).
-GEMV_MAX_M = 256
+# Decode GEMM handles M<=64 (stacked MFMA M-tiles); beyond that the
+# dequant+hipBLASLt path wins (measured: looping 64-row passes at M=128
+# raises TTFT ~45% vs dequant+matmul).
+GEMV_MAX_M = 64
 
 
 class Linear:
