@@ -85,11 +85,12 @@ def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
               packed: bool = False) -> torch.Tensor:
     if x.is_cuda:
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
-        if packed:
+        if packed and x.shape[0] > 2:
             _native().gemm_m16(x, w, scales, bias, out, _get_scratch(x.device),
                                group, True)
         else:
-            _native().gemv_int8(x, w, scales, out, group, bias)
+            # M<=2: the scalar GEMV reads 16B/lane and hits 4-5.6 TB/s
+            _native().gemv_int8(x, w, scales, out, group, bias, packed)
         return out
     if packed:
         w = ref.unpack_int8_mfma(w)
@@ -108,9 +109,11 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
 
 
 def _attn_splits(b: int, hkv: int, smax: int) -> int:
-    # fill ~1024 blocks (4/CU); each split needs >= 1 chunk of 64 positions
+    # fill ~1024 blocks (4/CU); cap so each split owns >= ~512 positions of
+    # capacity (short actual sequences leave idle splits + combine overhead)
     splits = 1
-    while (b * hkv * splits * 2 <= 1024 and splits * 2 <= (smax + 63) // 64):
+    cap = max(1, smax // 512)
+    while (b * hkv * splits * 2 <= 1024 and splits * 2 <= cap):
         splits *= 2
     return splits
 

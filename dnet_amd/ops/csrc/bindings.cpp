@@ -8,7 +8,7 @@ void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out,
                c10::optional<torch::Tensor> bias);
 void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
                torch::Tensor out, int64_t group,
-               c10::optional<torch::Tensor> bias);
+               c10::optional<torch::Tensor> bias, bool packed);
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale,
                  int64_t window, c10::optional<torch::Tensor> sinks,

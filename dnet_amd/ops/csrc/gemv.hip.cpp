@@ -55,7 +55,10 @@ __global__ void gemv_bf16_kernel(const short* __restrict__ x,
   }
 }
 
-template <int M>
+// PACKED: weights in pack_int8_mfma chunk-pair order — the dot is
+// order-independent, so only the x vector indices change (vec i covers
+// orig k = pr*64 + sl*8 (+32): pr = i/4, sl = i%4).
+template <int M, bool PACKED>
 __global__ void gemv_int8_kernel(const short* __restrict__ x,
                                  const int8_t* __restrict__ w,
                                  const short* __restrict__ scales,
@@ -82,7 +85,15 @@ __global__ void gemv_int8_kernel(const short* __restrict__ x,
 #pragma unroll
     for (int m = 0; m < M; ++m) {
       const short8* xrow = reinterpret_cast<const short8*>(x + (int64_t)m * ldx);
-      const short8 x0 = xrow[2 * i], x1 = xrow[2 * i + 1];
+      short8 x0, x1;
+      if (PACKED) {
+        const int pr = i / 4, sl = i % 4;
+        x0 = xrow[pr * 8 + sl];
+        x1 = xrow[pr * 8 + sl + 4];
+      } else {
+        x0 = xrow[2 * i];
+        x1 = xrow[2 * i + 1];
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         acc[m] = fmaf(wq[j], bits2f(x0.x[j]), acc[m]);
@@ -145,10 +156,11 @@ void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out,
 
 void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
                torch::Tensor out, int64_t group,
-               c10::optional<torch::Tensor> bias) {
+               c10::optional<torch::Tensor> bias, bool packed) {
   const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
   DNET_CHECK(w.size(1) == K && out.size(1) == N, "shape");
   DNET_CHECK(K % 16 == 0 && group % 16 == 0 && K % group == 0, "K/group align");
+  if (packed) DNET_CHECK(K % 64 == 0 && group % 64 == 0, "packed align");
   DNET_CHECK(scales.size(0) == N && scales.size(1) == K / group, "scales shape");
   DNET_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous() &&
                  scales.is_contiguous(), "contig");
@@ -162,13 +174,22 @@ void gemv_int8(torch::Tensor x, torch::Tensor w, torch::Tensor scales,
     else if (mt > 12 && mt < 16) mt = 12;
     else if (mt == 7) mt = 6;
     dispatch_m(mt, [&](auto mc) {
-      hipLaunchKernelGGL((gemv_int8_kernel<decltype(mc)::value>), dim3(grid),
-                         dim3(kRowsPerBlock * kWave), 0, stream,
-                         (const short*)x.data_ptr() + m0 * K,
-                         (const int8_t*)w.data_ptr(),
-                         (const short*)scales.data_ptr(), bptr,
-                         (short*)out.data_ptr() + m0 * N, (int)K, (int)N,
-                         (int)group, (int)K);
+      if (packed)
+        hipLaunchKernelGGL((gemv_int8_kernel<decltype(mc)::value, true>),
+                           dim3(grid), dim3(kRowsPerBlock * kWave), 0, stream,
+                           (const short*)x.data_ptr() + m0 * K,
+                           (const int8_t*)w.data_ptr(),
+                           (const short*)scales.data_ptr(), bptr,
+                           (short*)out.data_ptr() + m0 * N, (int)K, (int)N,
+                           (int)group, (int)K);
+      else
+        hipLaunchKernelGGL((gemv_int8_kernel<decltype(mc)::value, false>),
+                           dim3(grid), dim3(kRowsPerBlock * kWave), 0, stream,
+                           (const short*)x.data_ptr() + m0 * K,
+                           (const int8_t*)w.data_ptr(),
+                           (const short*)scales.data_ptr(), bptr,
+                           (short*)out.data_ptr() + m0 * N, (int)K, (int)N,
+                           (int)group, (int)K);
     });
     m0 += mt;
   }
