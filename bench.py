@@ -35,7 +35,7 @@ def main():
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="qwen-2.5-32b")
-    ap.add_argument("--quant", default="int8", choices=["int8", "bf16"])
+    ap.add_argument("--quant", default="int8", choices=["int8", "int4", "bf16"])
     ap.add_argument("--mb-size", type=int, default=64,
                     help="sequences per microbatch")
     ap.add_argument("--mb-per-rank", type=int, default=0,
@@ -57,7 +57,11 @@ def main():
     rank, world, device = init_from_env()
     on_gpu = device.type == "cuda"
 
-    quant = QuantConfig(8, 128) if args.quant == "int8" else None
+    quant = None
+    if args.quant == "int8":
+        quant = QuantConfig(8, 128)
+    elif args.quant == "int4":
+        quant = QuantConfig(4, 128)
     hf = dict(PRESETS[args.model])
     if args.layers:
         hf["num_hidden_layers"] = args.layers
@@ -122,7 +126,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if quant is None else "int8-w8a16(bf16 compute)",
+            "dtype": "bf16" if quant is None else f"int{quant.bits}-g{quant.group} weights (bf16 compute)",
             "data": "synthetic (random tokens, random-init weights)",
             "ttft_ms": round(ttft_ms, 1),
             "config": {

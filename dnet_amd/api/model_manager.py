@@ -22,9 +22,10 @@ def resolve_model_config(entry: CatalogEntry, quant: str = "") -> ModelConfig:
     import json
     q = quant or entry.quant
     qc = None
-    if q.startswith("int8"):
+    if q.startswith("int"):
+        bits = int(q[3])
         group = int(q.split("-g")[1]) if "-g" in q else 128
-        qc = QuantConfig(8, group)
+        qc = QuantConfig(bits, group)
     if entry.preset:
         return ModelConfig.from_hf(dict(PRESETS[entry.preset]), quant=qc)
     p = Path(entry.repo).expanduser()

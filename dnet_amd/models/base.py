@@ -57,6 +57,8 @@ class Linear:
     def in_features(self) -> int:
         return self.w.shape[1]
 
+    bits: int = 16
+
     @classmethod
     def make(cls, w: torch.Tensor, bias: Optional[torch.Tensor],
              quant: Optional[QuantConfig]) -> "Linear":
@@ -66,18 +68,38 @@ class Linear:
                       and quant.group % 64 == 0)
             if packed:
                 q = ops.pack_int8_mfma(q)
-            return cls(q, bias, s, quant.group, packed)
+            l = cls(q, bias, s, quant.group, packed)
+            l.bits = 8
+            return l
+        if quant is not None and quant.bits == 4:
+            if w.shape[1] % 128 or quant.group % 128:
+                # int4 needs quad alignment; fall back to int8 for this proj
+                return cls.make(w, bias, QuantConfig(8, max(quant.group, 64)))
+            q, s = ops.quantize_int4(w, quant.group)
+            packed = w.is_cuda
+            if packed:
+                q = ops.pack_int4_mfma(q)
+            l = cls(q.to(w.device), bias, s.to(w.device), quant.group, packed)
+            l.bits = 4
+            return l
         return cls(w.to(torch.bfloat16), bias)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         m = x.shape[0]
         if m <= GEMV_MAX_M:
+            if self.bits == 4:
+                return ops.gemv_int4(x, self.w, self.scales, self.group,
+                                     self.bias, self.packed)
             if self.is_quant:
                 return ops.gemv_int8(x, self.w, self.scales, self.group,
                                      self.bias, self.packed)
             return ops.gemv_bf16(x, self.w, self.bias)
-        wd = (ops.dequant_int8(self.w, self.scales, self.group, self.packed)
-              if self.is_quant else self.w)
+        if self.bits == 4:
+            wd = ops.dequant_int4(self.w, self.scales, self.group, self.packed)
+        elif self.is_quant:
+            wd = ops.dequant_int8(self.w, self.scales, self.group, self.packed)
+        else:
+            wd = self.w
         y = x @ wd.t()
         if self.bias is not None:
             y = y + self.bias

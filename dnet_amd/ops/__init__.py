@@ -73,7 +73,7 @@ def gemv_bf16(x: torch.Tensor, w: torch.Tensor,
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
         if x.shape[1] % 64 == 0:
             _native().gemm_m16(x, w, None, bias, out, _get_scratch(x.device),
-                               0, False)
+                               0, False, 16)
         else:
             _native().gemv_bf16(x, w, out, bias)
         return out
@@ -87,7 +87,7 @@ def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
         out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
         if packed and x.shape[0] > 2:
             _native().gemm_m16(x, w, scales, bias, out, _get_scratch(x.device),
-                               group, True)
+                               group, True, 8)
         else:
             # M<=2: the scalar GEMV reads 16B/lane and hits 4-5.6 TB/s
             _native().gemv_int8(x, w, scales, out, group, bias, packed)
@@ -95,6 +95,29 @@ def gemv_int8(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
     if packed:
         w = ref.unpack_int8_mfma(w)
     return ref.gemv_int8(x, w, scales, group, bias)
+
+
+def gemv_int4(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+              group: int, bias: torch.Tensor | None = None,
+              packed: bool = False) -> torch.Tensor:
+    if x.is_cuda:
+        assert packed, "GPU int4 path needs the packed layout"
+        out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
+        _native().gemm_m16(x, w, scales, bias, out, _get_scratch(x.device),
+                           group, True, 4)
+        return out
+    return ref.gemv_int4(x, w, scales, group, bias)
+
+
+def dequant_int4(w: torch.Tensor, scales: torch.Tensor, group: int,
+                 packed: bool = True) -> torch.Tensor:
+    if w.is_cuda:
+        assert packed
+        out = torch.empty(w.shape[0], w.shape[1] * 2, dtype=torch.bfloat16,
+                          device=w.device)
+        _native().dequant_int4(w, scales, out, group)
+        return out
+    return ref.dequant_int4(w, scales, group)
 
 
 def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
@@ -158,7 +181,9 @@ def swiglu(gu: torch.Tensor) -> torch.Tensor:
 
 
 quantize_int8 = ref.quantize_int8
+quantize_int4 = ref.quantize_int4
 pack_int8_mfma = ref.pack_int8_mfma
+pack_int4_mfma = ref.pack_int4_mfma
 unpack_int8_mfma = ref.unpack_int8_mfma
 rope_tables = ref.rope_tables
 rope_apply = ref.rope_apply

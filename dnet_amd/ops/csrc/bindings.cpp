@@ -23,10 +23,13 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 void swiglu(torch::Tensor gu, torch::Tensor y);
 void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group, bool packed);
+void dequant_int4(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
+                  int64_t group);
 void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
-              c10::optional<torch::Tensor> scratch, int64_t group, bool packed);
+              c10::optional<torch::Tensor> scratch, int64_t group, bool packed,
+              int64_t bits);
 void col_norm2(torch::Tensor x, torch::Tensor norms);
 void gather_cols(torch::Tensor x, torch::Tensor idx, torch::Tensor out);
 void scatter_cols(torch::Tensor in, torch::Tensor idx, torch::Tensor out);
@@ -41,6 +44,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_append", &dnet::rope_append, "fused RoPE + KV append (decode)");
   m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");
+  m.def("dequant_int4", &dnet::dequant_int4, "packed-int4 -> bf16 dequant");
   m.def("gemm_m16", &dnet::gemm_m16, "MFMA decode GEMM (M<=16, bf16 or int8)");
   m.def("col_norm2", &dnet::col_norm2, "per-column L2 norms");
   m.def("gather_cols", &dnet::gather_cols, "pack kept columns");

@@ -31,9 +31,22 @@ __device__ __forceinline__ bf16x8 deq8(const int8_t* q, const float s) {
   return b;
 }
 
-// QUANT: int8 packed weights (pack_int8_mfma layout), else bf16.
+// 4 packed nibble-bytes (offset-8) -> 8 bf16 values * scale
+__device__ __forceinline__ bf16x8 deq4(const uint8_t* qb, const float s) {
+  bf16x8 b;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int byte = qb[j];
+    b[2 * j] = (__bf16)((float)((byte & 0xF) - 8) * s);
+    b[2 * j + 1] = (__bf16)((float)((byte >> 4) - 8) * s);
+  }
+  return b;
+}
+
+// QBITS: 16 = bf16 weights, 8 = packed int8 (pack_int8_mfma chunk-pair
+// layout), 4 = packed int4 (pack_int4_mfma chunk-quad layout).
 // MT: number of stacked 16-row A tiles (M <= 16*MT).
-template <bool QUANT, int MT>
+template <int QBITS, int MT>
 __global__ void gemm_m16_kernel(const short* __restrict__ x,
                                 const void* __restrict__ w,
                                 const short* __restrict__ scales,
@@ -55,9 +68,19 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   const int n_w = min(n0 + row, N - 1);     // this lane's W row
 
   const int pairs = K / 64;
-  const int pp = pairs / splitk;
-  const int p_begin = blockIdx.y * pp;
-  const int p_end = (blockIdx.y == splitk - 1) ? pairs : p_begin + pp;
+  int p_begin, p_end;
+  if (QBITS == 4) {
+    // partition at quad (128-k) granularity so every split starts on an
+    // even pair (the int4 layout packs four chunks per lane load)
+    const int quads = K / 128;
+    const int qq = quads / splitk;
+    p_begin = blockIdx.y * qq * 2;
+    p_end = (blockIdx.y == splitk - 1) ? quads * 2 : p_begin + qq * 2;
+  } else {
+    const int pp = pairs / splitk;
+    p_begin = blockIdx.y * pp;
+    p_end = (blockIdx.y == splitk - 1) ? pairs : p_begin + pp;
+  }
   const int woff = (lane >> 4) * 16;
 
   f32x4 acc[MT][2];
@@ -67,8 +90,9 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
     for (int u = 0; u < 2; ++u) acc[t][u] = {0.f, 0.f, 0.f, 0.f};
 
   const int8_t* wrow_q = (const int8_t*)w + (int64_t)n_w * K;
+  const uint8_t* wrow_q4 = (const uint8_t*)w + (int64_t)n_w * (K / 2);
   const short* wrow_b = (const short*)w + (int64_t)n_w * K;
-  const short* srow = QUANT ? scales + (int64_t)n_w * (K / G) : nullptr;
+  const short* srow = QBITS < 16 ? scales + (int64_t)n_w * (K / G) : nullptr;
 
   for (int k0 = p_begin * 64; k0 < p_end * 64; k0 += XT) {
     const int tk = min(XT, p_end * 64 - k0);
@@ -91,12 +115,26 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
         const int p = (k0 / 64) + pl + u;
-        if (QUANT) {
+        if (QBITS == 8) {
           const int4 wq = *reinterpret_cast<const int4*>(&wrow_q[p * 64 + woff]);
           const int8_t* q8 = reinterpret_cast<const int8_t*>(&wq);
           const float sv = bits2f(srow[(p * 64) / G]);
           b[u][0] = deq8(q8, sv);
           b[u][1] = deq8(q8 + 8, sv);
+        } else if (QBITS == 4) {
+          // chunk-quad layout: one b128 covers this lane's slices of four
+          // chunks; load once per quad (even u), use halves per pair.
+          if (u == 0) {
+            const int quad = p / 2;
+            const int4 wq = *reinterpret_cast<const int4*>(
+                &wrow_q4[quad * 64 + woff]);
+            const uint8_t* q4 = reinterpret_cast<const uint8_t*>(&wq);
+            const float sv = bits2f(srow[(quad * 128) / G]);
+            b[0][0] = deq4(q4, sv);
+            b[0][1] = deq4(q4 + 4, sv);
+            b[1][0] = deq4(q4 + 8, sv);
+            b[1][1] = deq4(q4 + 12, sv);
+          }
         } else {
           b[u][0] = *reinterpret_cast<const bf16x8*>(
               &wrow_b[p * 64 + ks]);
@@ -107,7 +145,7 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
         for (int t = 0; t < MT; ++t) {
           const short* at = arow + t * 16 * SE;
           a[u][0][t] = *reinterpret_cast<const bf16x8*>(
-              &at[(pl + u) * 64 + (QUANT ? 0 : 0) + ks]);
+              &at[(pl + u) * 64 + ks]);
           a[u][1][t] = *reinterpret_cast<const bf16x8*>(
               &at[(pl + u) * 64 + 32 + ks]);
         }
@@ -125,12 +163,21 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
     for (; pl < pl_end; ++pl) {
       const int p = (k0 / 64) + pl;
       bf16x8 b0, b1;
-      if (QUANT) {
+      if (QBITS == 8) {
         const int4 wq = *reinterpret_cast<const int4*>(&wrow_q[p * 64 + woff]);
         const int8_t* q8 = reinterpret_cast<const int8_t*>(&wq);
         const float sv = bits2f(srow[(p * 64) / G]);
         b0 = deq8(q8, sv);
         b1 = deq8(q8 + 8, sv);
+      } else if (QBITS == 4) {
+        const int quad = p / 2;
+        const int half = p & 1;
+        const int2 wq = *reinterpret_cast<const int2*>(
+            &wrow_q4[quad * 64 + woff + half * 8]);
+        const uint8_t* q4 = reinterpret_cast<const uint8_t*>(&wq);
+        const float sv = bits2f(srow[(quad * 128) / G]);
+        b0 = deq4(q4, sv);
+        b1 = deq4(q4 + 4, sv);
       } else {
         b0 = *reinterpret_cast<const bf16x8*>(&wrow_b[p * 64 + ks]);
         b1 = *reinterpret_cast<const bf16x8*>(&wrow_b[p * 64 + 32 + ks]);
@@ -196,10 +243,9 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
                        c10::optional<torch::Tensor> scales,
                        c10::optional<torch::Tensor> bias, torch::Tensor out,
                        c10::optional<torch::Tensor> scratch, int group,
-                       int64_t m0, int M) {
+                       int64_t m0, int M, int bits) {
   const int64_t K = x.size(1), N = w.size(0);
   auto stream = current_stream();
-  const bool quant = scales.has_value();
   const int64_t scratch_elems = scratch.has_value() ? scratch->numel() : 0;
   const int sk = pick_splitk(N, K, M, scratch_elems);
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
@@ -213,20 +259,24 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
     fp = (float*)scratch->data_ptr();
     DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * M * N, stream));
   }
-  const short* sp = quant ? (const short*)scales->data_ptr() : nullptr;
+  const short* sp = bits < 16 ? (const short*)scales->data_ptr() : nullptr;
   const short* bp1 = sk > 1 ? nullptr : bptr;
 #define LAUNCH(QQ, TT)                                                      \
   hipLaunchKernelGGL((gemm_m16_kernel<QQ, TT>), grid, dim3(256), 0, stream, \
                      xp, w.data_ptr(), sp, bp1, op, fp, M, (int)K, (int)N,  \
                      group, sk)
-  if (quant) {
-    if (M > 32) LAUNCH(true, 4);
-    else if (M > 16) LAUNCH(true, 2);
-    else LAUNCH(true, 1);
+  if (bits == 8) {
+    if (M > 32) LAUNCH(8, 4);
+    else if (M > 16) LAUNCH(8, 2);
+    else LAUNCH(8, 1);
+  } else if (bits == 4) {
+    if (M > 32) LAUNCH(4, 4);
+    else if (M > 16) LAUNCH(4, 2);
+    else LAUNCH(4, 1);
   } else {
-    if (M > 32) LAUNCH(false, 4);
-    else if (M > 16) LAUNCH(false, 2);
-    else LAUNCH(false, 1);
+    if (M > 32) LAUNCH(16, 4);
+    else if (M > 16) LAUNCH(16, 2);
+    else LAUNCH(16, 1);
   }
 #undef LAUNCH
   if (sk > 1) {
@@ -241,20 +291,28 @@ void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
               c10::optional<torch::Tensor> scratch, int64_t group,
-              bool packed) {
+              bool packed, int64_t bits) {
   const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
   DNET_CHECK(K % 64 == 0, "K % 64 == 0 required for the MFMA path");
-  DNET_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N, "shape");
+  DNET_CHECK(out.size(0) == M && out.size(1) == N, "shape");
   DNET_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous(), "contig");
-  if (scales.has_value()) {
-    DNET_CHECK(packed, "int8 MFMA path expects the packed weight layout");
+  if (bits == 16) {
+    DNET_CHECK(w.size(1) == K, "w shape");
+  } else {
+    DNET_CHECK(packed, "quantized MFMA path expects the packed weight layout");
+    DNET_CHECK(scales.has_value() && scales->is_contiguous(), "scales");
     DNET_CHECK(group % 64 == 0 && K % group == 0, "group align");
-    DNET_CHECK(scales->is_contiguous(), "scales contig");
+    if (bits == 8) DNET_CHECK(w.size(1) == K, "w shape (int8)");
+    if (bits == 4) {
+      DNET_CHECK(w.size(1) == K / 2 && K % 128 == 0 && group % 128 == 0,
+                 "w shape / align (int4)");
+    }
   }
   int64_t m0 = 0;
   while (m0 < M) {
     const int mt = (int)std::min<int64_t>(M - m0, 64);
-    launch_m16(x, w, scales, bias, out, scratch, (int)group, m0, mt);
+    launch_m16(x, w, scales, bias, out, scratch, (int)group, m0, mt,
+               (int)bits);
     m0 += mt;
   }
 }

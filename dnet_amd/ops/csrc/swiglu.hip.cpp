@@ -97,4 +97,40 @@ void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                        (short*)out.data_ptr(), K, (int)group, N);
 }
 
+// int4 chunk-quad packed -> bf16 (prefill path). Packed byte at
+// quad*64 + slice*16 + chunk*4 + jb holds orig k = quad*128 + chunk*32 +
+// slice*8 + 2*jb (low nibble) and +1 (high nibble), offset-8.
+__global__ void dequant_int4_kernel(const uint8_t* __restrict__ w,
+                                    const short* __restrict__ scales,
+                                    short* __restrict__ out, const int64_t K,
+                                    const int G, const int64_t N) {
+  const int64_t nbytes = N * K / 2;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       idx < nbytes; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t n = idx / (K / 2);
+    const int64_t bi = idx % (K / 2);
+    const int64_t quad = bi / 64;
+    const int rem = (int)(bi % 64);
+    const int slice = rem / 16, chunk = (rem % 16) / 4, jb = rem % 4;
+    const int64_t k0 = quad * 128 + chunk * 32 + slice * 8 + 2 * jb;
+    const uint8_t byte = w[idx];
+    const float s = bits2f(scales[n * (K / G) + k0 / G]);
+    out[n * K + k0] = f2bits((float)((byte & 0xF) - 8) * s);
+    out[n * K + k0 + 1] = f2bits((float)((byte >> 4) - 8) * s);
+  }
+}
+
+void dequant_int4(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
+                  int64_t group) {
+  const int64_t N = w.size(0), K = out.size(1);
+  DNET_CHECK(w.size(1) == K / 2 && K % 128 == 0 && group % 128 == 0, "shape");
+  DNET_CHECK(w.is_contiguous() && scales.is_contiguous() && out.is_contiguous(), "contig");
+  auto stream = current_stream();
+  const int64_t nbytes = N * K / 2;
+  const int grid = (int)std::min<int64_t>((nbytes + 255) / 256, 2048);
+  hipLaunchKernelGGL(dequant_int4_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const uint8_t*)w.data_ptr(), (const short*)scales.data_ptr(),
+                     (short*)out.data_ptr(), K, (int)group, N);
+}
+
 }  // namespace dnet

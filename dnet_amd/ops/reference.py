@@ -99,6 +99,62 @@ def dequant_kv(codes: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
     return xf.view(*lead, d).to(torch.bfloat16)
 
 
+def quantize_int4(w: torch.Tensor, group: int = 128
+                  ) -> tuple[torch.Tensor, torch.Tensor]:
+    """Symmetric per-group int4 quantization along K: values in [-7, 7]
+    stored offset-by-8 as nibbles, two per byte (even k = low nibble)."""
+    n, k = w.shape
+    assert k % group == 0 and k % 2 == 0
+    wf = w.float().view(n, k // group, group)
+    amax = wf.abs().amax(dim=-1).clamp_min(1e-8)
+    scales = (amax / 7.0).to(torch.bfloat16)
+    q = torch.round(wf / scales.float().unsqueeze(-1)).clamp(-7, 7)
+    q = (q + 8).to(torch.uint8).view(n, k)
+    packed = (q[:, 0::2] | (q[:, 1::2] << 4)).contiguous()
+    return packed, scales
+
+
+def unpack_int4(q4: torch.Tensor) -> torch.Tensor:
+    """[N, K/2] packed nibbles -> [N, K] int8 in [-7, 7]."""
+    n, kb = q4.shape
+    lo = (q4 & 0xF).to(torch.int8) - 8
+    hi = (q4 >> 4).to(torch.int8) - 8
+    out = torch.empty(n, kb * 2, dtype=torch.int8)
+    out[:, 0::2] = lo
+    out[:, 1::2] = hi
+    return out
+
+
+def dequant_int4(q4: torch.Tensor, scales: torch.Tensor, group: int
+                 ) -> torch.Tensor:
+    return dequant_int8(unpack_int4(q4.cpu()), scales, group)
+
+
+def gemv_int4(x: torch.Tensor, q4: torch.Tensor, scales: torch.Tensor,
+              group: int, bias: torch.Tensor | None = None) -> torch.Tensor:
+    wd = dequant_int4(q4, scales, group)
+    y = x.float() @ wd.float().t()
+    if bias is not None:
+        y = y + bias.float()
+    return y.to(x.dtype)
+
+
+def pack_int4_mfma(q4: torch.Tensor) -> torch.Tensor:
+    """Permute packed-nibble rows into MFMA chunk-quad order: a lane's
+    16 B load covers its 8-elem B slices of FOUR adjacent K=32 chunks.
+
+    unpacked orig [N, quad, chunk(4), slice(4), j(8)] ->
+    packed nibble bytes [N, quad, slice, chunk, j/2]."""
+    n, kb = q4.shape
+    k = kb * 2
+    assert k % 128 == 0
+    vals = unpack_int4(q4.cpu()) + 8         # [N, K] in [1,15]
+    v = vals.view(n, k // 128, 4, 4, 8)      # [N, quad, chunk, slice, j]
+    v = v.permute(0, 1, 3, 2, 4).reshape(n, k)  # slice-major
+    packed = (v[:, 0::2] | (v[:, 1::2] << 4)).to(torch.uint8)
+    return packed.contiguous().to(q4.device)
+
+
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 pos: torch.Tensor, scale: float, window: int = 0,
                 sinks: torch.Tensor | None = None,

@@ -108,9 +108,10 @@ class ShardRuntime:
 
     def _resolve_config(self, req: ShardLoadModelRequest) -> ModelConfig:
         quant = None
-        if req.quant.startswith("int8"):
+        if req.quant.startswith("int"):
+            bits = int(req.quant[3])
             group = int(req.quant.split("-g")[1]) if "-g" in req.quant else 128
-            quant = QuantConfig(8, group)
+            quant = QuantConfig(bits, group)
         p = Path(req.model_path).expanduser()
         if (p / "config.json").exists():
             return ModelConfig.from_hf(json.loads((p / "config.json").read_text()),

@@ -57,3 +57,17 @@ def test_pack_unpack_int8_mfma():
     p = ref.pack_int8_mfma(q)
     assert not torch.equal(p, q)
     assert torch.equal(ref.unpack_int8_mfma(p), q)
+
+
+def test_quantize_int4_roundtrip():
+    torch.manual_seed(2)
+    w = torch.randn(8, 256, dtype=torch.bfloat16)
+    q4, s = ref.quantize_int4(w, 128)
+    assert q4.shape == (8, 128) and q4.dtype == torch.uint8
+    wd = ref.dequant_int4(q4, s, 128)
+    assert (wd.float() - w.float()).abs().max().item() < 0.5
+    # pack/unpack consistency through the chunk-quad layout
+    p = ref.pack_int4_mfma(q4)
+    vals = ref.unpack_int4(p) + 8
+    v = vals.view(8, 2, 4, 4, 8).permute(0, 1, 3, 2, 4).reshape(8, 256)
+    assert torch.equal(v, ref.unpack_int4(q4) + 8)

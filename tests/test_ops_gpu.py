@@ -325,3 +325,40 @@ def test_attn_decode_split_s_long():
     out_ref = ref.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), pos.cpu(), D ** -0.5)
     assert torch.allclose(out.float().cpu(), out_ref.float(), atol=3e-2,
                           rtol=3e-2)
+
+
+@pytest.mark.parametrize("m,n,k", [(1, 512, 1024), (8, 1000, 5120),
+                                   (64, 4096, 27648)])
+def test_gemm_int4(m, n, k):
+    """int4 chunk-quad MFMA path vs the CPU reference."""
+    torch.manual_seed(31)
+    group = 128
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=_dev())
+    wf = torch.randn(n, k, dtype=torch.bfloat16, device=_dev()) / 30
+    q4, scales = ops.quantize_int4(wf.cpu(), group)
+    q4p = ops.pack_int4_mfma(q4).to(_dev())
+    out = ops.gemv_int4(x, q4p, scales.to(_dev()), group, None, packed=True)
+    out_ref = ops.ref.gemv_int4(x.cpu(), q4, scales, group)
+    assert torch.allclose(out.float().cpu(), out_ref.float(), atol=8e-2,
+                          rtol=4e-2)
+    wd = ops.dequant_int4(q4p, scales.to(_dev()), group)
+    wd_ref = ops.ref.dequant_int4(q4, scales, group)
+    assert torch.allclose(wd.float().cpu(), wd_ref.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_model_int4_consistency_gpu():
+    """End-to-end int4 model decode vs its own bf16 weights (quality check)."""
+    from dnet_amd.models import ModelConfig, PRESETS, QuantConfig, get_ring_model
+    hf = dict(PRESETS["tiny"])
+    cfg4 = ModelConfig.from_hf(hf, quant=QuantConfig(4, 128))
+    m = get_ring_model(cfg4.model_type)(cfg4, range(4), "cuda:0", True, True,
+                                        smax=64)
+    m.init_random(5)
+    kv = m.make_kv_cache(2, 64)
+    toks = torch.randint(0, cfg4.vocab_size, (2, 8), device="cuda:0")
+    h = m.embed_tokens(toks).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    kv.pos.fill_(8)
+    hd = m.embed_tokens(toks[:, -1]).clone()
+    m.decode_window(hd, m.layer_ids, kv)
+    assert torch.isfinite(m.normalize_project(hd).float()).all()
