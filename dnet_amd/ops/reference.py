@@ -115,11 +115,11 @@ def quantize_int4(w: torch.Tensor, group: int = 128
 
 
 def unpack_int4(q4: torch.Tensor) -> torch.Tensor:
-    """[N, K/2] packed nibbles -> [N, K] int8 in [-7, 7]."""
+    """[N, K/2] packed nibbles -> [N, K] int8 in [-7, 7] (device-preserving)."""
     n, kb = q4.shape
     lo = (q4 & 0xF).to(torch.int8) - 8
     hi = (q4 >> 4).to(torch.int8) - 8
-    out = torch.empty(n, kb * 2, dtype=torch.int8)
+    out = torch.empty(n, kb * 2, dtype=torch.int8, device=q4.device)
     out[:, 0::2] = lo
     out[:, 1::2] = hi
     return out
@@ -127,7 +127,7 @@ def unpack_int4(q4: torch.Tensor) -> torch.Tensor:
 
 def dequant_int4(q4: torch.Tensor, scales: torch.Tensor, group: int
                  ) -> torch.Tensor:
-    return dequant_int8(unpack_int4(q4.cpu()), scales, group)
+    return dequant_int8(unpack_int4(q4), scales.to(q4.device), group)
 
 
 def gemv_int4(x: torch.Tensor, q4: torch.Tensor, scales: torch.Tensor,
@@ -148,11 +148,11 @@ def pack_int4_mfma(q4: torch.Tensor) -> torch.Tensor:
     n, kb = q4.shape
     k = kb * 2
     assert k % 128 == 0
-    vals = unpack_int4(q4.cpu()) + 8         # [N, K] in [1,15]
+    vals = unpack_int4(q4) + 8               # [N, K] in [1,15]
     v = vals.view(n, k // 128, 4, 4, 8)      # [N, quad, chunk, slice, j]
     v = v.permute(0, 1, 3, 2, 4).reshape(n, k)  # slice-major
     packed = (v[:, 0::2] | (v[:, 1::2] << 4)).to(torch.uint8)
-    return packed.contiguous().to(q4.device)
+    return packed.contiguous()
 
 
 def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
