@@ -27,6 +27,10 @@ model_catalog: list[CatalogEntry] = [
     CatalogEntry("llama-3-8b-synthetic", preset="llama-3-8b", tokenizer="byte"),
     CatalogEntry("qwen-2.5-32b-int8-synthetic", preset="qwen-2.5-32b",
                  quant="int8-g128", tokenizer="byte"),
+    CatalogEntry("qwen-2.5-32b-int4-synthetic", preset="qwen-2.5-32b",
+                 quant="int4-g128", tokenizer="byte"),
+    CatalogEntry("llama-3-70b-int4-synthetic", preset="llama-3-70b",
+                 quant="int4-g128", tokenizer="byte"),
     CatalogEntry("llama-3-70b-synthetic", preset="llama-3-70b", tokenizer="byte"),
     CatalogEntry("mixtral-8x7b-synthetic", preset="mixtral-8x7b", tokenizer="byte"),
     # real checkpoints (local safetensors dir or pre-downloaded HF cache)
