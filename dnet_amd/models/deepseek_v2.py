@@ -272,15 +272,20 @@ class DeepseekV2RingModel(RingModel):
         if c.norm_topk_prob:
             weights = weights / weights.sum(-1, keepdim=True)
         out = torch.zeros_like(y, dtype=torch.float32)
+        dense = y.shape[0] <= 64  # graph-safe dense path for decode batches
         for e in range(c.num_experts):
-            mask = (idx == e).any(dim=-1)
+            we_full = (weights * (idx == e)).sum(dim=-1)
+            if dense:
+                d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](y)))
+                out += d.float() * we_full.unsqueeze(-1)
+                continue
+            mask = we_full > 0
             if not bool(mask.any()):
                 continue
             rows = mask.nonzero(as_tuple=True)[0]
             xe = y[rows].contiguous()
             d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](xe)))
-            we = (weights * (idx == e)).sum(dim=-1)[rows]
-            out[rows] += d.float() * we.unsqueeze(-1)
+            out[rows] += d.float() * we_full[rows].unsqueeze(-1)
         out *= c.routed_scaling_factor
         if getattr(lw, "shared_gateup", None) is not None:
             out += lw.shared_down(ops.swiglu(lw.shared_gateup(y))).float()

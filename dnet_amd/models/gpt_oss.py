@@ -122,14 +122,21 @@ class GptOssRingModel(MoERingModel):
         vals, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
         weights = torch.softmax(vals, dim=-1)
         out = torch.zeros_like(y, dtype=torch.float32)
+        dense = y.shape[0] <= self.DENSE_MOE_MAX_T
         for e in range(c.num_experts):
-            mask = (idx == e).any(dim=-1)
+            if self.tp_size > 1 and e % self.tp_size != self.tp_rank:
+                continue
+            we_full = (weights * (idx == e)).sum(dim=-1)
+            if dense:
+                d = lw.experts_down[e](gpt_oss_glu(lw.experts_gateup[e](y)))
+                out += d.float() * we_full.unsqueeze(-1)
+                continue
+            mask = we_full > 0
             if not bool(mask.any()):
                 continue
             rows = mask.nonzero(as_tuple=True)[0]
             xe = y[rows].contiguous()
             a = gpt_oss_glu(lw.experts_gateup[e](xe))
             d = lw.experts_down[e](a)
-            we = (weights * (idx == e)).sum(dim=-1)[rows]
-            out[rows] += d.float() * we.unsqueeze(-1)
+            out[rows] += d.float() * we_full[rows].unsqueeze(-1)
         return out.to(y.dtype)

@@ -39,6 +39,11 @@ class MoERingModel(RingModel):
             "MoE safetensors loading lands with the weight-cache milestone; "
             "use init_random for synthetic runs")
 
+    # below this many rows, run every expert on the whole batch and
+    # weight-sum (no data-dependent host syncs -> hipGraph-capturable; the
+    # expert weight read — the decode bound — is once per expert either way)
+    DENSE_MOE_MAX_T = 64
+
     def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
         c = self.cfg
         T = y.shape[0]
@@ -46,17 +51,20 @@ class MoERingModel(RingModel):
         weights, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
         weights = torch.softmax(weights, dim=-1)
         out = torch.zeros_like(y, dtype=torch.float32)
+        dense = T <= self.DENSE_MOE_MAX_T
         for e in range(c.num_experts):
             if self.tp_size > 1 and e % self.tp_size != self.tp_rank:
                 continue  # expert parallelism: partial sum reduced by caller
-            mask = (idx == e).any(dim=-1)
+            we_full = (weights * (idx == e)).sum(dim=-1)
+            if dense:
+                d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](y)))
+                out += d.float() * we_full.unsqueeze(-1)
+                continue
+            mask = we_full > 0
             if not bool(mask.any()):
                 continue
             rows = mask.nonzero(as_tuple=True)[0]
             xe = y[rows].contiguous()
-            gu = lw.experts_gateup[e](xe)
-            a = ops.swiglu(gu)
-            d = lw.experts_down[e](a)
-            we = (weights * (idx == e)).sum(dim=-1)[rows]
-            out[rows] += d.float() * we.unsqueeze(-1)
+            d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](xe)))
+            out[rows] += d.float() * we_full[rows].unsqueeze(-1)
         return out.to(y.dtype)
