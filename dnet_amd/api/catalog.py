@@ -33,6 +33,11 @@ model_catalog: list[CatalogEntry] = [
                  quant="int4-g128", tokenizer="byte"),
     CatalogEntry("llama-3-70b-synthetic", preset="llama-3-70b", tokenizer="byte"),
     CatalogEntry("mixtral-8x7b-synthetic", preset="mixtral-8x7b", tokenizer="byte"),
+    CatalogEntry("gpt-oss-20b-synthetic", preset="gpt-oss-20b", tokenizer="byte"),
+    CatalogEntry("deepseek-v2-lite-synthetic", preset="deepseek-v2-lite",
+                 tokenizer="byte"),
+    CatalogEntry("openai/gpt-oss-20b", repo="openai/gpt-oss-20b"),
+    CatalogEntry("deepseek-ai/DeepSeek-V2-Lite", repo="deepseek-ai/DeepSeek-V2-Lite"),
     # real checkpoints (local safetensors dir or pre-downloaded HF cache)
     CatalogEntry("meta-llama/Llama-3.1-8B-Instruct",
                  repo="meta-llama/Llama-3.1-8B-Instruct"),

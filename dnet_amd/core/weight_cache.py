@@ -37,9 +37,17 @@ class PinnedLayerStore:
         for k, t in tensors.items():
             t = t.detach().cpu().contiguous()
             if self.pin:
-                t = t.pin_memory()
+                try:
+                    t = t.pin_memory()
+                except RuntimeError:
+                    self.pin = False
+                    log.warning("pin_memory failed; host staging unpinned "
+                                "(H2D will be slow)")
             out[k] = t
         self.layers[lid] = out
+        if lid == min(self.layers):
+            log.info("layer store: pinned=%s layer_bytes=%.1fMB", self.pin,
+                     self.layer_bytes(lid) / 1e6)
 
     def layer_bytes(self, lid: int) -> int:
         return sum(t.numel() * t.element_size()

@@ -139,6 +139,23 @@ PRESETS: dict[str, dict] = {
                           num_attention_heads=12, num_key_value_heads=12,
                           head_dim=64, intermediate_size=3072, vocab_size=50272,
                           rope_theta=10000.0),
+    "gpt-oss-20b": dict(model_type="gpt_oss", hidden_size=2880,
+                        num_hidden_layers=24, num_attention_heads=64,
+                        num_key_value_heads=8, head_dim=64,
+                        intermediate_size=2880, vocab_size=201088,
+                        num_local_experts=32, num_experts_per_tok=4,
+                        sliding_window=128, rope_theta=150000.0,
+                        attention_bias=True, rms_norm_eps=1e-5),
+    "deepseek-v2-lite": dict(model_type="deepseek_v2", hidden_size=2048,
+                             num_hidden_layers=27, num_attention_heads=16,
+                             num_key_value_heads=16, vocab_size=102400,
+                             intermediate_size=10944, kv_lora_rank=512,
+                             qk_nope_head_dim=128, qk_rope_head_dim=64,
+                             v_head_dim=128, n_routed_experts=64,
+                             num_experts_per_tok=6, n_shared_experts=2,
+                             moe_intermediate_size=1408,
+                             first_k_dense_replace=1,
+                             routed_scaling_factor=1.0, rope_theta=10000.0),
     "tiny": dict(model_type="llama", hidden_size=128, num_hidden_layers=4,
                  num_attention_heads=2, num_key_value_heads=2, head_dim=64,
                  intermediate_size=256, vocab_size=256, rope_theta=10000.0),
