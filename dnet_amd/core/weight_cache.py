@@ -96,6 +96,13 @@ class WeightCache:
         dst = self.slots[slot]
         t0 = time.perf_counter()
         if self.on_gpu:
+            # order the overwrite after all compute enqueued so far — the
+            # evicted layer's kernels may still be in flight on the compute
+            # stream (copies for layer t+depth start only once compute
+            # through layer t has drained; overlap with t+1.. is preserved)
+            ev_order = torch.cuda.Event()
+            ev_order.record(torch.cuda.current_stream(self.device))
+            self.copy_stream.wait_event(ev_order)
             with torch.cuda.stream(self.copy_stream):
                 for k, t in src.items():
                     dst[k].copy_(t, non_blocking=True)

@@ -36,7 +36,7 @@ class OffloadBinder:
     """Weight provider over a WeightCache with ring-order prefetch."""
 
     def __init__(self, cache: WeightCache, order: Sequence[int], group: int,
-                 packed: bool, prefetch_depth: int = 2):
+                 packed: bool, prefetch_depth: int = 3):
         self.cache = cache
         self.order = list(order)
         self.group = group
@@ -56,7 +56,7 @@ class OffloadBinder:
 
 
 def enable_offload(model: RingModel, residency: int,
-                   prefetch_depth: int = 2) -> WeightCache:
+                   prefetch_depth: int = 3) -> WeightCache:
     """Move the model's layer weights to pinned host memory and install the
     windowed weight cache as the model's weight provider. Returns the cache
     (for stats). MoE layers are not yet offloadable."""
