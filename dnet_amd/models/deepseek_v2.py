@@ -42,7 +42,9 @@ class DeepseekV2RingModel(RingModel):
     model_types = ["deepseek_v2", "deepseek_v3"]
 
     def __init__(self, cfg: ModelConfig, layer_ids, device, is_first, is_last,
-                 smax: int = 4096):
+                 smax: int = 4096, tp_rank: int = 0, tp_size: int = 1,
+                 tp_group=None):
+        assert tp_size == 1, "deepseek MLA TP is roadmap"
         super().__init__(cfg, layer_ids, device, is_first, is_last, smax)
         # rope tables over the rope sub-dim only
         cos, sin = ops.rope_tables(smax, cfg.qk_rope_head_dim, cfg.rope_theta,
