@@ -115,6 +115,11 @@ class WeightCache:
             self.slot_event[slot] = None
         self.slot_layer[slot] = lid
         self.layer_slot[lid] = slot
+        # a fresh prefetch counts as a use — otherwise the LRU evicts the
+        # just-prefetched slot on the next prefetch (thrash: every bind
+        # becomes a synchronous miss)
+        self._clock += 1
+        self._use_tick[slot] = self._clock
         mb = self.store.layer_bytes(lid) / 1e6
         log.info("[PROFILE][PREFETCH] layer=%d ms=%.2f bytes=%.1fMB (issued)",
                  lid, (time.perf_counter() - t0) * 1e3, mb)

@@ -234,3 +234,24 @@ def test_serialization_roundtrip():
         data, name, shape = tensor_to_bytes(t)
         t2 = bytes_to_tensor(data, name, shape)
         assert torch.equal(t, t2), dtype
+
+
+def test_weight_cache_prefetch_no_thrash():
+    """Sequential ring access with prefetch depth 3 must make every bind a
+    hit after warmup (the LRU counts prefetches as uses — otherwise each
+    prefetch evicts the previous one and every bind becomes a sync miss)."""
+    from dnet_amd.core.weight_cache import PinnedLayerStore, WeightCache
+    store = PinnedLayerStore(pin=False)
+    for lid in range(24):
+        store.put_layer(lid, {"w": torch.full((4, 4), float(lid))})
+    cache = WeightCache(store, residency=8, device=torch.device("cpu"))
+    order = list(range(24))
+    for lid in order[:8]:
+        cache.prefetch(lid)
+    for epoch in range(2):
+        for i, lid in enumerate(order):
+            t = cache.bind(lid)["w"]
+            assert float(t[0, 0]) == lid
+            for d in range(1, 4):
+                cache.prefetch(order[(i + d) % 24])
+    assert cache.misses == 0, f"bind misses: {cache.misses}"
