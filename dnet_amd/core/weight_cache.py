@@ -64,11 +64,19 @@ class WeightCache:
     """
 
     def __init__(self, store: PinnedLayerStore, residency: int,
-                 device: torch.device):
+                 device: torch.device, order: Optional[list[int]] = None):
         self.store = store
         self.device = device
         self.on_gpu = device.type == "cuda"
         self.residency = max(residency, 2)
+        # ring order for Belady eviction (sequential cyclic access: evict
+        # the slot whose layer's next use is furthest ahead). Plain LRU
+        # thrashes here: the warm window's ticks are older than bind ticks,
+        # so prefetching layer i+d evicts layer i+1.
+        self.order = list(order) if order else None
+        self._oidx = ({lid: i for i, lid in enumerate(self.order)}
+                      if self.order else {})
+        self._pos = 0
         template = next(iter(store.layers.values()))
         self.slots: list[dict[str, torch.Tensor]] = []
         for _ in range(self.residency):
@@ -83,9 +91,20 @@ class WeightCache:
         self.hits = 0
         self.misses = 0
 
+    def _dist(self, lid: Optional[int]) -> int:
+        # distance (in ring order) to this layer's next use; empty slots
+        # are the best victims of all
+        if lid is None:
+            return 1 << 30
+        n = len(self.order)
+        return (self._oidx.get(lid, 0) - self._pos) % n
+
     def _pick_slot(self) -> int:
-        # LRU over slots (sequential layer access -> effectively a ring)
-        i = min(range(self.residency), key=lambda s: self._use_tick[s])
+        if self.order:
+            i = max(range(self.residency),
+                    key=lambda s: self._dist(self.slot_layer[s]))
+        else:  # LRU fallback for non-ring access patterns
+            i = min(range(self.residency), key=lambda s: self._use_tick[s])
         old = self.slot_layer[i]
         if old is not None:
             self.layer_slot.pop(old, None)
@@ -133,6 +152,8 @@ class WeightCache:
         """Device tensors for layer ``lid``; current stream waits on the
         in-flight copy if needed."""
         t0 = time.perf_counter()
+        if self.order:
+            self._pos = self._oidx.get(lid, self._pos)
         slot = self.layer_slot.get(lid)
         if slot is None:
             self.misses += 1

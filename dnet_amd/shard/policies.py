@@ -72,7 +72,7 @@ def enable_offload(model: RingModel, residency: int,
     model.layers = {}
     if model.device.type == "cuda":
         torch.cuda.empty_cache()
-    cache = WeightCache(store, residency, model.device)
+    cache = WeightCache(store, residency, model.device, order=order)
     model.weight_provider = OffloadBinder(cache, order, group, packed,
                                           prefetch_depth)
     # warm the first window

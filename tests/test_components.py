@@ -244,8 +244,9 @@ def test_weight_cache_prefetch_no_thrash():
     store = PinnedLayerStore(pin=False)
     for lid in range(24):
         store.put_layer(lid, {"w": torch.full((4, 4), float(lid))})
-    cache = WeightCache(store, residency=8, device=torch.device("cpu"))
     order = list(range(24))
+    cache = WeightCache(store, residency=8, device=torch.device("cpu"),
+                        order=order)
     for lid in order[:8]:
         cache.prefetch(lid)
     for epoch in range(2):
