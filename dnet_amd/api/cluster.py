@@ -12,6 +12,7 @@ from ..core.types import LayerAssignment, TopologyInfo
 from ..models import ModelConfig, PRESETS, QuantConfig
 from ..parallel.profiler import DeviceProfile
 from ..parallel.solver import (compute_layer_assignments, halda_solve,
+                               optimize_device_ordering,
                                postprocess_single_round)
 from ..utils.hostfile import DeviceProperties
 from ..utils.logger import get_logger
@@ -38,6 +39,7 @@ class ClusterManager:
         self.discovery = discovery
         self.devices: dict[str, DeviceProperties] = {}
         self.profiles: dict[str, DeviceProfile] = {}
+        self.link_ms: dict[tuple, float] = {}   # (src, dst) -> median ms
         self.topology: Optional[TopologyInfo] = None
 
     async def scan_devices(self) -> dict[str, DeviceProperties]:
@@ -68,23 +70,30 @@ class ClusterManager:
                 if isinstance(r, Exception):
                     continue
                 self.profiles[d.instance] = DeviceProfile.from_dict(r.json())
-            # latency sweep: ask the first shard to probe everyone
+            # latency sweep: every shard probes every other shard so the
+            # solver sees the full link matrix (ring ordering follows it)
             if len(healthy) > 1:
                 peers = [{"instance": d.instance, "host": d.local_ip,
                           "port": d.shard_port} for d in healthy]
-                try:
-                    r = await client.post(
-                        f"http://{healthy[0].local_ip}:{healthy[0].server_port}"
+                sweeps = await asyncio.gather(*[
+                    client.post(
+                        f"http://{d.local_ip}:{d.server_port}"
                         "/measure_latency",
-                        json={"peers": peers, "payload_sizes": [65536],
-                              "reps": 5})
+                        json={"peers": [p for p in peers
+                                        if p["instance"] != d.instance],
+                              "payload_sizes": [65536], "reps": 5})
+                    for d in healthy], return_exceptions=True)
+                for src, r in zip(healthy, sweeps):
+                    if isinstance(r, Exception):
+                        continue
                     lat = r.json().get("latencies", {})
                     for inst, sizes in lat.items():
                         med = next(iter(sizes.values()), {}).get("median_ms")
-                        if inst in self.profiles and med is not None:
+                        if med is None:
+                            continue
+                        self.link_ms[(src.instance, inst)] = med
+                        if inst in self.profiles:
                             self.profiles[inst].t_comm_ms = med
-                except httpx.HTTPError:
-                    pass
         return {k: v.to_dict() for k, v in self.profiles.items()}
 
     def get_head_node(self) -> Optional[DeviceProperties]:
@@ -101,6 +110,11 @@ class ClusterManager:
                        batch: int = 1, seq_len: int = 4096) -> TopologyInfo:
         shards = self.shard_devices()
         assert shards, "no shard devices discovered"
+        if self.link_ms:
+            by_name = {d.instance: d for d in shards}
+            order = optimize_device_ordering(
+                [d.instance for d in shards], self.link_ms)
+            shards = [by_name[i] for i in order]
         profs = []
         for d in shards:
             p = self.profiles.get(d.instance)

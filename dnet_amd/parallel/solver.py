@@ -113,6 +113,34 @@ def postprocess_single_round(w: list, devices: list[DeviceProfile]) -> list:
         w[i] = 0
 
 
+def optimize_device_ordering(instances: list, link_ms: dict) -> list:
+    """Greedy ring ordering: chain devices so consecutive hops ride the
+    fastest measured links (reference: src/dnet/api/utils.py
+    optimize_device_ordering — greedy Thunderbolt adjacency; here the edge
+    weights are median link latencies from the /measure_latency sweep, so
+    on an MI355X node the ordering follows the xGMI/TCP fabric instead of
+    Thunderbolt cables).
+
+    ``link_ms`` maps (src, dst) -> median ms; missing pairs are treated as
+    equally slow so an empty map preserves the input order. The head stays
+    first (layer-0 affinity)."""
+    if len(instances) <= 2 or not link_ms:
+        return list(instances)
+
+    def cost(a: str, b: str) -> float:
+        v = link_ms.get((a, b), link_ms.get((b, a)))
+        return float(v) if v is not None else 1e9
+
+    order = [instances[0]]
+    rest = list(instances[1:])
+    while rest:
+        cur = order[-1]
+        # stable: ties keep input order
+        nxt = min(range(len(rest)), key=lambda j: (cost(cur, rest[j]), j))
+        order.append(rest.pop(nxt))
+    return order
+
+
 def compute_layer_assignments(w: list, k: int, num_layers: int) -> list:
     """Round-robin k rounds of w[i] layers per device -> per-device list of
     per-round layer lists (reference: src/dnet/api/utils.py

@@ -256,3 +256,18 @@ def test_weight_cache_prefetch_no_thrash():
             for d in range(1, 4):
                 cache.prefetch(order[(i + d) % 24])
     assert cache.misses == 0, f"bind misses: {cache.misses}"
+
+
+def test_optimize_device_ordering():
+    from dnet_amd.parallel.solver import optimize_device_ordering
+    inst = ["a", "b", "c", "d"]
+    # no link info: order preserved
+    assert optimize_device_ordering(inst, {}) == inst
+    # a-c and c-b are the fast links; d is far from everyone
+    links = {("a", "c"): 0.1, ("c", "b"): 0.1, ("a", "b"): 5.0,
+             ("a", "d"): 9.0, ("b", "d"): 9.0, ("c", "d"): 9.0}
+    assert optimize_device_ordering(inst, links) == ["a", "c", "b", "d"]
+    # symmetric lookup: only (dst, src) present still found
+    assert optimize_device_ordering(["x", "y", "z"],
+                                    {("z", "x"): 0.1, ("y", "x"): 5.0,
+                                     ("y", "z"): 1.0}) == ["x", "z", "y"]
