@@ -17,7 +17,11 @@ from ..utils.logger import get_logger
 def main(argv=None):
     ap = argparse.ArgumentParser("dnet-api")
     s = get_settings()
-    ap.add_argument("--hostfile", required=True)
+    ap.add_argument("--hostfile", default=None,
+                    help="static discovery hostfile; omit for UDP discovery")
+    ap.add_argument("--discovery", choices=["hostfile", "udp"], default=None,
+                    help="default: hostfile if --hostfile given, else udp")
+    ap.add_argument("--discovery-port", type=int, default=52525)
     ap.add_argument("--host", default=s.api.host)
     ap.add_argument("--port", type=int, default=s.api.port)
     ap.add_argument("--wire-port", type=int, default=s.api.grpc_port)
@@ -39,8 +43,15 @@ async def serve(args):
 
     log = get_logger("api")
     s = get_settings()
-    discovery = StaticDiscovery(args.hostfile, own_instance="api",
-                                own_is_manager=True)
+    mode = args.discovery or ("hostfile" if args.hostfile else "udp")
+    if mode == "hostfile":
+        assert args.hostfile, "--hostfile required for hostfile discovery"
+        discovery = StaticDiscovery(args.hostfile, own_instance="api",
+                                    own_is_manager=True)
+    else:
+        from ..discovery import UdpDiscovery
+        discovery = UdpDiscovery("api", args.port, args.wire_port,
+                                 is_manager=True, port=args.discovery_port)
     await discovery.async_start()
     cluster = ClusterManager(discovery)
     state = ApiState(cluster, s)

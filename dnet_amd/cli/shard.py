@@ -25,6 +25,10 @@ def main(argv=None):
     ap.add_argument("--http-port", type=int, default=s.shard.http_port)
     ap.add_argument("--wire-port", type=int, default=s.shard.grpc_port)
     ap.add_argument("--tui", action="store_true")
+    ap.add_argument("--announce", action="store_true",
+                    help="announce this shard over UDP multicast discovery")
+    ap.add_argument("--discovery-port", type=int, default=52525)
+    ap.add_argument("--gpu-index", type=int, default=0)
     args = ap.parse_args(argv)
 
     from ..shard.runtime import ShardRuntime
@@ -34,6 +38,15 @@ def main(argv=None):
     faulthandler.register(signal.SIGUSR1)
     rt = ShardRuntime(instance=args.name)
     start_servers(rt, args.host, args.http_port, args.wire_port)
+    if args.announce:
+        from ..discovery import _load_ext
+        p2p = _load_ext().P2PInstance(
+            instance=args.name, http_port=args.http_port,
+            shard_port=args.wire_port, is_manager=False,
+            gpu_index=args.gpu_index, port=args.discovery_port)
+        p2p.start()
+        log.info("announcing %s on UDP discovery port %d", args.name,
+                 args.discovery_port)
     log.info("dnet-shard %s on http://%s:%d (wire %d)", args.name, args.host,
              args.http_port, args.wire_port)
     if args.tui:
