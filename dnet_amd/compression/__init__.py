@@ -20,6 +20,7 @@ import torch
 from .. import ops
 
 FMT_SPARSE_V1 = "sparse_v1"
+FMT_QSPARSE8_V1 = "qsparse8_v1"
 
 
 def keep_count(d: int, keep_ratio: float) -> int:
@@ -59,33 +60,71 @@ def column_unsparsify(packed: torch.Tensor, idx: torch.Tensor,
     return out
 
 
-def dtype_string(orig_dtype: str, keep_ratio: float) -> str:
+def dtype_string(orig_dtype: str, keep_ratio: float,
+                 fmt: str = FMT_SPARSE_V1) -> str:
     """Reference-style dtype metadata, e.g. 'bfloat16|90.0|fmt=sparse_v1'."""
-    return f"{orig_dtype}|{keep_ratio * 100:.1f}|fmt={FMT_SPARSE_V1}"
+    return f"{orig_dtype}|{keep_ratio * 100:.1f}|fmt={fmt}"
 
 
 def is_compressed_dtype(dtype: str) -> bool:
     return "|" in dtype
 
 
-def compress_tensor_to_bytes(x: torch.Tensor, keep_ratio: float) -> bytes:
-    """Self-describing wire blob: msgpack header + idx + packed values."""
+def _qgroup(k: int) -> int:
+    return 64 if k % 64 == 0 else k
+
+
+def quantize_packed_int8(packed: torch.Tensor
+                         ) -> tuple[torch.Tensor, torch.Tensor, int]:
+    """Kept columns -> int8 codes + fp16 per-(row, group) scales
+    (symmetric; the qsparse8 stage of the reference's qsparse8_v1)."""
+    k = packed.shape[-1]
+    g = _qgroup(k)
+    grp = packed.float().view(*packed.shape[:-1], k // g, g)
+    scales = grp.abs().amax(dim=-1) / 127.0
+    codes = torch.round(grp / scales.clamp_min(1e-12).unsqueeze(-1))
+    codes = codes.clamp(-127, 127).to(torch.int8).view(*packed.shape)
+    return codes, scales.to(torch.float16), g
+
+
+def dequantize_packed_int8(codes: torch.Tensor, scales: torch.Tensor,
+                           g: int) -> torch.Tensor:
+    k = codes.shape[-1]
+    grp = codes.float().view(*codes.shape[:-1], k // g, g)
+    out = grp * scales.float().unsqueeze(-1)
+    return out.view(*codes.shape).to(torch.bfloat16)
+
+
+def compress_tensor_to_bytes(x: torch.Tensor, keep_ratio: float,
+                             quantize: bool = False) -> bytes:
+    """Self-describing wire blob: msgpack header + idx + packed values.
+
+    ``quantize=True`` emits the int8-quantized variant (reference:
+    qsparse8_v1 — kept int8 codes + fp16 group scales), ~4x smaller than
+    sparse_v1 at the same keep ratio.
+    """
     idx, packed = column_sparsify(x, keep_ratio)
-    header = {"fmt": FMT_SPARSE_V1, "shape": list(x.shape),
+    fmt = FMT_QSPARSE8_V1 if quantize else FMT_SPARSE_V1
+    header = {"fmt": fmt, "shape": list(x.shape),
               "dtype": dtype_string(str(x.dtype).replace("torch.", ""),
-                                    keep_ratio),
+                                    keep_ratio, fmt),
               "k": int(idx.numel())}
+    if quantize:
+        codes, scales, g = quantize_packed_int8(packed)
+        header["g"] = g
+        body = (codes.contiguous().cpu().numpy().tobytes()
+                + scales.contiguous().cpu().view(torch.int16).numpy()
+                .tobytes())
+    else:
+        body = packed.contiguous().cpu().view(torch.int16).numpy().tobytes()
     h = msgpack.packb(header, use_bin_type=True)
-    return (struct.pack(">I", len(h)) + h
-            + idx.cpu().numpy().tobytes()
-            + packed.contiguous().cpu().view(torch.int16).numpy().tobytes())
+    return struct.pack(">I", len(h)) + h + idx.cpu().numpy().tobytes() + body
 
 
 def decompress_tensor_from_bytes(blob: bytes, device="cpu") -> torch.Tensor:
     import numpy as np
     (hlen,) = struct.unpack(">I", blob[:4])
     header = msgpack.unpackb(blob[4:4 + hlen], raw=False)
-    assert header["fmt"] == FMT_SPARSE_V1
     shape = header["shape"]
     k = header["k"]
     off = 4 + hlen
@@ -95,7 +134,20 @@ def decompress_tensor_from_bytes(blob: bytes, device="cpu") -> torch.Tensor:
     rows = 1
     for s in shape[:-1]:
         rows *= s
-    vals = torch.from_numpy(np.frombuffer(blob[off:off + rows * k * 2],
-                                          dtype=np.int16).copy())
-    packed = vals.view(*shape[:-1], k).view(torch.bfloat16).to(device)
-    return column_unsparsify(packed, idx.to(device), shape[-1])
+    if header["fmt"] == FMT_QSPARSE8_V1:
+        g = header["g"]
+        codes = torch.from_numpy(np.frombuffer(blob[off:off + rows * k],
+                                               dtype=np.int8).copy())
+        off += rows * k
+        ns = rows * (k // g)
+        scales = torch.from_numpy(np.frombuffer(blob[off:off + ns * 2],
+                                                dtype=np.int16).copy())
+        packed = dequantize_packed_int8(
+            codes.view(*shape[:-1], k),
+            scales.view(*shape[:-1], k // g).view(torch.float16), g)
+    else:
+        assert header["fmt"] == FMT_SPARSE_V1
+        vals = torch.from_numpy(np.frombuffer(blob[off:off + rows * k * 2],
+                                              dtype=np.int16).copy())
+        packed = vals.view(*shape[:-1], k).view(torch.bfloat16)
+    return column_unsparsify(packed.to(device), idx.to(device), shape[-1])

@@ -271,3 +271,26 @@ def test_optimize_device_ordering():
     assert optimize_device_ordering(["x", "y", "z"],
                                     {("z", "x"): 0.1, ("y", "x"): 5.0,
                                      ("y", "z"): 1.0}) == ["x", "z", "y"]
+
+
+def test_compression_qsparse8_roundtrip():
+    from dnet_amd import compression as cz
+    torch.manual_seed(1)
+    x = torch.randn(4, 128, dtype=torch.bfloat16)
+    x[:, :16] *= 10
+    blob8 = cz.compress_tensor_to_bytes(x, 0.5, quantize=True)
+    blob16 = cz.compress_tensor_to_bytes(x, 0.5)
+    assert len(blob8) < len(blob16) * 0.65   # int8 codes + g64 scales
+    y = cz.decompress_tensor_from_bytes(blob8)
+    # kept columns reconstruct within int8 quant error
+    idx, packed = cz.column_sparsify(x, 0.5)
+    kept = y.index_select(-1, idx.long()).float()
+    err = (kept - packed.float()).abs().max()
+    scale = packed.float().abs().max() / 127
+    assert err <= 2 * scale
+    # dropped columns are zero
+    mask = torch.ones(128, dtype=torch.bool)
+    mask[idx.long()] = False
+    assert y[:, mask].abs().sum() == 0
+    assert cz.is_compressed_dtype(
+        cz.dtype_string("bfloat16", 0.5, cz.FMT_QSPARSE8_V1))
