@@ -40,9 +40,23 @@ class ByteTokenizer:
         return ([self.BOS] if add_special_tokens else []) + ids
 
     def decode(self, ids) -> str:
-        return bytes(i for i in ids
-                     if 0 <= i < 256 and i not in (self.BOS, self.EOS)
-                     ).decode("utf-8", errors="replace")
+        # ids beyond the byte range (synthetic models sample the full model
+        # vocab) render as a placeholder so streams stay visible
+        out: list[str] = []
+        buf: list[int] = []
+        for i in ids:
+            if i in (self.BOS, self.EOS):
+                continue
+            if 0 <= i < 256:
+                buf.append(i)
+            else:
+                if buf:
+                    out.append(bytes(buf).decode("utf-8", errors="replace"))
+                    buf = []
+                out.append("·")
+        if buf:
+            out.append(bytes(buf).decode("utf-8", errors="replace"))
+        return "".join(out)
 
 
 class Detokenizer:

@@ -183,8 +183,8 @@ class ShardRuntime:
                              "bw=%.1fGB/s", size, r["latency_ms"], r["gbps"])
             except Exception:
                 log.exception("xGMI link probe failed (non-fatal)")
-        log.info("model loaded: %s rank %d/%d layers %s", self.model_name,
-                 req.rank, req.world_size, req.layers[:4])
+        log.info("model loaded: %s rank %d/%d %d layers %s...", self.model_name,
+                 req.rank, req.world_size, len(req.layers), req.layers[:4])
 
     def _load_weights(self, ex: RingExecutor, req: ShardLoadModelRequest) -> None:
         meta = get_model_metadata(req.model_path)
