@@ -116,27 +116,10 @@ class GptOssRingModel(MoERingModel):
             self.lm_head = Linear((head.to(torch.bfloat16)
                                    if head is not None else emb).to(self.device))
 
-    def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
-        c = self.cfg
-        logits = lw.router(y).float()
-        vals, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
-        weights = torch.softmax(vals, dim=-1)
-        out = torch.zeros_like(y, dtype=torch.float32)
-        dense = y.shape[0] <= self.DENSE_MOE_MAX_T
-        for e in range(c.num_experts):
-            if self.tp_size > 1 and e % self.tp_size != self.tp_rank:
-                continue
-            we_full = (weights * (idx == e)).sum(dim=-1)
-            if dense:
-                d = lw.experts_down[e](gpt_oss_glu(lw.experts_gateup[e](y)))
-                out += d.float() * we_full.unsqueeze(-1)
-                continue
-            mask = we_full > 0
-            if not bool(mask.any()):
-                continue
-            rows = mask.nonzero(as_tuple=True)[0]
-            xe = y[rows].contiguous()
-            a = gpt_oss_glu(lw.experts_gateup[e](xe))
-            d = lw.experts_down[e](a)
-            out[rows] += d.float() * we_full[rows].unsqueeze(-1)
-        return out.to(y.dtype)
+    # dense/sparse expert routing is inherited from MoERingModel._mlp;
+    # only the activation differs (clamped GLU, fused in the grouped kernel
+    # via GLU=1)
+    GLU = 1
+
+    def _act(self, gu: torch.Tensor) -> torch.Tensor:
+        return gpt_oss_glu(gu, self.GLU_ALPHA, self.GLU_LIMIT)

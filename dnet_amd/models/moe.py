@@ -39,10 +39,52 @@ class MoERingModel(RingModel):
             "MoE safetensors loading lands with the weight-cache milestone; "
             "use init_random for synthetic runs")
 
-    # below this many rows, run every expert on the whole batch and
-    # weight-sum (no data-dependent host syncs -> hipGraph-capturable; the
-    # expert weight read — the decode bound — is once per expert either way)
+    # below this many rows, route through the grouped-expert kernels (no
+    # data-dependent host syncs -> hipGraph-capturable; unrouted experts are
+    # skipped per-block on device, so a single stream reads only the top-k
+    # experts' weights)
     DENSE_MOE_MAX_T = 64
+    GLU = 0                  # 0 = SwiGLU, 1 = gpt-oss clamped GLU
+    GLU_ALPHA = 1.702
+    GLU_LIMIT = 7.0
+
+    def _act(self, gu: torch.Tensor) -> torch.Tensor:
+        return ops.swiglu(gu)
+
+    def _expert_stack(self, lw: LayerWeights) -> dict:
+        """Stack this rank's expert weights [E_local, ...] for the grouped
+        kernels; per-expert Linears become views of the stack (the sparse
+        large-T path keeps working on the same storage)."""
+        st = getattr(lw, "experts_stacked", None)
+        if st is not None:
+            return st
+        local = [e for e in range(self.cfg.num_experts)
+                 if self.tp_size <= 1 or e % self.tp_size == self.tp_rank]
+        gls = [lw.experts_gateup[e] for e in local]
+        dls = [lw.experts_down[e] for e in local]
+
+        def stack(attr, ls):
+            ts = [getattr(l, attr) for l in ls]
+            return (torch.stack(ts).contiguous()
+                    if ts[0] is not None else None)
+
+        gw, gs, gb = stack("w", gls), stack("scales", gls), stack("bias", gls)
+        dw, ds, db = stack("w", dls), stack("scales", dls), stack("bias", dls)
+        for j, e in enumerate(local):
+            lw.experts_gateup[e].w = gw[j]
+            lw.experts_down[e].w = dw[j]
+            for lin, s, b in ((lw.experts_gateup[e], gs, gb),
+                              (lw.experts_down[e], ds, db)):
+                if s is not None:
+                    lin.scales = s[j]
+                if b is not None:
+                    lin.bias = b[j]
+        st = {"local": torch.tensor(local, dtype=torch.long,
+                                    device=gw.device),
+              "gw": gw, "gs": gs, "gb": gb, "dw": dw, "ds": ds, "db": db,
+              "group": gls[0].group, "packed": gls[0].packed}
+        lw.experts_stacked = st
+        return st
 
     def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
         c = self.cfg
@@ -50,21 +92,30 @@ class MoERingModel(RingModel):
         logits = lw.router(y).float()
         weights, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
         weights = torch.softmax(weights, dim=-1)
+        if (T <= self.DENSE_MOE_MAX_T
+                and lw.experts_gateup[0].bits in (8, 16)):
+            st = self._expert_stack(lw)
+            we = torch.zeros(T, c.num_experts, dtype=torch.float32,
+                             device=y.device)
+            we.scatter_(1, idx, weights)
+            if self.tp_size > 1:
+                we = we.index_select(1, st["local"]).contiguous()
+            act = ops.moe_gateup(y, st["gw"], st["gs"], st["gb"], we,
+                                 st["group"], st["packed"], self.GLU,
+                                 self.GLU_ALPHA, self.GLU_LIMIT)
+            out = ops.moe_down(act, st["dw"], st["ds"], st["db"], we,
+                               st["group"], st["packed"])
+            return out.to(y.dtype)
         out = torch.zeros_like(y, dtype=torch.float32)
-        dense = T <= self.DENSE_MOE_MAX_T
         for e in range(c.num_experts):
             if self.tp_size > 1 and e % self.tp_size != self.tp_rank:
                 continue  # expert parallelism: partial sum reduced by caller
             we_full = (weights * (idx == e)).sum(dim=-1)
-            if dense:
-                d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](y)))
-                out += d.float() * we_full.unsqueeze(-1)
-                continue
             mask = we_full > 0
             if not bool(mask.any()):
                 continue
             rows = mask.nonzero(as_tuple=True)[0]
             xe = y[rows].contiguous()
-            d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](xe)))
+            d = lw.experts_down[e](self._act(lw.experts_gateup[e](xe)))
             out[rows] += d.float() * we_full[rows].unsqueeze(-1)
         return out.to(y.dtype)

@@ -180,6 +180,38 @@ def swiglu(gu: torch.Tensor) -> torch.Tensor:
     return ref.swiglu(gu)
 
 
+def moe_gateup(x: torch.Tensor, w: torch.Tensor, scales, bias,
+               we: torch.Tensor, group: int = 0, packed: bool = False,
+               glu: int = 0, alpha: float = 1.702,
+               limit: float = 7.0) -> torch.Tensor:
+    """Grouped expert gate+up with fused GLU: x [M,K], stacked w [E,2I,K]
+    (bf16 or grouped-int8), we [M,E] f32 routing weights -> act [E,M,I].
+    Experts with all-zero we are skipped on GPU (their act rows hold
+    garbage; moe_down never reads them)."""
+    if x.is_cuda:
+        e, i2, _ = w.shape
+        act = torch.empty(e, x.shape[0], i2 // 2, dtype=torch.bfloat16,
+                          device=x.device)
+        _native().moe_gateup(x, w, scales, bias, we, act, group, packed,
+                             glu, alpha, limit)
+        return act
+    return ref.moe_gateup(x, w, scales, bias, we, group, packed, glu,
+                          alpha, limit)
+
+
+def moe_down(act: torch.Tensor, w: torch.Tensor, scales, bias,
+             we: torch.Tensor, group: int = 0,
+             packed: bool = False) -> torch.Tensor:
+    """Grouped expert down projection, weighted f32 accumulation:
+    act [E,M,I], stacked w [E,H,I], we [M,E] -> out [M,H] f32."""
+    if act.is_cuda:
+        out = torch.zeros(act.shape[1], w.shape[1], dtype=torch.float32,
+                          device=act.device)
+        _native().moe_down(act, w, scales, bias, we, out, group, packed)
+        return out
+    return ref.moe_down(act, w, scales, bias, we, group, packed)
+
+
 quantize_int8 = ref.quantize_int8
 quantize_int4 = ref.quantize_int4
 pack_int8_mfma = ref.pack_int8_mfma

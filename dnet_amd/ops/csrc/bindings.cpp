@@ -30,6 +30,15 @@ void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
               c10::optional<torch::Tensor> scratch, int64_t group, bool packed,
               int64_t bits);
+void moe_gateup(torch::Tensor x, torch::Tensor w,
+                c10::optional<torch::Tensor> scales,
+                c10::optional<torch::Tensor> bias, torch::Tensor we,
+                torch::Tensor act, int64_t group, bool packed, int64_t glu,
+                double alpha, double limit);
+void moe_down(torch::Tensor act, torch::Tensor w,
+              c10::optional<torch::Tensor> scales,
+              c10::optional<torch::Tensor> bias, torch::Tensor we,
+              torch::Tensor out, int64_t group, bool packed);
 void col_norm2(torch::Tensor x, torch::Tensor norms);
 void gather_cols(torch::Tensor x, torch::Tensor idx, torch::Tensor out);
 void scatter_cols(torch::Tensor in, torch::Tensor idx, torch::Tensor out);
@@ -46,6 +55,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");
   m.def("dequant_int4", &dnet::dequant_int4, "packed-int4 -> bf16 dequant");
   m.def("gemm_m16", &dnet::gemm_m16, "MFMA decode GEMM (M<=16, bf16 or int8)");
+  m.def("moe_gateup", &dnet::moe_gateup,
+        "grouped MoE gate+up GEMV with fused GLU and expert skip");
+  m.def("moe_down", &dnet::moe_down,
+        "grouped MoE down GEMV with weighted f32 accumulation");
   m.def("col_norm2", &dnet::col_norm2, "per-column L2 norms");
   m.def("gather_cols", &dnet::gather_cols, "pack kept columns");
   m.def("scatter_cols", &dnet::scatter_cols, "zero + scatter kept columns");

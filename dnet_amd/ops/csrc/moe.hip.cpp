@@ -1,0 +1,347 @@
+// Grouped MoE expert kernels: the decode-side dense-expert path in two
+// launches per layer instead of O(E) tiny per-expert GEMVs.
+//
+// Motivation (measured): the per-expert python loop costs ~4000 small
+// kernels per token on gpt-oss-20b (24 layers x 32 experts x 3 ops) and
+// runs ~350 GB/s effective; these kernels put the whole expert bank behind
+// two launches with an early per-expert skip, so a single-stream decode
+// reads only the routed experts' weights (top-k/E of the bank).
+//
+//  - moe_gateup: act[e,m,i] = glu(x[m,:] . Wg[e,i,:], x[m,:] . Wu[e,i,:])
+//    with Wg/Wu the halves of the stacked gate_up weights [E, 2I, K]
+//    (bf16, or grouped-int8 with fused dequant, optionally in the MFMA
+//    chunk-pair packed order), glu = SwiGLU (mixtral) or the gpt-oss
+//    clamped GLU. A wave owns one (e, i) pair: two K-dots amortize the x
+//    read and keep the weight read a pure 16 B/lane stream.
+//  - moe_down: out[m,h] += we[m,e] * (act[e,m,:] . Wd[e,h,:] + b[e,h])
+//    accumulated with f32 atomics across experts.
+//
+// Both kernels skip expert e entirely when we[m,e] == 0 for all m in the
+// tile — data-dependent work without host syncs, so the pair stays
+// hipGraph-capturable (reference counterpart: the MLX MoE gather path in
+// src/dnet/core/models/gpt_oss.py).
+#include "common.h"
+
+namespace dnet {
+
+constexpr int kMoeWaves = 4;
+
+template <int M>
+__device__ __forceinline__ bool expert_routed(const float* __restrict__ we,
+                                              const int E, const int e,
+                                              const int m0) {
+  bool any = false;
+#pragma unroll
+  for (int m = 0; m < M; ++m) any |= (we[(int64_t)(m0 + m) * E + e] != 0.f);
+  return any;
+}
+
+// GLU = 0: silu(g) * u   GLU = 1: gpt-oss clamped (u+1) * g * sigmoid(g*a)
+template <int GLU>
+__device__ __forceinline__ float apply_glu(float g, float u, const float alpha,
+                                           const float limit) {
+  if (GLU == 1) {
+    g = fminf(g, limit);
+    u = fminf(fmaxf(u, -limit), limit);
+    return (u + 1.f) * g / (1.f + __expf(-g * alpha));
+  }
+  return g / (1.f + __expf(-g)) * u;
+}
+
+template <int M, bool Q8, bool PACKED, int GLU>
+__global__ void moe_gateup_kernel(
+    const short* __restrict__ x, const void* __restrict__ w,
+    const short* __restrict__ scales, const short* __restrict__ bias,
+    const float* __restrict__ we, short* __restrict__ act, const int K,
+    const int I, const int E, const int Mtot, const int m0, const int G,
+    const float alpha, const float limit) {
+  const int e = blockIdx.z;
+  if (!expert_routed<M>(we, E, e, m0)) return;
+  const int wid = threadIdx.x / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int i = blockIdx.x * kMoeWaves + wid;
+  if (i >= I) return;
+  const int64_t grow = (int64_t)e * 2 * I + i;      // gate row index
+  const int64_t urow = grow + I;                    // up row index
+  float ag[M], au[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) ag[m] = au[m] = 0.f;
+  if (Q8) {
+    const int4* wg = reinterpret_cast<const int4*>(
+        static_cast<const int8_t*>(w) + grow * K);
+    const int4* wu = reinterpret_cast<const int4*>(
+        static_cast<const int8_t*>(w) + urow * K);
+    const short* sg = scales + grow * (K / G);
+    const short* su = scales + urow * (K / G);
+    const int vecs = K / 16;
+    for (int v = lane; v < vecs; v += kWave) {
+      const int4 gv = wg[v], uv = wu[v];
+      const int8_t* gq = reinterpret_cast<const int8_t*>(&gv);
+      const int8_t* uq = reinterpret_cast<const int8_t*>(&uv);
+      const float gs = bits2f(sg[(v * 16) / G]);
+      const float us = bits2f(su[(v * 16) / G]);
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        const short8* xr =
+            reinterpret_cast<const short8*>(x + (int64_t)(m0 + m) * K);
+        short8 x0, x1;
+        if (PACKED) {
+          const int pr = v / 4, sl = v % 4;
+          x0 = xr[pr * 8 + sl];
+          x1 = xr[pr * 8 + sl + 4];
+        } else {
+          x0 = xr[2 * v];
+          x1 = xr[2 * v + 1];
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float xa = bits2f(x0.x[j]), xb = bits2f(x1.x[j]);
+          ag[m] = fmaf((float)gq[j] * gs, xa, ag[m]);
+          ag[m] = fmaf((float)gq[8 + j] * gs, xb, ag[m]);
+          au[m] = fmaf((float)uq[j] * us, xa, au[m]);
+          au[m] = fmaf((float)uq[8 + j] * us, xb, au[m]);
+        }
+      }
+    }
+  } else {
+    const short8* wg =
+        reinterpret_cast<const short8*>(static_cast<const short*>(w) + grow * K);
+    const short8* wu =
+        reinterpret_cast<const short8*>(static_cast<const short*>(w) + urow * K);
+    const int vecs = K / 8;
+    for (int v = lane; v < vecs; v += kWave) {
+      const short8 gv = wg[v], uv = wu[v];
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        const short8 xv = reinterpret_cast<const short8*>(
+            x + (int64_t)(m0 + m) * K)[v];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          ag[m] = fmaf(bits2f(gv.x[j]), bits2f(xv.x[j]), ag[m]);
+          au[m] = fmaf(bits2f(uv.x[j]), bits2f(xv.x[j]), au[m]);
+        }
+      }
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    float g = wave_reduce_sum(ag[m]);
+    float u = wave_reduce_sum(au[m]);
+    if (lane == 0) {
+      if (bias != nullptr) {
+        g += bits2f(bias[grow]);
+        u += bits2f(bias[urow]);
+      }
+      act[((int64_t)e * Mtot + m0 + m) * I + i] =
+          f2bits(apply_glu<GLU>(g, u, alpha, limit));
+    }
+  }
+}
+
+template <int M, bool Q8, bool PACKED>
+__global__ void moe_down_kernel(
+    const short* __restrict__ act, const void* __restrict__ w,
+    const short* __restrict__ scales, const short* __restrict__ bias,
+    const float* __restrict__ we, float* __restrict__ out, const int I,
+    const int H, const int E, const int Mtot, const int m0, const int G) {
+  const int e = blockIdx.z;
+  if (!expert_routed<M>(we, E, e, m0)) return;
+  const int wid = threadIdx.x / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int h = blockIdx.x * kMoeWaves + wid;
+  if (h >= H) return;
+  const int64_t row = (int64_t)e * H + h;
+  float acc[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) acc[m] = 0.f;
+  if (Q8) {
+    const int4* wr = reinterpret_cast<const int4*>(
+        static_cast<const int8_t*>(w) + row * I);
+    const short* sr = scales + row * (I / G);
+    const int vecs = I / 16;
+    for (int v = lane; v < vecs; v += kWave) {
+      const int4 wv = wr[v];
+      const int8_t* q = reinterpret_cast<const int8_t*>(&wv);
+      const float s = bits2f(sr[(v * 16) / G]);
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        const short8* xr = reinterpret_cast<const short8*>(
+            act + ((int64_t)e * Mtot + m0 + m) * I);
+        short8 x0, x1;
+        if (PACKED) {
+          const int pr = v / 4, sl = v % 4;
+          x0 = xr[pr * 8 + sl];
+          x1 = xr[pr * 8 + sl + 4];
+        } else {
+          x0 = xr[2 * v];
+          x1 = xr[2 * v + 1];
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          acc[m] = fmaf((float)q[j] * s, bits2f(x0.x[j]), acc[m]);
+          acc[m] = fmaf((float)q[8 + j] * s, bits2f(x1.x[j]), acc[m]);
+        }
+      }
+    }
+  } else {
+    const short8* wr =
+        reinterpret_cast<const short8*>(static_cast<const short*>(w) + row * I);
+    const int vecs = I / 8;
+    for (int v = lane; v < vecs; v += kWave) {
+      const short8 wv = wr[v];
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        const short8 xv = reinterpret_cast<const short8*>(
+            act + ((int64_t)e * Mtot + m0 + m) * I)[v];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[m] = fmaf(bits2f(wv.x[j]), bits2f(xv.x[j]), acc[m]);
+      }
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M; ++m) {
+    float r = wave_reduce_sum(acc[m]);
+    if (lane == 0) {
+      const float wme = we[(int64_t)(m0 + m) * E + e];
+      if (wme != 0.f) {
+        if (bias != nullptr) r += bits2f(bias[row]);
+        atomicAdd(&out[(int64_t)(m0 + m) * H + h], r * wme);
+      }
+    }
+  }
+}
+
+template <typename LaunchFn>
+static void moe_dispatch_m(int M, LaunchFn&& fn) {
+  switch (M) {
+    case 1: fn(std::integral_constant<int, 1>{}); break;
+    case 2: fn(std::integral_constant<int, 2>{}); break;
+    case 3: fn(std::integral_constant<int, 3>{}); break;
+    case 4: fn(std::integral_constant<int, 4>{}); break;
+    case 6: fn(std::integral_constant<int, 6>{}); break;
+    case 8: fn(std::integral_constant<int, 8>{}); break;
+    default: TORCH_CHECK(false, "moe: unsupported M tile ", M);
+  }
+}
+
+static int moe_mtile(int64_t rem) {
+  if (rem >= 8) return 8;
+  if (rem == 7 || rem == 5) return 4;
+  return (int)rem;
+}
+
+void moe_gateup(torch::Tensor x, torch::Tensor w,
+                c10::optional<torch::Tensor> scales,
+                c10::optional<torch::Tensor> bias, torch::Tensor we,
+                torch::Tensor act, int64_t group, bool packed, int64_t glu,
+                double alpha, double limit) {
+  const int64_t M = x.size(0), K = x.size(1);
+  const int64_t E = w.size(0), I2 = w.size(1), I = I2 / 2;
+  const bool q8 = w.scalar_type() == torch::kInt8;
+  DNET_CHECK(w.size(2) == K && act.size(0) == E && act.size(1) == M &&
+                 act.size(2) == I, "moe_gateup shape");
+  DNET_CHECK(we.size(0) == M && we.size(1) == E, "we shape");
+  DNET_CHECK(x.is_contiguous() && w.is_contiguous() && we.is_contiguous() &&
+                 act.is_contiguous(), "contig");
+  DNET_CHECK(K % (q8 ? 16 : 8) == 0, "K align");
+  if (packed) DNET_CHECK(K % 64 == 0 && group % 64 == 0, "packed align");
+  const short* sptr = nullptr;
+  if (q8) {
+    DNET_CHECK(scales.has_value() && group > 0 && K % group == 0, "scales");
+    DNET_CHECK(scales->is_contiguous() && scales->size(0) == E &&
+                   scales->size(1) == I2 && scales->size(2) == K / group,
+               "scales shape");
+    sptr = (const short*)scales->data_ptr();
+  }
+  const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
+  auto stream = current_stream();
+  const dim3 grid(cdiv((int)I, kMoeWaves), 1, (unsigned)E);
+  int64_t m0 = 0;
+  while (m0 < M) {
+    const int mt = moe_mtile(M - m0);
+    moe_dispatch_m(mt, [&](auto mc) {
+      constexpr int MV = decltype(mc)::value;
+      auto launch = [&](auto q8c, auto pkc, auto gluc) {
+        hipLaunchKernelGGL(
+            (moe_gateup_kernel<MV, decltype(q8c)::value, decltype(pkc)::value,
+                               decltype(gluc)::value>),
+            grid, dim3(kMoeWaves * kWave), 0, stream,
+            (const short*)x.data_ptr(), w.data_ptr(), sptr, bptr,
+            (const float*)we.data_ptr(), (short*)act.data_ptr(), (int)K,
+            (int)I, (int)E, (int)M, (int)m0, (int)std::max<int64_t>(group, 16),
+            (float)alpha, (float)limit);
+      };
+      auto with_glu = [&](auto q8c, auto pkc) {
+        if (glu == 1)
+          launch(q8c, pkc, std::integral_constant<int, 1>{});
+        else
+          launch(q8c, pkc, std::integral_constant<int, 0>{});
+      };
+      if (q8) {
+        if (packed)
+          with_glu(std::true_type{}, std::true_type{});
+        else
+          with_glu(std::true_type{}, std::false_type{});
+      } else {
+        with_glu(std::false_type{}, std::false_type{});
+      }
+    });
+    m0 += mt;
+  }
+}
+
+void moe_down(torch::Tensor act, torch::Tensor w,
+              c10::optional<torch::Tensor> scales,
+              c10::optional<torch::Tensor> bias, torch::Tensor we,
+              torch::Tensor out, int64_t group, bool packed) {
+  const int64_t E = w.size(0), H = w.size(1), I = w.size(2);
+  const int64_t M = act.size(1);
+  const bool q8 = w.scalar_type() == torch::kInt8;
+  DNET_CHECK(act.size(0) == E && act.size(2) == I, "moe_down act shape");
+  DNET_CHECK(out.size(0) == M && out.size(1) == H &&
+                 out.scalar_type() == torch::kFloat, "moe_down out f32");
+  DNET_CHECK(we.size(0) == M && we.size(1) == E, "we shape");
+  DNET_CHECK(act.is_contiguous() && w.is_contiguous() && we.is_contiguous() &&
+                 out.is_contiguous(), "contig");
+  DNET_CHECK(I % (q8 ? 16 : 8) == 0, "I align");
+  if (packed) DNET_CHECK(I % 64 == 0 && group % 64 == 0, "packed align");
+  const short* sptr = nullptr;
+  if (q8) {
+    DNET_CHECK(scales.has_value() && group > 0 && I % group == 0, "scales");
+    DNET_CHECK(scales->is_contiguous() && scales->size(0) == E &&
+                   scales->size(1) == H && scales->size(2) == I / group,
+               "scales shape");
+    sptr = (const short*)scales->data_ptr();
+  }
+  const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
+  auto stream = current_stream();
+  const dim3 grid(cdiv((int)H, kMoeWaves), 1, (unsigned)E);
+  int64_t m0 = 0;
+  while (m0 < M) {
+    const int mt = moe_mtile(M - m0);
+    moe_dispatch_m(mt, [&](auto mc) {
+      constexpr int MV = decltype(mc)::value;
+      auto launch = [&](auto q8c, auto pkc) {
+        hipLaunchKernelGGL(
+            (moe_down_kernel<MV, decltype(q8c)::value, decltype(pkc)::value>),
+            grid, dim3(kMoeWaves * kWave), 0, stream,
+            (const short*)act.data_ptr(), w.data_ptr(), sptr, bptr,
+            (const float*)we.data_ptr(), (float*)out.data_ptr(), (int)I,
+            (int)H, (int)E, (int)M, (int)m0,
+            (int)std::max<int64_t>(group, 16));
+      };
+      if (q8) {
+        if (packed)
+          launch(std::true_type{}, std::true_type{});
+        else
+          launch(std::true_type{}, std::false_type{});
+      } else {
+        launch(std::false_type{}, std::false_type{});
+      }
+    });
+    m0 += mt;
+  }
+}
+
+}  // namespace dnet

@@ -288,3 +288,50 @@ def swiglu(gu: torch.Tensor) -> torch.Tensor:
     g = gu[..., :i].float()
     u = gu[..., i:].float()
     return (torch.nn.functional.silu(g) * u).to(gu.dtype)
+
+
+def _moe_dense_w(w: torch.Tensor, scales, group: int, packed: bool
+                 ) -> torch.Tensor:
+    """Stacked expert weights [E, N, K] (bf16 or grouped-int8) -> f32."""
+    if w.dtype != torch.int8:
+        return w.float()
+    e, n, k = w.shape
+    q = w.reshape(e * n, k)
+    if packed:
+        q = unpack_int8_mfma(q)
+    return dequant_int8(q, scales.reshape(e * n, -1), group).float().view(
+        e, n, k)
+
+
+def moe_gateup(x: torch.Tensor, w: torch.Tensor, scales, bias,
+               we: torch.Tensor, group: int = 0, packed: bool = False,
+               glu: int = 0, alpha: float = 1.702, limit: float = 7.0
+               ) -> torch.Tensor:
+    """x [M,K], w [E,2I,K], we [M,E] -> act [E,M,I] (glu 0=SwiGLU,
+    1=gpt-oss clamped GLU). Experts with all-zero we still computed here
+    (the kernel skips them; their act rows are never read by moe_down)."""
+    wf = _moe_dense_w(w, scales, group, packed)
+    gu = torch.einsum("mk,enk->emn", x.float(), wf)
+    if bias is not None:
+        gu = gu + bias.float().unsqueeze(1)
+    i = gu.shape[-1] // 2
+    g, u = gu[..., :i], gu[..., i:]
+    if glu == 1:
+        g = g.clamp(max=limit)
+        u = u.clamp(min=-limit, max=limit)
+        act = (u + 1.0) * g * torch.sigmoid(g * alpha)
+    else:
+        act = torch.nn.functional.silu(g) * u
+    return act.to(torch.bfloat16)
+
+
+def moe_down(act: torch.Tensor, w: torch.Tensor, scales, bias,
+             we: torch.Tensor, group: int = 0, packed: bool = False
+             ) -> torch.Tensor:
+    """act [E,M,I], w [E,H,I], we [M,E] -> out [M,H] f32 (weighted sum of
+    per-expert down projections; bias applied inside the weighting)."""
+    wf = _moe_dense_w(w, scales, group, packed)
+    d = torch.einsum("emi,ehi->emh", act.float(), wf)
+    if bias is not None:
+        d = d + bias.float().unsqueeze(1)
+    return (we.t().unsqueeze(-1) * d).sum(dim=0)

@@ -362,3 +362,70 @@ def test_model_int4_consistency_gpu():
     hd = m.embed_tokens(toks[:, -1]).clone()
     m.decode_window(hd, m.layer_ids, kv)
     assert torch.isfinite(m.normalize_project(hd).float()).all()
+
+
+@pytest.mark.parametrize("q8,packed,glu", [
+    (False, False, 0), (False, False, 1), (True, False, 0), (True, True, 1)])
+def test_moe_grouped_kernels(q8, packed, glu):
+    """moe_gateup + moe_down vs fp32 reference (routed experts only — the
+    kernel skips unrouted experts, leaving their act rows unwritten)."""
+    torch.manual_seed(7)
+    E, I, K, H, M, G = 8, 256, 512, 512, 3, 64
+    dev = _dev()
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    gw = torch.randn(E, 2 * I, K, dtype=torch.bfloat16, device=dev) / 8
+    dw = torch.randn(E, H, I, dtype=torch.bfloat16, device=dev) / 8
+    gb = torch.randn(E, 2 * I, dtype=torch.bfloat16, device=dev) / 4
+    db = torch.randn(E, H, dtype=torch.bfloat16, device=dev) / 4
+    # route 2 experts per row
+    we = torch.zeros(M, E, dtype=torch.float32, device=dev)
+    idx = torch.stack([torch.randperm(E, device=dev)[:2] for _ in range(M)])
+    we.scatter_(1, idx, torch.rand(M, 2, device=dev) + 0.1)
+    gs = ds = None
+    if q8:
+        gq, gs = ops.quantize_int8(gw.reshape(E * 2 * I, K), G)
+        dq, ds = ops.quantize_int8(dw.reshape(E * H, I), G)
+        if packed:
+            gq = ops.pack_int8_mfma(gq)
+            dq = ops.pack_int8_mfma(dq)
+        gw = gq.view(E, 2 * I, K).contiguous()
+        dw = dq.view(E, H, I).contiguous()
+        gs = gs.view(E, 2 * I, K // G).contiguous()
+        ds = ds.view(E, H, I // G).contiguous()
+
+    def cpu(t):
+        return None if t is None else t.cpu()
+
+    act = ops.moe_gateup(x, gw, gs, gb, we, G, packed, glu)
+    act_ref = ref.moe_gateup(cpu(x), cpu(gw), cpu(gs), cpu(gb), cpu(we),
+                             G, packed, glu)
+    routed = we.cpu().t().bool()   # [E, M]
+    assert torch.allclose(act.cpu().float()[routed],
+                          act_ref.float()[routed], atol=5e-2, rtol=5e-2)
+    # feed the reference act into BOTH downs so gateup error doesn't compound
+    act_in = act_ref.to(dev).contiguous()
+    out = ops.moe_down(act_in, dw, ds, db, we, G, packed)
+    out_ref = ref.moe_down(act_ref, cpu(dw), cpu(ds), cpu(db), cpu(we),
+                           G, packed)
+    assert torch.allclose(out.cpu(), out_ref, atol=5e-2, rtol=5e-2)
+
+
+def test_moe_dense_matches_sparse_gpu():
+    """Grouped dense path == per-expert sparse loop on the same weights."""
+    from dnet_amd.models import ModelConfig, PRESETS
+    from dnet_amd.models.gpt_oss import GptOssRingModel
+    hf = dict(PRESETS["gpt-oss-20b"])
+    hf["num_hidden_layers"] = 2
+    cfg = ModelConfig.from_hf(hf)
+    m = GptOssRingModel(cfg, [0, 1], device=_dev())
+    m.init_random(seed=5)
+    torch.manual_seed(3)
+    y = torch.randn(4, cfg.hidden_size, dtype=torch.bfloat16, device=_dev())
+    lw = m.layers[0]
+    dense = m._mlp(y, lw)
+    try:
+        GptOssRingModel.DENSE_MOE_MAX_T = 0
+        sparse = m._mlp(y, lw)
+    finally:
+        GptOssRingModel.DENSE_MOE_MAX_T = 64
+    assert torch.allclose(dense.float(), sparse.float(), atol=4e-2, rtol=4e-2)
