@@ -417,7 +417,7 @@ def test_moe_dense_matches_sparse_gpu():
     hf = dict(PRESETS["gpt-oss-20b"])
     hf["num_hidden_layers"] = 2
     cfg = ModelConfig.from_hf(hf)
-    m = GptOssRingModel(cfg, [0, 1], False, False, device=_dev())
+    m = GptOssRingModel(cfg, [0, 1], _dev(), False, False)
     m.init_random(seed=5)
     torch.manual_seed(3)
     y = torch.randn(4, cfg.hidden_size, dtype=torch.bfloat16, device=_dev())
