@@ -273,22 +273,31 @@ class DeepseekV2RingModel(RingModel):
         weights, idx = torch.topk(scores, c.num_experts_per_tok, dim=-1)
         if c.norm_topk_prob:
             weights = weights / weights.sum(-1, keepdim=True)
-        out = torch.zeros_like(y, dtype=torch.float32)
-        dense = y.shape[0] <= 64  # graph-safe dense path for decode batches
-        for e in range(c.num_experts):
-            we_full = (weights * (idx == e)).sum(dim=-1)
-            if dense:
-                d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](y)))
-                out += d.float() * we_full.unsqueeze(-1)
-                continue
-            mask = we_full > 0
-            if not bool(mask.any()):
-                continue
-            rows = mask.nonzero(as_tuple=True)[0]
-            xe = y[rows].contiguous()
-            d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](xe)))
-            out[rows] += d.float() * we_full[rows].unsqueeze(-1)
-        out *= c.routed_scaling_factor
+        T = y.shape[0]
+        if T <= 64 and lw.experts_gateup[0].bits in (8, 16):
+            # grouped-expert kernels (graph-safe; unrouted experts skipped
+            # on device) — see models/moe.py
+            from .moe import stack_experts
+            st = stack_experts(lw, list(range(c.num_experts)))
+            we = torch.zeros(T, c.num_experts, dtype=torch.float32,
+                             device=y.device)
+            we.scatter_(1, idx, weights)
+            act = ops.moe_gateup(y, st["gw"], st["gs"], st["gb"], we,
+                                 st["group"], st["packed"], 0)
+            out = ops.moe_down(act, st["dw"], st["ds"], st["db"], we,
+                               st["group"], st["packed"])
+        else:
+            out = torch.zeros_like(y, dtype=torch.float32)
+            for e in range(c.num_experts):
+                we_full = (weights * (idx == e)).sum(dim=-1)
+                mask = we_full > 0
+                if not bool(mask.any()):
+                    continue
+                rows = mask.nonzero(as_tuple=True)[0]
+                xe = y[rows].contiguous()
+                d = lw.experts_down[e](ops.swiglu(lw.experts_gateup[e](xe)))
+                out[rows] += d.float() * we_full[rows].unsqueeze(-1)
+        out = out * c.routed_scaling_factor
         if getattr(lw, "shared_gateup", None) is not None:
             out += lw.shared_down(ops.swiglu(lw.shared_gateup(y))).float()
         return out.to(y.dtype)

@@ -14,6 +14,39 @@ from .base import LayerWeights, Linear, RingModel
 from .config import ModelConfig
 
 
+def stack_experts(lw: LayerWeights, local: list) -> dict:
+    """Stack the given experts' weights [E_local, ...] for the grouped MoE
+    kernels; the per-expert Linears become views of the stack (the sparse
+    large-T path keeps working on the same storage, nothing is duplicated
+    after the originals are freed)."""
+    st = getattr(lw, "experts_stacked", None)
+    if st is not None:
+        return st
+    gls = [lw.experts_gateup[e] for e in local]
+    dls = [lw.experts_down[e] for e in local]
+
+    def stack(attr, ls):
+        ts = [getattr(l, attr) for l in ls]
+        return torch.stack(ts).contiguous() if ts[0] is not None else None
+
+    gw, gs, gb = stack("w", gls), stack("scales", gls), stack("bias", gls)
+    dw, ds, db = stack("w", dls), stack("scales", dls), stack("bias", dls)
+    for j, e in enumerate(local):
+        lw.experts_gateup[e].w = gw[j]
+        lw.experts_down[e].w = dw[j]
+        for lin, s, b in ((lw.experts_gateup[e], gs, gb),
+                          (lw.experts_down[e], ds, db)):
+            if s is not None:
+                lin.scales = s[j]
+            if b is not None:
+                lin.bias = b[j]
+    st = {"local": torch.tensor(local, dtype=torch.long, device=gw.device),
+          "gw": gw, "gs": gs, "gb": gb, "dw": dw, "ds": ds, "db": db,
+          "group": gls[0].group, "packed": gls[0].packed}
+    lw.experts_stacked = st
+    return st
+
+
 class MoERingModel(RingModel):
     model_type = "mixtral"
     model_types = ["mixtral", "qwen2_moe", "qwen3_moe"]
@@ -52,39 +85,9 @@ class MoERingModel(RingModel):
         return ops.swiglu(gu)
 
     def _expert_stack(self, lw: LayerWeights) -> dict:
-        """Stack this rank's expert weights [E_local, ...] for the grouped
-        kernels; per-expert Linears become views of the stack (the sparse
-        large-T path keeps working on the same storage)."""
-        st = getattr(lw, "experts_stacked", None)
-        if st is not None:
-            return st
         local = [e for e in range(self.cfg.num_experts)
                  if self.tp_size <= 1 or e % self.tp_size == self.tp_rank]
-        gls = [lw.experts_gateup[e] for e in local]
-        dls = [lw.experts_down[e] for e in local]
-
-        def stack(attr, ls):
-            ts = [getattr(l, attr) for l in ls]
-            return (torch.stack(ts).contiguous()
-                    if ts[0] is not None else None)
-
-        gw, gs, gb = stack("w", gls), stack("scales", gls), stack("bias", gls)
-        dw, ds, db = stack("w", dls), stack("scales", dls), stack("bias", dls)
-        for j, e in enumerate(local):
-            lw.experts_gateup[e].w = gw[j]
-            lw.experts_down[e].w = dw[j]
-            for lin, s, b in ((lw.experts_gateup[e], gs, gb),
-                              (lw.experts_down[e], ds, db)):
-                if s is not None:
-                    lin.scales = s[j]
-                if b is not None:
-                    lin.bias = b[j]
-        st = {"local": torch.tensor(local, dtype=torch.long,
-                                    device=gw.device),
-              "gw": gw, "gs": gs, "gb": gb, "dw": dw, "ds": ds, "db": db,
-              "group": gls[0].group, "packed": gls[0].packed}
-        lw.experts_stacked = st
-        return st
+        return stack_experts(lw, local)
 
     def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
         c = self.cfg

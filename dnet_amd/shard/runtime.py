@@ -360,7 +360,11 @@ class ShardRuntime:
                 self._emit_token(nonce, int(tokt[0]),
                                  finished=last or int(tokt[0]) in stop_ids)
 
-        ex.decode_stream(max_tokens, stop_ids=stop_ids, on_token=on_token)
+        t0 = time.perf_counter()
+        n = ex.decode_stream(max_tokens, stop_ids=stop_ids, on_token=on_token)
+        dt = time.perf_counter() - t0
+        log.info("[PROFILE][DECODE] nonce=%s tokens=%d ms=%.1f tok_s=%.1f",
+                 nonce[:18], n, dt * 1e3, n / max(dt, 1e-9))
 
     def _emit_token(self, nonce: str, token_id: int, finished: bool = False):
         ex = self.executor

@@ -8,7 +8,7 @@ mkdir -p gpurun_out
 cat > gpurun_out/hosts1 <<HOSTS
 shard0 127.0.0.1 18081 15052 0
 HOSTS
-python -m dnet_amd.cli.shard --name shard0 --host 127.0.0.1 --http-port 18081 --wire-port 15052 > gpurun_out/soak_shard.log 2>&1 &
+DNET_OBS_PROFILE=true python -m dnet_amd.cli.shard --name shard0 --host 127.0.0.1 --http-port 18081 --wire-port 15052 > gpurun_out/soak_shard.log 2>&1 &
 SHARD_PID=$!
 python -m dnet_amd.cli.api --hostfile gpurun_out/hosts1 --host 127.0.0.1 --port 18080 --wire-port 15051 --callback-addr 127.0.0.1:15051 > gpurun_out/soak_api.log 2>&1 &
 API_PID=$!
@@ -47,4 +47,4 @@ curl -s -m 300 -X POST http://127.0.0.1:18080/v1/load_model \
 chat gpt-oss-20b-synthetic 32 "hello moe" \
   | python3 -c 'import json,sys; d=json.load(sys.stdin); m=d.get("metrics",{}); print("tokens:", d["usage"]["completion_tokens"], "ttfb_ms:", round(m.get("ttfb_ms",0),1), "tps_decoding:", round(m.get("tps_decoding",0),1))'
 echo "=== soak done ==="
-tail -n 2 gpurun_out/soak_shard.log
+grep "\[PROFILE\]\[DECODE\]" gpurun_out/soak_shard.log | tail -5
