@@ -6,8 +6,11 @@ This probe times both, plus the stream loop with sampling and with a
 host-callback, for a given model — run on a GPU box:
     python scripts/serve_probe.py gpt-oss-20b 32
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
