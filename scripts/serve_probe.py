@@ -33,14 +33,20 @@ def timed(label, ex, toks, f, n):
 def main():
     model = sys.argv[1] if len(sys.argv) > 1 else "gpt-oss-20b"
     n = int(sys.argv[2]) if len(sys.argv) > 2 else 32
+    graphs = "--no-graphs" not in sys.argv
     hf = dict(PRESETS[model])
     quant = QuantConfig(8, 128) if model.startswith("qwen") else None
     cfg = ModelConfig.from_hf(hf, quant=quant)
     ex = RingExecutor(cfg, 0, 1, "cuda:0", mb_count=1, mb_size=1, smax=1024,
-                      seed=0, use_graphs=True)
+                      seed=0, use_graphs=graphs)
+    print("init done", flush=True)
     toks = torch.randint(0, cfg.vocab_size, (1, 1, 64), device="cuda:0")
     ex.prefill(toks)
+    torch.cuda.synchronize()
+    print("prefill done", flush=True)
     ex.decode_rounds(4)  # warm graphs
+    torch.cuda.synchronize()
+    print("warm done", flush=True)
 
     timed("decode_rounds (async)", ex, toks, lambda: ex.decode_rounds(n), n)
     timed("stream greedy no-cb", ex, toks,
