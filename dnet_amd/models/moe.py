@@ -7,6 +7,8 @@ later milestone.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from .. import ops
@@ -76,7 +78,7 @@ class MoERingModel(RingModel):
     # data-dependent host syncs -> hipGraph-capturable; unrouted experts are
     # skipped per-block on device, so a single stream reads only the top-k
     # experts' weights)
-    DENSE_MOE_MAX_T = 64
+    DENSE_MOE_MAX_T = int(os.environ.get("DNET_DENSE_MOE_T", "64"))
     GLU = 0                  # 0 = SwiGLU, 1 = gpt-oss clamped GLU
     GLU_ALPHA = 1.702
     GLU_LIMIT = 7.0

@@ -134,6 +134,10 @@ def dequant_int8(w: torch.Tensor, scales: torch.Tensor, group: int,
 def _attn_splits(b: int, hkv: int, smax: int) -> int:
     # fill ~1024 blocks (4/CU); cap so each split owns >= ~512 positions of
     # capacity (short actual sequences leave idle splits + combine overhead)
+    import os
+    force = os.environ.get("DNET_ATTN_SPLITS")
+    if force:
+        return max(1, int(force))
     splits = 1
     cap = max(1, smax // 512)
     while (b * hkv * splits * 2 <= 1024 and splits * 2 <= cap):
