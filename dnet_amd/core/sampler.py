@@ -51,6 +51,15 @@ class Sampler:
                 x = x.masked_fill(p < c.min_p * p.amax(-1, keepdim=True),
                                   float("-inf"))
             probs = torch.softmax(x, dim=-1)
+            # degenerate rows (NaN logits, or everything masked) would trip
+            # multinomial's device-side assert and abort the whole process
+            # (HSA exception) — fall back to greedy for those rows, without
+            # a host sync (stays usable inside the serving loop)
+            probs = torch.nan_to_num(probs, nan=0.0, posinf=0.0)
+            ok = probs.sum(dim=-1, keepdim=True) > 0
+            fallback = torch.nn.functional.one_hot(
+                lf.nan_to_num(nan=0.0).argmax(dim=-1), lf.shape[-1]).float()
+            probs = torch.where(ok, probs, fallback)
             tok = torch.multinomial(probs, 1, generator=self.generator).squeeze(-1)
         if not c.logprobs:
             return tok, None, None

@@ -27,7 +27,8 @@ def timed(label, ex, toks, f, n):
     f()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
-    print(f"{label:28s} {dt * 1e3 / n:7.2f} ms/tok  {n / dt:8.1f} tok/s")
+    print(f"{label:28s} {dt * 1e3 / n:7.2f} ms/tok  {n / dt:8.1f} tok/s",
+          flush=True)
 
 
 def main():
@@ -47,6 +48,18 @@ def main():
     ex.decode_rounds(4)  # warm graphs
     torch.cuda.synchronize()
     print("warm done", flush=True)
+
+    lb = ex.logits_buf[0].float()
+    print("logits nan:", bool(torch.isnan(lb).any()),
+          "inf:", bool(torch.isinf(lb).any()),
+          "absmax:", float(lb.nan_to_num().abs().max()), flush=True)
+    if "--sampled-only" in sys.argv:
+        def cb0(step, tok, last):
+            int(tok[0])
+        ex.set_decoding(DecodingConfig(temperature=0.7, top_p=0.9))
+        timed("stream sampled cb", ex, toks,
+              lambda: ex.decode_stream(n + 1, stop_ids=[0], on_token=cb0), n)
+        return
 
     timed("decode_rounds (async)", ex, toks, lambda: ex.decode_rounds(n), n)
     timed("stream greedy no-cb", ex, toks,
