@@ -58,23 +58,23 @@ def main():
             int(tok[0])
         ex.set_decoding(DecodingConfig(temperature=0.7, top_p=0.9))
         timed("stream sampled cb", ex, toks,
-              lambda: ex.decode_stream(n + 1, stop_ids=[0], on_token=cb0), n)
+              lambda: ex.decode_stream(n + 1, stop_ids=[-1], on_token=cb0), n)
         return
 
     timed("decode_rounds (async)", ex, toks, lambda: ex.decode_rounds(n), n)
     timed("stream greedy no-cb", ex, toks,
           lambda: ex.decode_stream(n + 1), n)
     timed("stream greedy stop-check", ex, toks,
-          lambda: ex.decode_stream(n + 1, stop_ids=[0]), n)
+          lambda: ex.decode_stream(n + 1, stop_ids=[-1]), n)
 
     def cb(step, tok, last):
         int(tok[0])
 
     timed("stream greedy cb", ex, toks,
-          lambda: ex.decode_stream(n + 1, stop_ids=[0], on_token=cb), n)
+          lambda: ex.decode_stream(n + 1, stop_ids=[-1], on_token=cb), n)
     ex.set_decoding(DecodingConfig(temperature=0.7, top_p=0.9))
     timed("stream sampled cb", ex, toks,
-          lambda: ex.decode_stream(n + 1, stop_ids=[0], on_token=cb), n)
+          lambda: ex.decode_stream(n + 1, stop_ids=[-1], on_token=cb), n)
 
 
 if __name__ == "__main__":
