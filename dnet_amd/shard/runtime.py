@@ -360,16 +360,21 @@ class ShardRuntime:
                 self._emit_token(nonce, int(tokt[0]),
                                  finished=last or int(tokt[0]) in stop_ids)
 
+        self._emit_s = 0.0
         t0 = time.perf_counter()
         n = ex.decode_stream(max_tokens, stop_ids=stop_ids, on_token=on_token)
         dt = time.perf_counter() - t0
-        log.info("[PROFILE][DECODE] nonce=%s tokens=%d ms=%.1f tok_s=%.1f",
-                 nonce[:18], n, dt * 1e3, n / max(dt, 1e-9))
+        log.info("[PROFILE][DECODE] nonce=%s tokens=%d ms=%.1f tok_s=%.1f "
+                 "emit_ms=%.1f", nonce[:18], n, dt * 1e3, n / max(dt, 1e-9),
+                 self._emit_s * 1e3)
+
+    _emit_s = 0.0
 
     def _emit_token(self, nonce: str, token_id: int, finished: bool = False):
         ex = self.executor
         if self._callback is None:
             return
+        t0 = time.perf_counter()
         frame = {"t": "token", "nonce": nonce, "token_id": token_id,
                  "ts_ms": int(time.time() * 1e3), "finished": finished}
         if ex.last_logprob is not None:
@@ -381,6 +386,8 @@ class ShardRuntime:
         except OSError:
             log.warning("token callback failed (api down?)")
             self._callback.close()
+        finally:
+            self._emit_s += time.perf_counter() - t0
 
     def _send_error(self, nonce: str):
         if self._callback is not None:
