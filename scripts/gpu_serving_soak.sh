@@ -44,7 +44,10 @@ echo "=== hot-swap to gpt-oss-20b (MoE/sinks/sliding) ==="
 curl -s -m 300 -X POST http://127.0.0.1:18080/v1/load_model \
   -H 'content-type: application/json' \
   -d '{"model":"gpt-oss-20b-synthetic","max_seq":1024}' | head -c 200; echo
-chat gpt-oss-20b-synthetic 32 "hello moe" \
-  | python3 -c 'import json,sys; d=json.load(sys.stdin); m=d.get("metrics",{}); print("tokens:", d["usage"]["completion_tokens"], "ttfb_ms:", round(m.get("ttfb_ms",0),1), "tps_decoding:", round(m.get("tps_decoding",0),1))'
+for i in 1 2; do
+  echo "--- gpt-oss request $i ---"
+  chat gpt-oss-20b-synthetic 32 "hello moe $i" \
+    | python3 -c 'import json,sys; d=json.load(sys.stdin); m=d.get("metrics",{}); print("tokens:", d["usage"]["completion_tokens"], "ttfb_ms:", round(m.get("ttfb_ms",0),1), "tps_decoding:", round(m.get("tps_decoding",0),1))'
+done
 echo "=== soak done ==="
 grep "\[PROFILE\]\[DECODE\]" gpurun_out/soak_shard.log | tail -5
