@@ -96,7 +96,12 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
 
   for (int k0 = p_begin * 64; k0 < p_end * 64; k0 += XT) {
     const int tk = min(XT, p_end * 64 - k0);
-    __syncthreads();
+    // raw barrier (lgkmcnt only): __syncthreads() would wait vmcnt(0) and
+    // drain the in-flight weight stream at EVERY x-tile boundary — at MT=4
+    // (XT=256) that is a full memory-pipeline restart every 4 pairs
+    // (guide §6: counted waits, not vmcnt(0), across barriers)
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    asm volatile("s_barrier" ::: "memory");
     for (int idx = threadIdx.x; idx < 16 * MT * (tk / 8); idx += 256) {
       const int r = idx / (tk / 8);
       const int vec = idx % (tk / 8);
@@ -105,60 +110,68 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
           *reinterpret_cast<const short8*>(
               &x[(int64_t)min(r, M - 1) * K + k0 + vec * 8]);
     }
-    __syncthreads();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    asm volatile("s_barrier" ::: "memory");
     const short* arow = &x_lds[row * SE];
     int pl = 0;
     const int pl_end = tk / 64;
-    for (; pl + 2 <= pl_end; pl += 2) {   // 2 pairs = 4 chunks unrolled
-      bf16x8 b[2][2];
-      bf16x8 a[2][2][MT];
+    for (; pl + 4 <= pl_end; pl += 4) {  // 4 pairs, all weight loads first
+      int4 wq8[4];
+      int4 wq4[2];
+      bf16x8 wb[4][2];
+      float sv[4];
+      if (QBITS == 8) {
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
-        const int p = (k0 / 64) + pl + u;
-        if (QBITS == 8) {
-          const int4 wq = *reinterpret_cast<const int4*>(&wrow_q[p * 64 + woff]);
-          const int8_t* q8 = reinterpret_cast<const int8_t*>(&wq);
-          const float sv = bits2f(srow[(p * 64) / G]);
-          b[u][0] = deq8(q8, sv);
-          b[u][1] = deq8(q8 + 8, sv);
-        } else if (QBITS == 4) {
-          // chunk-quad layout: one b128 covers this lane's slices of four
-          // chunks; load once per quad (even u), use halves per pair.
-          if (u == 0) {
-            const int quad = p / 2;
-            const int4 wq = *reinterpret_cast<const int4*>(
-                &wrow_q4[quad * 64 + woff]);
-            const uint8_t* q4 = reinterpret_cast<const uint8_t*>(&wq);
-            const float sv = bits2f(srow[(quad * 128) / G]);
-            b[0][0] = deq4(q4, sv);
-            b[0][1] = deq4(q4 + 4, sv);
-            b[1][0] = deq4(q4 + 8, sv);
-            b[1][1] = deq4(q4 + 12, sv);
-          }
-        } else {
-          b[u][0] = *reinterpret_cast<const bf16x8*>(
-              &wrow_b[p * 64 + ks]);
-          b[u][1] = *reinterpret_cast<const bf16x8*>(
+        for (int u = 0; u < 4; ++u) {
+          const int p = (k0 / 64) + pl + u;
+          wq8[u] = *reinterpret_cast<const int4*>(&wrow_q[p * 64 + woff]);
+          sv[u] = bits2f(srow[(p * 64) / G]);
+        }
+      } else if (QBITS == 4) {
+#pragma unroll
+        for (int v = 0; v < 2; ++v) {
+          const int quad = ((k0 / 64) + pl) / 2 + v;
+          wq4[v] = *reinterpret_cast<const int4*>(&wrow_q4[quad * 64 + woff]);
+          sv[v] = bits2f(srow[(quad * 128) / G]);
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int p = (k0 / 64) + pl + u;
+          wb[u][0] = *reinterpret_cast<const bf16x8*>(&wrow_b[p * 64 + ks]);
+          wb[u][1] = *reinterpret_cast<const bf16x8*>(
               &wrow_b[p * 64 + 32 + ks]);
+        }
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        bf16x8 b0, b1;
+        if (QBITS == 8) {
+          const int8_t* q8 = reinterpret_cast<const int8_t*>(&wq8[u]);
+          b0 = deq8(q8, sv[u]);
+          b1 = deq8(q8 + 8, sv[u]);
+        } else if (QBITS == 4) {
+          const uint8_t* q4 =
+              reinterpret_cast<const uint8_t*>(&wq4[u / 2]) + (u & 1) * 8;
+          b0 = deq4(q4, sv[u / 2]);
+          b1 = deq4(q4 + 4, sv[u / 2]);
+        } else {
+          b0 = wb[u][0];
+          b1 = wb[u][1];
         }
 #pragma unroll
         for (int t = 0; t < MT; ++t) {
           const short* at = arow + t * 16 * SE;
-          a[u][0][t] = *reinterpret_cast<const bf16x8*>(
-              &at[(pl + u) * 64 + ks]);
-          a[u][1][t] = *reinterpret_cast<const bf16x8*>(
-              &at[(pl + u) * 64 + 32 + ks]);
+          const bf16x8 a0 =
+              *reinterpret_cast<const bf16x8*>(&at[(pl + u) * 64 + ks]);
+          const bf16x8 a1 =
+              *reinterpret_cast<const bf16x8*>(&at[(pl + u) * 64 + 32 + ks]);
+          acc[t][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a0, b0, acc[t][0], 0, 0, 0);
+          acc[t][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a1, b1, acc[t][1], 0, 0, 0);
         }
       }
-#pragma unroll
-      for (int u = 0; u < 2; ++u)
-#pragma unroll
-        for (int t = 0; t < MT; ++t) {
-          acc[t][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[u][0][t], b[u][0], acc[t][0], 0, 0, 0);
-          acc[t][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[u][1][t], b[u][1], acc[t][1], 0, 0, 0);
-        }
     }
     for (; pl < pl_end; ++pl) {
       const int p = (k0 / 64) + pl;

@@ -49,10 +49,10 @@ def main():
             scratch = ops._get_scratch(dev)
 
             t_pk = bench(lambda: nat.gemm_m16(x, qp, scales, None, out,
-                                              scratch, 128, True))
+                                              scratch, 128, True, 8))
             t_old = bench(lambda: nat.gemv_int8(x, q, scales, out, 128, None))
             t_mfma_bf = bench(lambda: nat.gemm_m16(x, wf, None, None, out,
-                                                   scratch, 0, False))
+                                                   scratch, 0, False, 16))
             print(f"{name:8s} N={N:6d} K={K:6d} M={M:2d}  "
                   f"i8-packed {t_pk*1e6:7.1f}us {wbytes_i8/t_pk/1e9:6.0f}GB/s | "
                   f"i8-scalar {wbytes_i8/t_old/1e9:6.0f} | "
