@@ -50,7 +50,7 @@ def main():
 
             t_pk = bench(lambda: nat.gemm_m16(x, qp, scales, None, out,
                                               scratch, 128, True, 8))
-            t_old = bench(lambda: nat.gemv_int8(x, q, scales, out, 128, None))
+            t_old = bench(lambda: nat.gemv_int8(x, q, scales, out, 128, None, False))
             t_mfma_bf = bench(lambda: nat.gemm_m16(x, wf, None, None, out,
                                                    scratch, 0, False, 16))
             print(f"{name:8s} N={N:6d} K={K:6d} M={M:2d}  "
