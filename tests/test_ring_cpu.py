@@ -238,3 +238,32 @@ def _rank_main_rounds(rank, world, port, q):
     if rank == 0:
         q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
     dist.destroy_process_group()
+
+
+def test_bench_contract_multiproc_cpu(tmp_path):
+    """The driver's SCALE invocation end-to-end: torch.distributed.run with
+    2 ranks over gloo runs bench.py and rank 0 prints ONE valid JSON line
+    with the contract fields."""
+    import json
+    import subprocess
+    import sys
+
+    port = _free_port()
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--model", "tiny",
+         "--quant", "bf16", "--mb-size", "2", "--prompt-len", "8",
+         "--smax", "32"],
+        capture_output=True, text=True, timeout=300, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    j = json.loads(lines[0])
+    assert j["n_gpus"] == 2 and j["steps"] == 2 and j["warmup"] == 1
+    assert j["value"] > 0 and j["ms_per_step"] > 0
+    assert j["scaling"] == "weak" and "config" in j
+    assert j["config"]["global_batch"] == 2 * 2 * 2  # mb_count(2*2) * mb_size
