@@ -72,6 +72,19 @@ class GptOssRingModel(MoERingModel):
                     return sd[pref + name].to(torch.bfloat16)
             return None
 
+        def get_mx(name):
+            # MXFP4 checkpoints ship <name>_blocks (uint8 nibble pairs) +
+            # <name>_scales (uint8 E8M0); dequantized to bf16 at load like
+            # the reference's MXFP4 sanitization. Layout is output-major
+            # [E, rows, K] (the bf16 layout is input-major [E, K, rows]).
+            for pref in ("model.", ""):
+                b = sd.get(pref + name + "_blocks")
+                s = sd.get(pref + name + "_scales")
+                if b is not None and s is not None:
+                    from .. import ops
+                    return ops.mxfp4_dequant(b, s)
+            return None
+
         inter = c.moe_intermediate_size or c.intermediate_size
         for lid in self.layer_ids:
             p = f"layers.{lid}."
@@ -92,9 +105,15 @@ class GptOssRingModel(MoERingModel):
             gub = get(p + "mlp.experts.gate_up_proj_bias")  # [E, 2I]
             dn = get(p + "mlp.experts.down_proj")           # [E, I, H]
             dnb = get(p + "mlp.experts.down_proj_bias")     # [E, H]
+            mx = gu is None
+            if mx:
+                gu = get_mx(p + "mlp.experts.gate_up_proj")  # [E, 2I, H]
+                dn = get_mx(p + "mlp.experts.down_proj")     # [E, H, I]
             lw.experts_gateup, lw.experts_down = [], []
             for e in range(c.num_experts):
-                w_e = gu[e].t().contiguous()                # [2I, H] interleaved
+                # bf16 layout is input-major (transpose); MXFP4 blocks are
+                # already output-major
+                w_e = (gu[e] if mx else gu[e].t()).contiguous()  # interleaved
                 # de-interleave rows: gate = even rows, up = odd rows
                 w_e = torch.cat([w_e[0::2], w_e[1::2]])     # [2I, H] concat
                 b_e = torch.cat([gub[e][0::2], gub[e][1::2]]) if gub is not None else None
@@ -102,7 +121,7 @@ class GptOssRingModel(MoERingModel):
                     w_e.to(self.device),
                     b_e.to(self.device) if b_e is not None else None, c.quant))
                 lw.experts_down.append(Linear.make(
-                    dn[e].t().contiguous().to(self.device),
+                    (dn[e] if mx else dn[e].t()).contiguous().to(self.device),
                     dnb[e].to(self.device) if dnb is not None else None,
                     c.quant))
             lw.sinks = get(p + "self_attn.sinks").to(self.device)

@@ -223,3 +223,4 @@ pack_int4_mfma = ref.pack_int4_mfma
 unpack_int8_mfma = ref.unpack_int8_mfma
 rope_tables = ref.rope_tables
 rope_apply = ref.rope_apply
+mxfp4_dequant = ref.mxfp4_dequant

@@ -335,3 +335,27 @@ def moe_down(act: torch.Tensor, w: torch.Tensor, scales, bias,
     if bias is not None:
         d = d + bias.float().unsqueeze(1)
     return (we.t().unsqueeze(-1) * d).sum(dim=0)
+
+
+_MXFP4_LUT = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+              -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0]
+
+
+def mxfp4_dequant(blocks: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    """OCP MXFP4 -> bf16 (gpt-oss checkpoint format; reference counterpart:
+    src/dnet/core/models/gpt_oss.py MXFP4 weight sanitization).
+
+    blocks: uint8 [..., B, 16] — two E2M1 nibbles per byte, low nibble
+    first, 32 values per block. scales: uint8 [..., B] — E8M0 power-of-two
+    exponents, bias 127. Returns bf16 [..., B*32].
+    """
+    lut = torch.tensor(_MXFP4_LUT, dtype=torch.float32, device=blocks.device)
+    lo = lut[(blocks & 0xF).long()]
+    hi = lut[(blocks >> 4).long()]
+    vals = torch.stack([lo, hi], dim=-1).reshape(*blocks.shape[:-1],
+                                                 blocks.shape[-1] * 2)
+    ex = torch.pow(2.0, scales.float() - 127.0)
+    out = vals * ex.unsqueeze(-1)
+    return out.reshape(*blocks.shape[:-2],
+                       blocks.shape[-2] * blocks.shape[-1] * 2).to(
+                           torch.bfloat16)

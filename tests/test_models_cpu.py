@@ -190,3 +190,91 @@ def test_kv_cache_quantized_close_to_fp():
     a, b = run(16), run(8)
     cos = torch.nn.functional.cosine_similarity(a, b, dim=-1)
     assert (cos > 0.99).all(), f"kv8 vs kv16: {cos}"
+
+
+def test_mxfp4_dequant_and_gpt_oss_load():
+    """MXFP4 (E2M1 nibbles + E8M0 block scales) dequant, and the gpt-oss
+    loader accepting *_blocks/*_scales checkpoints (reference: MXFP4 weight
+    sanitization in src/dnet/core/models/gpt_oss.py)."""
+    import dnet_amd.ops as ops
+    from dnet_amd.ops.reference import _MXFP4_LUT
+
+    g = torch.Generator().manual_seed(0)
+
+    def encode(target_idx, scales_exp):
+        # target_idx: [.., B, 32] LUT indices; scales_exp: [.., B] exponents
+        lo, hi = target_idx[..., 0::2], target_idx[..., 1::2]
+        blocks = (lo | (hi << 4)).to(torch.uint8)
+        scales = (scales_exp + 127).to(torch.uint8)
+        lut = torch.tensor(_MXFP4_LUT)
+        expect = (lut[target_idx] *
+                  torch.pow(2.0, scales_exp.float()).unsqueeze(-1))
+        return blocks, scales, expect.reshape(*target_idx.shape[:-2], -1)
+
+    idx = torch.randint(0, 16, (3, 4, 2, 32), generator=g)
+    exp = torch.randint(-3, 4, (3, 4, 2), generator=g)
+    blocks, scales, expect = encode(idx, exp)
+    got = ops.mxfp4_dequant(blocks, scales)
+    assert got.dtype == torch.bfloat16
+    assert torch.equal(got.float(), expect)
+
+    # loader: bf16 checkpoint vs the same weights as MXFP4 blocks
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.models.gpt_oss import GptOssRingModel
+    hf = dict(model_type="gpt_oss", hidden_size=64, num_hidden_layers=1,
+              num_attention_heads=2, num_key_value_heads=1, head_dim=32,
+              intermediate_size=64, vocab_size=64, num_local_experts=2,
+              num_experts_per_tok=1, sliding_window=16, rope_theta=10000.0,
+              attention_bias=True, rms_norm_eps=1e-5)
+    cfg = ModelConfig.from_hf(hf)
+    E, H, I2 = 2, 64, 128
+
+    def rnd(*s):
+        return torch.randn(*s, generator=g, dtype=torch.float32).to(
+            torch.bfloat16)
+
+    gi = torch.randint(0, 16, (E, I2, H // 32, 32), generator=g)
+    ge = torch.randint(-2, 3, (E, I2, H // 32), generator=g)
+    gub, gus, gu_vals = encode(gi, ge)     # gu_vals [E, 2I, H] output-major
+    di = torch.randint(0, 16, (E, H, (I2 // 2) // 32, 32), generator=g)
+    de = torch.randint(-2, 3, (E, H, (I2 // 2) // 32), generator=g)
+    dnb_, dns, dn_vals = encode(di, de)    # [E, H, I]
+    base = {
+        "model.layers.0.input_layernorm.weight": rnd(64),
+        "model.layers.0.post_attention_layernorm.weight": rnd(64),
+        "model.layers.0.self_attn.q_proj.weight": rnd(64, 64),
+        "model.layers.0.self_attn.k_proj.weight": rnd(32, 64),
+        "model.layers.0.self_attn.v_proj.weight": rnd(32, 64),
+        "model.layers.0.self_attn.q_proj.bias": rnd(64),
+        "model.layers.0.self_attn.k_proj.bias": rnd(32),
+        "model.layers.0.self_attn.v_proj.bias": rnd(32),
+        "model.layers.0.self_attn.o_proj.weight": rnd(64, 64),
+        "model.layers.0.self_attn.o_proj.bias": rnd(64),
+        "model.layers.0.self_attn.sinks": rnd(2),
+        "model.layers.0.mlp.router.weight": rnd(E, 64),
+        "model.layers.0.mlp.router.bias": rnd(E),
+        "model.layers.0.mlp.experts.gate_up_proj_bias": rnd(E, I2),
+        "model.layers.0.mlp.experts.down_proj_bias": rnd(E, H),
+        "model.embed_tokens.weight": rnd(64, 64),
+        "model.norm.weight": rnd(64),
+    }
+    sd_bf = dict(base)
+    sd_bf["model.layers.0.mlp.experts.gate_up_proj"] = \
+        gu_vals.to(torch.bfloat16).transpose(1, 2).contiguous()  # [E,H,2I]
+    sd_bf["model.layers.0.mlp.experts.down_proj"] = \
+        dn_vals.to(torch.bfloat16).transpose(1, 2).contiguous()  # [E,I,H]
+    sd_mx = dict(base)
+    sd_mx["model.layers.0.mlp.experts.gate_up_proj_blocks"] = gub
+    sd_mx["model.layers.0.mlp.experts.gate_up_proj_scales"] = gus
+    sd_mx["model.layers.0.mlp.experts.down_proj_blocks"] = dnb_
+    sd_mx["model.layers.0.mlp.experts.down_proj_scales"] = dns
+
+    m1 = GptOssRingModel(cfg, [0], "cpu", True, True)
+    m1.load_state_dict(sd_bf)
+    m2 = GptOssRingModel(cfg, [0], "cpu", True, True)
+    m2.load_state_dict(sd_mx)
+    for e in range(E):
+        assert torch.equal(m1.layers[0].experts_gateup[e].w,
+                           m2.layers[0].experts_gateup[e].w)
+        assert torch.equal(m1.layers[0].experts_down[e].w,
+                           m2.layers[0].experts_down[e].w)
