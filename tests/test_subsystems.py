@@ -198,3 +198,62 @@ def test_shard_runtime_loads_real_checkpoint(tmp_path):
     assert gen.shape == (1, 1, 3)
     rt._unload()
     assert rt.status == "idle"
+
+
+def test_inference_timeout_and_error_frames():
+    """Failure detection: a dead mid-ring shard surfaces as a timeout after
+    token_timeout_s (not a hang), and an error frame from a shard aborts
+    the request immediately (reference: 300 s await_token timeout,
+    inference.py:166; RingError equivalent actually sent here)."""
+    import time as _time
+
+    from dnet_amd.api.inference import InferenceManager
+    from dnet_amd.api.tokenizer import ByteTokenizer
+
+    class MM:
+        tokenizer = ByteTokenizer(512)
+        stop_ids = [ByteTokenizer(512).EOS]
+
+    class DeadHead:
+        async def request(self, frame):
+            return {"t": "ack"}   # acks, then never produces tokens
+
+    im = InferenceManager(MM(), token_timeout_s=0.3)
+    im.head_client = DeadHead()
+    im.callback_addr = "127.0.0.1:1"
+
+    async def run_dead():
+        req = ChatRequestModel(model="tiny-random",
+                               messages=[{"role": "user", "content": "x"}])
+        return await im.chat_completions(req)
+
+    t0 = _time.perf_counter()
+    with pytest.raises((asyncio.TimeoutError, RuntimeError, Exception)):
+        asyncio.run(run_dead())
+    assert _time.perf_counter() - t0 < 5.0   # bounded, no hang
+    assert not im.pending                    # nonce cleaned up
+
+    class ErrorHead:
+        def __init__(self, im):
+            self.im = im
+
+        async def request(self, frame):
+            self.im.resolve_token({"t": "error", "nonce": frame["nonce"],
+                                   "failed_node": "shard1",
+                                   "error": "boom"})
+            return {"t": "ack"}
+
+    im2 = InferenceManager(MM(), token_timeout_s=10)
+    im2.head_client = ErrorHead(im2)
+    im2.callback_addr = "127.0.0.1:1"
+
+    async def run_err():
+        req = ChatRequestModel(model="tiny-random",
+                               messages=[{"role": "user", "content": "x"}])
+        return await im2.chat_completions(req)
+
+    t0 = _time.perf_counter()
+    with pytest.raises(RuntimeError, match="boom"):
+        asyncio.run(run_err())
+    assert _time.perf_counter() - t0 < 2.0
+    assert not im2.pending
