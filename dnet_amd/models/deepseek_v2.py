@@ -3,10 +3,9 @@
 Reference counterpart: src/dnet/core/models/deepseek_v2.py (mlx_lm wrapper;
 qk_nope + qk_rope split head dims). The MLA projections (q, kv_a with
 shared rope key, kv_a layernorm, kv_b) run through the fused GEMV/GEMM
-linears; attention itself uses the torch path because the MLA head dim
-(qk_nope+qk_rope, e.g. 192) is outside the 64/128 decode-kernel shapes —
-a dedicated MLA decode kernel is roadmap. KV is cached per head
-(uncompressed k/v; latent-cache compression is roadmap). MoE layers use
+linears; decode attention runs on the native flash-decode kernel with the
+MLA head-dim pair (qk 192 / v 128). KV is cached per head (uncompressed
+k/v; latent-cache compression is roadmap). MoE layers use
 routed top-k (softmax scoring, greedy) * routed_scaling_factor + shared
 experts; the first ``first_k_dense_replace`` layers are dense.
 """
@@ -220,25 +219,14 @@ class DeepseekV2RingModel(RingModel):
             kv.v[li].scatter_(2, idx.expand(B, c.num_q_heads, 1, v.shape[-1]),
                               v.unsqueeze(2))
             len_t = kv.pos + 1
-            attn = self._attn_decode_torch(q, kv.k[li], kv.v[li], len_t)
+            # MLA decode attention on the native kernel (D=192 qk / 128 v)
+            attn = ops.attn_decode(q.contiguous(), kv.k[li], kv.v[li],
+                                   len_t, self.scale)
             o = lw.o(attn.reshape(B, -1))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
             delta = self._mlp_for(lid, y2, lw)
         h.add_(delta)
         return h
-
-    def _attn_decode_torch(self, q, kcache, vcache, len_t):
-        """q [B, Hq, kd]; per-head full cache; masked softmax attention."""
-        B, Hq, kd = q.shape
-        S = kcache.shape[2]
-        scores = torch.einsum("bhd,bhsd->bhs", q.float(), kcache.float())
-        scores *= self.scale
-        mask = (torch.arange(S, device=q.device).view(1, 1, S)
-                >= len_t.view(B, 1, 1))
-        scores.masked_fill_(mask, float("-inf"))
-        p = torch.softmax(scores, dim=-1)
-        out = torch.einsum("bhs,bhsd->bhd", p, vcache.float())
-        return out.to(q.dtype)
 
     def prefill_window(self, h, layer_ids, kv, p0: int):
         c = self.cfg

@@ -152,12 +152,13 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                 vscale: torch.Tensor | None = None) -> torch.Tensor:
     if q.is_cuda:
         # q may be a strided slice of the fused QKV buffer; out is contiguous.
-        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         B, Hq, D = q.shape
+        dv = vcache.shape[-1]           # MLA: v head dim < qk head dim
+        out = torch.empty(B, Hq, dv, dtype=q.dtype, device=q.device)
         splits = _attn_splits(B, kcache.shape[1], kcache.shape[2])
         partials = None
         if splits > 1:
-            partials = torch.empty(B * Hq * splits * (D + 2),
+            partials = torch.empty(B * Hq * splits * (dv + 2),
                                    dtype=torch.float32, device=q.device)
         _native().attn_decode(q, kcache, vcache, pos, out, scale, window,
                               sinks, kscale, vscale, partials, splits)

@@ -41,7 +41,7 @@ constexpr int kChunk = 64;
 // (acc, m, l) go to scratch [B, Hq, splits, D+2] f32 and a combine kernel
 // merges them (sink logit folded there). Raises the block count from
 // B*Hkv to B*Hkv*splits so small batches still fill 256 CUs.
-template <int D, bool Q8>
+template <int D, int DV, bool Q8>
 __global__ void attn_decode_kernel(const short* __restrict__ q,
                                    const void* __restrict__ kc,
                                    const void* __restrict__ vc,
@@ -56,7 +56,8 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
                                    float* __restrict__ partials,
                                    const int splits) {
   constexpr int P = lds_pitch(D);
-  constexpr int DPL = D / kWave;  // output dims per lane (1 or 2)
+  constexpr int PV = lds_pitch(DV);
+  constexpr int DPL = DV / kWave;  // output dims per lane (1 or 2)
   const int b = blockIdx.x / Hkv;
   const int hkv = blockIdx.x % Hkv;
   const int G = Hq / Hkv;  // query heads per kv head
@@ -65,12 +66,13 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);               // [kChunk][P]
-  short* v_lds = k_lds + kChunk * P;                           // [kChunk][P]
-  short* q_lds = v_lds + kChunk * P;                           // [G][D]
+  short* v_lds = k_lds + kChunk * P;                           // [kChunk][PV]
+  short* q_lds = v_lds + kChunk * PV;                          // [G][D]
 
   const int len = pos[b];
   const int start = (window > 0) ? max(0, len - window) : 0;  // sliding
   const int64_t kvbase = ((int64_t)b * Hkv + hkv) * Smax * D;
+  const int64_t vbase = ((int64_t)b * Hkv + hkv) * Smax * DV;
 
   // Stage this group's q rows.
   for (int i = threadIdx.x; i < G * D / 8; i += blockDim.x) {
@@ -101,37 +103,52 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
     for (int i = threadIdx.x; i < kChunk * D / 8; i += blockDim.x) {
       const int row = i / (D / 8);
       const int col = (i % (D / 8)) * 8;
-      short8 kv8, vv8;
+      short8 kv8;
       if (row < valid) {
         if (Q8) {
           constexpr int NG = D / 64;
           const int64_t rb = kvbase / D * (int64_t)NG
                              + (int64_t)(s0 + row) * NG + col / 64;
           const float ks = bits2f(kscale[rb]);
-          const float vs = bits2f(vscale[rb]);
           const int2 kq = *reinterpret_cast<const int2*>(
               (const int8_t*)kc + kvbase + (int64_t)(s0 + row) * D + col);
-          const int2 vq = *reinterpret_cast<const int2*>(
-              (const int8_t*)vc + kvbase + (int64_t)(s0 + row) * D + col);
           const int8_t* kb = reinterpret_cast<const int8_t*>(&kq);
-          const int8_t* vb = reinterpret_cast<const int8_t*>(&vq);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            kv8.x[j] = f2bits((float)kb[j] * ks);
-            vv8.x[j] = f2bits((float)vb[j] * vs);
-          }
+          for (int j = 0; j < 8; ++j) kv8.x[j] = f2bits((float)kb[j] * ks);
         } else {
           kv8 = *reinterpret_cast<const short8*>(
               (const short*)kc + kvbase + (int64_t)(s0 + row) * D + col);
-          vv8 = *reinterpret_cast<const short8*>(
-              (const short*)vc + kvbase + (int64_t)(s0 + row) * D + col);
         }
       } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) { kv8.x[j] = 0; vv8.x[j] = 0; }
+        for (int j = 0; j < 8; ++j) kv8.x[j] = 0;
       }
       *reinterpret_cast<short8*>(&k_lds[row * P + col]) = kv8;
-      *reinterpret_cast<short8*>(&v_lds[row * P + col]) = vv8;
+    }
+    for (int i = threadIdx.x; i < kChunk * DV / 8; i += blockDim.x) {
+      const int row = i / (DV / 8);
+      const int col = (i % (DV / 8)) * 8;
+      short8 vv8;
+      if (row < valid) {
+        if (Q8) {
+          constexpr int NGV = DV / 64;
+          const int64_t rb = vbase / DV * (int64_t)NGV
+                             + (int64_t)(s0 + row) * NGV + col / 64;
+          const float vs = bits2f(vscale[rb]);
+          const int2 vq = *reinterpret_cast<const int2*>(
+              (const int8_t*)vc + vbase + (int64_t)(s0 + row) * DV + col);
+          const int8_t* vb = reinterpret_cast<const int8_t*>(&vq);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vv8.x[j] = f2bits((float)vb[j] * vs);
+        } else {
+          vv8 = *reinterpret_cast<const short8*>(
+              (const short*)vc + vbase + (int64_t)(s0 + row) * DV + col);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vv8.x[j] = 0;
+      }
+      *reinterpret_cast<short8*>(&v_lds[row * PV + col]) = vv8;
     }
     __syncthreads();
 
@@ -163,7 +180,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
       // Accumulate P @ V: lane owns dims d = DPL*lane + j.
       for (int s = 0; s < valid; ++s) {
         const float ps = __shfl(p, s, 64);
-        const short* vrow = &v_lds[s * P + DPL * lane];
+        const short* vrow = &v_lds[s * PV + DPL * lane];
 #pragma unroll
         for (int j = 0; j < DPL; ++j)
           acc[hi][j] = fmaf(ps, bits2f(vrow[j]), acc[hi][j]);
@@ -177,12 +194,12 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
     for (int hi = 0; hi < nh; ++hi) {
       const int g = wid + hi * 4;
       float* prow = partials
-          + (((int64_t)b * Hq + hkv * G + g) * splits + split) * (D + 2);
+          + (((int64_t)b * Hq + hkv * G + g) * splits + split) * (DV + 2);
 #pragma unroll
       for (int j = 0; j < DPL; ++j) prow[DPL * lane + j] = acc[hi][j];
       if (lane == 0) {
-        prow[D] = m[hi];
-        prow[D + 1] = l[hi];
+        prow[DV] = m[hi];
+        prow[DV + 1] = l[hi];
       }
     }
     return;
@@ -201,7 +218,7 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
     } else {
       inv = (l[hi] > 0.f) ? 1.f / l[hi] : 0.f;
     }
-    short* orow = out + ((int64_t)b * Hq + hkv * G + g) * D + DPL * lane;
+    short* orow = out + ((int64_t)b * Hq + hkv * G + g) * DV + DPL * lane;
 #pragma unroll
     for (int j = 0; j < DPL; ++j) orow[j] = f2bits(acc[hi][j] * inv);
   }
@@ -247,20 +264,24 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  c10::optional<torch::Tensor> partials, int64_t splits) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = kcache.size(1), Smax = kcache.size(2);
+  const int64_t DV = vcache.size(3);
   DNET_CHECK(kcache.size(0) == B && kcache.size(3) == D, "kcache shape");
+  DNET_CHECK(out.size(2) == DV, "out width = v head dim");
   DNET_CHECK(Hq % Hkv == 0 && Hq / Hkv <= 16, "GQA group <= 16");
-  DNET_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  DNET_CHECK((D == 64 && DV == 64) || (D == 128 && DV == 128) ||
+                 (D == 192 && DV == 128),
+             "head dims must be 64/64, 128/128 or 192/128 (MLA)");
   DNET_CHECK(pos.dtype() == torch::kInt32, "pos int32");
   DNET_CHECK(q.stride(2) == 1 && q.stride(1) == D, "q inner dims contiguous");
   DNET_CHECK(kcache.is_contiguous() && vcache.is_contiguous() &&
                  out.is_contiguous(), "contig");
   auto stream = current_stream();
   const int G = (int)(Hq / Hkv);
-  const int P = (int)D + 4;
-  const size_t lds = (2 * kChunk * P + G * D) * sizeof(short);
+  const size_t lds = (kChunk * ((int)D + 4) + kChunk * ((int)DV + 4) + G * D)
+                     * sizeof(short);
   if (splits > 1) {
     DNET_CHECK(partials.has_value()
-                   && partials->numel() >= B * Hq * splits * (D + 2),
+                   && partials->numel() >= B * Hq * splits * (DV + 2),
                "split-S partials scratch required");
   }
   const dim3 grid((unsigned)(B * Hkv), (unsigned)splits);
@@ -271,20 +292,23 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const short* ksp = q8 ? (const short*)kscale->data_ptr() : nullptr;
   const short* vsp = q8 ? (const short*)vscale->data_ptr() : nullptr;
   float* pp = splits > 1 ? (float*)partials->data_ptr() : nullptr;
-#define LAUNCH_ATTN(DD, QQ)                                                   \
-  hipLaunchKernelGGL((attn_decode_kernel<DD, QQ>), grid, dim3(256), lds,      \
-                     stream, (const short*)q.data_ptr(), kcache.data_ptr(),   \
+#define LAUNCH_ATTN(DD, DDV, QQ)                                            \
+  hipLaunchKernelGGL((attn_decode_kernel<DD, DDV, QQ>), grid, dim3(256),      \
+                     lds, stream, (const short*)q.data_ptr(),                 \
+                     kcache.data_ptr(),                                       \
                      vcache.data_ptr(), ksp, vsp, (const int*)pos.data_ptr(), \
                      (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,    \
                      (float)scale, ldq, (int)window,                          \
                      (splits > 1 ? nullptr : skp), pp, (int)splits)
-  if (D == 128 && q8) LAUNCH_ATTN(128, true);
-  else if (D == 128) LAUNCH_ATTN(128, false);
-  else if (q8) LAUNCH_ATTN(64, true);
-  else LAUNCH_ATTN(64, false);
+  if (D == 192 && q8) LAUNCH_ATTN(192, 128, true);
+  else if (D == 192) LAUNCH_ATTN(192, 128, false);
+  else if (D == 128 && q8) LAUNCH_ATTN(128, 128, true);
+  else if (D == 128) LAUNCH_ATTN(128, 128, false);
+  else if (q8) LAUNCH_ATTN(64, 64, true);
+  else LAUNCH_ATTN(64, 64, false);
 #undef LAUNCH_ATTN
   if (splits > 1) {
-    if (D == 128)
+    if (DV == 128)
       hipLaunchKernelGGL((attn_combine_kernel<128>), dim3((unsigned)(B * Hq)),
                          dim3(kWave), 0, stream, pp, skp,
                          (short*)out.data_ptr(), (int)Hq, (int)splits);

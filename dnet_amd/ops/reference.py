@@ -164,8 +164,9 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     [max(0, len-window), len) (window=0 -> full)."""
     B, Hq, D = q.shape
     Hkv = kcache.shape[1]
+    Dv = vcache.shape[-1]
     G = Hq // Hkv
-    out = torch.zeros(q.shape, dtype=q.dtype, device=q.device)
+    out = torch.zeros(B, Hq, Dv, dtype=q.dtype, device=q.device)
     for b in range(B):
         ln = int(pos[b])
         if ln == 0:
@@ -185,7 +186,7 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         else:
             p = torch.softmax(s, dim=-1)
         o = torch.einsum("hgl,hld->hgd", p, v)
-        out[b] = o.reshape(Hq, D).to(q.dtype)
+        out[b] = o.reshape(Hq, Dv).to(q.dtype)
     return out
 
 

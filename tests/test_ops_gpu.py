@@ -429,3 +429,36 @@ def test_moe_dense_matches_sparse_gpu():
     finally:
         GptOssRingModel.DENSE_MOE_MAX_T = 64
     assert torch.allclose(dense.float(), sparse.float(), atol=4e-2, rtol=4e-2)
+
+
+@pytest.mark.parametrize("q8,force_splits", [(False, 1), (False, 4),
+                                             (True, 1)])
+def test_attn_decode_mla_shape(q8, force_splits, monkeypatch):
+    """MLA head dims (qk 192 / v 128) on the decode kernel vs fp32 ref."""
+    monkeypatch.setenv("DNET_ATTN_SPLITS", str(force_splits))
+    torch.manual_seed(11)
+    B, H, S, D, DV = 3, 4, 256, 192, 128
+    dev = _dev()
+    q = torch.randn(B, H, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, H, S, DV, dtype=torch.bfloat16, device=dev)
+    pos = torch.tensor([S, S // 2, 7], dtype=torch.int32, device=dev)
+    ks = vs = None
+    kq = vq = None
+    if q8:
+        kq, ks = ref.quantize_kv_rows(kc.reshape(-1, D))
+        vq, vs = ref.quantize_kv_rows(vc.reshape(-1, DV))
+        kq = kq.view(B, H, S, D)
+        ks = ks.view(B, H, S, D // 64)
+        vq = vq.view(B, H, S, DV)
+        vs = vs.view(B, H, S, DV // 64)
+    out = ops.attn_decode(q, kq if q8 else kc, vq if q8 else vc, pos,
+                          D ** -0.5, kscale=ks, vscale=vs)
+    out_ref = ref.attn_decode(q.cpu(), (kq if q8 else kc).cpu(),
+                              (vq if q8 else vc).cpu(), pos.cpu(),
+                              D ** -0.5,
+                              kscale=None if ks is None else ks.cpu(),
+                              vscale=None if vs is None else vs.cpu())
+    assert out.shape == (B, H, DV)
+    assert torch.allclose(out.float().cpu(), out_ref.float(),
+                          atol=4e-2, rtol=4e-2)
