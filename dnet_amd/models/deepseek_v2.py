@@ -43,8 +43,9 @@ class DeepseekV2RingModel(RingModel):
     def __init__(self, cfg: ModelConfig, layer_ids, device, is_first, is_last,
                  smax: int = 4096, tp_rank: int = 0, tp_size: int = 1,
                  tp_group=None):
-        assert tp_size == 1, "deepseek MLA TP is roadmap"
-        super().__init__(cfg, layer_ids, device, is_first, is_last, smax)
+        assert cfg.num_q_heads % tp_size == 0, "heads must divide tp"
+        super().__init__(cfg, layer_ids, device, is_first, is_last, smax,
+                         tp_rank=tp_rank, tp_size=tp_size, tp_group=tp_group)
         # rope tables over the rope sub-dim only
         cos, sin = ops.rope_tables(smax, cfg.qk_rope_head_dim, cfg.rope_theta,
                                    scaling=cfg.rope_scaling)
@@ -52,9 +53,31 @@ class DeepseekV2RingModel(RingModel):
         self.sin = sin.to(self.device)
         kd = cfg.qk_nope_head_dim + cfg.qk_rope_head_dim
         self.scale = kd ** -0.5
+        # MLA TP: attention heads shard across the stage's tp group (kv_a
+        # latent projection is replicated; q/kv_b/o are per-head sliced)
+        self.nh = cfg.num_q_heads // self.tp_size
+
+    def _slice_heads_rows(self, w: torch.Tensor, per_head: int) -> torch.Tensor:
+        if self.tp_size == 1:
+            return w
+        h0 = self.tp_rank * self.nh
+        return w.view(self.cfg.num_q_heads, per_head, -1)[h0:h0 + self.nh] \
+                .reshape(self.nh * per_head, -1)
+
+    def _slice_heads_cols(self, w: torch.Tensor, per_head: int) -> torch.Tensor:
+        if self.tp_size == 1:
+            return w
+        h0 = self.tp_rank * self.nh
+        return w.view(w.shape[0], self.cfg.num_q_heads, per_head) \
+                [:, h0:h0 + self.nh].reshape(w.shape[0], self.nh * per_head)
 
     def make_kv_cache(self, batch: int, smax: int) -> MLAKVCache:
-        return MLAKVCache(self.cfg, self.layer_ids, batch, smax, self.device)
+        cfg = self.cfg
+        if self.tp_size > 1:
+            import copy
+            cfg = copy.copy(cfg)
+            cfg.num_q_heads = self.nh
+        return MLAKVCache(cfg, self.layer_ids, batch, smax, self.device)
 
     # ---------- weights ----------
 
@@ -71,21 +94,24 @@ class DeepseekV2RingModel(RingModel):
             lw.q_a = Linear.make(rand(c.q_lora_rank, c.hidden_size).to(dev),
                                  None, c.quant)
             lw.q_a_norm = ones(c.q_lora_rank)
-            lw.q_b = Linear.make(rand(c.num_q_heads * kd, c.q_lora_rank).to(dev),
-                                 None, c.quant)
+            lw.q_b = Linear.make(self._slice_heads_rows(
+                rand(c.num_q_heads * kd, c.q_lora_rank), kd).to(dev),
+                None, c.quant)
         else:
-            lw.q = Linear.make(rand(c.num_q_heads * kd, c.hidden_size).to(dev),
-                               None, c.quant)
+            lw.q = Linear.make(self._slice_heads_rows(
+                rand(c.num_q_heads * kd, c.hidden_size), kd).to(dev),
+                None, c.quant)
         lw.kv_a = Linear.make(
             rand(c.kv_lora_rank + c.qk_rope_head_dim, c.hidden_size).to(dev),
             None, None)
         lw.kv_a_norm = ones(c.kv_lora_rank)
-        lw.kv_b = Linear.make(
+        lw.kv_b = Linear.make(self._slice_heads_rows(
             rand(c.num_q_heads * (c.qk_nope_head_dim + c.v_head_dim),
-                 c.kv_lora_rank).to(dev), None, c.quant)
-        lw.o = Linear.make(
-            rand(c.hidden_size, c.num_q_heads * c.v_head_dim).to(dev),
+                 c.kv_lora_rank), c.qk_nope_head_dim + c.v_head_dim).to(dev),
             None, c.quant)
+        lw.o = Linear.make(self._slice_heads_cols(
+            rand(c.hidden_size, c.num_q_heads * c.v_head_dim),
+            c.v_head_dim).to(dev), None, c.quant)
         inter = c.moe_intermediate_size or c.intermediate_size
         if c.num_experts and lid >= c.first_k_dense_replace:
             lw.router = Linear(rand(c.num_experts, c.hidden_size).to(dev))
@@ -97,15 +123,17 @@ class DeepseekV2RingModel(RingModel):
                 for _ in range(c.num_experts)]
             if c.n_shared_experts:
                 si = inter * c.n_shared_experts
-                lw.shared_gateup = Linear.make(
-                    rand(2 * si, c.hidden_size).to(dev), None, c.quant)
-                lw.shared_down = Linear.make(
-                    rand(c.hidden_size, si).to(dev), None, c.quant)
+                lw.shared_gateup = Linear.make(self._slice_gateup(
+                    rand(2 * si, c.hidden_size)).to(dev), None, c.quant)
+                lw.shared_down = Linear.make(self._slice_cols(
+                    rand(c.hidden_size, si)).to(dev), None, c.quant)
         else:
-            lw.gateup = Linear.make(rand(2 * c.intermediate_size,
-                                         c.hidden_size).to(dev), None, c.quant)
-            lw.down = Linear.make(rand(c.hidden_size,
-                                       c.intermediate_size).to(dev), None, c.quant)
+            lw.gateup = Linear.make(self._slice_gateup(
+                rand(2 * c.intermediate_size, c.hidden_size)).to(dev),
+                None, c.quant)
+            lw.down = Linear.make(self._slice_cols(
+                rand(c.hidden_size, c.intermediate_size)).to(dev),
+                None, c.quant)
         return lw
 
     def load_state_dict(self, sd: dict):
@@ -124,29 +152,34 @@ class DeepseekV2RingModel(RingModel):
             lw = LayerWeights(
                 attn_norm=get(p + "input_layernorm.weight").to(dev),
                 mlp_norm=get(p + "post_attention_layernorm.weight").to(dev))
+            kd = c.qk_nope_head_dim + c.qk_rope_head_dim
             if get(p + "self_attn.q_proj.weight") is not None:
-                lw.q = Linear.make(get(p + "self_attn.q_proj.weight").to(dev),
-                                   None, c.quant)
+                lw.q = Linear.make(self._slice_heads_rows(
+                    get(p + "self_attn.q_proj.weight"), kd).to(dev),
+                    None, c.quant)
             else:
                 lw.q_a = Linear.make(
                     get(p + "self_attn.q_a_proj.weight").to(dev), None, c.quant)
                 lw.q_a_norm = get(p + "self_attn.q_a_layernorm.weight").to(dev)
-                lw.q_b = Linear.make(
-                    get(p + "self_attn.q_b_proj.weight").to(dev), None, c.quant)
+                lw.q_b = Linear.make(self._slice_heads_rows(
+                    get(p + "self_attn.q_b_proj.weight"), kd).to(dev),
+                    None, c.quant)
             lw.kv_a = Linear.make(
                 get(p + "self_attn.kv_a_proj_with_mqa.weight").to(dev), None, None)
             lw.kv_a_norm = get(p + "self_attn.kv_a_layernorm.weight").to(dev)
-            lw.kv_b = Linear.make(get(p + "self_attn.kv_b_proj.weight").to(dev),
-                                  None, c.quant)
-            lw.o = Linear.make(get(p + "self_attn.o_proj.weight").to(dev),
-                               None, c.quant)
+            lw.kv_b = Linear.make(self._slice_heads_rows(
+                get(p + "self_attn.kv_b_proj.weight"),
+                c.qk_nope_head_dim + c.v_head_dim).to(dev), None, c.quant)
+            lw.o = Linear.make(self._slice_heads_cols(
+                get(p + "self_attn.o_proj.weight"), c.v_head_dim).to(dev),
+                None, c.quant)
             if get(p + "mlp.gate_proj.weight") is not None:     # dense layer
-                lw.gateup = Linear.make(
+                lw.gateup = Linear.make(self._slice_gateup(
                     torch.cat([get(p + "mlp.gate_proj.weight"),
-                               get(p + "mlp.up_proj.weight")]).to(dev),
+                               get(p + "mlp.up_proj.weight")])).to(dev),
                     None, c.quant)
-                lw.down = Linear.make(get(p + "mlp.down_proj.weight").to(dev),
-                                      None, c.quant)
+                lw.down = Linear.make(self._slice_cols(
+                    get(p + "mlp.down_proj.weight")).to(dev), None, c.quant)
             else:                                               # MoE layer
                 lw.router = Linear(get(p + "mlp.gate.weight").to(dev))
                 gu = get(p + "mlp.experts.gate_up_proj")    # [E, 2I, H]
@@ -158,12 +191,12 @@ class DeepseekV2RingModel(RingModel):
                     lw.experts_down.append(Linear.make(
                         dn[e].contiguous().to(dev), None, c.quant))
                 if get(p + "mlp.shared_experts.gate_proj.weight") is not None:
-                    lw.shared_gateup = Linear.make(
+                    lw.shared_gateup = Linear.make(self._slice_gateup(
                         torch.cat([get(p + "mlp.shared_experts.gate_proj.weight"),
                                    get(p + "mlp.shared_experts.up_proj.weight")]
-                                  ).to(dev), None, c.quant)
-                    lw.shared_down = Linear.make(
-                        get(p + "mlp.shared_experts.down_proj.weight").to(dev),
+                                  )).to(dev), None, c.quant)
+                    lw.shared_down = Linear.make(self._slice_cols(
+                        get(p + "mlp.shared_experts.down_proj.weight")).to(dev),
                         None, c.quant)
             self.layers[lid] = lw
         if self.is_first:
@@ -183,22 +216,23 @@ class DeepseekV2RingModel(RingModel):
         T = y.shape[0]
         nope, rope, vd = c.qk_nope_head_dim, c.qk_rope_head_dim, c.v_head_dim
         kd = nope + rope
+        nh = self.nh
         if getattr(lw, "q_a", None) is not None:
             qa = ops.rmsnorm(lw.q_a(y), None, lw.q_a_norm, c.rms_eps)
-            q = lw.q_b(qa).view(T, c.num_q_heads, kd)
+            q = lw.q_b(qa).view(T, nh, kd)
         else:
-            q = lw.q(y).view(T, c.num_q_heads, kd)
+            q = lw.q(y).view(T, nh, kd)
         comp = lw.kv_a(y)                              # [T, lora + rope]
         c_kv = ops.rmsnorm(comp[:, :c.kv_lora_rank].contiguous(), None,
                            lw.kv_a_norm, c.rms_eps)
         k_pe = comp[:, c.kv_lora_rank:].view(T, 1, rope)
-        kv = lw.kv_b(c_kv).view(T, c.num_q_heads, nope + vd)
+        kv = lw.kv_b(c_kv).view(T, nh, nope + vd)
         k_nope, v = kv[..., :nope], kv[..., nope:]
         q_pe = ops.rope_apply(q[..., nope:].contiguous(), self.cos, self.sin,
                               positions)
         k_pe = ops.rope_apply(k_pe.contiguous(), self.cos, self.sin, positions)
         q = torch.cat([q[..., :nope], q_pe], dim=-1)
-        k = torch.cat([k_nope, k_pe.expand(T, c.num_q_heads, rope)], dim=-1)
+        k = torch.cat([k_nope, k_pe.expand(T, nh, rope)], dim=-1)
         return q, k, v.contiguous()
 
     def decode_window(self, h, layer_ids, kv):
@@ -214,17 +248,17 @@ class DeepseekV2RingModel(RingModel):
             li = kv.local[lid]
             # vectorized append at per-batch positions
             idx = kv.pos.long().view(B, 1, 1, 1)
-            kv.k[li].scatter_(2, idx.expand(B, c.num_q_heads, 1, k.shape[-1]),
+            kv.k[li].scatter_(2, idx.expand(B, self.nh, 1, k.shape[-1]),
                               k.unsqueeze(2))
-            kv.v[li].scatter_(2, idx.expand(B, c.num_q_heads, 1, v.shape[-1]),
+            kv.v[li].scatter_(2, idx.expand(B, self.nh, 1, v.shape[-1]),
                               v.unsqueeze(2))
             len_t = kv.pos + 1
             # MLA decode attention on the native kernel (D=192 qk / 128 v)
             attn = ops.attn_decode(q.contiguous(), kv.k[li], kv.v[li],
                                    len_t, self.scale)
-            o = lw.o(attn.reshape(B, -1))
+            o = self._tp_reduce(lw.o(attn.reshape(B, -1)))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
-            delta = self._mlp_for(lid, y2, lw)
+            delta = self._tp_reduce(self._mlp_for(lid, y2, lw))
         h.add_(delta)
         return h
 
@@ -238,18 +272,19 @@ class DeepseekV2RingModel(RingModel):
             y = ops.rmsnorm(flat, None, lw.attn_norm, c.rms_eps)
             q, k, v = self._mla_qkv(y, lw, positions.repeat(B))
             kd, vd = k.shape[-1], v.shape[-1]
-            q = q.view(B, T, c.num_q_heads, kd).transpose(1, 2)
-            k = k.view(B, T, c.num_q_heads, kd).transpose(1, 2)
-            v = v.view(B, T, c.num_q_heads, vd).transpose(1, 2)
+            q = q.view(B, T, self.nh, kd).transpose(1, 2)
+            k = k.view(B, T, self.nh, kd).transpose(1, 2)
+            v = v.view(B, T, self.nh, vd).transpose(1, 2)
             li = kv.local[lid]
             kv.k[li][:, :, p0:p0 + T] = k
             kv.v[li][:, :, p0:p0 + T] = v
             attn = _chunked_causal_attention(
                 q, kv.k[li][:, :, :p0 + T], kv.v[li][:, :, :p0 + T],
                 self.scale, p0)
-            o = lw.o(attn.transpose(1, 2).reshape(B * T, -1).contiguous())
+            o = self._tp_reduce(
+                lw.o(attn.transpose(1, 2).reshape(B * T, -1).contiguous()))
             y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)
-            flat.add_(self._mlp_for(lid, y2, lw))
+            flat.add_(self._tp_reduce(self._mlp_for(lid, y2, lw)))
         return h
 
     def _mlp_for(self, lid: int, y: torch.Tensor, lw) -> torch.Tensor:
@@ -264,12 +299,17 @@ class DeepseekV2RingModel(RingModel):
         T = y.shape[0]
         if T <= 64 and lw.experts_gateup[0].bits in (8, 16):
             # grouped-expert kernels (graph-safe; unrouted experts skipped
-            # on device) — see models/moe.py
+            # on device) — see models/moe.py. EP: e % tp == rank; partial
+            # sums reduced by the caller.
             from .moe import stack_experts
-            st = stack_experts(lw, list(range(c.num_experts)))
+            local = [e for e in range(c.num_experts)
+                     if self.tp_size <= 1 or e % self.tp_size == self.tp_rank]
+            st = stack_experts(lw, local)
             we = torch.zeros(T, c.num_experts, dtype=torch.float32,
                              device=y.device)
             we.scatter_(1, idx, weights)
+            if self.tp_size > 1:
+                we = we.index_select(1, st["local"]).contiguous()
             act = ops.moe_gateup(y, st["gw"], st["gs"], st["gb"], we,
                                  st["group"], st["packed"], 0)
             out = ops.moe_down(act, st["dw"], st["ds"], st["db"], we,
@@ -277,6 +317,8 @@ class DeepseekV2RingModel(RingModel):
         else:
             out = torch.zeros_like(y, dtype=torch.float32)
             for e in range(c.num_experts):
+                if self.tp_size > 1 and e % self.tp_size != self.tp_rank:
+                    continue
                 we_full = (weights * (idx == e)).sum(dim=-1)
                 mask = we_full > 0
                 if not bool(mask.any()):

@@ -267,3 +267,63 @@ def test_bench_contract_multiproc_cpu(tmp_path):
     assert j["value"] > 0 and j["ms_per_step"] > 0
     assert j["scaling"] == "weak" and "config" in j
     assert j["config"]["global_batch"] == 2 * 2 * 2  # mb_count(2*2) * mb_size
+
+
+DS_CFG = dict(model_type="deepseek_v2", hidden_size=64, num_hidden_layers=3,
+              num_attention_heads=4, num_key_value_heads=4, vocab_size=128,
+              intermediate_size=128, kv_lora_rank=32, qk_nope_head_dim=16,
+              qk_rope_head_dim=8, v_head_dim=16, n_routed_experts=4,
+              num_experts_per_tok=2, n_shared_experts=1,
+              moe_intermediate_size=32, first_k_dense_replace=1,
+              routed_scaling_factor=1.0, rope_theta=10000.0)
+
+
+def _ds_run_single():
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    cfg = ModelConfig.from_hf(DS_CFG)
+    ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=MB_COUNT, mb_size=MB_SIZE,
+                      smax=64, seed=7, use_graphs=False)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    return torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+
+def _ds_rank_main_tp(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(DS_CFG)
+    ex = RingExecutor(cfg, rank, world, "cpu", mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False,
+                      tp=2)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_deepseek_tp2_matches_single():
+    """DeepSeek MLA with TP=2 (per-head q/kv_b/o sharding + EP experts +
+    sliced shared experts) == single-rank, token-exact."""
+    single = _ds_run_single()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_ds_rank_main_tp, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert torch.equal(out, single), f"ds tp2 != single:\n{out}\n{single}"
