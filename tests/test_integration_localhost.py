@@ -139,6 +139,23 @@ def test_full_flow(cluster):
                     chunks.append(json.loads(line[6:]))
         assert chunks and chunks[-1].get("usage") is not None
 
+        # two concurrent chats: the shard driver serializes them per ring;
+        # token frames demux by nonce and both requests complete
+        import concurrent.futures as cf
+
+        def one(i):
+            rr = httpx.post(f"{api}/v1/chat/completions", json={
+                "model": "tiny-random", "stream": False, "max_tokens": 6,
+                "messages": [{"role": "user", "content": f"concurrent {i}"}]},
+                timeout=120)
+            return rr.status_code, rr.json()
+
+        with cf.ThreadPoolExecutor(2) as pool:
+            results = list(pool.map(one, range(2)))
+        for code, body2 in results:
+            assert code == 200, body2
+            assert body2["usage"]["completion_tokens"] >= 1
+
         # unload
         r = c.post(f"{api}/v1/unload_model")
         assert r.status_code == 200
