@@ -225,3 +225,36 @@ unpack_int8_mfma = ref.unpack_int8_mfma
 rope_tables = ref.rope_tables
 rope_apply = ref.rope_apply
 mxfp4_dequant = ref.mxfp4_dequant
+
+
+def attn_decode_partials(q, kcache, vcache, pos, scale,
+                         kscale=None, vscale=None) -> torch.Tensor:
+    """Local flash-decode partials [B, Hq, splits, Dv+2] f32 over THIS
+    cache shard — the combinable form for context-parallel attention
+    (combine with ``attn_combine`` after gathering shards)."""
+    if q.is_cuda:
+        B, Hq, _ = q.shape
+        dv = vcache.shape[-1]
+        splits = _attn_splits(B, kcache.shape[1], kcache.shape[2])
+        partials = torch.zeros(B, Hq, splits, dv + 2, dtype=torch.float32,
+                               device=q.device)
+        partials[..., dv] = -1e30   # empty shards stay neutral in combine
+        out = torch.empty(B, Hq, dv, dtype=q.dtype, device=q.device)
+        _native().attn_decode(q, kcache, vcache, pos, out, scale, 0, None,
+                              kscale, vscale, partials.view(-1), splits,
+                              False)
+        return partials
+    return ref.attn_decode_partials(q, kcache, vcache, pos, scale,
+                                    kscale, vscale)
+
+
+def attn_combine(partials: torch.Tensor,
+                 sinks: torch.Tensor | None = None) -> torch.Tensor:
+    """partials [B, Hq, S, Dv+2] -> out [B, Hq, Dv]."""
+    if partials.is_cuda:
+        B, Hq, S, W = partials.shape
+        out = torch.empty(B, Hq, W - 2, dtype=torch.bfloat16,
+                          device=partials.device)
+        _native().attn_combine(partials.contiguous().view(-1), sinks, out, S)
+        return out
+    return ref.attn_combine(partials, sinks)

@@ -189,8 +189,9 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
     __syncthreads();
   }
 
-  if (splits > 1) {
-    // write unnormalized partials; combine kernel normalizes + sinks
+  if (partials != nullptr) {
+    // write unnormalized partials; the combine kernel (or a cross-rank
+    // context-parallel combine) normalizes + folds sinks
     for (int hi = 0; hi < nh; ++hi) {
       const int g = wid + hi * 4;
       float* prow = partials
@@ -255,13 +256,38 @@ __global__ void attn_combine_kernel(const float* __restrict__ partials,
   for (int j = 0; j < DPL; ++j) orow[j] = f2bits(acc[j] * inv);
 }
 
+// merge split/rank partials [B*Hq, splits, DV+2] -> out [B, Hq, DV]
+// (cross-rank context parallelism gathers partials and calls this with
+// splits = world * local_splits).
+void attn_combine(torch::Tensor partials, c10::optional<torch::Tensor> sinks,
+                  torch::Tensor out, int64_t splits) {
+  const int64_t B = out.size(0), Hq = out.size(1), DV = out.size(2);
+  DNET_CHECK(DV == 64 || DV == 128, "combine DV 64/128");
+  DNET_CHECK(partials.is_contiguous() && out.is_contiguous(), "contig");
+  DNET_CHECK(partials.numel() >= B * Hq * splits * (DV + 2), "partials size");
+  auto stream = current_stream();
+  const short* skp = sinks.has_value() ? (const short*)sinks->data_ptr()
+                                       : nullptr;
+  if (DV == 128)
+    hipLaunchKernelGGL((attn_combine_kernel<128>), dim3((unsigned)(B * Hq)),
+                       dim3(kWave), 0, stream,
+                       (const float*)partials.data_ptr(), skp,
+                       (short*)out.data_ptr(), (int)Hq, (int)splits);
+  else
+    hipLaunchKernelGGL((attn_combine_kernel<64>), dim3((unsigned)(B * Hq)),
+                       dim3(kWave), 0, stream,
+                       (const float*)partials.data_ptr(), skp,
+                       (short*)out.data_ptr(), (int)Hq, (int)splits);
+}
+
 // q may be a column slice of a fused-QKV buffer: strides (ldq, D, 1).
 void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  torch::Tensor pos, torch::Tensor out, double scale,
                  int64_t window, c10::optional<torch::Tensor> sinks,
                  c10::optional<torch::Tensor> kscale,
                  c10::optional<torch::Tensor> vscale,
-                 c10::optional<torch::Tensor> partials, int64_t splits) {
+                 c10::optional<torch::Tensor> partials, int64_t splits,
+                 bool combine) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = kcache.size(1), Smax = kcache.size(2);
   const int64_t DV = vcache.size(3);
@@ -279,7 +305,7 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const int G = (int)(Hq / Hkv);
   const size_t lds = (kChunk * ((int)D + 4) + kChunk * ((int)DV + 4) + G * D)
                      * sizeof(short);
-  if (splits > 1) {
+  if (splits > 1 || !combine) {
     DNET_CHECK(partials.has_value()
                    && partials->numel() >= B * Hq * splits * (DV + 2),
                "split-S partials scratch required");
@@ -291,7 +317,8 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const bool q8 = kcache.dtype() == torch::kInt8;
   const short* ksp = q8 ? (const short*)kscale->data_ptr() : nullptr;
   const short* vsp = q8 ? (const short*)vscale->data_ptr() : nullptr;
-  float* pp = splits > 1 ? (float*)partials->data_ptr() : nullptr;
+  float* pp = (splits > 1 || !combine) ? (float*)partials->data_ptr()
+                                       : nullptr;
 #define LAUNCH_ATTN(DD, DDV, QQ)                                            \
   hipLaunchKernelGGL((attn_decode_kernel<DD, DDV, QQ>), grid, dim3(256),      \
                      lds, stream, (const short*)q.data_ptr(),                 \
@@ -307,7 +334,7 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   else if (q8) LAUNCH_ATTN(64, 64, true);
   else LAUNCH_ATTN(64, 64, false);
 #undef LAUNCH_ATTN
-  if (splits > 1) {
+  if (splits > 1 && combine) {
     if (DV == 128)
       hipLaunchKernelGGL((attn_combine_kernel<128>), dim3((unsigned)(B * Hq)),
                          dim3(kWave), 0, stream, pp, skp,

@@ -14,7 +14,10 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  int64_t window, c10::optional<torch::Tensor> sinks,
                  c10::optional<torch::Tensor> kscale,
                  c10::optional<torch::Tensor> vscale,
-                 c10::optional<torch::Tensor> partials, int64_t splits);
+                 c10::optional<torch::Tensor> partials, int64_t splits,
+                 bool combine);
+void attn_combine(torch::Tensor partials, c10::optional<torch::Tensor> sinks,
+                  torch::Tensor out, int64_t splits);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
                  torch::Tensor cos_table, torch::Tensor sin_table,
@@ -49,7 +52,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &dnet::rmsnorm, "fused RMSNorm (+residual)");
   m.def("gemv_bf16", &dnet::gemv_bf16, "bf16 decode GEMV");
   m.def("gemv_int8", &dnet::gemv_int8, "grouped-int8 W8A16 decode GEMV");
-  m.def("attn_decode", &dnet::attn_decode, "GQA decode attention vs KV cache");
+  m.def("attn_decode", &dnet::attn_decode, "GQA decode attention vs KV cache",
+        py::arg("q"), py::arg("kcache"), py::arg("vcache"), py::arg("pos"),
+        py::arg("out"), py::arg("scale"), py::arg("window"), py::arg("sinks"),
+        py::arg("kscale"), py::arg("vscale"), py::arg("partials"),
+        py::arg("splits"), py::arg("combine") = true);
+  m.def("attn_combine", &dnet::attn_combine,
+        "merge split/rank flash-decode partials");
   m.def("rope_append", &dnet::rope_append, "fused RoPE + KV append (decode)");
   m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");

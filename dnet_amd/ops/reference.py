@@ -360,3 +360,53 @@ def mxfp4_dequant(blocks: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
     return out.reshape(*blocks.shape[:-2],
                        blocks.shape[-2] * blocks.shape[-1] * 2).to(
                            torch.bfloat16)
+
+
+def attn_decode_partials(q, kcache, vcache, pos, scale,
+                         kscale=None, vscale=None) -> torch.Tensor:
+    """Unnormalized flash-decode partials over this cache shard:
+    [B, Hq, 1, Dv+2] f32 with acc (exp(s-m) @ V), m, l — the cross-rank /
+    cross-split combinable form (matches the kernel's split-S partials)."""
+    B, Hq, D = q.shape
+    Hkv = kcache.shape[1]
+    Dv = vcache.shape[-1]
+    G = Hq // Hkv
+    out = torch.zeros(B, Hq, 1, Dv + 2, dtype=torch.float32)
+    out[..., Dv] = -1e30
+    for b in range(B):
+        ln = int(pos[b])
+        if ln == 0:
+            continue
+        if kcache.dtype == torch.int8:
+            k = dequant_kv(kcache[b, :, :ln], kscale[b, :, :ln]).float()
+            v = dequant_kv(vcache[b, :, :ln], vscale[b, :, :ln]).float()
+        else:
+            k = kcache[b, :, :ln].float()
+            v = vcache[b, :, :ln].float()
+        qq = q[b].float().view(Hkv, G, D)
+        s = torch.einsum("hgd,hld->hgl", qq, k) * scale
+        m = s.amax(dim=-1, keepdim=True)
+        p = torch.exp(s - m)
+        acc = torch.einsum("hgl,hld->hgd", p, v)
+        out[b, :, 0, :Dv] = acc.reshape(Hq, Dv)
+        out[b, :, 0, Dv] = m.reshape(Hq)
+        out[b, :, 0, Dv + 1] = p.sum(-1).reshape(Hq)
+    return out
+
+
+def attn_combine(partials: torch.Tensor,
+                 sinks: torch.Tensor | None = None) -> torch.Tensor:
+    """partials [B, Hq, S, Dv+2] -> out [B, Hq, Dv] bf16 (merge shards)."""
+    B, Hq, S, W = partials.shape
+    Dv = W - 2
+    acc, m, l = partials[..., :Dv], partials[..., Dv], partials[..., Dv + 1]
+    M = m.amax(dim=-1)
+    if sinks is not None:
+        M = torch.maximum(M, sinks.float().view(1, Hq))
+    w = torch.exp(m - M.unsqueeze(-1))
+    denom = (l * w).sum(-1)
+    if sinks is not None:
+        denom = denom + torch.exp(sinks.float().view(1, Hq) - M)
+    num = (acc * w.unsqueeze(-1)).sum(2)
+    denom = denom.clamp_min(1e-30)
+    return (num / denom.unsqueeze(-1)).to(torch.bfloat16)

@@ -71,3 +71,32 @@ def test_quantize_int4_roundtrip():
     vals = ref.unpack_int4(p) + 8
     v = vals.view(8, 2, 4, 4, 8).permute(0, 1, 3, 2, 4).reshape(8, 256)
     assert torch.equal(v, ref.unpack_int4(q4) + 8)
+
+
+def test_attn_partials_combine_matches_full():
+    """Sharding the KV sequence into partials + combine == full attention
+    (the context-parallel decomposition, single process)."""
+    import dnet_amd.ops as ops
+    from dnet_amd.ops import reference as ref
+    torch.manual_seed(0)
+    B, Hq, Hkv, S, D = 2, 8, 2, 96, 64
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16)
+    kc = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16)
+    vc = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16)
+    pos = torch.tensor([90, 33], dtype=torch.int32)
+    full = ref.attn_decode(q, kc, vc, pos, D ** -0.5)
+    # two S shards of 48; local lengths clamp per shard
+    parts = []
+    for r in range(2):
+        s0, s1 = r * 48, (r + 1) * 48
+        ln = (pos - s0).clamp(0, s1 - s0).to(torch.int32)
+        parts.append(ops.attn_decode_partials(
+            q, kc[:, :, s0:s1].contiguous(), vc[:, :, s0:s1].contiguous(),
+            ln, D ** -0.5))
+    out = ops.attn_combine(torch.cat(parts, dim=2))
+    assert torch.allclose(out.float(), full.float(), atol=3e-2, rtol=3e-2)
+    # sinks fold in at combine time (gpt-oss)
+    sinks = torch.randn(Hq, dtype=torch.bfloat16)
+    full_s = ref.attn_decode(q, kc, vc, pos, D ** -0.5, sinks=sinks)
+    out_s = ops.attn_combine(torch.cat(parts, dim=2), sinks)
+    assert torch.allclose(out_s.float(), full_s.float(), atol=3e-2, rtol=3e-2)

@@ -462,3 +462,26 @@ def test_attn_decode_mla_shape(q8, force_splits, monkeypatch):
     assert out.shape == (B, H, DV)
     assert torch.allclose(out.float().cpu(), out_ref.float(),
                           atol=4e-2, rtol=4e-2)
+
+
+def test_attn_partials_combine_gpu():
+    """Native partials + native combine across simulated sequence shards
+    == fp32 reference full attention (context-parallel decomposition)."""
+    torch.manual_seed(13)
+    B, Hq, Hkv, S, D = 2, 8, 2, 128, 128
+    dev = _dev()
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    pos = torch.tensor([120, 45], dtype=torch.int32, device=dev)
+    full = ref.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), pos.cpu(), D ** -0.5)
+    parts = []
+    for r in range(2):
+        s0, s1 = r * 64, (r + 1) * 64
+        ln = (pos - s0).clamp(0, 64).to(torch.int32)
+        parts.append(ops.attn_decode_partials(
+            q, kc[:, :, s0:s1].contiguous(), vc[:, :, s0:s1].contiguous(),
+            ln, D ** -0.5))
+    out = ops.attn_combine(torch.cat(parts, dim=2))
+    assert torch.allclose(out.float().cpu(), full.float(),
+                          atol=3e-2, rtol=3e-2)
