@@ -33,14 +33,10 @@ def shard_bounds(total_len: int, world: int, rank: int) -> tuple[int, int]:
     return s0, s1
 
 
-def local_lengths(pos: torch.Tensor, world: int, rank: int) -> torch.Tensor:
-    """Per-sequence local cache length on this rank given global lengths
-    ``pos`` and block sharding by capacity: rank r owns positions
-    [r*cap, (r+1)*cap) of each sequence."""
-    # used when every rank allocates `cap` positions; global position p
-    # lives on rank p // cap
-    raise NotImplementedError("capacity-block layout lands with executor "
-                              "integration; tests use shard_bounds")
+def local_lengths(pos: torch.Tensor, cap: int, rank: int) -> torch.Tensor:
+    """Per-sequence valid length on this rank for capacity-block sharding
+    (rank r owns global positions [r*cap, (r+1)*cap))."""
+    return (pos - rank * cap).clamp(0, cap).to(torch.int32)
 
 
 def cp_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
