@@ -49,6 +49,8 @@ def main():
     ap.add_argument("--kv-bits", type=int, default=16, choices=[8, 16])
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree inside each pipeline stage")
+    ap.add_argument("--cp", type=int, default=1,
+                    help="context-parallel degree (KV sequence sharding)")
     ap.add_argument("--residency", type=int, default=0,
                     help="GPU-resident layers per rank (0=all; <local layers "
                          "enables host-DRAM weight streaming)")
@@ -73,6 +75,7 @@ def main():
                       mb_size=args.mb_size, smax=args.smax, seed=1234,
                       use_graphs=on_gpu and not args.no_graphs and args.tp == 1,
                       residency=args.residency, tp=args.tp,
+                      cp=args.cp,
                       kv_bits=args.kv_bits)
 
     g = torch.Generator().manual_seed(7)
@@ -134,8 +137,9 @@ def main():
                 "global_batch": total_seqs,
                 "seq_len": args.prompt_len,
                 "gen_len": args.steps,
-                "parallelism": (f"ring-pp{world // args.tp}"
-                                + (f"xtp{args.tp}" if args.tp > 1 else ""))
+                "parallelism": (f"ring-pp{world // (args.tp * args.cp)}"
+                                + (f"xtp{args.tp}" if args.tp > 1 else "")
+                                + (f"xcp{args.cp}" if args.cp > 1 else ""))
                                + (f"+offload(res={args.residency})"
                                   if args.residency else ""),
                 "microbatches": mb_count,

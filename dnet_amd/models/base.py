@@ -287,7 +287,8 @@ class RingModel:
 
     def __init__(self, cfg: ModelConfig, layer_ids: Sequence[int], device,
                  is_first: bool, is_last: bool, smax: int = 4096,
-                 tp_rank: int = 0, tp_size: int = 1, tp_group=None):
+                 tp_rank: int = 0, tp_size: int = 1, tp_group=None,
+                 cp_rank: int = 0, cp_size: int = 1, cp_group=None):
         self.cfg = cfg
         self.layer_ids = list(layer_ids)
         self.device = torch.device(device)
@@ -301,6 +302,15 @@ class RingModel:
         self.tp_rank = tp_rank
         self.tp_size = tp_size
         self.tp_group = tp_group
+        # context parallelism: the KV cache shards along the SEQUENCE axis
+        # across cp_group (weights replicated); decode attention merges the
+        # ranks' flash-decode partials (parallel/context.py). Mutually
+        # exclusive with TP inside one stage (v1).
+        self.cp_rank = cp_rank
+        self.cp_size = cp_size
+        self.cp_group = cp_group
+        self.cp_cap = -(-smax // cp_size) if cp_size > 1 else smax
+        assert not (tp_size > 1 and cp_size > 1)
         if tp_size > 1:
             assert cfg.num_q_heads % tp_size == 0
             assert cfg.num_kv_heads % tp_size == 0
@@ -500,6 +510,8 @@ class RingModel:
             import copy
             cfg = copy.copy(cfg)
             cfg.num_kv_heads = self.nkv_local
+        if self.cp_size > 1:
+            smax = self.cp_cap      # this rank's sequence shard
         return KVCache(cfg, self.layer_ids, batch, smax, self.device,
                        kv_bits=self.kv_bits)
 
@@ -539,15 +551,33 @@ class RingModel:
             if c.qk_norm:
                 self._qk_norm(q, k, lw)
             li = kv.local[lid]
-            ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos,
-                            self.sin,
-                            kv.kscale[li] if kv.quantized else None,
-                            kv.vscale[li] if kv.quantized else None)
             window, sinks = self._attn_params(lid)
-            attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5,
-                                   window, sinks,
-                                   kv.kscale[li] if kv.quantized else None,
-                                   kv.vscale[li] if kv.quantized else None)
+            if self.cp_size > 1:
+                # sequence-sharded KV: write locally, attend via gathered
+                # flash-decode partials (numerically = full attention)
+                from ..parallel.context import cp_attn_decode, local_lengths
+                assert not window, "sliding window + CP is roadmap"
+                wpos = (kv.pos - self.cp_rank * self.cp_cap).int()
+                ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos,
+                                self.cos, self.sin,
+                                kv.kscale[li] if kv.quantized else None,
+                                kv.vscale[li] if kv.quantized else None,
+                                wpos=wpos)
+                ln = local_lengths(kv.pos + 1, self.cp_cap, self.cp_rank)
+                attn = cp_attn_decode(
+                    q, kv.k[li], kv.v[li], ln, d ** -0.5,
+                    group=self.cp_group, sinks=sinks,
+                    kscale=kv.kscale[li] if kv.quantized else None,
+                    vscale=kv.vscale[li] if kv.quantized else None)
+            else:
+                ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos,
+                                self.sin,
+                                kv.kscale[li] if kv.quantized else None,
+                                kv.vscale[li] if kv.quantized else None)
+                attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5,
+                                       window, sinks,
+                                       kv.kscale[li] if kv.quantized else None,
+                                       kv.vscale[li] if kv.quantized else None)
             o = self._tp_reduce(lw.o(attn.view(B, nq * d)))
             y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
             delta = self._tp_reduce(self._mlp(y2, lw))
@@ -555,6 +585,14 @@ class RingModel:
                 self._profile_layer_sync(lid)
         h.add_(delta)
         return h
+
+    def _cp_gather(self, t: torch.Tensor) -> torch.Tensor:
+        """All-gather sequence shards [B, H, cap, D] -> [B, H, cp*cap, D]
+        (rank order = global position order)."""
+        import torch.distributed as dist
+        parts = [torch.empty_like(t) for _ in range(self.cp_size)]
+        dist.all_gather(parts, t.contiguous(), group=self.cp_group)
+        return torch.cat(parts, dim=2)
 
     def _profile_layer_sync(self, lid: int):
         import time
@@ -599,21 +637,50 @@ class RingModel:
             q = ops.rope_apply(q, self.cos, self.sin, positions)
             k = ops.rope_apply(k, self.cos, self.sin, positions)
             li = kv.local[lid]
-            if kv.quantized:
-                from ..ops import reference as _r
-                kc, ks = _r.quantize_kv_rows(k.transpose(1, 2))
-                vc, vs = _r.quantize_kv_rows(v.transpose(1, 2))
-                kv.k[li][:, :, p0:p0 + T] = kc
-                kv.kscale[li][:, :, p0:p0 + T] = ks
-                kv.v[li][:, :, p0:p0 + T] = vc
-                kv.vscale[li][:, :, p0:p0 + T] = vs
+            kt, vt = k.transpose(1, 2), v.transpose(1, 2)
+            if self.cp_size > 1:
+                # write only this rank's sequence shard; attention gathers
+                # the shards transiently (per layer) — steady-state KV
+                # memory stays sharded
+                cap, r = self.cp_cap, self.cp_rank
+                ls0, ls1 = max(p0, r * cap), min(p0 + T, (r + 1) * cap)
+                if ls1 > ls0:
+                    ksl = kt[:, :, ls0 - p0:ls1 - p0]
+                    vsl = vt[:, :, ls0 - p0:ls1 - p0]
+                    if kv.quantized:
+                        from ..ops import reference as _r
+                        kc, ks = _r.quantize_kv_rows(ksl)
+                        vc, vs = _r.quantize_kv_rows(vsl)
+                        kv.k[li][:, :, ls0 - r * cap:ls1 - r * cap] = kc
+                        kv.kscale[li][:, :, ls0 - r * cap:ls1 - r * cap] = ks
+                        kv.v[li][:, :, ls0 - r * cap:ls1 - r * cap] = vc
+                        kv.vscale[li][:, :, ls0 - r * cap:ls1 - r * cap] = vs
+                    else:
+                        kv.k[li][:, :, ls0 - r * cap:ls1 - r * cap] = ksl
+                        kv.v[li][:, :, ls0 - r * cap:ls1 - r * cap] = vsl
+                window, sinks = self._attn_params(lid)
+                assert not window, "sliding window + CP is roadmap"
+                kfull = self._cp_gather(kv.k_deq(li))[:, :, :p0 + T]
+                vfull = self._cp_gather(kv.v_deq(li))[:, :, :p0 + T]
+                attn = _chunked_causal_attention(
+                    q.transpose(1, 2), kfull, vfull, d ** -0.5, p0,
+                    window, sinks)
             else:
-                kv.k[li][:, :, p0:p0 + T] = k.transpose(1, 2)
-                kv.v[li][:, :, p0:p0 + T] = v.transpose(1, 2)
-            window, sinks = self._attn_params(lid)
-            attn = _chunked_causal_attention(
-                q.transpose(1, 2), kv.k_deq(li)[:, :, :p0 + T],
-                kv.v_deq(li)[:, :, :p0 + T], d ** -0.5, p0, window, sinks)
+                if kv.quantized:
+                    from ..ops import reference as _r
+                    kc, ks = _r.quantize_kv_rows(kt)
+                    vc, vs = _r.quantize_kv_rows(vt)
+                    kv.k[li][:, :, p0:p0 + T] = kc
+                    kv.kscale[li][:, :, p0:p0 + T] = ks
+                    kv.v[li][:, :, p0:p0 + T] = vc
+                    kv.vscale[li][:, :, p0:p0 + T] = vs
+                else:
+                    kv.k[li][:, :, p0:p0 + T] = kt
+                    kv.v[li][:, :, p0:p0 + T] = vt
+                window, sinks = self._attn_params(lid)
+                attn = _chunked_causal_attention(
+                    q.transpose(1, 2), kv.k_deq(li)[:, :, :p0 + T],
+                    kv.v_deq(li)[:, :, :p0 + T], d ** -0.5, p0, window, sinks)
             o = self._tp_reduce(
                 lw.o(attn.transpose(1, 2).reshape(B * T, nq * d).contiguous()))
             y2 = ops.rmsnorm(o, flat, lw.mlp_norm, c.rms_eps)

@@ -168,12 +168,13 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
 
 
 def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
-                kscale=None, vscale=None) -> None:
+                kscale=None, vscale=None, wpos=None) -> None:
     if q.is_cuda:
         _native().rope_append(q, k, v, kcache, vcache, pos, cos, sin,
-                              kscale, vscale)
+                              kscale, vscale, wpos)
         return
-    ref.rope_append(q, k, v, kcache, vcache, pos, cos, sin, kscale, vscale)
+    ref.rope_append(q, k, v, kcache, vcache, pos, cos, sin, kscale, vscale,
+                    wpos)
 
 
 def swiglu(gu: torch.Tensor) -> torch.Tensor:

@@ -22,7 +22,8 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
                  torch::Tensor cos_table, torch::Tensor sin_table,
                  c10::optional<torch::Tensor> kscale,
-                 c10::optional<torch::Tensor> vscale);
+                 c10::optional<torch::Tensor> vscale,
+                 c10::optional<torch::Tensor> wpos);
 void swiglu(torch::Tensor gu, torch::Tensor y);
 void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group, bool packed);
@@ -59,7 +60,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("splits"), py::arg("combine") = true);
   m.def("attn_combine", &dnet::attn_combine,
         "merge split/rank flash-decode partials");
-  m.def("rope_append", &dnet::rope_append, "fused RoPE + KV append (decode)");
+  m.def("rope_append", &dnet::rope_append, "fused RoPE + KV append (decode)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("kcache"),
+        py::arg("vcache"), py::arg("pos"), py::arg("cos_table"),
+        py::arg("sin_table"), py::arg("kscale"), py::arg("vscale"),
+        py::arg("wpos") = c10::nullopt);
   m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");
   m.def("dequant_int4", &dnet::dequant_int4, "packed-int4 -> bf16 dequant");

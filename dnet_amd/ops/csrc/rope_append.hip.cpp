@@ -24,6 +24,7 @@ __global__ void rope_append_kernel(short* __restrict__ q,
                                    short* __restrict__ kscale,
                                    short* __restrict__ vscale,
                                    const int* __restrict__ pos,
+                                   const int* __restrict__ wpos,
                                    const float* __restrict__ cost,
                                    const float* __restrict__ sint,
                                    const int Hq, const int Hkv, const int Smax,
@@ -58,7 +59,13 @@ __global__ void rope_append_kernel(short* __restrict__ q,
     v1 = bits2f(vrow[i]);
     v2 = bits2f(vrow[i + half]);
   }
-  const int64_t rowbase = ((int64_t)b * Hkv + hk) * Smax + p;
+  // context parallelism: the cache write targets the LOCAL position
+  // wpos[b] (rope still uses the global p); rows outside this rank's
+  // shard are skipped (uniform per block, so the Q8 wave reductions
+  // below never diverge)
+  const int wp = (wpos != nullptr) ? wpos[b] : p;
+  if (wp < 0 || wp >= Smax) return;
+  const int64_t rowbase = ((int64_t)b * Hkv + hk) * Smax + wp;
   if (!Q8) {
     if (!act) return;
     short* kdst = (short*)kcache + rowbase * D;
@@ -114,7 +121,8 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
                  torch::Tensor cos_table, torch::Tensor sin_table,
                  c10::optional<torch::Tensor> kscale,
-                 c10::optional<torch::Tensor> vscale) {
+                 c10::optional<torch::Tensor> vscale,
+                 c10::optional<torch::Tensor> wpos) {
   const int64_t B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int64_t Hkv = k.size(1), Smax = kcache.size(2);
   DNET_CHECK(D % 2 == 0 && D / 2 <= 1024, "head_dim");
@@ -125,6 +133,7 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   DNET_CHECK(v.stride(2) == 1 && v.stride(1) == D, "v inner contiguous");
   auto stream = current_stream();
   const bool q8 = kcache.dtype() == torch::kInt8;
+  const int* wpp = wpos.has_value() ? (const int*)wpos->data_ptr() : nullptr;
   if (q8) {
     DNET_CHECK(kscale.has_value() && vscale.has_value(), "q8 needs scales");
     DNET_CHECK(D == 64 || D == 128, "q8 kv needs D 64/128");
@@ -134,7 +143,7 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        (short*)k.data_ptr(), (const short*)v.data_ptr(),
                        kcache.data_ptr(), vcache.data_ptr(),
                        (short*)kscale->data_ptr(), (short*)vscale->data_ptr(),
-                       (const int*)pos.data_ptr(),
+                       (const int*)pos.data_ptr(), wpp,
                        (const float*)cos_table.data_ptr(),
                        (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
                        (int)Smax, (int)D, (int)q.stride(0), (int)k.stride(0),
@@ -145,7 +154,7 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        dim3((unsigned)(D / 2)), 0, stream, (short*)q.data_ptr(),
                        (short*)k.data_ptr(), (const short*)v.data_ptr(),
                        kcache.data_ptr(), vcache.data_ptr(), nullptr, nullptr,
-                       (const int*)pos.data_ptr(),
+                       (const int*)pos.data_ptr(), wpp,
                        (const float*)cos_table.data_ptr(),
                        (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
                        (int)Smax, (int)D, (int)q.stride(0), (int)k.stride(0),

@@ -263,15 +263,20 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
 
 
 def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
-                kscale=None, vscale=None):
-    """Decode-step fused rope+append reference (in-place on q, k and caches)."""
+                kscale=None, vscale=None, wpos=None):
+    """Decode-step fused rope+append reference (in-place on q, k and caches).
+    ``wpos`` (context parallelism): local cache write positions; rows with
+    wpos outside [0, Smax) are rotated but not stored."""
     B = q.shape[0]
     positions = pos.long()
     q.copy_(rope_apply(q, cos, sin, positions))
     k.copy_(rope_apply(k, cos, sin, positions))
     q8 = kcache.dtype == torch.int8
+    smax = kcache.shape[2]
     for b in range(B):
-        p = int(pos[b])
+        p = int(pos[b]) if wpos is None else int(wpos[b])
+        if p < 0 or p >= smax:
+            continue
         if q8:
             kc, ks = quantize_kv_rows(k[b])
             vc, vs = quantize_kv_rows(v[b])

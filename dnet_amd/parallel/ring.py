@@ -64,19 +64,23 @@ class RingExecutor:
                  seed: int = 0, decoding: Optional[DecodingConfig] = None,
                  use_graphs: Optional[bool] = None, init_weights: bool = True,
                  residency: int = 0, compress_ratio: float = 0.0,
-                 tp: int = 1, kv_bits: int = 16):
+                 tp: int = 1, cp: int = 1, kv_bits: int = 16):
         self.cfg = cfg
         self.rank = rank
         self.world = world
         self.device = torch.device(device)
-        # tensor parallelism inside each pipeline stage: ranks are laid out
-        # stage-major (rank = stage*tp + tp_rank); ring hops are pairwise
-        # rank -> rank+tp; all-reduce runs in the per-stage group.
-        assert world % tp == 0, "world must be a multiple of tp"
-        self.tp = tp
-        self.stages = world // tp
-        self.stage = rank // tp
-        self.tp_rank = rank % tp
+        # tensor OR context parallelism inside each pipeline stage: ranks
+        # are laid out stage-major (rank = stage*grp + grp_rank); ring hops
+        # are pairwise rank -> rank+grp; collectives run in the per-stage
+        # group (all-reduce for TP, partials all-gather for CP).
+        assert tp == 1 or cp == 1, "tp and cp are mutually exclusive (v1)"
+        grp = tp * cp
+        assert world % grp == 0, "world must be a multiple of tp*cp"
+        self.tp = grp               # stage group width (topology)
+        self.cp = cp
+        self.stages = world // grp
+        self.stage = rank // grp
+        self.tp_rank = rank % grp
         self.plan = plan or RingPlan.contiguous(cfg.num_layers, self.stages)
         self.rounds = self.plan.rounds
         # per-round layer windows for this stage (padded to k with empty
@@ -93,23 +97,30 @@ class RingExecutor:
         self.is_first = self.stage == 0
         self.is_last = self.stage == self.stages - 1
         self.ring = None
-        tp_group = None
+        stage_group = None
         if world > 1:
             self.ring = Ring(rank, world, self.device)
-            self.ring.next = (rank + tp) % world
-            self.ring.prev = (rank - tp) % world
-        if tp > 1:
+            self.ring.next = (rank + grp) % world
+            self.ring.prev = (rank - grp) % world
+        if grp > 1:
             for st in range(self.stages):  # collective: create every group
-                g = dist.new_group(list(range(st * tp, (st + 1) * tp)))
+                g = dist.new_group(list(range(st * grp, (st + 1) * grp)))
                 if st == self.stage:
-                    tp_group = g
+                    stage_group = g
         # token return path: last-stage rank pairs with its first-stage peer
-        self.token_src = (self.stages - 1) * tp + self.tp_rank
+        self.token_src = (self.stages - 1) * grp + self.tp_rank
         self.token_dst = self.tp_rank
         cls = get_ring_model(cfg.model_type)
-        self.model = cls(cfg, self.my_layers, self.device, self.is_first,
-                         self.is_last, smax=smax, tp_rank=self.tp_rank,
-                         tp_size=tp, tp_group=tp_group)
+        if cp > 1:
+            use_graphs = False  # the partials all-gather is not captured
+            self.model = cls(cfg, self.my_layers, self.device, self.is_first,
+                             self.is_last, smax=smax,
+                             cp_rank=self.tp_rank, cp_size=cp,
+                             cp_group=stage_group)
+        else:
+            self.model = cls(cfg, self.my_layers, self.device, self.is_first,
+                             self.is_last, smax=smax, tp_rank=self.tp_rank,
+                             tp_size=tp, tp_group=stage_group)
         self.model.kv_bits = kv_bits
         if init_weights:
             self.model.init_random(seed)
