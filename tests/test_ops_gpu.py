@@ -485,3 +485,29 @@ def test_attn_partials_combine_gpu():
     out = ops.attn_combine(torch.cat(parts, dim=2))
     assert torch.allclose(out.float().cpu(), full.float(),
                           atol=3e-2, rtol=3e-2)
+
+
+def test_rope_append_wpos():
+    """CP local-write mode: out-of-shard rows are rotated but not stored."""
+    torch.manual_seed(21)
+    B, Hq, Hkv, D, Smax = 3, 4, 2, 128, 32
+    cos, sin = ops.rope_tables(64, D, 10000.0)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=_dev())
+    k = torch.randn(B, Hkv, D, dtype=torch.bfloat16, device=_dev())
+    v = torch.randn(B, Hkv, D, dtype=torch.bfloat16, device=_dev())
+    kc = torch.zeros(B, Hkv, Smax, D, dtype=torch.bfloat16, device=_dev())
+    vc = torch.zeros_like(kc)
+    pos = torch.tensor([40, 33, 20], dtype=torch.int32, device=_dev())
+    wpos = (pos - 32).int()     # rank-1 shard of cap 32: 8, 1, -12
+    qr, kr = q.cpu().clone(), k.cpu().clone()
+    kcr, vcr = kc.cpu().clone(), vc.cpu().clone()
+    ref.rope_append(qr, kr, v.cpu().clone(), kcr, vcr, pos.cpu(), cos, sin,
+                    wpos=wpos.cpu())
+    ops.rope_append(q, k, v, kc, vc, pos, cos.to(_dev()), sin.to(_dev()),
+                    wpos=wpos)
+    assert torch.allclose(q.float().cpu(), qr.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(kc.float().cpu(), kcr.float(), atol=3e-2, rtol=3e-2)
+    # in-shard rows landed at their local index; out-of-shard wrote nothing
+    assert kc[0, :, 8].abs().sum() > 0      # 40-32 = 8
+    assert kc[1, :, 1].abs().sum() > 0      # 33-32 = 1
+    assert kc[2].abs().sum() == 0           # 20-32 < 0: other rank's row
