@@ -208,6 +208,21 @@ class KVCache:
         self.smax = smax
         self.batch = batch
 
+    def slot(self, s: int) -> "KVCache":
+        """A batch-1 VIEW of slot ``s`` (shared storage): the serving slot
+        scheduler prefills one sequence through this while other slots keep
+        their state (continuous batching)."""
+        import copy
+        c = copy.copy(self)
+        c.k = self.k[:, s:s + 1]
+        c.v = self.v[:, s:s + 1]
+        if getattr(self, "kscale", None) is not None:
+            c.kscale = self.kscale[:, s:s + 1]
+            c.vscale = self.vscale[:, s:s + 1]
+        c.pos = self.pos[s:s + 1]
+        c.batch = 1
+        return c
+
     @property
     def quantized(self) -> bool:
         return self.k.dtype == torch.int8

@@ -70,6 +70,7 @@ class ShardRuntime:
         self.ctrl_q: "queue.Queue[tuple]" = queue.Queue()
         self.infer_q: "queue.Queue[dict]" = queue.Queue()
         self._callback: Optional[SyncWireClient] = None
+        self.slots: Optional[list] = None   # continuous batching (world==1)
         self._stop = threading.Event()
         self.status = "idle"
         self.last_error = ""
@@ -165,6 +166,11 @@ class ShardRuntime:
                 ex.weight_cache = enable_offload(ex.model, residency)
                 ex.use_graphs = False
         self.executor = ex
+        # slot-based continuous batching: several single-stream requests
+        # share the decode batch (each owns one KV slot); enabled by
+        # loading with max_batch > 1 on a single-rank ring
+        self.slots = ([None] * req.max_batch
+                      if req.max_batch > 1 and req.world_size == 1 else None)
         self.load_req = req
         self.model_name = req.model_name or req.model_path
         if ex.is_last and req.api_callback_address:
@@ -201,6 +207,7 @@ class ShardRuntime:
     def _unload(self) -> None:
         import torch.distributed as dist
         self.executor = None
+        self.slots = None
         self.load_req = None
         self.model_name = ""
         if self._callback:
@@ -238,6 +245,9 @@ class ShardRuntime:
             if ex is None:
                 continue
             if ex.rank == 0:
+                if self.slots is not None:
+                    self._slots_tick()
+                    continue
                 try:
                     frame = self.infer_q.get(timeout=0.02)
                 except queue.Empty:
@@ -367,6 +377,86 @@ class ShardRuntime:
         log.info("[PROFILE][DECODE] nonce=%s tokens=%d ms=%.1f tok_s=%.1f "
                  "emit_ms=%.1f", nonce[:18], n, dt * 1e3, n / max(dt, 1e-9),
                  self._emit_s * 1e3)
+
+    # ---------- slot-based continuous batching (world == 1) ----------
+
+    def _slots_tick(self) -> None:
+        """One scheduler iteration: admit queued requests into free KV
+        slots (prefill them while other slots hold their state), then run
+        one decode step for every active slot. Reference has no equivalent
+        (requests serialize there); this keeps N single-stream requests at
+        ~single-stream latency each."""
+        active = any(s is not None for s in self.slots)
+        block = not active
+        while any(s is None for s in self.slots):
+            free = next(i for i, s in enumerate(self.slots) if s is None)
+            try:
+                frame = (self.infer_q.get(timeout=0.02) if block
+                         else self.infer_q.get_nowait())
+            except queue.Empty:
+                break
+            block = False
+            try:
+                self._slot_admit(free, frame)
+            except Exception:
+                log.exception("slot admit failed")
+                self._send_error(frame.get("nonce", ""))
+        if any(s is not None for s in self.slots):
+            self._slot_step()
+
+    def _slot_admit(self, si: int, frame: dict) -> None:
+        import numpy as np
+        ex = self.executor
+        tokens = torch.from_numpy(
+            np.frombuffer(frame["tokens"], dtype=np.int32).copy()).long()
+        T = int(frame.get("prompt_len", tokens.numel()))
+        p = frame.get("params", {})
+        ex.set_decoding(DecodingConfig(
+            temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
+            top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0)))
+        kv = ex.kvs[0].slot(si)
+        kv.pos.fill_(0)
+        h = torch.nn.functional.embedding(
+            tokens.view(1, T).to(ex.device), ex.model.embed)
+        ex.model.prefill_window(h, ex.my_layers, kv, 0)
+        kv.pos.fill_(T)
+        logits = ex.model.normalize_project(h[:, -1].contiguous())
+        tok, _, _ = ex.sampler.sample(logits.float())
+        t0 = int(tok[0])
+        ex.tokbuf[0][si] = t0
+        st = {"nonce": frame.get("nonce", ""), "produced": 1,
+              "max_tokens": int(frame.get("max_tokens", 128)),
+              "stop_ids": set(frame.get("stop_ids", []))}
+        done = t0 in st["stop_ids"] or st["max_tokens"] <= 1
+        self._emit_token(st["nonce"], t0, finished=done)
+        if not done:
+            self.slots[si] = st
+        log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
+                 st["nonce"][:18], T)
+
+    def _slot_step(self) -> None:
+        ex = self.executor
+        for r in range(ex.rounds):
+            ex._run_decode(0, r)
+        toks_t, _, _ = ex.sampler.sample(ex.logits_buf[0].float())
+        ex.tokbuf[0].copy_(toks_t)
+        ex.kvs[0].pos.add_(1)
+        # park idle slots at 0 so their dummy writes stay in range
+        idle = [i for i, st in enumerate(self.slots) if st is None]
+        if idle:
+            ex.kvs[0].pos[torch.tensor(idle, dtype=torch.long,
+                                       device=ex.device)] = 0
+        toks = toks_t.tolist()
+        for i, st in enumerate(self.slots):
+            if st is None:
+                continue
+            t = int(toks[i])
+            st["produced"] += 1
+            done = (t in st["stop_ids"]
+                    or st["produced"] >= st["max_tokens"])
+            self._emit_token(st["nonce"], t, finished=done)
+            if done:
+                self.slots[i] = None
 
     _emit_s = 0.0
 
