@@ -151,12 +151,17 @@ class ShardRuntime:
         synthetic = not (Path(req.model_path).expanduser() / "config.json").exists()
         residency = req.residency_size if \
             0 < req.residency_size < len(req.layers) else 0
+        slots_mode = req.max_batch > 1 and req.world_size == 1
         ex = RingExecutor(cfg, req.rank, req.world_size, device, plan=plan,
                           mb_count=1, mb_size=req.max_batch,
                           smax=req.max_seq,
+                          # slot scheduler interleaves per-slot eager
+                          # prefills with decode steps — run decode eager
+                          # too (graph replay after foreign allocations
+                          # proved unstable; eager is within ~2% here)
                           use_graphs=(device.type == "cuda"
                                       and self.settings.compute.use_graphs
-                                      and residency == 0),
+                                      and residency == 0 and not slots_mode),
                           init_weights=synthetic, residency=residency,
                           kv_bits=req.kv_bits)
         if not synthetic:
