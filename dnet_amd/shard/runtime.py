@@ -213,6 +213,7 @@ class ShardRuntime:
         import torch.distributed as dist
         self.executor = None
         self.slots = None
+        self._pending = None
         self.load_req = None
         self.model_name = ""
         if self._callback:
@@ -407,7 +408,16 @@ class ShardRuntime:
                 log.exception("slot admit failed")
                 self._send_error(frame.get("nonce", ""))
         if any(s is not None for s in self.slots):
-            self._slot_step()
+            # pipeline: launch step n+1 (device-side deps only), THEN emit
+            # step n's tokens — the host sync overlaps the next step's GPU
+            # work instead of serializing with it
+            launched = self._slot_step_launch()
+            if self._pending is not None:
+                self._slot_emit(*self._pending)
+            self._pending = launched
+        elif self._pending is not None:
+            self._slot_emit(*self._pending)
+            self._pending = None
 
     def _slot_admit(self, si: int, frame: dict) -> None:
         import numpy as np
@@ -439,7 +449,11 @@ class ShardRuntime:
         log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
                  st["nonce"][:18], T)
 
-    def _slot_step(self) -> None:
+    _pending = None   # (device tokens, active slot list) of the in-flight step
+
+    def _slot_step_launch(self):
+        """Enqueue one decode step for the whole batch — device ops only,
+        no host sync. Returns (tokens device tensor, active slots)."""
         ex = self.executor
         for r in range(ex.rounds):
             ex._run_decode(0, r)
@@ -451,9 +465,14 @@ class ShardRuntime:
         if idle:
             ex.kvs[0].pos[torch.tensor(idle, dtype=torch.long,
                                        device=ex.device)] = 0
-        toks = toks_t.tolist()
-        for i, st in enumerate(self.slots):
-            if st is None:
+        return toks_t, [i for i, st in enumerate(self.slots)
+                        if st is not None]
+
+    def _slot_emit(self, toks_t, active) -> None:
+        toks = toks_t.tolist()   # syncs; overlaps the already-launched step
+        for i in active:
+            st = self.slots[i]
+            if st is None:       # freed by an earlier emit (stop lag)
                 continue
             t = int(toks[i])
             st["produced"] += 1
