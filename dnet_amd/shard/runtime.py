@@ -155,10 +155,11 @@ class ShardRuntime:
         ex = RingExecutor(cfg, req.rank, req.world_size, device, plan=plan,
                           mb_count=1, mb_size=req.max_batch,
                           smax=req.max_seq,
-                          # slot scheduler interleaves per-slot eager
-                          # prefills with decode steps — run decode eager
-                          # too (graph replay after foreign allocations
-                          # proved unstable; eager is within ~2% here)
+                          # slot scheduler v1 decodes eager: per-slot
+                          # prefills interleave with decode steps and the
+                          # eager gap is small at serving batch sizes;
+                          # graph capture under slot churn is a round-2
+                          # validation item
                           use_graphs=(device.type == "cuda"
                                       and self.settings.compute.use_graphs
                                       and residency == 0 and not slots_mode),

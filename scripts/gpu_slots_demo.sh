@@ -18,13 +18,17 @@ done
 
 fire4() {
   local t0=$(date +%s.%N)
-  for i in 1 2 3 4; do
-    curl -s -m 90 -X POST http://127.0.0.1:18080/v1/chat/completions \
-      -H 'content-type: application/json' \
-      -d "{\"model\":\"gpt-oss-20b-synthetic\",\"max_tokens\":64,\"profile\":true,\"messages\":[{\"role\":\"user\",\"content\":\"stream $i\"}]}" \
-      | python3 -c 'import json,sys; d=json.load(sys.stdin); m=d.get("metrics",{}); print("  tokens:", d["usage"]["completion_tokens"], "tps:", round(m.get("tps_decoding",0),1), "total_ms:", round(m.get("total_ms",0),1))' &
-  done
-  wait
+  # subshell so `wait` covers only the curls (a bare wait in the main
+  # shell would also wait on the server daemons — forever)
+  (
+    for i in 1 2 3 4; do
+      curl -s -m 90 -X POST http://127.0.0.1:18080/v1/chat/completions \
+        -H 'content-type: application/json' \
+        -d "{\"model\":\"gpt-oss-20b-synthetic\",\"max_tokens\":64,\"profile\":true,\"messages\":[{\"role\":\"user\",\"content\":\"stream $i\"}]}" \
+        | python3 -c 'import json,sys; d=json.load(sys.stdin); m=d.get("metrics",{}); print("  tokens:", d["usage"]["completion_tokens"], "tps:", round(m.get("tps_decoding",0),1), "total_ms:", round(m.get("total_ms",0),1))' &
+    done
+    wait
+  )
   python3 -c "import time; print('  wall: %.2fs' % ($(date +%s.%N) - $t0))"
 }
 
