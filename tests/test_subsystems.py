@@ -377,3 +377,44 @@ def test_slot_admission_mid_decode():
     assert emitted["b"] == emitted2["b"]
     rt._unload()
     rt2._unload()
+
+
+def test_slot_stop_id_mid_stream():
+    """A slot hitting a stop id frees early while the other keeps going."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    emitted: dict[str, list] = {}
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(
+                (frame["token_id"], frame["finished"]))
+
+        def close(self):
+            pass
+
+    rt = ShardRuntime("probe")
+    rt._load(ShardLoadModelRequest(
+        model_path="tiny", model_name="tiny", total_layers=4,
+        layers=[0, 1, 2, 3], rank=0, world_size=1, max_batch=2, max_seq=64))
+    rt._callback = Cap()
+    prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    # run once unstopped to learn the greedy sequence, then use its 3rd
+    # token as a stop id
+    rt.infer_q.put({"nonce": "probe0", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 8, "stop_ids": [], "params": {}})
+    for _ in range(15):
+        rt._slots_tick()
+    seq = [t for t, _ in emitted["probe0"]]
+    stop = seq[2]
+    rt.infer_q.put({"nonce": "stopped", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 8, "stop_ids": [stop], "params": {}})
+    rt.infer_q.put({"nonce": "runs", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 8, "stop_ids": [], "params": {}})
+    for _ in range(20):
+        rt._slots_tick()
+    st = emitted["stopped"]
+    assert len(st) == 3 and st[-1][1]            # stopped at the stop id
+    assert len(emitted["runs"]) == 8             # unaffected neighbor
+    rt._unload()
