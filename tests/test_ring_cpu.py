@@ -453,3 +453,51 @@ def test_pp2_cp2_matches_single():
         p.join(timeout=60)
         assert p.exitcode == 0
     assert torch.equal(out, single), f"pp2xcp2 != single:\n{out}\n{single}"
+
+
+def _cp2_kv8_rank_main(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, rank, world, "cpu", mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False,
+                      cp=2, kv_bits=8)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_cp2_kv8_matches_single_kv8():
+    """CP + int8 KV cache: quantized shard writes and dequant-on-read
+    combine must equal the single-rank int8-KV run token-exactly."""
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=MB_COUNT, mb_size=MB_SIZE,
+                      smax=64, seed=7, use_graphs=False, kv_bits=8)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks)
+    gen = ex.decode_rounds(NGEN)
+    single = torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_cp2_kv8_rank_main, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert torch.equal(out, single), f"cp2-kv8 != single-kv8:\n{out}\n{single}"
