@@ -70,9 +70,78 @@ class MoERingModel(RingModel):
         return lw
 
     def load_state_dict(self, sd: dict):
-        raise NotImplementedError(
-            "MoE safetensors loading lands with the weight-cache milestone; "
-            "use init_random for synthetic runs")
+        """HF mixtral layout (`block_sparse_moe.gate` +
+        `experts.E.w1/w3/w2` = gate/up/down) and qwen-moe layout
+        (`mlp.gate` + `mlp.experts.E.{gate,up,down}_proj`); attention and
+        norms follow the llama-family naming (reference:
+        src/dnet/core/models/gpt_oss.py MoE weight mapping)."""
+        c = self.cfg
+
+        def get(name):
+            for pref in ("model.", ""):
+                if pref + name in sd:
+                    return sd[pref + name].to(torch.bfloat16)
+            return None
+
+        dev = self.device
+        for lid in self.layer_ids:
+            p = f"layers.{lid}."
+            qw, kw, vw = (get(p + f"self_attn.{x}_proj.weight") for x in "qkv")
+            qb = get(p + "self_attn.q_proj.bias")
+            bias = None
+            if qb is not None:
+                bias = torch.cat([qb, get(p + "self_attn.k_proj.bias"),
+                                  get(p + "self_attn.v_proj.bias")]).to(dev)
+            lw = LayerWeights(
+                attn_norm=get(p + "input_layernorm.weight").to(dev),
+                qkv=Linear.make(torch.cat([qw, kw, vw]).to(dev), bias,
+                                c.quant),
+                o=Linear.make(get(p + "self_attn.o_proj.weight").to(dev),
+                              None, c.quant),
+                mlp_norm=get(p + "post_attention_layernorm.weight").to(dev),
+            )
+            qn = get(p + "self_attn.q_norm.weight")
+            if qn is not None:
+                lw.q_norm = qn.to(dev)
+                lw.k_norm = get(p + "self_attn.k_norm.weight").to(dev)
+            router = get(p + "block_sparse_moe.gate.weight")
+            if router is None:
+                router = get(p + "mlp.gate.weight")
+            lw.router = Linear(router.to(dev))
+            lw.gateup = None
+            lw.down = None
+            lw.experts_gateup, lw.experts_down = [], []
+            gub = get(p + "mlp.experts.gate_up_proj")   # batched [E, 2I, H]
+            dnb = get(p + "mlp.experts.down_proj")      # batched [E, H, I]
+            for e in range(c.num_experts):
+                if gub is not None:     # batched layout (gate/up halves)
+                    gu_e, dn_e = gub[e].contiguous(), dnb[e].contiguous()
+                else:                   # per-expert tensors
+                    mx = f"{p}block_sparse_moe.experts.{e}."
+                    qw2 = f"{p}mlp.experts.{e}."
+                    gate = get(mx + "w1.weight")
+                    if gate is not None:               # legacy mixtral
+                        up = get(mx + "w3.weight")
+                        dn_e = get(mx + "w2.weight")
+                    else:                              # qwen-moe naming
+                        gate = get(qw2 + "gate_proj.weight")
+                        up = get(qw2 + "up_proj.weight")
+                        dn_e = get(qw2 + "down_proj.weight")
+                    gu_e = torch.cat([gate, up])
+                lw.experts_gateup.append(Linear.make(gu_e.to(dev), None,
+                                                     c.quant))
+                lw.experts_down.append(Linear.make(dn_e.to(dev), None,
+                                                   c.quant))
+            self.layers[lid] = lw
+        if self.is_first:
+            self.embed = get("embed_tokens.weight").to(dev)
+        if self.is_last:
+            self.final_norm = get("norm.weight").to(dev)
+            head = sd.get("lm_head.weight")
+            if head is None or c.tie_word_embeddings:
+                self.lm_head = Linear(get("embed_tokens.weight").to(dev))
+            else:
+                self.lm_head = Linear(head.to(torch.bfloat16).to(dev))
 
     # below this many rows, route through the grouped-expert kernels (no
     # data-dependent host syncs -> hipGraph-capturable; unrouted experts are

@@ -278,3 +278,29 @@ def test_mxfp4_dequant_and_gpt_oss_load():
                            m2.layers[0].experts_gateup[e].w)
         assert torch.equal(m1.layers[0].experts_down[e].w,
                            m2.layers[0].experts_down[e].w)
+
+
+def test_vs_transformers_mixtral():
+    transformers = pytest.importorskip("transformers")
+    torch.manual_seed(4)
+    tc = transformers.MixtralConfig(
+        hidden_size=128, intermediate_size=256, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, head_dim=32,
+        vocab_size=256, rope_theta=10000.0, rms_norm_eps=1e-5,
+        num_local_experts=4, num_experts_per_tok=2,
+        tie_word_embeddings=False, max_position_embeddings=128)
+    hf = transformers.MixtralForCausalLM(tc).eval().float()
+    cfg = ModelConfig.from_hf(tc.to_dict())
+    m = get_ring_model(cfg.model_type)(cfg, range(cfg.num_layers), "cpu",
+                                       True, True, smax=64)
+    m.load_state_dict({k: v for k, v in hf.state_dict().items()})
+    kv = KVCache(cfg, range(cfg.num_layers), 1, 64, "cpu")
+
+    tokens = torch.randint(0, 256, (1, 10))
+    with torch.no_grad():
+        ref_logits = hf(tokens).logits[:, -1].float()
+    h = m.embed_tokens(tokens).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    ours = m.normalize_project(h[:, -1].contiguous()).float()
+    cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
+    assert (cos > 0.99).all(), f"vs transformers mixtral: cos={cos}"
