@@ -438,17 +438,23 @@ class RingModel:
             qb, kb, vb = (get(p + f"self_attn.{x}_proj.bias") for x in "qkv")
             bias = None
             if qb is not None:
-                bias = torch.cat([qb, kb, vb]).to(self.device)
+                bias = self._slice_qkv(
+                    torch.cat([qb, kb, vb]).view(-1, 1)).view(-1).to(self.device)
             lw = LayerWeights(
                 attn_norm=get(p + "input_layernorm.weight").to(self.device),
-                qkv=Linear.make(torch.cat([qw, kw, vw]).to(self.device), bias, c.quant),
-                o=Linear.make(get(p + "self_attn.o_proj.weight").to(self.device), None, c.quant),
-                mlp_norm=get(p + "post_attention_layernorm.weight").to(self.device),
-                gateup=Linear.make(
-                    torch.cat([get(p + "mlp.gate_proj.weight"),
-                               get(p + "mlp.up_proj.weight")]).to(self.device),
+                qkv=Linear.make(self._slice_qkv(
+                    torch.cat([qw, kw, vw])).to(self.device), bias, c.quant),
+                o=Linear.make(self._slice_cols(
+                    get(p + "self_attn.o_proj.weight")).to(self.device),
                     None, c.quant),
-                down=Linear.make(get(p + "mlp.down_proj.weight").to(self.device), None, c.quant),
+                mlp_norm=get(p + "post_attention_layernorm.weight").to(self.device),
+                gateup=Linear.make(self._slice_gateup(
+                    torch.cat([get(p + "mlp.gate_proj.weight"),
+                               get(p + "mlp.up_proj.weight")])).to(self.device),
+                    None, c.quant),
+                down=Linear.make(self._slice_cols(
+                    get(p + "mlp.down_proj.weight")).to(self.device),
+                    None, c.quant),
             )
             qn = get(p + "self_attn.q_norm.weight")
             if qn is not None:
