@@ -45,11 +45,18 @@ class GptOssRingModel(MoERingModel):
         lw = super()._init_layer(rand, lid)
         c = self.cfg
         dev = self.device
-        # gpt-oss: biases everywhere + learned sink logit per q head
-        lw.qkv.bias = rand(c.qkv_out).to(dev)
-        lw.o.bias = rand(c.hidden_size).to(dev)
+        # gpt-oss: biases everywhere + learned sink logit per q head.
+        # TP: qkv bias and sinks slice with the heads; the o bias lives on
+        # rank 0 only (the stage all-reduce sums partials — a copy per
+        # rank would count it tp times)
+        lw.qkv.bias = self._slice_qkv(
+            rand(c.qkv_out).view(-1, 1)).view(-1).to(dev)
+        ob = rand(c.hidden_size).to(dev)
+        lw.o.bias = ob if self.tp_rank == 0 else None
         lw.router.bias = rand(c.num_experts).to(dev)
-        lw.sinks = rand(c.num_q_heads).to(dev)
+        nh = c.num_q_heads // self.tp_size
+        lw.sinks = rand(c.num_q_heads)[self.tp_rank * nh:
+                                       (self.tp_rank + 1) * nh].to(dev)
         lw.expert_biases_gu = [rand(2 * (c.moe_intermediate_size
                                          or c.intermediate_size)).to(dev)
                                for _ in range(c.num_experts)]
@@ -90,13 +97,18 @@ class GptOssRingModel(MoERingModel):
             p = f"layers.{lid}."
             qw, kw, vw = (get(p + f"self_attn.{x}_proj.weight") for x in "qkv")
             qb, kb, vb = (get(p + f"self_attn.{x}_proj.bias") for x in "qkv")
-            bias = torch.cat([qb, kb, vb]).to(self.device) if qb is not None else None
+            bias = (self._slice_qkv(torch.cat([qb, kb, vb]).view(-1, 1))
+                    .view(-1).to(self.device) if qb is not None else None)
+            ob = get(p + "self_attn.o_proj.bias")
+            if ob is not None and self.tp_rank != 0:
+                ob = None          # bias counted once across the stage sum
             lw = LayerWeights(
                 attn_norm=get(p + "input_layernorm.weight").to(self.device),
-                qkv=Linear.make(torch.cat([qw, kw, vw]).to(self.device), bias,
-                                c.quant),
-                o=Linear.make(get(p + "self_attn.o_proj.weight").to(self.device),
-                              get(p + "self_attn.o_proj.bias"), c.quant),
+                qkv=Linear.make(self._slice_qkv(
+                    torch.cat([qw, kw, vw])).to(self.device), bias, c.quant),
+                o=Linear.make(self._slice_cols(
+                    get(p + "self_attn.o_proj.weight")).to(self.device),
+                    ob, c.quant),
                 mlp_norm=get(p + "post_attention_layernorm.weight").to(self.device),
             )
             lw.router = Linear(get(p + "mlp.router.weight").to(self.device),
@@ -124,7 +136,10 @@ class GptOssRingModel(MoERingModel):
                     (dn[e] if mx else dn[e].t()).contiguous().to(self.device),
                     dnb[e].to(self.device) if dnb is not None else None,
                     c.quant))
-            lw.sinks = get(p + "self_attn.sinks").to(self.device)
+            sk = get(p + "self_attn.sinks")
+            nh = c.num_q_heads // self.tp_size
+            lw.sinks = sk[self.tp_rank * nh:
+                          (self.tp_rank + 1) * nh].to(self.device)
             self.layers[lid] = lw
         if self.is_first:
             self.embed = get("embed_tokens.weight").to(self.device)
