@@ -8,12 +8,9 @@ from typing import Optional
 
 import httpx
 
-from ..core.types import LayerAssignment, TopologyInfo
+from ..core.types import TopologyInfo
 from ..models import ModelConfig, PRESETS, QuantConfig
 from ..parallel.profiler import DeviceProfile
-from ..parallel.solver import (compute_layer_assignments, halda_solve,
-                               optimize_device_ordering,
-                               postprocess_single_round)
 from ..utils.hostfile import DeviceProperties
 from ..utils.logger import get_logger
 
@@ -35,7 +32,9 @@ def estimate_layer_bytes(cfg: ModelConfig) -> int:
 
 
 class ClusterManager:
-    def __init__(self, discovery, solver_settings=None):
+    def __init__(self, discovery, solver_settings=None, strategy=None):
+        from .strategies import ring_strategy
+        self.strategy = strategy or ring_strategy()
         self.discovery = discovery
         self.devices: dict[str, DeviceProperties] = {}
         self.profiles: dict[str, DeviceProfile] = {}
@@ -110,48 +109,18 @@ class ClusterManager:
                        batch: int = 1, seq_len: int = 4096) -> TopologyInfo:
         shards = self.shard_devices()
         assert shards, "no shard devices discovered"
-        if self.link_ms:
-            by_name = {d.instance: d for d in shards}
-            order = optimize_device_ordering(
-                [d.instance for d in shards], self.link_ms)
-            shards = [by_name[i] for i in order]
-        profs = []
+        profiles = {}
         for d in shards:
             p = self.profiles.get(d.instance)
             if p is None:
                 p = DeviceProfile(instance=d.instance, hbm_gbps=5000.0,
                                   h2d_gbps=50.0, hbm_free_gb=280.0)
-            profs.append(p)
+            profiles[d.instance] = p
         lb = estimate_layer_bytes(cfg)
         kv_per_layer = (2 * batch * cfg.num_kv_heads * seq_len * cfg.head_dim
                         * (kv_bits / 8))
-        res = halda_solve(profs, cfg.num_layers, lb, kv_per_layer,
-                          kv_bits=kv_bits)
-        w = res.w
-        if res.k == 1:
-            w = postprocess_single_round(w, profs)
-        # drop devices with zero layers, keep order
-        active = [i for i, x in enumerate(w) if x > 0]
-        w_active = [w[i] for i in active]
-        assigns_lists = compute_layer_assignments(w_active, res.k,
-                                                  cfg.num_layers)
-        assignments = []
-        for j, i in enumerate(active):
-            d = shards[i]
-            nxt = shards[active[(j + 1) % len(active)]].instance
-            assignments.append(LayerAssignment(
-                instance=d.instance, layers=assigns_lists[j],
-                next_instance=nxt,
-                window_size=min(4, max(w_active[j], 1)),
-                residency_size=res.n[i], gpu_index=max(d.gpu_index, 0)))
-        head = assignments[0].instance if assignments else ""
-        head_dev = next((d for d in shards if d.instance == head), shards[0])
-        topo = TopologyInfo(
-            model=model_id, kv_bits=kv_bits, num_layers=cfg.num_layers,
-            devices=[a.instance for a in assignments],
-            assignments=assignments,
-            solution={"w": res.w, "n": res.n, "k": res.k,
-                      "obj_value_ms": res.obj_value, "sets": res.sets},
-            master_addr=head_dev.local_ip, master_port=master_port)
+        topo = self.strategy.solver.solve(
+            model_id, cfg, shards, profiles, self.link_ms, lb, kv_per_layer,
+            master_port, kv_bits)
         self.topology = topo
         return topo

@@ -1,0 +1,97 @@
+"""Strategy layer: solver + transport bundle per parallelism strategy.
+
+Reference counterpart: src/dnet/api/strategies/ (`Strategy` = {solver,
+adapter} bundle, strategies/base.py:43-54; `RingTopologySolver` +
+`RingApiAdapter`, strategies/ring.py). Here the ring strategy is the
+shipped one (like the reference in practice); the ABCs keep the seam so a
+future strategy (e.g. context-parallel-first placement) plugs in without
+touching ClusterManager / InferenceManager.
+"""
+from __future__ import annotations
+
+from abc import ABC, abstractmethod
+from typing import Optional
+
+from ..core.types import LayerAssignment, TopologyInfo
+from ..parallel.solver import (compute_layer_assignments, halda_solve,
+                               optimize_device_ordering,
+                               postprocess_single_round)
+
+
+class TopologySolver(ABC):
+    """Turns device profiles + link measurements into a TopologyInfo
+    (reference: src/dnet/core/topology.py TopologySolver ABC)."""
+
+    @abstractmethod
+    def solve(self, model_id: str, cfg, shards, profiles: dict,
+              link_ms: dict, layer_bytes: int, kv_bytes_per_layer: float,
+              master_port: int, kv_bits: int) -> TopologyInfo:
+        ...
+
+
+class RingTopologySolver(TopologySolver):
+    """Pipelined-ring placement: link-aware device ordering -> HALDA-style
+    w/n/k layer distribution -> k-round assignments with ring closure
+    (reference: src/dnet/api/strategies/ring.py RingTopologySolver)."""
+
+    def solve(self, model_id, cfg, shards, profiles, link_ms, layer_bytes,
+              kv_bytes_per_layer, master_port, kv_bits) -> TopologyInfo:
+        if link_ms:
+            by_name = {d.instance: d for d in shards}
+            order = optimize_device_ordering(
+                [d.instance for d in shards], link_ms)
+            shards = [by_name[i] for i in order]
+        profs = [profiles[d.instance] for d in shards]
+        res = halda_solve(profs, cfg.num_layers, layer_bytes,
+                          kv_bytes_per_layer, kv_bits=kv_bits)
+        w = res.w
+        if res.k == 1:
+            w = postprocess_single_round(w, profs)
+        active = [i for i, x in enumerate(w) if x > 0]
+        w_active = [w[i] for i in active]
+        assigns_lists = compute_layer_assignments(w_active, res.k,
+                                                  cfg.num_layers)
+        assignments = []
+        for j, i in enumerate(active):
+            d = shards[i]
+            nxt = shards[active[(j + 1) % len(active)]].instance
+            assignments.append(LayerAssignment(
+                instance=d.instance, layers=assigns_lists[j],
+                next_instance=nxt,
+                window_size=min(4, max(w_active[j], 1)),
+                residency_size=res.n[i], gpu_index=max(d.gpu_index, 0)))
+        head = assignments[0].instance if assignments else ""
+        head_dev = next((d for d in shards if d.instance == head), shards[0])
+        return TopologyInfo(
+            model=model_id, kv_bits=kv_bits, num_layers=cfg.num_layers,
+            devices=[a.instance for a in assignments],
+            assignments=assignments,
+            solution={"w": res.w, "n": res.n, "k": res.k,
+                      "obj_value_ms": res.obj_value, "sets": res.sets},
+            master_addr=head_dev.local_ip, master_port=master_port)
+
+
+class ApiAdapterBase(ABC):
+    """API-side transport to the head shard (reference:
+    strategies/base.py ApiAdapterBase): connect, send an inference
+    request, resolve per-token frames. The shipped implementation is
+    InferenceManager (api/inference.py) over the msgpack/TCP wire."""
+
+    @abstractmethod
+    def connect_head(self, host: str, port: int, callback_addr: str): ...
+
+    @abstractmethod
+    def resolve_token(self, frame: dict): ...
+
+
+class Strategy:
+    """{solver, adapter} bundle (reference: strategies/base.py Strategy)."""
+
+    def __init__(self, solver: TopologySolver,
+                 adapter: Optional[ApiAdapterBase] = None):
+        self.solver = solver
+        self.adapter = adapter
+
+
+def ring_strategy(adapter=None) -> Strategy:
+    return Strategy(RingTopologySolver(), adapter)
