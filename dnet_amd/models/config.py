@@ -52,6 +52,7 @@ class ModelConfig:
     v_head_dim: int = 0
     first_k_dense_replace: int = 0
     n_shared_experts: int = 0
+    shared_expert_intermediate_size: int = 0   # qwen2-moe gated shared expert
     routed_scaling_factor: float = 1.0
     norm_topk_prob: bool = False
     quant: Optional[QuantConfig] = None
@@ -108,6 +109,9 @@ class ModelConfig:
             mc.num_experts_per_tok = cfg.get("num_experts_per_tok", 2)
             mc.moe_intermediate_size = cfg.get("moe_intermediate_size",
                                                cfg.get("intermediate_size"))
+            mc.norm_topk_prob = cfg.get("norm_topk_prob", False)
+            mc.shared_expert_intermediate_size = cfg.get(
+                "shared_expert_intermediate_size", 0)
         return mc
 
 

@@ -67,6 +67,13 @@ class MoERingModel(RingModel):
         lw.experts_down = [
             Linear.make(rand(c.hidden_size, inter).to(dev), None, c.quant)
             for _ in range(c.num_experts)]
+        if c.shared_expert_intermediate_size:       # qwen2-moe
+            si = c.shared_expert_intermediate_size
+            lw.shared_gateup = Linear.make(self._slice_gateup(
+                rand(2 * si, c.hidden_size)).to(dev), None, c.quant)
+            lw.shared_down = Linear.make(self._slice_cols(
+                rand(c.hidden_size, si)).to(dev), None, c.quant)
+            lw.shared_gate = Linear(rand(1, c.hidden_size).to(dev))
         return lw
 
     def load_state_dict(self, sd: dict):
@@ -132,6 +139,16 @@ class MoERingModel(RingModel):
                                                      c.quant))
                 lw.experts_down.append(Linear.make(dn_e.to(dev), None,
                                                    c.quant))
+            sg = get(p + "mlp.shared_expert.gate_proj.weight")
+            if sg is not None:                      # qwen2-moe shared expert
+                lw.shared_gateup = Linear.make(self._slice_gateup(
+                    torch.cat([sg, get(p + "mlp.shared_expert.up_proj.weight")]
+                              )).to(dev), None, c.quant)
+                lw.shared_down = Linear.make(self._slice_cols(
+                    get(p + "mlp.shared_expert.down_proj.weight")).to(dev),
+                    None, c.quant)
+                lw.shared_gate = Linear(
+                    get(p + "mlp.shared_expert_gate.weight").to(dev))
             self.layers[lid] = lw
         if self.is_first:
             self.embed = get("embed_tokens.weight").to(dev)
@@ -160,12 +177,33 @@ class MoERingModel(RingModel):
                  if self.tp_size <= 1 or e % self.tp_size == self.tp_rank]
         return stack_experts(lw, local)
 
+    def _route(self, logits: torch.Tensor):
+        """(weights, idx). mixtral: softmax over the top-k logits;
+        qwen-moe: softmax over ALL experts -> top-k (-> renorm when
+        norm_topk_prob)."""
+        c = self.cfg
+        if c.model_type in ("qwen2_moe", "qwen3_moe"):
+            probs = torch.softmax(logits, dim=-1)
+            weights, idx = torch.topk(probs, c.num_experts_per_tok, dim=-1)
+            if c.norm_topk_prob:
+                weights = weights / weights.sum(-1, keepdim=True)
+            return weights, idx
+        weights, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
+        return torch.softmax(weights, dim=-1), idx
+
+    def _shared_expert(self, y: torch.Tensor, lw: LayerWeights):
+        """qwen2-moe always-on shared expert, sigmoid-gated; returns a
+        PARTIAL under TP (rows/cols sliced; the caller's all-reduce sums).
+        """
+        sh = lw.shared_down(ops.swiglu(lw.shared_gateup(y)))
+        gate = torch.sigmoid(lw.shared_gate(y).float())
+        return sh.float() * gate
+
     def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
         c = self.cfg
         T = y.shape[0]
         logits = lw.router(y).float()
-        weights, idx = torch.topk(logits, c.num_experts_per_tok, dim=-1)
-        weights = torch.softmax(weights, dim=-1)
+        weights, idx = self._route(logits)
         if (T <= self.DENSE_MOE_MAX_T
                 and lw.experts_gateup[0].bits in (8, 16)):
             st = self._expert_stack(lw)
@@ -179,6 +217,8 @@ class MoERingModel(RingModel):
                                  self.GLU_ALPHA, self.GLU_LIMIT)
             out = ops.moe_down(act, st["dw"], st["ds"], st["db"], we,
                                st["group"], st["packed"])
+            if getattr(lw, "shared_gateup", None) is not None:
+                out = out + self._shared_expert(y, lw)
             return out.to(y.dtype)
         out = torch.zeros_like(y, dtype=torch.float32)
         for e in range(c.num_experts):
@@ -192,4 +232,6 @@ class MoERingModel(RingModel):
             xe = y[rows].contiguous()
             d = lw.experts_down[e](self._act(lw.experts_gateup[e](xe)))
             out[rows] += d.float() * we_full[rows].unsqueeze(-1)
+        if getattr(lw, "shared_gateup", None) is not None:
+            out = out + self._shared_expert(y, lw)
         return out.to(y.dtype)

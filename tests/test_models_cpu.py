@@ -304,3 +304,33 @@ def test_vs_transformers_mixtral():
     ours = m.normalize_project(h[:, -1].contiguous()).float()
     cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
     assert (cos > 0.99).all(), f"vs transformers mixtral: cos={cos}"
+
+
+def test_vs_transformers_qwen2_moe():
+    """Qwen2-MoE: softmax-before-topk routing + sigmoid-gated shared
+    expert (differs from mixtral on both counts)."""
+    transformers = pytest.importorskip("transformers")
+    torch.manual_seed(6)
+    tc = transformers.Qwen2MoeConfig(
+        hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, vocab_size=256,
+        rope_theta=10000.0, rms_norm_eps=1e-5, num_experts=4,
+        num_experts_per_tok=2, moe_intermediate_size=48,
+        shared_expert_intermediate_size=96, norm_topk_prob=True,
+        decoder_sparse_step=1, tie_word_embeddings=False,
+        max_position_embeddings=128)
+    hf = transformers.Qwen2MoeForCausalLM(tc).eval().float()
+    cfg = ModelConfig.from_hf(tc.to_dict())
+    m = get_ring_model(cfg.model_type)(cfg, range(cfg.num_layers), "cpu",
+                                       True, True, smax=64)
+    m.load_state_dict({k: v for k, v in hf.state_dict().items()})
+    kv = KVCache(cfg, range(cfg.num_layers), 1, 64, "cpu")
+
+    tokens = torch.randint(0, 256, (1, 9))
+    with torch.no_grad():
+        ref_logits = hf(tokens).logits[:, -1].float()
+    h = m.embed_tokens(tokens).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    ours = m.normalize_project(h[:, -1].contiguous()).float()
+    cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
+    assert (cos > 0.99).all(), f"vs transformers qwen2-moe: cos={cos}"
