@@ -14,3 +14,22 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip)
+
+
+def retry_flaky(times=2):
+    """Spawn + gloo rendezvous can transiently fail under machine load;
+    retry once. A genuine regression still fails ``times`` times."""
+    import functools
+
+    def deco(f):
+        @functools.wraps(f)
+        def wrapped(*a, **k):
+            last = None
+            for i in range(times):
+                try:
+                    return f(*a, **k)
+                except Exception as e:  # noqa: BLE001
+                    last = e
+            raise last
+        return wrapped
+    return deco

@@ -17,6 +17,7 @@ def _free_port():
     s.close()
     return p
 
+from conftest import retry_flaky
 from dnet_amd.models import ModelConfig, PRESETS
 from dnet_amd.parallel.ring import RingExecutor, RingPlan, split_layers
 
@@ -68,7 +69,8 @@ def _rank_main(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(360)
+@retry_flaky()
 def test_two_rank_ring_matches_single():
     single = _run_single()
     ctx = mp.get_context("spawn")
@@ -85,7 +87,8 @@ def test_two_rank_ring_matches_single():
     assert torch.equal(out, single), f"ring-2 != single:\n{out}\n{single}"
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(360)
+@retry_flaky()
 def test_two_rank_ring_compressed_hops():
     """Ring with column-sparsified activation hops still generates sane
     tokens (lossy, so no exact-match; shapes/flow must hold)."""
@@ -123,7 +126,8 @@ def _rank_main_compressed(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(360)
+@retry_flaky()
 def test_tp2_matches_single():
     """One stage with TP=2 (sharded heads/MLP + gloo all-reduce) must
     generate exactly the single-rank tokens (same deterministic weights)."""
@@ -162,7 +166,8 @@ def _rank_main_tp(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_pp2_tp2_matches_single():
     """4 ranks = 2 pipeline stages x TP 2 must equal the single-rank run."""
     single = _run_single()
@@ -200,7 +205,8 @@ def _rank_main_pp_tp(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(360)
+@retry_flaky()
 def test_two_rank_two_rounds_matches_single():
     """k=2 ring rounds (interleaved layer windows, two laps per token) must
     equal the single-rank run (prima.cpp-style k-round pipelining)."""
@@ -310,7 +316,8 @@ def _ds_rank_main_tp(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_deepseek_tp2_matches_single():
     """DeepSeek MLA with TP=2 (per-head q/kv_b/o sharding + EP experts +
     sliced shared experts) == single-rank, token-exact."""
@@ -350,7 +357,8 @@ def _cp_rank_main(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_context_parallel_attn_matches_full():
     """KV sequence sharded across 2 gloo ranks: gathered flash-decode
     partials combine to the full-attention result."""
@@ -397,7 +405,8 @@ def _cp2_rank_main(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_cp2_matches_single():
     """One stage with context parallelism (KV sequence sharded across 2
     ranks, partials-combined attention) must generate the single-rank
@@ -437,7 +446,8 @@ def _ppcp_rank_main(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_pp2_cp2_matches_single():
     """4 ranks = 2 pipeline stages x CP 2 == the single-rank run."""
     single = _run_single()
@@ -475,7 +485,8 @@ def _cp2_kv8_rank_main(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_cp2_kv8_matches_single_kv8():
     """CP + int8 KV cache: quantized shard writes and dequant-on-read
     combine must equal the single-rank int8-KV run token-exactly."""
@@ -549,7 +560,8 @@ def _ckpt_tp_rank_main(rank, world, port, q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_tp2_checkpoint_load_matches_single():
     """Loading the SAME HF-style checkpoint under TP=2 (loader slices
     heads/MLP per rank) must reproduce the single-rank tokens."""
@@ -610,7 +622,8 @@ def _gptoss_run(rank, world, port, q, tp):
     return out
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(480)
+@retry_flaky()
 def test_gpt_oss_tp2_matches_single():
     """gpt-oss under TP=2: sliced qkv bias + per-head sinks + rank-0-only
     o bias + EP experts must reproduce the single-rank tokens."""
