@@ -53,6 +53,9 @@ class ModelConfig:
     first_k_dense_replace: int = 0
     n_shared_experts: int = 0
     shared_expert_intermediate_size: int = 0   # qwen2-moe gated shared expert
+    scoring_func: str = "softmax"              # deepseek-v3: "sigmoid"
+    n_group: int = 1                           # deepseek-v3 group-limited
+    topk_group: int = 1                        #   routing (noaux_tc)
     routed_scaling_factor: float = 1.0
     norm_topk_prob: bool = False
     quant: Optional[QuantConfig] = None
@@ -103,6 +106,9 @@ class ModelConfig:
             mc.num_experts = cfg.get("n_routed_experts") or 0
             mc.num_experts_per_tok = cfg.get("num_experts_per_tok", 0)
             mc.moe_intermediate_size = cfg.get("moe_intermediate_size", 0)
+            mc.scoring_func = cfg.get("scoring_func", "softmax")
+            mc.n_group = cfg.get("n_group") or 1
+            mc.topk_group = cfg.get("topk_group") or 1
         if mt in ("mixtral", "qwen2_moe", "qwen3_moe", "gpt_oss"):
             mc.num_experts = cfg.get("num_local_experts",
                                      cfg.get("num_experts", 8))

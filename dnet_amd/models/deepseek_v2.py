@@ -115,6 +115,8 @@ class DeepseekV2RingModel(RingModel):
         inter = c.moe_intermediate_size or c.intermediate_size
         if c.num_experts and lid >= c.first_k_dense_replace:
             lw.router = Linear(rand(c.num_experts, c.hidden_size).to(dev))
+            if c.scoring_func == "sigmoid":
+                lw.router_bias = (rand(c.num_experts) / 10).to(dev)
             lw.experts_gateup = [
                 Linear.make(rand(2 * inter, c.hidden_size).to(dev), None, c.quant)
                 for _ in range(c.num_experts)]
@@ -182,6 +184,9 @@ class DeepseekV2RingModel(RingModel):
                     get(p + "mlp.down_proj.weight")).to(dev), None, c.quant)
             else:                                               # MoE layer
                 lw.router = Linear(get(p + "mlp.gate.weight").to(dev))
+                eb = get(p + "mlp.gate.e_score_correction_bias")
+                if eb is not None:
+                    lw.router_bias = eb.to(dev)
                 gu = get(p + "mlp.experts.gate_up_proj")    # [E, 2I, H]
                 dn = get(p + "mlp.experts.down_proj")       # [E, H, I]
                 lw.experts_gateup, lw.experts_down = [], []
@@ -291,11 +296,32 @@ class DeepseekV2RingModel(RingModel):
         c = self.cfg
         if lw.gateup is not None:
             return self._mlp(y, lw)
-        # MoE: softmax scoring -> top-k (greedy) * routed_scaling + shared
-        scores = torch.softmax(lw.router(y).float(), dim=-1)
-        weights, idx = torch.topk(scores, c.num_experts_per_tok, dim=-1)
-        if c.norm_topk_prob:
-            weights = weights / weights.sum(-1, keepdim=True)
+        logits = lw.router(y).float()
+        if c.scoring_func == "sigmoid":
+            # deepseek-v3 noaux_tc: sigmoid scores + learned correction
+            # bias for CHOICE only, group-limited top-k; returned weights
+            # come from the uncorrected scores
+            T0 = logits.shape[0]
+            scores = torch.sigmoid(logits)
+            choice = scores + lw.router_bias.float()
+            eg = c.num_experts // c.n_group
+            gs = choice.view(T0, c.n_group, eg).topk(
+                min(2, eg), dim=-1)[0].sum(-1)
+            gidx = gs.topk(c.topk_group, dim=-1)[1]
+            gmask = torch.zeros_like(gs).scatter_(1, gidx, 1.0)
+            choice = choice.masked_fill(
+                ~gmask.unsqueeze(-1).expand(T0, c.n_group, eg)
+                .reshape(T0, c.num_experts).bool(), float("-inf"))
+            idx = choice.topk(c.num_experts_per_tok, dim=-1)[1]
+            weights = scores.gather(1, idx)
+            if c.norm_topk_prob:
+                weights = weights / (weights.sum(-1, keepdim=True) + 1e-20)
+        else:
+            # deepseek-v2: softmax scoring -> greedy top-k
+            scores = torch.softmax(logits, dim=-1)
+            weights, idx = torch.topk(scores, c.num_experts_per_tok, dim=-1)
+            if c.norm_topk_prob:
+                weights = weights / weights.sum(-1, keepdim=True)
         T = y.shape[0]
         if T <= 64 and lw.experts_gateup[0].bits in (8, 16):
             # grouped-expert kernels (graph-safe; unrouted experts skipped

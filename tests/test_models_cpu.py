@@ -334,3 +334,67 @@ def test_vs_transformers_qwen2_moe():
     ours = m.normalize_project(h[:, -1].contiguous()).float()
     cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
     assert (cos > 0.99).all(), f"vs transformers qwen2-moe: cos={cos}"
+
+
+def test_vs_transformers_deepseek_v3():
+    """DeepSeek-V3: sigmoid scoring + e_score_correction_bias +
+    group-limited top-k routing (noaux_tc) — differs from v2's softmax."""
+    transformers = pytest.importorskip("transformers")
+    if not hasattr(transformers, "DeepseekV3ForCausalLM"):
+        pytest.skip("no deepseek_v3 in transformers")
+    torch.manual_seed(8)
+    tc = transformers.DeepseekV3Config(
+        hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=4, vocab_size=256,
+        q_lora_rank=48, kv_lora_rank=32, qk_nope_head_dim=16,
+        qk_rope_head_dim=8, v_head_dim=16, n_routed_experts=8,
+        num_experts_per_tok=2, n_shared_experts=1, n_group=2, topk_group=1,
+        moe_intermediate_size=32, first_k_dense_replace=1,
+        routed_scaling_factor=1.5, norm_topk_prob=True,
+        rope_interleave=False,
+        scoring_func="sigmoid", rope_theta=10000.0, rms_norm_eps=1e-5,
+        tie_word_embeddings=False, max_position_embeddings=128)
+    hf = transformers.DeepseekV3ForCausalLM(tc).eval().float()
+    cfg = ModelConfig.from_hf(tc.to_dict())
+    m = get_ring_model(cfg.model_type)(cfg, range(cfg.num_layers), "cpu",
+                                       True, True, smax=64)
+    m.load_state_dict({k: v for k, v in hf.state_dict().items()})
+    kv = m.make_kv_cache(1, 64)
+
+    tokens = torch.randint(0, 256, (1, 8))
+    with torch.no_grad():
+        ref_logits = hf(tokens).logits[:, -1].float()
+    h = m.embed_tokens(tokens).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    ours = m.normalize_project(h[:, -1].contiguous()).float()
+    cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
+    assert (cos > 0.99).all(), f"vs transformers deepseek-v3: cos={cos}"
+
+
+def test_vs_transformers_qwen3_moe():
+    transformers = pytest.importorskip("transformers")
+    if not hasattr(transformers, "Qwen3MoeForCausalLM"):
+        pytest.skip("no qwen3_moe in transformers")
+    torch.manual_seed(9)
+    tc = transformers.Qwen3MoeConfig(
+        hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, head_dim=16,
+        vocab_size=256, rope_theta=10000.0, rms_norm_eps=1e-5,
+        num_experts=4, num_experts_per_tok=2, moe_intermediate_size=48,
+        norm_topk_prob=True, decoder_sparse_step=1,
+        tie_word_embeddings=False, max_position_embeddings=128)
+    hf = transformers.Qwen3MoeForCausalLM(tc).eval().float()
+    cfg = ModelConfig.from_hf(tc.to_dict())
+    m = get_ring_model(cfg.model_type)(cfg, range(cfg.num_layers), "cpu",
+                                       True, True, smax=64)
+    m.load_state_dict({k: v for k, v in hf.state_dict().items()})
+    kv = KVCache(cfg, range(cfg.num_layers), 1, 64, "cpu")
+
+    tokens = torch.randint(0, 256, (1, 8))
+    with torch.no_grad():
+        ref_logits = hf(tokens).logits[:, -1].float()
+    h = m.embed_tokens(tokens).clone()
+    m.prefill_window(h, m.layer_ids, kv, 0)
+    ours = m.normalize_project(h[:, -1].contiguous()).float()
+    cos = torch.nn.functional.cosine_similarity(ours, ref_logits, dim=-1)
+    assert (cos > 0.99).all(), f"vs transformers qwen3-moe: cos={cos}"
