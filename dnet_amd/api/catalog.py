@@ -36,6 +36,8 @@ model_catalog: list[CatalogEntry] = [
     CatalogEntry("gpt-oss-20b-synthetic", preset="gpt-oss-20b", tokenizer="byte"),
     CatalogEntry("deepseek-v2-lite-synthetic", preset="deepseek-v2-lite",
                  tokenizer="byte"),
+    CatalogEntry("deepseek-v3-lite-synthetic", preset="deepseek-v3-lite",
+                 tokenizer="byte"),
     CatalogEntry("openai/gpt-oss-20b", repo="openai/gpt-oss-20b"),
     CatalogEntry("deepseek-ai/DeepSeek-V2-Lite", repo="deepseek-ai/DeepSeek-V2-Lite"),
     # real checkpoints (local safetensors dir or pre-downloaded HF cache)

@@ -166,6 +166,18 @@ PRESETS: dict[str, dict] = {
                              moe_intermediate_size=1408,
                              first_k_dense_replace=1,
                              routed_scaling_factor=1.0, rope_theta=10000.0),
+    "deepseek-v3-lite": dict(model_type="deepseek_v3", hidden_size=2048,
+                             num_hidden_layers=27, num_attention_heads=16,
+                             num_key_value_heads=16, vocab_size=102400,
+                             intermediate_size=10944, q_lora_rank=0,
+                             kv_lora_rank=512, qk_nope_head_dim=128,
+                             qk_rope_head_dim=64, v_head_dim=128,
+                             n_routed_experts=64, num_experts_per_tok=6,
+                             n_shared_experts=2, moe_intermediate_size=1408,
+                             first_k_dense_replace=1, n_group=8,
+                             topk_group=4, scoring_func="sigmoid",
+                             norm_topk_prob=True,
+                             routed_scaling_factor=2.5, rope_theta=10000.0),
     "tiny": dict(model_type="llama", hidden_size=128, num_hidden_layers=4,
                  num_attention_heads=2, num_key_value_heads=2, head_dim=64,
                  intermediate_size=256, vocab_size=256, rope_theta=10000.0),
