@@ -430,6 +430,8 @@ class ShardRuntime:
         ex.set_decoding(DecodingConfig(
             temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
             top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0)))
+        ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
+        ex.last_tops = None
         kv = ex.kvs[0].slot(si)
         kv.pos.fill_(0)
         h = torch.nn.functional.embedding(
