@@ -39,6 +39,7 @@ class ClusterManager:
         self.devices: dict[str, DeviceProperties] = {}
         self.profiles: dict[str, DeviceProfile] = {}
         self.link_ms: dict[tuple, float] = {}   # (src, dst) -> median ms
+        self.excluded: set = set()               # failed shards (recovery)
         self.topology: Optional[TopologyInfo] = None
 
     async def scan_devices(self) -> dict[str, DeviceProperties]:
@@ -46,7 +47,8 @@ class ClusterManager:
         return self.devices
 
     def shard_devices(self) -> list[DeviceProperties]:
-        return [d for d in self.devices.values() if not d.is_manager]
+        return [d for d in self.devices.values()
+                if not d.is_manager and d.instance not in self.excluded]
 
     async def profile_cluster(self, parallel: bool = True) -> dict:
         """Health-check then /profile each shard; merge latency medians."""
@@ -94,6 +96,22 @@ class ClusterManager:
                         if inst in self.profiles:
                             self.profiles[inst].t_comm_ms = med
         return {k: v.to_dict() for k, v in self.profiles.items()}
+
+    async def healthy_shards(self) -> list[DeviceProperties]:
+        """Quick /health sweep (reference: cluster.py health filtering) —
+        used by failure recovery to re-solve over the survivors."""
+        await self.scan_devices()
+        out = []
+        async with httpx.AsyncClient(timeout=5.0) as client:
+            for d in [d for d in self.devices.values() if not d.is_manager]:
+                try:
+                    r = await client.get(
+                        f"http://{d.local_ip}:{d.server_port}/health")
+                    if r.status_code == 200:
+                        out.append(d)
+                except httpx.HTTPError:
+                    pass
+        return out
 
     def get_head_node(self) -> Optional[DeviceProperties]:
         """Owner of layer 0 (reference: cluster.py get_head_node)."""

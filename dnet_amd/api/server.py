@@ -29,6 +29,7 @@ log = get_logger("api")
 class ApiState:
     def __init__(self, cluster: ClusterManager, settings):
         self.cluster = cluster
+        self.last_load = None          # APILoadModelRequest for recovery
         self.models = ModelManager(cluster)
         self.inference = InferenceManager(
             self.models, token_timeout_s=settings.api.request_timeout_s)
@@ -125,6 +126,7 @@ def build_api_app(state: ApiState) -> FastAPI:
             raise HTTPException(500, str(e))
         head = s.cluster.get_head_node()
         s.inference.connect_head(head.local_ip, head.shard_port, cb)
+        s.last_load = req
         return {"status": "ok", "model": req.model,
                 "topology": s.cluster.topology.model_dump()}
 
@@ -132,6 +134,38 @@ def build_api_app(state: ApiState) -> FastAPI:
     async def unload_model():
         await s.models.unload_model()
         return {"status": "ok"}
+
+    @app.post("/v1/recover")
+    async def recover():
+        """Elastic recovery after a shard failure: health-sweep, exclude
+        dead shards, re-solve the ring over the survivors and reload the
+        last-loaded model (drop-and-reload; in-flight requests error out
+        via the token timeout). Reference defines RingError but never
+        recovers — a dead mid-ring shard stays a timeout there."""
+        if s.last_load is None:
+            raise HTTPException(400, "nothing was loaded")
+        healthy = await s.cluster.healthy_shards()
+        if not healthy:
+            raise HTTPException(503, "no healthy shards")
+        all_shards = {d.instance for d in s.cluster.devices.values()
+                      if not d.is_manager}
+        s.cluster.excluded = all_shards - {d.instance for d in healthy}
+        await s.models.unload_model()          # best-effort on survivors
+        req = s.last_load
+        entry = get_entry(req.model)
+        cfg = resolve_model_config(entry, req.quant)
+        s.cluster.solve_topology(
+            req.model, cfg, master_port=s.settings.transport.master_port)
+        cb = s.settings.api.callback_addr or \
+            f"127.0.0.1:{s.settings.api.grpc_port}"
+        await s.models.load_model(
+            s.cluster.topology, entry, quant=req.quant,
+            max_batch=req.max_batch, max_seq=req.max_seq,
+            api_callback_address=cb)
+        head = s.cluster.get_head_node()
+        s.inference.connect_head(head.local_ip, head.shard_port, cb)
+        return {"status": "ok", "excluded": sorted(s.cluster.excluded),
+                "topology": s.cluster.topology.model_dump()}
 
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatRequestModel):

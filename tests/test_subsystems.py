@@ -418,3 +418,53 @@ def test_slot_stop_id_mid_stream():
     assert len(st) == 3 and st[-1][1]            # stopped at the stop id
     assert len(emitted["runs"]) == 8             # unaffected neighbor
     rt._unload()
+
+
+def test_recover_excludes_dead_shard(monkeypatch):
+    """/v1/recover: health-sweep drops the dead shard, the ring re-solves
+    over the survivors and the model reloads there."""
+    from dnet_amd.api import server as srv
+
+    cluster = ClusterManager(FakeDiscovery(_devices(2)))
+    state = ApiState(cluster, get_settings())
+    app = build_api_app(state)
+    client = TestClient(app)
+
+    loads = []
+
+    async def fake_load(topology, entry, **kw):
+        loads.append(topology)
+        state.models.loaded_model = entry.id
+
+    async def fake_unload():
+        pass
+
+    async def fake_profile(parallel=True):
+        await cluster.scan_devices()
+        return {}
+
+    async def fake_healthy():
+        await cluster.scan_devices()
+        return [d for d in cluster.devices.values()
+                if not d.is_manager and d.instance != "shard1"]
+
+    monkeypatch.setattr(state.models, "load_model", fake_load)
+    monkeypatch.setattr(state.models, "unload_model", fake_unload)
+    monkeypatch.setattr(cluster, "profile_cluster", fake_profile)
+    monkeypatch.setattr(cluster, "healthy_shards", fake_healthy)
+    monkeypatch.setattr(state.inference, "connect_head", lambda *a, **k: None)
+
+    r = client.post("/v1/load_model", json={"model": "tiny-random"})
+    assert r.status_code == 200, r.text
+    assert len(loads) == 1
+    assert set(loads[0].devices) == {"shard0", "shard1"}
+
+    r = client.post("/v1/recover")
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["excluded"] == ["shard1"]
+    assert len(loads) == 2
+    assert set(loads[1].devices) == {"shard0"}
+    # coverage: all layers still assigned across the survivors
+    total = sum(len(rd) for a in loads[1].assignments for rd in a.layers)
+    assert total == loads[1].num_layers
