@@ -71,3 +71,60 @@ class Sampler:
             tops = [{int(i): float(v) for v, i in zip(tv[b], ti[b])}
                     for b in range(logits.shape[0])]
         return tok, chosen, tops
+
+
+class RowSampler:
+    """Per-row sampling for slot-batched serving: each row of the logits
+    batch carries its own request's decoding params (temperature / top_k /
+    top_p / min_p), vectorized in one pass. Greedy rows (temperature 0)
+    take the argmax; sampled rows go through per-row filtered multinomial
+    with the same degenerate-row guard as ``Sampler``."""
+
+    def __init__(self, batch: int, device=None):
+        self.batch = batch
+        self.device = device
+        self.temp = torch.zeros(batch, device=device)
+        self.top_p = torch.ones(batch, device=device)
+        self.top_k = torch.zeros(batch, dtype=torch.long, device=device)
+        self.min_p = torch.zeros(batch, device=device)
+        self._has_min_p = False    # python-side (no device sync per step)
+
+    def set_row(self, i: int, cfg: DecodingConfig):
+        self.temp[i] = cfg.temperature
+        self.top_p[i] = cfg.top_p
+        self.top_k[i] = cfg.top_k
+        self.min_p[i] = cfg.min_p
+        if cfg.min_p > 0:
+            self._has_min_p = True
+
+    def sample(self, logits: torch.Tensor) -> torch.Tensor:
+        lf = logits.float()
+        B, V = lf.shape
+        greedy_tok = lf.argmax(dim=-1)
+        t = self.temp.clamp_min(1e-6).unsqueeze(-1)
+        x = lf / t
+        sx, si = torch.sort(x, descending=True, dim=-1)
+        ranks = torch.arange(V, device=lf.device).expand(B, V)
+        keep = torch.ones_like(sx, dtype=torch.bool)
+        # per-row top-k (0 = off)
+        k = torch.where(self.top_k > 0, self.top_k,
+                        torch.full_like(self.top_k, V))
+        keep &= ranks < k.unsqueeze(-1)
+        # per-row top-p (keep at least the top token)
+        probs_s = torch.softmax(sx, dim=-1)
+        cum = probs_s.cumsum(dim=-1)
+        keep &= (cum - probs_s) < self.top_p.unsqueeze(-1)
+        sx = sx.masked_fill(~keep, float("-inf"))
+        x = torch.full_like(x, float("-inf")).scatter_(-1, si, sx)
+        if self._has_min_p:
+            p = torch.softmax(x, dim=-1)
+            thresh = self.min_p.unsqueeze(-1) * p.amax(-1, keepdim=True)
+            x = x.masked_fill(p < thresh, float("-inf"))
+        probs = torch.softmax(x, dim=-1)
+        probs = torch.nan_to_num(probs, nan=0.0, posinf=0.0)
+        ok = probs.sum(dim=-1, keepdim=True) > 0
+        fallback = torch.nn.functional.one_hot(
+            lf.nan_to_num(nan=0.0).argmax(dim=-1), V).float()
+        probs = torch.where(ok, probs, fallback)
+        tok = torch.multinomial(probs, 1).squeeze(-1)
+        return torch.where(self.temp <= 0.0, greedy_tok, tok)

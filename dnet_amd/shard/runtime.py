@@ -177,6 +177,9 @@ class ShardRuntime:
         # loading with max_batch > 1 on a single-rank ring
         self.slots = ([None] * req.max_batch
                       if req.max_batch > 1 and req.world_size == 1 else None)
+        if self.slots is not None:
+            from ..core.sampler import RowSampler
+            self._row_sampler = RowSampler(req.max_batch, device=ex.device)
         self.load_req = req
         self.model_name = req.model_name or req.model_path
         if ex.is_last and req.api_callback_address:
@@ -427,9 +430,11 @@ class ShardRuntime:
             np.frombuffer(frame["tokens"], dtype=np.int32).copy()).long()
         T = int(frame.get("prompt_len", tokens.numel()))
         p = frame.get("params", {})
-        ex.set_decoding(DecodingConfig(
+        cfg = DecodingConfig(
             temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
-            top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0)))
+            top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0))
+        # per-slot sampling params (each request keeps its own)
+        self._row_sampler.set_row(si, cfg)
         ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
         ex.last_tops = None
         kv = ex.kvs[0].slot(si)
@@ -439,7 +444,8 @@ class ShardRuntime:
         ex.model.prefill_window(h, ex.my_layers, kv, 0)
         kv.pos.fill_(T)
         logits = ex.model.normalize_project(h[:, -1].contiguous())
-        tok, _, _ = ex.sampler.sample(logits.float())
+        from ..core.sampler import Sampler
+        tok, _, _ = Sampler(cfg).sample(logits.float())
         t0 = int(tok[0])
         ex.tokbuf[0][si] = t0
         st = {"nonce": frame.get("nonce", ""), "produced": 1,
@@ -460,7 +466,7 @@ class ShardRuntime:
         ex = self.executor
         for r in range(ex.rounds):
             ex._run_decode(0, r)
-        toks_t, _, _ = ex.sampler.sample(ex.logits_buf[0].float())
+        toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
         ex.tokbuf[0].copy_(toks_t)
         ex.kvs[0].pos.add_(1)
         # park idle slots at 0 so their dummy writes stay in range

@@ -468,3 +468,44 @@ def test_recover_excludes_dead_shard(monkeypatch):
     # coverage: all layers still assigned across the survivors
     total = sum(len(rd) for a in loads[1].assignments for rd in a.layers)
     assert total == loads[1].num_layers
+
+
+def test_slot_per_request_sampling_params():
+    """Each slot samples with ITS OWN request's params: a greedy request
+    and a top-k=1 request produce the greedy sequence; a wild high-temp
+    request does not perturb them."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    emitted: dict[str, list] = {}
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(frame["token_id"])
+
+        def close(self):
+            pass
+
+    rt = ShardRuntime("probe")
+    rt._load(ShardLoadModelRequest(
+        model_path="tiny", model_name="tiny", total_layers=4,
+        layers=[0, 1, 2, 3], rank=0, world_size=1, max_batch=3, max_seq=64))
+    rt._callback = Cap()
+    prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    rt.infer_q.put({"nonce": "greedy", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 6, "stop_ids": [], "params": {}})
+    rt.infer_q.put({"nonce": "topk1", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 6, "stop_ids": [],
+                    "params": {"temperature": 0.7, "top_k": 1}})
+    rt.infer_q.put({"nonce": "hot", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 6, "stop_ids": [],
+                    "params": {"temperature": 5.0}})
+    for _ in range(25):
+        rt._slots_tick()
+        if all(s is None for s in rt.slots) and rt._pending is None:
+            break
+    assert len(emitted["greedy"]) == 6
+    # top-k=1 at any temperature IS greedy -> identical sequence
+    assert emitted["topk1"] == emitted["greedy"]
+    assert len(emitted["hot"]) == 6
+    rt._unload()
