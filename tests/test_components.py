@@ -295,3 +295,20 @@ def test_compression_qsparse8_roundtrip():
     assert y[:, mask].abs().sum() == 0
     assert cz.is_compressed_dtype(
         cz.dtype_string("bfloat16", 0.5, cz.FMT_QSPARSE8_V1))
+
+
+def test_row_sampler():
+    from dnet_amd.core.sampler import DecodingConfig, RowSampler
+    torch.manual_seed(0)
+    rs = RowSampler(3)
+    rs.set_row(0, DecodingConfig(temperature=0.0))
+    rs.set_row(1, DecodingConfig(temperature=1.0, top_k=1))
+    rs.set_row(2, DecodingConfig(temperature=0.8, top_p=0.01))
+    logits = torch.tensor([[0.1, 5.0, 0.2, 0.3]] * 3)
+    for _ in range(8):
+        assert rs.sample(logits).tolist() == [1, 1, 1]
+    # min-p path: mass below min_p * max prob is filtered
+    rs2 = RowSampler(1)
+    rs2.set_row(0, DecodingConfig(temperature=1.0, min_p=0.9))
+    for _ in range(8):
+        assert int(rs2.sample(logits[:1])[0]) == 1
