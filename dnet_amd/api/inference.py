@@ -74,6 +74,7 @@ class InferenceManager:
                 "prompt_len": len(prompt_ids),
                 "max_tokens": request.effective_max_tokens,
                 "params": {
+                    "seed": request.seed,
                     "temperature": request.temperature, "top_p": request.top_p,
                     "top_k": request.top_k, "min_p": request.min_p,
                     "logprobs": request.logprobs,

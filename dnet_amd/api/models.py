@@ -23,6 +23,7 @@ class ChatRequestModel(BaseModel):
     max_tokens: Optional[int] = None
     max_completion_tokens: Optional[int] = None
     temperature: float = 0.0
+    seed: int | None = None
     top_p: float = 1.0
     top_k: int = 0
     min_p: float = 0.0

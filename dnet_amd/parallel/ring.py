@@ -334,8 +334,12 @@ class RingExecutor:
         for kv in self.kvs:
             kv.reset()
 
-    def set_decoding(self, cfg: DecodingConfig):
-        self.sampler = Sampler(cfg)
+    def set_decoding(self, cfg: DecodingConfig, seed=None):
+        g = None
+        if seed is not None:
+            g = torch.Generator(device=self.device)
+            g.manual_seed(int(seed))
+        self.sampler = Sampler(cfg, generator=g)
 
     def decode_stream(self, max_tokens: int, stop_ids=(), on_token=None,
                       mb: int = 0):

@@ -357,7 +357,8 @@ class ShardRuntime:
             temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
             top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0),
             logprobs=bool(p.get("logprobs", False)),
-            top_logprobs=int(p.get("top_logprobs", 0))))
+            top_logprobs=int(p.get("top_logprobs", 0))),
+            seed=p.get("seed"))
         # pad the single sequence to the executor's batch width
         B = ex.mb_size
         toks = tokens.expand(1, B, tokens.shape[-1]).contiguous().to(ex.device)

@@ -509,3 +509,38 @@ def test_slot_per_request_sampling_params():
     assert emitted["topk1"] == emitted["greedy"]
     assert len(emitted["hot"]) == 6
     rt._unload()
+
+
+def test_seeded_sampling_reproducible():
+    """OpenAI `seed`: identical seeded requests sample identical tokens at
+    temperature > 0 (legacy serial path)."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    emitted: dict[str, list] = {}
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(frame["token_id"])
+
+        def close(self):
+            pass
+
+    rt = ShardRuntime("probe")
+    rt._load(ShardLoadModelRequest(
+        model_path="tiny", model_name="tiny", total_layers=4,
+        layers=[0, 1, 2, 3], rank=0, world_size=1, max_batch=1, max_seq=64))
+    rt._callback = Cap()
+    prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+
+    def infer(nonce, seed):
+        rt._execute_infer(nonce, torch.frombuffer(
+            bytearray(prompt), dtype=torch.int32).long().view(1, 1, -1),
+            8, [], {"temperature": 1.0, "seed": seed})
+
+    infer("a", 42)
+    infer("b", 42)
+    infer("c", 7)
+    assert emitted["a"] == emitted["b"]
+    assert len(emitted["c"]) == 8
+    rt._unload()
