@@ -334,6 +334,55 @@ class RingExecutor:
         for kv in self.kvs:
             kv.reset()
 
+    # ------------- slot-batched serving (continuous batching) -------------
+
+    def prefill_slot(self, si: int, tokens: torch.Tensor):
+        """Collective single-slot prefill through the ring: one sequence
+        flows stage to stage while other slots keep decoding state.
+        ``tokens`` [T] int64 (significant on the first stage). Returns the
+        last-position logits [1, V] on the LAST stage, else None."""
+        T = int(tokens.shape[-1])
+        H = self.cfg.hidden_size
+        kvslot = self.kvs[0].slot(si)
+        kvslot.pos.fill_(0)
+        h = None
+        for r in range(self.rounds):
+            if self.is_first and r == 0:
+                h = self.model.embed_tokens(
+                    tokens.view(1, T).to(self.device)).clone()
+            else:
+                if h is None:
+                    h = torch.empty(1, T, H, dtype=torch.bfloat16,
+                                    device=self.device)
+                self.ring.recv(h)
+            if self.windows[r]:
+                self.model.prefill_window(h, self.windows[r], kvslot, 0)
+            if not (self.is_last and r == self.rounds - 1):
+                if self.stages > 1:
+                    self.ring.send(h)
+        kvslot.pos.fill_(T)
+        if self.is_last:
+            return self.model.normalize_project(h[:, -1].contiguous())
+        return None
+
+    def slot_step_compute(self, mb: int = 0) -> None:
+        """One decode step over the whole slot batch, including the ring
+        hops — no sampling (the caller samples on the last stage)."""
+        last_r = self.rounds - 1
+        if self.is_first:
+            for r in range(self.rounds):
+                if r > 0:
+                    self._recv_hidden(mb)
+                self._run_decode(mb, r)
+                if self.stages > 1 and not (self.is_last and r == last_r):
+                    self._send_hidden(mb)
+        elif self.stages > 1:
+            for r in range(self.rounds):
+                self._recv_hidden(mb)
+                self._run_decode(mb, r)
+                if not (self.is_last and r == last_r):
+                    self._send_hidden(mb)
+
     def set_decoding(self, cfg: DecodingConfig, seed=None):
         g = None
         if seed is not None:

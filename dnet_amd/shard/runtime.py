@@ -36,6 +36,7 @@ from ..utils.model_meta import get_model_metadata, load_tensors
 log = get_logger("shard")
 
 CMD_NOOP, CMD_INFER, CMD_UNLOAD, CMD_SHUTDOWN = 0, 1, 2, 3
+CMD_SLOT_ADMIT, CMD_SLOT_STEP = 4, 5
 
 
 class SyncWireClient:
@@ -174,9 +175,10 @@ class ShardRuntime:
         self.executor = ex
         # slot-based continuous batching: several single-stream requests
         # share the decode batch (each owns one KV slot); enabled by
-        # loading with max_batch > 1 on a single-rank ring
-        self.slots = ([None] * req.max_batch
-                      if req.max_batch > 1 and req.world_size == 1 else None)
+        # loading with max_batch > 1. Multi-rank rings coordinate slot
+        # admits/steps via the command broadcasts (every rank keeps the
+        # same slot state; the last stage samples and emits).
+        self.slots = [None] * req.max_batch if req.max_batch > 1 else None
         if self.slots is not None:
             from ..core.sampler import RowSampler
             self._row_sampler = RowSampler(req.max_batch, device=ex.device)
@@ -274,6 +276,10 @@ class ShardRuntime:
                 cmd = self._recv_cmd()
                 if cmd[0] == CMD_INFER:
                     self._run_infer_follower(cmd)
+                elif cmd[0] == CMD_SLOT_ADMIT:
+                    self._slot_admit_follower(cmd)
+                elif cmd[0] == CMD_SLOT_STEP:
+                    self._slot_step_exec()
                 elif cmd[0] == CMD_UNLOAD:
                     self._unload()
 
@@ -389,7 +395,7 @@ class ShardRuntime:
                  "emit_ms=%.1f", nonce[:18], n, dt * 1e3, n / max(dt, 1e-9),
                  self._emit_s * 1e3)
 
-    # ---------- slot-based continuous batching (world == 1) ----------
+    # ---------- slot-based continuous batching ----------
 
     def _slots_tick(self) -> None:
         """One scheduler iteration: admit queued requests into free KV
@@ -397,8 +403,10 @@ class ShardRuntime:
         one decode step for every active slot. Reference has no equivalent
         (requests serialize there); this keeps N single-stream requests at
         ~single-stream latency each."""
+        ex = self.executor
         active = any(s is not None for s in self.slots)
         block = not active
+        progressed = False
         while any(s is None for s in self.slots):
             free = next(i for i, s in enumerate(self.slots) if s is None)
             try:
@@ -407,24 +415,35 @@ class ShardRuntime:
             except queue.Empty:
                 break
             block = False
+            progressed = True
             try:
                 self._slot_admit(free, frame)
             except Exception:
                 log.exception("slot admit failed")
                 self._send_error(frame.get("nonce", ""))
         if any(s is not None for s in self.slots):
-            # pipeline: launch step n+1 (device-side deps only), THEN emit
-            # step n's tokens — the host sync overlaps the next step's GPU
-            # work instead of serializing with it
-            launched = self._slot_step_launch()
-            if self._pending is not None:
-                self._slot_emit(*self._pending)
-            self._pending = launched
+            progressed = True
+            if ex.world > 1:
+                # multi-rank: broadcast the step command; emit synchronously
+                self._broadcast_cmd(CMD_SLOT_STEP)
+                self._slot_step_exec()
+            else:
+                # single rank: pipeline — launch step n+1 (device-side deps
+                # only), THEN emit step n's tokens so the host sync overlaps
+                # the next step's GPU work
+                launched = self._slot_step_launch()
+                if self._pending is not None:
+                    self._slot_emit(*self._pending)
+                self._pending = launched
         elif self._pending is not None:
             self._slot_emit(*self._pending)
             self._pending = None
+        if not progressed and ex.world > 1:
+            self._broadcast_cmd(CMD_NOOP)
 
     def _slot_admit(self, si: int, frame: dict) -> None:
+        """Rank-0 admit: broadcast the slot command + payload, then run the
+        collective admit on this rank too."""
         import numpy as np
         ex = self.executor
         tokens = torch.from_numpy(
@@ -434,30 +453,88 @@ class ShardRuntime:
         cfg = DecodingConfig(
             temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
             top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0))
-        # per-slot sampling params (each request keeps its own)
+        stop_ids = list(frame.get("stop_ids", []))
+        max_tokens = int(frame.get("max_tokens", 128))
+        nonce = frame.get("nonce", "")
+        nonce_ids = list(nonce.encode("utf-8"))[:64]
+        if ex.world > 1:
+            import torch.distributed as dist
+            self._broadcast_cmd(CMD_SLOT_ADMIT, si, T, max_tokens,
+                                len(stop_ids), len(nonce_ids),
+                                cfg.temperature, cfg.top_p, cfg.top_k,
+                                cfg.min_p)
+            payload = torch.cat([
+                tokens.flatten().to(self._comm_device()),
+                torch.tensor(stop_ids + nonce_ids, dtype=torch.int64,
+                             device=self._comm_device())])
+            dist.broadcast(payload, src=0)
+        self._slot_admit_exec(si, tokens.view(-1), max_tokens, stop_ids,
+                              nonce, cfg)
+
+    def _slot_admit_follower(self, cmd) -> None:
+        import torch.distributed as dist
+        si, T, max_tokens = int(cmd[1]), int(cmd[2]), int(cmd[3])
+        n_stop, n_nonce = int(cmd[4]), int(cmd[5])
+        cfg = DecodingConfig(temperature=cmd[6], top_p=cmd[7],
+                             top_k=int(cmd[8]), min_p=cmd[9])
+        payload = torch.zeros(T + n_stop + n_nonce, dtype=torch.int64,
+                              device=self._comm_device())
+        dist.broadcast(payload, src=0)
+        tokens = payload[:T].cpu()
+        stop_ids = payload[T:T + n_stop].tolist()
+        nonce = bytes(payload[T + n_stop:].tolist()).decode("utf-8", "replace")
+        self._slot_admit_exec(si, tokens, max_tokens, stop_ids, nonce, cfg)
+
+    def _slot_admit_exec(self, si, tokens, max_tokens, stop_ids, nonce,
+                         cfg) -> None:
+        """Collective slot admit (every rank): single-slot ring prefill,
+        sample on the last stage, broadcast the first token, register the
+        slot state identically everywhere."""
+        ex = self.executor
         self._row_sampler.set_row(si, cfg)
         ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
         ex.last_tops = None
-        kv = ex.kvs[0].slot(si)
-        kv.pos.fill_(0)
-        h = torch.nn.functional.embedding(
-            tokens.view(1, T).to(ex.device), ex.model.embed)
-        ex.model.prefill_window(h, ex.my_layers, kv, 0)
-        kv.pos.fill_(T)
-        logits = ex.model.normalize_project(h[:, -1].contiguous())
-        from ..core.sampler import Sampler
-        tok, _, _ = Sampler(cfg).sample(logits.float())
-        t0 = int(tok[0])
+        logits = ex.prefill_slot(si, tokens)
+        t0_t = torch.zeros(1, dtype=torch.int64, device=self._comm_device())
+        if ex.is_last:
+            from ..core.sampler import Sampler
+            tok, _, _ = Sampler(cfg).sample(logits.float())
+            t0_t[0] = int(tok[0])
+        if ex.world > 1:
+            import torch.distributed as dist
+            dist.broadcast(t0_t, src=(ex.stages - 1) * ex.tp)
+        t0 = int(t0_t[0])
         ex.tokbuf[0][si] = t0
-        st = {"nonce": frame.get("nonce", ""), "produced": 1,
-              "max_tokens": int(frame.get("max_tokens", 128)),
-              "stop_ids": set(frame.get("stop_ids", []))}
-        done = t0 in st["stop_ids"] or st["max_tokens"] <= 1
-        self._emit_token(st["nonce"], t0, finished=done)
+        st = {"nonce": nonce, "produced": 1, "max_tokens": max_tokens,
+              "stop_ids": set(stop_ids)}
+        done = t0 in st["stop_ids"] or max_tokens <= 1
+        if ex.is_last:
+            self._emit_token(nonce, t0, finished=done)
         if not done:
             self.slots[si] = st
         log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
-                 st["nonce"][:18], T)
+                 nonce[:18], int(tokens.shape[-1]))
+
+    def _slot_step_exec(self) -> None:
+        """Collective decode step for all slots (multi-rank path, emits
+        synchronously): hops + compute, sample on the last stage, token
+        broadcast, identical slot-state update on every rank."""
+        ex = self.executor
+        ex.slot_step_compute()
+        if ex.is_last:
+            toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
+            ex.tokbuf[0].copy_(toks_t)
+        if ex.world > 1:
+            import torch.distributed as dist
+            dist.broadcast(ex.tokbuf[0], src=(ex.stages - 1) * ex.tp)
+        ex.kvs[0].pos.add_(1)
+        idle = [i for i, st in enumerate(self.slots) if st is None]
+        if idle:
+            ex.kvs[0].pos[torch.tensor(idle, dtype=torch.long,
+                                       device=ex.device)] = 0
+        self._slot_emit(ex.tokbuf[0],
+                        [i for i, st in enumerate(self.slots)
+                         if st is not None])
 
     _pending = None   # (device tokens, active slot list) of the in-flight step
 
@@ -465,8 +542,7 @@ class ShardRuntime:
         """Enqueue one decode step for the whole batch — device ops only,
         no host sync. Returns (tokens device tensor, active slots)."""
         ex = self.executor
-        for r in range(ex.rounds):
-            ex._run_decode(0, r)
+        ex.slot_step_compute()
         toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
         ex.tokbuf[0].copy_(toks_t)
         ex.kvs[0].pos.add_(1)

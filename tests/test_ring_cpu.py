@@ -645,3 +645,109 @@ def test_gpt_oss_tp2_matches_single():
     # FIRST token
     assert torch.equal(out[..., :3], single[..., :3]), \
         f"gpt-oss tp2 != single:\n{out}\n{single}"
+
+
+def _slots_pp2_rank_main(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+    layer_rounds = [[0, 1]] if rank == 0 else [[2, 3]]
+    rt = ShardRuntime(f"s{rank}")
+    rt._load(ShardLoadModelRequest(
+        model_path="tiny", model_name="tiny", total_layers=4,
+        layers=layer_rounds[0], layer_rounds=layer_rounds, rank=rank,
+        world_size=2, master_addr="127.0.0.1", master_port=port + 1,
+        max_batch=2, max_seq=64))
+    emitted: dict = {}
+    finished = [0]
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(frame["token_id"])
+            if frame.get("finished"):
+                finished[0] += 1
+
+        def close(self):
+            pass
+
+    if rank == 1:
+        rt._callback = Cap()
+    prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    if rank == 0:
+        rt.infer_q.put({"nonce": "a", "tokens": prompt, "prompt_len": 8,
+                        "max_tokens": 6, "stop_ids": [], "params": {}})
+        rt.infer_q.put({"nonce": "b", "tokens": prompt, "prompt_len": 8,
+                        "max_tokens": 4, "stop_ids": [], "params": {}})
+        while (any(s is not None for s in rt.slots)
+               or not rt.infer_q.empty()):
+            rt._slots_tick()
+        q.put(("rank0", None))
+    else:
+        while finished[0] < 2:
+            cmd = rt._recv_cmd()
+            from dnet_amd.shard import runtime as R
+            if cmd[0] == R.CMD_SLOT_ADMIT:
+                rt._slot_admit_follower(cmd)
+            elif cmd[0] == R.CMD_SLOT_STEP:
+                rt._slot_step_exec()
+        q.put(("rank1", emitted))
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(480)
+@retry_flaky()
+def test_slots_pp2_matches_single():
+    """Multi-rank continuous batching: 2 concurrent requests over a
+    2-stage ring produce exactly the single-rank slot scheduler's tokens
+    (same deterministic weights; the last stage samples and emits)."""
+    # single-rank slots reference
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+    emitted: dict = {}
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(frame["token_id"])
+
+        def close(self):
+            pass
+
+    rt = ShardRuntime("single")
+    rt._load(ShardLoadModelRequest(
+        model_path="tiny", model_name="tiny", total_layers=4,
+        layers=[0, 1, 2, 3], rank=0, world_size=1, max_batch=2, max_seq=64))
+    rt._callback = Cap()
+    prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    rt.infer_q.put({"nonce": "a", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 6, "stop_ids": [], "params": {}})
+    rt.infer_q.put({"nonce": "b", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 4, "stop_ids": [], "params": {}})
+    for _ in range(30):
+        rt._slots_tick()
+        if (all(s is None for s in rt.slots) and rt._pending is None
+                and rt.infer_q.empty()):
+            break
+    rt._unload()
+    assert len(emitted["a"]) == 6 and len(emitted["b"]) == 4
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_slots_pp2_rank_main, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        name, data = q.get(timeout=200)
+        results[name] = data
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    pp2 = results["rank1"]
+    assert pp2["a"] == emitted["a"], f"{pp2}\n{emitted}"
+    assert pp2["b"] == emitted["b"]
