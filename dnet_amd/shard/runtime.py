@@ -184,7 +184,10 @@ class ShardRuntime:
             self._row_sampler = RowSampler(req.max_batch, device=ex.device)
         self.load_req = req
         self.model_name = req.model_name or req.model_path
-        if ex.is_last and req.api_callback_address:
+        # one emitter per ring: the grp-rank-0 member of the last stage
+        # (TP/CP peers compute identical logits; two callbacks would
+        # double-emit every token)
+        if ex.is_last and ex.tp_rank == 0 and req.api_callback_address:
             host, _, port = req.api_callback_address.rpartition(":")
             self._callback = SyncWireClient(host or "127.0.0.1", int(port))
         self.status = "loaded"
