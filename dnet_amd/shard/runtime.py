@@ -13,6 +13,7 @@ the GPU (SURVEY.md §7 hard-part (3)).
 from __future__ import annotations
 
 import json
+import os
 import queue
 import socket
 import struct
@@ -156,14 +157,17 @@ class ShardRuntime:
         ex = RingExecutor(cfg, req.rank, req.world_size, device, plan=plan,
                           mb_count=1, mb_size=req.max_batch,
                           smax=req.max_seq,
-                          # slot scheduler v1 decodes eager: per-slot
-                          # prefills interleave with decode steps and the
-                          # eager gap is small at serving batch sizes;
-                          # graph capture under slot churn is a round-2
-                          # validation item
+                          # slot scheduler decodes eager by default (the
+                          # eager gap is small at serving batch sizes);
+                          # DNET_SLOTS_GRAPHS=1 opts into hipGraph decode
+                          # under slot churn (replays interleaved with
+                          # eager per-slot prefills — validate before
+                          # relying on it)
                           use_graphs=(device.type == "cuda"
                                       and self.settings.compute.use_graphs
-                                      and residency == 0 and not slots_mode),
+                                      and residency == 0
+                                      and (not slots_mode or os.environ.get(
+                                          "DNET_SLOTS_GRAPHS") == "1")),
                           init_weights=synthetic, residency=residency,
                           kv_bits=req.kv_bits)
         if not synthetic:
