@@ -51,6 +51,9 @@ def main():
                     help="tensor-parallel degree inside each pipeline stage")
     ap.add_argument("--cp", type=int, default=1,
                     help="context-parallel degree (KV sequence sharding)")
+    ap.add_argument("--prefill-chunk", type=int, default=0,
+                    help="prefill position-chunk size (bounds activation "
+                         "memory for long prompts; 0 = one shot)")
     ap.add_argument("--residency", type=int, default=0,
                     help="GPU-resident layers per rank (0=all; <local layers "
                          "enables host-DRAM weight streaming)")
@@ -92,7 +95,7 @@ def main():
     # TTFT: prefill + first token
     barrier_sync()
     t0 = time.perf_counter()
-    ex.prefill(tokens)
+    ex.prefill(tokens, chunk=args.prefill_chunk)
     barrier_sync()
     ttft_ms = (time.perf_counter() - t0) * 1e3
 

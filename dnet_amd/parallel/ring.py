@@ -232,33 +232,43 @@ class RingExecutor:
 
     # ------------- collective schedules -------------
 
-    def prefill(self, tokens: torch.Tensor) -> torch.Tensor | None:
+    def prefill(self, tokens: torch.Tensor,
+                chunk: int = 0) -> torch.Tensor | None:
         """tokens: [mb_count, mb_size, T] int64 (significant on rank 0; other
         ranks use it for shape only). Fills KV, samples the first token per
         sequence. Returns first-token tensor [mb_count, mb_size] on rank 0
-        (and on the last rank), else None."""
+        (and on the last rank), else None.
+
+        ``chunk`` > 0 prefills in position chunks of that size (row-wise
+        identical math — same tokens — while bounding the activation
+        buffer to [B, chunk, H] for long prompts)."""
         M, B, T = tokens.shape
         assert M == self.mb_count and B == self.mb_size
         H = self.cfg.hidden_size
         tok_reqs = []
         first_tokens = torch.zeros(M, B, dtype=torch.int64, device=self.device)
         last_r = self.rounds - 1
+        step = chunk if 0 < chunk < T else T
         for mb in range(M):
-            h = None
-            for r in range(self.rounds):
-                if self.is_first and r == 0:
-                    h = self.model.embed_tokens(tokens[mb].to(self.device)).clone()
-                else:
-                    if h is None:
-                        h = torch.empty(B, T, H, dtype=torch.bfloat16,
-                                        device=self.device)
-                    self.ring.recv(h)
-                if self.windows[r]:
-                    self.model.prefill_window(h, self.windows[r],
-                                              self.kvs[mb], 0)
-                if not (self.is_last and r == last_r):
-                    if self.stages > 1:
-                        self.ring.send(h)
+            for p0 in range(0, T, step):
+                p1 = min(p0 + step, T)
+                h = None
+                for r in range(self.rounds):
+                    if self.is_first and r == 0:
+                        h = self.model.embed_tokens(
+                            tokens[mb, :, p0:p1].to(self.device)).clone()
+                    else:
+                        if h is None:
+                            h = torch.empty(B, p1 - p0, H,
+                                            dtype=torch.bfloat16,
+                                            device=self.device)
+                        self.ring.recv(h)
+                    if self.windows[r]:
+                        self.model.prefill_window(h, self.windows[r],
+                                                  self.kvs[mb], p0)
+                    if not (self.is_last and r == last_r):
+                        if self.stages > 1:
+                            self.ring.send(h)
             self.kvs[mb].pos.fill_(T)
             if not self.is_last:
                 if self.is_first:

@@ -375,7 +375,7 @@ class ShardRuntime:
         # pad the single sequence to the executor's batch width
         B = ex.mb_size
         toks = tokens.expand(1, B, tokens.shape[-1]).contiguous().to(ex.device)
-        first = ex.prefill(toks)
+        first = ex.prefill(toks, chunk=2048)   # bound activation memory
         # every rank must agree on EOS-after-first-token before entering the
         # collective decode loop; rank 0 holds the first token in tokbuf.
         tok0_t = ex.tokbuf[0][:1].clone()

@@ -751,3 +751,51 @@ def test_slots_pp2_matches_single():
     pp2 = results["rank1"]
     assert pp2["a"] == emitted["a"], f"{pp2}\n{emitted}"
     assert pp2["b"] == emitted["b"]
+
+
+def _chunk_rank_main(rank, world, port, q, chunk):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    import torch.distributed as dist
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, rank, world, "cpu", mb_count=MB_COUNT,
+                      mb_size=MB_SIZE, smax=64, seed=7, use_graphs=False)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks, chunk=chunk)
+    gen = ex.decode_rounds(NGEN)
+    if rank == 0:
+        q.put(torch.cat([first.unsqueeze(-1), gen], dim=-1))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(480)
+@retry_flaky()
+def test_chunked_prefill_matches_full():
+    """Prefill in 3-token position chunks (bounded activation memory for
+    long prompts) == one-shot prefill, token-exact — single rank and over
+    a 2-stage ring."""
+    single = _run_single()
+    cfg = ModelConfig.from_hf(CFG)
+    ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=MB_COUNT, mb_size=MB_SIZE,
+                      smax=64, seed=7, use_graphs=False)
+    toks = _tokens(cfg)
+    first = ex.prefill(toks, chunk=3)
+    out = torch.cat([first.unsqueeze(-1), ex.decode_rounds(NGEN)], dim=-1)
+    assert torch.equal(out, single)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_chunk_rank_main, args=(r, 2, port, q, 3))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out2 = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert torch.equal(out2, single)
