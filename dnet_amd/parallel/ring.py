@@ -346,30 +346,36 @@ class RingExecutor:
 
     # ------------- slot-batched serving (continuous batching) -------------
 
-    def prefill_slot(self, si: int, tokens: torch.Tensor):
+    def prefill_slot(self, si: int, tokens: torch.Tensor,
+                     chunk: int = 2048):
         """Collective single-slot prefill through the ring: one sequence
-        flows stage to stage while other slots keep decoding state.
-        ``tokens`` [T] int64 (significant on the first stage). Returns the
-        last-position logits [1, V] on the LAST stage, else None."""
+        flows stage to stage (position-chunked to bound activations) while
+        other slots keep decoding state. ``tokens`` [T] int64 (significant
+        on the first stage). Returns the last-position logits [1, V] on
+        the LAST stage, else None."""
         T = int(tokens.shape[-1])
         H = self.cfg.hidden_size
         kvslot = self.kvs[0].slot(si)
         kvslot.pos.fill_(0)
+        step = chunk if 0 < chunk < T else T
         h = None
-        for r in range(self.rounds):
-            if self.is_first and r == 0:
-                h = self.model.embed_tokens(
-                    tokens.view(1, T).to(self.device)).clone()
-            else:
-                if h is None:
-                    h = torch.empty(1, T, H, dtype=torch.bfloat16,
-                                    device=self.device)
-                self.ring.recv(h)
-            if self.windows[r]:
-                self.model.prefill_window(h, self.windows[r], kvslot, 0)
-            if not (self.is_last and r == self.rounds - 1):
-                if self.stages > 1:
-                    self.ring.send(h)
+        for p0 in range(0, T, step):
+            p1 = min(p0 + step, T)
+            h = None
+            for r in range(self.rounds):
+                if self.is_first and r == 0:
+                    h = self.model.embed_tokens(
+                        tokens.view(1, T)[:, p0:p1].to(self.device)).clone()
+                else:
+                    if h is None:
+                        h = torch.empty(1, p1 - p0, H, dtype=torch.bfloat16,
+                                        device=self.device)
+                    self.ring.recv(h)
+                if self.windows[r]:
+                    self.model.prefill_window(h, self.windows[r], kvslot, p0)
+                if not (self.is_last and r == self.rounds - 1):
+                    if self.stages > 1:
+                        self.ring.send(h)
         kvslot.pos.fill_(T)
         if self.is_last:
             return self.model.normalize_project(h[:, -1].contiguous())
