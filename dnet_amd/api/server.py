@@ -67,6 +67,7 @@ def build_api_app(state: ApiState) -> FastAPI:
         if entry is None:
             raise HTTPException(404, f"unknown model {req.model}")
         cfg = resolve_model_config(entry, req.quant)
+        s.cluster.excluded = set()   # explicit re-prepare forgives failures
         await s.cluster.profile_cluster()
         topo = s.cluster.solve_topology(
             req.model, cfg, master_port=s.settings.transport.master_port,
