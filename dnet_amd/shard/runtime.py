@@ -366,6 +366,10 @@ class ShardRuntime:
                        max_tokens: int, stop_ids: list, p: dict) -> None:
         ex = self.executor
         ex.reset()
+        # never decode past the KV capacity (rope_append at pos >= smax
+        # would write out of range)
+        T_prompt = tokens.shape[-1]
+        max_tokens = max(1, min(max_tokens, ex.smax - T_prompt))
         ex.set_decoding(DecodingConfig(
             temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
             top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0),
@@ -498,6 +502,8 @@ class ShardRuntime:
         sample on the last stage, broadcast the first token, register the
         slot state identically everywhere."""
         ex = self.executor
+        max_tokens = max(1, min(max_tokens,
+                                ex.smax - int(tokens.shape[-1])))
         self._row_sampler.set_row(si, cfg)
         ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
         ex.last_tops = None

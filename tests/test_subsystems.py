@@ -544,3 +544,42 @@ def test_seeded_sampling_reproducible():
     assert emitted["a"] == emitted["b"]
     assert len(emitted["c"]) == 8
     rt._unload()
+
+
+def test_decode_never_exceeds_kv_capacity():
+    """max_tokens is clamped so decode never writes past smax (an
+    unclamped request would rope_append out of cache range)."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    emitted: dict[str, list] = {}
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(frame["token_id"])
+
+        def close(self):
+            pass
+
+    for max_batch in (1, 2):
+        emitted.clear()
+        rt = ShardRuntime("probe")
+        rt._load(ShardLoadModelRequest(
+            model_path="tiny", model_name="tiny", total_layers=4,
+            layers=[0, 1, 2, 3], rank=0, world_size=1,
+            max_batch=max_batch, max_seq=16))
+        rt._callback = Cap()
+        prompt = torch.arange(1, 11, dtype=torch.int32).numpy().tobytes()
+        if max_batch == 1:
+            rt._execute_infer("x", torch.frombuffer(
+                bytearray(prompt), dtype=torch.int32).long().view(1, 1, -1),
+                500, [], {})
+        else:
+            rt.infer_q.put({"nonce": "x", "tokens": prompt, "prompt_len": 10,
+                            "max_tokens": 500, "stop_ids": [], "params": {}})
+            for _ in range(30):
+                rt._slots_tick()
+                if all(s is None for s in rt.slots) and rt._pending is None:
+                    break
+        assert len(emitted["x"]) == 6    # smax 16 - prompt 10
+        rt._unload()
