@@ -432,7 +432,16 @@ class ShardRuntime:
             except Exception:
                 log.exception("slot admit failed")
                 self._send_error(frame.get("nonce", ""))
-        if any(s is not None for s in self.slots):
+        prefilling = [i for i, st in enumerate(self.slots)
+                      if st is not None and st.get("state") == "prefill"]
+        if prefilling:
+            # interleaved admission: one prompt chunk per tick, so long
+            # prompts trickle in BETWEEN decode steps instead of blocking
+            # the in-flight streams (single-rank path)
+            progressed = True
+            self._slot_prefill_chunk(prefilling[0])
+        if any(st is not None and st.get("state") != "prefill"
+               for st in self.slots):
             progressed = True
             if ex.world > 1:
                 # multi-rank: broadcast the step command; emit synchronously
@@ -507,6 +516,17 @@ class ShardRuntime:
         self._row_sampler.set_row(si, cfg)
         ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
         ex.last_tops = None
+        chunk = int(os.environ.get("DNET_PREFILL_CHUNK", "2048"))
+        if ex.world == 1 and int(tokens.shape[-1]) > chunk:
+            # long prompt: admit in "prefill" state — the tick loop feeds
+            # one chunk per iteration between decode steps
+            self.slots[si] = {"state": "prefill", "tokens": tokens,
+                              "p0": 0, "cfg": cfg, "nonce": nonce,
+                              "produced": 0, "max_tokens": max_tokens,
+                              "stop_ids": set(stop_ids)}
+            log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d "
+                     "(chunked)", si, nonce[:18], int(tokens.shape[-1]))
+            return
         logits = ex.prefill_slot(si, tokens)
         t0_t = torch.zeros(1, dtype=torch.int64, device=self._comm_device())
         if ex.is_last:
@@ -528,6 +548,38 @@ class ShardRuntime:
         log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
                  nonce[:18], int(tokens.shape[-1]))
 
+    def _slot_prefill_chunk(self, si: int) -> None:
+        """Feed one position chunk of a prefilling slot's prompt; on the
+        final chunk, sample the first token and flip the slot active."""
+        ex = self.executor
+        st = self.slots[si]
+        chunk = int(os.environ.get("DNET_PREFILL_CHUNK", "2048"))
+        toks = st["tokens"]
+        T = int(toks.shape[-1])
+        p0 = st["p0"]
+        p1 = min(p0 + chunk, T)
+        kvslot = ex.kvs[0].slot(si)
+        h = ex.model.embed_tokens(
+            toks.view(1, T)[:, p0:p1].to(ex.device)).clone()
+        ex.model.prefill_window(h, ex.my_layers, kvslot, p0)
+        st["p0"] = p1
+        if p1 < T:
+            return
+        kvslot.pos.fill_(T)
+        logits = ex.model.normalize_project(h[:, -1].contiguous())
+        from ..core.sampler import Sampler
+        tok, _, _ = Sampler(st["cfg"]).sample(logits.float())
+        t0 = int(tok[0])
+        ex.tokbuf[0][si] = t0
+        st["produced"] = 1
+        st["state"] = "active"
+        st.pop("tokens")
+        done = t0 in st["stop_ids"] or st["max_tokens"] <= 1
+        if ex.is_last:
+            self._emit_token(st["nonce"], t0, finished=done)
+        if done:
+            self.slots[si] = None
+
     def _slot_step_exec(self) -> None:
         """Collective decode step for all slots (multi-rank path, emits
         synchronously): hops + compute, sample on the last stage, token
@@ -541,10 +593,10 @@ class ShardRuntime:
             import torch.distributed as dist
             dist.broadcast(ex.tokbuf[0], src=(ex.stages - 1) * ex.tp)
         ex.kvs[0].pos.add_(1)
-        idle = [i for i, st in enumerate(self.slots) if st is None]
-        if idle:
-            ex.kvs[0].pos[torch.tensor(idle, dtype=torch.long,
-                                       device=ex.device)] = 0
+        park = [i for i, st in enumerate(self.slots) if st is None]
+        if park:
+            ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
+                                       device=ex.device)] = ex.smax - 1
         self._slot_emit(ex.tokbuf[0],
                         [i for i, st in enumerate(self.slots)
                          if st is not None])
@@ -559,13 +611,16 @@ class ShardRuntime:
         toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
         ex.tokbuf[0].copy_(toks_t)
         ex.kvs[0].pos.add_(1)
-        # park idle slots at 0 so their dummy writes stay in range
-        idle = [i for i, st in enumerate(self.slots) if st is None]
-        if idle:
-            ex.kvs[0].pos[torch.tensor(idle, dtype=torch.long,
-                                       device=ex.device)] = 0
+        # park idle/prefilling slots at the last row: the dummy append
+        # stays in range and that row is rewritten by a real token before
+        # any active slot ever attends to it
+        park = [i for i, st in enumerate(self.slots)
+                if st is None or st.get("state") == "prefill"]
+        if park:
+            ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
+                                       device=ex.device)] = ex.smax - 1
         return toks_t, [i for i, st in enumerate(self.slots)
-                        if st is not None]
+                        if st is not None and st.get("state") != "prefill"]
 
     def _slot_emit(self, toks_t, active) -> None:
         toks = toks_t.tolist()   # syncs; overlaps the already-launched step

@@ -583,3 +583,67 @@ def test_decode_never_exceeds_kv_capacity():
                     break
         assert len(emitted["x"]) == 6    # smax 16 - prompt 10
         rt._unload()
+
+
+def test_interleaved_chunked_admission(monkeypatch):
+    """A long prompt trickles in between decode steps (DNET_PREFILL_CHUNK)
+    without perturbing the in-flight stream, and its tokens equal the
+    unchunked run."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    def load(cap_dict):
+        rt = ShardRuntime("probe")
+        rt._load(ShardLoadModelRequest(
+            model_path="tiny", model_name="tiny", total_layers=4,
+            layers=[0, 1, 2, 3], rank=0, world_size=1, max_batch=2,
+            max_seq=64))
+
+        class Cap:
+            def send(self, frame):
+                cap_dict.setdefault(frame["nonce"], []).append(
+                    frame["token_id"])
+
+            def close(self):
+                pass
+
+        rt._callback = Cap()
+        return rt
+
+    short = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    longp = torch.arange(1, 13, dtype=torch.int32).numpy().tobytes()
+
+    def run_all(rt):
+        for _ in range(40):
+            rt._slots_tick()
+            if (all(s is None for s in rt.slots) and rt._pending is None
+                    and rt.infer_q.empty()):
+                break
+
+    # solo baselines (unchunked)
+    solo: dict = {}
+    rt = load(solo)
+    rt.infer_q.put({"nonce": "B", "tokens": short, "prompt_len": 8,
+                    "max_tokens": 8, "stop_ids": [], "params": {}})
+    run_all(rt)
+    rt._unload()
+    rt = load(solo)
+    rt.infer_q.put({"nonce": "A", "tokens": longp, "prompt_len": 12,
+                    "max_tokens": 5, "stop_ids": [], "params": {}})
+    run_all(rt)
+    rt._unload()
+
+    # interleaved: B decoding, A's 12-token prompt arrives in 4-token chunks
+    monkeypatch.setenv("DNET_PREFILL_CHUNK", "4")
+    mixed: dict = {}
+    rt = load(mixed)
+    rt.infer_q.put({"nonce": "B", "tokens": short, "prompt_len": 8,
+                    "max_tokens": 8, "stop_ids": [], "params": {}})
+    rt._slots_tick()            # B admitted + starts decoding
+    rt._slots_tick()
+    rt.infer_q.put({"nonce": "A", "tokens": longp, "prompt_len": 12,
+                    "max_tokens": 5, "stop_ids": [], "params": {}})
+    run_all(rt)
+    rt._unload()
+    assert mixed["B"] == solo["B"]          # undisturbed by A's admission
+    assert mixed["A"] == solo["A"]          # chunked == unchunked
