@@ -130,18 +130,21 @@ class LayerWeights:
     experts_down: Optional[list] = None
     sinks: Optional[torch.Tensor] = None    # gpt-oss attention sinks [Hq]
 
-    _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm")
-    _LINEAR_FIELDS = ("qkv", "o", "gateup", "down")
+    _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm", "sinks")
+    _LINEAR_FIELDS = ("qkv", "o", "gateup", "down", "router",
+                      "shared_gateup", "shared_down", "shared_gate")
 
     def to_tensor_dict(self) -> dict:
-        """Flat name->tensor dict (weight-cache slot format)."""
+        """Flat name->tensor dict (weight-cache slot format). MoE expert
+        banks serialize in the STACKED [E, ...] form (uniform shapes per
+        layer — the weight cache's slot template requirement)."""
         out = {}
         for f in self._TENSOR_FIELDS:
-            t = getattr(self, f)
+            t = getattr(self, f, None)
             if t is not None:
                 out[f] = t
         for f in self._LINEAR_FIELDS:
-            l = getattr(self, f)
+            l = getattr(self, f, None)
             if l is None:
                 continue
             out[f + ".w"] = l.w
@@ -149,6 +152,14 @@ class LayerWeights:
                 out[f + ".scales"] = l.scales
             if l.bias is not None:
                 out[f + ".bias"] = l.bias
+        if getattr(self, "experts_gateup", None):
+            from .moe import stack_experts  # lazy: avoids import cycle
+            st = stack_experts(self, list(range(len(self.experts_gateup))))
+            for key, name in (("gw", "experts.gw"), ("gs", "experts.gs"),
+                              ("gb", "experts.gb"), ("dw", "experts.dw"),
+                              ("ds", "experts.ds"), ("db", "experts.db")):
+                if st[key] is not None:
+                    out[name] = st[key]
         return out
 
     @classmethod
@@ -157,11 +168,37 @@ class LayerWeights:
         for f in cls._TENSOR_FIELDS:
             if f in d:
                 setattr(lw, f, d[f])
+        def _mk(w, bias, scales, pk):
+            l = Linear(w, bias, scales, group, pk)
+            if scales is not None:
+                l.bits = 4 if w.dtype == torch.uint8 else 8
+            return l
+
         for f in cls._LINEAR_FIELDS:
             if f + ".w" in d:
-                setattr(lw, f, Linear(d[f + ".w"], d.get(f + ".bias"),
-                                      d.get(f + ".scales"), group,
-                                      packed and (f + ".scales") in d))
+                setattr(lw, f, _mk(d[f + ".w"], d.get(f + ".bias"),
+                                   d.get(f + ".scales"),
+                                   packed and (f + ".scales") in d))
+        if "experts.gw" in d:
+            gw, dw = d["experts.gw"], d["experts.dw"]
+            gs, gb = d.get("experts.gs"), d.get("experts.gb")
+            ds, db = d.get("experts.ds"), d.get("experts.db")
+            E = gw.shape[0]
+            q = gs is not None
+            lw.experts_gateup = [
+                _mk(gw[e], None if gb is None else gb[e],
+                    None if gs is None else gs[e], packed and q)
+                for e in range(E)]
+            lw.experts_down = [
+                _mk(dw[e], None if db is None else db[e],
+                    None if ds is None else ds[e], packed and q)
+                for e in range(E)]
+            lw.experts_stacked = {
+                "local": torch.arange(E, device=gw.device),
+                "gw": gw, "gs": gs, "gb": gb, "dw": dw, "ds": ds, "db": db,
+                "group": group, "packed": packed and q}
+            lw.gateup = None
+            lw.down = None
         return lw
 
     def nbytes(self) -> int:
@@ -536,9 +573,10 @@ class RingModel:
         return KVCache(cfg, self.layer_ids, batch, smax, self.device,
                        kv_bits=self.kv_bits)
 
-    def _attn_params(self, lid: int):
+    def _attn_params(self, lid: int, lw=None):
         """(window, sinks) for layer lid — overridden by sliding-window /
-        sink models (gpt-oss)."""
+        sink models (gpt-oss). ``lw`` is the BOUND layer weights (works
+        under the offload weight provider too)."""
         return self.cfg.sliding_window or 0, None
 
     def _qk_norm(self, q, k, lw):
@@ -572,7 +610,7 @@ class RingModel:
             if c.qk_norm:
                 self._qk_norm(q, k, lw)
             li = kv.local[lid]
-            window, sinks = self._attn_params(lid)
+            window, sinks = self._attn_params(lid, lw)
             if self.cp_size > 1:
                 # sequence-sharded KV: write locally, attend via gathered
                 # flash-decode partials (numerically = full attention)
@@ -679,7 +717,7 @@ class RingModel:
                     else:
                         kv.k[li][:, :, ls0 - r * cap:ls1 - r * cap] = ksl
                         kv.v[li][:, :, ls0 - r * cap:ls1 - r * cap] = vsl
-                window, sinks = self._attn_params(lid)
+                window, sinks = self._attn_params(lid, lw)
                 assert not window, "sliding window + CP is roadmap"
                 kfull = self._cp_gather(kv.k_deq(li))[:, :, :p0 + T]
                 vfull = self._cp_gather(kv.v_deq(li))[:, :, :p0 + T]
@@ -698,7 +736,7 @@ class RingModel:
                 else:
                     kv.k[li][:, :, p0:p0 + T] = kt
                     kv.v[li][:, :, p0:p0 + T] = vt
-                window, sinks = self._attn_params(lid)
+                window, sinks = self._attn_params(lid, lw)
                 attn = _chunked_causal_attention(
                     q.transpose(1, 2), kv.k_deq(li)[:, :, :p0 + T],
                     kv.v_deq(li)[:, :, :p0 + T], d ** -0.5, p0, window, sinks)

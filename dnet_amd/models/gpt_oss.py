@@ -31,13 +31,14 @@ class GptOssRingModel(MoERingModel):
     model_type = "gpt_oss"
     model_types = ["gpt_oss"]
 
-    def _attn_params(self, lid: int):
+    def _attn_params(self, lid: int, lw=None):
         types = self.cfg.sliding_window_pattern
         if types is not None and lid < len(types):
             sliding = types[lid] == "sliding_attention"
         else:
             sliding = lid % 2 == 0  # gpt-oss default: even layers slide
-        lw = self.layers.get(lid) if self.weight_provider is None else None
+        if lw is None:
+            lw = self.layers.get(lid)
         sinks = getattr(lw, "sinks", None) if lw is not None else None
         return (self.cfg.sliding_window if sliding else 0), sinks
 

@@ -59,15 +59,27 @@ def enable_offload(model: RingModel, residency: int,
                    prefetch_depth: int = 3) -> WeightCache:
     """Move the model's layer weights to pinned host memory and install the
     windowed weight cache as the model's weight provider. Returns the cache
-    (for stats). MoE layers are not yet offloadable."""
+    (for stats). MoE expert banks stream in their stacked [E, ...] form;
+    layers must share one tensor-shape template (true for uniform stacks —
+    deepseek's mixed dense/MoE stack is not offloadable yet)."""
     assert model.layers, "load or init weights first"
     group = model.cfg.quant.group if model.cfg.quant else 0
     packed = False
     store = PinnedLayerStore(pin=True)
+    template = None
     for lid, lw in sorted(model.layers.items()):
         if lw.qkv is not None and lw.qkv.is_quant:
             packed = lw.qkv.packed
-        store.put_layer(lid, lw.to_tensor_dict())
+        td = lw.to_tensor_dict()
+        shapes = {k: tuple(t.shape) for k, t in td.items()}
+        if template is None:
+            template = shapes
+        elif shapes != template:
+            raise ValueError(
+                "offload needs one tensor-shape template across layers "
+                f"(layer {lid} differs — mixed dense/MoE stacks like "
+                "deepseek's first_k_dense_replace are not offloadable yet)")
+        store.put_layer(lid, td)
     order = sorted(model.layers.keys())
     model.layers = {}
     if model.device.type == "cuda":

@@ -312,3 +312,52 @@ def test_row_sampler():
     rs2.set_row(0, DecodingConfig(temperature=1.0, min_p=0.9))
     for _ in range(8):
         assert int(rs2.sample(logits[:1])[0]) == 1
+
+
+@pytest.mark.parametrize("model_type", ["mixtral", "gpt_oss"])
+def test_moe_offload_matches_fit(model_type):
+    """MoE layers stream through the weight cache (stacked expert banks in
+    the slot template) and produce the fit path's exact tokens."""
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+
+    hf = dict(model_type=model_type, hidden_size=64, num_hidden_layers=4,
+              num_attention_heads=4, num_key_value_heads=2, head_dim=16,
+              vocab_size=128, intermediate_size=64, num_local_experts=4,
+              num_experts_per_tok=2, rope_theta=10000.0, rms_norm_eps=1e-5)
+    if model_type == "gpt_oss":
+        hf["sliding_window"] = 16
+        hf["attention_bias"] = True
+    cfg = ModelConfig.from_hf(hf)
+    toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
+                         generator=torch.Generator().manual_seed(5))
+
+    def run(residency):
+        ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=2, smax=32,
+                          seed=13, use_graphs=False, residency=residency)
+        first = ex.prefill(toks.clone())
+        gen = ex.decode_rounds(4)
+        return torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+    fit = run(0)
+    off = run(2)
+    assert torch.equal(fit, off)
+
+
+def test_offload_rejects_mixed_template():
+    """Mixed dense/MoE stacks (deepseek first_k_dense_replace) raise a
+    clear error instead of corrupting the slot cache."""
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+
+    hf = dict(model_type="deepseek_v2", hidden_size=64, num_hidden_layers=3,
+              num_attention_heads=4, num_key_value_heads=4, vocab_size=128,
+              intermediate_size=64, kv_lora_rank=32, qk_nope_head_dim=16,
+              qk_rope_head_dim=8, v_head_dim=16, n_routed_experts=4,
+              num_experts_per_tok=2, n_shared_experts=1,
+              moe_intermediate_size=32, first_k_dense_replace=1,
+              routed_scaling_factor=1.0, rope_theta=10000.0)
+    cfg = ModelConfig.from_hf(hf)
+    with pytest.raises(ValueError, match="template"):
+        RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=1, smax=32,
+                     seed=1, use_graphs=False, residency=2)
