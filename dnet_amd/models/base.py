@@ -130,9 +130,11 @@ class LayerWeights:
     experts_down: Optional[list] = None
     sinks: Optional[torch.Tensor] = None    # gpt-oss attention sinks [Hq]
 
-    _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm", "sinks")
+    _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm", "sinks",
+                      "q_a_norm", "kv_a_norm")
     _LINEAR_FIELDS = ("qkv", "o", "gateup", "down", "router",
-                      "shared_gateup", "shared_down", "shared_gate")
+                      "shared_gateup", "shared_down", "shared_gate",
+                      "q", "q_a", "q_b", "kv_a", "kv_b")
 
     def to_tensor_dict(self) -> dict:
         """Flat name->tensor dict (weight-cache slot format). MoE expert
@@ -386,9 +388,14 @@ class RingModel:
         self._obs_sync = obs.enabled and obs.sync_per_layer
 
     def _layer(self, lid: int) -> "LayerWeights":
+        # resident layers win (partial offload keeps non-uniform layers —
+        # e.g. deepseek's leading dense layers — pinned on the GPU)
+        lw = self.layers.get(lid)
+        if lw is not None:
+            return lw
         if self.weight_provider is not None:
             return self.weight_provider(lid)
-        return self.layers[lid]
+        raise KeyError(f"layer {lid} neither resident nor provided")
 
     # ---------- weight init / loading ----------
 

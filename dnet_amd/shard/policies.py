@@ -65,23 +65,27 @@ def enable_offload(model: RingModel, residency: int,
     assert model.layers, "load or init weights first"
     group = model.cfg.quant.group if model.cfg.quant else 0
     packed = False
+    # group layers by tensor-shape template: the LARGEST uniform group
+    # streams through the slot cache; the rest (e.g. deepseek's leading
+    # dense layers) stay resident (partial offload)
+    dicts = {lid: lw.to_tensor_dict()
+             for lid, lw in sorted(model.layers.items())}
+    groups: dict = {}
+    for lid, td in dicts.items():
+        key = tuple(sorted((k, tuple(t.shape)) for k, t in td.items()))
+        groups.setdefault(key, []).append(lid)
+    order = max(groups.values(), key=len)
+    resident = [lid for lid in dicts if lid not in set(order)]
+    if resident:
+        log.info("partial offload: %d non-uniform layer(s) stay resident "
+                 "(%s)", len(resident), resident)
     store = PinnedLayerStore(pin=True)
-    template = None
-    for lid, lw in sorted(model.layers.items()):
+    for lid in order:
+        lw = model.layers[lid]
         if lw.qkv is not None and lw.qkv.is_quant:
             packed = lw.qkv.packed
-        td = lw.to_tensor_dict()
-        shapes = {k: tuple(t.shape) for k, t in td.items()}
-        if template is None:
-            template = shapes
-        elif shapes != template:
-            raise ValueError(
-                "offload needs one tensor-shape template across layers "
-                f"(layer {lid} differs — mixed dense/MoE stacks like "
-                "deepseek's first_k_dense_replace are not offloadable yet)")
-        store.put_layer(lid, td)
-    order = sorted(model.layers.keys())
-    model.layers = {}
+        store.put_layer(lid, dicts[lid])
+    model.layers = {lid: model.layers[lid] for lid in resident}
     if model.device.type == "cuda":
         torch.cuda.empty_cache()
     cache = WeightCache(store, residency, model.device, order=order)

@@ -344,13 +344,14 @@ def test_moe_offload_matches_fit(model_type):
     assert torch.equal(fit, off)
 
 
-def test_offload_rejects_mixed_template():
-    """Mixed dense/MoE stacks (deepseek first_k_dense_replace) raise a
-    clear error instead of corrupting the slot cache."""
+def test_deepseek_partial_offload_matches_fit():
+    """Mixed dense/MoE stacks (deepseek first_k_dense_replace): the
+    uniform MoE suffix streams, the dense layer stays resident — tokens
+    match the fully-resident run."""
     from dnet_amd.models import ModelConfig
     from dnet_amd.parallel.ring import RingExecutor
 
-    hf = dict(model_type="deepseek_v2", hidden_size=64, num_hidden_layers=3,
+    hf = dict(model_type="deepseek_v2", hidden_size=64, num_hidden_layers=4,
               num_attention_heads=4, num_key_value_heads=4, vocab_size=128,
               intermediate_size=64, kv_lora_rank=32, qk_nope_head_dim=16,
               qk_rope_head_dim=8, v_head_dim=16, n_routed_experts=4,
@@ -358,6 +359,16 @@ def test_offload_rejects_mixed_template():
               moe_intermediate_size=32, first_k_dense_replace=1,
               routed_scaling_factor=1.0, rope_theta=10000.0)
     cfg = ModelConfig.from_hf(hf)
-    with pytest.raises(ValueError, match="template"):
-        RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=1, smax=32,
-                     seed=1, use_graphs=False, residency=2)
+    toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
+                         generator=torch.Generator().manual_seed(5))
+
+    def run(residency):
+        ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=2, smax=32,
+                          seed=13, use_graphs=False, residency=residency)
+        first = ex.prefill(toks.clone())
+        gen = ex.decode_rounds(4)
+        return torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+    fit = run(0)
+    off = run(2)
+    assert torch.equal(fit, off)
