@@ -314,7 +314,8 @@ def test_row_sampler():
         assert int(rs2.sample(logits[:1])[0]) == 1
 
 
-@pytest.mark.parametrize("model_type", ["mixtral", "gpt_oss"])
+@pytest.mark.parametrize("model_type", ["mixtral", "gpt_oss",
+                                        "qwen2_moe"])
 def test_moe_offload_matches_fit(model_type):
     """MoE layers stream through the weight cache (stacked expert banks in
     the slot template) and produce the fit path's exact tokens."""
@@ -328,6 +329,11 @@ def test_moe_offload_matches_fit(model_type):
     if model_type == "gpt_oss":
         hf["sliding_window"] = 16
         hf["attention_bias"] = True
+    if model_type == "qwen2_moe":
+        hf["num_experts"] = hf.pop("num_local_experts")
+        hf["moe_intermediate_size"] = 32
+        hf["shared_expert_intermediate_size"] = 48
+        hf["norm_topk_prob"] = True
     cfg = ModelConfig.from_hf(hf)
     toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
                          generator=torch.Generator().manual_seed(5))
