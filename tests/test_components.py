@@ -314,9 +314,9 @@ def test_row_sampler():
         assert int(rs2.sample(logits[:1])[0]) == 1
 
 
-@pytest.mark.parametrize("model_type", ["mixtral", "gpt_oss",
-                                        "qwen2_moe"])
-def test_moe_offload_matches_fit(model_type):
+@pytest.mark.parametrize("model_type,qbits", [
+    ("mixtral", 0), ("mixtral", 8), ("gpt_oss", 0), ("qwen2_moe", 0)])
+def test_moe_offload_matches_fit(model_type, qbits):
     """MoE layers stream through the weight cache (stacked expert banks in
     the slot template) and produce the fit path's exact tokens."""
     from dnet_amd.models import ModelConfig
@@ -334,7 +334,9 @@ def test_moe_offload_matches_fit(model_type):
         hf["moe_intermediate_size"] = 32
         hf["shared_expert_intermediate_size"] = 48
         hf["norm_topk_prob"] = True
-    cfg = ModelConfig.from_hf(hf)
+    from dnet_amd.models import QuantConfig
+    quant = QuantConfig(8, 16) if qbits == 8 else None
+    cfg = ModelConfig.from_hf(hf, quant=quant)
     toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
                          generator=torch.Generator().manual_seed(5))
 
