@@ -67,6 +67,29 @@ class InferenceManager:
         ttfb_ms = None
         n_tokens = 0
         finish_reason = "stop"
+        # OpenAI `stop` strings: matched API-side in the detokenized text
+        # (reference ignores them silently — advisor r1). Text that could
+        # still grow into a stop string is held back until disambiguated.
+        stops = request.stop
+        stops = ([stops] if isinstance(stops, str) else list(stops or []))
+        stops = [s for s in stops if s]
+        pend = ""
+
+        def _split_pending(buf: str):
+            """(emit_now, keep, hit): earliest stop occurrence wins; else
+            hold back the longest tail that is a proper prefix of a stop."""
+            cut = min((i for i in (buf.find(s) for s in stops) if i >= 0),
+                      default=-1)
+            if cut >= 0:
+                return buf[:cut], "", True
+            hold = 0
+            for s in stops:
+                for k in range(min(len(s) - 1, len(buf)), hold, -1):
+                    if buf.endswith(s[:k]):
+                        hold = k
+                        break
+            return (buf[:-hold] if hold else buf), (buf[-hold:] if hold
+                                                    else ""), False
         try:
             await self.head_client.request({
                 "t": "infer", "nonce": nonce,
@@ -93,8 +116,12 @@ class InferenceManager:
                 if ttfb_ms is None:
                     ttfb_ms = (time.perf_counter() - t_start) * 1e3
                 is_stop = tid in self.mm.stop_ids
+                hit_stop_str = False
                 if not is_stop:
                     delta = detok.add_token(tid)
+                    if delta and stops:
+                        pend += delta
+                        delta, pend, hit_stop_str = _split_pending(pend)
                     if delta:
                         lp = None
                         if request.logprobs and "logprob" in frame:
@@ -110,12 +137,25 @@ class InferenceManager:
                             choices=[StreamChoice(
                                 delta=ChoiceDelta(content=delta),
                                 logprobs=lp)])
+                if hit_stop_str:
+                    # stop the ring early (frees the slot / legacy loop);
+                    # any in-flight tokens after this are dropped with the
+                    # pending queue
+                    try:
+                        await self.head_client.request(
+                            {"t": "cancel", "nonce": nonce})
+                    except Exception:
+                        log.warning("cancel send failed", exc_info=True)
+                    break
                 if frame.get("finished") or is_stop:
                     if not is_stop and n_tokens >= request.effective_max_tokens:
                         finish_reason = "length"
                     break
         finally:
             self.pending.pop(nonce, None)
+        if pend:     # held-back text that never completed a stop string
+            yield ChatChunkModel(id=nonce, model=request.model, choices=[
+                StreamChoice(delta=ChoiceDelta(content=pend))])
         total_ms = (time.perf_counter() - t_start) * 1e3
         usage = UsageModel(prompt_tokens=len(prompt_ids),
                            completion_tokens=n_tokens,

@@ -88,14 +88,25 @@ class RowSampler:
         self.top_k = torch.zeros(batch, dtype=torch.long, device=device)
         self.min_p = torch.zeros(batch, device=device)
         self._has_min_p = False    # python-side (no device sync per step)
+        # per-row RNG for seeded requests (OpenAI `seed`); None = shared
+        # default stream. Seeded rows sample via inverse-CDF with a u drawn
+        # from their own generator, so one request's stream is reproducible
+        # regardless of which other slots are active.
+        self.gens: list = [None] * batch
 
-    def set_row(self, i: int, cfg: DecodingConfig):
+    def set_row(self, i: int, cfg: DecodingConfig, seed=None):
         self.temp[i] = cfg.temperature
         self.top_p[i] = cfg.top_p
         self.top_k[i] = cfg.top_k
         self.min_p[i] = cfg.min_p
         if cfg.min_p > 0:
             self._has_min_p = True
+        g = None
+        if seed is not None:
+            g = torch.Generator(device=self.device)
+            g.manual_seed(int(seed))
+        self.gens[i] = g
+        return g
 
     def sample(self, logits: torch.Tensor) -> torch.Tensor:
         lf = logits.float()
@@ -126,5 +137,16 @@ class RowSampler:
         fallback = torch.nn.functional.one_hot(
             lf.nan_to_num(nan=0.0).argmax(dim=-1), V).float()
         probs = torch.where(ok, probs, fallback)
-        tok = torch.multinomial(probs, 1).squeeze(-1)
+        if any(g is not None for g in self.gens[:B]):
+            # inverse-CDF with per-row uniforms: seeded rows draw from
+            # their own generator, unseeded rows from the default stream
+            u = torch.rand(B, 1, device=lf.device)
+            for i, g in enumerate(self.gens[:B]):
+                if g is not None:
+                    u[i] = torch.rand(1, 1, device=lf.device, generator=g)
+            cdf = probs.cumsum(-1)
+            tok = torch.searchsorted(
+                cdf, u * cdf[..., -1:]).squeeze(-1).clamp_(0, V - 1)
+        else:
+            tok = torch.multinomial(probs, 1).squeeze(-1)
         return torch.where(self.temp <= 0.0, greedy_tok, tok)

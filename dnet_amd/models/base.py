@@ -131,7 +131,7 @@ class LayerWeights:
     sinks: Optional[torch.Tensor] = None    # gpt-oss attention sinks [Hq]
 
     _TENSOR_FIELDS = ("attn_norm", "mlp_norm", "q_norm", "k_norm", "sinks",
-                      "q_a_norm", "kv_a_norm")
+                      "q_a_norm", "kv_a_norm", "router_bias")
     _LINEAR_FIELDS = ("qkv", "o", "gateup", "down", "router",
                       "shared_gateup", "shared_down", "shared_gate",
                       "q", "q_a", "q_b", "kv_a", "kv_b")
@@ -197,6 +197,7 @@ class LayerWeights:
                 for e in range(E)]
             lw.experts_stacked = {
                 "local": torch.arange(E, device=gw.device),
+                "local_list": list(range(E)),
                 "gw": gw, "gs": gs, "gb": gb, "dw": dw, "ds": ds, "db": db,
                 "group": group, "packed": packed and q}
             lw.gateup = None

@@ -327,15 +327,14 @@ class DeepseekV2RingModel(RingModel):
             # grouped-expert kernels (graph-safe; unrouted experts skipped
             # on device) — see models/moe.py. EP: e % tp == rank; partial
             # sums reduced by the caller.
-            from .moe import stack_experts
+            from .moe import stack_experts, stack_route_weights
             local = [e for e in range(c.num_experts)
                      if self.tp_size <= 1 or e % self.tp_size == self.tp_rank]
             st = stack_experts(lw, local)
             we = torch.zeros(T, c.num_experts, dtype=torch.float32,
                              device=y.device)
             we.scatter_(1, idx, weights)
-            if self.tp_size > 1:
-                we = we.index_select(1, st["local"]).contiguous()
+            we = stack_route_weights(st, we, self.tp_size, self.tp_rank)
             act = ops.moe_gateup(y, st["gw"], st["gs"], st["gb"], we,
                                  st["group"], st["packed"], 0)
             out = ops.moe_down(act, st["dw"], st["ds"], st["db"], we,

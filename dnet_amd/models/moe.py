@@ -20,10 +20,27 @@ def stack_experts(lw: LayerWeights, local: list) -> dict:
     """Stack the given experts' weights [E_local, ...] for the grouped MoE
     kernels; the per-expert Linears become views of the stack (the sparse
     large-T path keeps working on the same storage, nothing is duplicated
-    after the originals are freed)."""
+    after the originals are freed).
+
+    The returned dict's "local"/"local_list" name the experts actually in
+    the stack IN STACK ORDER — possibly a superset of the request (the
+    offload path restores the FULL bank per slot); callers align router
+    weights with stack_route_weights, which zeroes non-owned experts under
+    EP. A cached stack that does not COVER the request (advisor r1: a
+    TP-subset cache vs a full-bank serialization, or vice versa) is
+    rebuilt from the per-expert Linears instead of being returned as-is —
+    returning the wrong subset silently double-counted experts under
+    TP+offload."""
+    local = list(local)
     st = getattr(lw, "experts_stacked", None)
     if st is not None:
-        return st
+        cached = st.get("local_list")
+        if cached is None:
+            cached = st["local"].tolist()
+            st["local_list"] = cached
+        if set(local).issubset(cached):
+            return st
+        # fall through: rebuild a stack covering the request
     gls = [lw.experts_gateup[e] for e in local]
     dls = [lw.experts_down[e] for e in local]
 
@@ -43,10 +60,30 @@ def stack_experts(lw: LayerWeights, local: list) -> dict:
             if b is not None:
                 lin.bias = b[j]
     st = {"local": torch.tensor(local, dtype=torch.long, device=gw.device),
+          "local_list": local,
           "gw": gw, "gs": gs, "gb": gb, "dw": dw, "ds": ds, "db": db,
           "group": gls[0].group, "packed": gls[0].packed}
     lw.experts_stacked = st
     return st
+
+
+def stack_route_weights(st: dict, we: torch.Tensor, tp_size: int,
+                        tp_rank: int) -> torch.Tensor:
+    """Align dense router weights [T, E_total] with the stack's expert
+    rows and zero experts this rank does not own (EP rule: e % tp == rank).
+    Device-only ops — graph-safe. The zeroed columns make the grouped
+    kernels' per-block early exit skip non-owned experts, so a full-bank
+    stack under EP reads only the owned experts' weights."""
+    if len(st["local_list"]) != we.shape[1]:
+        we = we.index_select(1, st["local"]).contiguous()
+    if tp_size > 1:
+        key = ("own", tp_size, tp_rank)
+        own = st.get(key)
+        if own is None:
+            own = ((st["local"] % tp_size) == tp_rank).to(we.dtype)
+            st[key] = own
+        we = we * own
+    return we
 
 
 class MoERingModel(RingModel):
@@ -210,8 +247,7 @@ class MoERingModel(RingModel):
             we = torch.zeros(T, c.num_experts, dtype=torch.float32,
                              device=y.device)
             we.scatter_(1, idx, weights)
-            if self.tp_size > 1:
-                we = we.index_select(1, st["local"]).contiguous()
+            we = stack_route_weights(st, we, self.tp_size, self.tp_rank)
             act = ops.moe_gateup(y, st["gw"], st["gs"], st["gb"], we,
                                  st["group"], st["packed"], self.GLU,
                                  self.GLU_ALPHA, self.GLU_LIMIT)

@@ -407,7 +407,7 @@ class RingExecutor:
         self.sampler = Sampler(cfg, generator=g)
 
     def decode_stream(self, max_tokens: int, stop_ids=(), on_token=None,
-                      mb: int = 0):
+                      mb: int = 0, should_stop=None):
         """Serving decode: one microbatch, token broadcast from the last rank
         to ALL ranks each step (so every rank stops identically on EOS), and
         ``on_token(step, tokens_tensor)`` called per step on every rank.
@@ -427,6 +427,11 @@ class RingExecutor:
 
         produced = 0
         for s in range(max_tokens - 1):
+            # external early stop (API cancel). Single-process only: with
+            # world > 1 the stop decision must stay collective (every rank
+            # breaks on the same broadcast token), so callers pass None.
+            if should_stop is not None and should_stop():
+                break
             last_r = self.rounds - 1
             if self.is_first:
                 for r in range(self.rounds):

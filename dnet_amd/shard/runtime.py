@@ -37,7 +37,7 @@ from ..utils.model_meta import get_model_metadata, load_tensors
 log = get_logger("shard")
 
 CMD_NOOP, CMD_INFER, CMD_UNLOAD, CMD_SHUTDOWN = 0, 1, 2, 3
-CMD_SLOT_ADMIT, CMD_SLOT_STEP = 4, 5
+CMD_SLOT_ADMIT, CMD_SLOT_STEP, CMD_SLOT_CANCEL = 4, 5, 6
 
 
 class SyncWireClient:
@@ -73,6 +73,7 @@ class ShardRuntime:
         self.infer_q: "queue.Queue[dict]" = queue.Queue()
         self._callback: Optional[SyncWireClient] = None
         self.slots: Optional[list] = None   # continuous batching (world==1)
+        self._cancelled: dict = {}          # nonce -> ts, cancelled by API
         self._stop = threading.Event()
         self.status = "idle"
         self.last_error = ""
@@ -99,6 +100,13 @@ class ShardRuntime:
         log.debug("infer queued nonce=%s prompt_len=%s", frame.get("nonce"),
                  frame.get("prompt_len"))
         self.infer_q.put(frame)
+
+    def submit_cancel(self, nonce: str) -> None:
+        """API-side early stop (user stop strings matched in the detok
+        text): frees the request's slot in slots mode; stops the legacy
+        single-rank decode loop at the next token. Entries expire after
+        60 s so cancels for already-finished requests don't accumulate."""
+        self._cancelled[nonce] = time.monotonic()
 
     def shutdown(self) -> None:
         self._stop.set()
@@ -186,6 +194,11 @@ class ShardRuntime:
         if self.slots is not None:
             from ..core.sampler import RowSampler
             self._row_sampler = RowSampler(req.max_batch, device=ex.device)
+            # per-slot generation counter: bumped on every admit so a
+            # pending (pipelined) emit can detect that its slot was freed
+            # and re-admitted and must not deliver the stale token under
+            # the new request's nonce
+            self._slot_gen = [0] * req.max_batch
         self.load_req = req
         self.model_name = req.model_name or req.model_path
         # one emitter per ring: the grp-rank-0 member of the last stage
@@ -287,6 +300,8 @@ class ShardRuntime:
                     self._slot_admit_follower(cmd)
                 elif cmd[0] == CMD_SLOT_STEP:
                     self._slot_step_exec()
+                elif cmd[0] == CMD_SLOT_CANCEL:
+                    self.slots[int(cmd[1])] = None
                 elif cmd[0] == CMD_UNLOAD:
                     self._unload()
 
@@ -400,7 +415,11 @@ class ShardRuntime:
 
         self._emit_s = 0.0
         t0 = time.perf_counter()
-        n = ex.decode_stream(max_tokens, stop_ids=stop_ids, on_token=on_token)
+        stopper = ((lambda: nonce in self._cancelled)
+                   if ex.world == 1 else None)
+        n = ex.decode_stream(max_tokens, stop_ids=stop_ids, on_token=on_token,
+                             should_stop=stopper)
+        self._cancelled.pop(nonce, None)
         dt = time.perf_counter() - t0
         log.info("[PROFILE][DECODE] nonce=%s tokens=%d ms=%.1f tok_s=%.1f "
                  "emit_ms=%.1f", nonce[:18], n, dt * 1e3, n / max(dt, 1e-9),
@@ -415,6 +434,26 @@ class ShardRuntime:
         (requests serialize there); this keeps N single-stream requests at
         ~single-stream latency each."""
         ex = self.executor
+        # Emit the pipelined step launched LAST tick before admitting:
+        # emission may free slots (stop/max_tokens), and admitting first
+        # would re-populate a slot the pending active list still names —
+        # its stale token would then be delivered under the new nonce.
+        if self._pending is not None:
+            self._slot_emit(*self._pending)
+            self._pending = None
+        if self._cancelled:
+            for i, st in enumerate(self.slots):
+                if st is not None and st.get("nonce") in self._cancelled:
+                    self._cancelled.pop(st["nonce"], None)
+                    if ex.world > 1:
+                        self._broadcast_cmd(CMD_SLOT_CANCEL, i)
+                    self.slots[i] = None
+                    log.info("[PROFILE][SLOT] cancel slot=%d nonce=%s", i,
+                             st.get("nonce", "")[:18])
+            now = time.monotonic()
+            for n, ts in list(self._cancelled.items()):
+                if now - ts > 60.0:
+                    del self._cancelled[n]
         active = any(s is not None for s in self.slots)
         block = not active
         progressed = False
@@ -449,15 +488,10 @@ class ShardRuntime:
                 self._slot_step_exec()
             else:
                 # single rank: pipeline — launch step n+1 (device-side deps
-                # only), THEN emit step n's tokens so the host sync overlaps
-                # the next step's GPU work
-                launched = self._slot_step_launch()
-                if self._pending is not None:
-                    self._slot_emit(*self._pending)
-                self._pending = launched
-        elif self._pending is not None:
-            self._slot_emit(*self._pending)
-            self._pending = None
+                # only); its tokens are emitted at the TOP of the next tick
+                # (before admission) so the host sync overlaps this step's
+                # GPU work without racing slot reuse
+                self._pending = self._slot_step_launch()
         if not progressed and ex.world > 1:
             self._broadcast_cmd(CMD_NOOP)
 
@@ -477,19 +511,21 @@ class ShardRuntime:
         max_tokens = int(frame.get("max_tokens", 128))
         nonce = frame.get("nonce", "")
         nonce_ids = list(nonce.encode("utf-8"))[:64]
+        seed = p.get("seed")
         if ex.world > 1:
             import torch.distributed as dist
             self._broadcast_cmd(CMD_SLOT_ADMIT, si, T, max_tokens,
                                 len(stop_ids), len(nonce_ids),
                                 cfg.temperature, cfg.top_p, cfg.top_k,
-                                cfg.min_p)
+                                cfg.min_p,
+                                -1.0 if seed is None else int(seed))
             payload = torch.cat([
                 tokens.flatten().to(self._comm_device()),
                 torch.tensor(stop_ids + nonce_ids, dtype=torch.int64,
                              device=self._comm_device())])
             dist.broadcast(payload, src=0)
         self._slot_admit_exec(si, tokens.view(-1), max_tokens, stop_ids,
-                              nonce, cfg)
+                              nonce, cfg, seed)
 
     def _slot_admit_follower(self, cmd) -> None:
         import torch.distributed as dist
@@ -497,23 +533,26 @@ class ShardRuntime:
         n_stop, n_nonce = int(cmd[4]), int(cmd[5])
         cfg = DecodingConfig(temperature=cmd[6], top_p=cmd[7],
                              top_k=int(cmd[8]), min_p=cmd[9])
+        seed = None if cmd[10] < 0 else int(cmd[10])
         payload = torch.zeros(T + n_stop + n_nonce, dtype=torch.int64,
                               device=self._comm_device())
         dist.broadcast(payload, src=0)
         tokens = payload[:T].cpu()
         stop_ids = payload[T:T + n_stop].tolist()
         nonce = bytes(payload[T + n_stop:].tolist()).decode("utf-8", "replace")
-        self._slot_admit_exec(si, tokens, max_tokens, stop_ids, nonce, cfg)
+        self._slot_admit_exec(si, tokens, max_tokens, stop_ids, nonce, cfg,
+                              seed)
 
     def _slot_admit_exec(self, si, tokens, max_tokens, stop_ids, nonce,
-                         cfg) -> None:
+                         cfg, seed=None) -> None:
         """Collective slot admit (every rank): single-slot ring prefill,
         sample on the last stage, broadcast the first token, register the
         slot state identically everywhere."""
         ex = self.executor
+        self._slot_gen[si] += 1
         max_tokens = max(1, min(max_tokens,
                                 ex.smax - int(tokens.shape[-1])))
-        self._row_sampler.set_row(si, cfg)
+        gen = self._row_sampler.set_row(si, cfg, seed)
         ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
         ex.last_tops = None
         chunk = int(os.environ.get("DNET_PREFILL_CHUNK", "2048"))
@@ -531,7 +570,7 @@ class ShardRuntime:
         t0_t = torch.zeros(1, dtype=torch.int64, device=self._comm_device())
         if ex.is_last:
             from ..core.sampler import Sampler
-            tok, _, _ = Sampler(cfg).sample(logits.float())
+            tok, _, _ = Sampler(cfg, generator=gen).sample(logits.float())
             t0_t[0] = int(tok[0])
         if ex.world > 1:
             import torch.distributed as dist
@@ -568,7 +607,9 @@ class ShardRuntime:
         kvslot.pos.fill_(T)
         logits = ex.model.normalize_project(h[:, -1].contiguous())
         from ..core.sampler import Sampler
-        tok, _, _ = Sampler(st["cfg"]).sample(logits.float())
+        tok, _, _ = Sampler(st["cfg"],
+                            generator=self._row_sampler.gens[si]
+                            ).sample(logits.float())
         t0 = int(tok[0])
         ex.tokbuf[0][si] = t0
         st["produced"] = 1
@@ -598,10 +639,12 @@ class ShardRuntime:
             ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
                                        device=ex.device)] = ex.smax - 1
         self._slot_emit(ex.tokbuf[0],
-                        [i for i, st in enumerate(self.slots)
+                        [(i, self._slot_gen[i])
+                         for i, st in enumerate(self.slots)
                          if st is not None])
 
-    _pending = None   # (device tokens, active slot list) of the in-flight step
+    _pending = None   # (device tokens, [(slot, gen)]) of the in-flight step
+    _slot_gen: list = []   # per-slot admit generation (see _slot_emit)
 
     def _slot_step_launch(self):
         """Enqueue one decode step for the whole batch — device ops only,
@@ -619,14 +662,17 @@ class ShardRuntime:
         if park:
             ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
                                        device=ex.device)] = ex.smax - 1
-        return toks_t, [i for i, st in enumerate(self.slots)
+        return toks_t, [(i, self._slot_gen[i])
+                        for i, st in enumerate(self.slots)
                         if st is not None and st.get("state") != "prefill"]
 
     def _slot_emit(self, toks_t, active) -> None:
         toks = toks_t.tolist()   # syncs; overlaps the already-launched step
-        for i in active:
+        for i, gen in active:
             st = self.slots[i]
-            if st is None:       # freed by an earlier emit (stop lag)
+            if st is None or self._slot_gen[i] != gen:
+                # freed since launch (stop lag), or freed AND re-admitted —
+                # either way this row's token belongs to the old request
                 continue
             t = int(toks[i])
             st["produced"] += 1

@@ -128,6 +128,9 @@ async def wire_handler_factory(rt: ShardRuntime):
             return {"t": "ack"}
         if t == "reset":
             return {"t": "ack"}
+        if t == "cancel":
+            rt.submit_cancel(frame.get("nonce", ""))
+            return {"t": "ack"}
         if t == "ping":
             return {"t": "pong"}
         if t == "latency_probe":

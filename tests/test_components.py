@@ -352,10 +352,13 @@ def test_moe_offload_matches_fit(model_type, qbits):
     assert torch.equal(fit, off)
 
 
-def test_deepseek_partial_offload_matches_fit():
+@pytest.mark.parametrize("scoring", ["softmax", "sigmoid"])
+def test_deepseek_partial_offload_matches_fit(scoring):
     """Mixed dense/MoE stacks (deepseek first_k_dense_replace): the
     uniform MoE suffix streams, the dense layer stays resident — tokens
-    match the fully-resident run."""
+    match the fully-resident run. The sigmoid case (deepseek-v3 noaux_tc)
+    additionally covers router_bias surviving the weight-cache round-trip
+    (advisor r1: router_bias was missing from _TENSOR_FIELDS)."""
     from dnet_amd.models import ModelConfig
     from dnet_amd.parallel.ring import RingExecutor
 
@@ -366,6 +369,9 @@ def test_deepseek_partial_offload_matches_fit():
               num_experts_per_tok=2, n_shared_experts=1,
               moe_intermediate_size=32, first_k_dense_replace=1,
               routed_scaling_factor=1.0, rope_theta=10000.0)
+    if scoring == "sigmoid":
+        hf.update(scoring_func="sigmoid", topk_method="noaux_tc",
+                  n_group=2, topk_group=1, norm_topk_prob=True)
     cfg = ModelConfig.from_hf(hf)
     toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
                          generator=torch.Generator().manual_seed(5))

@@ -139,6 +139,66 @@ def test_inference_manager_token_flow():
     assert resp.metrics["tokens_generated"] == 3
 
 
+def test_stop_strings_truncate_and_cancel():
+    """OpenAI `stop` strings (advisor r1): matched API-side in detok text,
+    stream truncated BEFORE the stop string, a cancel frame goes to the
+    head, and partial-prefix holdback text is flushed when no stop hits."""
+    from dnet_amd.api.inference import InferenceManager
+    from dnet_amd.api.tokenizer import ByteTokenizer
+
+    class MM:
+        tokenizer = ByteTokenizer(512)
+        stop_ids = [ByteTokenizer(512).EOS]
+
+    sent = []
+
+    class FakeHead:
+        def __init__(self, im, text):
+            self.im = im
+            self.text = text
+
+        async def request(self, frame):
+            sent.append(frame)
+            if frame.get("t") != "infer":
+                return {"t": "ack"}
+
+            async def feed():
+                nonce = frame["nonce"]
+                for ch in self.text:
+                    self.im.resolve_token({"t": "token", "nonce": nonce,
+                                           "token_id": int(ch),
+                                           "finished": False})
+                self.im.resolve_token({"t": "token", "nonce": nonce,
+                                       "token_id": MM.stop_ids[0],
+                                       "finished": True})
+            asyncio.get_event_loop().create_task(feed())
+            return {"t": "ack"}
+
+    async def run(text, stop):
+        im = InferenceManager(MM(), token_timeout_s=10)
+        im.head_client = FakeHead(im, text)
+        im.callback_addr = "127.0.0.1:1"
+        req = ChatRequestModel(model="tiny-random", stop=stop,
+                               messages=[{"role": "user", "content": "hi"}])
+        return await im.chat_completions(req)
+
+    # stop string mid-stream: truncate before it, cancel sent
+    resp = asyncio.run(run(b"hello STOP world", ["STOP"]))
+    assert resp.choices[0].message.content == "hello "
+    assert resp.choices[0].finish_reason == "stop"
+    assert any(f.get("t") == "cancel" for f in sent)
+
+    # stop prefix that never completes: held-back text is flushed
+    sent.clear()
+    resp = asyncio.run(run(b"abcST", ["STOP"]))
+    assert resp.choices[0].message.content == "abcST"
+    assert not any(f.get("t") == "cancel" for f in sent)
+
+    # string form of `stop`
+    resp = asyncio.run(run(b"xxByy", "B"))
+    assert resp.choices[0].message.content == "xx"
+
+
 def test_sampler_modes():
     torch.manual_seed(0)
     logits = torch.tensor([[0.1, 5.0, 0.2, 0.3]])
@@ -418,6 +478,108 @@ def test_slot_stop_id_mid_stream():
     assert len(st) == 3 and st[-1][1]            # stopped at the stop id
     assert len(emitted["runs"]) == 8             # unaffected neighbor
     rt._unload()
+
+
+def test_slot_reuse_no_stale_token():
+    """Regression (advisor r1, high): a slot freed by a stop and re-admitted
+    must not receive the old request's post-stop token from the pipelined
+    pending step. The re-admitted request's stream must equal a solo run."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    emitted: dict[str, list] = {}
+
+    class Cap:
+        def send(self, frame):
+            emitted.setdefault(frame["nonce"], []).append(frame["token_id"])
+
+        def close(self):
+            pass
+
+    def mk(name):
+        rt = ShardRuntime(name)
+        rt._load(ShardLoadModelRequest(
+            model_path="tiny", model_name="tiny", total_layers=4,
+            layers=[0, 1, 2, 3], rank=0, world_size=1,
+            max_batch=2, max_seq=64))
+        rt._callback = Cap()
+        return rt
+
+    pa = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    pc = torch.arange(5, 13, dtype=torch.int32).numpy().tobytes()
+
+    # solo reference for c's prompt
+    rt0 = mk("solo")
+    rt0.infer_q.put({"nonce": "c-solo", "tokens": pc, "prompt_len": 8,
+                     "max_tokens": 6, "stop_ids": [], "params": {}})
+    for _ in range(15):
+        rt0._slots_tick()
+    rt0._unload()
+
+    # a (short) + b (long) fill both slots; c queues and takes a's slot
+    # the moment it frees — exactly the reuse window of the old bug
+    rt = mk("probe")
+    rt.infer_q.put({"nonce": "a", "tokens": pa, "prompt_len": 8,
+                    "max_tokens": 2, "stop_ids": [], "params": {}})
+    rt.infer_q.put({"nonce": "b", "tokens": pa, "prompt_len": 8,
+                    "max_tokens": 12, "stop_ids": [], "params": {}})
+    rt.infer_q.put({"nonce": "c", "tokens": pc, "prompt_len": 8,
+                    "max_tokens": 6, "stop_ids": [], "params": {}})
+    for _ in range(40):
+        rt._slots_tick()
+        if (all(s is None for s in rt.slots) and rt._pending is None
+                and len(emitted.get("c", [])) >= 6):
+            break
+    assert len(emitted["a"]) == 2
+    assert len(emitted["b"]) == 12
+    assert emitted["c"] == emitted["c-solo"]
+    rt._unload()
+
+
+def test_slot_seeded_request_reproducible():
+    """Per-request `seed` in slots mode (advisor r1): the same seeded
+    sampled request produces identical tokens whether it runs alone or
+    next to another (unseeded, sampled) stream."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    def run(extra_neighbor):
+        emitted: dict[str, list] = {}
+
+        class Cap:
+            def send(self, frame):
+                emitted.setdefault(frame["nonce"], []).append(
+                    frame["token_id"])
+
+            def close(self):
+                pass
+
+        rt = ShardRuntime("probe")
+        rt._load(ShardLoadModelRequest(
+            model_path="tiny", model_name="tiny", total_layers=4,
+            layers=[0, 1, 2, 3], rank=0, world_size=1,
+            max_batch=2, max_seq=64))
+        rt._callback = Cap()
+        prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+        rt.infer_q.put({"nonce": "seeded", "tokens": prompt, "prompt_len": 8,
+                        "max_tokens": 8, "stop_ids": [],
+                        "params": {"temperature": 0.9, "seed": 1234}})
+        if extra_neighbor:
+            rt.infer_q.put({"nonce": "other", "tokens": prompt,
+                            "prompt_len": 8, "max_tokens": 6, "stop_ids": [],
+                            "params": {"temperature": 1.3}})
+        for _ in range(30):
+            rt._slots_tick()
+            if len(emitted.get("seeded", [])) >= 8:
+                break
+        rt._unload()
+        return emitted["seeded"]
+
+    solo = run(False)
+    torch.manual_seed(999)   # perturb the default stream between runs
+    with_neighbor = run(True)
+    assert len(solo) == 8
+    assert solo == with_neighbor
 
 
 def test_recover_excludes_dead_shard(monkeypatch):
