@@ -230,6 +230,285 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Streamed int8/int4 schedule (round-2).
+//
+// The generic kernel above is pair-rate bound on quantized weights: PMC
+// showed SQ_WAIT 69% with int8 wall time == bf16 wall time at every shape
+// (profiles/r01_gemm_ubench.md) — hipcc keeps only ~2 weight loads in
+// flight and serializes the x staging as load;vmcnt(0);ds_write per chunk.
+// Every compiler-level restructure was measured and rejected; this is the
+// hand-scheduled counted-vmcnt plan: all VMEM in the hot loop is either a
+// glds DMA or issued by inline asm, and drained with counted s_waitcnt
+// vmcnt(N) (never 0 against the weight stream — guide "what does break
+// it"). Per 256-k x-tile:
+//
+//   raw s_barrier (lgkmcnt only — previous tile's consumers done)
+//   issue x staging as global_load_lds DMA (no VGPR round trip) [W_i, S_i]
+//   issue NEXT tile's weight prefetch W_{i+1}            [W_i, S_i, W_{i+1}]
+//   s_waitcnt vmcnt(WP)  -- drains W_i + S_i, W_{i+1} stays IN FLIGHT
+//   raw s_barrier
+//   dequant + MFMA from the W_i registers (no waits at all)
+//
+// so the weight stream is continuously one full tile deep (4-6 VMEM ops,
+// 64-96 B per lane) across both barriers and the staging drain, and the
+// staging costs ~1 round trip + 1 wait per tile. vmcnt retires in issue
+// order, which is what makes the counted wait exact.
+//
+// glds writes LDS at (wave-uniform base + lane*16), so the LDS x image is
+// UNPADDED (16*MT rows x 32 16B-chunks); bank conflicts on the 16-lane
+// fragment reads are killed by an XOR swizzle applied on the SOURCE
+// address instead of row padding (guide §5 rule 21): source chunk c lands
+// at image chunk c ^ (row & 15) — the 16 reader lanes (rows 0..15, same
+// c) then touch 16 different 16B slots = all 64 LDS banks.
+// ---------------------------------------------------------------------------
+
+using i32x4 = __attribute__((ext_vector_type(4))) int;
+
+// NSC = scales per 256-k tile = 256/G (G compile-time via dispatch).
+template <int QBITS, int NSC>
+struct WSet {
+  i32x4 w[QBITS == 8 ? 4 : 2];
+  unsigned s[NSC];
+};
+
+// Issue one tile's weight+scale loads. wb already includes the lane's
+// row offset (n_w*K + woff) plus the tile's k byte offset; sb points at
+// the tile's first scale.
+template <int QBITS, int NSC>
+__device__ __forceinline__ void wset_issue(WSet<QBITS, NSC>& o,
+                                           const void* wb, const short* sb) {
+#pragma unroll
+  for (int u = 0; u < (QBITS == 8 ? 4 : 2); ++u)
+    asm volatile("global_load_dwordx4 %0, %1, off offset:%c2"
+                 : "=v"(o.w[u])
+                 : "v"(wb), "i"(u * 64));
+#pragma unroll
+  for (int c = 0; c < NSC; ++c)
+    asm volatile("global_load_ushort %0, %1, off offset:%c2"
+                 : "=v"(o.s[c])
+                 : "v"(sb), "i"(c * 2));
+}
+
+template <int QBITS, int NSC>
+__device__ __forceinline__ void wset_issue_at(WSet<QBITS, NSC>& o,
+                                              const void* wrow,
+                                              const short* srow, int k) {
+  constexpr int GS = 256 / NSC;
+  if (QBITS == 8)
+    wset_issue<QBITS, NSC>(o, (const int8_t*)wrow + k, srow + k / GS);
+  else
+    wset_issue<QBITS, NSC>(o, (const uint8_t*)wrow + k / 2, srow + k / GS);
+}
+
+template <int QBITS, int MT, int NSC>
+__device__ __forceinline__ void stream_tile(
+    WSet<QBITS, NSC>& cur, WSet<QBITS, NSC>& nxt, const short* __restrict__ x,
+    short* x_lds, const void* wrow, const short* srow, const int k0,
+    const int knext, const int M, const int K, f32x4 (&acc)[MT][2],
+    const int row, const int ks) {
+  constexpr int NSG = MT * 2;  // glds per wave (1 KB each)
+  constexpr int TPW = (QBITS == 8) ? 4 : 2;
+  constexpr int WP = TPW + NSC;
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  asm volatile("s_barrier" ::: "memory");
+  const int wavei = (int)threadIdx.x >> 6, lanei = (int)threadIdx.x & 63;
+#pragma unroll
+  for (int g = 0; g < NSG; ++g) {
+    const int L = (wavei * NSG + g) * 64 + lanei;  // image chunk
+    const int r = L >> 5;
+    const int cs = (L & 31) ^ (r & 15);            // source chunk (swizzle)
+    const short* ga = &x[(int64_t)min(r, M - 1) * K + k0 + cs * 8];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)ga,
+        (__attribute__((address_space(3))) void*)&x_lds[(wavei * NSG + g) *
+                                                        64 * 8],
+        16, 0, 0);
+  }
+  wset_issue_at<QBITS, NSC>(nxt, wrow, srow, knext);
+  // drain staging DMA + W_i (older); W_{i+1}'s WP ops stay in flight
+  asm volatile("s_waitcnt vmcnt(%c0)" ::"i"(WP) : "memory");
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  asm volatile("s_barrier" ::: "memory");
+  // opaque tie: without it the IR optimizer byte-extracts the whole W set
+  // right after its issue asm (one tile EARLY), keeping 16 unpacked bytes
+  // live through the tile -> scratch spills in the hot loop
+#pragma unroll
+  for (int u = 0; u < TPW; ++u) asm("" : "+v"(cur.w[u]));
+#pragma unroll
+  for (int c = 0; c < NSC; ++c) asm("" : "+v"(cur.s[c]));
+  const int ln4 = ks >> 3;  // lane>>4
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    bf16x8 b0, b1;
+    if (QBITS == 8) {
+      const int8_t* q8 = reinterpret_cast<const int8_t*>(&cur.w[u]);
+      const float sv = bits2f((short)cur.s[u * NSC / 4]);
+      b0 = deq8(q8, sv);
+      b1 = deq8(q8 + 8, sv);
+    } else {
+      const uint8_t* q4 =
+          reinterpret_cast<const uint8_t*>(&cur.w[u / 2]) + (u & 1) * 8;
+      const float sv = bits2f((short)cur.s[(u / 2) * NSC / 2]);
+      b0 = deq4(q4, sv);
+      b1 = deq4(q4 + 4, sv);
+    }
+    const int c0 = (u * 8 + ln4) ^ row;      // swizzled image chunks
+    const int c1 = (u * 8 + 4 + ln4) ^ row;
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+      const short* rb = &x_lds[(t * 16 + row) * 256];
+      const bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&rb[c0 * 8]);
+      const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(&rb[c1 * 8]);
+      acc[t][0] =
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[t][0], 0, 0, 0);
+      acc[t][1] =
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[t][1], 0, 0, 0);
+    }
+  }
+}
+
+// MINW: min waves per EU hint. MT=4 at the default (4, matching the
+// 4-blocks/CU LDS cap) spills ~26 regs; MINW=3 trades 16->12 waves/CU for
+// a spill-free 160-VGPR allocation. Both instantiated; DNET_GEMM_MT4OCC
+// picks (measured on hardware, see profiles/).
+template <int QBITS, int MT, int NSC, int MINW>
+__global__ __launch_bounds__(256, MINW) void gemm_m16_stream_kernel(
+    const short* __restrict__ x, const void* __restrict__ w,
+    const short* __restrict__ scales, const short* __restrict__ bias,
+    short* __restrict__ out, float* __restrict__ out_f32, const int M,
+    const int K, const int N, const int G, const int splitk) {
+  constexpr int XT = 256;
+  __shared__ short x_lds[16 * MT * XT];  // unpadded (glds image, swizzled)
+  const int wave = threadIdx.x / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int n0 = (blockIdx.x * 4 + wave) * 16;
+  const int row = lane & 15;
+  const int ks = (lane >> 4) * 8;
+  const int n_w = min(n0 + row, N - 1);
+
+  const int pairs = K / 64;
+  int p_begin, p_end;
+  if (QBITS == 4) {
+    const int quads = K / 128;
+    const int qq = quads / splitk;
+    p_begin = blockIdx.y * qq * 2;
+    p_end = (blockIdx.y == splitk - 1) ? quads * 2 : p_begin + qq * 2;
+  } else {
+    const int pp = pairs / splitk;
+    p_begin = blockIdx.y * pp;
+    p_end = (blockIdx.y == splitk - 1) ? pairs : p_begin + pp;
+  }
+  const int woff = (lane >> 4) * 16;
+
+  f32x4 acc[MT][2];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+#pragma unroll
+    for (int u = 0; u < 2; ++u) acc[t][u] = {0.f, 0.f, 0.f, 0.f};
+
+  const void* wrow =
+      QBITS == 8
+          ? (const void*)((const int8_t*)w + (int64_t)n_w * K + woff)
+          : (const void*)((const uint8_t*)w + (int64_t)n_w * (K / 2) + woff);
+  const short* srow = scales + (int64_t)n_w * (K / G);
+
+  const int kbeg = p_begin * 64, kend = p_end * 64;
+  const int kfull = kbeg + ((kend - kbeg) / XT) * XT;
+  int k0 = kbeg;
+  WSet<QBITS, NSC> wa, wb2;
+  if (k0 < kfull) {
+    wset_issue_at<QBITS, NSC>(wa, wrow, srow, k0);
+    while (true) {
+      int kn = (k0 + XT < kfull) ? k0 + XT : k0;
+      stream_tile<QBITS, MT, NSC>(wa, wb2, x, x_lds, wrow, srow, k0, kn, M, K,
+                                  acc, row, ks);
+      k0 += XT;
+      if (k0 >= kfull) break;
+      kn = (k0 + XT < kfull) ? k0 + XT : k0;
+      stream_tile<QBITS, MT, NSC>(wb2, wa, x, x_lds, wrow, srow, k0, kn, M, K,
+                                  acc, row, ks);
+      k0 += XT;
+      if (k0 >= kfull) break;
+    }
+  }
+
+  // tail (< XT k): generic serial staging + pair loop, executed by the
+  // whole block (condition uniform). Any still-in-flight prefetch regs
+  // are simply never read; the compiler's own waits for the loads below
+  // are counted against a longer FIFO, which is stricter, never wrong.
+  const int ln4 = ks >> 3;
+  for (; k0 < kend; k0 += XT) {
+    const int tk = kend - k0;  // < XT, multiple of 64
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    asm volatile("s_barrier" ::: "memory");
+    for (int idx = threadIdx.x; idx < 16 * MT * (tk / 8); idx += 256) {
+      const int r = idx / (tk / 8);
+      const int vec = idx % (tk / 8);
+      // same swizzled image as the glds path (rows are full 32-chunk
+      // strides even when tk < XT)
+      *reinterpret_cast<short8*>(&x_lds[(r * 32 + (vec ^ (r & 15))) * 8]) =
+          *reinterpret_cast<const short8*>(
+              &x[(int64_t)min(r, M - 1) * K + k0 + vec * 8]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    asm volatile("s_barrier" ::: "memory");
+    for (int pl = 0; pl < tk / 64; ++pl) {
+      const int p = (k0 / 64) + pl;
+      bf16x8 b0, b1;
+      if (QBITS == 8) {
+        const int8_t* wq = (const int8_t*)wrow;
+        const int4 wv = *reinterpret_cast<const int4*>(&wq[p * 64]);
+        const int8_t* q8 = reinterpret_cast<const int8_t*>(&wv);
+        const float sv = bits2f(srow[(p * 64) / G]);
+        b0 = deq8(q8, sv);
+        b1 = deq8(q8 + 8, sv);
+      } else {
+        const int quad = p / 2;
+        const int half = p & 1;
+        const uint8_t* wq = (const uint8_t*)wrow;
+        const int2 wv =
+            *reinterpret_cast<const int2*>(&wq[quad * 64 + half * 8]);
+        const uint8_t* q4 = reinterpret_cast<const uint8_t*>(&wv);
+        const float sv = bits2f(srow[(quad * 128) / G]);
+        b0 = deq4(q4, sv);
+        b1 = deq4(q4 + 4, sv);
+      }
+      const int c0 = (pl * 8 + ln4) ^ row;
+      const int c1 = (pl * 8 + 4 + ln4) ^ row;
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        const short* rb = &x_lds[(t * 16 + row) * 256];
+        const bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&rb[c0 * 8]);
+        const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(&rb[c1 * 8]);
+        acc[t][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[t][0],
+                                                            0, 0, 0);
+        acc[t][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[t][1],
+                                                            0, 0, 0);
+      }
+    }
+  }
+
+  const int n = n0 + (lane & 15);
+  if (n >= N) return;
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    const f32x4 a2 = acc[t][0] + acc[t][1];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = t * 16 + (lane >> 4) * 4 + r;
+      if (m >= M) continue;
+      if (splitk > 1) {
+        atomicAdd(out_f32 + (int64_t)m * N + n, a2[r]);
+      } else {
+        float v = a2[r];
+        if (bias != nullptr) v += bits2f(bias[n]);
+        out[(int64_t)m * N + n] = f2bits(v);
+      }
+    }
+  }
+}
+
 __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
                                         const short* __restrict__ bias,
                                         short* __restrict__ out,
@@ -278,20 +557,49 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   hipLaunchKernelGGL((gemm_m16_kernel<QQ, TT>), grid, dim3(256), 0, stream, \
                      xp, w.data_ptr(), sp, bp1, op, fp, M, (int)K, (int)N,  \
                      group, sk)
+  static const bool mt4occ3 = []() {
+    const char* e = getenv("DNET_GEMM_MT4OCC");
+    return e == nullptr || e[0] == '3';  // default: spill-free 3-wave MT4
+  }();
+#define LAUNCH_STREAM(QQ, TT, NSC, MW)                                    \
+  hipLaunchKernelGGL((gemm_m16_stream_kernel<QQ, TT, NSC, MW>), grid,     \
+                     dim3(256), 0, stream, xp, w.data_ptr(), sp, bp1, op, \
+                     fp, M, (int)K, (int)N, group, sk)
+#define LAUNCH_STREAM_G(QQ, TT)                                           \
+  do {                                                                    \
+    if (TT == 4 && mt4occ3) {                                             \
+      if (group == 64) LAUNCH_STREAM(QQ, TT, 4, 3);                       \
+      else if (group == 128) LAUNCH_STREAM(QQ, TT, 2, 3);                 \
+      else LAUNCH_STREAM(QQ, TT, 1, 3);                                   \
+    } else {                                                              \
+      if (group == 64) LAUNCH_STREAM(QQ, TT, 4, 4);                       \
+      else if (group == 128) LAUNCH_STREAM(QQ, TT, 2, 4);                 \
+      else LAUNCH_STREAM(QQ, TT, 1, 4);                                   \
+    }                                                                     \
+  } while (0)
+  // the streamed counted-vmcnt schedule needs a compile-time scale count
+  // per 256-k tile; other group sizes fall back to the generic kernel
+  const bool can_stream =
+      (bits == 8 && (group == 64 || group == 128 || group == 256)) ||
+      (bits == 4 && (group == 128 || group == 256));
+  static const bool nostream = getenv("DNET_GEMM_NOSTREAM") != nullptr;
+  const bool use_stream = can_stream && !nostream;
   if (bits == 8) {
-    if (M > 32) LAUNCH(8, 4);
-    else if (M > 16) LAUNCH(8, 2);
-    else LAUNCH(8, 1);
+    if (M > 32) { if (use_stream) LAUNCH_STREAM_G(8, 4); else LAUNCH(8, 4); }
+    else if (M > 16) { if (use_stream) LAUNCH_STREAM_G(8, 2); else LAUNCH(8, 2); }
+    else { if (use_stream) LAUNCH_STREAM_G(8, 1); else LAUNCH(8, 1); }
   } else if (bits == 4) {
-    if (M > 32) LAUNCH(4, 4);
-    else if (M > 16) LAUNCH(4, 2);
-    else LAUNCH(4, 1);
+    if (M > 32) { if (use_stream) LAUNCH_STREAM_G(4, 4); else LAUNCH(4, 4); }
+    else if (M > 16) { if (use_stream) LAUNCH_STREAM_G(4, 2); else LAUNCH(4, 2); }
+    else { if (use_stream) LAUNCH_STREAM_G(4, 1); else LAUNCH(4, 1); }
   } else {
     if (M > 32) LAUNCH(16, 4);
     else if (M > 16) LAUNCH(16, 2);
     else LAUNCH(16, 1);
   }
 #undef LAUNCH
+#undef LAUNCH_STREAM
+#undef LAUNCH_STREAM_G
   if (sk > 1) {
     const int64_t total = (int64_t)M * N;
     const int cgrid = (int)std::min<int64_t>((total + 255) / 256, 2048);

@@ -35,6 +35,11 @@ def bench(fn, reps=50):
 
 
 def main():
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--m", type=str, default="1,16,32,64")
+    args = ap.parse_args()
+    ms = [int(v) for v in args.m.split(",")]
     dev = "cuda:0"
     nat = _native()
     for name, N, K in SHAPES:
@@ -43,7 +48,7 @@ def main():
         qp = ops.pack_int8_mfma(q)
         wbytes_i8 = q.numel() + scales.numel() * 2
         wbytes_bf = wf.numel() * 2
-        for M in (1, 16, 32, 64):
+        for M in ms:
             x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
             out = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
             scratch = ops._get_scratch(dev)

@@ -511,3 +511,46 @@ def test_rope_append_wpos():
     assert kc[0, :, 8].abs().sum() > 0      # 40-32 = 8
     assert kc[1, :, 1].abs().sum() > 0      # 33-32 = 1
     assert kc[2].abs().sum() == 0           # 20-32 < 0: other rank's row
+
+
+@pytest.mark.parametrize("m", [1, 8, 16, 24, 33, 64])
+@pytest.mark.parametrize("k,group", [
+    (320, 64),     # 1 full 256-k tile + 1 tail pair
+    (1088, 64),    # 4 full tiles + tail
+    (2048, 128),   # full tiles only
+    (1280, 256),   # NSC=1
+    (704, 128),    # odd tail (2 full tiles + 3 tail pairs)
+])
+def test_gemm_stream_matches_generic_int8(m, k, group):
+    """The streamed counted-vmcnt schedule must produce bit-identical
+    results to the generic kernel for every MT tier, scale-group width,
+    and split/tail geometry (incl. the partial last tile served by the
+    serial tail path)."""
+
+    torch.manual_seed(m * 1000 + k)
+    n = 1000
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=_dev())
+    wf = torch.randn(n, k, dtype=torch.bfloat16, device=_dev()) / 30
+    q, scales = ops.quantize_int8(wf, group)
+    qp = ops.pack_int8_mfma(q)
+    bias = torch.randn(n, dtype=torch.bfloat16, device=_dev())
+    out_s = ops.gemv_int8(x, qp, scales, group, bias, packed=True)
+    out_ref = ref.gemv_int8(x.cpu(), q.cpu(), scales.cpu(), group, bias.cpu())
+    assert torch.allclose(out_s.float().cpu(), out_ref.float(), atol=6e-2,
+                          rtol=3e-2), \
+        (out_s.float().cpu() - out_ref.float()).abs().max()
+
+
+@pytest.mark.parametrize("m", [4, 40, 64])
+@pytest.mark.parametrize("k,group", [(384, 128), (1152, 128), (1536, 256)])
+def test_gemm_stream_matches_generic_int4(m, k, group):
+    torch.manual_seed(m + k)
+    n = 768
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=_dev())
+    wf = torch.randn(n, k, dtype=torch.bfloat16, device=_dev()) / 30
+    q4, scales = ops.quantize_int4(wf.cpu(), group)
+    q4p = ops.pack_int4_mfma(q4).to(_dev())
+    out_s = ops.gemv_int4(x, q4p, scales.to(_dev()), group, None, packed=True)
+    out_ref = ops.ref.gemv_int4(x.cpu(), q4, scales, group)
+    assert torch.allclose(out_s.float().cpu(), out_ref.float(), atol=8e-2,
+                          rtol=4e-2)
