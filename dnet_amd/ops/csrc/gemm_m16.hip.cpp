@@ -524,7 +524,13 @@ __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
 static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems) {
   // Target ~1024 blocks (~4 blocks / 16 waves per CU) so HBM latency is
   // covered by wave overlap; each split keeps >= 8 K-chunk-pairs of work.
+  static const int forced = []() {
+    const char* e = getenv("DNET_GEMM_SPLITK");  // debug/bench override
+    return e ? atoi(e) : 0;
+  }();
   const int blocks = (int)((N + 63) / 64);
+  if (forced > 0)
+    return (M * N > scratch_elems || (K / 64) < forced) ? 1 : forced;
   int sk = 1;
   while (sk < 32 && blocks * sk < 1024 && (K / 64) / (sk * 2) >= 8) sk *= 2;
   if (sk > 1 && M * N > scratch_elems) sk = 1;
