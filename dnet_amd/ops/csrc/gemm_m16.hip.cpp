@@ -236,24 +236,26 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
 // The generic kernel above is pair-rate bound on quantized weights: PMC
 // showed SQ_WAIT 69% with int8 wall time == bf16 wall time at every shape
 // (profiles/r01_gemm_ubench.md) — hipcc keeps only ~2 weight loads in
-// flight and serializes the x staging as load;vmcnt(0);ds_write per chunk.
-// Every compiler-level restructure was measured and rejected; this is the
-// hand-scheduled counted-vmcnt plan: all VMEM in the hot loop is either a
-// glds DMA or issued by inline asm, and drained with counted s_waitcnt
-// vmcnt(N) (never 0 against the weight stream — guide "what does break
-// it"). Per 256-k x-tile:
+// flight and serializes the x staging as load;vmcnt(0);ds_write per chunk
+// (one full round trip per 16-byte chunk). Per 256-k x-tile this schedule
+// instead takes ONE combined round trip:
 //
 //   raw s_barrier (lgkmcnt only — previous tile's consumers done)
-//   issue x staging as global_load_lds DMA (no VGPR round trip) [W_i, S_i]
-//   issue NEXT tile's weight prefetch W_{i+1}            [W_i, S_i, W_{i+1}]
-//   s_waitcnt vmcnt(WP)  -- drains W_i + S_i, W_{i+1} stays IN FLIGHT
+//   issue x staging as global_load_lds DMA (no VGPR round trip)
+//   ONE asm block: issue the tile's weight+scale loads, s_waitcnt vmcnt(0)
 //   raw s_barrier
-//   dequant + MFMA from the W_i registers (no waits at all)
+//   dequant + MFMA from registers (no VMEM waits at all)
 //
-// so the weight stream is continuously one full tile deep (4-6 VMEM ops,
-// 64-96 B per lane) across both barriers and the staging drain, and the
-// staging costs ~1 round trip + 1 wait per tile. vmcnt retires in issue
-// order, which is what makes the counted wait exact.
+// The weight loads and the staging DMA are all in flight together
+// (10-14 VMEM ops, 5-9 KB per lane-group), so their latencies overlap in
+// one drain; wave overlap (4 blocks x 4 waves / CU) covers the per-wave
+// consume phase. Every async VMEM value is produced AND waited for
+// INSIDE a single asm statement: hipcc models asm outputs as ready when
+// the statement ends, so an async result that escapes a statement before
+// its s_waitcnt can be copied/rematerialized from an in-flight register
+// (measured: garbage numerics and corrupted-address page faults from the
+// earlier cross-tile ping-pong prefetch). Outputs are early-clobber so
+// the in-flight destinations can never alias the address operands.
 //
 // glds writes LDS at (wave-uniform base + lane*16), so the LDS x image is
 // UNPADDED (16*MT rows x 32 16B-chunks); bank conflicts on the 16-lane
@@ -265,91 +267,122 @@ __global__ void gemm_m16_kernel(const short* __restrict__ x,
 
 using i32x4 = __attribute__((ext_vector_type(4))) int;
 
-// NSC = scales per 256-k tile = 256/G (G compile-time via dispatch).
+// NSC = scales per 256-k tile = 256/G (G compile-time via dispatch; the
+// host streams only G in {64,128}, and requires the split start k to be
+// G-aligned so the in-tile scale index u*NSC/4 is exact).
 template <int QBITS, int NSC>
 struct WSet {
   i32x4 w[QBITS == 8 ? 4 : 2];
   unsigned s[NSC];
 };
 
-// Issue one tile's weight+scale loads. wb already includes the lane's
-// row offset (n_w*K + woff) plus the tile's k byte offset; sb points at
-// the tile's first scale.
+// Load one tile's weights+scales and drain: single asm statement (see
+// header comment). wb already includes the lane's row offset
+// (n_w*K + woff) plus the tile's k byte offset; sb points at the tile's
+// first scale.
 template <int QBITS, int NSC>
-__device__ __forceinline__ void wset_issue(WSet<QBITS, NSC>& o,
-                                           const void* wb, const short* sb) {
-#pragma unroll
-  for (int u = 0; u < (QBITS == 8 ? 4 : 2); ++u)
-    asm volatile("global_load_dwordx4 %0, %1, off offset:%c2"
-                 : "=v"(o.w[u])
-                 : "v"(wb), "i"(u * 64));
-#pragma unroll
-  for (int c = 0; c < NSC; ++c)
-    asm volatile("global_load_ushort %0, %1, off offset:%c2"
-                 : "=v"(o.s[c])
-                 : "v"(sb), "i"(c * 2));
+__device__ __forceinline__ void wset_load(WSet<QBITS, NSC>& o,
+                                          const void* wb, const short* sb) {
+  if constexpr (QBITS == 8 && NSC == 2) {
+    asm volatile(
+        "global_load_dwordx4 %0, %6, off\n\t"
+        "global_load_dwordx4 %1, %6, off offset:64\n\t"
+        "global_load_dwordx4 %2, %6, off offset:128\n\t"
+        "global_load_dwordx4 %3, %6, off offset:192\n\t"
+        "global_load_ushort %4, %7, off\n\t"
+        "global_load_ushort %5, %7, off offset:2\n\t"
+        "s_waitcnt vmcnt(0)"
+        : "=&v"(o.w[0]), "=&v"(o.w[1]), "=&v"(o.w[2]), "=&v"(o.w[3]),
+          "=&v"(o.s[0]), "=&v"(o.s[1])
+        : "v"(wb), "v"(sb)
+        : "memory");
+  } else if constexpr (QBITS == 8 && NSC == 4) {
+    asm volatile(
+        "global_load_dwordx4 %0, %8, off\n\t"
+        "global_load_dwordx4 %1, %8, off offset:64\n\t"
+        "global_load_dwordx4 %2, %8, off offset:128\n\t"
+        "global_load_dwordx4 %3, %8, off offset:192\n\t"
+        "global_load_ushort %4, %9, off\n\t"
+        "global_load_ushort %5, %9, off offset:2\n\t"
+        "global_load_ushort %6, %9, off offset:4\n\t"
+        "global_load_ushort %7, %9, off offset:6\n\t"
+        "s_waitcnt vmcnt(0)"
+        : "=&v"(o.w[0]), "=&v"(o.w[1]), "=&v"(o.w[2]), "=&v"(o.w[3]),
+          "=&v"(o.s[0]), "=&v"(o.s[1]), "=&v"(o.s[2]), "=&v"(o.s[3])
+        : "v"(wb), "v"(sb)
+        : "memory");
+  } else {
+    static_assert(QBITS == 4 && NSC == 2, "unsupported stream variant");
+    asm volatile(
+        "global_load_dwordx4 %0, %4, off\n\t"
+        "global_load_dwordx4 %1, %4, off offset:64\n\t"
+        "global_load_ushort %2, %5, off\n\t"
+        "global_load_ushort %3, %5, off offset:2\n\t"
+        "s_waitcnt vmcnt(0)"
+        : "=&v"(o.w[0]), "=&v"(o.w[1]), "=&v"(o.s[0]), "=&v"(o.s[1])
+        : "v"(wb), "v"(sb)
+        : "memory");
+  }
 }
 
-template <int QBITS, int NSC>
-__device__ __forceinline__ void wset_issue_at(WSet<QBITS, NSC>& o,
-                                              const void* wrow,
-                                              const short* srow, int k) {
-  constexpr int GS = 256 / NSC;
-  if (QBITS == 8)
-    wset_issue<QBITS, NSC>(o, (const int8_t*)wrow + k, srow + k / GS);
-  else
-    wset_issue<QBITS, NSC>(o, (const uint8_t*)wrow + k / 2, srow + k / GS);
-}
-
-template <int QBITS, int MT, int NSC>
+template <int QBITS, int MT, int NSC, bool GLDS>
 __device__ __forceinline__ void stream_tile(
-    WSet<QBITS, NSC>& cur, WSet<QBITS, NSC>& nxt, const short* __restrict__ x,
-    short* x_lds, const void* wrow, const short* srow, const int k0,
-    const int knext, const int M, const int K, f32x4 (&acc)[MT][2],
-    const int row, const int ks) {
+    const short* __restrict__ x, short* x_lds, const void* wrow,
+    const short* srow, const int k0, const int M, const int K,
+    f32x4 (&acc)[MT][2], const int row, const int ks) {
   constexpr int NSG = MT * 2;  // glds per wave (1 KB each)
-  constexpr int TPW = (QBITS == 8) ? 4 : 2;
-  constexpr int WP = TPW + NSC;
+  constexpr int GS = 256 / NSC;
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   asm volatile("s_barrier" ::: "memory");
   const int wavei = (int)threadIdx.x >> 6, lanei = (int)threadIdx.x & 63;
+  if constexpr (GLDS) {
 #pragma unroll
-  for (int g = 0; g < NSG; ++g) {
-    const int L = (wavei * NSG + g) * 64 + lanei;  // image chunk
-    const int r = L >> 5;
-    const int cs = (L & 31) ^ (r & 15);            // source chunk (swizzle)
-    const short* ga = &x[(int64_t)min(r, M - 1) * K + k0 + cs * 8];
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)ga,
-        (__attribute__((address_space(3))) void*)&x_lds[(wavei * NSG + g) *
-                                                        64 * 8],
-        16, 0, 0);
+    for (int g = 0; g < NSG; ++g) {
+      const int L = (wavei * NSG + g) * 64 + lanei;  // image chunk
+      const int r = L >> 5;
+      const int cs = (L & 31) ^ (r & 15);            // source chunk (swizzle)
+      const short* ga = &x[(int64_t)min(r, M - 1) * K + k0 + cs * 8];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)&x_lds[(wavei * NSG + g) *
+                                                          64 * 8],
+          16, 0, 0);
+    }
+  } else {
+    // debug/fallback: same swizzled image via plain loads + ds_write
+#pragma unroll
+    for (int g = 0; g < NSG; ++g) {
+      const int L = (wavei * NSG + g) * 64 + lanei;
+      const int r = L >> 5;
+      const int cs = (L & 31) ^ (r & 15);
+      *reinterpret_cast<short8*>(&x_lds[L * 8]) =
+          *reinterpret_cast<const short8*>(
+              &x[(int64_t)min(r, M - 1) * K + k0 + cs * 8]);
+    }
   }
-  wset_issue_at<QBITS, NSC>(nxt, wrow, srow, knext);
-  // drain staging DMA + W_i (older); W_{i+1}'s WP ops stay in flight
-  asm volatile("s_waitcnt vmcnt(%c0)" ::"i"(WP) : "memory");
+  WSet<QBITS, NSC> w;
+  if (QBITS == 8)
+    wset_load<QBITS, NSC>(w, (const int8_t*)wrow + k0, srow + k0 / GS);
+  else
+    wset_load<QBITS, NSC>(w, (const uint8_t*)wrow + k0 / 2, srow + k0 / GS);
+  // the wset_load drain (vmcnt 0) also covered the staging DMA
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   asm volatile("s_barrier" ::: "memory");
-  // opaque tie: without it the IR optimizer byte-extracts the whole W set
-  // right after its issue asm (one tile EARLY), keeping 16 unpacked bytes
-  // live through the tile -> scratch spills in the hot loop
-#pragma unroll
-  for (int u = 0; u < TPW; ++u) asm("" : "+v"(cur.w[u]));
-#pragma unroll
-  for (int c = 0; c < NSC; ++c) asm("" : "+v"(cur.s[c]));
+  if constexpr (GLDS)  // DS ops bound-check against m0 on gfx9-family
+    asm volatile("s_mov_b32 m0, -1" ::: "memory");
   const int ln4 = ks >> 3;  // lane>>4
 #pragma unroll
   for (int u = 0; u < 4; ++u) {
     bf16x8 b0, b1;
     if (QBITS == 8) {
-      const int8_t* q8 = reinterpret_cast<const int8_t*>(&cur.w[u]);
-      const float sv = bits2f((short)cur.s[u * NSC / 4]);
+      const int8_t* q8 = reinterpret_cast<const int8_t*>(&w.w[u]);
+      const float sv = bits2f((short)w.s[u * NSC / 4]);
       b0 = deq8(q8, sv);
       b1 = deq8(q8 + 8, sv);
     } else {
       const uint8_t* q4 =
-          reinterpret_cast<const uint8_t*>(&cur.w[u / 2]) + (u & 1) * 8;
-      const float sv = bits2f((short)cur.s[(u / 2) * NSC / 2]);
+          reinterpret_cast<const uint8_t*>(&w.w[u / 2]) + (u & 1) * 8;
+      const float sv = bits2f((short)w.s[(u / 2) * NSC / 2]);
       b0 = deq4(q4, sv);
       b1 = deq4(q4 + 4, sv);
     }
@@ -368,11 +401,8 @@ __device__ __forceinline__ void stream_tile(
   }
 }
 
-// MINW: min waves per EU hint. MT=4 at the default (4, matching the
-// 4-blocks/CU LDS cap) spills ~26 regs; MINW=3 trades 16->12 waves/CU for
-// a spill-free 160-VGPR allocation. Both instantiated; DNET_GEMM_MT4OCC
-// picks (measured on hardware, see profiles/).
-template <int QBITS, int MT, int NSC, int MINW>
+// MINW: min waves per EU hint (occupancy/VGPR trade, see profiles/).
+template <int QBITS, int MT, int NSC, int MINW, bool GLDS = true>
 __global__ __launch_bounds__(256, MINW) void gemm_m16_stream_kernel(
     const short* __restrict__ x, const void* __restrict__ w,
     const short* __restrict__ scales, const short* __restrict__ bias,
@@ -416,27 +446,12 @@ __global__ __launch_bounds__(256, MINW) void gemm_m16_stream_kernel(
   const int kbeg = p_begin * 64, kend = p_end * 64;
   const int kfull = kbeg + ((kend - kbeg) / XT) * XT;
   int k0 = kbeg;
-  WSet<QBITS, NSC> wa, wb2;
-  if (k0 < kfull) {
-    wset_issue_at<QBITS, NSC>(wa, wrow, srow, k0);
-    while (true) {
-      int kn = (k0 + XT < kfull) ? k0 + XT : k0;
-      stream_tile<QBITS, MT, NSC>(wa, wb2, x, x_lds, wrow, srow, k0, kn, M, K,
-                                  acc, row, ks);
-      k0 += XT;
-      if (k0 >= kfull) break;
-      kn = (k0 + XT < kfull) ? k0 + XT : k0;
-      stream_tile<QBITS, MT, NSC>(wb2, wa, x, x_lds, wrow, srow, k0, kn, M, K,
-                                  acc, row, ks);
-      k0 += XT;
-      if (k0 >= kfull) break;
-    }
-  }
+  for (; k0 < kfull; k0 += XT)
+    stream_tile<QBITS, MT, NSC, GLDS>(x, x_lds, wrow, srow, k0, M, K, acc,
+                                      row, ks);
 
   // tail (< XT k): generic serial staging + pair loop, executed by the
-  // whole block (condition uniform). Any still-in-flight prefetch regs
-  // are simply never read; the compiler's own waits for the loads below
-  // are counted against a longer FIFO, which is stricter, never wrong.
+  // whole block (condition uniform)
   const int ln4 = ks >> 3;
   for (; k0 < kend; k0 += XT) {
     const int tk = kend - k0;  // < XT, multiple of 64
@@ -567,37 +582,54 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
     const char* e = getenv("DNET_GEMM_MT4OCC");
     return e == nullptr || e[0] == '3';  // default: spill-free 3-wave MT4
   }();
-#define LAUNCH_STREAM(QQ, TT, NSC, MW)                                    \
-  hipLaunchKernelGGL((gemm_m16_stream_kernel<QQ, TT, NSC, MW>), grid,     \
-                     dim3(256), 0, stream, xp, w.data_ptr(), sp, bp1, op, \
-                     fp, M, (int)K, (int)N, group, sk)
-#define LAUNCH_STREAM_G(QQ, TT)                                           \
+  static const bool noglds = getenv("DNET_GEMM_NOGLDS") != nullptr;
+#define LAUNCH_STREAM(QQ, TT, NSC, MW)                                      \
+  do {                                                                      \
+    if (noglds)                                                             \
+      hipLaunchKernelGGL((gemm_m16_stream_kernel<QQ, TT, NSC, MW, false>), \
+                         grid, dim3(256), 0, stream, xp, w.data_ptr(), sp,  \
+                         bp1, op, fp, M, (int)K, (int)N, group, sk);        \
+    else                                                                    \
+      hipLaunchKernelGGL((gemm_m16_stream_kernel<QQ, TT, NSC, MW, true>),   \
+                         grid, dim3(256), 0, stream, xp, w.data_ptr(), sp,  \
+                         bp1, op, fp, M, (int)K, (int)N, group, sk);        \
+  } while (0)
+#define LAUNCH_STREAM_G8(TT)                                            \
   do {                                                                    \
     if (TT == 4 && mt4occ3) {                                             \
-      if (group == 64) LAUNCH_STREAM(QQ, TT, 4, 3);                       \
-      else if (group == 128) LAUNCH_STREAM(QQ, TT, 2, 3);                 \
-      else LAUNCH_STREAM(QQ, TT, 1, 3);                                   \
+      if (group == 64) LAUNCH_STREAM(8, TT, 4, 3);                        \
+      else LAUNCH_STREAM(8, TT, 2, 3);                                    \
     } else {                                                              \
-      if (group == 64) LAUNCH_STREAM(QQ, TT, 4, 4);                       \
-      else if (group == 128) LAUNCH_STREAM(QQ, TT, 2, 4);                 \
-      else LAUNCH_STREAM(QQ, TT, 1, 4);                                   \
+      if (group == 64) LAUNCH_STREAM(8, TT, 4, 4);                        \
+      else LAUNCH_STREAM(8, TT, 2, 4);                                    \
     }                                                                     \
+  } while (0)
+#define LAUNCH_STREAM_G4(TT)                                              \
+  do {                                                                    \
+    if (TT == 4 && mt4occ3) LAUNCH_STREAM(4, TT, 2, 3);                   \
+    else LAUNCH_STREAM(4, TT, 2, 4);                                      \
   } while (0)
   // the streamed counted-vmcnt schedule needs a compile-time scale count
   // per 256-k tile; other group sizes fall back to the generic kernel
+  // G must divide every tile start (k0 is any multiple of 64 under
+  // split-k partitioning, and the in-tile scale index u*NSC/4 assumes
+  // k0 % G == 0), so only G <= 128 streams; G=256 falls back
+  // additionally every split start kbeg = by*(pairs/sk)*64 must be
+  // G-aligned (the in-tile scale index assumes k0 % G == 0)
+  const bool split_aligned = sk == 1 || ((K / 64 / sk) * 64) % group == 0;
   const bool can_stream =
-      (bits == 8 && (group == 64 || group == 128 || group == 256)) ||
-      (bits == 4 && (group == 128 || group == 256));
+      split_aligned && ((bits == 8 && (group == 64 || group == 128)) ||
+                        (bits == 4 && group == 128));
   static const bool nostream = getenv("DNET_GEMM_NOSTREAM") != nullptr;
   const bool use_stream = can_stream && !nostream;
   if (bits == 8) {
-    if (M > 32) { if (use_stream) LAUNCH_STREAM_G(8, 4); else LAUNCH(8, 4); }
-    else if (M > 16) { if (use_stream) LAUNCH_STREAM_G(8, 2); else LAUNCH(8, 2); }
-    else { if (use_stream) LAUNCH_STREAM_G(8, 1); else LAUNCH(8, 1); }
+    if (M > 32) { if (use_stream) LAUNCH_STREAM_G8(4); else LAUNCH(8, 4); }
+    else if (M > 16) { if (use_stream) LAUNCH_STREAM_G8(2); else LAUNCH(8, 2); }
+    else { if (use_stream) LAUNCH_STREAM_G8(1); else LAUNCH(8, 1); }
   } else if (bits == 4) {
-    if (M > 32) { if (use_stream) LAUNCH_STREAM_G(4, 4); else LAUNCH(4, 4); }
-    else if (M > 16) { if (use_stream) LAUNCH_STREAM_G(4, 2); else LAUNCH(4, 2); }
-    else { if (use_stream) LAUNCH_STREAM_G(4, 1); else LAUNCH(4, 1); }
+    if (M > 32) { if (use_stream) LAUNCH_STREAM_G4(4); else LAUNCH(4, 4); }
+    else if (M > 16) { if (use_stream) LAUNCH_STREAM_G4(2); else LAUNCH(4, 2); }
+    else { if (use_stream) LAUNCH_STREAM_G4(1); else LAUNCH(4, 1); }
   } else {
     if (M > 32) LAUNCH(16, 4);
     else if (M > 16) LAUNCH(16, 2);
@@ -605,7 +637,8 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   }
 #undef LAUNCH
 #undef LAUNCH_STREAM
-#undef LAUNCH_STREAM_G
+#undef LAUNCH_STREAM_G8
+#undef LAUNCH_STREAM_G4
   if (sk > 1) {
     const int64_t total = (int64_t)M * N;
     const int cgrid = (int)std::min<int64_t>((total + 255) / 256, 2048);
