@@ -616,7 +616,8 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   // k0 % G == 0), so only G <= 128 streams; G=256 falls back
   // additionally every split start kbeg = by*(pairs/sk)*64 must be
   // G-aligned (the in-tile scale index assumes k0 % G == 0)
-  const bool split_aligned = sk == 1 || ((K / 64 / sk) * 64) % group == 0;
+  const bool split_aligned =
+      group > 0 && (sk == 1 || ((K / 64 / sk) * 64) % group == 0);
   const bool can_stream =
       split_aligned && ((bits == 8 && (group == 64 || group == 128)) ||
                         (bits == 4 && group == 128));
