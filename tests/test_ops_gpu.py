@@ -519,7 +519,7 @@ def test_rope_append_wpos():
     (1088, 64),    # 4 full tiles + tail
     (2048, 128),   # full tiles only
     (1280, 256),   # NSC=1
-    (704, 128),    # odd tail (2 full tiles + 3 tail pairs)
+    (1920, 128),   # odd tail (7 full tiles + 2 tail pairs)
 ])
 def test_gemm_stream_matches_generic_int8(m, k, group):
     """The streamed counted-vmcnt schedule must produce bit-identical
