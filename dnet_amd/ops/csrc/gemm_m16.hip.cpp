@@ -325,13 +325,89 @@ __device__ __forceinline__ void wset_load(WSet<QBITS, NSC>& o,
   }
 }
 
-template <int QBITS, int MT, int NSC, bool GLDS>
-__device__ __forceinline__ void stream_tile(
-    const short* __restrict__ x, short* x_lds, const void* wrow,
-    const short* srow, const int k0, const int M, const int K,
-    f32x4 (&acc)[MT][2], const int row, const int ks) {
+// 2-tile weight batch: one statement issues BOTH tiles' weights (+ the
+// first tile's staging already in flight) and drains once — halves the
+// exposed weight round trips vs per-tile loads. Values are final at the
+// statement end, so holding tile B's registers across tile A's consume
+// and tile B's staging is safe (only IN-FLIGHT values must not escape).
+template <int QBITS, int NSC>
+struct WSet2 {
+  i32x4 w[QBITS == 8 ? 8 : 4];
+  unsigned s[2 * NSC];
+};
+
+template <int QBITS, int NSC>
+__device__ __forceinline__ void wset_load2(WSet2<QBITS, NSC>& o,
+                                           const void* wb, const short* sb) {
+  if constexpr (QBITS == 8 && NSC == 2) {
+    asm volatile(
+        "global_load_dwordx4 %0, %12, off\n\t"
+        "global_load_dwordx4 %1, %12, off offset:64\n\t"
+        "global_load_dwordx4 %2, %12, off offset:128\n\t"
+        "global_load_dwordx4 %3, %12, off offset:192\n\t"
+        "global_load_dwordx4 %4, %12, off offset:256\n\t"
+        "global_load_dwordx4 %5, %12, off offset:320\n\t"
+        "global_load_dwordx4 %6, %12, off offset:384\n\t"
+        "global_load_dwordx4 %7, %12, off offset:448\n\t"
+        "global_load_ushort %8, %13, off\n\t"
+        "global_load_ushort %9, %13, off offset:2\n\t"
+        "global_load_ushort %10, %13, off offset:4\n\t"
+        "global_load_ushort %11, %13, off offset:6\n\t"
+        "s_waitcnt vmcnt(0)"
+        : "=&v"(o.w[0]), "=&v"(o.w[1]), "=&v"(o.w[2]), "=&v"(o.w[3]),
+          "=&v"(o.w[4]), "=&v"(o.w[5]), "=&v"(o.w[6]), "=&v"(o.w[7]),
+          "=&v"(o.s[0]), "=&v"(o.s[1]), "=&v"(o.s[2]), "=&v"(o.s[3])
+        : "v"(wb), "v"(sb)
+        : "memory");
+  } else if constexpr (QBITS == 8 && NSC == 4) {
+    asm volatile(
+        "global_load_dwordx4 %0, %16, off\n\t"
+        "global_load_dwordx4 %1, %16, off offset:64\n\t"
+        "global_load_dwordx4 %2, %16, off offset:128\n\t"
+        "global_load_dwordx4 %3, %16, off offset:192\n\t"
+        "global_load_dwordx4 %4, %16, off offset:256\n\t"
+        "global_load_dwordx4 %5, %16, off offset:320\n\t"
+        "global_load_dwordx4 %6, %16, off offset:384\n\t"
+        "global_load_dwordx4 %7, %16, off offset:448\n\t"
+        "global_load_ushort %8, %17, off\n\t"
+        "global_load_ushort %9, %17, off offset:2\n\t"
+        "global_load_ushort %10, %17, off offset:4\n\t"
+        "global_load_ushort %11, %17, off offset:6\n\t"
+        "global_load_ushort %12, %17, off offset:8\n\t"
+        "global_load_ushort %13, %17, off offset:10\n\t"
+        "global_load_ushort %14, %17, off offset:12\n\t"
+        "global_load_ushort %15, %17, off offset:14\n\t"
+        "s_waitcnt vmcnt(0)"
+        : "=&v"(o.w[0]), "=&v"(o.w[1]), "=&v"(o.w[2]), "=&v"(o.w[3]),
+          "=&v"(o.w[4]), "=&v"(o.w[5]), "=&v"(o.w[6]), "=&v"(o.w[7]),
+          "=&v"(o.s[0]), "=&v"(o.s[1]), "=&v"(o.s[2]), "=&v"(o.s[3]),
+          "=&v"(o.s[4]), "=&v"(o.s[5]), "=&v"(o.s[6]), "=&v"(o.s[7])
+        : "v"(wb), "v"(sb)
+        : "memory");
+  } else {
+    static_assert(QBITS == 4 && NSC == 2, "unsupported stream variant");
+    asm volatile(
+        "global_load_dwordx4 %0, %8, off\n\t"
+        "global_load_dwordx4 %1, %8, off offset:64\n\t"
+        "global_load_dwordx4 %2, %8, off offset:128\n\t"
+        "global_load_dwordx4 %3, %8, off offset:192\n\t"
+        "global_load_ushort %4, %9, off\n\t"
+        "global_load_ushort %5, %9, off offset:2\n\t"
+        "global_load_ushort %6, %9, off offset:4\n\t"
+        "global_load_ushort %7, %9, off offset:6\n\t"
+        "s_waitcnt vmcnt(0)"
+        : "=&v"(o.w[0]), "=&v"(o.w[1]), "=&v"(o.w[2]), "=&v"(o.w[3]),
+          "=&v"(o.s[0]), "=&v"(o.s[1]), "=&v"(o.s[2]), "=&v"(o.s[3])
+        : "v"(wb), "v"(sb)
+        : "memory");
+  }
+}
+
+template <int MT, bool GLDS>
+__device__ __forceinline__ void stage_tile(const short* __restrict__ x,
+                                           short* x_lds, const int k0,
+                                           const int M, const int K) {
   constexpr int NSG = MT * 2;  // glds per wave (1 KB each)
-  constexpr int GS = 256 / NSC;
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   asm volatile("s_barrier" ::: "memory");
   const int wavei = (int)threadIdx.x >> 6, lanei = (int)threadIdx.x & 63;
@@ -360,29 +436,36 @@ __device__ __forceinline__ void stream_tile(
               &x[(int64_t)min(r, M - 1) * K + k0 + cs * 8]);
     }
   }
-  WSet<QBITS, NSC> w;
-  if (QBITS == 8)
-    wset_load<QBITS, NSC>(w, (const int8_t*)wrow + k0, srow + k0 / GS);
-  else
-    wset_load<QBITS, NSC>(w, (const uint8_t*)wrow + k0 / 2, srow + k0 / GS);
-  // the wset_load drain (vmcnt 0) also covered the staging DMA
+}
+
+template <bool GLDS>
+__device__ __forceinline__ void end_stage() {
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   asm volatile("s_barrier" ::: "memory");
   if constexpr (GLDS)  // DS ops bound-check against m0 on gfx9-family
     asm volatile("s_mov_b32 m0, -1" ::: "memory");
+}
+
+// dequant + MFMA for one tile's 4 pairs from already-final registers
+template <int QBITS, int MT, int NSC>
+__device__ __forceinline__ void consume_tile(const i32x4* __restrict__ wv,
+                                             const unsigned* __restrict__ sc,
+                                             short* x_lds,
+                                             f32x4 (&acc)[MT][2],
+                                             const int row, const int ks) {
   const int ln4 = ks >> 3;  // lane>>4
 #pragma unroll
   for (int u = 0; u < 4; ++u) {
     bf16x8 b0, b1;
     if (QBITS == 8) {
-      const int8_t* q8 = reinterpret_cast<const int8_t*>(&w.w[u]);
-      const float sv = bits2f((short)w.s[u * NSC / 4]);
+      const int8_t* q8 = reinterpret_cast<const int8_t*>(&wv[u]);
+      const float sv = bits2f((short)sc[u * NSC / 4]);
       b0 = deq8(q8, sv);
       b1 = deq8(q8 + 8, sv);
     } else {
       const uint8_t* q4 =
-          reinterpret_cast<const uint8_t*>(&w.w[u / 2]) + (u & 1) * 8;
-      const float sv = bits2f((short)w.s[(u / 2) * NSC / 2]);
+          reinterpret_cast<const uint8_t*>(&wv[u / 2]) + (u & 1) * 8;
+      const float sv = bits2f((short)sc[(u / 2) * NSC / 2]);
       b0 = deq4(q4, sv);
       b1 = deq4(q4 + 4, sv);
     }
@@ -445,10 +528,37 @@ __global__ __launch_bounds__(256, MINW) void gemm_m16_stream_kernel(
 
   const int kbeg = p_begin * 64, kend = p_end * 64;
   const int kfull = kbeg + ((kend - kbeg) / XT) * XT;
+  constexpr int GS = 256 / NSC;
   int k0 = kbeg;
-  for (; k0 < kfull; k0 += XT)
-    stream_tile<QBITS, MT, NSC, GLDS>(x, x_lds, wrow, srow, k0, M, K, acc,
-                                      row, ks);
+  // 2-tile groups: one weight statement serves both tiles (one exposed
+  // round trip instead of two); tile B's staging waits alone
+  for (; k0 + 2 * XT <= kfull; k0 += 2 * XT) {
+    WSet2<QBITS, NSC> w2;
+    stage_tile<MT, GLDS>(x, x_lds, k0, M, K);
+    if (QBITS == 8)
+      wset_load2<QBITS, NSC>(w2, (const int8_t*)wrow + k0, srow + k0 / GS);
+    else
+      wset_load2<QBITS, NSC>(w2, (const uint8_t*)wrow + k0 / 2,
+                             srow + k0 / GS);
+    end_stage<GLDS>();
+    consume_tile<QBITS, MT, NSC>(&w2.w[0], &w2.s[0], x_lds, acc, row, ks);
+    stage_tile<MT, GLDS>(x, x_lds, k0 + XT, M, K);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // staging only
+    end_stage<GLDS>();
+    consume_tile<QBITS, MT, NSC>(&w2.w[QBITS == 8 ? 4 : 2], &w2.s[NSC],
+                                 x_lds, acc, row, ks);
+  }
+  for (; k0 < kfull; k0 += XT) {
+    WSet<QBITS, NSC> w1;
+    stage_tile<MT, GLDS>(x, x_lds, k0, M, K);
+    if (QBITS == 8)
+      wset_load<QBITS, NSC>(w1, (const int8_t*)wrow + k0, srow + k0 / GS);
+    else
+      wset_load<QBITS, NSC>(w1, (const uint8_t*)wrow + k0 / 2,
+                            srow + k0 / GS);
+    end_stage<GLDS>();
+    consume_tile<QBITS, MT, NSC>(&w1.w[0], &w1.s[0], x_lds, acc, row, ks);
+  }
 
   // tail (< XT k): generic serial staging + pair loop, executed by the
   // whole block (condition uniform)
@@ -536,7 +646,8 @@ __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
   }
 }
 
-static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems) {
+static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
+                       int min_pairs) {
   // Target ~1024 blocks (~4 blocks / 16 waves per CU) so HBM latency is
   // covered by wave overlap; each split keeps >= 8 K-chunk-pairs of work.
   static const int forced = []() {
@@ -547,7 +658,8 @@ static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems) {
   if (forced > 0)
     return (M * N > scratch_elems || (K / 64) < forced) ? 1 : forced;
   int sk = 1;
-  while (sk < 32 && blocks * sk < 1024 && (K / 64) / (sk * 2) >= 8) sk *= 2;
+  while (sk < 32 && blocks * sk < 1024 && (K / 64) / (sk * 2) >= min_pairs)
+    sk *= 2;
   if (sk > 1 && M * N > scratch_elems) sk = 1;
   return sk;
 }
@@ -560,7 +672,10 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   const int64_t K = x.size(1), N = w.size(0);
   auto stream = current_stream();
   const int64_t scratch_elems = scratch.has_value() ? scratch->numel() : 0;
-  const int sk = pick_splitk(N, K, M, scratch_elems);
+  // the streamed schedule tolerates short splits (one combined round
+  // trip per 4-pair tile), so quantized paths split deeper to fill the
+  // 256 CUs on small-N shapes (qkv/o run at <2 blocks/CU otherwise)
+  const int sk = pick_splitk(N, K, M, scratch_elems, bits < 16 ? 4 : 8);
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   const dim3 grid((unsigned)((N + 63) / 64), sk);
   const short* xp = (const short*)x.data_ptr() + m0 * K;
