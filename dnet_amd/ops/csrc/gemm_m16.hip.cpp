@@ -647,7 +647,7 @@ __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
 }
 
 static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
-                       int min_pairs) {
+                       int min_pairs, int g_align) {
   // Target ~1024 blocks (~4 blocks / 16 waves per CU) so HBM latency is
   // covered by wave overlap; each split keeps >= 8 K-chunk-pairs of work.
   static const int forced = []() {
@@ -658,7 +658,9 @@ static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
   if (forced > 0)
     return (M * N > scratch_elems || (K / 64) < forced) ? 1 : forced;
   int sk = 1;
-  while (sk < 32 && blocks * sk < 1024 && (K / 64) / (sk * 2) >= min_pairs)
+  while (sk < 32 && blocks * sk < 1024 &&
+         (K / 64) / (sk * 2) >= min_pairs &&
+         ((K / 64) / (sk * 2)) * 64 % g_align == 0)  // split starts G-aligned
     sk *= 2;
   if (sk > 1 && M * N > scratch_elems) sk = 1;
   return sk;
@@ -675,7 +677,8 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   // the streamed schedule tolerates short splits (one combined round
   // trip per 4-pair tile), so quantized paths split deeper to fill the
   // 256 CUs on small-N shapes (qkv/o run at <2 blocks/CU otherwise)
-  const int sk = pick_splitk(N, K, M, scratch_elems, bits < 16 ? 4 : 8);
+  const int sk = pick_splitk(N, K, M, scratch_elems, bits < 16 ? 4 : 8,
+                             bits < 16 ? std::max<int>((int)group, 64) : 64);
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   const dim3 grid((unsigned)((N + 63) / 64), sk);
   const short* xp = (const short*)x.data_ptr() + m0 * K;
