@@ -657,17 +657,11 @@ static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
   const int blocks = (int)((N + 63) / 64);
   if (forced > 0)
     return (M * N > scratch_elems || (K / 64) < forced) ? 1 : forced;
-  // smallest sk (any integer, not just powers of 2) that fills the chip
-  // (~1024 blocks = 4/CU), subject to: >= min_pairs per split and
-  // G-aligned split starts; else the largest feasible sk
-  const int pairs = (int)(K / 64);
   int sk = 1;
-  for (int c = 2; c <= 32 && blocks * sk < 1024; ++c) {
-    const int pp = pairs / c;
-    if (pp < min_pairs) break;
-    if ((int64_t)pp * 64 % g_align != 0) continue;
-    sk = c;
-  }
+  while (sk < 32 && blocks * sk < 1024 &&
+         (K / 64) / (sk * 2) >= min_pairs &&
+         ((K / 64) / (sk * 2)) * 64 % g_align == 0)  // split starts G-aligned
+    sk *= 2;
   if (sk > 1 && M * N > scratch_elems) sk = 1;
   return sk;
 }
