@@ -14,6 +14,8 @@ remap).
 """
 from __future__ import annotations
 
+import os
+
 from dataclasses import dataclass, field
 from typing import Optional, Sequence
 
@@ -296,7 +298,16 @@ def _chunked_causal_attention(q, k, v, scale, q_offsets, window=0,
     q: [B, Hq, T, D]; k/v: [B, Hkv, S, D] where S >= T and queries occupy
     positions q_offsets..q_offsets+T-1 (same offset for all batches).
     Chunked over queries to bound the score matrix.
+
+    On GPU with supported head dims this dispatches to the fused MFMA
+    flash kernel (ops.attn_prefill): QK^T/PV on matrix cores, online fp32
+    softmax, score tiles never materialized (round-2 TTFT lever).
     """
+    if (q.is_cuda and q.dtype == torch.bfloat16 and ops.has_native()
+            and ops.attn_prefill_supported(q.shape[-1], v.shape[-1])
+            and os.environ.get("DNET_EINSUM_PREFILL") is None):
+        return ops.attn_prefill(q, k, v, scale, q_offsets, window or 0,
+                                sinks)
     B, Hq, T, D = q.shape
     Hkv = k.shape[1]
     S = k.shape[2]

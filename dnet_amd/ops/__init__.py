@@ -167,6 +167,29 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                            kscale, vscale)
 
 
+_PREFILL_DIMS = {(128, 128), (64, 64), (192, 128), (96, 96)}
+
+
+def attn_prefill_supported(d: int, dv: int) -> bool:
+    return (d, dv) in _PREFILL_DIMS
+
+
+def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 scale: float, q_off: int, window: int = 0,
+                 sinks: torch.Tensor | None = None) -> torch.Tensor:
+    """Fused MFMA flash prefill attention (causal + window + sinks).
+
+    q [B,Hq,T,D], k/v [B,Hkv,S,D(v)] -> [B,Hq,T,Dv]. GPU-only; callers
+    gate on attn_prefill_supported and fall back to the chunked-einsum
+    path otherwise (models/base.py)."""
+    B, Hq, T, D = q.shape
+    Dv = v.shape[-1]
+    out = torch.empty(B, Hq, T, Dv, dtype=q.dtype, device=q.device)
+    _native().attn_prefill(q.contiguous(), k.contiguous(), v.contiguous(),
+                           sinks, out, q_off, window, scale)
+    return out
+
+
 def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
                 kscale=None, vscale=None, wpos=None) -> None:
     if q.is_cuda:

@@ -16,6 +16,9 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  c10::optional<torch::Tensor> vscale,
                  c10::optional<torch::Tensor> partials, int64_t splits,
                  bool combine);
+void attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                  c10::optional<torch::Tensor> sinks, torch::Tensor out,
+                  int64_t q_off, int64_t window, double scale);
 void attn_combine(torch::Tensor partials, c10::optional<torch::Tensor> sinks,
                   torch::Tensor out, int64_t splits);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -58,6 +61,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("scale"), py::arg("window"), py::arg("sinks"),
         py::arg("kscale"), py::arg("vscale"), py::arg("partials"),
         py::arg("splits"), py::arg("combine") = true);
+  m.def("attn_prefill", &dnet::attn_prefill,
+        "MFMA flash prefill attention (causal + window + sinks)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("sinks"),
+        py::arg("out"), py::arg("q_off"), py::arg("window"),
+        py::arg("scale"));
   m.def("attn_combine", &dnet::attn_combine,
         "merge split/rank flash-decode partials");
   m.def("rope_append", &dnet::rope_append, "fused RoPE + KV append (decode)",

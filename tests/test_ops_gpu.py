@@ -554,3 +554,39 @@ def test_gemm_stream_matches_generic_int4(m, k, group):
     out_ref = ops.ref.gemv_int4(x.cpu(), q4, scales, group)
     assert torch.allclose(out_s.float().cpu(), out_ref.float(), atol=8e-2,
                           rtol=4e-2)
+
+
+@pytest.mark.parametrize("hq,hkv,d,dv,t,s,qoff,window,use_sinks", [
+    (8, 2, 128, 128, 128, 128, 0, 0, False),      # GQA causal from 0
+    (8, 8, 64, 64, 96, 160, 64, 0, False),        # continuation (qoff>0)
+    (4, 4, 128, 128, 33, 61, 28, 0, False),       # odd T/S, partial q tile
+    (8, 2, 128, 128, 128, 128, 0, 48, False),     # sliding window
+    (8, 2, 128, 128, 120, 120, 0, 32, True),      # window + sinks (gpt-oss)
+    (10, 10, 192, 128, 64, 90, 26, 0, False),     # MLA dims (qk192/v128)
+])
+def test_attn_prefill_flash_vs_einsum(hq, hkv, d, dv, t, s, qoff, window,
+                                      use_sinks):
+    """The fused MFMA flash prefill kernel must match the chunked-einsum
+    reference path (bf16 GEMMs + fp32 softmax) on every geometry the
+    serving path uses: GQA, continuation offsets, partial tiles, sliding
+    window, sinks, MLA head dims."""
+    from dnet_amd.models.base import _chunked_causal_attention
+    torch.manual_seed(hq * 100 + t)
+    B = 3
+    q = torch.randn(B, hq, t, d, dtype=torch.bfloat16, device=_dev())
+    k = torch.randn(B, hkv, s, d, dtype=torch.bfloat16, device=_dev())
+    v = torch.randn(B, hkv, s, dv, dtype=torch.bfloat16, device=_dev())
+    sinks = (torch.randn(hq, dtype=torch.bfloat16, device=_dev())
+             if use_sinks else None)
+    scale = d ** -0.5
+    out = ops.attn_prefill(q, k, v, scale, qoff, window, sinks)
+    import os
+    os.environ["DNET_EINSUM_PREFILL"] = "1"
+    try:
+        ref_out = _chunked_causal_attention(q, k, v, scale, qoff, window,
+                                            sinks)
+    finally:
+        os.environ.pop("DNET_EINSUM_PREFILL", None)
+    assert torch.allclose(out.float(), ref_out.float(), atol=4e-2,
+                          rtol=3e-2), \
+        (out.float() - ref_out.float()).abs().max()
