@@ -161,6 +161,38 @@ class RingExecutor:
                                          device=self.device)
                              for _ in range(mb_count)]
 
+    class _RecvPipe:
+        """Depth-2 pre-posted receive pipeline for a fixed-count sequence
+        of identical ring hops (the decode loop used blocking
+        recv per hop; pre-posting lets the upstream rank's send proceed
+        while this rank still computes the previous microbatch). `total`
+        must be EXACT — every posted receive is matched before the loop
+        exits, so no dangling requests survive (gloo would hang on
+        shutdown otherwise)."""
+
+        def __init__(self, ring, like: torch.Tensor, total: int,
+                     depth: int = 2):
+            self.ring = ring
+            self.remaining = total
+            self.q: list = []
+            self.bufs = [torch.empty_like(like) for _ in range(min(depth,
+                                                                  total))]
+            self._free = list(self.bufs)
+            self._post()
+
+        def _post(self):
+            while self._free and len(self.q) < self.remaining:
+                b = self._free.pop()
+                self.q.append((self.ring.irecv(b), b))
+
+        def next_into(self, dst: torch.Tensor):
+            req, b = self.q.pop(0)
+            req.wait()
+            dst.copy_(b)
+            self.remaining -= 1
+            self._free.append(b)
+            self._post()
+
     def _send_hidden(self, mb: int):
         if not self.compress_ratio:
             self.ring.send(self.hbuf[mb])
@@ -300,6 +332,40 @@ class RingExecutor:
                else None)
         tok_req: dict[int, object] = {}
         last_r = self.rounds - 1
+        # pre-posted hidden receives (exact total: the loop consumes
+        # per_step hops per (s, mb)) + async sends drained just before
+        # their buffer is rewritten — no host-blocking comm between the
+        # compute launches (reference pipelined its
+        # in-flight frames, src/dnet/shard/adapters/ring.py:226-299)
+        # first stage excluded: its lap-wrap hops and the token frames
+        # arrive on the SAME (last->first) channel, and p2p matching is
+        # FIFO per pair — a pre-posted hidden recv would swallow a token
+        # frame (size mismatch/hang). Middle/last stages only ever
+        # receive hidden frames from prev, so pre-posting is safe there.
+        pipe = None
+        if self.stages > 1 and not self.is_first and not self.compress_ratio:
+            if n > 0:
+                pipe = self._RecvPipe(self.ring, self.hbuf[0],
+                                      n * M * self.rounds)
+        send_req: dict[int, list] = {}
+
+        def recv_h(mb):
+            if pipe is not None:
+                pipe.next_into(self.hbuf[mb])
+            else:
+                self._recv_hidden(mb)
+
+        def send_h(mb):
+            if pipe is not None:   # same no-compress fast path
+                send_req.setdefault(mb, []).append(
+                    self.ring.isend(self.hbuf[mb]))
+            else:
+                self._send_hidden(mb)
+
+        def drain_sends(mb):
+            for rq in send_req.pop(mb, ()):
+                rq.wait()
+
         for s in range(n):
             for mb in range(M):
                 if self.is_first:
@@ -309,19 +375,21 @@ class RingExecutor:
                         out[mb, :, s - 1] = self.tokbuf[mb]
                     for r in range(self.rounds):
                         if r > 0:
-                            self._recv_hidden(mb)   # lap wrap from last stage
+                            recv_h(mb)          # lap wrap from last stage
+                        drain_sends(mb)         # hbuf[mb] rewritten below
                         self._run_decode(mb, r)
                         if self.stages > 1 and not (self.is_last and r == last_r):
-                            self._send_hidden(mb)
+                            send_h(mb)
                     if self.stages > 1:
                         tok_req[mb] = self.ring.irecv(self.tokbuf[mb],
                                                       src=self.token_src)
                 if self.stages > 1 and not self.is_first:
                     for r in range(self.rounds):
-                        self._recv_hidden(mb)
+                        recv_h(mb)
+                        drain_sends(mb)
                         self._run_decode(mb, r)
                         if not (self.is_last and r == last_r):
-                            self._send_hidden(mb)
+                            send_h(mb)
                 if self.is_last:
                     tok, _, _ = self.sampler.sample(self.logits_buf[mb].float())
                     if self.stages > 1:
@@ -331,6 +399,8 @@ class RingExecutor:
                         if out is not None and self.rank == 0:
                             out[mb, :, s] = tok
                 self.kvs[mb].pos.add_(1)
+        for mb in list(send_req):
+            drain_sends(mb)
         # drain last round's tokens on rank 0
         if self.is_first and self.stages > 1:
             for mb in range(M):
