@@ -76,6 +76,10 @@ class TransportSettings(BaseModel):
 class StorageSettings(BaseModel):
     model_dir: str = "~/.dnet_amd/models"
     repack_dir: str = "~/.dnet_amd/repacked_layers"
+    # write per-layer repacked safetensors on first cold load and load
+    # from them on later loads (reference wires repack into OffloadPolicy,
+    # src/dnet/shard/policies/offload.py:46-77; here it serves every load)
+    repack_on_load: bool = True
 
 
 class TopologySettings(BaseModel):

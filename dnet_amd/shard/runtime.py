@@ -224,6 +224,16 @@ class ShardRuntime:
                  req.rank, req.world_size, len(req.layers), req.layers[:4])
 
     def _load_weights(self, ex: RingExecutor, req: ShardLoadModelRequest) -> None:
+        from ..utils import repack as rp
+        model_id = req.model_name or req.model_path
+        need_api = ex.is_first or ex.is_last
+        if self.settings.storage.repack_on_load:
+            sd = rp.load_repacked(model_id, ex.my_layers, include_api=need_api)
+            if sd is not None:
+                log.info("repack fastpath: loaded %d tensors from %s",
+                         len(sd), rp.repack_dir_for(model_id, ex.my_layers))
+                ex.model.load_state_dict(sd)
+                return
         meta = get_model_metadata(req.model_path)
         names: list[str] = []
         for lid in ex.my_layers:
@@ -233,6 +243,15 @@ class ShardRuntime:
         if ex.is_last:
             names += meta.final_norm + meta.lm_head + meta.embed
         sd = load_tensors(meta, sorted(set(names)))
+        if self.settings.storage.repack_on_load:
+            try:
+                # repack for free from the just-loaded tensors: the next
+                # load of this assignment skips the full-safetensors parse
+                rp.ensure_repacked_for_layers(req.model_path, model_id,
+                                              ex.my_layers,
+                                              include_api=need_api, sd=sd)
+            except Exception:
+                log.exception("repack write failed (non-fatal)")
         ex.model.load_state_dict(sd)
 
     def _unload(self) -> None:

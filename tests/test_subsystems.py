@@ -809,3 +809,54 @@ def test_interleaved_chunked_admission(monkeypatch):
     rt._unload()
     assert mixed["B"] == solo["B"]          # undisturbed by A's admission
     assert mixed["A"] == solo["A"]          # chunked == unchunked
+
+
+def test_repack_fastpath_on_reload(tmp_path, monkeypatch):
+    """the first cold load of a real checkpoint writes
+    per-layer repacked files; a second load of the same assignment reads
+    THEM (fastpath) — proven by deleting the source safetensors between
+    loads — and produces identical tokens."""
+    transformers = pytest.importorskip("transformers")
+    from safetensors.torch import save_file
+
+    from dnet_amd.config import reset_settings
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    monkeypatch.setenv("DNET_STORAGE_REPACK_DIR", str(tmp_path / "repack"))
+    reset_settings()
+    try:
+        tc = transformers.LlamaConfig(
+            hidden_size=128, intermediate_size=256, num_hidden_layers=2,
+            num_attention_heads=2, num_key_value_heads=2, head_dim=64,
+            vocab_size=256, rope_theta=10000.0, max_position_embeddings=64,
+            tie_word_embeddings=False)
+        hf = transformers.LlamaForCausalLM(tc).eval()
+        mdir = tmp_path / "ckpt"
+        mdir.mkdir()
+        (mdir / "config.json").write_text(tc.to_json_string())
+        src = mdir / "model.safetensors"
+        save_file({k: v.contiguous() for k, v in hf.state_dict().items()},
+                  str(src))
+
+        def load_and_decode(tag):
+            rt = ShardRuntime(tag)
+            rt._load(ShardLoadModelRequest(
+                model_path=str(mdir), model_name="tiny-rp", total_layers=2,
+                layers=[0, 1], rank=0, world_size=1, max_batch=1,
+                max_seq=64))
+            ex = rt.executor
+            toks = torch.randint(0, 256, (1, 1, 6),
+                                 generator=torch.Generator().manual_seed(4))
+            first = ex.prefill(toks)
+            gen = ex.decode_rounds(3)
+            rt._unload()
+            return torch.cat([first.unsqueeze(-1), gen], dim=-1)
+
+        cold = load_and_decode("cold")
+        assert (tmp_path / "repack").exists()
+        src.unlink()     # source gone: only the repack can serve the load
+        warm = load_and_decode("warm")
+        assert torch.equal(cold, warm)
+    finally:
+        reset_settings()
