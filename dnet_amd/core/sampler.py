@@ -93,6 +93,14 @@ class RowSampler:
         # from their own generator, so one request's stream is reproducible
         # regardless of which other slots are active.
         self.gens: list = [None] * batch
+        # per-row logprob config (OpenAI `logprobs`/`top_logprobs`); the
+        # sampler computes them for the WHOLE batch when any row wants
+        # them, and the emitter picks per row
+        self.want_lp: list = [False] * batch
+        self.n_top: list = [0] * batch
+        self.last_logp = None    # [B] device, chosen-token logprob
+        self.last_topv = None    # [B, K] device
+        self.last_topi = None
 
     def set_row(self, i: int, cfg: DecodingConfig, seed=None):
         self.temp[i] = cfg.temperature
@@ -101,6 +109,8 @@ class RowSampler:
         self.min_p[i] = cfg.min_p
         if cfg.min_p > 0:
             self._has_min_p = True
+        self.want_lp[i] = bool(cfg.logprobs)
+        self.n_top[i] = int(cfg.top_logprobs or 0)
         g = None
         if seed is not None:
             g = torch.Generator(device=self.device)
@@ -149,4 +159,16 @@ class RowSampler:
                 cdf, u * cdf[..., -1:]).squeeze(-1).clamp_(0, V - 1)
         else:
             tok = torch.multinomial(probs, 1).squeeze(-1)
-        return torch.where(self.temp <= 0.0, greedy_tok, tok)
+        tok = torch.where(self.temp <= 0.0, greedy_tok, tok)
+        if any(self.want_lp[:B]):
+            logp = lf - torch.logsumexp(lf, dim=-1, keepdim=True)
+            self.last_logp = logp.gather(-1, tok.unsqueeze(-1)).squeeze(-1)
+            kmax = max(self.n_top[:B])
+            if kmax > 0:
+                self.last_topv, self.last_topi = torch.topk(logp, kmax,
+                                                            dim=-1)
+            else:
+                self.last_topv = self.last_topi = None
+        else:
+            self.last_logp = self.last_topv = self.last_topi = None
+        return tok

@@ -308,7 +308,8 @@ class RingExecutor:
                                                     src=self.token_src))
             else:
                 logits = self.model.normalize_project(h[:, -1].contiguous())
-                tok, _, _ = self.sampler.sample(logits)
+                tok, logprob, tops = self.sampler.sample(logits.float())
+                self.last_logprob, self.last_tops = logprob, tops
                 first_tokens[mb] = tok
                 if not self.is_first:
                     self.ring.send(first_tokens[mb], dst=self.token_dst)

@@ -165,17 +165,17 @@ class ShardRuntime:
         ex = RingExecutor(cfg, req.rank, req.world_size, device, plan=plan,
                           mb_count=1, mb_size=req.max_batch,
                           smax=req.max_seq,
-                          # slot scheduler decodes eager by default (the
-                          # eager gap is small at serving batch sizes);
-                          # DNET_SLOTS_GRAPHS=1 opts into hipGraph decode
-                          # under slot churn (replays interleaved with
-                          # eager per-slot prefills — validate before
-                          # relying on it)
+                          # hipGraph decode is ON by default, including
+                          # under slot churn (admissions prefill eagerly
+                          # into the same static KV storage the captured
+                          # graph reads — verified token-exact vs eager
+                          # on hardware, tests/test_ops_gpu.py slots
+                          # churn test). DNET_SLOTS_GRAPHS=0 opts out.
                           use_graphs=(device.type == "cuda"
                                       and self.settings.compute.use_graphs
                                       and residency == 0
                                       and (not slots_mode or os.environ.get(
-                                          "DNET_SLOTS_GRAPHS") == "1")),
+                                          "DNET_SLOTS_GRAPHS", "1") == "1")),
                           init_weights=synthetic, residency=residency,
                           kv_bits=req.kv_bits)
         if not synthetic:
@@ -525,7 +525,9 @@ class ShardRuntime:
         p = frame.get("params", {})
         cfg = DecodingConfig(
             temperature=p.get("temperature", 0.0), top_p=p.get("top_p", 1.0),
-            top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0))
+            top_k=int(p.get("top_k", 0)), min_p=p.get("min_p", 0.0),
+            logprobs=bool(p.get("logprobs", False)),
+            top_logprobs=int(p.get("top_logprobs", 0)))
         stop_ids = list(frame.get("stop_ids", []))
         max_tokens = int(frame.get("max_tokens", 128))
         nonce = frame.get("nonce", "")
@@ -537,7 +539,9 @@ class ShardRuntime:
                                 len(stop_ids), len(nonce_ids),
                                 cfg.temperature, cfg.top_p, cfg.top_k,
                                 cfg.min_p,
-                                -1.0 if seed is None else int(seed))
+                                -1.0 if seed is None else int(seed),
+                                1.0 if cfg.logprobs else 0.0,
+                                cfg.top_logprobs)
             payload = torch.cat([
                 tokens.flatten().to(self._comm_device()),
                 torch.tensor(stop_ids + nonce_ids, dtype=torch.int64,
@@ -551,7 +555,9 @@ class ShardRuntime:
         si, T, max_tokens = int(cmd[1]), int(cmd[2]), int(cmd[3])
         n_stop, n_nonce = int(cmd[4]), int(cmd[5])
         cfg = DecodingConfig(temperature=cmd[6], top_p=cmd[7],
-                             top_k=int(cmd[8]), min_p=cmd[9])
+                             top_k=int(cmd[8]), min_p=cmd[9],
+                             logprobs=bool(cmd[11]),
+                             top_logprobs=int(cmd[12]))
         seed = None if cmd[10] < 0 else int(cmd[10])
         payload = torch.zeros(T + n_stop + n_nonce, dtype=torch.int64,
                               device=self._comm_device())
@@ -572,8 +578,6 @@ class ShardRuntime:
         max_tokens = max(1, min(max_tokens,
                                 ex.smax - int(tokens.shape[-1])))
         gen = self._row_sampler.set_row(si, cfg, seed)
-        ex.last_logprob = None   # slots mode: no per-token logprobs (yet)
-        ex.last_tops = None
         chunk = int(os.environ.get("DNET_PREFILL_CHUNK", "2048"))
         if ex.world == 1 and int(tokens.shape[-1]) > chunk:
             # long prompt: admit in "prefill" state — the tick loop feeds
@@ -587,10 +591,14 @@ class ShardRuntime:
             return
         logits = ex.prefill_slot(si, tokens)
         t0_t = torch.zeros(1, dtype=torch.int64, device=self._comm_device())
+        lp0 = tops0 = None
         if ex.is_last:
             from ..core.sampler import Sampler
-            tok, _, _ = Sampler(cfg, generator=gen).sample(logits.float())
+            tok, lp, tops = Sampler(cfg, generator=gen).sample(logits.float())
             t0_t[0] = int(tok[0])
+            if lp is not None:
+                lp0 = float(lp[0])
+                tops0 = tops[0] if tops else None
         if ex.world > 1:
             import torch.distributed as dist
             dist.broadcast(t0_t, src=(ex.stages - 1) * ex.tp)
@@ -600,7 +608,8 @@ class ShardRuntime:
               "stop_ids": set(stop_ids)}
         done = t0 in st["stop_ids"] or max_tokens <= 1
         if ex.is_last:
-            self._emit_token(nonce, t0, finished=done)
+            self._emit_token(nonce, t0, finished=done, logprob=lp0,
+                             tops=tops0)
         if not done:
             self.slots[si] = st
         log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
@@ -626,17 +635,20 @@ class ShardRuntime:
         kvslot.pos.fill_(T)
         logits = ex.model.normalize_project(h[:, -1].contiguous())
         from ..core.sampler import Sampler
-        tok, _, _ = Sampler(st["cfg"],
-                            generator=self._row_sampler.gens[si]
-                            ).sample(logits.float())
+        tok, lp, tops = Sampler(st["cfg"],
+                                generator=self._row_sampler.gens[si]
+                                ).sample(logits.float())
         t0 = int(tok[0])
+        lp0 = float(lp[0]) if lp is not None else None
+        tops0 = tops[0] if tops else None
         ex.tokbuf[0][si] = t0
         st["produced"] = 1
         st["state"] = "active"
         st.pop("tokens")
         done = t0 in st["stop_ids"] or st["max_tokens"] <= 1
         if ex.is_last:
-            self._emit_token(st["nonce"], t0, finished=done)
+            self._emit_token(st["nonce"], t0, finished=done, logprob=lp0,
+                             tops=tops0)
         if done:
             self.slots[si] = None
 
@@ -645,6 +657,7 @@ class ShardRuntime:
         synchronously): hops + compute, sample on the last stage, token
         broadcast, identical slot-state update on every rank."""
         ex = self.executor
+        rs = self._row_sampler
         ex.slot_step_compute()
         if ex.is_last:
             toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
@@ -660,15 +673,17 @@ class ShardRuntime:
         self._slot_emit(ex.tokbuf[0],
                         [(i, self._slot_gen[i])
                          for i, st in enumerate(self.slots)
-                         if st is not None])
+                         if st is not None],
+                        rs.last_logp, rs.last_topv, rs.last_topi)
 
     _pending = None   # (device tokens, [(slot, gen)]) of the in-flight step
     _slot_gen: list = []   # per-slot admit generation (see _slot_emit)
 
     def _slot_step_launch(self):
         """Enqueue one decode step for the whole batch — device ops only,
-        no host sync. Returns (tokens device tensor, active slots)."""
+        no host sync. Returns (tokens, active slots, logprob tensors)."""
         ex = self.executor
+        rs = self._row_sampler
         ex.slot_step_compute()
         toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
         ex.tokbuf[0].copy_(toks_t)
@@ -681,12 +696,18 @@ class ShardRuntime:
         if park:
             ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
                                        device=ex.device)] = ex.smax - 1
-        return toks_t, [(i, self._slot_gen[i])
-                        for i, st in enumerate(self.slots)
-                        if st is not None and st.get("state") != "prefill"]
+        return (toks_t, [(i, self._slot_gen[i])
+                         for i, st in enumerate(self.slots)
+                         if st is not None and st.get("state") != "prefill"],
+                rs.last_logp, rs.last_topv, rs.last_topi)
 
-    def _slot_emit(self, toks_t, active) -> None:
+    def _slot_emit(self, toks_t, active, logp=None, topv=None,
+                   topi=None) -> None:
         toks = toks_t.tolist()   # syncs; overlaps the already-launched step
+        rs = self._row_sampler
+        lp_l = logp.tolist() if logp is not None else None
+        tv_l = topv.tolist() if topv is not None else None
+        ti_l = topi.tolist() if topi is not None else None
         for i, gen in active:
             st = self.slots[i]
             if st is None or self._slot_gen[i] != gen:
@@ -697,23 +718,34 @@ class ShardRuntime:
             st["produced"] += 1
             done = (t in st["stop_ids"]
                     or st["produced"] >= st["max_tokens"])
-            self._emit_token(st["nonce"], t, finished=done)
+            lp = lp_l[i] if (lp_l is not None and rs.want_lp[i]) else None
+            tops = None
+            if tv_l is not None and rs.n_top[i] > 0:
+                k = rs.n_top[i]
+                tops = {int(ti_l[i][j]): float(tv_l[i][j]) for j in range(k)}
+            self._emit_token(st["nonce"], t, finished=done, logprob=lp,
+                             tops=tops)
             if done:
                 self.slots[i] = None
 
     _emit_s = 0.0
 
-    def _emit_token(self, nonce: str, token_id: int, finished: bool = False):
+    def _emit_token(self, nonce: str, token_id: int, finished: bool = False,
+                    logprob=None, tops=None):
         ex = self.executor
         if self._callback is None:
             return
         t0 = time.perf_counter()
         frame = {"t": "token", "nonce": nonce, "token_id": token_id,
                  "ts_ms": int(time.time() * 1e3), "finished": finished}
-        if ex.last_logprob is not None:
-            frame["logprob"] = float(ex.last_logprob[0])
-        if ex.last_tops is not None:
-            frame["top_logprobs"] = ex.last_tops[0]
+        if logprob is None and ex.last_logprob is not None:
+            logprob = float(ex.last_logprob[0])     # legacy serial path
+        if tops is None and ex.last_tops is not None:
+            tops = ex.last_tops[0]
+        if logprob is not None:
+            frame["logprob"] = logprob
+        if tops is not None:
+            frame["top_logprobs"] = tops
         try:
             self._callback.send(frame)
         except OSError:

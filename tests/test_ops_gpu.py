@@ -590,3 +590,65 @@ def test_attn_prefill_flash_vs_einsum(hq, hkv, d, dv, t, s, qoff, window,
     assert torch.allclose(out.float(), ref_out.float(), atol=4e-2,
                           rtol=3e-2), \
         (out.float() - ref_out.float()).abs().max()
+
+
+def test_slots_graph_churn_matches_eager():
+    """Continuous batching with hipGraph decode ON (now the default) must
+    be token-exact vs eager under slot churn: staggered admits, early
+    stops, slot reuse, a chunked long-prompt admission mid-stream."""
+    import os
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    def run(graphs: bool):
+        os.environ["DNET_SLOTS_GRAPHS"] = "1" if graphs else "0"
+        frames: dict[str, list] = {}
+
+        class Cap:
+            def send(self, fr):
+                frames.setdefault(fr["nonce"], []).append(fr["token_id"])
+
+            def close(self):
+                pass
+
+        try:
+            rt = ShardRuntime("churn")
+            rt._load(ShardLoadModelRequest(
+                model_path="tiny", model_name="tiny", total_layers=4,
+                layers=[0, 1, 2, 3], rank=0, world_size=1,
+                max_batch=3, max_seq=96))
+            rt._callback = Cap()
+            pa = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+            pb = torch.arange(3, 11, dtype=torch.int32).numpy().tobytes()
+            plong = torch.arange(1, 41, dtype=torch.int32).numpy().tobytes()
+            rt.infer_q.put({"nonce": "a", "tokens": pa, "prompt_len": 8,
+                            "max_tokens": 3, "stop_ids": [], "params": {}})
+            rt.infer_q.put({"nonce": "b", "tokens": pb, "prompt_len": 8,
+                            "max_tokens": 12, "stop_ids": [], "params": {}})
+            for _ in range(4):
+                rt._slots_tick()
+            # a finishes -> its slot is reused by c; long prompt d admits
+            # CHUNKED between decode steps (DNET_PREFILL_CHUNK)
+            os.environ["DNET_PREFILL_CHUNK"] = "16"
+            rt.infer_q.put({"nonce": "c", "tokens": pa, "prompt_len": 8,
+                            "max_tokens": 6, "stop_ids": [], "params": {}})
+            rt.infer_q.put({"nonce": "d", "tokens": plong, "prompt_len": 40,
+                            "max_tokens": 5, "stop_ids": [], "params": {}})
+            for _ in range(40):
+                rt._slots_tick()
+                if (all(s is None for s in rt.slots) and rt._pending is None
+                        and len(frames.get("d", [])) >= 5):
+                    break
+            rt._unload()
+            return frames
+        finally:
+            os.environ.pop("DNET_SLOTS_GRAPHS", None)
+            os.environ.pop("DNET_PREFILL_CHUNK", None)
+
+    eager = run(False)
+    graphed = run(True)
+    assert set(eager) == set(graphed) == {"a", "b", "c", "d"}
+    for k in eager:
+        assert eager[k] == graphed[k], (k, eager[k], graphed[k])
+    assert len(graphed["a"]) == 3 and len(graphed["b"]) == 12
+    assert len(graphed["c"]) == 6 and len(graphed["d"]) == 5

@@ -860,3 +860,59 @@ def test_repack_fastpath_on_reload(tmp_path, monkeypatch):
         assert torch.equal(cold, warm)
     finally:
         reset_settings()
+
+
+def test_slot_logprobs_match_serial():
+    """per-token logprobs in slots mode. A logprobs
+    request through the slot scheduler gets per-token logprob +
+    top_logprobs frames that match the legacy serial path's values."""
+    from dnet_amd.core.types import ShardLoadModelRequest
+    from dnet_amd.shard.runtime import ShardRuntime
+
+    frames: dict[str, list] = {}
+
+    class Cap:
+        def send(self, fr):
+            frames.setdefault(fr["nonce"], []).append(fr)
+
+        def close(self):
+            pass
+
+    def load(max_batch, tag):
+        rt = ShardRuntime(tag)
+        rt._load(ShardLoadModelRequest(
+            model_path="tiny", model_name="tiny", total_layers=4,
+            layers=[0, 1, 2, 3], rank=0, world_size=1,
+            max_batch=max_batch, max_seq=64))
+        rt._callback = Cap()
+        return rt
+
+    prompt = torch.arange(1, 9, dtype=torch.int32).numpy().tobytes()
+    params = {"logprobs": True, "top_logprobs": 3}
+
+    rt1 = load(1, "serial")
+    rt1._execute_infer("serial", torch.frombuffer(
+        bytearray(prompt), dtype=torch.int32).long().view(1, 1, -1), 5, [],
+        dict(params))
+    rt1._unload()
+
+    rt = load(2, "slots")
+    rt.infer_q.put({"nonce": "lp", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 5, "stop_ids": [], "params": dict(params)})
+    rt.infer_q.put({"nonce": "plain", "tokens": prompt, "prompt_len": 8,
+                    "max_tokens": 5, "stop_ids": [], "params": {}})
+    for _ in range(30):
+        rt._slots_tick()
+        if len(frames.get("lp", [])) >= 5 and len(frames.get("plain", [])) >= 5:
+            break
+    rt._unload()
+
+    ser, slot = frames["serial"], frames["lp"]
+    assert len(ser) == 5 and len(slot) == 5
+    for a, b in zip(ser, slot):
+        assert a["token_id"] == b["token_id"]
+        assert "logprob" in b and "top_logprobs" in b
+        assert abs(a["logprob"] - b["logprob"]) < 1e-3
+        assert len(b["top_logprobs"]) == 3
+    # the no-logprobs neighbor got none
+    assert all("logprob" not in f for f in frames["plain"])
