@@ -52,6 +52,11 @@ class Linear:
         return self.scales is not None
 
     @property
+    def mxfp4(self) -> bool:
+        # MXFP4 carries e8m0 uint8 scales (int4/int8 scales are bf16)
+        return self.scales is not None and self.scales.dtype == torch.uint8
+
+    @property
     def out_features(self) -> int:
         return self.w.shape[0]
 
@@ -86,8 +91,24 @@ class Linear:
             return l
         return cls(w.to(torch.bfloat16), bias)
 
+    @classmethod
+    def make_mxfp4(cls, packed: torch.Tensor, scales: torch.Tensor,
+                   bias: Optional[torch.Tensor]) -> "Linear":
+        """Native MXFP4 weights: e2m1 nibbles [N, K/2] + e8m0 [N, K/32].
+        Decode runs through the grouped MoE kernels (fused in-kernel
+        dequant); this Linear's own __call__ (big-T sparse path) dequants
+        to bf16 transiently."""
+        l = cls(packed, bias, scales, 32, False)
+        l.bits = 4
+        return l
+
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         m = x.shape[0]
+        if self.mxfp4:
+            y = x @ ops.dequant_mxfp4(self.w, self.scales).t()
+            if self.bias is not None:
+                y = y + self.bias
+            return y
         if m <= GEMV_MAX_M:
             if self.bits == 4:
                 return ops.gemv_int4(x, self.w, self.scales, self.group,
@@ -173,6 +194,8 @@ class LayerWeights:
             if f in d:
                 setattr(lw, f, d[f])
         def _mk(w, bias, scales, pk):
+            if scales is not None and scales.dtype == torch.uint8:
+                return Linear.make_mxfp4(w, scales, bias)
             l = Linear(w, bias, scales, group, pk)
             if scales is not None:
                 l.bits = 4 if w.dtype == torch.uint8 else 8

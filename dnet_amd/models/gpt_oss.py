@@ -80,16 +80,28 @@ class GptOssRingModel(MoERingModel):
                     return sd[pref + name].to(torch.bfloat16)
             return None
 
+        # MXFP4 checkpoints ship <name>_blocks (uint8 nibble pairs) +
+        # <name>_scales (uint8 E8M0), output-major [E, rows, K/32, 16] /
+        # [E, rows, K/32]. Default: keep them PACKED and execute natively
+        # in the grouped MoE kernels (~4.25 bit/weight resident;
+        # reference executes MXFP4 via MLX nn.quantize,
+        # src/dnet/core/models/gpt_oss.py:216-287). An explicit int8/int4
+        # quant request, or DNET_MXFP4_DEQUANT=1, dequantizes to bf16 at
+        # load instead (round-1 behavior).
+        import os
+        mx_native = c.quant is None and not os.environ.get(
+            "DNET_MXFP4_DEQUANT")
+
         def get_mx(name):
-            # MXFP4 checkpoints ship <name>_blocks (uint8 nibble pairs) +
-            # <name>_scales (uint8 E8M0); dequantized to bf16 at load like
-            # the reference's MXFP4 sanitization. Layout is output-major
-            # [E, rows, K] (the bf16 layout is input-major [E, K, rows]).
             for pref in ("model.", ""):
                 b = sd.get(pref + name + "_blocks")
                 s = sd.get(pref + name + "_scales")
                 if b is not None and s is not None:
                     from .. import ops
+                    if mx_native:
+                        e, r, g, _ = b.shape
+                        return (b.reshape(e, r, g * 16).contiguous(),
+                                s.reshape(e, r, g).contiguous())
                     return ops.mxfp4_dequant(b, s)
             return None
 
@@ -119,20 +131,35 @@ class GptOssRingModel(MoERingModel):
             dn = get(p + "mlp.experts.down_proj")           # [E, I, H]
             dnb = get(p + "mlp.experts.down_proj_bias")     # [E, H]
             mx = gu is None
+            gu_s = dn_s = None
             if mx:
-                gu = get_mx(p + "mlp.experts.gate_up_proj")  # [E, 2I, H]
-                dn = get_mx(p + "mlp.experts.down_proj")     # [E, H, I]
+                gu = get_mx(p + "mlp.experts.gate_up_proj")  # [E, 2I, H(/2)]
+                dn = get_mx(p + "mlp.experts.down_proj")     # [E, H, I(/2)]
+                if mx_native:
+                    gu, gu_s = gu
+                    dn, dn_s = dn
             lw.experts_gateup, lw.experts_down = [], []
             for e in range(c.num_experts):
                 # bf16 layout is input-major (transpose); MXFP4 blocks are
-                # already output-major
+                # already output-major. De-interleave rows: gate = even
+                # rows, up = odd rows (nibble packing is along K, so row
+                # gathers work unchanged on the packed layout).
                 w_e = (gu[e] if mx else gu[e].t()).contiguous()  # interleaved
-                # de-interleave rows: gate = even rows, up = odd rows
-                w_e = torch.cat([w_e[0::2], w_e[1::2]])     # [2I, H] concat
+                w_e = torch.cat([w_e[0::2], w_e[1::2]])     # [2I, ...] concat
                 b_e = torch.cat([gub[e][0::2], gub[e][1::2]]) if gub is not None else None
+                b_e = b_e.to(self.device) if b_e is not None else None
+                if gu_s is not None:
+                    s_e = torch.cat([gu_s[e][0::2], gu_s[e][1::2]])
+                    lw.experts_gateup.append(Linear.make_mxfp4(
+                        w_e.to(self.device), s_e.contiguous().to(self.device),
+                        b_e))
+                    lw.experts_down.append(Linear.make_mxfp4(
+                        dn[e].contiguous().to(self.device),
+                        dn_s[e].contiguous().to(self.device),
+                        dnb[e].to(self.device) if dnb is not None else None))
+                    continue
                 lw.experts_gateup.append(Linear.make(
-                    w_e.to(self.device),
-                    b_e.to(self.device) if b_e is not None else None, c.quant))
+                    w_e.to(self.device), b_e, c.quant))
                 lw.experts_down.append(Linear.make(
                     (dn[e] if mx else dn[e].t()).contiguous().to(self.device),
                     dnb[e].to(self.device) if dnb is not None else None,

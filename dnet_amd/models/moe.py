@@ -241,8 +241,9 @@ class MoERingModel(RingModel):
         T = y.shape[0]
         logits = lw.router(y).float()
         weights, idx = self._route(logits)
+        eg0 = lw.experts_gateup[0]
         if (T <= self.DENSE_MOE_MAX_T
-                and lw.experts_gateup[0].bits in (8, 16)):
+                and (eg0.bits in (8, 16) or eg0.mxfp4)):
             st = self._expert_stack(lw)
             we = torch.zeros(T, c.num_experts, dtype=torch.float32,
                              device=y.device)

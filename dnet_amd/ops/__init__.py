@@ -249,6 +249,17 @@ unpack_int8_mfma = ref.unpack_int8_mfma
 rope_tables = ref.rope_tables
 rope_apply = ref.rope_apply
 mxfp4_dequant = ref.mxfp4_dequant
+quantize_mxfp4 = ref.quantize_mxfp4
+
+
+def dequant_mxfp4(w: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    """MXFP4 row-major (nibbles [N, K/2] + e8m0 [N, K/32]) -> bf16."""
+    if w.is_cuda and has_native():
+        out = torch.empty(w.shape[0], w.shape[1] * 2, dtype=torch.bfloat16,
+                          device=w.device)
+        _native().dequant_mxfp4(w, scales, out)
+        return out
+    return ref.dequant_mxfp4(w, scales)
 
 
 def attn_decode_partials(q, kcache, vcache, pos, scale,

@@ -32,6 +32,7 @@ void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group, bool packed);
 void dequant_int4(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group);
+void dequant_mxfp4(torch::Tensor w, torch::Tensor scales, torch::Tensor out);
 void gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
@@ -61,6 +62,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("scale"), py::arg("window"), py::arg("sinks"),
         py::arg("kscale"), py::arg("vscale"), py::arg("partials"),
         py::arg("splits"), py::arg("combine") = true);
+  m.def("dequant_mxfp4", &dnet::dequant_mxfp4,
+        "MXFP4 (e2m1 + e8m0 blocks) -> bf16");
   m.def("attn_prefill", &dnet::attn_prefill,
         "MFMA flash prefill attention (causal + window + sinks)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("sinks"),

@@ -36,6 +36,19 @@ __device__ __forceinline__ bool expert_routed(const float* __restrict__ we,
   return any;
 }
 
+// MXFP4: e2m1 nibble * 2^(e8m0-127) block scale -> float, via bf16 bit
+// construction with the block exponent folded into the exponent field
+// (no table lookup). mag==1 is the e2m1 subnormal 0.5.
+__device__ __forceinline__ float mxfp4_val(const int nib, const int ebits) {
+  const int mag = nib & 7;
+  const int e = mag >> 1, m = mag & 1;
+  int bits = ((126 + e) << 7) | (m << 6);
+  bits = (mag == 1) ? 0x3F00 : bits;
+  bits = (mag == 0) ? 0 : (bits + ((ebits - 127) << 7));
+  bits |= (nib >> 3) << 15;
+  return bits2f((short)bits);
+}
+
 // GLU = 0: silu(g) * u   GLU = 1: gpt-oss clamped (u+1) * g * sigmoid(g*a)
 template <int GLU>
 __device__ __forceinline__ float apply_glu(float g, float u, const float alpha,
@@ -48,7 +61,8 @@ __device__ __forceinline__ float apply_glu(float g, float u, const float alpha,
   return g / (1.f + __expf(-g)) * u;
 }
 
-template <int M, bool Q8, bool PACKED, int GLU>
+// QMODE: 0 = bf16, 1 = grouped int8, 2 = mxfp4 (nibble rows + e8m0)
+template <int M, int QMODE, bool PACKED, int GLU>
 __global__ void moe_gateup_kernel(
     const short* __restrict__ x, const void* __restrict__ w,
     const short* __restrict__ scales, const short* __restrict__ bias,
@@ -66,7 +80,40 @@ __global__ void moe_gateup_kernel(
   float ag[M], au[M];
 #pragma unroll
   for (int m = 0; m < M; ++m) ag[m] = au[m] = 0.f;
-  if (Q8) {
+  if (QMODE == 2) {
+    const uint8_t* base = static_cast<const uint8_t*>(w);
+    const int4* wg = reinterpret_cast<const int4*>(base + grow * (K / 2));
+    const int4* wu = reinterpret_cast<const int4*>(base + urow * (K / 2));
+    const uint8_t* sb = reinterpret_cast<const uint8_t*>(scales);
+    const uint8_t* sg = sb + grow * (K / 32);
+    const uint8_t* su = sb + urow * (K / 32);
+    const int vecs = K / 32;   // one 16 B load = one 32-value block
+    for (int v = lane; v < vecs; v += kWave) {
+      const int4 gv = wg[v], uv = wu[v];
+      const uint8_t* gq = reinterpret_cast<const uint8_t*>(&gv);
+      const uint8_t* uq = reinterpret_cast<const uint8_t*>(&uv);
+      const int ge = sg[v], ue = su[v];
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        const short8* xr =
+            reinterpret_cast<const short8*>(x + (int64_t)(m0 + m) * K);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {       // 4 x 8 values per block
+          const short8 xv = xr[v * 4 + c];
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {     // byte = 2 values
+            const int gb = gq[c * 4 + j], ub = uq[c * 4 + j];
+            const float xa = bits2f(xv.x[2 * j]);
+            const float xb = bits2f(xv.x[2 * j + 1]);
+            ag[m] = fmaf(mxfp4_val(gb & 0xF, ge), xa, ag[m]);
+            ag[m] = fmaf(mxfp4_val(gb >> 4, ge), xb, ag[m]);
+            au[m] = fmaf(mxfp4_val(ub & 0xF, ue), xa, au[m]);
+            au[m] = fmaf(mxfp4_val(ub >> 4, ue), xb, au[m]);
+          }
+        }
+      }
+    }
+  } else if (QMODE == 1) {
     const int4* wg = reinterpret_cast<const int4*>(
         static_cast<const int8_t*>(w) + grow * K);
     const int4* wu = reinterpret_cast<const int4*>(
@@ -138,7 +185,7 @@ __global__ void moe_gateup_kernel(
   }
 }
 
-template <int M, bool Q8, bool PACKED>
+template <int M, int QMODE, bool PACKED>
 __global__ void moe_down_kernel(
     const short* __restrict__ act, const void* __restrict__ w,
     const short* __restrict__ scales, const short* __restrict__ bias,
@@ -154,7 +201,35 @@ __global__ void moe_down_kernel(
   float acc[M];
 #pragma unroll
   for (int m = 0; m < M; ++m) acc[m] = 0.f;
-  if (Q8) {
+  if (QMODE == 2) {
+    const uint8_t* base = static_cast<const uint8_t*>(w);
+    const int4* wr = reinterpret_cast<const int4*>(base + row * (I / 2));
+    const uint8_t* sr = reinterpret_cast<const uint8_t*>(scales) +
+                        row * (I / 32);
+    const int vecs = I / 32;
+    for (int v = lane; v < vecs; v += kWave) {
+      const int4 wv = wr[v];
+      const uint8_t* q = reinterpret_cast<const uint8_t*>(&wv);
+      const int eb = sr[v];
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        const short8* xr = reinterpret_cast<const short8*>(
+            act + ((int64_t)e * Mtot + m0 + m) * I);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const short8 xv = xr[v * 4 + c];
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const int b = q[c * 4 + j];
+            acc[m] = fmaf(mxfp4_val(b & 0xF, eb), bits2f(xv.x[2 * j]),
+                          acc[m]);
+            acc[m] = fmaf(mxfp4_val(b >> 4, eb), bits2f(xv.x[2 * j + 1]),
+                          acc[m]);
+          }
+        }
+      }
+    }
+  } else if (QMODE == 1) {
     const int4* wr = reinterpret_cast<const int4*>(
         static_cast<const int8_t*>(w) + row * I);
     const short* sr = scales + row * (I / G);
@@ -238,13 +313,16 @@ void moe_gateup(torch::Tensor x, torch::Tensor w,
                 double alpha, double limit) {
   const int64_t M = x.size(0), K = x.size(1);
   const int64_t E = w.size(0), I2 = w.size(1), I = I2 / 2;
-  const bool q8 = w.scalar_type() == torch::kInt8;
-  DNET_CHECK(w.size(2) == K && act.size(0) == E && act.size(1) == M &&
-                 act.size(2) == I, "moe_gateup shape");
+  // weight dtype selects the mode: bf16, int8 (kChar), mxfp4 nibble rows
+  // (kByte, [E, 2I, K/2] + e8m0 uint8 scales [E, 2I, K/32])
+  const bool mx4 = w.scalar_type() == torch::kByte;
+  const bool q8 = w.scalar_type() == torch::kChar;
+  DNET_CHECK(w.size(2) == (mx4 ? K / 2 : K) && act.size(0) == E &&
+                 act.size(1) == M && act.size(2) == I, "moe_gateup shape");
   DNET_CHECK(we.size(0) == M && we.size(1) == E, "we shape");
   DNET_CHECK(x.is_contiguous() && w.is_contiguous() && we.is_contiguous() &&
                  act.is_contiguous(), "contig");
-  DNET_CHECK(K % (q8 ? 16 : 8) == 0, "K align");
+  DNET_CHECK(K % (q8 ? 16 : (mx4 ? 32 : 8)) == 0, "K align");
   if (packed) DNET_CHECK(K % 64 == 0 && group % 64 == 0, "packed align");
   const short* sptr = nullptr;
   if (q8) {
@@ -252,6 +330,12 @@ void moe_gateup(torch::Tensor x, torch::Tensor w,
     DNET_CHECK(scales->is_contiguous() && scales->size(0) == E &&
                    scales->size(1) == I2 && scales->size(2) == K / group,
                "scales shape");
+    sptr = (const short*)scales->data_ptr();
+  } else if (mx4) {
+    DNET_CHECK(scales.has_value() && scales->scalar_type() == torch::kByte &&
+                   scales->is_contiguous() && scales->size(0) == E &&
+                   scales->size(1) == I2 && scales->size(2) == K / 32,
+               "mxfp4 scales shape");
     sptr = (const short*)scales->data_ptr();
   }
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
@@ -262,9 +346,9 @@ void moe_gateup(torch::Tensor x, torch::Tensor w,
     const int mt = moe_mtile(M - m0);
     moe_dispatch_m(mt, [&](auto mc) {
       constexpr int MV = decltype(mc)::value;
-      auto launch = [&](auto q8c, auto pkc, auto gluc) {
+      auto launch = [&](auto qmc, auto pkc, auto gluc) {
         hipLaunchKernelGGL(
-            (moe_gateup_kernel<MV, decltype(q8c)::value, decltype(pkc)::value,
+            (moe_gateup_kernel<MV, decltype(qmc)::value, decltype(pkc)::value,
                                decltype(gluc)::value>),
             grid, dim3(kMoeWaves * kWave), 0, stream,
             (const short*)x.data_ptr(), w.data_ptr(), sptr, bptr,
@@ -272,19 +356,21 @@ void moe_gateup(torch::Tensor x, torch::Tensor w,
             (int)I, (int)E, (int)M, (int)m0, (int)std::max<int64_t>(group, 16),
             (float)alpha, (float)limit);
       };
-      auto with_glu = [&](auto q8c, auto pkc) {
+      auto with_glu = [&](auto qmc, auto pkc) {
         if (glu == 1)
-          launch(q8c, pkc, std::integral_constant<int, 1>{});
+          launch(qmc, pkc, std::integral_constant<int, 1>{});
         else
-          launch(q8c, pkc, std::integral_constant<int, 0>{});
+          launch(qmc, pkc, std::integral_constant<int, 0>{});
       };
       if (q8) {
         if (packed)
-          with_glu(std::true_type{}, std::true_type{});
+          with_glu(std::integral_constant<int, 1>{}, std::true_type{});
         else
-          with_glu(std::true_type{}, std::false_type{});
+          with_glu(std::integral_constant<int, 1>{}, std::false_type{});
+      } else if (mx4) {
+        with_glu(std::integral_constant<int, 2>{}, std::false_type{});
       } else {
-        with_glu(std::false_type{}, std::false_type{});
+        with_glu(std::integral_constant<int, 0>{}, std::false_type{});
       }
     });
     m0 += mt;
@@ -295,16 +381,18 @@ void moe_down(torch::Tensor act, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor we,
               torch::Tensor out, int64_t group, bool packed) {
-  const int64_t E = w.size(0), H = w.size(1), I = w.size(2);
-  const int64_t M = act.size(1);
-  const bool q8 = w.scalar_type() == torch::kInt8;
-  DNET_CHECK(act.size(0) == E && act.size(2) == I, "moe_down act shape");
+  const int64_t E = w.size(0), H = w.size(1);
+  const int64_t M = act.size(1), I = act.size(2);
+  const bool mx4 = w.scalar_type() == torch::kByte;
+  const bool q8 = w.scalar_type() == torch::kChar;
+  DNET_CHECK(act.size(0) == E && w.size(2) == (mx4 ? I / 2 : I),
+             "moe_down act/w shape");
   DNET_CHECK(out.size(0) == M && out.size(1) == H &&
                  out.scalar_type() == torch::kFloat, "moe_down out f32");
   DNET_CHECK(we.size(0) == M && we.size(1) == E, "we shape");
   DNET_CHECK(act.is_contiguous() && w.is_contiguous() && we.is_contiguous() &&
                  out.is_contiguous(), "contig");
-  DNET_CHECK(I % (q8 ? 16 : 8) == 0, "I align");
+  DNET_CHECK(I % (q8 ? 16 : (mx4 ? 32 : 8)) == 0, "I align");
   if (packed) DNET_CHECK(I % 64 == 0 && group % 64 == 0, "packed align");
   const short* sptr = nullptr;
   if (q8) {
@@ -312,6 +400,12 @@ void moe_down(torch::Tensor act, torch::Tensor w,
     DNET_CHECK(scales->is_contiguous() && scales->size(0) == E &&
                    scales->size(1) == H && scales->size(2) == I / group,
                "scales shape");
+    sptr = (const short*)scales->data_ptr();
+  } else if (mx4) {
+    DNET_CHECK(scales.has_value() && scales->scalar_type() == torch::kByte &&
+                   scales->is_contiguous() && scales->size(0) == E &&
+                   scales->size(1) == H && scales->size(2) == I / 32,
+               "mxfp4 scales shape");
     sptr = (const short*)scales->data_ptr();
   }
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
@@ -322,9 +416,9 @@ void moe_down(torch::Tensor act, torch::Tensor w,
     const int mt = moe_mtile(M - m0);
     moe_dispatch_m(mt, [&](auto mc) {
       constexpr int MV = decltype(mc)::value;
-      auto launch = [&](auto q8c, auto pkc) {
+      auto launch = [&](auto qmc, auto pkc) {
         hipLaunchKernelGGL(
-            (moe_down_kernel<MV, decltype(q8c)::value, decltype(pkc)::value>),
+            (moe_down_kernel<MV, decltype(qmc)::value, decltype(pkc)::value>),
             grid, dim3(kMoeWaves * kWave), 0, stream,
             (const short*)act.data_ptr(), w.data_ptr(), sptr, bptr,
             (const float*)we.data_ptr(), (float*)out.data_ptr(), (int)I,
@@ -333,15 +427,54 @@ void moe_down(torch::Tensor act, torch::Tensor w,
       };
       if (q8) {
         if (packed)
-          launch(std::true_type{}, std::true_type{});
+          launch(std::integral_constant<int, 1>{}, std::true_type{});
         else
-          launch(std::true_type{}, std::false_type{});
+          launch(std::integral_constant<int, 1>{}, std::false_type{});
+      } else if (mx4) {
+        launch(std::integral_constant<int, 2>{}, std::false_type{});
       } else {
-        launch(std::false_type{}, std::false_type{});
+        launch(std::integral_constant<int, 0>{}, std::false_type{});
       }
     });
     m0 += mt;
   }
+}
+
+__global__ void dequant_mxfp4_kernel(const uint8_t* __restrict__ w,
+                                     const uint8_t* __restrict__ scales,
+                                     short* __restrict__ out,
+                                     const int64_t nblocks, const int bpr) {
+  // one thread = one 32-value block (16 B in, 64 B out)
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < nblocks; i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = i / bpr;
+    const int blk = (int)(i % bpr);
+    const int4 v = reinterpret_cast<const int4*>(w)[i];
+    const uint8_t* q = reinterpret_cast<const uint8_t*>(&v);
+    const int eb = scales[row * bpr + blk];
+    short* o = out + (row * bpr + blk) * 32;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      o[2 * j] = f2bits(mxfp4_val(q[j] & 0xF, eb));
+      o[2 * j + 1] = f2bits(mxfp4_val(q[j] >> 4, eb));
+    }
+  }
+}
+
+void dequant_mxfp4(torch::Tensor w, torch::Tensor scales, torch::Tensor out) {
+  const int64_t rows = w.size(0), kb = w.size(1);
+  DNET_CHECK(w.scalar_type() == torch::kByte &&
+                 scales.scalar_type() == torch::kByte, "mxfp4 dtypes");
+  DNET_CHECK(kb % 16 == 0 && scales.size(1) == kb / 16 &&
+                 out.size(0) == rows && out.size(1) == kb * 2, "shapes");
+  DNET_CHECK(w.is_contiguous() && scales.is_contiguous() &&
+                 out.is_contiguous(), "contig");
+  const int64_t nblocks = rows * (kb / 16);
+  const int grid = (int)std::min<int64_t>((nblocks + 255) / 256, 4096);
+  hipLaunchKernelGGL(dequant_mxfp4_kernel, dim3(grid), dim3(256), 0,
+                     current_stream(), (const uint8_t*)w.data_ptr(),
+                     (const uint8_t*)scales.data_ptr(),
+                     (short*)out.data_ptr(), nblocks, (int)(kb / 16));
 }
 
 }  // namespace dnet

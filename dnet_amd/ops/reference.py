@@ -139,6 +139,56 @@ def gemv_int4(x: torch.Tensor, q4: torch.Tensor, scales: torch.Tensor,
     return y.to(x.dtype)
 
 
+# ---- MXFP4 (OCP microscaling fp4): e2m1 nibbles + one e8m0 (power-of-2
+# uint8, bias 127) scale per 32-element block. The gpt-oss checkpoint
+# format (reference executes it via MLX nn.quantize,
+# src/dnet/core/models/gpt_oss.py:216-287); here it executes natively in
+# the grouped MoE kernels at ~4.25 bits/weight resident.
+
+_E2M1 = torch.tensor([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0])
+
+
+def quantize_mxfp4(w: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """[N, K] -> (packed nibbles [N, K/2] uint8 (even k = low nibble),
+    e8m0 scales [N, K/32] uint8). Round-to-nearest onto the e2m1 grid."""
+    n, k = w.shape
+    assert k % 32 == 0
+    wf = w.float().view(n, k // 32, 32)
+    amax = wf.abs().amax(dim=-1).clamp_min(1e-30)
+    # scale = 2^e with amax/2^e <= 6 (largest e2m1 magnitude)
+    e = torch.ceil(torch.log2(amax / 6.0)).clamp(-127, 127)
+    scales = (e + 127).to(torch.uint8)
+    v = wf / torch.pow(2.0, e).unsqueeze(-1)
+    mag = v.abs().unsqueeze(-1)                     # [n, nb, 32, 1]
+    idx = (mag - _E2M1.to(w.device).view(1, 1, 1, 8)).abs().argmin(-1)
+    code = (idx + torch.where(v < 0, 8, 0)).to(torch.uint8).view(n, k)
+    packed = (code[:, 0::2] | (code[:, 1::2] << 4)).contiguous()
+    return packed, scales.view(n, k // 32).contiguous()
+
+
+def dequant_mxfp4(packed: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    """(nibbles [N, K/2], e8m0 [N, K/32]) -> bf16 [N, K]."""
+    n, kb = packed.shape
+    k = kb * 2
+    code = torch.empty(n, k, dtype=torch.uint8, device=packed.device)
+    code[:, 0::2] = packed & 0xF
+    code[:, 1::2] = packed >> 4
+    lut = _E2M1.to(packed.device)
+    val = lut[(code & 7).long()] * torch.where(code >= 8, -1.0, 1.0)
+    sc = torch.pow(2.0, scales.float() - 127.0)
+    out = val.view(n, k // 32, 32) * sc.unsqueeze(-1)
+    return out.view(n, k).to(torch.bfloat16)
+
+
+def gemv_mxfp4(x: torch.Tensor, packed: torch.Tensor, scales: torch.Tensor,
+               bias: torch.Tensor | None = None) -> torch.Tensor:
+    wd = dequant_mxfp4(packed, scales)
+    y = x.float() @ wd.float().t()
+    if bias is not None:
+        y = y + bias.float()
+    return y.to(x.dtype)
+
+
 def pack_int4_mfma(q4: torch.Tensor) -> torch.Tensor:
     """Permute packed-nibble rows into MFMA chunk-quad order: a lane's
     16 B load covers its 8-elem B slices of FOUR adjacent K=32 chunks.
@@ -298,7 +348,13 @@ def swiglu(gu: torch.Tensor) -> torch.Tensor:
 
 def _moe_dense_w(w: torch.Tensor, scales, group: int, packed: bool
                  ) -> torch.Tensor:
-    """Stacked expert weights [E, N, K] (bf16 or grouped-int8) -> f32."""
+    """Stacked expert weights [E, N, K(/2)] (bf16, grouped-int8, or
+    mxfp4 nibble rows) -> f32."""
+    if w.dtype == torch.uint8:     # mxfp4: [E, N, K/2] + e8m0 [E, N, K/32]
+        e, n, kb = w.shape
+        return dequant_mxfp4(w.reshape(e * n, kb),
+                             scales.reshape(e * n, -1)).float().view(
+                                 e, n, kb * 2)
     if w.dtype != torch.int8:
         return w.float()
     e, n, k = w.shape

@@ -271,13 +271,45 @@ def test_mxfp4_dequant_and_gpt_oss_load():
 
     m1 = GptOssRingModel(cfg, [0], "cpu", True, True)
     m1.load_state_dict(sd_bf)
+    # native path (default): experts stay PACKED (~4.25 bit/weight
+    # resident), dequant is fused in the kernels / done transiently
     m2 = GptOssRingModel(cfg, [0], "cpu", True, True)
     m2.load_state_dict(sd_mx)
+    import dnet_amd.ops as dops
     for e in range(E):
+        eg = m2.layers[0].experts_gateup[e]
+        ed = m2.layers[0].experts_down[e]
+        assert eg.mxfp4 and eg.w.dtype == torch.uint8
         assert torch.equal(m1.layers[0].experts_gateup[e].w,
-                           m2.layers[0].experts_gateup[e].w)
+                           dops.dequant_mxfp4(eg.w, eg.scales))
         assert torch.equal(m1.layers[0].experts_down[e].w,
-                           m2.layers[0].experts_down[e].w)
+                           dops.dequant_mxfp4(ed.w, ed.scales))
+    # resident footprint: nibbles + 1 scale byte / 32 weights
+    assert (m2.layers[0].experts_gateup[0].nbytes()
+            < m1.layers[0].experts_gateup[0].nbytes() / 3)
+
+    # golden: native execution == dequantize-at-load execution
+    import os
+    os.environ["DNET_MXFP4_DEQUANT"] = "1"
+    try:
+        m3 = GptOssRingModel(cfg, [0], "cpu", True, True)
+        m3.load_state_dict(sd_mx)
+    finally:
+        os.environ.pop("DNET_MXFP4_DEQUANT", None)
+    from dnet_amd.models import KVCache
+    toks = torch.randint(0, 64, (1, 5), generator=g)
+    for m in (m2, m3):
+        m.final_norm = m1.final_norm
+        m.lm_head = m1.lm_head
+        m.embed = m1.embed
+    def logits_of(m):
+        kv = m.make_kv_cache(1, 16)
+        h = m.embed_tokens(toks)
+        m.prefill_window(h, [0], kv, 0)
+        return m.normalize_project(h[:, -1].contiguous())
+    l_native, l_deq = logits_of(m2), logits_of(m3)
+    assert torch.allclose(l_native.float(), l_deq.float(), atol=2e-2,
+                          rtol=2e-2), (l_native - l_deq).abs().max()
 
 
 def test_vs_transformers_mixtral():

@@ -652,3 +652,39 @@ def test_slots_graph_churn_matches_eager():
         assert eager[k] == graphed[k], (k, eager[k], graphed[k])
     assert len(graphed["a"]) == 3 and len(graphed["b"]) == 12
     assert len(graphed["c"]) == 6 and len(graphed["d"]) == 5
+
+
+def test_mxfp4_kernels_vs_ref():
+    """Native MXFP4 execution: the dequant kernel and the grouped MoE
+    kernels (fused e2m1+e8m0 dequant) vs the CPU reference."""
+    torch.manual_seed(12)
+    N, K = 96, 256
+    w = torch.randn(N, K) / 4
+    p, s = ops.quantize_mxfp4(w)
+    got = ops.dequant_mxfp4(p.to(_dev()), s.to(_dev())).cpu()
+    want = ref.dequant_mxfp4(p, s)
+    assert torch.equal(got, want)
+
+    # grouped MoE kernels with a packed mxfp4 expert bank
+    E, I, H, M = 4, 64, 128, 5
+    gw = torch.randn(E, 2 * I, H) / 4
+    dw = torch.randn(E, H, I) / 4
+    gp = torch.stack([ops.quantize_mxfp4(gw[e])[0] for e in range(E)])
+    gs = torch.stack([ops.quantize_mxfp4(gw[e])[1] for e in range(E)])
+    dp = torch.stack([ops.quantize_mxfp4(dw[e])[0] for e in range(E)])
+    ds = torch.stack([ops.quantize_mxfp4(dw[e])[1] for e in range(E)])
+    x = torch.randn(M, H, dtype=torch.bfloat16)
+    we = torch.zeros(M, E)
+    we[torch.arange(M), torch.randint(0, E, (M,))] = 1.0
+    gb = torch.randn(E, 2 * I, dtype=torch.bfloat16)
+    db = torch.randn(E, H, dtype=torch.bfloat16)
+    act = ops.moe_gateup(x.to(_dev()), gp.to(_dev()), gs.to(_dev()),
+                         gb.to(_dev()), we.to(_dev()), glu=1)
+    out = ops.moe_down(act, dp.to(_dev()), ds.to(_dev()), db.to(_dev()),
+                       we.to(_dev()))
+    act_ref = ref.moe_gateup(x, gp, gs, gb, we, glu=1)
+    out_ref = ref.moe_down(act_ref, dp, ds, db, we)
+    m = we.bool().any(0)
+    assert torch.allclose(act.cpu().float()[m], act_ref.float()[m],
+                          atol=5e-2, rtol=3e-2)
+    assert torch.allclose(out.cpu(), out_ref, atol=8e-2, rtol=3e-2)
