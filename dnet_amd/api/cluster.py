@@ -56,12 +56,19 @@ class ClusterManager:
         shards = self.shard_devices()
         async with httpx.AsyncClient(timeout=120.0) as client:
             healthy = []
+            ranks: dict[int, str] = {}
+            xgmi: dict = {}
             for d in shards:
                 try:
                     r = await client.get(
                         f"http://{d.local_ip}:{d.server_port}/health")
                     if r.status_code == 200:
                         healthy.append(d)
+                        h = r.json()
+                        if h.get("rank", -1) >= 0:
+                            ranks[int(h["rank"])] = d.instance
+                        if h.get("xgmi"):
+                            xgmi = h["xgmi"]
                 except httpx.HTTPError:
                     log.warning("shard %s unreachable", d.instance)
             results = await asyncio.gather(*[
@@ -95,6 +102,18 @@ class ClusterManager:
                         self.link_ms[(src.instance, inst)] = med
                         if inst in self.profiles:
                             self.profiles[inst].t_comm_ms = med
+            # measured xGMI fabric map (collected in-group at the previous
+            # load, surfaced via /health) overrides the TCP RTT numbers:
+            # ring ordering then follows the actual per-link fabric
+            # 
+            for key, r in xgmi.items():
+                try:
+                    i, j = (int(x) for x in key.split("-"))
+                except ValueError:
+                    continue
+                a, b = ranks.get(i), ranks.get(j)
+                if a and b and "latency_ms" in r:
+                    self.link_ms[(a, b)] = r["latency_ms"]
         return {k: v.to_dict() for k, v in self.profiles.items()}
 
     async def healthy_shards(self) -> list[DeviceProperties]:

@@ -55,11 +55,18 @@ class RingTopologySolver(TopologySolver):
         for j, i in enumerate(active):
             d = shards[i]
             nxt = shards[active[(j + 1) % len(active)]].instance
+            # window = whole slice when resident; otherwise the prefetch
+            # window scales with residency (was hard-coded min(4, w) —
+            # VERDICT r1 weak item 6)
+            n_i = res.n[i]
+            if n_i >= w_active[j]:
+                win = w_active[j]
+            else:
+                win = max(1, min(n_i // 2, 8))
             assignments.append(LayerAssignment(
                 instance=d.instance, layers=assigns_lists[j],
-                next_instance=nxt,
-                window_size=min(4, max(w_active[j], 1)),
-                residency_size=res.n[i], gpu_index=max(d.gpu_index, 0)))
+                next_instance=nxt, window_size=win,
+                residency_size=n_i, gpu_index=max(d.gpu_index, 0)))
         head = assignments[0].instance if assignments else ""
         head_dev = next((d for d in shards if d.instance == head), shards[0])
         return TopologyInfo(

@@ -11,15 +11,16 @@ Decode is weight-read bound, so per-layer cost on device i is
     c_i = layer_bytes/hbm_bw_i              (resident layer)
     c_i' = layer_bytes/h2d_bw_i             (host-swapped layer; copy-stream
                                              overlap hides compute)
-The makespan objective min max_i t_i with identical layers is solved exactly
-by greedy list assignment on uniform machines; residency n_i then follows
-from HBM capacity, swapped layers re-priced at c_i', and a second greedy
-pass rebalances if swapping changed the ordering. k>1 (multiple ring rounds)
-is chosen so each round's window fits residency.
+The makespan objective min max_i t_i (bottleneck stage = pipelined decode
+throughput) is solved EXACTLY by a convex-cost DP over (device, count),
+with total time (single-stream latency) as the lexicographic tiebreak;
+residency n_i follows from HBM capacity with swapped layers priced at
+c_i'. k>1 (multiple ring rounds) is chosen so each round's window fits
+residency.
 """
 from __future__ import annotations
 
-import heapq
+
 from dataclasses import dataclass, field
 
 from .profiler import DeviceProfile
@@ -55,25 +56,46 @@ def halda_solve(devices: list[DeviceProfile], num_layers: int,
         usable = max(d.hbm_free_gb - overhead_gb, 0.5) * 1e9
         cap_layers.append(max(int(usable // max(layer_bytes + kv_bytes_per_layer, 1)), 1))
 
-    def per_layer_cost(i: int, count: int) -> float:
-        """Average per-layer cost if device i holds `count` layers."""
+    def dev_time(i: int, count: int) -> float:
+        """Per-token time if device i holds `count` layers (resident up
+        to capacity, host-swapped beyond)."""
         resident = min(count, cap_layers[i])
-        swapped = count - resident
-        return (resident * c_res[i] + swapped * c_swap[i]) / max(count, 1)
+        return resident * c_res[i] + (count - resident) * c_swap[i]
 
-    # Greedy list scheduling on uniform machines: repeatedly give the next
-    # layer to the device whose finish time after taking it is smallest.
+    # Exact min-makespan DP over (device, layer-count) — O(nd * L^2),
+    # trivially cheap for <=16 devices x <=128 layers. Minimizes the
+    # bottleneck stage time (pipelined decode throughput), with total
+    # time (single-stream latency) as the lexicographic tiebreak.
+    # (VERDICT r1 item 7: the round-1 greedy list scheduler was exact
+    # only for identical layers on a homogeneous node; this is optimal
+    # for any profile mix, asserted vs brute force in tests.)
+    L = num_layers
+    INF = float("inf")
+    f = [(INF, INF)] * (L + 1)
+    f[0] = (0.0, 0.0)
+    choice = [[0] * (L + 1) for _ in range(nd)]
+    for i in range(nd):
+        g = [(INF, INF)] * (L + 1)
+        for l in range(L + 1):
+            best = (INF, INF)
+            bx = 0
+            for x in range(l + 1):
+                prev = f[l - x]
+                if prev[0] == INF:
+                    continue
+                t = dev_time(i, x)
+                cand = (max(prev[0], t), prev[1] + t)
+                if cand < best:
+                    best = cand
+                    bx = x
+            g[l] = best
+            choice[i][l] = bx
+        f = g
     w = [0] * nd
-    heap = [(c_res[i], i) for i in range(nd)]
-    heapq.heapify(heap)
-    for _ in range(num_layers):
-        _, i = heapq.heappop(heap)
-        w[i] += 1
-        nxt = w[i] + 1
-        resident = min(nxt, cap_layers[i])
-        t_next = (min(nxt, cap_layers[i]) * c_res[i]
-                  + max(nxt - cap_layers[i], 0) * c_swap[i])
-        heapq.heappush(heap, (t_next, i))
+    rem = L
+    for i in range(nd - 1, -1, -1):
+        w[i] = choice[i][rem]
+        rem -= w[i]
 
     n = [min(w[i], cap_layers[i]) for i in range(nd)]
     # k rounds: if any device swaps, use enough rounds that one round's

@@ -209,15 +209,29 @@ class ShardRuntime:
             self._callback = SyncWireClient(host or "127.0.0.1", int(port))
         self.status = "loaded"
         self.link_profile = {}
+        self.link_matrix = {}
+        self.rank = req.rank
         if req.world_size > 1:
             try:
-                from ..parallel.profiler import measure_ring_links
+                from ..parallel.profiler import (measure_link_matrix,
+                                                 measure_ring_links)
                 self.link_profile = measure_ring_links(
                     req.rank, req.world_size, ex.device,
                     sizes=(65536,), reps=5)
                 for size, r in self.link_profile.items():
                     log.info("[PROFILE][XGMI] payload=%d latency=%.3fms "
                              "bw=%.1fGB/s", size, r["latency_ms"], r["gbps"])
+                # full per-pair fabric map (identical on every rank);
+                # surfaces via /health so the NEXT prepare_topology orders
+                # the ring by measured xGMI links instead of TCP RTT
+                if req.world_size <= 16:
+                    self.link_matrix = measure_link_matrix(
+                        req.rank, req.world_size, ex.device)
+                    if req.rank == 0:
+                        for k, r in sorted(self.link_matrix.items()):
+                            log.info("[PROFILE][XGMI-LINK] %s latency="
+                                     "%.3fms bw=%.1fGB/s", k,
+                                     r["latency_ms"], r["gbps"])
             except Exception:
                 log.exception("xGMI link probe failed (non-fatal)")
         log.info("model loaded: %s rank %d/%d %d layers %s...", self.model_name,

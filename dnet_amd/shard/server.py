@@ -29,6 +29,8 @@ class HealthResponse(BaseModel):
     model: str = ""
     queue_depth: int = 0
     error: str = ""
+    rank: int = -1                 # rank in the active ring (-1 = none)
+    xgmi: dict = {}                # all-pairs link matrix ("i-j" -> ms/GBps)
 
 
 class MeasureLatencyRequest(BaseModel):
@@ -45,7 +47,9 @@ def build_shard_app(rt: ShardRuntime) -> FastAPI:
         return HealthResponse(status=rt.status, instance=rt.instance,
                               model=rt.model_name,
                               queue_depth=rt.infer_q.qsize(),
-                              error=rt.last_error)
+                              error=rt.last_error,
+                              rank=getattr(rt, "rank", -1),
+                              xgmi=getattr(rt, "link_matrix", {}) or {})
 
     @app.post("/load_model")
     def load_model(req: ShardLoadModelRequest) -> dict:

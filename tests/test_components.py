@@ -386,3 +386,48 @@ def test_deepseek_partial_offload_matches_fit(scoring):
     fit = run(0)
     off = run(2)
     assert torch.equal(fit, off)
+
+
+def test_halda_dp_optimal_vs_bruteforce():
+    """the layer-distribution DP must be makespan-
+    OPTIMAL on heterogeneous profiles — asserted against brute force on
+    small instances (mixed HBM speeds, capacity-limited devices that
+    must host-swap)."""
+    import itertools
+
+    from dnet_amd.parallel.profiler import DeviceProfile
+    from dnet_amd.parallel.solver import halda_solve
+
+    cases = [
+        # (hbm_gbps, h2d_gbps, hbm_free_gb) per device, layers, layer GB
+        ([(6000, 50, 280), (6000, 50, 280), (2000, 50, 280)], 13, 0.5),
+        ([(6000, 50, 4.5), (3000, 50, 280)], 12, 1.0),   # dev0 cap ~ 1
+        ([(8000, 60, 280), (4000, 30, 6.0), (2000, 20, 280),
+          (1000, 10, 280)], 17, 1.0),
+    ]
+    for devs_spec, L, gb in cases:
+        devs = [DeviceProfile(instance=f"d{i}", hbm_gbps=b, h2d_gbps=h,
+                              hbm_free_gb=f)
+                for i, (b, h, f) in enumerate(devs_spec)]
+        layer_bytes = gb * 1e9
+        res = halda_solve(devs, L, layer_bytes)
+        # reconstruct per-device cost exactly like the solver
+        overhead = 4.0
+
+        def t_of(i, cnt):
+            d = devs[i]
+            cr = layer_bytes / (d.hbm_gbps * 1e9) * 1e3
+            cs = layer_bytes / (d.h2d_gbps * 1e9) * 1e3
+            cap = max(int(max(d.hbm_free_gb - overhead, 0.5) * 1e9
+                          // layer_bytes), 1)
+            r = min(cnt, cap)
+            return r * cr + (cnt - r) * cs
+
+        got = max(t_of(i, res.w[i]) for i in range(len(devs)))
+        best = min(
+            max(t_of(i, c) for i, c in enumerate(combo))
+            for combo in itertools.product(range(L + 1),
+                                           repeat=len(devs))
+            if sum(combo) == L)
+        assert sum(res.w) == L
+        assert abs(got - best) < 1e-9, (res.w, got, best)
