@@ -431,3 +431,42 @@ def test_halda_dp_optimal_vs_bruteforce():
             if sum(combo) == L)
         assert sum(res.w) == L
         assert abs(got - best) < 1e-9, (res.w, got, best)
+
+
+def test_mla_latent_cache_matches_perhead():
+    """the compressed (latent) MLA KV cache — c_kv +
+    shared roped key, 576/token instead of nh*(192+128) — must produce
+    the per-head path's tokens via weight absorption, at ~nh*320/576 the
+    cache footprint."""
+    from dnet_amd.models import ModelConfig
+    from dnet_amd.parallel.ring import RingExecutor
+
+    hf = dict(model_type="deepseek_v2", hidden_size=64, num_hidden_layers=3,
+              num_attention_heads=8, num_key_value_heads=8, vocab_size=128,
+              intermediate_size=64, kv_lora_rank=32, qk_nope_head_dim=16,
+              qk_rope_head_dim=8, v_head_dim=16, n_routed_experts=4,
+              num_experts_per_tok=2, n_shared_experts=1,
+              moe_intermediate_size=32, first_k_dense_replace=1,
+              routed_scaling_factor=1.0, rope_theta=10000.0)
+    cfg = ModelConfig.from_hf(hf)
+    toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
+                         generator=torch.Generator().manual_seed(9))
+
+    def run(perhead):
+        if perhead:
+            os.environ["DNET_MLA_PERHEAD"] = "1"
+        try:
+            ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=2,
+                              smax=32, seed=21, use_graphs=False)
+            first = ex.prefill(toks.clone(), chunk=4)  # continuation path
+            gen = ex.decode_rounds(5)
+            return (torch.cat([first.unsqueeze(-1), gen], dim=-1),
+                    ex.kvs[0].nbytes())
+        finally:
+            os.environ.pop("DNET_MLA_PERHEAD", None)
+
+    lat, lat_bytes = run(False)
+    ph, ph_bytes = run(True)
+    assert torch.equal(lat, ph), (lat, ph)
+    # nh*(nope+rope+vd) = 8*40 = 320 per token vs lora+rope = 40
+    assert lat_bytes * 4 < ph_bytes
