@@ -662,19 +662,27 @@ class RingModel:
                 # sequence-sharded KV: write locally, attend via gathered
                 # flash-decode partials (numerically = full attention)
                 from ..parallel.context import cp_attn_decode, local_lengths
-                assert not window, "sliding window + CP is roadmap"
                 wpos = (kv.pos - self.cp_rank * self.cp_cap).int()
                 ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos,
                                 self.cos, self.sin,
                                 kv.kscale[li] if kv.quantized else None,
                                 kv.vscale[li] if kv.quantized else None,
                                 wpos=wpos)
-                ln = local_lengths(kv.pos + 1, self.cp_cap, self.cp_rank)
-                attn = cp_attn_decode(
-                    q, kv.k[li], kv.v[li], ln, d ** -0.5,
-                    group=self.cp_group, sinks=sinks,
-                    kscale=kv.kscale[li] if kv.quantized else None,
-                    vscale=kv.vscale[li] if kv.quantized else None)
+                if window and window > 0:
+                    from ..parallel.context import cp_attn_decode_windowed
+                    attn = cp_attn_decode_windowed(
+                        q, kv.k[li], kv.v[li], kv.pos, d ** -0.5,
+                        self.cp_cap, self.cp_rank, window,
+                        group=self.cp_group, sinks=sinks,
+                        kscale=kv.kscale[li] if kv.quantized else None,
+                        vscale=kv.vscale[li] if kv.quantized else None)
+                else:
+                    ln = local_lengths(kv.pos + 1, self.cp_cap, self.cp_rank)
+                    attn = cp_attn_decode(
+                        q, kv.k[li], kv.v[li], ln, d ** -0.5,
+                        group=self.cp_group, sinks=sinks,
+                        kscale=kv.kscale[li] if kv.quantized else None,
+                        vscale=kv.vscale[li] if kv.quantized else None)
             else:
                 ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos,
                                 self.sin,
@@ -769,12 +777,16 @@ class RingModel:
                         kv.k[li][:, :, ls0 - r * cap:ls1 - r * cap] = ksl
                         kv.v[li][:, :, ls0 - r * cap:ls1 - r * cap] = vsl
                 window, sinks = self._attn_params(lid, lw)
-                assert not window, "sliding window + CP is roadmap"
-                kfull = self._cp_gather(kv.k_deq(li))[:, :, :p0 + T]
-                vfull = self._cp_gather(kv.v_deq(li))[:, :, :p0 + T]
-                attn = _chunked_causal_attention(
-                    q.transpose(1, 2), kfull, vfull, d ** -0.5, p0,
-                    window, sinks)
+                # gather-free: each rank computes partials of this query
+                # chunk vs its LOCAL shard; the full KV is never
+                # materialized (transient memory T x cap per layer)
+                from ..parallel.context import cp_prefill_attention
+                attn = cp_prefill_attention(
+                    q.transpose(1, 2).contiguous(), kv.k[li], kv.v[li],
+                    p0 + T, p0, d ** -0.5, self.cp_cap, self.cp_rank,
+                    window=window, group=self.cp_group, sinks=sinks,
+                    kscale=kv.kscale[li] if kv.quantized else None,
+                    vscale=kv.vscale[li] if kv.quantized else None)
             else:
                 if kv.quantized:
                     from ..ops import reference as _r
