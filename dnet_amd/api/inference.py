@@ -33,6 +33,16 @@ class InferenceManager:
         self.pending: dict[str, asyncio.Queue] = {}
         self.head_client: Optional[WireClient] = None
         self.callback_addr: str = ""
+        # set by the API server when auto-recover is enabled: invoked on
+        # error frames / token timeouts (fire-and-forget)
+        self.on_failure = None
+
+    def _notify_failure(self):
+        if self.on_failure is not None:
+            try:
+                self.on_failure()
+            except Exception:
+                log.exception("on_failure hook failed")
 
     def connect_head(self, host: str, port: int, callback_addr: str):
         self.head_client = WireClient(host, port)
@@ -107,9 +117,14 @@ class InferenceManager:
             yield ChatChunkModel(id=nonce, model=request.model, choices=[
                 StreamChoice(delta=ChoiceDelta(role="assistant", content=""))])
             while True:
-                frame = await asyncio.wait_for(q.get(),
-                                               timeout=self.token_timeout_s)
+                try:
+                    frame = await asyncio.wait_for(
+                        q.get(), timeout=self.token_timeout_s)
+                except asyncio.TimeoutError:
+                    self._notify_failure()
+                    raise
                 if frame.get("t") == "error":
+                    self._notify_failure()
                     raise RuntimeError(frame.get("error", "ring error"))
                 tid = int(frame["token_id"])
                 n_tokens += 1

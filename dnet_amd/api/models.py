@@ -129,6 +129,9 @@ class PrepareTopologyManualRequest(BaseModel):
 class APILoadModelRequest(BaseModel):
     model: str
     quant: str = ""
-    max_batch: int = 1
+    # >1 = slot-scheduled continuous batching (the production serving
+    # mode: concurrent requests each own a KV slot; per-slot sampling,
+    # seeds, logprobs and hipGraph decode all supported). 1 = serial.
+    max_batch: int = 8
     max_seq: int = 4096
     warmup: bool = False

@@ -7,6 +7,8 @@ Reference counterpart: src/dnet/api/http_api.py (routes /health,
 """
 from __future__ import annotations
 
+import asyncio
+
 import json
 from typing import Optional
 
@@ -136,8 +138,7 @@ def build_api_app(state: ApiState) -> FastAPI:
         await s.models.unload_model()
         return {"status": "ok"}
 
-    @app.post("/v1/recover")
-    async def recover():
+    async def _do_recover():
         """Elastic recovery after a shard failure: health-sweep, exclude
         dead shards, re-solve the ring over the survivors and reload the
         last-loaded model (drop-and-reload; in-flight requests error out
@@ -167,6 +168,31 @@ def build_api_app(state: ApiState) -> FastAPI:
         s.inference.connect_head(head.local_ip, head.shard_port, cb)
         return {"status": "ok", "excluded": sorted(s.cluster.excluded),
                 "topology": s.cluster.topology.model_dump()}
+
+    @app.post("/v1/recover")
+    async def recover():
+        return await _do_recover()
+
+    # failure -> automatic re-solve (recovery was
+    # operator-only): the inference manager reports error frames / token
+    # timeouts; one debounced background recover runs at a time
+    if s.settings.api.auto_recover:
+        async def _auto_recover():
+            if getattr(s, "_recovering", False):
+                return
+            s._recovering = True
+            try:
+                log.warning("auto-recover: failure reported, re-solving")
+                out = await _do_recover()
+                log.warning("auto-recover done: excluded=%s",
+                            out.get("excluded"))
+            except Exception:
+                log.exception("auto-recover failed")
+            finally:
+                s._recovering = False
+
+        s.inference.on_failure = lambda: asyncio.get_event_loop(
+            ).create_task(_auto_recover())
 
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatRequestModel):

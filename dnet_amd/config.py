@@ -44,6 +44,10 @@ class ApiSettings(BaseModel):
     grpc_port: int = 50051          # token-callback data plane
     callback_addr: str = ""          # override for SendToken target
     request_timeout_s: float = 300.0  # per-token timeout (reference: 300 s)
+    # on an error frame or token timeout, automatically health-sweep,
+    # exclude dead shards, re-solve and reload (the /v1/recover flow) —
+    # the reference never recovers (RingError defined but unsent)
+    auto_recover: bool = True
 
 
 class ShardSettings(BaseModel):
