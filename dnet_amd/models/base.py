@@ -102,14 +102,31 @@ class Linear:
         l.bits = 4
         return l
 
+    def dequant(self) -> torch.Tensor:
+        if self.mxfp4:
+            return ops.dequant_mxfp4(self.w, self.scales)
+        if self.bits == 4:
+            return ops.dequant_int4(self.w, self.scales, self.group,
+                                    self.packed)
+        if self.is_quant:
+            return ops.dequant_int8(self.w, self.scales, self.group,
+                                    self.packed)
+        return self.w
+
+    _wd = None   # transient dequantized weights (prefill side-stream
+                 # prefetch; consumed once by the next big-M call)
+
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         m = x.shape[0]
         if self.mxfp4:
-            y = x @ ops.dequant_mxfp4(self.w, self.scales).t()
+            wd, self._wd = (self._wd, None) if self._wd is not None \
+                else (self.dequant(), None)
+            y = x @ wd.t()
             if self.bias is not None:
                 y = y + self.bias
             return y
         if m <= GEMV_MAX_M:
+            self._wd = None
             if self.bits == 4:
                 return ops.gemv_int4(x, self.w, self.scales, self.group,
                                      self.bias, self.packed)
@@ -117,12 +134,8 @@ class Linear:
                 return ops.gemv_int8(x, self.w, self.scales, self.group,
                                      self.bias, self.packed)
             return ops.gemv_bf16(x, self.w, self.bias)
-        if self.bits == 4:
-            wd = ops.dequant_int4(self.w, self.scales, self.group, self.packed)
-        elif self.is_quant:
-            wd = ops.dequant_int8(self.w, self.scales, self.group, self.packed)
-        else:
-            wd = self.w
+        wd, self._wd = (self._wd, None) if self._wd is not None \
+            else (self.dequant(), None)
         y = x @ wd.t()
         if self.bias is not None:
             y = y + self.bias
@@ -729,6 +742,30 @@ class RingModel:
         a = ops.swiglu(gu)
         return lw.down(a)
 
+    def _dequant_prefetch(self, layer_ids: Sequence[int], i: int):
+        """Side-stream dequant of layer i+1's dense weights while layer i
+        computes (prefill only; ~65 ms of the 64x128 TTFT was serial
+        dequant). Returns an event the consumer waits on, or None."""
+        if (self.device.type != "cuda" or i + 1 >= len(layer_ids)
+                or self.weight_provider is not None):  # offload streams
+            return None
+        nxt = self.layers.get(layer_ids[i + 1])
+        if nxt is None:
+            return None
+        if self._dq_stream is None:
+            self._dq_stream = torch.cuda.Stream()
+        ev = torch.cuda.Event()
+        with torch.cuda.stream(self._dq_stream):
+            for lin in (nxt.qkv, nxt.o, nxt.gateup, nxt.down):
+                if lin is not None and lin.is_quant:
+                    wd = lin.dequant()
+                    wd.record_stream(torch.cuda.current_stream())
+                    lin._wd = wd
+            ev.record(self._dq_stream)
+        return ev
+
+    _dq_stream = None
+
     def prefill_window(self, h: torch.Tensor, layer_ids: Sequence[int],
                        kv: KVCache, p0: int) -> torch.Tensor:
         """Prefill T tokens. h: [B, T, H]; tokens occupy positions p0..p0+T-1.
@@ -739,8 +776,13 @@ class RingModel:
         B, T, H = h.shape
         nq, nkv, d = self.nq_local, self.nkv_local, c.head_dim
         positions = torch.arange(p0, p0 + T, device=h.device)
-        for lid in layer_ids:
+        layer_ids = list(layer_ids)
+        dq_ev = self._dequant_prefetch(layer_ids, -1)  # prefetch layer 0
+        for i, lid in enumerate(layer_ids):
             lw = self._layer(lid)
+            if dq_ev is not None:
+                torch.cuda.current_stream().wait_event(dq_ev)
+            dq_ev = self._dequant_prefetch(layer_ids, i)
             flat = h.view(B * T, H)
             y = ops.rmsnorm(flat, None, lw.attn_norm, c.rms_eps)
             qkv = lw.qkv(y)

@@ -63,7 +63,10 @@ _scratch: dict = {}
 def _get_scratch(device) -> torch.Tensor:
     key = str(device)
     if key not in _scratch:
-        _scratch[key] = torch.empty(64 * 8192, dtype=torch.float32, device=device)
+        # zero-initialized: the split-k path relies on it (the combine
+        # kernel re-zeroes after each use instead of a per-launch memset)
+        _scratch[key] = torch.zeros(64 * 8192, dtype=torch.float32,
+                                    device=device)
     return _scratch[key]
 
 

@@ -634,13 +634,16 @@ __global__ __launch_bounds__(256, MINW) void gemm_m16_stream_kernel(
   }
 }
 
-__global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
+__global__ void f32_to_bf16_bias_kernel(float* __restrict__ in,
                                         const short* __restrict__ bias,
                                         short* __restrict__ out,
                                         const int64_t total, const int N) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
     float v = in[i];
+    in[i] = 0.f;   // re-zero for the next split-k use: the combine pass
+                   // replaces a per-launch hipMemsetAsync of the scratch
+                   // (~3 launches + full passes per layer per step)
     if (bias != nullptr) v += bits2f(bias[i % N]);
     out[i] = f2bits(v);
   }
@@ -685,10 +688,10 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
   short* op = (short*)out.data_ptr() + m0 * N;
   float* fp = nullptr;
   if (sk > 1) {
+    // scratch is allocated zeroed and re-zeroed by the combine kernel
     TORCH_CHECK(scratch.has_value() && scratch->numel() >= (int64_t)M * N,
                 "split-k scratch required");
     fp = (float*)scratch->data_ptr();
-    DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * M * N, stream));
   }
   const short* sp = bits < 16 ? (const short*)scales->data_ptr() : nullptr;
   const short* bp1 = sk > 1 ? nullptr : bptr;
