@@ -35,7 +35,8 @@ def main():
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--model", default="qwen-2.5-32b")
-    ap.add_argument("--quant", default="int8", choices=["int8", "int4", "bf16"])
+    ap.add_argument("--quant", default="int8",
+                    choices=["int8", "int4", "bf16", "mxfp4"])
     ap.add_argument("--mb-size", type=int, default=64,
                     help="sequences per microbatch")
     ap.add_argument("--mb-per-rank", type=int, default=0,
@@ -67,6 +68,8 @@ def main():
         quant = QuantConfig(8, 128)
     elif args.quant == "int4":
         quant = QuantConfig(4, 128)
+    elif args.quant == "mxfp4":
+        quant = QuantConfig(4, 32, fmt="mxfp4")   # expert banks only
     hf = dict(PRESETS[args.model])
     if args.layers:
         hf["num_hidden_layers"] = args.layers
@@ -150,7 +153,10 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if quant is None else f"int{quant.bits}-g{quant.group} weights (bf16 compute)",
+            "dtype": ("bf16" if quant is None else
+                      ("mxfp4 experts (bf16 attention/compute)"
+                       if quant.fmt == "mxfp4" else
+                       f"int{quant.bits}-g{quant.group} weights (bf16 compute)")),
             "data": "synthetic (random tokens, random-init weights)",
             "ttft_ms": round(ttft_ms, 1),
             "ttft_samples_ms": [round(t, 1) for t in ttft_samples],

@@ -69,6 +69,12 @@ class Linear:
     @classmethod
     def make(cls, w: torch.Tensor, bias: Optional[torch.Tensor],
              quant: Optional[QuantConfig]) -> "Linear":
+        if quant is not None and quant.fmt == "mxfp4":
+            if w.shape[1] % 32:
+                return cls(w.to(torch.bfloat16), bias)
+            packed, scales = ops.quantize_mxfp4(w.float())
+            return cls.make_mxfp4(packed.to(w.device), scales.to(w.device),
+                                  bias)
         if quant is not None and quant.bits == 8:
             q, s = ops.quantize_int8(w, quant.group)
             packed = (w.is_cuda and w.shape[1] % 64 == 0

@@ -11,12 +11,17 @@ from typing import Optional
 
 @dataclass
 class QuantConfig:
-    """Grouped symmetric int8 weight quantization (W8A16)."""
+    """Weight quantization: grouped symmetric int8/int4 (W8A16/W4A16), or
+    OCP MXFP4 (fmt="mxfp4": e2m1 nibbles + e8m0 32-block scales — applied
+    to MoE expert banks only, like gpt-oss checkpoints)."""
     bits: int = 8
     group: int = 128
+    fmt: str = ""          # "" = grouped-int, "mxfp4"
 
     @property
     def name(self) -> str:
+        if self.fmt == "mxfp4":
+            return "mxfp4"
         return f"int{self.bits}-g{self.group}"
 
 

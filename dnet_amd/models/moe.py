@@ -91,8 +91,19 @@ class MoERingModel(RingModel):
     model_types = ["mixtral", "qwen2_moe", "qwen3_moe"]
 
     def _init_layer(self, rand, lid: int) -> LayerWeights:
-        lw = super()._init_layer(rand, lid)
+        # mxfp4 applies to the EXPERT banks only (gpt-oss checkpoint
+        # shape); attention/router stay bf16 — so the base init runs
+        # unquantized in that mode
         c = self.cfg
+        mx = c.quant is not None and c.quant.fmt == "mxfp4"
+        if mx:
+            import copy
+            base_self = copy.copy(self)
+            base_self.cfg = copy.copy(c)
+            base_self.cfg.quant = None
+            lw = RingModel._init_layer(base_self, rand, lid)
+        else:
+            lw = super()._init_layer(rand, lid)
         inter = c.moe_intermediate_size or c.intermediate_size
         dev = self.device
         lw.gateup = None

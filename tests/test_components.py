@@ -315,7 +315,8 @@ def test_row_sampler():
 
 
 @pytest.mark.parametrize("model_type,qbits", [
-    ("mixtral", 0), ("mixtral", 8), ("gpt_oss", 0), ("qwen2_moe", 0)])
+    ("mixtral", 0), ("mixtral", 8), ("gpt_oss", 0), ("gpt_oss", "mxfp4"),
+    ("qwen2_moe", 0)])
 def test_moe_offload_matches_fit(model_type, qbits):
     """MoE layers stream through the weight cache (stacked expert banks in
     the slot template) and produce the fit path's exact tokens."""
@@ -335,7 +336,8 @@ def test_moe_offload_matches_fit(model_type, qbits):
         hf["shared_expert_intermediate_size"] = 48
         hf["norm_topk_prob"] = True
     from dnet_amd.models import QuantConfig
-    quant = QuantConfig(8, 16) if qbits == 8 else None
+    quant = (QuantConfig(8, 16) if qbits == 8 else
+             QuantConfig(4, 32, fmt="mxfp4") if qbits == "mxfp4" else None)
     cfg = ModelConfig.from_hf(hf, quant=quant)
     toks = torch.randint(0, cfg.vocab_size, (1, 2, 6),
                          generator=torch.Generator().manual_seed(5))
