@@ -26,6 +26,8 @@ namespace dnet {
 
 constexpr int kMoeWaves = 4;
 
+using i32x4 = __attribute__((ext_vector_type(4))) int;
+
 template <int M>
 __device__ __forceinline__ bool expert_routed(const float* __restrict__ we,
                                               const int E, const int e,
@@ -89,26 +91,40 @@ __global__ void moe_gateup_kernel(
     const uint8_t* su = sb + urow * (K / 32);
     const int vecs = K / 32;   // one 16 B load = one 32-value block
     for (int v = lane; v < vecs; v += kWave) {
-      const int4 gv = wg[v], uv = wu[v];
+      i32x4 gv = *reinterpret_cast<const i32x4*>(&wg[v]);
+      i32x4 uv = *reinterpret_cast<const i32x4*>(&wu[v]);
+      // volatile tie: stops the software pipeliner from pre-extracting
+      // the NEXT iterations' nibbles (it spilled 200+ regs otherwise);
+      // the loads themselves still prefetch ahead
+      asm volatile("" : "+v"(gv), "+v"(uv));
       const uint8_t* gq = reinterpret_cast<const uint8_t*>(&gv);
       const uint8_t* uq = reinterpret_cast<const uint8_t*>(&uv);
       const int ge = sg[v], ue = su[v];
+      // dequant per 8-value chunk, outside the M loop (whole-block
+      // arrays at gateup's 2 rows = 64 floats blew past 128 VGPR; doing
+      // it inside the M loop hoisted M copies -> 1000+ scratch spills).
+      // c is a REAL loop (unroll 1): only one chunk's dequant + x
+      // values are live at a time
+#pragma unroll 1
+      for (int c = 0; c < 4; ++c) {         // 4 x 8 values per block
+        float gd[8], ud[8];
 #pragma unroll
-      for (int m = 0; m < M; ++m) {
-        const short8* xr =
-            reinterpret_cast<const short8*>(x + (int64_t)(m0 + m) * K);
+        for (int j = 0; j < 4; ++j) {
+          const int gb = gq[c * 4 + j], ub = uq[c * 4 + j];
+          gd[2 * j] = mxfp4_val(gb & 0xF, ge);
+          gd[2 * j + 1] = mxfp4_val(gb >> 4, ge);
+          ud[2 * j] = mxfp4_val(ub & 0xF, ue);
+          ud[2 * j + 1] = mxfp4_val(ub >> 4, ue);
+        }
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {       // 4 x 8 values per block
-          const short8 xv = xr[v * 4 + c];
+        for (int m = 0; m < M; ++m) {
+          const short8 xv = reinterpret_cast<const short8*>(
+              x + (int64_t)(m0 + m) * K)[v * 4 + c];
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {     // byte = 2 values
-            const int gb = gq[c * 4 + j], ub = uq[c * 4 + j];
-            const float xa = bits2f(xv.x[2 * j]);
-            const float xb = bits2f(xv.x[2 * j + 1]);
-            ag[m] = fmaf(mxfp4_val(gb & 0xF, ge), xa, ag[m]);
-            ag[m] = fmaf(mxfp4_val(gb >> 4, ge), xb, ag[m]);
-            au[m] = fmaf(mxfp4_val(ub & 0xF, ue), xa, au[m]);
-            au[m] = fmaf(mxfp4_val(ub >> 4, ue), xb, au[m]);
+          for (int j = 0; j < 8; ++j) {
+            const float xa = bits2f(xv.x[j]);
+            ag[m] = fmaf(gd[j], xa, ag[m]);
+            au[m] = fmaf(ud[j], xa, au[m]);
           }
         }
       }
@@ -208,24 +224,26 @@ __global__ void moe_down_kernel(
                         row * (I / 32);
     const int vecs = I / 32;
     for (int v = lane; v < vecs; v += kWave) {
-      const int4 wv = wr[v];
+      i32x4 wv = *reinterpret_cast<const i32x4*>(&wr[v]);
+      asm volatile("" : "+v"(wv));   // see gateup: stops pre-extraction
       const uint8_t* q = reinterpret_cast<const uint8_t*>(&wv);
       const int eb = sr[v];
+#pragma unroll 1
+      for (int c = 0; c < 4; ++c) {
+        float qd[8];
 #pragma unroll
-      for (int m = 0; m < M; ++m) {
-        const short8* xr = reinterpret_cast<const short8*>(
-            act + ((int64_t)e * Mtot + m0 + m) * I);
+        for (int j = 0; j < 4; ++j) {
+          const int b = q[c * 4 + j];
+          qd[2 * j] = mxfp4_val(b & 0xF, eb);
+          qd[2 * j + 1] = mxfp4_val(b >> 4, eb);
+        }
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          const short8 xv = xr[v * 4 + c];
+        for (int m = 0; m < M; ++m) {
+          const short8 xv = reinterpret_cast<const short8*>(
+              act + ((int64_t)e * Mtot + m0 + m) * I)[v * 4 + c];
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            const int b = q[c * 4 + j];
-            acc[m] = fmaf(mxfp4_val(b & 0xF, eb), bits2f(xv.x[2 * j]),
-                          acc[m]);
-            acc[m] = fmaf(mxfp4_val(b >> 4, eb), bits2f(xv.x[2 * j + 1]),
-                          acc[m]);
-          }
+          for (int j = 0; j < 8; ++j)
+            acc[m] = fmaf(qd[j], bits2f(xv.x[j]), acc[m]);
         }
       }
     }
@@ -300,7 +318,8 @@ static void moe_dispatch_m(int M, LaunchFn&& fn) {
   }
 }
 
-static int moe_mtile(int64_t rem) {
+static int moe_mtile(int64_t rem, bool mx4 = false) {
+  if (mx4 && rem >= 6) return 6;   // M=8 mxfp4 spills ~30 regs
   if (rem >= 8) return 8;
   if (rem == 7 || rem == 5) return 4;
   return (int)rem;
@@ -343,7 +362,7 @@ void moe_gateup(torch::Tensor x, torch::Tensor w,
   const dim3 grid(cdiv((int)I, kMoeWaves), 1, (unsigned)E);
   int64_t m0 = 0;
   while (m0 < M) {
-    const int mt = moe_mtile(M - m0);
+    const int mt = moe_mtile(M - m0, mx4);
     moe_dispatch_m(mt, [&](auto mc) {
       constexpr int MV = decltype(mc)::value;
       auto launch = [&](auto qmc, auto pkc, auto gluc) {
@@ -413,7 +432,7 @@ void moe_down(torch::Tensor act, torch::Tensor w,
   const dim3 grid(cdiv((int)H, kMoeWaves), 1, (unsigned)E);
   int64_t m0 = 0;
   while (m0 < M) {
-    const int mt = moe_mtile(M - m0);
+    const int mt = moe_mtile(M - m0, mx4);
     moe_dispatch_m(mt, [&](auto mc) {
       constexpr int MV = decltype(mc)::value;
       auto launch = [&](auto qmc, auto pkc) {
