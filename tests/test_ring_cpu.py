@@ -970,3 +970,44 @@ def test_cp_windowed_partials_math():
                                vc[:, :, :q0 + T].contiguous(), q0 + T, q0,
                                D ** -0.5, q0 + T, 0, window=window)
     assert torch.allclose(got.float(), want.float(), atol=2e-2, rtol=2e-2)
+
+
+def _linkmx_rank_main(rank, world, port, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    import torch.distributed as dist
+    from dnet_amd.parallel.profiler import measure_link_matrix
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    m = measure_link_matrix(rank, world, torch.device("cpu"),
+                            size=4096, reps=3)
+    q.put((rank, m))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(360)
+@retry_flaky()
+def test_link_matrix_all_pairs():
+    """All-pairs fabric probe: every ordered pair measured, full matrix
+    identical on every rank (this is what /health surfaces for xGMI-aware
+    ring ordering)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    world = 3
+    procs = [ctx.Process(target=_linkmx_rank_main, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(world):
+        r, m = q.get(timeout=150)
+        res[r] = m
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    keys = {f"{i}-{j}" for i in range(world) for j in range(world) if i != j}
+    assert set(res[0]) == keys
+    assert set(res[1]) == keys == set(res[2])
+    for k in keys:
+        assert res[0][k]["latency_ms"] > 0
+        assert res[0][k] == res[1][k] == res[2][k]
