@@ -335,6 +335,7 @@ class ShardRuntime:
                     self._slot_step_exec()
                 elif cmd[0] == CMD_SLOT_CANCEL:
                     self.slots[int(cmd[1])] = None
+                    self._row_sampler.clear_row(int(cmd[1]))
                 elif cmd[0] == CMD_UNLOAD:
                     self._unload()
 
@@ -481,6 +482,7 @@ class ShardRuntime:
                     if ex.world > 1:
                         self._broadcast_cmd(CMD_SLOT_CANCEL, i)
                     self.slots[i] = None
+                    self._row_sampler.clear_row(i)
                     log.info("[PROFILE][SLOT] cancel slot=%d nonce=%s", i,
                              st.get("nonce", "")[:18])
             now = time.monotonic()
@@ -665,6 +667,7 @@ class ShardRuntime:
                              tops=tops0)
         if done:
             self.slots[si] = None
+            self._row_sampler.clear_row(si)
 
     def _slot_step_exec(self) -> None:
         """Collective decode step for all slots (multi-rank path, emits
@@ -741,6 +744,7 @@ class ShardRuntime:
                              tops=tops)
             if done:
                 self.slots[i] = None
+                self._row_sampler.clear_row(i)
 
     _emit_s = 0.0
 
