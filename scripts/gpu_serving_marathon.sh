@@ -45,6 +45,12 @@ cycle qwen-2.5-32b-int8-synthetic ',"quant":"int8-g128"'
 cycle gpt-oss-20b-synthetic
 cycle deepseek-v2-lite-synthetic
 cycle mixtral-8x7b-synthetic ',"quant":"int8-g128"'
+echo "=== concurrent streams (qwen-32b int8, slots) ==="
+curl -s -m 600 -X POST http://127.0.0.1:18080/v1/load_model -H 'content-type: application/json' \
+  -d '{"model":"qwen-2.5-32b-int8-synthetic","max_batch":8}' > /dev/null
+python scripts/serving_concurrent_bench.py --streams 1 --max-tokens 48
+python scripts/serving_concurrent_bench.py --streams 4 --max-tokens 48
+python scripts/serving_concurrent_bench.py --streams 8 --max-tokens 48
 echo "=== marathon done ==="
 grep -c "PROFILE..DECODE" gpurun_out/mar_shard.log | xargs echo "decode runs:"
 grep -ciE "traceback|error" gpurun_out/mar_shard.log | xargs echo "shard errors:"
