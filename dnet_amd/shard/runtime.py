@@ -290,7 +290,18 @@ class ShardRuntime:
         """Main driver loop — owns the model and the process group."""
         while not self._stop.is_set():
             try:
-                kind, arg, done, box = self.ctrl_q.get(timeout=0.02)
+                # BUSY (active slots / queued work): poll the control
+                # queue without blocking — the 20 ms timeout here ran
+                # once per slot TICK and capped serving at ~26 tok/s
+                # (each token paid the control poll)
+                busy = (self.slots is not None
+                        and (any(st is not None for st in self.slots)
+                             or not self.infer_q.empty()
+                             or self._pending is not None))
+                if busy:
+                    kind, arg, done, box = self.ctrl_q.get_nowait()
+                else:
+                    kind, arg, done, box = self.ctrl_q.get(timeout=0.02)
                 try:
                     if kind == "load":
                         self._load(arg)
