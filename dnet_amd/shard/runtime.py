@@ -484,7 +484,9 @@ class ShardRuntime:
         # would re-populate a slot the pending active list still names —
         # its stale token would then be delivered under the new nonce.
         if self._pending is not None:
+            t0e = time.perf_counter()
             self._slot_emit(*self._pending)
+            self._tick_emit_s += time.perf_counter() - t0e
             self._pending = None
         if self._cancelled:
             for i, st in enumerate(self.slots):
@@ -537,7 +539,17 @@ class ShardRuntime:
                 # only); its tokens are emitted at the TOP of the next tick
                 # (before admission) so the host sync overlaps this step's
                 # GPU work without racing slot reuse
+                t0 = time.perf_counter()
                 self._pending = self._slot_step_launch()
+                self._tick_launch_s += time.perf_counter() - t0
+                self._tick_n += 1
+                if self._tick_n >= 128:
+                    log.info("[PROFILE][TICK] n=%d launch_ms=%.2f "
+                             "emit_ms=%.2f", self._tick_n,
+                             self._tick_launch_s / self._tick_n * 1e3,
+                             self._tick_emit_s / self._tick_n * 1e3)
+                    self._tick_n = 0
+                    self._tick_launch_s = self._tick_emit_s = 0.0
         if not progressed and ex.world > 1:
             self._broadcast_cmd(CMD_NOOP)
 
@@ -705,6 +717,9 @@ class ShardRuntime:
                         rs.last_logp, rs.last_topv, rs.last_topi)
 
     _pending = None   # (device tokens, [(slot, gen)]) of the in-flight step
+    _tick_n = 0
+    _tick_launch_s = 0.0
+    _tick_emit_s = 0.0
     _slot_gen: list = []   # per-slot admit generation (see _slot_emit)
 
     def _slot_step_launch(self):
