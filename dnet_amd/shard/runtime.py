@@ -479,15 +479,6 @@ class ShardRuntime:
         (requests serialize there); this keeps N single-stream requests at
         ~single-stream latency each."""
         ex = self.executor
-        # Emit the pipelined step launched LAST tick before admitting:
-        # emission may free slots (stop/max_tokens), and admitting first
-        # would re-populate a slot the pending active list still names —
-        # its stale token would then be delivered under the new nonce.
-        if self._pending is not None:
-            t0e = time.perf_counter()
-            self._slot_emit(*self._pending)
-            self._tick_emit_s += time.perf_counter() - t0e
-            self._pending = None
         if self._cancelled:
             for i, st in enumerate(self.slots):
                 if st is not None and st.get("nonce") in self._cancelled:
@@ -535,13 +526,20 @@ class ShardRuntime:
                 self._broadcast_cmd(CMD_SLOT_STEP)
                 self._slot_step_exec()
             else:
-                # single rank: pipeline — launch step n+1 (device-side deps
-                # only); its tokens are emitted at the TOP of the next tick
-                # (before admission) so the host sync overlaps this step's
-                # GPU work without racing slot reuse
+                # single rank: pipeline — launch step n+1 (device-side
+                # deps only) FIRST, then emit step n's tokens: the host
+                # sync inside emit overlaps the just-launched GPU work.
+                # Slot-reuse safety comes from the per-slot generation
+                # guard in _slot_emit (admission may have re-populated a
+                # slot the pending active list still names).
                 t0 = time.perf_counter()
-                self._pending = self._slot_step_launch()
+                launched = self._slot_step_launch()
                 self._tick_launch_s += time.perf_counter() - t0
+                if self._pending is not None:
+                    t0e = time.perf_counter()
+                    self._slot_emit(*self._pending)
+                    self._tick_emit_s += time.perf_counter() - t0e
+                self._pending = launched
                 self._tick_n += 1
                 if self._tick_n >= 128:
                     log.info("[PROFILE][TICK] n=%d launch_ms=%.2f "
@@ -550,6 +548,12 @@ class ShardRuntime:
                              self._tick_emit_s / self._tick_n * 1e3)
                     self._tick_n = 0
                     self._tick_launch_s = self._tick_emit_s = 0.0
+        if (self._pending is not None
+                and not any(st is not None and st.get("state") != "prefill"
+                            for st in self.slots)):
+            # nothing left to launch: flush the final pipelined step
+            self._slot_emit(*self._pending)
+            self._pending = None
         if not progressed and ex.world > 1:
             self._broadcast_cmd(CMD_NOOP)
 
