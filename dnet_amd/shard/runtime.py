@@ -194,6 +194,13 @@ class ShardRuntime:
         if self.slots is not None:
             from ..core.sampler import RowSampler
             self._row_sampler = RowSampler(req.max_batch, device=ex.device)
+            # device-resident park mask (True = keep the dummy KV append
+            # at smax-1). Updated only on slot-state CHANGES: building a
+            # park index tensor per tick was a synchronous H2D that
+            # serialized the host against the in-flight decode graph
+            # (~17 ms/tick measured)
+            self._park_dev = torch.ones(req.max_batch, dtype=torch.bool,
+                                        device=ex.device)
             # per-slot generation counter: bumped on every admit so a
             # pending (pipelined) emit can detect that its slot was freed
             # and re-admitted and must not deliver the stale token under
@@ -347,6 +354,7 @@ class ShardRuntime:
                 elif cmd[0] == CMD_SLOT_CANCEL:
                     self.slots[int(cmd[1])] = None
                     self._row_sampler.clear_row(int(cmd[1]))
+                    self._park_dev[int(cmd[1])] = True
                 elif cmd[0] == CMD_UNLOAD:
                     self._unload()
 
@@ -487,6 +495,7 @@ class ShardRuntime:
                         self._broadcast_cmd(CMD_SLOT_CANCEL, i)
                     self.slots[i] = None
                     self._row_sampler.clear_row(i)
+                    self._park_dev[i] = True
                     log.info("[PROFILE][SLOT] cancel slot=%d nonce=%s", i,
                              st.get("nonce", "")[:18])
             now = time.monotonic()
@@ -624,7 +633,9 @@ class ShardRuntime:
         chunk = int(os.environ.get("DNET_PREFILL_CHUNK", "2048"))
         if ex.world == 1 and int(tokens.shape[-1]) > chunk:
             # long prompt: admit in "prefill" state — the tick loop feeds
-            # one chunk per iteration between decode steps
+            # one chunk per iteration between decode steps (slot stays
+            # PARKED until the prefill completes)
+            self._park_dev[si] = True
             self.slots[si] = {"state": "prefill", "tokens": tokens,
                               "p0": 0, "cfg": cfg, "nonce": nonce,
                               "produced": 0, "max_tokens": max_tokens,
@@ -655,6 +666,7 @@ class ShardRuntime:
                              tops=tops0)
         if not done:
             self.slots[si] = st
+            self._park_dev[si] = False
         log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
                  nonce[:18], int(tokens.shape[-1]))
 
@@ -687,6 +699,7 @@ class ShardRuntime:
         ex.tokbuf[0][si] = t0
         st["produced"] = 1
         st["state"] = "active"
+        self._park_dev[si] = False
         st.pop("tokens")
         done = t0 in st["stop_ids"] or st["max_tokens"] <= 1
         if ex.is_last:
@@ -695,6 +708,7 @@ class ShardRuntime:
         if done:
             self.slots[si] = None
             self._row_sampler.clear_row(si)
+            self._park_dev[si] = True
 
     def _slot_step_exec(self) -> None:
         """Collective decode step for all slots (multi-rank path, emits
@@ -710,10 +724,7 @@ class ShardRuntime:
             import torch.distributed as dist
             dist.broadcast(ex.tokbuf[0], src=(ex.stages - 1) * ex.tp)
         ex.kvs[0].pos.add_(1)
-        park = [i for i, st in enumerate(self.slots) if st is None]
-        if park:
-            ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
-                                       device=ex.device)] = ex.smax - 1
+        ex.kvs[0].pos.masked_fill_(self._park_dev, ex.smax - 1)
         self._slot_emit(ex.tokbuf[0],
                         [(i, self._slot_gen[i])
                          for i, st in enumerate(self.slots)
@@ -775,6 +786,7 @@ class ShardRuntime:
             if done:
                 self.slots[i] = None
                 self._row_sampler.clear_row(i)
+                self._park_dev[i] = True
 
     _emit_s = 0.0
 
