@@ -240,14 +240,25 @@ class RingExecutor:
         # warmup advanced nothing persistent except KV garbage at pos; KV pos
         # unchanged (we don't advance pos in the body), cache rows at pos get
         # rewritten by real steps.
+        # TWO alternating execs per step: relaunching the SAME graphExec
+        # while its previous launch is still running blocks the host in
+        # hipGraphLaunch — the slot scheduler's pipelined tick (launch
+        # n+1, then emit n) measured the whole step time inside the
+        # "launch" otherwise. Same static buffers; stream order keeps
+        # them correct.
         self._graphs = {}
+        self._graph_flip = {}
         for mb in range(self.mb_count):
             for r in range(self.rounds):
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g, pool=pool):
-                    self._decode_body(mb, r)
-                self._graphs[(mb, r)] = g
-        log.info("captured %d decode graphs", len(self._graphs))
+                pair = []
+                for _ in range(2):
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g, pool=pool):
+                        self._decode_body(mb, r)
+                    pair.append(g)
+                self._graphs[(mb, r)] = pair
+                self._graph_flip[(mb, r)] = 0
+        log.info("captured %d decode graphs (x2 execs)", len(self._graphs))
 
     def _run_decode(self, mb: int, r: int = 0):
         if self.use_graphs and not self._graphs:
@@ -258,7 +269,9 @@ class RingExecutor:
                 self.use_graphs = False
                 self._graphs = {}
         if self.use_graphs:
-            self._graphs[(mb, r)].replay()
+            f = self._graph_flip[(mb, r)]
+            self._graphs[(mb, r)][f].replay()
+            self._graph_flip[(mb, r)] = f ^ 1
         else:
             self._decode_body(mb, r)
 
