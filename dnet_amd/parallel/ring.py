@@ -465,9 +465,17 @@ class RingExecutor:
             return self.model.normalize_project(h[:, -1].contiguous())
         return None
 
+    t_replay = 0.0   # cumulative; slot-tick profiling
+
     def slot_step_compute(self, mb: int = 0) -> None:
         """One decode step over the whole slot batch, including the ring
         hops — no sampling (the caller samples on the last stage)."""
+        import time as _t
+        _t0 = _t.perf_counter()
+        self._slot_step_compute_inner(mb)
+        self.t_replay += _t.perf_counter() - _t0
+
+    def _slot_step_compute_inner(self, mb: int = 0) -> None:
         last_r = self.rounds - 1
         if self.is_first:
             for r in range(self.rounds):

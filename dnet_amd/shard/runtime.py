@@ -552,11 +552,13 @@ class ShardRuntime:
                 self._tick_n += 1
                 if self._tick_n >= 128:
                     log.info("[PROFILE][TICK] n=%d launch_ms=%.2f "
-                             "emit_ms=%.2f", self._tick_n,
+                             "emit_ms=%.2f replay_ms=%.2f", self._tick_n,
                              self._tick_launch_s / self._tick_n * 1e3,
-                             self._tick_emit_s / self._tick_n * 1e3)
+                             self._tick_emit_s / self._tick_n * 1e3,
+                             ex.t_replay / self._tick_n * 1e3)
                     self._tick_n = 0
                     self._tick_launch_s = self._tick_emit_s = 0.0
+                    ex.t_replay = 0.0
         if (self._pending is not None
                 and not any(st is not None and st.get("state") != "prefill"
                             for st in self.slots)):
