@@ -556,9 +556,13 @@ class ShardRuntime:
                              self._tick_launch_s / self._tick_n * 1e3,
                              self._tick_emit_s / self._tick_n * 1e3,
                              ex.t_replay / self._tick_n * 1e3)
+                    log.info("[PROFILE][TICK2] sample_ms=%.2f aux_ms=%.2f",
+                             self._t_sample / self._tick_n * 1e3,
+                             self._t_aux / self._tick_n * 1e3)
                     self._tick_n = 0
                     self._tick_launch_s = self._tick_emit_s = 0.0
                     ex.t_replay = 0.0
+                    self._t_sample = self._t_aux = 0.0
         if (self._pending is not None
                 and not any(st is not None and st.get("state") != "prefill"
                             for st in self.slots)):
@@ -739,13 +743,19 @@ class ShardRuntime:
     _tick_emit_s = 0.0
     _slot_gen: list = []   # per-slot admit generation (see _slot_emit)
 
+    _t_sample = 0.0
+    _t_aux = 0.0
+
     def _slot_step_launch(self):
         """Enqueue one decode step for the whole batch — device ops only,
         no host sync. Returns (tokens, active slots, logprob tensors)."""
         ex = self.executor
         rs = self._row_sampler
         ex.slot_step_compute()
+        t0 = time.perf_counter()
         toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
+        self._t_sample += time.perf_counter() - t0
+        t0 = time.perf_counter()
         ex.tokbuf[0].copy_(toks_t)
         ex.kvs[0].pos.add_(1)
         # park idle/prefilling slots at the last row: the dummy append
