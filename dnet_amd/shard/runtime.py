@@ -760,12 +760,12 @@ class ShardRuntime:
         ex.kvs[0].pos.add_(1)
         # park idle/prefilling slots at the last row: the dummy append
         # stays in range and that row is rewritten by a real token before
-        # any active slot ever attends to it
-        park = [i for i, st in enumerate(self.slots)
-                if st is None or st.get("state") == "prefill"]
-        if park:
-            ex.kvs[0].pos[torch.tensor(park, dtype=torch.long,
-                                       device=ex.device)] = ex.smax - 1
+        # any active slot ever attends to it. Device-resident mask:
+        # building a park index tensor here was a synchronous H2D that
+        # serialized the host against the in-flight decode graph
+        # (~17 ms/tick measured).
+        ex.kvs[0].pos.masked_fill_(self._park_dev, ex.smax - 1)
+        self._t_aux += time.perf_counter() - t0
         return (toks_t, [(i, self._slot_gen[i])
                          for i, st in enumerate(self.slots)
                          if st is not None and st.get("state") != "prefill"],
