@@ -194,12 +194,19 @@ class ShardRuntime:
         if self.slots is not None:
             from ..core.sampler import RowSampler
             self._row_sampler = RowSampler(req.max_batch, device=ex.device)
-            # device-resident park mask (True = keep the dummy KV append
-            # at smax-1). Updated only on slot-state CHANGES: building a
-            # park index tensor per tick was a synchronous H2D that
-            # serialized the host against the in-flight decode graph
-            # (~17 ms/tick measured)
+            # device-resident park masks (updated only on slot-state
+            # CHANGES: building a park index tensor per tick was a
+            # synchronous H2D that serialized the host against the
+            # in-flight decode graph, ~17 ms/tick measured).
+            #   _park_dev: row's dummy append pinned in place (freed or
+            #              mid-chunked-prefill slots)
+            #   _free_dev: additionally FREE -> pinned at pos 0, so the
+            #              dead row's attention scans 1 position, not
+            #              smax (prefilling slots must stay at smax-1:
+            #              their low rows hold real prefill KV)
             self._park_dev = torch.ones(req.max_batch, dtype=torch.bool,
+                                        device=ex.device)
+            self._free_dev = torch.ones(req.max_batch, dtype=torch.bool,
                                         device=ex.device)
             # per-slot generation counter: bumped on every admit so a
             # pending (pipelined) emit can detect that its slot was freed
@@ -355,6 +362,7 @@ class ShardRuntime:
                     self.slots[int(cmd[1])] = None
                     self._row_sampler.clear_row(int(cmd[1]))
                     self._park_dev[int(cmd[1])] = True
+                    self._free_dev[int(cmd[1])] = True
                 elif cmd[0] == CMD_UNLOAD:
                     self._unload()
 
@@ -496,6 +504,7 @@ class ShardRuntime:
                     self.slots[i] = None
                     self._row_sampler.clear_row(i)
                     self._park_dev[i] = True
+                    self._free_dev[i] = True
                     log.info("[PROFILE][SLOT] cancel slot=%d nonce=%s", i,
                              st.get("nonce", "")[:18])
             now = time.monotonic()
@@ -548,14 +557,21 @@ class ShardRuntime:
                     t0e = time.perf_counter()
                     self._slot_emit(*self._pending)
                     self._tick_emit_s += time.perf_counter() - t0e
+                    if self._pending_evs is not None:
+                        # emit synced the pending step -> both events done
+                        self._tick_gpu_ms += self._pending_evs[0].\
+                            elapsed_time(self._pending_evs[1])
                 self._pending = launched
+                self._pending_evs = self._launch_evs
                 self._tick_n += 1
                 if self._tick_n >= 128:
                     log.info("[PROFILE][TICK] n=%d launch_ms=%.2f "
-                             "emit_ms=%.2f replay_ms=%.2f", self._tick_n,
+                             "emit_ms=%.2f replay_ms=%.2f gpu_ms=%.2f",
+                             self._tick_n,
                              self._tick_launch_s / self._tick_n * 1e3,
                              self._tick_emit_s / self._tick_n * 1e3,
-                             ex.t_replay / self._tick_n * 1e3)
+                             ex.t_replay / self._tick_n * 1e3,
+                             self._tick_gpu_ms / self._tick_n)
                     log.info("[PROFILE][TICK2] sample_ms=%.2f aux_ms=%.2f",
                              self._t_sample / self._tick_n * 1e3,
                              self._t_aux / self._tick_n * 1e3)
@@ -563,6 +579,7 @@ class ShardRuntime:
                     self._tick_launch_s = self._tick_emit_s = 0.0
                     ex.t_replay = 0.0
                     self._t_sample = self._t_aux = 0.0
+                    self._tick_gpu_ms = 0.0
         if (self._pending is not None
                 and not any(st is not None and st.get("state") != "prefill"
                             for st in self.slots)):
@@ -642,6 +659,12 @@ class ShardRuntime:
             # one chunk per iteration between decode steps (slot stays
             # PARKED until the prefill completes)
             self._park_dev[si] = True
+            self._free_dev[si] = False
+            # the row was free-parked at pos 0; move it to smax-1 NOW —
+            # the next decode step appends at the CURRENT pos before the
+            # masks are re-applied, and an append at 0 would overwrite
+            # the first prefill chunk
+            ex.kvs[0].pos[si] = ex.smax - 1
             self.slots[si] = {"state": "prefill", "tokens": tokens,
                               "p0": 0, "cfg": cfg, "nonce": nonce,
                               "produced": 0, "max_tokens": max_tokens,
@@ -673,6 +696,7 @@ class ShardRuntime:
         if not done:
             self.slots[si] = st
             self._park_dev[si] = False
+            self._free_dev[si] = False
         log.info("[PROFILE][SLOT] admit slot=%d nonce=%s prompt=%d", si,
                  nonce[:18], int(tokens.shape[-1]))
 
@@ -706,6 +730,7 @@ class ShardRuntime:
         st["produced"] = 1
         st["state"] = "active"
         self._park_dev[si] = False
+        self._free_dev[si] = False
         st.pop("tokens")
         done = t0 in st["stop_ids"] or st["max_tokens"] <= 1
         if ex.is_last:
@@ -715,6 +740,7 @@ class ShardRuntime:
             self.slots[si] = None
             self._row_sampler.clear_row(si)
             self._park_dev[si] = True
+            self._free_dev[si] = True
 
     def _slot_step_exec(self) -> None:
         """Collective decode step for all slots (multi-rank path, emits
@@ -731,6 +757,7 @@ class ShardRuntime:
             dist.broadcast(ex.tokbuf[0], src=(ex.stages - 1) * ex.tp)
         ex.kvs[0].pos.add_(1)
         ex.kvs[0].pos.masked_fill_(self._park_dev, ex.smax - 1)
+        ex.kvs[0].pos.masked_fill_(self._free_dev, 0)
         self._slot_emit(ex.tokbuf[0],
                         [(i, self._slot_gen[i])
                          for i, st in enumerate(self.slots)
@@ -738,9 +765,12 @@ class ShardRuntime:
                         rs.last_logp, rs.last_topv, rs.last_topi)
 
     _pending = None   # (device tokens, [(slot, gen)]) of the in-flight step
+    _pending_evs = None   # CUDA events bracketing the pending step
+    _launch_evs = None
     _tick_n = 0
     _tick_launch_s = 0.0
     _tick_emit_s = 0.0
+    _tick_gpu_ms = 0.0
     _slot_gen: list = []   # per-slot admit generation (see _slot_emit)
 
     _t_sample = 0.0
@@ -751,6 +781,10 @@ class ShardRuntime:
         no host sync. Returns (tokens, active slots, logprob tensors)."""
         ex = self.executor
         rs = self._row_sampler
+        if ex.device.type == "cuda":
+            self._launch_evs = (torch.cuda.Event(enable_timing=True),
+                                torch.cuda.Event(enable_timing=True))
+            self._launch_evs[0].record()
         ex.slot_step_compute()
         t0 = time.perf_counter()
         toks_t = self._row_sampler.sample(ex.logits_buf[0].float())
@@ -765,6 +799,9 @@ class ShardRuntime:
         # serialized the host against the in-flight decode graph
         # (~17 ms/tick measured).
         ex.kvs[0].pos.masked_fill_(self._park_dev, ex.smax - 1)
+        ex.kvs[0].pos.masked_fill_(self._free_dev, 0)
+        if self._launch_evs is not None:
+            self._launch_evs[1].record()
         self._t_aux += time.perf_counter() - t0
         return (toks_t, [(i, self._slot_gen[i])
                          for i, st in enumerate(self.slots)
@@ -799,6 +836,7 @@ class ShardRuntime:
                 self.slots[i] = None
                 self._row_sampler.clear_row(i)
                 self._park_dev[i] = True
+                self._free_dev[i] = True
 
     _emit_s = 0.0
 
