@@ -65,7 +65,11 @@ def _get_scratch(device) -> torch.Tensor:
     if key not in _scratch:
         # zero-initialized: the split-k path relies on it (the combine
         # kernel re-zeroes after each use instead of a per-launch memset)
-        _scratch[key] = torch.zeros(64 * 8192, dtype=torch.float32,
+        # sized for the largest M=64 pass (lm_head-class N): too small a
+        # scratch silently disables split-k (launch_m16 falls back to
+        # sk=1) — measured on gateup N=55296: 864 blocks quantize badly
+        # onto 256 CUs at 3 blocks/CU
+        _scratch[key] = torch.zeros(64 * 152064, dtype=torch.float32,
                                     device=device)
     return _scratch[key]
 
