@@ -665,6 +665,20 @@ static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
          (K / 64) / (sk * 2) >= min_pairs &&
          ((K / 64) / (sk * 2)) * 64 % g_align == 0)  // split starts G-aligned
     sk *= 2;
+  // Measured (profiles/r02, sk sweep): a split span that is not a
+  // multiple of 256 pushes its remainder through the serial tail path;
+  // on SHORT spans that tail dominates (qkv K=5120 sk=8: span 640,
+  // 20% tail, 27.7us vs 23.6us at sk=4). Halve sk when the tail is
+  // >=15% of the span, the halved span is tail-free, and the halved
+  // grid still covers the chip (o N=5120 shows 320 blocks is too few).
+  if (sk > 1) {
+    const int64_t span = ((K / 64) / sk) * 64;
+    const int64_t span2 = ((K / 64) / (sk / 2)) * 64;
+    if (span % 256 != 0 && (double)(span % 256) / span >= 0.15 &&
+        span2 % 256 == 0 && span2 % g_align == 0 &&
+        blocks * (sk / 2) >= 440)
+      sk /= 2;
+  }
   if (sk > 1 && M * N > scratch_elems) sk = 1;
   return sk;
 }
