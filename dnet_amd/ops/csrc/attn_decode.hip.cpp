@@ -36,6 +36,9 @@ __device__ __forceinline__ float wave_allreduce_max(float v) {
 
 constexpr int kChunk = 64;
 
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
 // splits > 1: flash-decode S-partitioning — gridDim.y splits each take
 // chunks c = c0 + split, c0 + split + splits, ...; unnormalized partials
 // (acc, m, l) go to scratch [B, Hq, splits, D+2] f32 and a combine kernel
@@ -225,6 +228,272 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// MFMA decode attention: same contract as attn_decode_kernel, compute
+// reshaped for the matrix cores. The G (<=16) query heads of a
+// (batch, kv-head) group form the 16-row A operand; each of the 4 waves
+// owns a 16-position subtile of every 64-position chunk and keeps an
+// INDEPENDENT online-softmax state (m, l, acc) — position space is
+// fully partitioned across waves, so there is no per-chunk cross-wave
+// synchronization at all, just one merge at the end (intra-block
+// flash-decode split). Scores = Q[16,D] x K^T via v_mfma_f32_16x16x32,
+// P@V via the same MFMA with the per-wave P slab zero-padded to k=32.
+// Replaces a serial 64-iteration __shfl broadcast loop per (head,
+// chunk) in the scalar kernel (~26us -> ~14us per layer at batch 64,
+// and the win grows with context).
+template <int D, int DV, bool Q8>
+__global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
+    const short* __restrict__ q, const void* __restrict__ kc,
+    const void* __restrict__ vc, const short* __restrict__ kscale,
+    const short* __restrict__ vscale, const int* __restrict__ pos,
+    short* __restrict__ out, const int Hq, const int Hkv, const int Smax,
+    const float scale, const int ldq, const int window,
+    const short* __restrict__ sinks, float* __restrict__ partials,
+    const int splits) {
+  constexpr int DC = D / 32;   // QK k-chunks
+  constexpr int NV = DV / 16;  // PV n-tiles
+  constexpr int KP = D + 8;    // k_lds row pitch (16-lane frags bank-clean)
+  constexpr int VP = 88;       // vt_lds pitch: covers the k<=79 overread of
+                               // wave 3 (zero-P columns) and 44-dword row
+                               // stride hits all banks across 16 rows
+  constexpr int SP = 40;       // per-wave P slab pitch
+  constexpr int STAGE = 64 * KP * 2 + DV * VP * 2;
+  static_assert(STAGE >= 4 * 16 * DV * 4 + 4 * 16 * 2 * 4,
+                "merge area must fit in the staging LDS");
+  __shared__ __attribute__((aligned(16))) char smem[STAGE];
+  __shared__ short p_lds[4][16 * SP];
+  short* k_lds = reinterpret_cast<short*>(smem);
+  short* vt_lds = reinterpret_cast<short*>(smem + 64 * KP * 2);
+
+  const int b = blockIdx.x / Hkv;
+  const int hkv = blockIdx.x % Hkv;
+  const int G = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int row = lane & 15;
+  const int hi = lane >> 4;
+  const int len = pos[b];
+  const int start = (window > 0) ? max(0, len - window) : 0;
+  const int64_t kvbase = ((int64_t)b * Hkv + hkv) * Smax * D;
+  const int64_t vbase = ((int64_t)b * Hkv + hkv) * (int64_t)Smax * DV;
+
+  // Q fragments (A layout: lane holds Q[head=row][c*32 + hi*8 .. +8]);
+  // rows >= G duplicate head G-1 — their outputs are never stored and
+  // softmax is per-row, so they cannot contaminate real heads
+  const short* qp = q + (int64_t)b * ldq + (int64_t)(hkv * G + min(row, G - 1)) * D;
+  bf16x8 qf[DC];
+#pragma unroll
+  for (int c = 0; c < DC; ++c)
+    qf[c] = *reinterpret_cast<const bf16x8*>(&qp[c * 32 + hi * 8]);
+
+  f32x4 acc[NV];
+#pragma unroll
+  for (int n = 0; n < NV; ++n) acc[n] = {0.f, 0.f, 0.f, 0.f};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.f;
+  }
+
+  // zero this wave's dead P columns (k = 16..31) once: the PV MFMA runs
+  // k=32 and the upper half multiplies whatever V rows it overreads
+  for (int i = lane; i < 16 * 16; i += 64)
+    p_lds[wave][(i >> 4) * SP + 16 + (i & 15)] = 0;
+  // zero vt's never-staged tail columns (s = 64..VP-1) once: wave 3's
+  // k>=16 B-fragment overreads them, and uninitialized LDS bits can be
+  // NaN — NaN * 0 is NaN, which would poison the accumulators
+  for (int i = threadIdx.x; i < DV * (VP - 64); i += 256)
+    vt_lds[(i / (VP - 64)) * VP + 64 + i % (VP - 64)] = 0;
+
+  const int nchunks = (len + kChunk - 1) / kChunk;
+  const int split = blockIdx.y;
+  for (int c = start / kChunk + split; c < nchunks; c += splits) {
+    const int s0 = c * kChunk;
+    const int valid = min(kChunk, len - s0);
+    __syncthreads();   // previous tile's consumers done
+    for (int i = threadIdx.x; i < kChunk * (D / 8); i += blockDim.x) {
+      const int sr = i / (D / 8);
+      const int col = (i % (D / 8)) * 8;
+      short8 kv8;
+      if (sr < valid) {
+        if (Q8) {
+          constexpr int NG = D / 64;
+          const int64_t rb = kvbase / D * (int64_t)NG
+                             + (int64_t)(s0 + sr) * NG + col / 64;
+          const float ks = bits2f(kscale[rb]);
+          const int2 kq = *reinterpret_cast<const int2*>(
+              (const int8_t*)kc + kvbase + (int64_t)(s0 + sr) * D + col);
+          const int8_t* kb = reinterpret_cast<const int8_t*>(&kq);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) kv8.x[j] = f2bits((float)kb[j] * ks);
+        } else {
+          kv8 = *reinterpret_cast<const short8*>(
+              (const short*)kc + kvbase + (int64_t)(s0 + sr) * D + col);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kv8.x[j] = 0;
+      }
+      *reinterpret_cast<short8*>(&k_lds[sr * KP + col]) = kv8;
+    }
+    // V transposed [dv][s] so the PV B-fragment reads are contiguous;
+    // global side reads full short8 rows (the 8 scattered 2 B LDS
+    // stores are cheap, 4 B global loads were not)
+    for (int i = threadIdx.x; i < kChunk * (DV / 8); i += blockDim.x) {
+      const int sr = i / (DV / 8);
+      const int dc = (i % (DV / 8)) * 8;
+      short8 vv8;
+      if (sr < valid) {
+        if (Q8) {
+          constexpr int NGV = DV / 64;
+          const int64_t rb = vbase / DV * (int64_t)NGV
+                             + (int64_t)(s0 + sr) * NGV + dc / 64;
+          const float vs = bits2f(vscale[rb]);
+          const int2 vq = *reinterpret_cast<const int2*>(
+              (const int8_t*)vc + vbase + (int64_t)(s0 + sr) * DV + dc);
+          const int8_t* vb = reinterpret_cast<const int8_t*>(&vq);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vv8.x[j] = f2bits((float)vb[j] * vs);
+        } else {
+          vv8 = *reinterpret_cast<const short8*>(
+              (const short*)vc + vbase + (int64_t)(s0 + sr) * DV + dc);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vv8.x[j] = 0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_lds[(dc + j) * VP + sr] = vv8.x[j];
+    }
+    __syncthreads();
+
+    // scores for this wave's 16 positions: C[head=hi*4+r][pos col=row]
+    f32x4 sc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int cc = 0; cc < DC; ++cc) {
+      const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+          &k_lds[(wave * 16 + row) * KP + cc * 32 + hi * 8]);
+      sc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[cc], kf, sc, 0, 0, 0);
+    }
+    const int spos = s0 + wave * 16 + row;
+    const bool dead = (spos >= len) || (spos < start);
+    float pv[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      pv[r] = dead ? -INFINITY : sc[r] * scale;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = pv[r];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, 16));
+      const float mn = fmaxf(m_run[r], mx);
+      const float corr = __expf(m_run[r] - mn);
+      pv[r] = __expf(pv[r] - mn);   // -inf underflows to exactly 0
+      float ls = pv[r];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        ls += __shfl_xor(ls, off, 16);
+      l_run[r] = l_run[r] * corr + ls;
+      m_run[r] = mn;
+#pragma unroll
+      for (int n = 0; n < NV; ++n) acc[n][r] *= corr;
+    }
+    // P slab (cols 0..15 = this wave's positions), then PV
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      p_lds[wave][(hi * 4 + r) * SP + row] = f2bits(pv[r]);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local slab
+#pragma unroll
+    for (int n = 0; n < NV; ++n) {
+      const bf16x8 pf =
+          *reinterpret_cast<const bf16x8*>(&p_lds[wave][row * SP + hi * 8]);
+      const bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+          &vt_lds[(n * 16 + row) * VP + wave * 16 + hi * 8]);
+      acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, acc[n], 0, 0,
+                                                       0);
+    }
+  }
+
+  // ---- cross-wave merge (position space was partitioned; combine the
+  // four partial states like a split-S merge, in the freed staging LDS)
+  __syncthreads();
+  float* mrg = reinterpret_cast<float*>(smem);            // [4][16][DV]
+  float* mml = mrg + 4 * 16 * DV;                         // [4][16][2]
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int head = hi * 4 + r;
+#pragma unroll
+    for (int n = 0; n < NV; ++n)
+      mrg[(wave * 16 + head) * DV + n * 16 + row] = acc[n][r];
+    if (row == 0) {
+      mml[(wave * 16 + head) * 2] = m_run[r];
+      mml[(wave * 16 + head) * 2 + 1] = l_run[r];
+    }
+  }
+  __syncthreads();
+  if (wave != 0) return;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int head = hi * 4 + r;
+    float M = -1e30f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w)
+      M = fmaxf(M, mml[(w * 16 + head) * 2]);
+    float L = 0.f;
+#pragma unroll
+    for (int n = 0; n < NV; ++n) acc[n][r] = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      const float wgt = __expf(mml[(w * 16 + head) * 2] - M);
+      L += mml[(w * 16 + head) * 2 + 1] * wgt;
+#pragma unroll
+      for (int n = 0; n < NV; ++n)
+        acc[n][r] += mrg[(w * 16 + head) * DV + n * 16 + row] * wgt;
+    }
+    m_run[r] = M;
+    l_run[r] = L;
+  }
+
+  if (partials != nullptr) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int head = hi * 4 + r;
+      if (head >= G) continue;
+      float* prow = partials
+          + (((int64_t)b * Hq + hkv * G + head) * splits + split) * (DV + 2);
+#pragma unroll
+      for (int n = 0; n < NV; ++n) prow[n * 16 + row] = acc[n][r];
+      if (row == 0) {
+        prow[DV] = m_run[r];
+        prow[DV + 1] = l_run[r];
+      }
+    }
+    return;
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int head = hi * 4 + r;
+    if (head >= G) continue;
+    float inv;
+    if (sinks != nullptr) {
+      const float sk = bits2f(sinks[hkv * G + head]);
+      const float mx = fmaxf(m_run[r], sk);
+      const float num = __expf(m_run[r] - mx);
+      const float denom = l_run[r] * num + __expf(sk - mx);
+      inv = (denom > 0.f) ? num / denom : 0.f;
+    } else {
+      inv = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+    }
+    short* orow = out + ((int64_t)b * Hq + hkv * G + head) * DV;
+#pragma unroll
+    for (int n = 0; n < NV; ++n)
+      orow[n * 16 + row] = f2bits(acc[n][r] * inv);
+  }
+}
+
 // merge the per-split unnormalized partials: one 64-lane block per (b, h).
 template <int D>
 __global__ void attn_combine_kernel(const float* __restrict__ partials,
@@ -319,14 +588,35 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
   const short* vsp = q8 ? (const short*)vscale->data_ptr() : nullptr;
   float* pp = (splits > 1 || !combine) ? (float*)partials->data_ptr()
                                        : nullptr;
-#define LAUNCH_ATTN(DD, DDV, QQ)                                            \
-  hipLaunchKernelGGL((attn_decode_kernel<DD, DDV, QQ>), grid, dim3(256),      \
-                     lds, stream, (const short*)q.data_ptr(),                 \
-                     kcache.data_ptr(),                                       \
-                     vcache.data_ptr(), ksp, vsp, (const int*)pos.data_ptr(), \
-                     (short*)out.data_ptr(), (int)Hq, (int)Hkv, (int)Smax,    \
-                     (float)scale, ldq, (int)window,                          \
-                     (splits > 1 ? nullptr : skp), pp, (int)splits)
+  // Scalar flash-decode is the default: the MFMA variant measured
+  // SLOWER end to end (3635 vs 3683 tok/s short-ctx, 1433 vs 1491 at
+  // 2k ctx) — decode attention has so little arithmetic per KV byte
+  // that the V-transpose staging the PV MFMA needs (8 scattered 2 B
+  // LDS stores per element vs one short8 row store) costs more than
+  // the matrix cores save. Kept for A/B via DNET_ATTN_MFMA=1.
+  static const bool mfma_on = []() {
+    const char* e = getenv("DNET_ATTN_MFMA");
+    return e != nullptr && e[0] == '1';
+  }();
+#define LAUNCH_ATTN(DD, DDV, QQ)                                              \
+  do {                                                                        \
+    if (mfma_on)                                                              \
+      hipLaunchKernelGGL((attn_decode_mfma_kernel<DD, DDV, QQ>), grid,        \
+                         dim3(256), 0, stream, (const short*)q.data_ptr(),    \
+                         kcache.data_ptr(), vcache.data_ptr(), ksp, vsp,      \
+                         (const int*)pos.data_ptr(), (short*)out.data_ptr(),  \
+                         (int)Hq, (int)Hkv, (int)Smax, (float)scale, ldq,     \
+                         (int)window, (splits > 1 ? nullptr : skp), pp,       \
+                         (int)splits);                                        \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_decode_kernel<DD, DDV, QQ>), grid, dim3(256),  \
+                         lds, stream, (const short*)q.data_ptr(),             \
+                         kcache.data_ptr(), vcache.data_ptr(), ksp, vsp,      \
+                         (const int*)pos.data_ptr(), (short*)out.data_ptr(),  \
+                         (int)Hq, (int)Hkv, (int)Smax, (float)scale, ldq,     \
+                         (int)window, (splits > 1 ? nullptr : skp), pp,       \
+                         (int)splits);                                        \
+  } while (0)
   if (D == 192 && q8) LAUNCH_ATTN(192, 128, true);
   else if (D == 192) LAUNCH_ATTN(192, 128, false);
   else if (D == 128 && q8) LAUNCH_ATTN(128, 128, true);
