@@ -20,6 +20,7 @@ import logging
 import os
 import time
 
+import dnet_amd  # noqa: F401  (sets the TunableOp env BEFORE torch loads)
 import torch
 
 logging.basicConfig(level=logging.INFO)
