@@ -744,8 +744,13 @@ class RingModel:
         self._obs_t0 = time.perf_counter()
 
     def _mlp(self, y: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
-        gu = lw.gateup(y)
-        a = ops.swiglu(gu)
+        g = lw.gateup
+        if (y.is_cuda and g.is_quant and not g.mxfp4 and g.packed
+                and g.bias is None and 2 < y.shape[0] <= 64):
+            # decode fast path: split-k combine fused into SwiGLU
+            a = ops.gemv_swiglu(y, g.w, g.scales, g.group, g.bits)
+        else:
+            a = ops.swiglu(g(y))
         return lw.down(a)
 
     def _dequant_prefetch(self, layer_ids: Sequence[int], i: int):

@@ -211,6 +211,29 @@ def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
                     wpos)
 
 
+def gemv_swiglu(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+                group: int, bits: int = 8,
+                packed: bool = True) -> torch.Tensor:
+    """Fused gate/up GEMM + SwiGLU for the decode path: when the GEMM
+    runs split-k, SwiGLU reads (and re-zeroes) the f32 scratch directly
+    instead of going through the f32->bf16 combine kernel — one fewer
+    dependent ~5 us kernel per layer per step."""
+    if not (x.is_cuda and packed and 2 < x.shape[0] <= 64):
+        gemv = gemv_int8 if bits == 8 else gemv_int4
+        return swiglu(gemv(x, w, scales, group, None, packed))
+    M, N = x.shape[0], w.shape[0]
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    scratch = _get_scratch(x.device)
+    deferred = _native().gemm_m16(x, w, scales, None, out, scratch, group,
+                                  True, bits, True)
+    y = torch.empty(M, N // 2, dtype=x.dtype, device=x.device)
+    if deferred:
+        _native().swiglu_f32(scratch, y, N)
+    else:
+        _native().swiglu(out, y)
+    return y
+
+
 def swiglu(gu: torch.Tensor) -> torch.Tensor:
     if gu.is_cuda:
         i = gu.shape[-1] // 2

@@ -1,5 +1,6 @@
 // Python bindings for the dnet_amd gfx950 kernel extension.
 #include <torch/extension.h>
+namespace py = pybind11;
 
 namespace dnet {
 void rmsnorm(torch::Tensor x, c10::optional<torch::Tensor> residual,
@@ -28,16 +29,17 @@ void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  c10::optional<torch::Tensor> vscale,
                  c10::optional<torch::Tensor> wpos);
 void swiglu(torch::Tensor gu, torch::Tensor y);
+void swiglu_f32(torch::Tensor gu_f32, torch::Tensor y, int64_t N);
 void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group, bool packed);
 void dequant_int4(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group);
 void dequant_mxfp4(torch::Tensor w, torch::Tensor scales, torch::Tensor out);
-void gemm_m16(torch::Tensor x, torch::Tensor w,
+bool gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
               c10::optional<torch::Tensor> scratch, int64_t group, bool packed,
-              int64_t bits);
+              int64_t bits, bool defer_combine);
 void moe_gateup(torch::Tensor x, torch::Tensor w,
                 c10::optional<torch::Tensor> scales,
                 c10::optional<torch::Tensor> bias, torch::Tensor we,
@@ -77,9 +79,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("sin_table"), py::arg("kscale"), py::arg("vscale"),
         py::arg("wpos") = c10::nullopt);
   m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
+  m.def("swiglu_f32", &dnet::swiglu_f32,
+        "SwiGLU reading+re-zeroing the split-k f32 scratch");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");
   m.def("dequant_int4", &dnet::dequant_int4, "packed-int4 -> bf16 dequant");
-  m.def("gemm_m16", &dnet::gemm_m16, "MFMA decode GEMM (M<=16, bf16 or int8)");
+  m.def("gemm_m16", &dnet::gemm_m16, "MFMA decode GEMM (M<=16, bf16 or int8)",
+        py::arg("x"), py::arg("w"), py::arg("scales"), py::arg("bias"),
+        py::arg("out"), py::arg("scratch"), py::arg("group"),
+        py::arg("packed"), py::arg("bits"),
+        py::arg("defer_combine") = false);
   m.def("moe_gateup", &dnet::moe_gateup,
         "grouped MoE gate+up GEMV with fused GLU and expert skip");
   m.def("moe_down", &dnet::moe_down,

@@ -683,11 +683,11 @@ static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
   return sk;
 }
 
-static void launch_m16(torch::Tensor x, torch::Tensor w,
+static bool launch_m16(torch::Tensor x, torch::Tensor w,
                        c10::optional<torch::Tensor> scales,
                        c10::optional<torch::Tensor> bias, torch::Tensor out,
                        c10::optional<torch::Tensor> scratch, int group,
-                       int64_t m0, int M, int bits) {
+                       int64_t m0, int M, int bits, bool defer_combine) {
   const int64_t K = x.size(1), N = w.size(0);
   auto stream = current_stream();
   const int64_t scratch_elems = scratch.has_value() ? scratch->numel() : 0;
@@ -776,18 +776,22 @@ static void launch_m16(torch::Tensor x, torch::Tensor w,
 #undef LAUNCH_STREAM_G8
 #undef LAUNCH_STREAM_G4
   if (sk > 1) {
+    // defer_combine: the caller's next kernel reads (and re-zeroes) the
+    // f32 scratch itself — skip the convert/bias pass
+    if (defer_combine && bptr == nullptr) return true;
     const int64_t total = (int64_t)M * N;
     const int cgrid = (int)std::min<int64_t>((total + 255) / 256, 2048);
     hipLaunchKernelGGL(f32_to_bf16_bias_kernel, dim3(cgrid), dim3(256), 0,
                        stream, fp, bptr, op, total, (int)N);
   }
+  return false;
 }
 
-void gemm_m16(torch::Tensor x, torch::Tensor w,
+bool gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
               c10::optional<torch::Tensor> scratch, int64_t group,
-              bool packed, int64_t bits) {
+              bool packed, int64_t bits, bool defer_combine) {
   const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
   DNET_CHECK(K % 64 == 0, "K % 64 == 0 required for the MFMA path");
   DNET_CHECK(out.size(0) == M && out.size(1) == N, "shape");
@@ -804,13 +808,18 @@ void gemm_m16(torch::Tensor x, torch::Tensor w,
                  "w shape / align (int4)");
     }
   }
+  // defer only meaningful for the single-chunk (M <= 64) decode path:
+  // multiple chunks could pick different sk and leave a mixed state
+  const bool defer = defer_combine && M <= 64 && !bias.has_value();
+  bool deferred = false;
   int64_t m0 = 0;
   while (m0 < M) {
     const int mt = (int)std::min<int64_t>(M - m0, 64);
-    launch_m16(x, w, scales, bias, out, scratch, (int)group, m0, mt,
-               (int)bits);
+    deferred = launch_m16(x, w, scales, bias, out, scratch, (int)group, m0,
+                          mt, (int)bits, defer);
     m0 += mt;
   }
+  return deferred;
 }
 
 }  // namespace dnet

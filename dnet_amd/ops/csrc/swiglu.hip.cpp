@@ -30,6 +30,50 @@ __global__ void swiglu_kernel(const short* __restrict__ gu,
   }
 }
 
+// Split-k-fused variant: reads the UN-COMBINED f32 split-k scratch of
+// the gate/up GEMM directly (bias-free), re-zeroing it for the next
+// split-k use — removes the f32->bf16 combine kernel from the decode
+// dependency chain (each ~5 us launch-bound hop, 64 per step).
+__global__ void swiglu_f32_kernel(float* __restrict__ gu,
+                                  short* __restrict__ y, const int64_t I,
+                                  const int64_t T) {
+  const int64_t nvec = T * I / 4;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       idx < nvec; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = idx / (I / 4);
+    const int64_t col = idx % (I / 4);
+    float4 g = reinterpret_cast<float4*>(gu + row * 2 * I)[col];
+    float4 u = reinterpret_cast<float4*>(gu + row * 2 * I + I)[col];
+    reinterpret_cast<float4*>(gu + row * 2 * I)[col] =
+        make_float4(0.f, 0.f, 0.f, 0.f);
+    reinterpret_cast<float4*>(gu + row * 2 * I + I)[col] =
+        make_float4(0.f, 0.f, 0.f, 0.f);
+    short4v o;
+    const float gf[4] = {g.x, g.y, g.z, g.w};
+    const float uf[4] = {u.x, u.y, u.z, u.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float silu = gf[j] / (1.f + __expf(-gf[j]));
+      o.x[j] = f2bits(silu * uf[j]);
+    }
+    reinterpret_cast<short4v*>(y + row * I)[col] = o;
+  }
+}
+
+void swiglu_f32(torch::Tensor gu_f32, torch::Tensor y, int64_t N) {
+  const int64_t I = y.size(-1);
+  const int64_t T = y.numel() / I;
+  DNET_CHECK(N == 2 * I, "gu width must be 2*I");
+  DNET_CHECK(I % 4 == 0, "I % 4");
+  DNET_CHECK(gu_f32.numel() >= T * N, "scratch too small");
+  DNET_CHECK(gu_f32.is_contiguous() && y.is_contiguous(), "contig");
+  auto stream = current_stream();
+  const int64_t nvec = T * I / 4;
+  const int grid = (int)std::min<int64_t>((nvec + 255) / 256, 2048);
+  hipLaunchKernelGGL(swiglu_f32_kernel, dim3(grid), dim3(256), 0, stream,
+                     (float*)gu_f32.data_ptr(), (short*)y.data_ptr(), I, T);
+}
+
 void swiglu(torch::Tensor gu, torch::Tensor y) {
   const int64_t I = y.size(-1);
   const int64_t T = y.numel() / I;
