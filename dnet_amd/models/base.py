@@ -669,24 +669,37 @@ class RingModel:
             y = ops.rmsnorm(delta if delta is not None else h,
                             h if delta is not None else None,
                             lw.attn_norm, c.rms_eps)
-            qkv = lw.qkv(y)
-            q = qkv[:, :nq * d].view(B, nq, d)
-            k = qkv[:, nq * d:(nq + nkv) * d].view(B, nkv, d)
-            v = qkv[:, (nq + nkv) * d:].view(B, nkv, d)
-            if c.qk_norm:
-                self._qk_norm(q, k, lw)
             li = kv.local[lid]
             window, sinks = self._attn_params(lid, lw)
-            if self.cp_size > 1:
-                # sequence-sharded KV: write locally, attend via gathered
-                # flash-decode partials (numerically = full attention)
-                from ..parallel.context import cp_attn_decode, local_lengths
-                wpos = (kv.pos - self.cp_rank * self.cp_cap).int()
+            ql = lw.qkv
+            wpos = ((kv.pos - self.cp_rank * self.cp_cap).int()
+                    if self.cp_size > 1 else None)
+            if (y.is_cuda and ql.is_quant and not ql.mxfp4 and ql.packed
+                    and not c.qk_norm and 2 < B <= 64):
+                # decode fast path: qkv split-k combine fused into
+                # RoPE+append (bias applied there, k/v go straight to
+                # the cache)
+                q = ops.gemv_qkv_rope(
+                    y, ql.w, ql.scales, ql.group, ql.bits, ql.bias, nq,
+                    nkv, d, kv.k[li], kv.v[li], kv.pos, self.cos, self.sin,
+                    kv.kscale[li] if kv.quantized else None,
+                    kv.vscale[li] if kv.quantized else None, wpos)
+            else:
+                qkv = lw.qkv(y)
+                q = qkv[:, :nq * d].view(B, nq, d)
+                k = qkv[:, nq * d:(nq + nkv) * d].view(B, nkv, d)
+                v = qkv[:, (nq + nkv) * d:].view(B, nkv, d)
+                if c.qk_norm:
+                    self._qk_norm(q, k, lw)
                 ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos,
                                 self.cos, self.sin,
                                 kv.kscale[li] if kv.quantized else None,
                                 kv.vscale[li] if kv.quantized else None,
                                 wpos=wpos)
+            if self.cp_size > 1:
+                # sequence-sharded KV: write locally, attend via gathered
+                # flash-decode partials (numerically = full attention)
+                from ..parallel.context import cp_attn_decode, local_lengths
                 if window and window > 0:
                     from ..parallel.context import cp_attn_decode_windowed
                     attn = cp_attn_decode_windowed(
@@ -703,10 +716,6 @@ class RingModel:
                         kscale=kv.kscale[li] if kv.quantized else None,
                         vscale=kv.vscale[li] if kv.quantized else None)
             else:
-                ops.rope_append(q, k, v, kv.k[li], kv.v[li], kv.pos, self.cos,
-                                self.sin,
-                                kv.kscale[li] if kv.quantized else None,
-                                kv.vscale[li] if kv.quantized else None)
                 attn = ops.attn_decode(q, kv.k[li], kv.v[li], len_t, d ** -0.5,
                                        window, sinks,
                                        kv.kscale[li] if kv.quantized else None,

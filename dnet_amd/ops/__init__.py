@@ -211,6 +211,50 @@ def rope_append(q, k, v, kcache, vcache, pos, cos, sin,
                     wpos)
 
 
+_defer_cache: dict = {}
+
+
+def _will_defer(M: int, N: int, K: int, group: int, bits: int,
+                scratch_elems: int) -> bool:
+    key = (M, N, K, group, bits)
+    v = _defer_cache.get(key)
+    if v is None:
+        v = _native().gemm_m16_will_defer(M, N, K, group, bits, scratch_elems)
+        _defer_cache[key] = v
+    return v
+
+
+def gemv_qkv_rope(y: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+                  group: int, bits: int, bias, nq: int, nkv: int, d: int,
+                  kcache, vcache, pos, cos, sin, kscale=None, vscale=None,
+                  wpos=None):
+    """Fused qkv GEMM + RoPE + KV append for the decode path: when the
+    GEMM runs split-k, RoPE reads (and re-zeroes) the f32 scratch
+    directly, applies the qkv bias itself, writes the rotated q to a
+    dedicated buffer and k/v straight to the cache — the combine kernel
+    and the bf16 k/v round trip disappear from the dependency chain.
+    Returns the rotated q [B, nq, d]."""
+    B = y.shape[0]
+    scratch = _get_scratch(y.device)
+    if _will_defer(B, w.shape[0], y.shape[1], group, bits, scratch.numel()):
+        out = torch.empty(B, w.shape[0], dtype=y.dtype, device=y.device)
+        deferred = _native().gemm_m16(y, w, scales, None, out, scratch,
+                                      group, True, bits, True)
+        assert deferred, "will_defer disagreed with gemm_m16"
+        q = torch.empty(B, nq, d, dtype=y.dtype, device=y.device)
+        _native().rope_append_f32(scratch, bias, q, nkv, kcache, vcache,
+                                  pos, cos, sin, kscale, vscale, wpos)
+        return q
+    gemv = gemv_int8 if bits == 8 else gemv_int4
+    qkv = gemv(y, w, scales, group, bias, True)
+    q = qkv[:, :nq * d].view(B, nq, d)
+    k = qkv[:, nq * d:(nq + nkv) * d].view(B, nkv, d)
+    v = qkv[:, (nq + nkv) * d:].view(B, nkv, d)
+    rope_append(q, k, v, kcache, vcache, pos, cos, sin, kscale, vscale,
+                wpos=wpos)
+    return q
+
+
 def gemv_swiglu(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
                 group: int, bits: int = 8,
                 packed: bool = True) -> torch.Tensor:

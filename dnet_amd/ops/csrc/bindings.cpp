@@ -22,6 +22,13 @@ void attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                   int64_t q_off, int64_t window, double scale);
 void attn_combine(torch::Tensor partials, c10::optional<torch::Tensor> sinks,
                   torch::Tensor out, int64_t splits);
+void rope_append_f32(torch::Tensor src, c10::optional<torch::Tensor> bias,
+                     torch::Tensor qout, int64_t Hkv, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor pos,
+                     torch::Tensor cos_table, torch::Tensor sin_table,
+                     c10::optional<torch::Tensor> kscale,
+                     c10::optional<torch::Tensor> vscale,
+                     c10::optional<torch::Tensor> wpos);
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor kcache, torch::Tensor vcache, torch::Tensor pos,
                  torch::Tensor cos_table, torch::Tensor sin_table,
@@ -35,6 +42,8 @@ void dequant_int8(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
 void dequant_int4(torch::Tensor w, torch::Tensor scales, torch::Tensor out,
                   int64_t group);
 void dequant_mxfp4(torch::Tensor w, torch::Tensor scales, torch::Tensor out);
+bool gemm_m16_will_defer(int64_t M, int64_t N, int64_t K, int64_t group,
+                         int64_t bits, int64_t scratch_elems);
 bool gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,
@@ -79,6 +88,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("sin_table"), py::arg("kscale"), py::arg("vscale"),
         py::arg("wpos") = c10::nullopt);
   m.def("swiglu", &dnet::swiglu, "fused SwiGLU");
+  m.def("gemm_m16_will_defer", &dnet::gemm_m16_will_defer,
+        "would gemm_m16(defer_combine=true) defer for this shape");
+  m.def("rope_append_f32", &dnet::rope_append_f32,
+        "RoPE+append reading+re-zeroing the qkv split-k f32 scratch");
   m.def("swiglu_f32", &dnet::swiglu_f32,
         "SwiGLU reading+re-zeroing the split-k f32 scratch");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");

@@ -787,6 +787,17 @@ static bool launch_m16(torch::Tensor x, torch::Tensor w,
   return false;
 }
 
+// Would gemm_m16(defer_combine=true) actually defer for this shape?
+// (Pure: lets callers decide between the fused-consumer and the
+// bias-in-combine paths BEFORE launching.)
+bool gemm_m16_will_defer(int64_t M, int64_t N, int64_t K, int64_t group,
+                         int64_t bits, int64_t scratch_elems) {
+  if (M > 64 || M <= 2) return false;
+  const int sk = pick_splitk(N, K, M, scratch_elems, bits < 16 ? 4 : 8,
+                             bits < 16 ? std::max<int>((int)group, 64) : 64);
+  return sk > 1;
+}
+
 bool gemm_m16(torch::Tensor x, torch::Tensor w,
               c10::optional<torch::Tensor> scales,
               c10::optional<torch::Tensor> bias, torch::Tensor out,

@@ -15,7 +15,13 @@ namespace dnet {
 // grid: (B, Hq + Hkv); block: D/2 lanes, lane i rotates pair (i, i + D/2).
 // Q8: KV cache stored int8 with one bf16 scale per 64-dim group
 // (reference: utils/model.py make_cache kv quantization, 8-bit group 64).
-template <bool Q8>
+// F32: split-k-fused variant — q/k/v come from the UN-COMBINED f32
+// split-k scratch of the qkv GEMM (layout [B, (Hq+2*Hkv)*D], row stride
+// ldq), bias (bf16, optional) is applied here, the rotated q goes to a
+// dedicated bf16 buffer `qout`, and every scratch element read is
+// re-zeroed for the next split-k use. Removes the f32->bf16 combine
+// kernel from the decode dependency chain.
+template <bool Q8, bool F32 = false>
 __global__ void rope_append_kernel(short* __restrict__ q,
                                    short* __restrict__ k,
                                    const short* __restrict__ v,
@@ -29,7 +35,10 @@ __global__ void rope_append_kernel(short* __restrict__ q,
                                    const float* __restrict__ sint,
                                    const int Hq, const int Hkv, const int Smax,
                                    const int D, const int ldq, const int ldk,
-                                   const int ldv) {
+                                   const int ldv,
+                                   float* __restrict__ src = nullptr,
+                                   const short* __restrict__ biasv = nullptr,
+                                   short* __restrict__ qout = nullptr) {
   const int b = blockIdx.x;
   const int h = blockIdx.y;
   const int i = threadIdx.x;  // 0 .. max(D/2,64)-1; lanes >= D/2 idle
@@ -40,6 +49,20 @@ __global__ void rope_append_kernel(short* __restrict__ q,
   const float s = act ? sint[(int64_t)p * half + i] : 0.f;
   if (h < Hq) {
     if (!act) return;
+    if (F32) {
+      float* row = src + (int64_t)b * ldq + h * D;
+      float x1 = row[i], x2 = row[i + half];
+      if (biasv != nullptr) {
+        x1 += bits2f(biasv[h * D + i]);
+        x2 += bits2f(biasv[h * D + i + half]);
+      }
+      row[i] = 0.f;
+      row[i + half] = 0.f;
+      short* qrow = qout + (int64_t)b * Hq * D + h * D;
+      qrow[i] = f2bits(x1 * c - x2 * s);
+      qrow[i + half] = f2bits(x2 * c + x1 * s);
+      return;
+    }
     short* row = q + (int64_t)b * ldq + h * D;
     const float x1 = bits2f(row[i]), x2 = bits2f(row[i + half]);
     row[i] = f2bits(x1 * c - x2 * s);
@@ -49,7 +72,26 @@ __global__ void rope_append_kernel(short* __restrict__ q,
   const int hk = h - Hq;
   short* krow = k + (int64_t)b * ldk + hk * D;
   float k1 = 0.f, k2 = 0.f, v1 = 0.f, v2 = 0.f;
-  if (act) {
+  if (act && F32) {
+    const int kcol = (Hq + hk) * D, vcol = (Hq + Hkv + hk) * D;
+    float* krowf = src + (int64_t)b * ldq + kcol;
+    float* vrowf = src + (int64_t)b * ldq + vcol;
+    float x1 = krowf[i], x2 = krowf[i + half];
+    v1 = vrowf[i];
+    v2 = vrowf[i + half];
+    if (biasv != nullptr) {
+      x1 += bits2f(biasv[kcol + i]);
+      x2 += bits2f(biasv[kcol + i + half]);
+      v1 += bits2f(biasv[vcol + i]);
+      v2 += bits2f(biasv[vcol + i + half]);
+    }
+    krowf[i] = 0.f;
+    krowf[i + half] = 0.f;
+    vrowf[i] = 0.f;
+    vrowf[i + half] = 0.f;
+    k1 = x1 * c - x2 * s;
+    k2 = x2 * c + x1 * s;
+  } else if (act) {
     const float x1 = bits2f(krow[i]), x2 = bits2f(krow[i + half]);
     k1 = x1 * c - x2 * s;
     k2 = x2 * c + x1 * s;
@@ -115,6 +157,51 @@ __global__ void rope_append_kernel(short* __restrict__ q,
       if (i == 0 && ng == 2) vsrow[1] = f2bits(q8scale(a1));
     }
   }
+}
+
+// Fused qkv-split-k-combine + RoPE + append: src = f32 scratch of the
+// qkv GEMM (un-combined), qout receives the rotated+biased q rows.
+void rope_append_f32(torch::Tensor src, c10::optional<torch::Tensor> bias,
+                     torch::Tensor qout, int64_t Hkv, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor pos,
+                     torch::Tensor cos_table, torch::Tensor sin_table,
+                     c10::optional<torch::Tensor> kscale,
+                     c10::optional<torch::Tensor> vscale,
+                     c10::optional<torch::Tensor> wpos) {
+  const int64_t B = qout.size(0), Hq = qout.size(1), D = qout.size(2);
+  const int64_t Smax = kcache.size(2);
+  const int64_t N = (Hq + 2 * Hkv) * D;
+  DNET_CHECK(D % 2 == 0 && D / 2 <= 1024, "head_dim");
+  DNET_CHECK(qout.is_contiguous() && src.is_contiguous(), "contig");
+  DNET_CHECK(src.numel() >= B * N, "scratch too small");
+  DNET_CHECK(cos_table.dtype() == torch::kFloat32, "cos table f32");
+  auto stream = current_stream();
+  const bool q8 = kcache.dtype() == torch::kInt8;
+  const int* wpp = wpos.has_value() ? (const int*)wpos->data_ptr() : nullptr;
+  const short* bp = bias.has_value() ? (const short*)bias->data_ptr()
+                                     : nullptr;
+  const unsigned thr = q8 ? 64u : (unsigned)(D / 2);
+  if (q8)
+    hipLaunchKernelGGL((rope_append_kernel<true, true>),
+                       dim3((unsigned)B, (unsigned)(Hq + Hkv)), dim3(thr), 0,
+                       stream, nullptr, nullptr, nullptr, kcache.data_ptr(),
+                       vcache.data_ptr(), (short*)kscale->data_ptr(),
+                       (short*)vscale->data_ptr(),
+                       (const int*)pos.data_ptr(), wpp,
+                       (const float*)cos_table.data_ptr(),
+                       (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
+                       (int)Smax, (int)D, (int)N, 0, 0,
+                       (float*)src.data_ptr(), bp, (short*)qout.data_ptr());
+  else
+    hipLaunchKernelGGL((rope_append_kernel<false, true>),
+                       dim3((unsigned)B, (unsigned)(Hq + Hkv)), dim3(thr), 0,
+                       stream, nullptr, nullptr, nullptr, kcache.data_ptr(),
+                       vcache.data_ptr(), nullptr, nullptr,
+                       (const int*)pos.data_ptr(), wpp,
+                       (const float*)cos_table.data_ptr(),
+                       (const float*)sin_table.data_ptr(), (int)Hq, (int)Hkv,
+                       (int)Smax, (int)D, (int)N, 0, 0,
+                       (float*)src.data_ptr(), bp, (short*)qout.data_ptr());
 }
 
 void rope_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
