@@ -664,11 +664,15 @@ class RingModel:
         nq, nkv, d = self.nq_local, self.nkv_local, c.head_dim
         len_t = kv.pos + 1  # attend over lengths including the token being written
         delta = None
+        self._mlp_defer_ok = True   # only this loop consumes DEFERRED
         for lid in layer_ids:
             lw = self._layer(lid)
-            y = ops.rmsnorm(delta if delta is not None else h,
-                            h if delta is not None else None,
-                            lw.attn_norm, c.rms_eps)
+            if delta is ops.DEFERRED:
+                y = ops.rmsnorm_f32_scratch(h, lw.attn_norm, c.rms_eps)
+            else:
+                y = ops.rmsnorm(delta if delta is not None else h,
+                                h if delta is not None else None,
+                                lw.attn_norm, c.rms_eps)
             li = kv.local[lid]
             window, sinks = self._attn_params(lid, lw)
             ql = lw.qkv
@@ -720,16 +724,32 @@ class RingModel:
                                        window, sinks,
                                        kv.kscale[li] if kv.quantized else None,
                                        kv.vscale[li] if kv.quantized else None)
-            o = self._tp_reduce(lw.o(attn.view(B, nq * d)))
-            y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
-            delta = self._tp_reduce(self._mlp(y2, lw))
+            ol = lw.o
+            if (y.is_cuda and ol.is_quant and not ol.mxfp4 and ol.packed
+                    and ol.bias is None and self.tp_size == 1
+                    and 2 < B <= 64):
+                # decode fast path: o split-k combine fused into the
+                # residual RMSNorm
+                y2 = ops.gemv_rmsnorm(attn.view(B, nq * d), ol.w, ol.scales,
+                                      ol.group, ol.bits, h, lw.mlp_norm,
+                                      c.rms_eps)
+            else:
+                o = self._tp_reduce(lw.o(attn.view(B, nq * d)))
+                y2 = ops.rmsnorm(o, h, lw.mlp_norm, c.rms_eps)
+            delta = self._mlp(y2, lw)
+            if delta is not ops.DEFERRED:
+                delta = self._tp_reduce(delta)
             if self._obs_sync:
                 self._profile_layer_sync(lid)
             elif self._obs_every > 0:
                 self._obs_count += 1
                 if self._obs_count % self._obs_every == 0:
                     self._profile_layer_sync(lid)
-        h.add_(delta)
+        self._mlp_defer_ok = False
+        if delta is ops.DEFERRED:
+            ops.resid_add_scratch(h)
+        else:
+            h.add_(delta)
         return h
 
     def _cp_gather(self, t: torch.Tensor) -> torch.Tensor:
@@ -760,7 +780,18 @@ class RingModel:
             a = ops.gemv_swiglu(y, g.w, g.scales, g.group, g.bits)
         else:
             a = ops.swiglu(g(y))
-        return lw.down(a)
+        dl = lw.down
+        if (getattr(self, "_mlp_defer_ok", False) and a.is_cuda
+                and dl.is_quant and not dl.mxfp4 and dl.packed
+                and dl.bias is None and self.tp_size == 1
+                and 2 < a.shape[0] <= 64
+                and ops.linear_will_defer(a, dl.w, dl.group, dl.bits)):
+            # leave the down projection un-combined in the f32 scratch;
+            # the decode loop's next RMSNorm (or the final residual add)
+            # consumes + re-zeroes it
+            ops.gemv_defer(a, dl.w, dl.scales, dl.group, dl.bits)
+            return ops.DEFERRED
+        return dl(a)
 
     def _dequant_prefetch(self, layer_ids: Sequence[int], i: int):
         """Side-stream dequant of layer i+1's dense weights while layer i

@@ -255,6 +255,66 @@ def gemv_qkv_rope(y: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
     return q
 
 
+class _Deferred:
+    """Sentinel: the projection's result lives (un-combined, f32) in the
+    split-k scratch; the consumer must read+re-zero it."""
+
+
+DEFERRED = _Deferred()
+
+
+def gemv_defer(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+               group: int, bits: int) -> bool:
+    """Launch the GEMM leaving the f32 split-k result in the scratch
+    (caller checked _will_defer). Returns True."""
+    out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype, device=x.device)
+    d = _native().gemm_m16(x, w, scales, None, out, _get_scratch(x.device),
+                           group, True, bits, True)
+    assert d, "will_defer disagreed with gemm_m16"
+    return True
+
+
+def rmsnorm_f32_scratch(residual: torch.Tensor, wn: torch.Tensor,
+                        eps: float) -> torch.Tensor:
+    """RMSNorm(+residual) whose x input is the deferred f32 scratch."""
+    y = torch.empty_like(residual)
+    _native().rmsnorm_f32(_get_scratch(residual.device), residual, wn, y,
+                          eps)
+    return y
+
+
+def resid_add_scratch(h: torch.Tensor) -> None:
+    _native().resid_add_f32(h, _get_scratch(h.device))
+
+
+def linear_will_defer(x: torch.Tensor, w: torch.Tensor, group: int,
+                      bits: int) -> bool:
+    return _will_defer(x.shape[0], w.shape[0], x.shape[1], group, bits,
+                       _get_scratch(x.device).numel())
+
+
+def gemv_rmsnorm(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
+                 group: int, bits: int, residual: torch.Tensor,
+                 wn: torch.Tensor, eps: float) -> torch.Tensor:
+    """Fused projection GEMM + RMSNorm(+residual) for the decode path:
+    when the GEMM runs split-k, the norm reads (and re-zeroes) the f32
+    scratch directly — the combine kernel disappears from the chain.
+    Updates `residual` in place (residual += proj) like ops.rmsnorm."""
+    scratch = _get_scratch(x.device)
+    if _will_defer(x.shape[0], w.shape[0], x.shape[1], group, bits,
+                   scratch.numel()):
+        out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype,
+                          device=x.device)
+        _native().gemm_m16(x, w, scales, None, out, scratch, group, True,
+                           bits, True)
+        y = torch.empty_like(residual)
+        _native().rmsnorm_f32(scratch, residual, wn, y, eps)
+        return y
+    gemv = gemv_int8 if bits == 8 else gemv_int4
+    o = gemv(x, w, scales, group, None, True)
+    return rmsnorm(o, residual, wn, eps)
+
+
 def gemv_swiglu(x: torch.Tensor, w: torch.Tensor, scales: torch.Tensor,
                 group: int, bits: int = 8,
                 packed: bool = True) -> torch.Tensor:

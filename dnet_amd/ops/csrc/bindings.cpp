@@ -3,6 +3,9 @@
 namespace py = pybind11;
 
 namespace dnet {
+void rmsnorm_f32(torch::Tensor xf, torch::Tensor residual, torch::Tensor w,
+                 torch::Tensor y, double eps);
+void resid_add_f32(torch::Tensor h, torch::Tensor xf);
 void rmsnorm(torch::Tensor x, c10::optional<torch::Tensor> residual,
              torch::Tensor w, torch::Tensor y, double eps);
 void gemv_bf16(torch::Tensor x, torch::Tensor w, torch::Tensor out,
@@ -92,6 +95,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "would gemm_m16(defer_combine=true) defer for this shape");
   m.def("rope_append_f32", &dnet::rope_append_f32,
         "RoPE+append reading+re-zeroing the qkv split-k f32 scratch");
+  m.def("resid_add_f32", &dnet::resid_add_f32,
+        "h += split-k f32 scratch (re-zeroed)");
+  m.def("rmsnorm_f32", &dnet::rmsnorm_f32,
+        "RMSNorm+residual reading+re-zeroing the split-k f32 scratch");
   m.def("swiglu_f32", &dnet::swiglu_f32,
         "SwiGLU reading+re-zeroing the split-k f32 scratch");
   m.def("dequant_int8", &dnet::dequant_int8, "grouped-int8 -> bf16 dequant");

@@ -10,12 +10,16 @@
 
 namespace dnet {
 
-template <bool HAS_RES>
+// F32X: x comes from the UN-COMBINED f32 split-k scratch of the
+// producing GEMM (o/down projection) and is re-zeroed after the read —
+// the f32->bf16 combine kernel disappears from the decode chain.
+template <bool HAS_RES, bool F32X = false>
 __global__ void rmsnorm_kernel(const short* __restrict__ x,
                                short* __restrict__ res,  // in/out residual
                                const short* __restrict__ w,
                                short* __restrict__ y,
-                               const int H, const float eps, const int T) {
+                               const int H, const float eps, const int T,
+                               float* __restrict__ xf = nullptr) {
   __shared__ float scratch[16];
   const int vecs = H / 8;  // H % 8 == 0 enforced on host
   for (int row = blockIdx.x; row < T; row += gridDim.x) {
@@ -23,7 +27,18 @@ __global__ void rmsnorm_kernel(const short* __restrict__ x,
     short8* rv = HAS_RES ? reinterpret_cast<short8*>(res + (int64_t)row * H) : nullptr;
     float ss = 0.f;
     for (int i = threadIdx.x; i < vecs; i += blockDim.x) {
-      short8 v = xv[i];
+      short8 v;
+      if (F32X) {
+        float4* xr = reinterpret_cast<float4*>(xf + (int64_t)row * H + i * 8);
+        const float4 a = xr[0], bq = xr[1];
+        xr[0] = make_float4(0.f, 0.f, 0.f, 0.f);
+        xr[1] = make_float4(0.f, 0.f, 0.f, 0.f);
+        const float av[8] = {a.x, a.y, a.z, a.w, bq.x, bq.y, bq.z, bq.w};
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v.x[j] = f2bits(av[j]);
+      } else {
+        v = xv[i];
+      }
       if (HAS_RES) {
         short8 rr = rv[i];
 #pragma unroll
@@ -55,6 +70,47 @@ __global__ void rmsnorm_kernel(const short* __restrict__ x,
     }
     __syncthreads();
   }
+}
+
+// h += x where x is the f32 split-k scratch (re-zeroed): the final
+// residual update of a decode window whose last down-projection was
+// left deferred in scratch.
+__global__ void resid_add_f32_kernel(short* __restrict__ h,
+                                     float* __restrict__ xf,
+                                     const int64_t total) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const float v = xf[i];
+    xf[i] = 0.f;
+    h[i] = f2bits(bits2f(h[i]) + v);
+  }
+}
+
+void resid_add_f32(torch::Tensor h, torch::Tensor xf) {
+  const int64_t total = h.numel();
+  DNET_CHECK(h.is_contiguous() && xf.is_contiguous(), "contig");
+  DNET_CHECK(xf.numel() >= total, "scratch too small");
+  auto stream = current_stream();
+  const int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(resid_add_f32_kernel, dim3(grid), dim3(256), 0, stream,
+                     (short*)h.data_ptr(), (float*)xf.data_ptr(), total);
+}
+
+// Split-k-fused variant: x read (and re-zeroed) from the f32 scratch.
+void rmsnorm_f32(torch::Tensor xf, torch::Tensor residual, torch::Tensor w,
+                 torch::Tensor y, double eps) {
+  const int64_t H = residual.size(-1);
+  const int64_t T = residual.numel() / H;
+  DNET_CHECK(H % 8 == 0, "H % 8");
+  DNET_CHECK(xf.numel() >= T * H, "scratch too small");
+  DNET_CHECK(residual.is_contiguous() && y.is_contiguous() &&
+                 w.is_contiguous() && xf.is_contiguous(), "contig");
+  auto stream = current_stream();
+  const int grid = (int)std::min<int64_t>(T, 2048);
+  hipLaunchKernelGGL((rmsnorm_kernel<true, true>), dim3(grid), dim3(256), 0,
+                     stream, nullptr, (short*)residual.data_ptr(),
+                     (const short*)w.data_ptr(), (short*)y.data_ptr(),
+                     (int)H, (float)eps, (int)T, (float*)xf.data_ptr());
 }
 
 // y = rmsnorm(x) * w ; if residual is given: residual += x (in place), then
