@@ -37,6 +37,7 @@ __device__ __forceinline__ float wave_allreduce_max(float v) {
 constexpr int kChunk = 64;
 
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using bf16x2 = __attribute__((ext_vector_type(2))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 // splits > 1: flash-decode S-partitioning — gridDim.y splits each take
@@ -71,6 +72,11 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
   short* k_lds = reinterpret_cast<short*>(smem);               // [kChunk][P]
   short* v_lds = k_lds + kChunk * P;                           // [kChunk][PV]
   short* q_lds = v_lds + kChunk * PV;                          // [G][D]
+  float* p_lds = reinterpret_cast<float*>(q_lds + ((G * D + 7) & ~7));
+  // p_lds: [4 waves][kChunk] exp'd scores — LDS broadcast reads replace
+  // the per-iteration __shfl (long dependency chain). (An f32 V tile
+  // was measured: the +17 KB LDS costs a block/CU of occupancy and
+  // LOSES 13% at 2k ctx — V stays bf16.)
 
   const int len = pos[b];
   const int start = (window > 0) ? max(0, len - window) : 0;  // sliding
@@ -157,16 +163,20 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
 
     for (int hi = 0; hi < nh; ++hi) {
       const int g = wid + hi * 4;
-      // Score for position s = lane.
+      // Score for position s = lane (packed bf16 dot2: one VALU per
+      // 2 dims instead of 2 converts + 2 fmas).
       float s_val = 0.f;
       const short* krow = &k_lds[lane * P];
       const short* qrow = &q_lds[g * D];
       for (int d = 0; d < D; d += 4) {
-        const short4v kv = *reinterpret_cast<const short4v*>(&krow[d]);
-        const short4v qv = *reinterpret_cast<const short4v*>(&qrow[d]);  // broadcast
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          s_val = fmaf(bits2f(kv.x[j]), bits2f(qv.x[j]), s_val);
+        const short4v kq = *reinterpret_cast<const short4v*>(&krow[d]);
+        const short4v qq = *reinterpret_cast<const short4v*>(&qrow[d]);
+        s_val = __builtin_amdgcn_fdot2_f32_bf16(
+            *reinterpret_cast<const bf16x2*>(&kq.x[0]),
+            *reinterpret_cast<const bf16x2*>(&qq.x[0]), s_val, false);
+        s_val = __builtin_amdgcn_fdot2_f32_bf16(
+            *reinterpret_cast<const bf16x2*>(&kq.x[2]),
+            *reinterpret_cast<const bf16x2*>(&qq.x[2]), s_val, false);
       }
       s_val *= scale;
       const bool in_win = (lane < valid) && (s0 + lane >= start);
@@ -180,9 +190,13 @@ __global__ void attn_decode_kernel(const short* __restrict__ q,
       m[hi] = mn;
 #pragma unroll
       for (int j = 0; j < DPL; ++j) acc[hi][j] *= alpha;
-      // Accumulate P @ V: lane owns dims d = DPL*lane + j.
+      // Accumulate P @ V: lane owns dims d = DPL*lane + j; p comes
+      // from a per-wave LDS slab (broadcast reads — no serial __shfl
+      // dependency chain)
+      p_lds[wid * kChunk + lane] = p;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // wave-local
       for (int s = 0; s < valid; ++s) {
-        const float ps = __shfl(p, s, 64);
+        const float ps = p_lds[wid * kChunk + s];
         const short* vrow = &v_lds[s * PV + DPL * lane];
 #pragma unroll
         for (int j = 0; j < DPL; ++j)
@@ -572,8 +586,9 @@ void attn_decode(torch::Tensor q, torch::Tensor kcache, torch::Tensor vcache,
                  out.is_contiguous(), "contig");
   auto stream = current_stream();
   const int G = (int)(Hq / Hkv);
-  const size_t lds = (kChunk * ((int)D + 4) + kChunk * ((int)DV + 4) + G * D)
-                     * sizeof(short);
+  const size_t lds = (kChunk * ((int)D + 4) + kChunk * ((int)DV + 4)
+                      + (((int64_t)G * D + 7) & ~7)) * sizeof(short)
+                     + 4 * kChunk * sizeof(float);
   if (splits > 1 || !combine) {
     DNET_CHECK(partials.has_value()
                    && partials->numel() >= B * Hq * splits * (DV + 2),
