@@ -494,7 +494,18 @@ class RingModel:
                     rand_for(99991)(c.vocab_size, c.hidden_size).to(self.device)
                 self.lm_head = Linear(emb)
             else:
-                self.lm_head = Linear(rand_for(99992)(c.vocab_size, c.hidden_size).to(self.device))
+                self.lm_head = self._make_lm_head(
+                    rand_for(99992)(c.vocab_size, c.hidden_size).to(self.device))
+
+    def _make_lm_head(self, w: torch.Tensor) -> Linear:
+        """Untied lm_head follows the model's weight quantization (the
+        reference quantizes every linear incl. the head): the decode-path
+        logits GEMM reads 0.78 GB bf16 per step on a 152k vocab — int8
+        halves it (~415 -> ~260 us/step measured)."""
+        q = self.cfg.quant
+        if q is not None and q.fmt != "mxfp4" and w.shape[1] % 128 == 0:
+            return Linear.make(w, None, q)
+        return Linear(w)
 
     def _init_layer(self, rand, lid: int) -> LayerWeights:
         c = self.cfg
@@ -573,7 +584,8 @@ class RingModel:
                 emb = get("embed_tokens.weight").to(self.device)
                 self.lm_head = Linear(emb)
             else:
-                self.lm_head = Linear(head.to(torch.bfloat16).to(self.device))
+                self.lm_head = self._make_lm_head(
+                    head.to(torch.bfloat16).to(self.device))
 
     # ---------- forward ----------
 
