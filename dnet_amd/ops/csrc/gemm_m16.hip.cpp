@@ -634,20 +634,21 @@ __global__ __launch_bounds__(256, MINW) void gemm_m16_stream_kernel(
   }
 }
 
-__global__ void f32_to_bf16_bias_kernel(float* __restrict__ in,
+__global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
                                         const short* __restrict__ bias,
                                         short* __restrict__ out,
                                         const int64_t total, const int N) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
     float v = in[i];
-    in[i] = 0.f;   // re-zero for the next split-k use: the combine pass
-                   // replaces a per-launch hipMemsetAsync of the scratch
-                   // (~3 launches + full passes per layer per step)
     if (bias != nullptr) v += bits2f(bias[i % N]);
     out[i] = f2bits(v);
   }
 }
+
+// One device per process (shard design): tracks whether the split-k
+// scratch holds un-zeroed partials from a plain (non-deferred) combine.
+static bool g_scratch_dirty = false;
 
 static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
                        int min_pairs, int g_align) {
@@ -702,10 +703,21 @@ static bool launch_m16(torch::Tensor x, torch::Tensor w,
   short* op = (short*)out.data_ptr() + m0 * N;
   float* fp = nullptr;
   if (sk > 1) {
-    // scratch is allocated zeroed and re-zeroed by the combine kernel
     TORCH_CHECK(scratch.has_value() && scratch->numel() >= (int64_t)M * N,
                 "split-k scratch required");
     fp = (float*)scratch->data_ptr();
+    // The scratch must be zero before the atomics. Deferred-combine
+    // consumers (rope/swiglu/rmsnorm f32 variants) re-zero what they
+    // read, so a chain of deferred GEMMs never needs a memset; the
+    // plain combine does NOT re-zero — a combine-side in[i]=0 write
+    // measured a 3.5x MoE decode slowdown (the grouped expert kernels
+    // dropped ~3.9 -> ~2.5 TB/s whenever the combine left dirtied
+    // scratch lines behind; bisected to that single change) — so the
+    // NEXT split-k launch after a combine memsets first.
+    if (g_scratch_dirty) {
+      DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * M * N, stream));
+      g_scratch_dirty = false;
+    }
   }
   const short* sp = bits < 16 ? (const short*)scales->data_ptr() : nullptr;
   const short* bp1 = sk > 1 ? nullptr : bptr;
@@ -783,6 +795,7 @@ static bool launch_m16(torch::Tensor x, torch::Tensor w,
     const int cgrid = (int)std::min<int64_t>((total + 255) / 256, 2048);
     hipLaunchKernelGGL(f32_to_bf16_bias_kernel, dim3(cgrid), dim3(256), 0,
                        stream, fp, bptr, op, total, (int)N);
+    g_scratch_dirty = true;   // combine leaves partials in the scratch
   }
   return false;
 }
