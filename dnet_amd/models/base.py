@@ -691,7 +691,11 @@ class RingModel:
             wpos = ((kv.pos - self.cp_rank * self.cp_cap).int()
                     if self.cp_size > 1 else None)
             if (y.is_cuda and ql.is_quant and not ql.mxfp4 and ql.packed
-                    and not c.qk_norm and 2 < B <= 64):
+                    and not c.qk_norm and 2 < B <= 64
+                    and self.cp_size == 1):
+                # (CP excluded: the wpos-offset variant of the fused
+                # kernel has no single-GPU test path — the unfused
+                # chain is the gloo-verified one)
                 # decode fast path: qkv split-k combine fused into
                 # RoPE+append (bias applied there, k/v go straight to
                 # the cache)
