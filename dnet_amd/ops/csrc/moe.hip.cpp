@@ -64,8 +64,11 @@ __device__ __forceinline__ float apply_glu(float g, float u, const float alpha,
 }
 
 // QMODE: 0 = bf16, 1 = grouped int8, 2 = mxfp4 (nibble rows + e8m0)
+// launch bounds: M<=8 keeps the tight 128-VGPR budget (measured faster
+// at high occupancy); the wide tiles trade occupancy for bank-read
+// amortization and need the bigger register file.
 template <int M, int QMODE, bool PACKED, int GLU>
-__global__ void moe_gateup_kernel(
+__global__ __launch_bounds__(256, M <= 8 ? 4 : 2) void moe_gateup_kernel(
     const short* __restrict__ x, const void* __restrict__ w,
     const short* __restrict__ scales, const short* __restrict__ bias,
     const float* __restrict__ we, short* __restrict__ act, const int K,
@@ -202,7 +205,7 @@ __global__ void moe_gateup_kernel(
 }
 
 template <int M, int QMODE, bool PACKED>
-__global__ void moe_down_kernel(
+__global__ __launch_bounds__(256, 4) void moe_down_kernel(
     const short* __restrict__ act, const void* __restrict__ w,
     const short* __restrict__ scales, const short* __restrict__ bias,
     const float* __restrict__ we, float* __restrict__ out, const int I,
@@ -314,15 +317,31 @@ static void moe_dispatch_m(int M, LaunchFn&& fn) {
     case 4: fn(std::integral_constant<int, 4>{}); break;
     case 6: fn(std::integral_constant<int, 6>{}); break;
     case 8: fn(std::integral_constant<int, 8>{}); break;
+    case 16: fn(std::integral_constant<int, 16>{}); break;
+    case 32: fn(std::integral_constant<int, 32>{}); break;
     default: TORCH_CHECK(false, "moe: unsupported M tile ", M);
   }
 }
 
-static int moe_mtile(int64_t rem, bool mx4 = false) {
+static int moe_mtile(int64_t rem, bool mx4, int cap) {
+  // Wider M tiles amortize the expert-bank read (at batch 32, M=8
+  // re-reads every routed bank 4x per layer) but MEASURED SLOWER:
+  // gpt-oss b32 gu8/dn8 2434 tok/s vs dn32 2386 / gu16 2141 — the
+  // occupancy cost (VGPR 2M accumulators) outweighs the traffic cut;
+  // these kernels are latency-bound. M=16/32 tiers kept behind
+  // DNET_MOE_MT_GU / DNET_MOE_MT_DN; the structural fix is an
+  // LDS-staged-x MFMA expert GEMM (round-3 lever).
   if (mx4 && rem >= 6) return 6;   // M=8 mxfp4 spills ~30 regs
+  if (cap >= 32 && rem >= 32) return 32;
+  if (cap >= 16 && rem >= 16) return 16;
   if (rem >= 8) return 8;
   if (rem == 7 || rem == 5) return 4;
   return (int)rem;
+}
+
+static int moe_cap(const char* env, int dflt) {
+  const char* e = getenv(env);
+  return e ? atoi(e) : dflt;
 }
 
 void moe_gateup(torch::Tensor x, torch::Tensor w,
@@ -360,9 +379,10 @@ void moe_gateup(torch::Tensor x, torch::Tensor w,
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   auto stream = current_stream();
   const dim3 grid(cdiv((int)I, kMoeWaves), 1, (unsigned)E);
+  static const int gu_cap = moe_cap("DNET_MOE_MT_GU", 8);
   int64_t m0 = 0;
   while (m0 < M) {
-    const int mt = moe_mtile(M - m0, mx4);
+    const int mt = moe_mtile(M - m0, mx4, gu_cap);
     moe_dispatch_m(mt, [&](auto mc) {
       constexpr int MV = decltype(mc)::value;
       auto launch = [&](auto qmc, auto pkc, auto gluc) {
@@ -430,9 +450,10 @@ void moe_down(torch::Tensor act, torch::Tensor w,
   const short* bptr = bias.has_value() ? (const short*)bias->data_ptr() : nullptr;
   auto stream = current_stream();
   const dim3 grid(cdiv((int)H, kMoeWaves), 1, (unsigned)E);
+  static const int dn_cap = moe_cap("DNET_MOE_MT_DN", 8);
   int64_t m0 = 0;
   while (m0 < M) {
-    const int mt = moe_mtile(M - m0, mx4);
+    const int mt = moe_mtile(M - m0, mx4, dn_cap);
     moe_dispatch_m(mt, [&](auto mc) {
       constexpr int MV = decltype(mc)::value;
       auto launch = [&](auto qmc, auto pkc) {
