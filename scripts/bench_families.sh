@@ -18,3 +18,4 @@ run gpt-oss-mxfp4-b32   --model gpt-oss-20b --quant mxfp4 --mb-size 32 --steps 6
 run mixtral-int8-b32    --model mixtral-8x7b --quant int8 --mb-size 32 --steps 6 --warmup 2
 run deepseek-bf16-b16   --model deepseek-v2-lite --quant bf16 --mb-size 16 --steps 6 --warmup 2
 run llama8b-bf16-b64    --model llama-3-8b --quant bf16 --steps 8 --warmup 3
+run qwen3-8b-int8-b64   --model qwen3-8b --quant int8 --steps 8 --warmup 3
