@@ -20,7 +20,16 @@ import logging
 import os
 import time
 
-import dnet_amd  # noqa: F401  (sets the TunableOp env BEFORE torch loads)
+import sys as _sys
+
+import dnet_amd
+
+# TunableOp algo cache covers the HEADLINE model's prefill shapes only;
+# enabling it globally taxes eager-decode models (see enable_tunableop_cache)
+_mi = _sys.argv.index("--model") + 1 if "--model" in _sys.argv else 0
+if _mi == 0 or (_mi < len(_sys.argv)
+                and _sys.argv[_mi].startswith("qwen-2.5-32b")):
+    dnet_amd.enable_tunableop_cache()
 import torch
 
 logging.basicConfig(level=logging.INFO)

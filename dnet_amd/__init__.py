@@ -1,16 +1,26 @@
 """dnet_amd: MI355X-native distributed LLM inference engine."""
-import os as _os
 
-# Pre-tuned hipBLASLt/rocBLAS algorithm cache for the prefill GEMM shapes
-# (PyTorch TunableOp): measured TTFT 523 -> 475 ms on the qwen-32b
-# 64x128 batched prefill. TUNING=0 means cached shapes use the tuned
-# algorithm and uncached shapes fall back to the normal heuristics —
-# no runtime tuning cost ever. Opt out by setting
-# PYTORCH_TUNABLEOP_ENABLED yourself before import.
-_tun = _os.path.join(_os.path.dirname(_os.path.abspath(__file__)), "..",
-                     "profiles", "gfx950_tunableop.csv")
-_tun0 = _tun.replace(".csv", "0.csv")
-if _os.path.exists(_tun0) and "PYTORCH_TUNABLEOP_ENABLED" not in _os.environ:
-    _os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
-    _os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
-    _os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _tun
+
+def enable_tunableop_cache() -> bool:
+    """Point PyTorch TunableOp at the committed gfx950 algorithm cache
+    (profiles/gfx950_tunableop0.csv — prefill GEMM shapes of the
+    qwen-32b headline bench; TTFT 523 -> 476 ms measured). MUST be
+    called BEFORE torch is first imported.
+
+    NOT enabled automatically: TunableOp wraps every GEMM dispatch and
+    measured a 17% decode tax on eager-path models whose shapes are not
+    in the cache (deepseek-v2-lite 1493 -> 1239 tok/s) — so only
+    bench.py enables it, and only for the headline model. Returns True
+    if the cache file exists and the env was set."""
+    import os
+    import sys
+    if "torch" in sys.modules:  # too late — torch reads the env at load
+        return False
+    tun = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..",
+                       "profiles", "gfx950_tunableop.csv")
+    if not os.path.exists(tun.replace(".csv", "0.csv")):
+        return False
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", tun)
+    return True
