@@ -455,8 +455,9 @@ def test_mla_latent_cache_matches_perhead():
                          generator=torch.Generator().manual_seed(9))
 
     def run(perhead):
-        if perhead:
-            os.environ["DNET_MLA_PERHEAD"] = "1"
+        # explicit: the default is context-aware (latent only for
+        # smax >= 2048), so force each mode
+        os.environ["DNET_MLA_PERHEAD"] = "1" if perhead else "0"
         try:
             ex = RingExecutor(cfg, 0, 1, "cpu", mb_count=1, mb_size=2,
                               smax=32, seed=21, use_graphs=False)
