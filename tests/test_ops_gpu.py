@@ -749,3 +749,58 @@ def test_fused_qkv_rope_matches_separate(kvq):
                           rtol=3e-2)
     assert torch.allclose(dq(vc1, vs1).cpu(), dq(vc2, vs2).cpu(), atol=4e-2,
                           rtol=3e-2)
+
+
+def test_deferred_combine_consumers_match_plain():
+    """The split-k deferred-combine consumers (gemv_swiglu reading the
+    f32 scratch, gemv_rmsnorm, and the resid-add) must match the plain
+    GEMM -> combine -> op chain, and must leave the scratch re-zeroed
+    for the next split-k user."""
+    from dnet_amd.ops import _get_scratch, _will_defer, _native
+    torch.manual_seed(11)
+    dev = _dev()
+    M, K, I = 16, 1024, 512
+    N = 2 * I
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    wf = torch.randn(N, K, dtype=torch.bfloat16, device=dev) / 30
+    q8, sc = ops.quantize_int8(wf, 128)
+    qp = ops.pack_int8_mfma(q8)
+    if not _will_defer(M, N, K, 128, 8, _get_scratch(dev).numel()):
+        pytest.skip("shape does not split-k")
+
+    # plain chain
+    gu = ops.gemv_int8(x, qp, sc, 128, None, packed=True)
+    y_ref = ops.swiglu(gu)
+    # fused chain
+    y_fused = ops.gemv_swiglu(x, qp, sc, 128, 8)
+    assert torch.allclose(y_fused.float().cpu(), y_ref.float().cpu(),
+                          atol=4e-2, rtol=3e-2)
+
+    # rmsnorm variant: residual updated in place + y match
+    wproj = torch.randn(K, K, dtype=torch.bfloat16, device=dev) / 30
+    pq, psc = ops.quantize_int8(wproj, 128)
+    pqp = ops.pack_int8_mfma(pq)
+    wn = torch.randn(K, dtype=torch.bfloat16, device=dev)
+    h1 = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    h2 = h1.clone()
+    o = ops.gemv_int8(x, pqp, psc, 128, None, packed=True)
+    y1 = ops.rmsnorm(o, h1, wn, 1e-6)
+    y2 = ops.gemv_rmsnorm(x, pqp, psc, 128, 8, h2, wn, 1e-6)
+    assert torch.allclose(y2.float().cpu(), y1.float().cpu(), atol=4e-2,
+                          rtol=3e-2)
+    assert torch.allclose(h2.float().cpu(), h1.float().cpu(), atol=4e-2,
+                          rtol=3e-2)
+
+    # resid-add variant: h += deferred projection
+    h3, h4 = h1.clone(), h1.clone()
+    d_ref = ops.gemv_int8(x, pqp, psc, 128, None, packed=True)
+    h3.add_(d_ref)
+    ok = ops.gemv_defer(x, pqp, psc, 128, 8)
+    assert ok
+    ops.resid_add_scratch(h4)
+    assert torch.allclose(h4.float().cpu(), h3.float().cpu(), atol=4e-2,
+                          rtol=3e-2)
+
+    # all consumers must have re-zeroed what they read
+    scr = _get_scratch(dev)
+    assert int((scr != 0).sum()) == 0, "scratch not re-zeroed by consumers"
