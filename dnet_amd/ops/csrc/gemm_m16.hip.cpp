@@ -646,9 +646,12 @@ __global__ void f32_to_bf16_bias_kernel(const float* __restrict__ in,
   }
 }
 
-// One device per process (shard design): tracks whether the split-k
-// scratch holds un-zeroed partials from a plain (non-deferred) combine.
-static bool g_scratch_dirty = false;
+// One device per process (shard design): number of leading scratch
+// elements holding un-zeroed partials from plain (non-deferred)
+// combines. A bool is NOT enough: a small GEMM's memset after a big
+// combine would clear the flag while stale partials remain beyond its
+// own M*N region and corrupt the next bigger user.
+static int64_t g_scratch_dirty_elems = 0;
 
 static int pick_splitk(int64_t N, int64_t K, int64_t M, int64_t scratch_elems,
                        int min_pairs, int g_align) {
@@ -714,9 +717,11 @@ static bool launch_m16(torch::Tensor x, torch::Tensor w,
     // dropped ~3.9 -> ~2.5 TB/s whenever the combine left dirtied
     // scratch lines behind; bisected to that single change) — so the
     // NEXT split-k launch after a combine memsets first.
-    if (g_scratch_dirty) {
-      DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * M * N, stream));
-      g_scratch_dirty = false;
+    if (g_scratch_dirty_elems > 0) {
+      const int64_t span = std::max<int64_t>(g_scratch_dirty_elems,
+                                             (int64_t)M * N);
+      DNET_CHECK_HIP(hipMemsetAsync(fp, 0, sizeof(float) * span, stream));
+      g_scratch_dirty_elems = 0;
     }
   }
   const short* sp = bits < 16 ? (const short*)scales->data_ptr() : nullptr;
@@ -795,7 +800,8 @@ static bool launch_m16(torch::Tensor x, torch::Tensor w,
     const int cgrid = (int)std::min<int64_t>((total + 255) / 256, 2048);
     hipLaunchKernelGGL(f32_to_bf16_bias_kernel, dim3(cgrid), dim3(256), 0,
                        stream, fp, bptr, op, total, (int)N);
-    g_scratch_dirty = true;   // combine leaves partials in the scratch
+    g_scratch_dirty_elems =
+        std::max<int64_t>(g_scratch_dirty_elems, (int64_t)M * N);
   }
   return false;
 }

@@ -768,13 +768,22 @@ def test_deferred_combine_consumers_match_plain():
     if not _will_defer(M, N, K, 128, 8, _get_scratch(dev).numel()):
         pytest.skip("shape does not split-k")
 
+    # suite-order hazard: leave a LARGE stale combine region first (this
+    # is exactly the case that broke the boolean dirty flag — a small
+    # memset cleared it while stale partials remained beyond M*N)
+    xb = torch.randn(64, K, dtype=torch.bfloat16, device=dev)
+    wb = torch.randn(4096, K, dtype=torch.bfloat16, device=dev) / 30
+    qb, sb = ops.quantize_int8(wb, 128)
+    ops.gemv_int8(xb, ops.pack_int8_mfma(qb), sb, 128, None, packed=True)
+
     # plain chain
     gu = ops.gemv_int8(x, qp, sc, 128, None, packed=True)
     y_ref = ops.swiglu(gu)
-    # fused chain
+    # fused chain (reads f32 pre-combine values — y_ref went through a
+    # bf16 rounding at the combine, so tolerance covers one bf16 ulp)
     y_fused = ops.gemv_swiglu(x, qp, sc, 128, 8)
     assert torch.allclose(y_fused.float().cpu(), y_ref.float().cpu(),
-                          atol=4e-2, rtol=3e-2)
+                          atol=8e-2, rtol=4e-2)
 
     # rmsnorm variant: residual updated in place + y match
     wproj = torch.randn(K, K, dtype=torch.bfloat16, device=dev) / 30
