@@ -26,9 +26,17 @@ import dnet_amd
 
 # TunableOp algo cache covers the HEADLINE model's prefill shapes only;
 # enabling it globally taxes eager-decode models (see enable_tunableop_cache)
-_mi = _sys.argv.index("--model") + 1 if "--model" in _sys.argv else 0
-if _mi == 0 or (_mi < len(_sys.argv)
-                and _sys.argv[_mi].startswith("qwen-2.5-32b")):
+def _bench_model_arg() -> str:
+    av = _sys.argv
+    for i, a in enumerate(av):
+        if a == "--model" and i + 1 < len(av):
+            return av[i + 1]
+        if a.startswith("--model="):
+            return a.split("=", 1)[1]
+    return "qwen-2.5-32b"
+
+
+if _bench_model_arg().startswith("qwen-2.5-32b"):
     dnet_amd.enable_tunableop_cache()
 import torch
 
