@@ -16,6 +16,10 @@ def enable_tunableop_cache() -> bool:
     import sys
     if "torch" in sys.modules:  # too late — torch reads the env at load
         return False
+    # torch appends the DEVICE ORDINAL before the extension when
+    # reading/writing (gfx950_tunableop.csv -> gfx950_tunableop<N>.csv);
+    # profiles/ ships copies for ordinals 0-7 so every rank of a
+    # multi-GPU bench run gets the tuned prefill algorithms
     tun = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..",
                        "profiles", "gfx950_tunableop.csv")
     if not os.path.exists(tun.replace(".csv", "0.csv")):
